@@ -1,0 +1,197 @@
+"""Pre-forked sandbox runner ("zygote") for the MI355X executor.
+
+Started once per executor-server (server.cpp). It:
+
+1. pre-imports the heavy sandbox runtime (numpy, the import hooks) so every
+   forked child gets them copy-on-write instead of paying a cold import;
+2. immediately pre-forks ONE warm child that additionally initializes the
+   HIP runtime (_hipops: device context + pinned staging buffers) while the
+   executor is still sitting in the warm pool -- so GPU start-up cost is
+   paid before any request arrives;
+3. on a "run" job from the server, hands it to the warm child (or forks a
+   fresh one), reports {start,pid} and later {exit,code}.
+
+This replaces the reference's per-request `upm guess` + `pip install` +
+`xonsh` cold start (reference executor/server.rs:120-169) with a fork-based
+warm path; dependency auto-install still happens (sandbox_runtime.depscan)
+but only costs an AST scan when everything is already importable.
+
+Protocol (newline-delimited JSON over --fd):
+  server -> zygote: {"event":"run","id":N,"script":...,"cwd":...,
+                     "stdout":...,"stderr":...,"env":{...}}
+  zygote -> server: {"event":"warm"}                 (warm child is ready)
+                    {"event":"start","id":N,"pid":P}
+                    {"event":"exit","id":N,"code":C}
+"""
+
+import argparse
+import json
+import os
+import select
+import signal
+import socket
+import sys
+import time
+
+RUNTIME_DIR = os.path.dirname(os.path.abspath(__file__))
+if RUNTIME_DIR not in sys.path:
+    sys.path.insert(0, RUNTIME_DIR)
+
+import sandbox_runtime  # noqa: E402
+
+
+def _run_job_in_child(job: dict) -> None:
+    """Executed in the forked child. Never returns."""
+    exit_code = 0
+    try:
+        os.setsid()
+        stdin_fd = os.open("/dev/null", os.O_RDONLY)
+        stdout_fd = os.open(job["stdout"], os.O_WRONLY | os.O_CREAT | os.O_TRUNC)
+        stderr_fd = os.open(job["stderr"], os.O_WRONLY | os.O_CREAT | os.O_TRUNC)
+        os.dup2(stdin_fd, 0)
+        os.dup2(stdout_fd, 1)
+        os.dup2(stderr_fd, 2)
+        # line-buffered text layer over the new fds
+        sys.stdout = os.fdopen(1, "w", buffering=1, closefd=False)
+        sys.stderr = os.fdopen(2, "w", buffering=1, closefd=False)
+        os.environ.update(job.get("env") or {})
+        os.chdir(job["cwd"])
+        exit_code = sandbox_runtime.run_user_script(job["script"])
+    except SystemExit as e:
+        exit_code = e.code if isinstance(e.code, int) else (0 if e.code is None else 1)
+    except BaseException:
+        import traceback
+
+        traceback.print_exc()
+        exit_code = 1
+    finally:
+        try:
+            sys.stdout.flush()
+            sys.stderr.flush()
+        except Exception:
+            pass
+        os._exit(exit_code & 0xFF if exit_code >= 0 else 1)
+
+
+class WarmChild:
+    """A pre-forked child that finishes expensive init (HIP context, pinned
+    buffers) before any request, then blocks waiting for one job."""
+
+    def __init__(self):
+        parent_sock, child_sock = socket.socketpair()
+        pid = os.fork()
+        if pid == 0:
+            parent_sock.close()
+            try:
+                sandbox_runtime.prewarm()  # HIP init happens HERE, post-fork
+                child_sock.sendall(b"ready\n")
+                data = b""
+                while not data.endswith(b"\n"):
+                    chunk = child_sock.recv(65536)
+                    if not chunk:
+                        os._exit(0)  # zygote dropped us
+                    data += chunk
+                job = json.loads(data)
+                child_sock.close()
+                _run_job_in_child(job)  # never returns
+            except BaseException:
+                os._exit(1)
+        child_sock.close()
+        self.pid = pid
+        self.sock = parent_sock
+        self.ready = False
+
+    def poll_ready(self) -> bool:
+        if self.ready:
+            return True
+        r, _, _ = select.select([self.sock], [], [], 0)
+        if r:
+            data = self.sock.recv(16)
+            if data:
+                self.ready = True
+        return self.ready
+
+    def submit(self, job: dict) -> None:
+        self.sock.sendall(json.dumps(job).encode() + b"\n")
+        self.sock.close()
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--fd", type=int, required=True)
+    args = parser.parse_args()
+
+    signal.signal(signal.SIGCHLD, signal.SIG_DFL)
+
+    server = socket.socket(fileno=args.fd)
+    server.setblocking(True)
+
+    def send(msg: dict) -> None:
+        try:
+            server.sendall(json.dumps(msg).encode() + b"\n")
+        except OSError:
+            os._exit(0)
+
+    warm = WarmChild()
+    announced_warm = False
+    running: dict[int, int] = {}  # pid -> job id
+    buf = b""
+
+    while True:
+        if not announced_warm and warm is not None and warm.poll_ready():
+            announced_warm = True
+            send({"event": "warm"})
+
+        # reap finished children
+        while running:
+            try:
+                pid, status = os.waitpid(-1, os.WNOHANG)
+            except ChildProcessError:
+                break
+            if pid == 0:
+                break
+            job_id = running.pop(pid, None)
+            if job_id is not None:
+                code = os.WEXITSTATUS(status) if os.WIFEXITED(status) else -1
+                send({"event": "exit", "id": job_id, "code": code})
+
+        r, _, _ = select.select([server], [], [], 0.02)
+        if not r:
+            continue
+        chunk = server.recv(65536)
+        if not chunk:
+            # server went away: kill children and exit
+            for pid in running:
+                try:
+                    os.killpg(pid, signal.SIGKILL)
+                except OSError:
+                    pass
+            os._exit(0)
+        buf += chunk
+        while b"\n" in buf:
+            line, buf = buf.split(b"\n", 1)
+            if not line.strip():
+                continue
+            try:
+                msg = json.loads(line)
+            except json.JSONDecodeError:
+                continue
+            if msg.get("event") != "run":
+                continue
+            job_id = msg["id"]
+            if warm is not None and warm.poll_ready():
+                warm.submit(msg)
+                running[warm.pid] = job_id
+                send({"event": "start", "id": job_id, "pid": warm.pid})
+                warm = None  # single-use (fresh-sandbox semantics)
+            else:
+                pid = os.fork()
+                if pid == 0:
+                    server.close()
+                    _run_job_in_child(msg)  # never returns
+                running[pid] = job_id
+                send({"event": "start", "id": job_id, "pid": pid})
+
+
+if __name__ == "__main__":
+    main()

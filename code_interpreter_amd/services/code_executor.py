@@ -1,0 +1,115 @@
+"""Shared executor-side types and the client for the in-sandbox HTTP API.
+
+The internal sandbox API (served by executor/server.cpp, wire-compatible
+with the reference's in-pod API, server.rs:189-191):
+
+  PUT  /workspace/{rel}   upload a file
+  GET  /workspace/{rel}   download a file
+  POST /execute           {source_code, timeout?, env?} ->
+                          {stdout, stderr, exit_code, files: [abs paths]}
+
+Both executor backends (local process pool, kubernetes pods) drive a
+sandbox through SandboxClient.run(): parallel-upload input files from
+storage, execute, parallel-download changed files back into storage
+(reference orchestration shape: kubernetes_code_executor.py:81-149).
+"""
+
+import asyncio
+from dataclasses import dataclass, field
+from typing import Mapping, Optional
+
+import httpx
+
+from code_interpreter_amd.services.storage import Storage
+
+WORKSPACE_PREFIX = "/workspace/"
+
+
+@dataclass
+class Result:
+    stdout: str
+    stderr: str
+    exit_code: int
+    files: Mapping[str, str] = field(default_factory=dict)
+
+
+class ExecutorError(RuntimeError):
+    """Sandbox-level failure (unreachable, bad response, ...). Retryable."""
+
+
+class SandboxClient:
+    """HTTP client for one sandbox (TCP base_url or unix socket)."""
+
+    def __init__(
+        self,
+        base_url: str = "http://executor",
+        uds: Optional[str] = None,
+        timeout: float = 60.0,
+    ):
+        transport = httpx.AsyncHTTPTransport(uds=uds) if uds else None
+        self._client = httpx.AsyncClient(
+            base_url=base_url, transport=transport, timeout=timeout
+        )
+
+    async def aclose(self) -> None:
+        await self._client.aclose()
+
+    async def healthy(self) -> Optional[dict]:
+        try:
+            resp = await self._client.get("/healthz", timeout=2.0)
+            if resp.status_code == 200:
+                return resp.json()
+        except (httpx.HTTPError, OSError):
+            pass
+        return None
+
+    async def run(
+        self,
+        storage: Storage,
+        source_code: str,
+        files: Mapping[str, str] = {},
+        env: Mapping[str, str] = {},
+        timeout: Optional[float] = None,
+    ) -> Result:
+        try:
+            return await self._run(storage, source_code, files, env, timeout)
+        except (httpx.HTTPError, OSError) as e:
+            raise ExecutorError(f"sandbox request failed: {e!r}") from e
+
+    async def _run(self, storage, source_code, files, env, timeout) -> Result:
+        async def upload(path: str, object_hash: str):
+            rel = path[len(WORKSPACE_PREFIX):] if path.startswith(WORKSPACE_PREFIX) else path.lstrip("/")
+            async with storage.reader(object_hash) as reader:
+                data = await reader.read()
+            resp = await self._client.put(f"/workspace/{rel}", content=data)
+            if resp.status_code not in (200, 204):
+                raise ExecutorError(f"upload of {path} failed: {resp.status_code}")
+
+        await asyncio.gather(*(upload(p, h) for p, h in files.items()))
+
+        body: dict = {"source_code": source_code, "env": dict(env)}
+        if timeout is not None:
+            body["timeout"] = timeout
+        resp = await self._client.post("/execute", json=body)
+        if resp.status_code != 200:
+            raise ExecutorError(f"execute failed: {resp.status_code} {resp.text!r}")
+        payload = resp.json()
+
+        async def download(path: str):
+            rel = path[len(WORKSPACE_PREFIX):] if path.startswith(WORKSPACE_PREFIX) else path.lstrip("/")
+            async with storage.writer() as writer:
+                async with self._client.stream("GET", f"/workspace/{rel}") as file_resp:
+                    file_resp.raise_for_status()
+                    async for chunk in file_resp.aiter_bytes():
+                        await writer.write(chunk)
+                return path, writer.hash
+
+        stored = dict(
+            await asyncio.gather(*(download(p) for p in payload["files"]))
+        )
+        return Result(
+            stdout=payload["stdout"],
+            stderr=payload["stderr"],
+            exit_code=payload["exit_code"],
+            files=stored,
+        )

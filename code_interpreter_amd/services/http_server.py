@@ -1,0 +1,153 @@
+"""HTTP API frontend.
+
+Wire-contract parity with the reference (http_server.py:36-162; see
+SURVEY.md section 1 for the byte-level contracts):
+
+  POST /v1/execute             {source_code, files?, env?} ->
+                               {stdout, stderr, exit_code, files}
+                               errors -> 500 {detail}
+  POST /v1/parse-custom-tool   {tool_source_code} ->
+                               {tool_name, tool_input_schema_json,
+                                tool_description}
+                               parse errors -> 400 {error_messages}
+  POST /v1/execute-custom-tool {tool_source_code, tool_input_json, env?} ->
+                               {tool_output_json}
+                               tool errors -> 400 {stderr}
+"""
+
+import json
+import logging
+import uuid
+from contextvars import ContextVar
+from typing import Dict, List
+
+from fastapi import Depends, FastAPI, HTTPException, status
+from fastapi.responses import JSONResponse
+from pydantic import BaseModel
+
+from code_interpreter_amd.services.custom_tool_executor import (
+    CustomToolExecuteError,
+    CustomToolExecutor,
+    CustomToolParseError,
+)
+from code_interpreter_amd.utils.validation import AbsolutePath, Hash
+
+logger = logging.getLogger("code_interpreter_service")
+
+
+class ExecuteRequest(BaseModel):
+    source_code: str
+    files: Dict[AbsolutePath, Hash] = {}
+    env: Dict[str, str] = {}
+
+
+class ExecuteResponse(BaseModel):
+    stdout: str
+    stderr: str
+    exit_code: int
+    files: Dict[AbsolutePath, Hash]
+
+
+class ParseCustomToolRequest(BaseModel):
+    tool_source_code: str
+
+
+class ParseCustomToolResponse(BaseModel):
+    tool_name: str
+    tool_input_schema_json: str
+    tool_description: str
+
+
+class ParseCustomToolErrorResponse(BaseModel):
+    error_messages: List[str]
+
+
+class ExecuteCustomToolRequest(BaseModel):
+    tool_source_code: str
+    tool_input_json: str
+    env: Dict[str, str] = {}
+
+
+class ExecuteCustomToolResponse(BaseModel):
+    tool_output_json: str
+
+
+class ExecuteCustomToolErrorResponse(BaseModel):
+    stderr: str
+
+
+def create_http_server(
+    code_executor,
+    custom_tool_executor: CustomToolExecutor,
+    request_id_context_var: ContextVar,
+) -> FastAPI:
+    app = FastAPI(title="code-interpreter-amd")
+
+    def set_request_id() -> str:
+        request_id = str(uuid.uuid4())
+        request_id_context_var.set(request_id)
+        return request_id
+
+    @app.post("/v1/execute", response_model=ExecuteResponse)
+    async def execute(request: ExecuteRequest, request_id: str = Depends(set_request_id)):
+        logger.info(
+            "Executing code with files %s: %s", request.files, request.source_code
+        )
+        try:
+            result = await code_executor.execute(
+                source_code=request.source_code,
+                files=request.files,
+                env=request.env,
+            )
+        except Exception as e:
+            logger.exception("Error executing code")
+            raise HTTPException(status_code=500, detail=str(e))
+        logger.info("Code execution completed with result %s", result)
+        return ExecuteResponse(
+            stdout=result.stdout,
+            stderr=result.stderr,
+            exit_code=result.exit_code,
+            files=dict(result.files),
+        )
+
+    @app.post("/v1/parse-custom-tool", response_model=ParseCustomToolResponse)
+    async def parse_custom_tool(
+        request: ParseCustomToolRequest, request_id: str = Depends(set_request_id)
+    ):
+        logger.info("Parsing custom tool")
+        tool = custom_tool_executor.parse(tool_source_code=request.tool_source_code)
+        return ParseCustomToolResponse(
+            tool_name=tool.name,
+            tool_input_schema_json=json.dumps(tool.input_schema),
+            tool_description=tool.description,
+        )
+
+    @app.exception_handler(CustomToolParseError)
+    async def parse_error_handler(request, e: CustomToolParseError):
+        logger.warning("Invalid custom tool: %s", e.errors)
+        return JSONResponse(
+            status_code=status.HTTP_400_BAD_REQUEST,
+            content=ParseCustomToolErrorResponse(error_messages=e.errors).model_dump(),
+        )
+
+    @app.post("/v1/execute-custom-tool", response_model=ExecuteCustomToolResponse)
+    async def execute_custom_tool(
+        request: ExecuteCustomToolRequest, request_id: str = Depends(set_request_id)
+    ):
+        logger.info("Executing custom tool")
+        result = await custom_tool_executor.execute(
+            tool_source_code=request.tool_source_code,
+            tool_input_json=request.tool_input_json,
+            env=request.env,
+        )
+        return ExecuteCustomToolResponse(tool_output_json=json.dumps(result))
+
+    @app.exception_handler(CustomToolExecuteError)
+    async def execute_error_handler(request, e: CustomToolExecuteError):
+        logger.warning("Error executing custom tool: %s", e)
+        return JSONResponse(
+            status_code=status.HTTP_400_BAD_REQUEST,
+            content=ExecuteCustomToolErrorResponse(stderr=e.stderr).model_dump(),
+        )
+
+    return app
