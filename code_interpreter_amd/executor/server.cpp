@@ -419,6 +419,51 @@ static Zygote* g_zygote = nullptr;
 static std::atomic<uint64_t> g_job_id{1};
 
 // ---------------------------------------------------------------------------
+// sandbox sessions: one fresh workspace per execution request, served by one
+// long-lived engine process (the per-request isolation unit is the forked
+// single-use interpreter, not the server)
+// ---------------------------------------------------------------------------
+struct SandboxRegistry {
+  std::mutex mu;
+  std::map<std::string, std::string> workspaces;  // id -> dir
+};
+static SandboxRegistry g_sandboxes;
+static std::string g_sessions_root;
+
+static std::string random_hex(int nbytes) {
+  static const char* hexd = "0123456789abcdef";
+  unsigned char buf[32];
+  FILE* f = fopen("/dev/urandom", "rb");
+  size_t got = f ? fread(buf, 1, (size_t)nbytes, f) : 0;
+  if (f) fclose(f);
+  std::string out;
+  for (size_t i = 0; i < got; i++) {
+    out += hexd[buf[i] >> 4];
+    out += hexd[buf[i] & 15];
+  }
+  return out;
+}
+
+static void rmtree(const std::string& path) {
+  DIR* d = opendir(path.c_str());
+  if (d) {
+    struct dirent* ent;
+    while ((ent = readdir(d)) != nullptr) {
+      std::string name = ent->d_name;
+      if (name == "." || name == "..") continue;
+      std::string full = path + "/" + name;
+      struct stat st;
+      if (lstat(full.c_str(), &st) == 0 && S_ISDIR(st.st_mode))
+        rmtree(full);
+      else
+        unlink(full.c_str());
+    }
+    closedir(d);
+  }
+  rmdir(path.c_str());
+}
+
+// ---------------------------------------------------------------------------
 // filesystem helpers
 // ---------------------------------------------------------------------------
 static bool mkdirs(const std::string& path) {
@@ -526,7 +571,7 @@ static ExecOutcome run_cold(const std::string& script_path,
                             const std::string& stdout_path,
                             const std::string& stderr_path,
                             const std::map<std::string, std::string>& extra_env,
-                            double timeout_s) {
+                            double timeout_s, const std::string& workspace) {
   ExecOutcome out;
   pid_t pid = fork();
   if (pid < 0) {
@@ -542,7 +587,7 @@ static ExecOutcome run_cold(const std::string& script_path,
     if (so >= 0) dup2(so, 1);
     if (se >= 0) dup2(se, 2);
     for (auto& kv : extra_env) setenv(kv.first.c_str(), kv.second.c_str(), 1);
-    if (chdir(g_cfg.workspace.c_str()) != 0) _exit(126);
+    if (chdir(workspace.c_str()) != 0) _exit(126);
     execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), script_path.c_str(),
            (char*)nullptr);
     _exit(127);
@@ -581,11 +626,11 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
                                   const std::string& stdout_path,
                                   const std::string& stderr_path,
                                   const std::map<std::string, std::string>& extra_env,
-                                  double timeout_s) {
+                                  double timeout_s, const std::string& workspace) {
   uint64_t id = g_job_id.fetch_add(1);
   std::string req = "{\"event\":\"run\",\"id\":" + std::to_string(id) +
                     ",\"script\":" + json::quote(script_path) +
-                    ",\"cwd\":" + json::quote(g_cfg.workspace) +
+                    ",\"cwd\":" + json::quote(workspace) +
                     ",\"stdout\":" + json::quote(stdout_path) +
                     ",\"stderr\":" + json::quote(stderr_path) + ",\"env\":{";
   bool first = true;
@@ -785,7 +830,8 @@ static std::string url_decode(const std::string& s) {
   return out;
 }
 
-static void handle_execute(Conn& conn, const HttpRequest& req) {
+static void handle_execute(Conn& conn, const HttpRequest& req,
+                           const std::string& workspace) {
   json::Parser parser(req.body);
   json::Value body = parser.parse();
   if (!parser.ok || !body.is_object() || !body.get("source_code") ||
@@ -802,6 +848,8 @@ static void handle_execute(Conn& conn, const HttpRequest& req) {
     if (env->is_object())
       for (auto& kv : env->obj)
         if (kv.second.is_string()) extra_env[kv.first] = kv.second.str;
+
+  mkdirs(workspace);
 
   // execution start time for the changed-file scan
   struct timespec now;
@@ -822,14 +870,14 @@ static void handle_execute(Conn& conn, const HttpRequest& req) {
   ExecOutcome outcome;
   if (g_cfg.zygote && g_zygote && g_zygote->alive()) {
     outcome = run_via_zygote(script_path, stdout_path, stderr_path, extra_env,
-                             timeout_s);
+                             timeout_s, workspace);
   } else {
-    outcome =
-        run_cold(script_path, stdout_path, stderr_path, extra_env, timeout_s);
+    outcome = run_cold(script_path, stdout_path, stderr_path, extra_env,
+                       timeout_s, workspace);
   }
 
   std::vector<std::string> changed;
-  scan_changed(g_cfg.workspace, "", start, g_cfg.scan_recursive, changed);
+  scan_changed(workspace, "", start, g_cfg.scan_recursive, changed);
 
   std::string resp = "{\"stdout\":" + json::quote(outcome.stdout_text) +
                      ",\"stderr\":" + json::quote(outcome.stderr_text) +
@@ -850,6 +898,44 @@ static void handle_execute(Conn& conn, const HttpRequest& req) {
   conn.respond(200, "OK", resp);
 }
 
+// workspace file routes (PUT/GET), shared by the legacy pod-style routes
+// and the per-session routes
+static void handle_workspace_io(Conn& conn, const HttpRequest& req,
+                                const std::string& workspace,
+                                const std::string& rel_encoded) {
+  std::string rel = url_decode(rel_encoded);
+  if (!safe_rel_path(rel)) {
+    conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
+    return;
+  }
+  std::string full = workspace + "/" + rel;
+  if (req.method == "PUT") {
+    size_t slash = full.rfind('/');
+    if (slash != std::string::npos) mkdirs(full.substr(0, slash));
+    if (write_file(full, req.body))
+      conn.respond(204, "No Content", "");
+    else
+      conn.respond(500, "Internal Server Error", "{\"error\":\"write failed\"}");
+  } else if (req.method == "GET") {
+    struct stat st;
+    if (stat(full.c_str(), &st) != 0 || !S_ISREG(st.st_mode)) {
+      conn.respond(404, "Not Found", "{\"error\":\"not found\"}");
+    } else {
+      conn.respond(200, "OK", read_file(full), "application/octet-stream");
+    }
+  } else {
+    conn.respond(405, "Method Not Allowed", "{\"error\":\"method\"}");
+  }
+}
+
+static bool lookup_sandbox(const std::string& id, std::string& workspace) {
+  std::lock_guard<std::mutex> lk(g_sandboxes.mu);
+  auto it = g_sandboxes.workspaces.find(id);
+  if (it == g_sandboxes.workspaces.end()) return false;
+  workspace = it->second;
+  return true;
+}
+
 static void handle_conn(int fd) {
   Conn conn(fd);
   while (true) {
@@ -862,30 +948,50 @@ static void handle_conn(int fd) {
                    std::string("{\"status\":\"ok\",\"warm\":") +
                        (warm ? "true" : "false") + "}");
     } else if (req.path.rfind("/workspace/", 0) == 0) {
-      std::string rel = url_decode(req.path.substr(strlen("/workspace/")));
-      if (!safe_rel_path(rel)) {
-        conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
-      } else if (req.method == "PUT") {
-        std::string full = g_cfg.workspace + "/" + rel;
-        size_t slash = full.rfind('/');
-        if (slash != std::string::npos) mkdirs(full.substr(0, slash));
-        if (write_file(full, req.body))
+      // legacy pod-style routes: the default workspace
+      handle_workspace_io(conn, req, g_cfg.workspace,
+                          req.path.substr(strlen("/workspace/")));
+    } else if (req.method == "POST" && req.path == "/execute") {
+      handle_execute(conn, req, g_cfg.workspace);
+    } else if (req.method == "POST" && req.path == "/sandboxes") {
+      // fresh single-use workspace session
+      std::string id = random_hex(12);
+      std::string ws = g_sessions_root + "/" + id;
+      mkdirs(ws);
+      {
+        std::lock_guard<std::mutex> lk(g_sandboxes.mu);
+        g_sandboxes.workspaces[id] = ws;
+      }
+      conn.respond(200, "OK", "{\"id\":" + json::quote(id) + "}");
+    } else if (req.path.rfind("/sandboxes/", 0) == 0) {
+      std::string rest = req.path.substr(strlen("/sandboxes/"));
+      size_t slash = rest.find('/');
+      std::string id = rest.substr(0, slash == std::string::npos ? rest.size()
+                                                                 : slash);
+      std::string ws;
+      if (!lookup_sandbox(id, ws)) {
+        conn.respond(404, "Not Found", "{\"error\":\"no such sandbox\"}");
+      } else if (slash == std::string::npos) {
+        if (req.method == "DELETE") {
+          {
+            std::lock_guard<std::mutex> lk(g_sandboxes.mu);
+            g_sandboxes.workspaces.erase(id);
+          }
+          rmtree(ws);
           conn.respond(204, "No Content", "");
-        else
-          conn.respond(500, "Internal Server Error", "{\"error\":\"write failed\"}");
-      } else if (req.method == "GET") {
-        std::string full = g_cfg.workspace + "/" + rel;
-        struct stat st;
-        if (stat(full.c_str(), &st) != 0 || !S_ISREG(st.st_mode)) {
-          conn.respond(404, "Not Found", "{\"error\":\"not found\"}");
         } else {
-          conn.respond(200, "OK", read_file(full), "application/octet-stream");
+          conn.respond(405, "Method Not Allowed", "{\"error\":\"method\"}");
         }
       } else {
-        conn.respond(405, "Method Not Allowed", "{\"error\":\"method\"}");
+        std::string sub = rest.substr(slash + 1);
+        if (sub.rfind("workspace/", 0) == 0) {
+          handle_workspace_io(conn, req, ws, sub.substr(strlen("workspace/")));
+        } else if (req.method == "POST" && sub == "execute") {
+          handle_execute(conn, req, ws);
+        } else {
+          conn.respond(404, "Not Found", "{\"error\":\"no route\"}");
+        }
       }
-    } else if (req.method == "POST" && req.path == "/execute") {
-      handle_execute(conn, req);
     } else {
       conn.respond(404, "Not Found", "{\"error\":\"no route\"}");
     }
@@ -960,6 +1066,8 @@ int main(int, char**) {
   g_cfg.runtime_dir = env_or("APP_RUNTIME_DIR", self_dir);
 
   mkdirs(g_cfg.workspace);
+  g_sessions_root = env_or("APP_SESSIONS_DIR", "/tmp/sandboxes");
+  mkdirs(g_sessions_root);
 
   if (g_cfg.zygote) {
     g_zygote = new Zygote();

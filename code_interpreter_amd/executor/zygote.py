@@ -44,6 +44,11 @@ def _run_job_in_child(job: dict) -> None:
     """Executed in the forked child. Never returns."""
     exit_code = 0
     try:
+        try:
+            signal.set_wakeup_fd(-1)
+            signal.signal(signal.SIGCHLD, signal.SIG_DFL)
+        except (ValueError, OSError):
+            pass
         os.setsid()
         stdin_fd = os.open("/dev/null", os.O_RDONLY)
         stdout_fd = os.open(job["stdout"], os.O_WRONLY | os.O_CREAT | os.O_TRUNC)
@@ -83,6 +88,11 @@ class WarmChild:
         if pid == 0:
             parent_sock.close()
             try:
+                try:
+                    signal.set_wakeup_fd(-1)
+                    signal.signal(signal.SIGCHLD, signal.SIG_DFL)
+                except (ValueError, OSError):
+                    pass
                 sandbox_runtime.prewarm()  # HIP init happens HERE, post-fork
                 child_sock.sendall(b"ready\n")
                 data = b""
@@ -121,7 +131,13 @@ def main() -> None:
     parser.add_argument("--fd", type=int, required=True)
     args = parser.parse_args()
 
-    signal.signal(signal.SIGCHLD, signal.SIG_DFL)
+    # SIGCHLD self-pipe: child exits interrupt select() immediately instead
+    # of waiting out the poll tick (request latency, not correctness)
+    wake_r, wake_w = os.pipe()
+    os.set_blocking(wake_r, False)
+    os.set_blocking(wake_w, False)
+    signal.set_wakeup_fd(wake_w)
+    signal.signal(signal.SIGCHLD, lambda *_: None)
 
     server = socket.socket(fileno=args.fd)
     server.setblocking(True)
@@ -132,18 +148,27 @@ def main() -> None:
         except OSError:
             os._exit(0)
 
-    warm = WarmChild()
-    announced_warm = False
+    # Warm-children pool: each is a pre-forked, HIP-initialized interpreter
+    # waiting for exactly one job (single-use sandbox semantics); the pool
+    # refills in the background so steady-state request cost is a handoff.
+    pool_target = max(0, int(os.environ.get("APP_WARM_CHILDREN", "2")))
+    warm_pool: list[WarmChild] = []
+    announced = 0
     running: dict[int, int] = {}  # pid -> job id
     buf = b""
 
     while True:
-        if not announced_warm and warm is not None and warm.poll_ready():
-            announced_warm = True
-            send({"event": "warm"})
+        # refill: forking is cheap; the expensive HIP init happens inside
+        # the children concurrently
+        while len(warm_pool) < pool_target:
+            warm_pool.append(WarmChild())
+        ready_count = sum(1 for w in warm_pool if w.poll_ready())
+        if ready_count != announced:
+            announced = ready_count
+            send({"event": "warm", "ready": ready_count})
 
         # reap finished children
-        while running:
+        while True:
             try:
                 pid, status = os.waitpid(-1, os.WNOHANG)
             except ChildProcessError:
@@ -154,14 +179,25 @@ def main() -> None:
             if job_id is not None:
                 code = os.WEXITSTATUS(status) if os.WIFEXITED(status) else -1
                 send({"event": "exit", "id": job_id, "code": code})
+            else:
+                # a warm child died before use: drop it from the pool
+                warm_pool = [w for w in warm_pool if w.pid != pid]
 
-        r, _, _ = select.select([server], [], [], 0.02)
-        if not r:
+        try:
+            r, _, _ = select.select([server, wake_r], [], [], 0.25)
+        except InterruptedError:
+            continue
+        if wake_r in r:
+            try:
+                os.read(wake_r, 4096)
+            except BlockingIOError:
+                pass
+        if server not in r:
             continue
         chunk = server.recv(65536)
         if not chunk:
             # server went away: kill children and exit
-            for pid in running:
+            for pid in list(running) + [w.pid for w in warm_pool]:
                 try:
                     os.killpg(pid, signal.SIGKILL)
                 except OSError:
@@ -179,11 +215,15 @@ def main() -> None:
             if msg.get("event") != "run":
                 continue
             job_id = msg["id"]
-            if warm is not None and warm.poll_ready():
-                warm.submit(msg)
-                running[warm.pid] = job_id
-                send({"event": "start", "id": job_id, "pid": warm.pid})
-                warm = None  # single-use (fresh-sandbox semantics)
+            child = None
+            for i, w in enumerate(warm_pool):
+                if w.poll_ready():
+                    child = warm_pool.pop(i)
+                    break
+            if child is not None:
+                child.submit(msg)
+                running[child.pid] = job_id
+                send({"event": "start", "id": job_id, "pid": child.pid})
             else:
                 pid = os.fork()
                 if pid == 0:

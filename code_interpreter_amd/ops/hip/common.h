@@ -1,0 +1,52 @@
+// Shared declarations for the gfx950 HIP kernel library.
+// All kernels are written CDNA4-first: wave64, MFMA matrix cores, LDS
+// tiling sized for 160 KiB/CU, grid sizing for 256 CUs / 8 XCDs.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) {                                                 \
+      throw std::runtime_error(std::string("HIP error at " __FILE__ ":") +  \
+                               std::to_string(__LINE__) + ": " +            \
+                               hipGetErrorString(_e));                      \
+    }                                                                       \
+  } while (0)
+
+// element dtypes understood by the generic elementwise/reduce entry points
+enum class DType : int { F32 = 0, F64 = 1 };
+
+enum class UnaryOp : int { Square = 0, Neg = 1, Abs = 2, Sqrt = 3, Exp = 4 };
+enum class BinOp : int { Add = 0, Sub = 1, Mul = 2, Div = 3 };
+
+// launchers (defined in kernels_ew.hip / gemm_*.hip); all take raw device
+// pointers and run on `stream`
+void launch_unary(DType dt, UnaryOp op, const void* in, void* out,
+                  int64_t n, hipStream_t stream);
+void launch_binary(DType dt, BinOp op, const void* a, const void* b, void* out,
+                   int64_t n, hipStream_t stream);
+void launch_binary_scalar(DType dt, BinOp op, const void* a, double scalar,
+                          void* out, int64_t n, hipStream_t stream);
+// reduction: out_partials must hold >= reduce_num_partials(n) elements of dt;
+// final scalar (in dt) is written to out_scalar (device ptr) by stage 2
+int reduce_num_partials(int64_t n);
+void launch_sum(DType dt, bool square_inputs, const void* in, void* partials,
+                void* out_scalar, int64_t n, hipStream_t stream);
+// philox4x32-10 uniform doubles/floats in [0, 1)
+void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
+                         uint64_t offset, hipStream_t stream);
+
+// row-major GEMM: C[M,N] = A[M,K] @ B[K,N]
+void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,
+                     int k, hipStream_t stream);
+void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
+                     int k, hipStream_t stream);
+// bf16 in (as uint16 storage), f32 accumulate, bf16 out
+void launch_gemm_bf16(const uint16_t* a, const uint16_t* b, uint16_t* c, int m,
+                      int n, int k, hipStream_t stream);

@@ -1,0 +1,368 @@
+// Elementwise, reduction and RNG kernels for gfx950.
+//
+// Design notes (per the CDNA4 performance rules):
+//  - memory-bound kernels are vectorized to 16 B/lane (double2 / float4):
+//    scalar loads leave >2x bandwidth on the table;
+//  - grids are capped at 8 blocks/CU x 256 CUs = 2048 workgroups with
+//    grid-stride loops (fills all 8 XCDs, avoids launch overhead);
+//  - reductions: wave64 __shfl_down tree -> per-wave LDS slot -> block
+//    reduce -> one partial per block -> single-block stage 2 (no global
+//    atomics, deterministic for a fixed grid);
+//  - RNG is Philox4x32-10 (counter-based): stateless, any index range can
+//    be generated independently at full bandwidth.
+
+#include "common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxBlocks = 2048;  // 256 CUs x 8 blocks
+
+inline int grid_for(int64_t work_items) {
+  int64_t blocks = (work_items + kBlock - 1) / kBlock;
+  if (blocks > kMaxBlocks) blocks = kMaxBlocks;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+// ---------------------------------------------------------------------------
+// elementwise
+// ---------------------------------------------------------------------------
+template <typename T>
+struct Vec2 {
+  T x, y;
+};
+
+template <typename T, int OP>
+__device__ __forceinline__ T apply_unary(T v) {
+  if constexpr (OP == (int)UnaryOp::Square) return v * v;
+  if constexpr (OP == (int)UnaryOp::Neg) return -v;
+  if constexpr (OP == (int)UnaryOp::Abs) return v < T(0) ? -v : v;
+  if constexpr (OP == (int)UnaryOp::Sqrt) return (T)sqrt((double)v);
+  if constexpr (OP == (int)UnaryOp::Exp) return (T)exp((double)v);
+  return v;
+}
+
+template <typename T, int OP>
+__device__ __forceinline__ T apply_bin(T a, T b) {
+  if constexpr (OP == (int)BinOp::Add) return a + b;
+  if constexpr (OP == (int)BinOp::Sub) return a - b;
+  if constexpr (OP == (int)BinOp::Mul) return a * b;
+  if constexpr (OP == (int)BinOp::Div) return a / b;
+  return a;
+}
+
+// processes 2 elements of T per lane per step (16 B for f64, 8 B for f32;
+// f32 uses 4/lane below)
+template <typename T, int OP>
+__global__ void unary_kernel(const T* __restrict__ in, T* __restrict__ out,
+                             int64_t n) {
+  using V2 = Vec2<T>;
+  int64_t n2 = n / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n2;
+       i += stride) {
+    V2 v = reinterpret_cast<const V2*>(in)[i];
+    v.x = apply_unary<T, OP>(v.x);
+    v.y = apply_unary<T, OP>(v.y);
+    reinterpret_cast<V2*>(out)[i] = v;
+  }
+  // tail
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    out[n - 1] = apply_unary<T, OP>(in[n - 1]);
+  }
+}
+
+template <typename T, int OP>
+__global__ void binary_kernel(const T* __restrict__ a, const T* __restrict__ b,
+                              T* __restrict__ out, int64_t n) {
+  using V2 = Vec2<T>;
+  int64_t n2 = n / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n2;
+       i += stride) {
+    V2 va = reinterpret_cast<const V2*>(a)[i];
+    V2 vb = reinterpret_cast<const V2*>(b)[i];
+    va.x = apply_bin<T, OP>(va.x, vb.x);
+    va.y = apply_bin<T, OP>(va.y, vb.y);
+    reinterpret_cast<V2*>(out)[i] = va;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    out[n - 1] = apply_bin<T, OP>(a[n - 1], b[n - 1]);
+  }
+}
+
+template <typename T, int OP>
+__global__ void binary_scalar_kernel(const T* __restrict__ a, T scalar,
+                                     T* __restrict__ out, int64_t n) {
+  using V2 = Vec2<T>;
+  int64_t n2 = n / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n2;
+       i += stride) {
+    V2 va = reinterpret_cast<const V2*>(a)[i];
+    va.x = apply_bin<T, OP>(va.x, scalar);
+    va.y = apply_bin<T, OP>(va.y, scalar);
+    reinterpret_cast<V2*>(out)[i] = va;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    out[n - 1] = apply_bin<T, OP>(a[n - 1], scalar);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// reduction (sum / sum-of-squares)
+// ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ double wave_reduce_sum(double v) {
+  // wave64 butterfly; __shfl_down over the full 64-lane wave
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+template <typename T, bool SQUARE>
+__global__ void sum_stage1(const T* __restrict__ in, T* __restrict__ partials,
+                           int64_t n) {
+  using V2 = Vec2<T>;
+  int64_t n2 = n / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  double acc = 0.0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n2;
+       i += stride) {
+    V2 v = reinterpret_cast<const V2*>(in)[i];
+    if constexpr (SQUARE) {
+      acc += (double)v.x * (double)v.x + (double)v.y * (double)v.y;
+    } else {
+      acc += (double)v.x + (double)v.y;
+    }
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    double t = (double)in[n - 1];
+    acc += SQUARE ? t * t : t;
+  }
+  acc = wave_reduce_sum<T>(acc);
+  __shared__ double wave_sums[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wave_sums[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double total = 0.0;
+    for (int w = 0; w < (int)blockDim.x / 64; w++) total += wave_sums[w];
+    partials[blockIdx.x] = (T)total;
+  }
+}
+
+template <typename T>
+__global__ void sum_stage2(const T* __restrict__ partials, T* __restrict__ out,
+                           int n) {
+  double acc = 0.0;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) acc += (double)partials[i];
+  acc = wave_reduce_sum<T>(acc);
+  __shared__ double wave_sums[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wave_sums[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double total = 0.0;
+    for (int w = 0; w < (int)blockDim.x / 64; w++) total += wave_sums[w];
+    out[0] = (T)total;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Philox4x32-10 uniform RNG
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void philox_round(uint32_t& c0, uint32_t& c1,
+                                             uint32_t& c2, uint32_t& c3,
+                                             uint32_t k0, uint32_t k1) {
+  const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  uint32_t hi0 = __umulhi(M0, c0), lo0 = M0 * c0;
+  uint32_t hi1 = __umulhi(M1, c2), lo1 = M1 * c2;
+  uint32_t n0 = hi1 ^ c1 ^ k0;
+  uint32_t n1 = lo1;
+  uint32_t n2 = hi0 ^ c3 ^ k1;
+  uint32_t n3 = lo0;
+  c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+}
+
+__device__ __forceinline__ void philox10(uint64_t ctr, uint64_t seed,
+                                         uint32_t out[4]) {
+  const uint32_t W0 = 0x9E3779B9u, W1 = 0xBB67AE85u;
+  uint32_t c0 = (uint32_t)ctr, c1 = (uint32_t)(ctr >> 32), c2 = 0x85EBCA6Bu,
+           c3 = 0xC2B2AE35u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; r++) {
+    philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += W0;
+    k1 += W1;
+  }
+  out[0] = c0; out[1] = c1; out[2] = c2; out[3] = c3;
+}
+
+// each counter yields 2 doubles (53-bit mantissas) or 4 floats
+__global__ void rand_f64_kernel(double* __restrict__ out, int64_t n,
+                                uint64_t seed, uint64_t offset) {
+  int64_t pairs = (n + 1) / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const double scale = 1.0 / 9007199254740992.0;  // 2^-53
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < pairs;
+       i += stride) {
+    uint32_t r[4];
+    philox10(offset + (uint64_t)i, seed, r);
+    uint64_t u0 = ((uint64_t)r[1] << 32) | r[0];
+    uint64_t u1 = ((uint64_t)r[3] << 32) | r[2];
+    double d0 = (double)(u0 >> 11) * scale;
+    double d1 = (double)(u1 >> 11) * scale;
+    int64_t j = i * 2;
+    if (j + 1 < n) {
+      reinterpret_cast<Vec2<double>*>(out)[i] = {d0, d1};
+    } else if (j < n) {
+      out[j] = d0;
+    }
+  }
+}
+
+__global__ void rand_f32_kernel(float* __restrict__ out, int64_t n,
+                                uint64_t seed, uint64_t offset) {
+  int64_t quads = (n + 3) / 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float scale = 1.0f / 16777216.0f;  // 2^-24
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < quads;
+       i += stride) {
+    uint32_t r[4];
+    philox10(offset + (uint64_t)i, seed, r);
+    int64_t j = i * 4;
+    if (j + 3 < n) {
+      float4 v = {(float)(r[0] >> 8) * scale, (float)(r[1] >> 8) * scale,
+                  (float)(r[2] >> 8) * scale, (float)(r[3] >> 8) * scale};
+      reinterpret_cast<float4*>(out)[i] = v;
+    } else {
+      for (int q = 0; q < 4 && j + q < n; q++)
+        out[j + q] = (float)(r[q] >> 8) * scale;
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+template <typename T>
+static void launch_unary_t(UnaryOp op, const T* in, T* out, int64_t n,
+                           hipStream_t s) {
+  int grid = grid_for(n / 2 + 1);
+  switch (op) {
+#define CASE(OP)                                                         \
+  case UnaryOp::OP:                                                      \
+    hipLaunchKernelGGL((unary_kernel<T, (int)UnaryOp::OP>), dim3(grid),  \
+                       dim3(kBlock), 0, s, in, out, n);                  \
+    break;
+    CASE(Square) CASE(Neg) CASE(Abs) CASE(Sqrt) CASE(Exp)
+#undef CASE
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_unary(DType dt, UnaryOp op, const void* in, void* out, int64_t n,
+                  hipStream_t stream) {
+  if (dt == DType::F64)
+    launch_unary_t(op, (const double*)in, (double*)out, n, stream);
+  else
+    launch_unary_t(op, (const float*)in, (float*)out, n, stream);
+}
+
+template <typename T>
+static void launch_binary_t(BinOp op, const T* a, const T* b, T* out,
+                            int64_t n, hipStream_t s) {
+  int grid = grid_for(n / 2 + 1);
+  switch (op) {
+#define CASE(OP)                                                          \
+  case BinOp::OP:                                                         \
+    hipLaunchKernelGGL((binary_kernel<T, (int)BinOp::OP>), dim3(grid),    \
+                       dim3(kBlock), 0, s, a, b, out, n);                 \
+    break;
+    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div)
+#undef CASE
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_binary(DType dt, BinOp op, const void* a, const void* b, void* out,
+                   int64_t n, hipStream_t stream) {
+  if (dt == DType::F64)
+    launch_binary_t(op, (const double*)a, (const double*)b, (double*)out, n,
+                    stream);
+  else
+    launch_binary_t(op, (const float*)a, (const float*)b, (float*)out, n,
+                    stream);
+}
+
+template <typename T>
+static void launch_binary_scalar_t(BinOp op, const T* a, T scalar, T* out,
+                                   int64_t n, hipStream_t s) {
+  int grid = grid_for(n / 2 + 1);
+  switch (op) {
+#define CASE(OP)                                                             \
+  case BinOp::OP:                                                            \
+    hipLaunchKernelGGL((binary_scalar_kernel<T, (int)BinOp::OP>), dim3(grid), \
+                       dim3(kBlock), 0, s, a, scalar, out, n);               \
+    break;
+    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div)
+#undef CASE
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_binary_scalar(DType dt, BinOp op, const void* a, double scalar,
+                          void* out, int64_t n, hipStream_t stream) {
+  if (dt == DType::F64)
+    launch_binary_scalar_t(op, (const double*)a, scalar, (double*)out, n,
+                           stream);
+  else
+    launch_binary_scalar_t(op, (const float*)a, (float)scalar, (float*)out, n,
+                           stream);
+}
+
+int reduce_num_partials(int64_t n) { return grid_for(n / 2 + 1); }
+
+template <typename T>
+static void launch_sum_t(bool square, const T* in, T* partials, T* out_scalar,
+                         int64_t n, hipStream_t s) {
+  int grid = grid_for(n / 2 + 1);
+  if (square)
+    hipLaunchKernelGGL((sum_stage1<T, true>), dim3(grid), dim3(kBlock), 0, s,
+                       in, partials, n);
+  else
+    hipLaunchKernelGGL((sum_stage1<T, false>), dim3(grid), dim3(kBlock), 0, s,
+                       in, partials, n);
+  HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
+                     out_scalar, grid);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_sum(DType dt, bool square_inputs, const void* in, void* partials,
+                void* out_scalar, int64_t n, hipStream_t stream) {
+  if (dt == DType::F64)
+    launch_sum_t(square_inputs, (const double*)in, (double*)partials,
+                 (double*)out_scalar, n, stream);
+  else
+    launch_sum_t(square_inputs, (const float*)in, (float*)partials,
+                 (float*)out_scalar, n, stream);
+}
+
+void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
+                         uint64_t offset, hipStream_t stream) {
+  if (dt == DType::F64) {
+    int grid = grid_for((n + 1) / 2);
+    hipLaunchKernelGGL(rand_f64_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       (double*)out, n, seed, offset);
+  } else {
+    int grid = grid_for((n + 3) / 4);
+    hipLaunchKernelGGL(rand_f32_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       (float*)out, n, seed, offset);
+  }
+  HIP_CHECK(hipGetLastError());
+}

@@ -1,0 +1,447 @@
+"""numpy -> gfx950 HIP kernel routing for sandboxed user code.
+
+Two mechanisms, installed by the sandbox runtime (executor/sandbox_runtime
+.py) when a GPU is visible:
+
+1. patched module-level entry points (numpy.random.rand, numpy.matmul,
+   numpy.dot, numpy.square, numpy.sum): above a size threshold the work
+   runs on the MI355X through _hipops and the result stays device-resident;
+2. DeviceArray: a duck array (NEP 13/18 __array_ufunc__ +
+   __array_function__) so follow-on numpy calls on a device-resident
+   result keep running on the GPU (numpy.sum(numpy.square(x)) never
+   round-trips the intermediate through host memory); any unsupported
+   operation transparently materializes to a host ndarray and falls back.
+
+Everything computes the same values user code would get on the CPU (same
+dtype; RNG is Philox instead of MT19937 -- a documented backend change,
+seeded from numpy's global RNG so np.random.seed still fixes the stream).
+
+This file is imported standalone (sys.path) inside sandbox children; it
+must not import the control-plane package or torch.
+"""
+
+import os
+import sys
+
+_OPS_DIR = os.path.dirname(os.path.abspath(__file__))
+if _OPS_DIR not in sys.path:
+    sys.path.insert(0, _OPS_DIR)
+
+import _hipops
+
+import numpy as _np
+
+# dtype codes shared with _hipops
+_F32, _F64 = 0, 1
+_UNARY = {"square": 0, "negative": 1, "absolute": 2, "sqrt": 3, "exp": 4}
+_BINARY = {"add": 0, "subtract": 1, "multiply": 2, "divide": 3, "true_divide": 3}
+
+MIN_ELEMS = int(os.environ.get("APP_HIP_NUMPY_MIN_ELEMS", 2_000_000))
+MIN_MATMUL_FLOPS = float(os.environ.get("APP_HIP_NUMPY_MIN_MATMUL_FLOPS", 5e7))
+
+_state = {"ready": False, "failed": None}
+
+
+def available() -> bool:
+    return _hipops.is_available()
+
+
+def _ensure_ready() -> None:
+    if _state["ready"]:
+        return
+    if _state["failed"]:
+        raise RuntimeError(_state["failed"])
+    try:
+        _hipops.init(0)
+        _state["ready"] = True
+    except Exception as e:
+        _state["failed"] = str(e)
+        raise
+
+
+def warmup() -> None:
+    """Bring up the HIP context + staging pools and touch every hot kernel
+    once (pre-forked warm child calls this before any request)."""
+    if not available():
+        raise RuntimeError("no AMD GPU visible")
+    _ensure_ready()
+    h = _hipops.rand(4096, _F64, 12345)
+    h2 = _hipops.unary(h, _UNARY["square"], _F64, 4096)
+    _hipops.sum(h, _F64, 4096, 1)
+    a = _np.zeros((64, 64), dtype=_np.float32)
+    ha = _hipops.upload(a)
+    hc = _hipops.gemm(ha, ha, 64, 64, 64, _F32)
+    for handle in (h, h2, ha, hc):
+        _hipops.free(handle)
+    _hipops.synchronize()
+
+
+def _dtype_code(dtype):
+    if dtype == _np.float64:
+        return _F64
+    if dtype == _np.float32:
+        return _F32
+    return None
+
+
+class DeviceArray:
+    """Device-resident array, duck-typed against numpy via NEP 13/18."""
+
+    __array_priority__ = 1000.0
+
+    def __init__(self, handle, shape, dtype):
+        self._handle = handle
+        self.shape = tuple(shape)
+        self.dtype = _np.dtype(dtype)
+        self._host = None  # materialized cache
+
+    # -- basics ---------------------------------------------------------
+    @property
+    def size(self):
+        n = 1
+        for s in self.shape:
+            n *= s
+        return n
+
+    @property
+    def ndim(self):
+        return len(self.shape)
+
+    def __len__(self):
+        if not self.shape:
+            raise TypeError("len() of unsized object")
+        return self.shape[0]
+
+    def __repr__(self):
+        return repr(self.materialize())
+
+    def __str__(self):
+        return str(self.materialize())
+
+    def __del__(self):
+        try:
+            if self._handle is not None:
+                _hipops.free(self._handle)
+        except Exception:
+            pass
+
+    def materialize(self) -> _np.ndarray:
+        if self._host is None:
+            out = _np.empty(self.shape, dtype=self.dtype)
+            _hipops.download(self._handle, out)
+            self._host = out
+        return self._host
+
+    def __array__(self, dtype=None, copy=None):
+        host = self.materialize()
+        if dtype is not None and dtype != host.dtype:
+            return host.astype(dtype)
+        return host
+
+    # anything we don't implement: materialize and delegate
+    def __getattr__(self, name):
+        return getattr(self.materialize(), name)
+
+    def __getitem__(self, idx):
+        return self.materialize()[idx]
+
+    def __float__(self):
+        return float(self.materialize())
+
+    def __iter__(self):
+        return iter(self.materialize())
+
+    # -- device compute -------------------------------------------------
+    def _unary(self, opname):
+        out = _hipops.unary(self._handle, _UNARY[opname], _dtype_code(self.dtype), self.size)
+        return DeviceArray(out, self.shape, self.dtype)
+
+    def _binary(self, opname, other, reverse=False):
+        code = _dtype_code(self.dtype)
+        if isinstance(other, DeviceArray):
+            if other.shape != self.shape or other.dtype != self.dtype:
+                return NotImplemented
+            a, b = (other, self) if reverse else (self, other)
+            out = _hipops.binary(a._handle, b._handle, _BINARY[opname], code, self.size)
+            return DeviceArray(out, self.shape, self.dtype)
+        if isinstance(other, (int, float)):
+            if reverse and opname in ("subtract", "divide", "true_divide"):
+                return NotImplemented  # scalar-first sub/div: fall back
+            out = _hipops.binary_scalar(
+                self._handle, float(other), _BINARY[opname], code, self.size
+            )
+            return DeviceArray(out, self.shape, self.dtype)
+        return NotImplemented
+
+    def sum(self, axis=None, **kwargs):
+        if axis is None and not kwargs.get("keepdims"):
+            return self.dtype.type(
+                _hipops.sum(self._handle, _dtype_code(self.dtype), self.size, 0)
+            )
+        return self.materialize().sum(axis=axis, **kwargs)
+
+    def mean(self, axis=None, **kwargs):
+        if axis is None and not kwargs.get("keepdims"):
+            return self.dtype.type(float(self.sum()) / self.size)
+        return self.materialize().mean(axis=axis, **kwargs)
+
+    def square_sum(self):
+        """Fused sum(x*x) -- no intermediate array."""
+        return self.dtype.type(
+            _hipops.sum(self._handle, _dtype_code(self.dtype), self.size, 1)
+        )
+
+    # -- NEP 13: ufuncs --------------------------------------------------
+    def __array_ufunc__(self, ufunc, method, *inputs, **kwargs):
+        if kwargs.get("out") is not None:
+            return self._fallback_ufunc(ufunc, method, inputs, kwargs)
+        name = ufunc.__name__
+        if method == "__call__":
+            if name in _UNARY and len(inputs) == 1 and inputs[0] is self:
+                return self._unary(name)
+            if name in _BINARY and len(inputs) == 2:
+                a, b = inputs
+                if a is self:
+                    r = self._binary(name, b)
+                elif b is self:
+                    r = self._binary(name, a, reverse=True)
+                else:
+                    r = NotImplemented
+                if r is not NotImplemented:
+                    return r
+        elif method == "reduce" and name == "add" and len(inputs) == 1:
+            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
+                return inputs[0].sum()
+        return self._fallback_ufunc(ufunc, method, inputs, kwargs)
+
+    def _fallback_ufunc(self, ufunc, method, inputs, kwargs):
+        host_inputs = [
+            x.materialize() if isinstance(x, DeviceArray) else x for x in inputs
+        ]
+        return getattr(ufunc, method)(*host_inputs, **kwargs)
+
+    # -- NEP 18: numpy functions -----------------------------------------
+    def __array_function__(self, func, types, args, kwargs):
+        if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
+            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
+                return args[0].sum()
+        if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
+            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
+                return args[0].mean()
+        if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
+            r = matmul(*args, _force=True)
+            if r is not NotImplemented:
+                return r
+        if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
+            return args[0]._unary("square")
+        # generic fallback: materialize every DeviceArray
+        host_args = [
+            x.materialize() if isinstance(x, DeviceArray) else x for x in args
+        ]
+        return func(*host_args, **kwargs)
+
+    # -- operators -------------------------------------------------------
+    def __add__(self, o):
+        return self._coerce(self._binary("add", o))
+
+    def __radd__(self, o):
+        return self._coerce(self._binary("add", o, reverse=True))
+
+    def __sub__(self, o):
+        return self._coerce(self._binary("subtract", o))
+
+    def __mul__(self, o):
+        return self._coerce(self._binary("multiply", o))
+
+    def __rmul__(self, o):
+        return self._coerce(self._binary("multiply", o, reverse=True))
+
+    def __truediv__(self, o):
+        return self._coerce(self._binary("divide", o))
+
+    def __matmul__(self, o):
+        r = matmul(self, o, _force=True)
+        return self.materialize() @ _asarray(o) if r is NotImplemented else r
+
+    def _coerce(self, r):
+        if r is NotImplemented:
+            raise TypeError("unsupported operand for device array")
+        return r
+
+
+def _asarray(x):
+    return x.materialize() if isinstance(x, DeviceArray) else _np.asarray(x)
+
+
+def _to_device(x) -> "DeviceArray | None":
+    """Upload a host ndarray (f32/f64, C-contiguous) to the device."""
+    if isinstance(x, DeviceArray):
+        return x
+    arr = _np.asarray(x)
+    if _dtype_code(arr.dtype) is None:
+        return None
+    arr = _np.ascontiguousarray(arr)
+    return DeviceArray(_hipops.upload(arr), arr.shape, arr.dtype)
+
+
+# ---------------------------------------------------------------------------
+# module-level compute entry points (used by the numpy patches and tests)
+# ---------------------------------------------------------------------------
+def rand(*shape, seed=None):
+    """Uniform [0,1) float64 of the given shape, generated on-device
+    (Philox4x32-10)."""
+    _ensure_ready()
+    n = 1
+    for s in shape:
+        n *= int(s)
+    if seed is None:
+        seed = int(_np.random.randint(0, 2**63 - 1, dtype=_np.int64))
+    h = _hipops.rand(n, _F64, int(seed))
+    return DeviceArray(h, shape if shape else (), _np.float64)
+
+
+def square(x):
+    _ensure_ready()
+    d = _to_device(x)
+    if d is None:
+        return _np.square(_asarray(x))
+    return d._unary("square")
+
+
+def sum_(x):
+    _ensure_ready()
+    d = _to_device(x)
+    if d is None:
+        return _np.sum(_asarray(x))
+    return d.sum()
+
+
+def square_sum(x):
+    _ensure_ready()
+    d = _to_device(x)
+    if d is None:
+        return _np.sum(_np.square(_asarray(x)))
+    return d.square_sum()
+
+
+def matmul(a, b, _force=False):
+    """Row-major 2D matmul on the MFMA matrix cores (f32: 32x32x2 f32 MFMA,
+    f64: 16x16x4 f64 MFMA). Returns NotImplemented when the shape/dtype is
+    not routable (caller falls back)."""
+    _ensure_ready()
+    a_shape = a.shape if hasattr(a, "shape") else _np.asarray(a).shape
+    b_shape = b.shape if hasattr(b, "shape") else _np.asarray(b).shape
+    if len(a_shape) != 2 or len(b_shape) != 2 or a_shape[1] != b_shape[0]:
+        return NotImplemented
+    m, k = a_shape
+    n = b_shape[1]
+    if not _force and 2.0 * m * n * k < MIN_MATMUL_FLOPS:
+        return NotImplemented
+    da = _to_device(a)
+    db = _to_device(b)
+    if da is None or db is None or da.dtype != db.dtype:
+        return NotImplemented
+    code = _dtype_code(da.dtype)
+    hc = _hipops.gemm(da._handle, db._handle, m, n, k, code)
+    return DeviceArray(hc, (m, n), da.dtype)
+
+
+# ---------------------------------------------------------------------------
+# numpy patching
+# ---------------------------------------------------------------------------
+_installed = {"done": False}
+
+
+def install(numpy_module, mode: str = "auto") -> None:
+    """Patch numpy's hot entry points to route large work to the GPU.
+    mode="require" raises if the GPU/extension is unusable."""
+    if _installed["done"]:
+        return
+    if not available():
+        if mode == "require":
+            raise RuntimeError("APP_HIP_NUMPY=require but no AMD GPU is visible")
+        return
+
+    np = numpy_module
+    orig_rand = np.random.rand
+    orig_matmul = np.matmul
+    orig_dot = np.dot
+    orig_square = np.square
+    orig_sum = np.sum
+
+    def patched_rand(*shape):
+        n = 1
+        for s in shape:
+            n *= int(s)
+        if n >= MIN_ELEMS:
+            try:
+                return rand(*shape)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_rand(*shape)
+
+    def patched_matmul(a, b, *args, **kwargs):
+        if not args and not kwargs:
+            try:
+                r = matmul(a, b)
+                if r is not NotImplemented:
+                    return r
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_matmul(_asarray(a), _asarray(b), *args, **kwargs)
+
+    def patched_dot(a, b, *args, **kwargs):
+        if not args and not kwargs:
+            try:
+                r = matmul(a, b)
+                if r is not NotImplemented:
+                    return r
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_dot(_asarray(a), _asarray(b), *args, **kwargs)
+
+    def patched_square(x, *args, **kwargs):
+        if not args and not kwargs:
+            if isinstance(x, DeviceArray):
+                return x._unary("square")
+            arr = x if isinstance(x, _np.ndarray) else None
+            if (
+                arr is not None
+                and arr.size >= MIN_ELEMS
+                and _dtype_code(arr.dtype) is not None
+                and arr.flags.c_contiguous
+            ):
+                try:
+                    return square(arr)
+                except Exception:
+                    if mode == "require":
+                        raise
+        return orig_square(_asarray(x), *args, **kwargs)
+
+    def patched_sum(x, *args, **kwargs):
+        if not args and (not kwargs or set(kwargs) <= {"axis"}) and kwargs.get("axis") is None:
+            if isinstance(x, DeviceArray):
+                return x.sum()
+            arr = x if isinstance(x, _np.ndarray) else None
+            if (
+                arr is not None
+                and arr.size >= MIN_ELEMS
+                and _dtype_code(arr.dtype) is not None
+                and arr.flags.c_contiguous
+            ):
+                try:
+                    return sum_(arr)
+                except Exception:
+                    if mode == "require":
+                        raise
+        return orig_sum(_asarray(x), *args, **kwargs)
+
+    np.random.rand = patched_rand
+    np.matmul = patched_matmul
+    np.dot = patched_dot
+    np.square = patched_square
+    np.sum = patched_sum
+    _installed["done"] = True

@@ -1,17 +1,26 @@
 """Shared executor-side types and the client for the in-sandbox HTTP API.
 
-The internal sandbox API (served by executor/server.cpp, wire-compatible
-with the reference's in-pod API, server.rs:189-191):
+The internal sandbox API (served by executor/server.cpp):
 
-  PUT  /workspace/{rel}   upload a file
-  GET  /workspace/{rel}   download a file
-  POST /execute           {source_code, timeout?, env?} ->
-                          {stdout, stderr, exit_code, files: [abs paths]}
+  legacy pod-style (wire-compatible with the reference's in-pod API,
+  server.rs:189-191; used by the kubernetes pod backend where the pod
+  itself is single-use):
+    PUT/GET /workspace/{rel}         file staging
+    POST    /execute                 {source_code, timeout?, env?} ->
+                                     {stdout, stderr, exit_code, files:[...]}
 
-Both executor backends (local process pool, kubernetes pods) drive a
-sandbox through SandboxClient.run(): parallel-upload input files from
-storage, execute, parallel-download changed files back into storage
-(reference orchestration shape: kubernetes_code_executor.py:81-149).
+  engine-style sessions (used by the local per-GPU engine backend; the
+  single-use unit is a forked interpreter + fresh workspace, not the
+  server):
+    POST    /sandboxes               -> {id}
+    PUT/GET /sandboxes/{id}/workspace/{rel}
+    POST    /sandboxes/{id}/execute
+    DELETE  /sandboxes/{id}
+
+Both backends drive a sandbox through SandboxClient.run(): parallel-upload
+input files from storage, execute, parallel-download changed files back
+into storage (reference orchestration shape:
+kubernetes_code_executor.py:81-149).
 """
 
 import asyncio
@@ -37,8 +46,14 @@ class ExecutorError(RuntimeError):
     """Sandbox-level failure (unreachable, bad response, ...). Retryable."""
 
 
+def _rel(path: str) -> str:
+    if path.startswith(WORKSPACE_PREFIX):
+        return path[len(WORKSPACE_PREFIX):]
+    return path.lstrip("/")
+
+
 class SandboxClient:
-    """HTTP client for one sandbox (TCP base_url or unix socket)."""
+    """HTTP client for one executor server (TCP base_url or unix socket)."""
 
     def __init__(
         self,
@@ -46,7 +61,11 @@ class SandboxClient:
         uds: Optional[str] = None,
         timeout: float = 60.0,
     ):
-        transport = httpx.AsyncHTTPTransport(uds=uds) if uds else None
+        transport = (
+            httpx.AsyncHTTPTransport(uds=uds, limits=httpx.Limits(max_connections=64))
+            if uds
+            else None
+        )
         self._client = httpx.AsyncClient(
             base_url=base_url, transport=transport, timeout=timeout
         )
@@ -63,6 +82,18 @@ class SandboxClient:
             pass
         return None
 
+    async def create_sandbox(self) -> str:
+        resp = await self._client.post("/sandboxes")
+        if resp.status_code != 200:
+            raise ExecutorError(f"sandbox create failed: {resp.status_code}")
+        return resp.json()["id"]
+
+    async def delete_sandbox(self, sandbox_id: str) -> None:
+        try:
+            await self._client.delete(f"/sandboxes/{sandbox_id}")
+        except (httpx.HTTPError, OSError, RuntimeError):
+            pass  # best-effort teardown (client may already be closed)
+
     async def run(
         self,
         storage: Storage,
@@ -70,18 +101,44 @@ class SandboxClient:
         files: Mapping[str, str] = {},
         env: Mapping[str, str] = {},
         timeout: Optional[float] = None,
+        session: Optional[str] = None,
     ) -> Result:
+        """Run one execution. With ``session``, all routes are scoped to
+        that sandbox session's fresh workspace."""
+        prefix = f"/sandboxes/{session}" if session else ""
         try:
-            return await self._run(storage, source_code, files, env, timeout)
+            return await self._run(storage, source_code, files, env, timeout, prefix)
         except (httpx.HTTPError, OSError) as e:
             raise ExecutorError(f"sandbox request failed: {e!r}") from e
 
-    async def _run(self, storage, source_code, files, env, timeout) -> Result:
+    async def run_single_use(
+        self,
+        storage: Storage,
+        source_code: str,
+        files: Mapping[str, str] = {},
+        env: Mapping[str, str] = {},
+        timeout: Optional[float] = None,
+    ) -> Result:
+        """Create a fresh sandbox session, run, tear it down."""
+        try:
+            session = await self.create_sandbox()
+        except (httpx.HTTPError, OSError) as e:
+            raise ExecutorError(f"sandbox create failed: {e!r}") from e
+        try:
+            return await self.run(
+                storage, source_code, files=files, env=env, timeout=timeout,
+                session=session,
+            )
+        finally:
+            asyncio.ensure_future(self.delete_sandbox(session))
+
+    async def _run(self, storage, source_code, files, env, timeout, prefix) -> Result:
         async def upload(path: str, object_hash: str):
-            rel = path[len(WORKSPACE_PREFIX):] if path.startswith(WORKSPACE_PREFIX) else path.lstrip("/")
             async with storage.reader(object_hash) as reader:
                 data = await reader.read()
-            resp = await self._client.put(f"/workspace/{rel}", content=data)
+            resp = await self._client.put(
+                f"{prefix}/workspace/{_rel(path)}", content=data
+            )
             if resp.status_code not in (200, 204):
                 raise ExecutorError(f"upload of {path} failed: {resp.status_code}")
 
@@ -90,23 +147,22 @@ class SandboxClient:
         body: dict = {"source_code": source_code, "env": dict(env)}
         if timeout is not None:
             body["timeout"] = timeout
-        resp = await self._client.post("/execute", json=body)
+        resp = await self._client.post(f"{prefix}/execute", json=body)
         if resp.status_code != 200:
             raise ExecutorError(f"execute failed: {resp.status_code} {resp.text!r}")
         payload = resp.json()
 
         async def download(path: str):
-            rel = path[len(WORKSPACE_PREFIX):] if path.startswith(WORKSPACE_PREFIX) else path.lstrip("/")
             async with storage.writer() as writer:
-                async with self._client.stream("GET", f"/workspace/{rel}") as file_resp:
+                async with self._client.stream(
+                    "GET", f"{prefix}/workspace/{_rel(path)}"
+                ) as file_resp:
                     file_resp.raise_for_status()
                     async for chunk in file_resp.aiter_bytes():
                         await writer.write(chunk)
                 return path, writer.hash
 
-        stored = dict(
-            await asyncio.gather(*(download(p) for p in payload["files"]))
-        )
+        stored = dict(await asyncio.gather(*(download(p) for p in payload["files"])))
         return Result(
             stdout=payload["stdout"],
             stderr=payload["stderr"],

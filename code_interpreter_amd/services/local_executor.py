@@ -1,25 +1,24 @@
-"""Local-process executor backend: warm pools of single-use sandboxes fanned
-across the node's MI355X GPUs.
+"""Local executor backend: one long-lived engine per MI355X GPU, warm
+forked sandboxes inside each engine.
 
-The MI355X-native equivalent of the reference's warm Kubernetes pod queue
-(kubernetes_code_executor.py:151-264), with the pod replaced by a local
-executor-server process (executor/server.cpp + zygote) listening on a unix
-socket:
+The MI355X-native redesign of the reference's warm Kubernetes pod queue
+(kubernetes_code_executor.py:151-264). The reference's isolation/warmth
+unit is a single-use pod (seconds to spawn, so it keeps 5 warm). Here:
 
-- each sandbox is pinned to one GPU via HIP_VISIBLE_DEVICES (round-robin
-  across `gpu_count` devices), so concurrent /v1/execute requests run
-  data-parallel across the 8 GPUs of a node;
-- sandboxes are single-use: taken from the warm deque, torn down after one
-  execution, refilled asynchronously off the critical path;
-- a freshly spawned sandbox pre-imports numpy and pre-initializes HIP
-  (zygote warm child) while it waits in the pool, so those costs are never
-  on the request path;
-- whole-execution retry x3 with exponential backoff on sandbox failure
+- ENGINE: a long-lived executor-server process (executor/server.cpp +
+  zygote), one per GPU, pinned via HIP_VISIBLE_DEVICES -- concurrent
+  /v1/execute requests fan out data-parallel across the node's 8 GPUs;
+- SANDBOX: per request, a fresh workspace session plus a fresh single-use
+  interpreter forked from the engine's zygote. The zygote keeps a pool of
+  pre-forked children that have ALREADY imported numpy and initialized the
+  HIP runtime (device context + pinned staging buffers), refilled in the
+  background -- so the per-request cost is a fork handoff, not a cold
+  start (the reference pays upm+pip+xonsh per request, server.rs:126-169);
+- engines that die are respawned; whole-execution retry x3 with backoff
   (parity: kubernetes_code_executor.py:75-79).
 
-Also usable as the data plane INSIDE a GPU executor pod, where the pod-level
-scheduler is Kubernetes (services/pod_executor.py) and this pool runs with
-gpu_count=1.
+Also usable INSIDE a GPU executor pod (kubernetes backend) with
+gpu_count=1, where the pod-level scheduler is services/pod_executor.py.
 """
 
 import asyncio
@@ -32,7 +31,6 @@ import sys
 import tempfile
 import time
 import uuid
-from collections import deque
 from dataclasses import dataclass
 from pathlib import Path
 from typing import Mapping, Optional
@@ -55,28 +53,34 @@ DEFAULT_SERVER_BIN = RUNTIME_DIR / "build" / "executor-server"
 
 
 @dataclass
-class Sandbox:
+class Engine:
     proc: subprocess.Popen
     client: SandboxClient
     root: str
     gpu: Optional[int]
+    inflight: int = 0
+
+    def alive(self) -> bool:
+        return self.proc.poll() is None
 
     async def aclose(self) -> None:
         await self.client.aclose()
         try:
-            self.proc.send_signal(signal.SIGTERM)
-        except ProcessLookupError:
-            pass
+            os.killpg(self.proc.pid, signal.SIGKILL)
+        except (ProcessLookupError, PermissionError, OSError):
+            try:
+                self.proc.kill()
+            except ProcessLookupError:
+                pass
         try:
             await asyncio.to_thread(self.proc.wait, 5)
         except subprocess.TimeoutExpired:
-            self.proc.kill()
-            await asyncio.to_thread(self.proc.wait)
+            pass
         shutil.rmtree(self.root, ignore_errors=True)
 
 
 class LocalPoolExecutor:
-    """Warm pool of single-use local sandboxes, one execution each."""
+    """Per-GPU engine pool; each request runs in a fresh forked sandbox."""
 
     def __init__(
         self,
@@ -99,8 +103,8 @@ class LocalPoolExecutor:
             gpu_count = detect_gpu_count()
         self.gpu_count = gpu_count
         self.gpu_pinning = gpu_pinning and gpu_count > 0
-        # pool target scales with GPUs so every device has warm capacity
-        self.pool_target_length = pool_target_length * max(1, gpu_count)
+        # warm single-use interpreters kept ready inside each engine
+        self.warm_children = max(1, pool_target_length)
         self.executor_root = executor_root or tempfile.mkdtemp(prefix="ci-amd-")
         self.server_bin = server_bin or str(DEFAULT_SERVER_BIN)
         self.execute_timeout = execute_timeout
@@ -111,57 +115,48 @@ class LocalPoolExecutor:
         self.hip_numpy = hip_numpy
         self.spawn_ready_timeout = spawn_ready_timeout
 
-        self._pool: deque[Sandbox] = deque()
-        self._spawning_count = 0
-        self._next_gpu = 0
+        self.n_engines = max(1, gpu_count)
+        self._engines: list[Optional[Engine]] = [None] * self.n_engines
+        self._spawn_locks = [asyncio.Lock() for _ in range(self.n_engines)]
+        self._rr = 0
         self._closed = False
 
     # -- lifecycle ---------------------------------------------------------
 
     async def fill_pool(self) -> None:
-        """Top the warm pool up to the target length (async, off the
-        request path; parity: fill_executor_pod_queue,
-        kubernetes_code_executor.py:151-189)."""
-        while (
-            not self._closed
-            and len(self._pool) + self._spawning_count < self.pool_target_length
-        ):
-            self._spawning_count += 1
-            try:
-                sandbox = await self.spawn_sandbox()
-                self._pool.append(sandbox)
-            except Exception as e:
-                logger.warning("sandbox prewarm failed: %s", e)
-                await asyncio.sleep(1.0)
-                return
-            finally:
-                self._spawning_count -= 1
+        """Bring up one engine per GPU (concurrently); called at startup
+        and after engine failures, off the request path."""
+        await asyncio.gather(
+            *(self._ensure_engine(i) for i in range(self.n_engines)),
+            return_exceptions=True,
+        )
 
-    def _pick_gpu(self) -> Optional[int]:
-        if not self.gpu_pinning:
-            return None
-        gpu = self._next_gpu % self.gpu_count
-        self._next_gpu += 1
-        return gpu
+    async def _ensure_engine(self, idx: int) -> Engine:
+        async with self._spawn_locks[idx]:
+            engine = self._engines[idx]
+            if engine is not None and engine.alive():
+                return engine
+            if engine is not None:
+                await engine.aclose()
+                self._engines[idx] = None
+            if self._closed:
+                raise ExecutorError("executor closed")
+            gpu = idx % self.gpu_count if self.gpu_pinning else None
+            engine = await self._spawn_engine(gpu)
+            self._engines[idx] = engine
+            return engine
 
-    async def spawn_sandbox(self) -> Sandbox:
-        async def attempt() -> Sandbox:
-            return await self._spawn_once()
-
-        return await async_retry(attempt, attempts=3, retry_on=(ExecutorError,),
-                                 min_backoff=0.5, max_backoff=2.0)
-
-    async def _spawn_once(self) -> Sandbox:
+    async def _spawn_engine(self, gpu: Optional[int]) -> Engine:
         if not os.path.exists(self.server_bin):
             raise FileNotFoundError(
                 f"executor-server binary not found at {self.server_bin}; "
-                "build it with `python -m code_interpreter_amd.ops.build` "
-                "or `make -C code_interpreter_amd/executor`"
+                "build it with `make -C code_interpreter_amd/executor`"
             )
-        gpu = self._pick_gpu()
-        root = os.path.join(self.executor_root, f"sbx-{uuid.uuid4().hex[:12]}")
+        root = os.path.join(self.executor_root, f"eng-{uuid.uuid4().hex[:12]}")
         workspace = os.path.join(root, "workspace")
+        sessions = os.path.join(root, "sandboxes")
         os.makedirs(workspace, exist_ok=True)
+        os.makedirs(sessions, exist_ok=True)
         sock = os.path.join(root, "exec.sock")
 
         env = dict(os.environ)
@@ -169,10 +164,12 @@ class LocalPoolExecutor:
             {
                 "APP_LISTEN_UNIX": sock,
                 "APP_WORKSPACE": workspace,
+                "APP_SESSIONS_DIR": sessions,
                 "APP_PYTHON": sys.executable,
                 "APP_RUNTIME_DIR": str(RUNTIME_DIR),
                 "APP_OPS_DIR": str(OPS_DIR),
                 "APP_ZYGOTE": "1" if self.zygote_enabled else "0",
+                "APP_WARM_CHILDREN": str(self.warm_children),
                 "APP_SCAN_RECURSIVE": "1" if self.scan_recursive else "0",
                 "APP_DEP_INSTALL": "1" if self.dep_install else "0",
                 "APP_PIP_EXTRA_ARGS": self.pip_extra_args,
@@ -199,7 +196,8 @@ class LocalPoolExecutor:
                     f"executor-server exited early (code {proc.returncode})"
                 )
             if await client.healthy() is not None:
-                return Sandbox(proc=proc, client=client, root=root, gpu=gpu)
+                logger.info("engine up (gpu=%s, root=%s)", gpu, root)
+                return Engine(proc=proc, client=client, root=root, gpu=gpu)
             await asyncio.sleep(0.05)
         proc.kill()
         await client.aclose()
@@ -208,11 +206,28 @@ class LocalPoolExecutor:
 
     async def aclose(self) -> None:
         self._closed = True
-        while self._pool:
-            await self._pool.popleft().aclose()
+        for engine in self._engines:
+            if engine is not None:
+                await engine.aclose()
+        self._engines = [None] * self.n_engines
         shutil.rmtree(self.executor_root, ignore_errors=True)
 
     # -- execution ---------------------------------------------------------
+
+    def _pick_engine_idx(self) -> int:
+        # least-loaded live engine; round-robin tiebreak
+        best, best_load = None, None
+        n = self.n_engines
+        for off in range(n):
+            i = (self._rr + off) % n
+            e = self._engines[i]
+            load = e.inflight if e is not None and e.alive() else 0
+            if best_load is None or load < best_load:
+                best, best_load = i, load
+                if load == 0:
+                    break
+        self._rr = (best + 1) % n
+        return best
 
     async def execute(
         self,
@@ -221,23 +236,22 @@ class LocalPoolExecutor:
         env: Mapping[str, str] = {},
     ) -> Result:
         async def attempt() -> Result:
-            sandbox = await self._take_sandbox()
+            idx = self._pick_engine_idx()
+            engine = await self._ensure_engine(idx)
+            engine.inflight += 1
             try:
-                return await sandbox.client.run(
+                return await engine.client.run_single_use(
                     self.file_storage,
                     source_code,
                     files=files,
                     env=env,
                     timeout=self.execute_timeout,
                 )
+            except ExecutorError:
+                if not engine.alive():
+                    asyncio.ensure_future(self._ensure_engine(idx))
+                raise
             finally:
-                # single-use teardown + async refill, off the critical path
-                asyncio.create_task(sandbox.aclose())
-                asyncio.create_task(self.fill_pool())
+                engine.inflight -= 1
 
         return await async_retry(attempt, attempts=3, retry_on=(ExecutorError,))
-
-    async def _take_sandbox(self) -> Sandbox:
-        if self._pool:
-            return self._pool.popleft()
-        return await self.spawn_sandbox()

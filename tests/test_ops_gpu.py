@@ -1,0 +1,248 @@
+"""GPU numerics tests for the gfx950 kernel library: every HIP kernel is
+checked against a plain numpy (fp64/fp32) reference on random data, with
+transpose-detecting inputs for the GEMMs."""
+
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+OPS_DIR = Path(__file__).resolve().parent.parent / "code_interpreter_amd" / "ops"
+sys.path.insert(0, str(OPS_DIR))
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def hip():
+    import _hipops
+
+    if not _hipops.is_available():
+        pytest.skip("no AMD GPU visible")
+    _hipops.init(0)
+    return _hipops
+
+
+@pytest.fixture(scope="module")
+def hnp(hip):
+    import hipnp
+
+    return hipnp
+
+
+def f32_to_bf16_np(a: np.ndarray) -> np.ndarray:
+    u = a.astype(np.float32).view(np.uint32)
+    lsb = (u >> 16) & 1
+    return ((u + 0x7FFF + lsb) >> 16).astype(np.uint16)
+
+
+def bf16_to_f32_np(b: np.ndarray) -> np.ndarray:
+    return (b.astype(np.uint32) << 16).view(np.float32)
+
+
+# ---------------------------------------------------------------------------
+# elementwise + reduce + rng
+# ---------------------------------------------------------------------------
+def test_upload_download_roundtrip(hip):
+    rng = np.random.default_rng(0)
+    a = rng.standard_normal(1_000_003)  # odd size exercises the tail
+    h = hip.upload(a)
+    out = np.empty_like(a)
+    hip.download(h, out)
+    hip.free(h)
+    np.testing.assert_array_equal(a, out)
+
+
+def test_large_pinned_staged_upload(hip):
+    # > 2x the 32 MB staging buffer: exercises the double-buffer path
+    rng = np.random.default_rng(1)
+    a = rng.standard_normal(12_000_000)  # 96 MB
+    h = hip.upload(a)
+    out = np.empty_like(a)
+    hip.download(h, out)
+    hip.free(h)
+    np.testing.assert_array_equal(a, out)
+
+
+@pytest.mark.parametrize("dtype,code", [(np.float64, 1), (np.float32, 0)])
+def test_square_matches_numpy(hip, dtype, code):
+    rng = np.random.default_rng(2)
+    a = rng.standard_normal(500_001).astype(dtype) * 3
+    h = hip.upload(a)
+    h2 = hip.unary(h, 0, code, a.size)
+    out = np.empty_like(a)
+    hip.download(h2, out)
+    hip.free(h)
+    hip.free(h2)
+    np.testing.assert_allclose(out, np.square(a), rtol=0)
+
+
+@pytest.mark.parametrize("op,npop", [(0, np.add), (1, np.subtract), (2, np.multiply)])
+def test_binary_matches_numpy(hip, op, npop):
+    rng = np.random.default_rng(3)
+    a = rng.standard_normal(100_001)
+    b = rng.standard_normal(100_001) + 2.0
+    ha, hb = hip.upload(a), hip.upload(b)
+    hc = hip.binary(ha, hb, op, 1, a.size)
+    out = np.empty_like(a)
+    hip.download(hc, out)
+    for h in (ha, hb, hc):
+        hip.free(h)
+    np.testing.assert_array_equal(out, npop(a, b))
+
+
+def test_sum_matches_numpy(hip):
+    rng = np.random.default_rng(4)
+    a = rng.standard_normal(10_000_001)
+    h = hip.upload(a)
+    s = hip.sum(h, 1, a.size, 0)
+    sq = hip.sum(h, 1, a.size, 1)
+    hip.free(h)
+    np.testing.assert_allclose(s, a.sum(), rtol=1e-12)
+    np.testing.assert_allclose(sq, np.square(a).sum(), rtol=1e-12)
+
+
+def test_rand_statistics(hip):
+    n = 10_000_000
+    h = hip.rand(n, 1, 424242)
+    out = np.empty(n)
+    hip.download(h, out)
+    hip.free(h)
+    assert out.min() >= 0.0 and out.max() < 1.0
+    np.testing.assert_allclose(out.mean(), 0.5, atol=2e-3)
+    np.testing.assert_allclose(out.var(), 1.0 / 12, atol=2e-3)
+    # counter-based: consecutive draws must differ
+    assert np.unique(out[:1000]).size > 990
+
+
+def test_rand_streams_differ(hip):
+    h1 = hip.rand(1000, 1, 111)
+    h2 = hip.rand(1000, 1, 222)
+    a, b = np.empty(1000), np.empty(1000)
+    hip.download(h1, a)
+    hip.download(h2, b)
+    hip.free(h1)
+    hip.free(h2)
+    assert not np.array_equal(a, b)
+
+
+# ---------------------------------------------------------------------------
+# GEMMs (transpose-detecting: asymmetric random inputs + identity checks)
+# ---------------------------------------------------------------------------
+def _gemm_host(hip, a, b, code, out_dtype):
+    m, k = a.shape
+    n = b.shape[1]
+    ha, hb = hip.upload(np.ascontiguousarray(a)), hip.upload(np.ascontiguousarray(b))
+    hc = hip.gemm(ha, hb, m, n, k, code)
+    out = np.empty((m, n), dtype=out_dtype)
+    hip.download(hc, out)
+    for h in (ha, hb, hc):
+        hip.free(h)
+    return out
+
+
+@pytest.mark.parametrize("shape", [(256, 256, 256), (300, 130, 70), (129, 257, 65)])
+def test_gemm_f32(hip, shape):
+    m, n, k = shape
+    rng = np.random.default_rng(5)
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    b = rng.uniform(-1, 1, (k, n)).astype(np.float32)
+    c = _gemm_host(hip, a, b, 0, np.float32)
+    ref = a.astype(np.float64) @ b.astype(np.float64)
+    np.testing.assert_allclose(c, ref, rtol=1e-5, atol=1e-4 * np.sqrt(k))
+
+
+def test_gemm_f32_identity_asymmetric(hip):
+    # A = I with an ASYMMETRIC B catches row/col-swapped C writes
+    n = 128
+    a = np.eye(n, dtype=np.float32)
+    b = np.arange(n * n, dtype=np.float32).reshape(n, n) / (n * n)
+    c = _gemm_host(hip, a, b, 0, np.float32)
+    np.testing.assert_allclose(c, b, rtol=1e-6)
+
+
+@pytest.mark.parametrize("shape", [(256, 256, 256), (192, 100, 35)])
+def test_gemm_f64(hip, shape):
+    m, n, k = shape
+    rng = np.random.default_rng(6)
+    a = rng.uniform(-1, 1, (m, k))
+    b = rng.uniform(-1, 1, (k, n))
+    c = _gemm_host(hip, a, b, 1, np.float64)
+    np.testing.assert_allclose(c, a @ b, rtol=1e-13, atol=1e-12 * k)
+
+
+def test_gemm_f64_identity_asymmetric(hip):
+    n = 64
+    a = np.eye(n)
+    b = np.arange(n * n, dtype=np.float64).reshape(n, n)
+    c = _gemm_host(hip, a, b, 1, np.float64)
+    np.testing.assert_allclose(c, b)
+
+
+@pytest.mark.parametrize("shape", [(256, 256, 256), (128, 256, 96)])
+def test_gemm_bf16(hip, shape):
+    m, n, k = shape
+    rng = np.random.default_rng(7)
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    b = rng.uniform(-1, 1, (k, n)).astype(np.float32)
+    a_bf = f32_to_bf16_np(a)
+    b_bf = f32_to_bf16_np(b)
+    c_bf = _gemm_host(hip, a_bf, b_bf, 2, np.uint16)
+    c = bf16_to_f32_np(c_bf)
+    # reference: same quantized inputs, f32 accumulate, then bf16 round
+    ref = bf16_to_f32_np(a_bf).astype(np.float64) @ bf16_to_f32_np(b_bf).astype(
+        np.float64
+    )
+    np.testing.assert_allclose(c, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k))
+
+
+def test_gemm_bf16_identity_asymmetric(hip):
+    n = 128
+    a = f32_to_bf16_np(np.eye(n, dtype=np.float32))
+    b_f = (np.arange(n * n, dtype=np.float32).reshape(n, n) % 251) / 256.0
+    b = f32_to_bf16_np(b_f)
+    c = bf16_to_f32_np(_gemm_host(hip, a, b, 2, np.uint16))
+    np.testing.assert_allclose(c, bf16_to_f32_np(b), rtol=1e-2, atol=1e-2)
+
+
+# ---------------------------------------------------------------------------
+# hipnp DeviceArray + numpy routing
+# ---------------------------------------------------------------------------
+def test_device_array_square_sum_chain(hnp):
+    x = hnp.rand(1_000_000, seed=99)
+    s = np.sum(np.square(x))  # dispatches via __array_function__/__array_ufunc__
+    host = np.asarray(x)
+    np.testing.assert_allclose(float(s), np.square(host).sum(), rtol=1e-10)
+
+
+def test_device_array_fused_square_sum(hnp):
+    x = hnp.rand(2_000_000, seed=7)
+    np.testing.assert_allclose(
+        float(x.square_sum()), np.square(np.asarray(x)).sum(), rtol=1e-10
+    )
+
+
+def test_device_array_fallback(hnp):
+    x = hnp.rand(10_000, seed=1)
+    # unsupported numpy op: transparently materializes
+    med = np.median(x)
+    assert 0.4 < float(med) < 0.6
+
+
+def test_device_array_arith(hnp):
+    x = hnp.rand(100_000, seed=2)
+    y = (x * 2.0 + 1.0) - x
+    np.testing.assert_allclose(
+        np.asarray(y), np.asarray(x) + 1.0, rtol=1e-12
+    )
+
+
+def test_hipnp_matmul_matches_numpy(hnp):
+    rng = np.random.default_rng(8)
+    a = rng.uniform(-1, 1, (512, 384)).astype(np.float32)
+    b = rng.uniform(-1, 1, (384, 256)).astype(np.float32)
+    c = hnp.matmul(a, b, _force=True)
+    np.testing.assert_allclose(
+        np.asarray(c), a.astype(np.float64) @ b.astype(np.float64), rtol=1e-4, atol=1e-3
+    )
