@@ -1,0 +1,261 @@
+"""Flagship serving benchmark: exec requests/sec + p50 latency for the
+reference's benchmark-numpy.py workload through the full service stack.
+
+Per BASELINE.json: each "step" is one POST /v1/execute of the
+benchmark-numpy workload (10^8-element uniform rand -> square -> sum;
+reference examples/benchmark-numpy.py:16-28), served over real HTTP by the
+FastAPI control plane and executed in a fresh single-use GPU sandbox
+(engine + forked warm child), with the numpy compute routed to the gfx950
+HIP kernels (Philox RNG, fused square+sum reduction).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched by torch.distributed.run with one rank per GPU; each rank
+runs its own service instance pinned to HIP device LOCAL_RANK and the
+printed value is the whole-job aggregate requests/sec (max-over-ranks
+elapsed). Rank 0 prints exactly one JSON line.
+"""
+
+import argparse
+import asyncio
+import json
+import os
+import socket
+import sys
+import tempfile
+import threading
+import time
+from pathlib import Path
+
+REPO_ROOT = Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO_ROOT))
+
+WORKLOAD = """
+import numpy
+import time
+
+def compute():
+    array_size = {array_size}
+    large_array = numpy.random.rand(array_size)
+    result = numpy.sum(numpy.square(large_array))
+    return result
+
+start_time = time.time()
+result = compute()
+end_time = time.time()
+print("Result:", result)
+print("Execution Time:", end_time - start_time, "seconds")
+"""
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _percentile(sorted_vals, p):
+    if not sorted_vals:
+        return 0.0
+    idx = min(len(sorted_vals) - 1, int(round(p / 100.0 * (len(sorted_vals) - 1))))
+    return sorted_vals[idx]
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=16)
+    parser.add_argument("--warmup", type=int, default=4)
+    parser.add_argument("--concurrency", type=int, default=4)
+    parser.add_argument("--array-size", type=int, default=10**8)
+    parser.add_argument("--workload", default="benchmark-numpy.py")
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+
+    import torch
+
+    use_gpu = torch.cuda.is_available()
+    distributed = world_size > 1
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+
+    # pin this rank's service (and its sandbox engines) to one GPU
+    if use_gpu:
+        os.environ["HIP_VISIBLE_DEVICES"] = os.environ.get(
+            "HIP_VISIBLE_DEVICES", str(local_rank)
+        )
+
+    result = asyncio.run(run_rank(args, rank, world_size, use_gpu))
+
+    if distributed:
+        import torch.distributed as dist
+
+        elapsed = torch.tensor([result["elapsed"]], dtype=torch.float64)
+        if use_gpu:
+            elapsed = elapsed.cuda()
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+        result["elapsed"] = float(elapsed.item())
+        lat = torch.tensor(
+            [result["p50_ms"], result["p95_ms"]], dtype=torch.float64
+        )
+        if use_gpu:
+            lat = lat.cuda()
+        dist.all_reduce(lat, op=dist.ReduceOp.SUM)
+        result["p50_ms"] = float(lat[0].item()) / world_size
+        result["p95_ms"] = float(lat[1].item()) / world_size
+
+    if rank == 0:
+        total_requests = args.steps * world_size
+        value = total_requests / result["elapsed"]
+        print(
+            json.dumps(
+                {
+                    "metric": "exec requests/sec (benchmark-numpy.py)",
+                    "value": round(value, 3),
+                    "unit": "req/s",
+                    "n_gpus": world_size if use_gpu else 0,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(result["elapsed"] / args.steps * 1000, 3),
+                    "p50_ms": round(result["p50_ms"], 2),
+                    "p95_ms": round(result["p95_ms"], 2),
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "float64",
+                    "data": "synthetic",
+                    "config": {
+                        "model": args.workload,
+                        "array_size": args.array_size,
+                        "concurrency": args.concurrency,
+                        "parallelism": f"dp{world_size}",
+                        "sandbox": "fresh single-use interpreter per request",
+                        "compute": "hip" if use_gpu else "cpu",
+                    },
+                }
+            ),
+            flush=True,
+        )
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
+    import httpx
+    import torch
+    import uvicorn
+
+    from code_interpreter_amd.application_context import ApplicationContext
+    from code_interpreter_amd.config import Config
+
+    tmp = tempfile.mkdtemp(prefix=f"bench-r{rank}-")
+    port = _free_port()
+    config = Config(
+        http_listen_addr=f"127.0.0.1:{port}",
+        file_storage_path=os.path.join(tmp, "storage"),
+        executor_root=os.path.join(tmp, "executors"),
+        executor_backend="local",
+        executor_pool_target_length=max(2, args.concurrency),
+        gpu_count=1 if use_gpu else 0,
+        gpu_pinning=False,  # engines inherit this rank's HIP_VISIBLE_DEVICES
+        hip_numpy="require" if use_gpu else "off",
+        dep_install=False,
+    )
+    ctx = ApplicationContext(config)
+
+    server = uvicorn.Server(
+        uvicorn.Config(
+            app=ctx.http_server, host="127.0.0.1", port=port, log_config=None
+        )
+    )
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+
+    source = WORKLOAD.format(array_size=args.array_size)
+    async with httpx.AsyncClient(
+        base_url=f"http://127.0.0.1:{port}", timeout=300.0
+    ) as client:
+        # wait for the service + at least one engine
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            try:
+                r = await client.post(
+                    "/v1/execute", json={"source_code": "print('ready')"}
+                )
+                if r.status_code == 200 and r.json()["exit_code"] == 0:
+                    break
+            except httpx.HTTPError:
+                pass
+            await asyncio.sleep(0.25)
+        else:
+            raise RuntimeError("service did not become ready")
+
+        async def one_request() -> float:
+            t0 = time.perf_counter()
+            resp = await client.post("/v1/execute", json={"source_code": source})
+            dt = time.perf_counter() - t0
+            body = resp.json()
+            if resp.status_code != 200 or body["exit_code"] != 0:
+                raise RuntimeError(
+                    f"execute failed: {resp.status_code} "
+                    f"{body.get('stderr', '')[:500]}"
+                )
+            if "Result:" not in body["stdout"]:
+                raise RuntimeError(f"unexpected stdout: {body['stdout'][:200]}")
+            return dt
+
+        async def run_phase(n: int) -> list:
+            sem = asyncio.Semaphore(args.concurrency)
+            latencies = []
+
+            async def guarded():
+                async with sem:
+                    latencies.append(await one_request())
+
+            await asyncio.gather(*(guarded() for _ in range(n)))
+            return latencies
+
+        # warmup (untimed)
+        await run_phase(args.warmup)
+
+        # timed region, bracketed by barrier + device sync on both sides
+        if world_size > 1:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+        t_start = time.perf_counter()
+        latencies = await run_phase(args.steps)
+        if use_gpu:
+            torch.cuda.synchronize()
+        if world_size > 1:
+            import torch.distributed as dist
+
+            dist.barrier()
+        elapsed = time.perf_counter() - t_start
+
+    server.should_exit = True
+    thread.join(timeout=10)
+    await ctx.code_executor.aclose()
+
+    latencies.sort()
+    return {
+        "elapsed": elapsed,
+        "p50_ms": _percentile(latencies, 50) * 1000,
+        "p95_ms": _percentile(latencies, 95) * 1000,
+    }
+
+
+if __name__ == "__main__":
+    main()
