@@ -101,7 +101,13 @@ def _maybe_install_hip_numpy() -> None:
     if _hipnp_state["attempted"]:
         return
     mode = os.environ.get("APP_HIP_NUMPY", "auto").lower()
-    if mode == "off" or "numpy" not in sys.modules:
+    if mode == "off":
+        return
+    # the import hook also fires for numpy-internal imports while numpy is
+    # still initializing (e.g. `from numpy import dtypes`); installing then
+    # would import a partially initialized numpy into hipnp
+    np_module = sys.modules.get("numpy")
+    if np_module is None or not hasattr(np_module, "ndarray"):
         return
     _hipnp_state["attempted"] = True
     ops_dir = os.environ.get("APP_OPS_DIR")
