@@ -107,7 +107,9 @@ def _maybe_install_hip_numpy() -> None:
     # still initializing (e.g. `from numpy import dtypes`); installing then
     # would import a partially initialized numpy into hipnp
     np_module = sys.modules.get("numpy")
-    if np_module is None or not hasattr(np_module, "ndarray"):
+    if np_module is None or not all(
+        hasattr(np_module, attr) for attr in ("ndarray", "random", "matmul", "sum")
+    ):
         return
     _hipnp_state["attempted"] = True
     ops_dir = os.environ.get("APP_OPS_DIR")

@@ -11,7 +11,9 @@
 // Operand layout for mfma_f64_16x16x4_f64 (one f64 per lane for A/B):
 //   A: lane l supplies A[i = l&15][k = l>>4]
 //   B: lane l supplies B[k = l>>4][j = l&15]
-//   C/D (4 regs): col = lane&15, row = 4*(lane>>4) + reg
+//   C/D (4 regs): col = lane&15, row = (lane>>4) + 4*reg
+// (empirically verified on gfx950 with scripts/mfma_f64_probe.cpp -- note
+// the f64 C/D row map differs from the bf16/f16 16x16 map)
 
 #include "common.h"
 
@@ -93,7 +95,7 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
     __syncthreads();
   }
 
-  const int crow0 = row0 + wave_m * 32 + 4 * (lane >> 4);
+  const int crow0 = row0 + wave_m * 32 + (lane >> 4);
   const int ccol0 = col0 + wave_n * 32 + l15;
 #pragma unroll
   for (int mt = 0; mt < 2; mt++) {
@@ -103,7 +105,7 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
       if (col >= N) continue;
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
-        int row = crow0 + mt * 16 + reg;
+        int row = crow0 + mt * 16 + 4 * reg;
         if (row < M) C[(int64_t)row * N + col] = acc[mt][nt][reg];
       }
     }
