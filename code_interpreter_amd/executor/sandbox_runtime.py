@@ -70,8 +70,17 @@ def _patched_import(name, globals=None, locals=None, fromlist=(), level=0):
 
             clip.write_videofile = quiet_write
             clip._ci_amd_quiet = True
-    elif name == "numpy":
-        _maybe_install_hip_numpy()
+    elif name == "numpy" or name.startswith("numpy."):
+        # only act on imports from OUTSIDE numpy: by the time a non-numpy
+        # caller's `import numpy` returns, numpy is fully initialized
+        # (numpy-internal imports fire this hook mid-init, and numpy 2.x
+        # lazy-loads submodules via module __getattr__, so attribute
+        # probing from inside the hook would re-enter the importer)
+        caller = ""
+        if isinstance(globals, dict):
+            caller = globals.get("__name__") or ""
+        if not caller.startswith("numpy"):
+            _maybe_install_hip_numpy()
 
     return module
 
@@ -107,9 +116,9 @@ def _maybe_install_hip_numpy() -> None:
     # still initializing (e.g. `from numpy import dtypes`); installing then
     # would import a partially initialized numpy into hipnp
     np_module = sys.modules.get("numpy")
-    if np_module is None or not all(
-        hasattr(np_module, attr) for attr in ("ndarray", "random", "matmul", "sum")
-    ):
+    # __dict__ lookup only: hasattr would trigger numpy's lazy submodule
+    # __getattr__ re-entrantly
+    if np_module is None or "ndarray" not in np_module.__dict__:
         return
     _hipnp_state["attempted"] = True
     ops_dir = os.environ.get("APP_OPS_DIR")
