@@ -116,9 +116,14 @@ def _maybe_install_hip_numpy() -> None:
     # still initializing (e.g. `from numpy import dtypes`); installing then
     # would import a partially initialized numpy into hipnp
     np_module = sys.modules.get("numpy")
-    # __dict__ lookup only: hasattr would trigger numpy's lazy submodule
-    # __getattr__ re-entrantly
-    if np_module is None or "ndarray" not in np_module.__dict__:
+    # __dict__ lookups only (hasattr would trigger numpy's lazy submodule
+    # __getattr__ re-entrantly). "test" is assigned at the tail of
+    # numpy/__init__.py, so its presence means numpy finished initializing
+    # (C extensions calling import_array() re-enter the import hook with
+    # globals=None mid-init, which the caller filter cannot catch).
+    if np_module is None or not (
+        "ndarray" in np_module.__dict__ and "test" in np_module.__dict__
+    ):
         return
     _hipnp_state["attempted"] = True
     ops_dir = os.environ.get("APP_OPS_DIR")
