@@ -90,6 +90,19 @@ void pool_free(void* p, int64_t size) {
   g.free_list[round_size(size)].push_back(p);
 }
 
+void ensure_staging() {
+  if (g.pin[0]) return;
+  for (int i = 0; i < 2; i++) {
+    HIP_CHECK(hipHostMalloc(&g.pin[i], State::kStage, hipHostMallocDefault));
+    HIP_CHECK(hipEventCreateWithFlags(&g.pin_evt[i], hipEventDisableTiming));
+  }
+}
+
+void ensure_reduce_scratch() {
+  if (g.reduce_scratch) return;
+  HIP_CHECK(hipMalloc(&g.reduce_scratch, 4096 * sizeof(double)));
+}
+
 uint64_t register_buf(void* p, int64_t size) {
   uint64_t h = g.next_handle++;
   g.bufs[h] = {p, size};
@@ -141,11 +154,9 @@ PyObject* py_init(PyObject*, PyObject* args) {
   HIP_CHECK(hipSetDevice(device));
   HIP_CHECK(hipStreamCreateWithFlags(&g.compute, hipStreamNonBlocking));
   HIP_CHECK(hipStreamCreateWithFlags(&g.copy, hipStreamNonBlocking));
-  for (int i = 0; i < 2; i++) {
-    HIP_CHECK(hipHostMalloc(&g.pin[i], State::kStage, hipHostMallocDefault));
-    HIP_CHECK(hipEventCreateWithFlags(&g.pin_evt[i], hipEventDisableTiming));
-  }
-  HIP_CHECK(hipMalloc(&g.reduce_scratch, 4096 * sizeof(double)));
+  // pinned staging + reduce scratch are allocated lazily on first use:
+  // hipHostMalloc of 64 MB costs tens of ms and sandbox prewarm latency
+  // bounds the service's sustained request throughput
   HIP_CHECK(hipMalloc(&g.scalar_dev, sizeof(double)));
   HIP_CHECK(hipHostMalloc((void**)&g.scalar_pin, sizeof(double),
                           hipHostMallocDefault));
@@ -192,6 +203,7 @@ PyObject* py_upload(PyObject*, PyObject* args) {
   int64_t nbytes = view.len;
   void* dev = nullptr;
   Py_BEGIN_ALLOW_THREADS;
+  ensure_staging();
   dev = pool_alloc(nbytes);
   const char* src = (const char*)view.buf;
   int64_t off = 0;
@@ -232,6 +244,7 @@ PyObject* py_download(PyObject*, PyObject* args) {
     throw std::runtime_error("download target too small");
   }
   Py_BEGIN_ALLOW_THREADS;
+  ensure_staging();
   HIP_CHECK(hipStreamSynchronize(g.compute));
   char* dst = (char*)view.buf;
   int64_t off = 0;
@@ -353,6 +366,7 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   DevBuf& in = get_buf(h);
   double result = 0;
   Py_BEGIN_ALLOW_THREADS;
+  ensure_reduce_scratch();
   launch_sum(dtype, square != 0, in.ptr, g.reduce_scratch, g.scalar_dev, n,
              g.compute);
   HIP_CHECK(hipMemcpyAsync(g.scalar_pin, g.scalar_dev, 8,
