@@ -1069,6 +1069,23 @@ int main(int, char**) {
   g_sessions_root = env_or("APP_SESSIONS_DIR", "/tmp/sandboxes");
   mkdirs(g_sessions_root);
 
+  // GPU compute daemon: ONE HIP context per engine (per-process context
+  // creation serializes in the driver); sandbox children RPC to it via
+  // hipnp's remote backend. If there is no GPU the daemon exits(3), the
+  // socket never appears, and children fall back (ops/hipnp.py).
+  std::string ops_dir = env_or("APP_OPS_DIR", "");
+  if (env_or("APP_HIP_DAEMON", "1") != "0" && !ops_dir.empty()) {
+    std::string gpu_sock = g_sessions_root + "/gpu.sock";
+    pid_t pid = fork();
+    if (pid == 0) {
+      std::string script = ops_dir + "/hipd.py";
+      execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), "-u", script.c_str(),
+             "--socket", gpu_sock.c_str(), (char*)nullptr);
+      _exit(127);
+    }
+    if (pid > 0) setenv("APP_GPU_SERVICE", gpu_sock.c_str(), 1);
+  }
+
   if (g_cfg.zygote) {
     g_zygote = new Zygote();
     if (!g_zygote->start()) {

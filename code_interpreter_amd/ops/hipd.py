@@ -1,0 +1,177 @@
+"""Per-engine GPU compute daemon.
+
+Holds the engine's ONE HIP context (per-process context creation
+serializes in the kernel driver at ~13/s -- measured, see
+profiles/NOTES.md), a shared device-memory free-list, and the pinned
+staging buffers. Sandbox children don't own a GPU context: their numpy
+ops RPC here over a unix socket (hipnp's remote backend), so a fresh
+sandbox costs a fork, and repeated same-shape allocations across requests
+hit the warm pool.
+
+One thread per connection; _hipops releases the GIL around device work.
+All handles opened by a connection are freed when it closes (sandbox
+exit == resource cleanup).
+
+Protocol (little-endian): [u32 header_len][json header][payload bytes]
+both directions; header carries "plen" when a payload follows.
+
+Usage: python hipd.py --socket PATH   (exits 3 if no GPU is visible)
+"""
+
+import argparse
+import json
+import os
+import socket
+import struct
+import sys
+import threading
+
+OPS_DIR = os.path.dirname(os.path.abspath(__file__))
+if OPS_DIR not in sys.path:
+    sys.path.insert(0, OPS_DIR)
+
+import _hipops
+
+
+def read_exact(sock: socket.socket, n: int) -> bytes:
+    buf = bytearray(n)
+    view = memoryview(buf)
+    got = 0
+    while got < n:
+        r = sock.recv_into(view[got:], n - got)
+        if r == 0:
+            raise ConnectionError("peer closed")
+        got += r
+    return bytes(buf)
+
+
+def recv_msg(sock):
+    (hlen,) = struct.unpack("<I", read_exact(sock, 4))
+    header = json.loads(read_exact(sock, hlen))
+    payload = b""
+    plen = header.get("plen", 0)
+    if plen:
+        payload = read_exact(sock, plen)
+    return header, payload
+
+
+def send_msg(sock, header: dict, payload: bytes = b"") -> None:
+    if payload:
+        header = {**header, "plen": len(payload)}
+    hb = json.dumps(header).encode()
+    sock.sendall(struct.pack("<I", len(hb)) + hb + payload)
+
+
+class Connection(threading.Thread):
+    def __init__(self, sock):
+        super().__init__(daemon=True)
+        self.sock = sock
+        self.handles = set()
+
+    def run(self):
+        try:
+            while True:
+                header, payload = recv_msg(self.sock)
+                try:
+                    resp, out_payload = self.dispatch(header, payload)
+                except Exception as e:  # per-op errors back to client
+                    resp, out_payload = {"ok": False, "error": str(e)}, b""
+                send_msg(self.sock, resp, out_payload)
+        except (ConnectionError, OSError):
+            pass
+        finally:
+            for h in self.handles:
+                try:
+                    _hipops.free(h)
+                except Exception:
+                    pass
+            try:
+                self.sock.close()
+            except OSError:
+                pass
+
+    def dispatch(self, m: dict, payload: bytes):
+        op = m["op"]
+        if op == "ping":
+            return {"ok": True}, b""
+        if op == "upload":
+            h = _hipops.upload(payload)
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "download":
+            out = bytearray(m["nbytes"])
+            _hipops.download(m["h"], out)
+            return {"ok": True}, bytes(out)
+        if op == "alloc":
+            h = _hipops.alloc(m["nbytes"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "free":
+            self.handles.discard(m["h"])
+            _hipops.free(m["h"])
+            return {"ok": True}, b""
+        if op == "rand":
+            h = _hipops.rand(m["n"], m["dtype"], m["seed"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "unary":
+            h = _hipops.unary(m["h"], m["uop"], m["dtype"], m["n"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "binary":
+            h = _hipops.binary(m["ha"], m["hb"], m["bop"], m["dtype"], m["n"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "binary_scalar":
+            h = _hipops.binary_scalar(
+                m["h"], m["scalar"], m["bop"], m["dtype"], m["n"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "sum":
+            v = _hipops.sum(m["h"], m["dtype"], m["n"], m["square"])
+            return {"ok": True, "value": v}, b""
+        if op == "gemm":
+            h = _hipops.gemm(m["ha"], m["hb"], m["m"], m["n"], m["k"], m["dtype"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "sync":
+            _hipops.synchronize()
+            return {"ok": True}, b""
+        if op == "mem_info":
+            free_b, total_b = _hipops.mem_info()
+            return {"ok": True, "free": free_b, "total": total_b}, b""
+        raise ValueError(f"unknown op {op!r}")
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--socket", required=True)
+    parser.add_argument("--device", type=int, default=0)
+    args = parser.parse_args()
+
+    if not _hipops.is_available():
+        sys.exit(3)
+    _hipops.init(args.device)
+    # warm the hot kernels + staging once, before accepting connections
+    h = _hipops.rand(1 << 20, 1, 7)
+    _hipops.sum(h, 1, 1 << 20, 1)
+    _hipops.free(h)
+    _hipops.synchronize()
+
+    try:
+        os.unlink(args.socket)
+    except FileNotFoundError:
+        pass
+    server = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    server.bind(args.socket)
+    server.listen(128)
+    # readiness marker for the spawning server
+    print("READY", flush=True)
+    while True:
+        conn, _ = server.accept()
+        Connection(conn).start()
+
+
+if __name__ == "__main__":
+    main()

@@ -16,6 +16,12 @@ Everything computes the same values user code would get on the CPU (same
 dtype; RNG is Philox instead of MT19937 -- a documented backend change,
 seeded from numpy's global RNG so np.random.seed still fixes the stream).
 
+Compute backends: inside an engine, sandbox children talk to the per-GPU
+compute daemon (ops/hipd.py, socket from APP_GPU_SERVICE) -- one HIP
+context per GPU, shared memory pool, fork-cheap sandboxes (per-process
+HIP context creation serializes in the driver at ~13/s). Without a
+daemon, the local backend owns the context directly through _hipops.
+
 This file is imported standalone (sys.path) inside sandbox children; it
 must not import the control-plane package or torch.
 """
@@ -31,6 +37,139 @@ import _hipops
 
 import numpy as _np
 
+import json as _json
+import socket as _socket
+import struct as _struct
+import threading as _threading
+
+
+class LocalBackend:
+    """Direct _hipops calls: this process owns the HIP context."""
+
+    name = "local"
+
+    def __init__(self):
+        _hipops.init(0)
+
+    def __getattr__(self, item):
+        return getattr(_hipops, item)
+
+
+class RemoteBackend:
+    """RPC to the engine's GPU daemon (ops/hipd.py) over a unix socket."""
+
+    name = "remote"
+
+    def __init__(self, path: str):
+        self._sock = _socket.socket(_socket.AF_UNIX, _socket.SOCK_STREAM)
+        self._sock.connect(path)
+        self._lock = _threading.Lock()
+        self._call({"op": "ping"})
+
+    def _send(self, header: dict, payload=None) -> None:
+        if payload is not None:
+            header = {**header, "plen": len(payload)}
+        hb = _json.dumps(header).encode()
+        self._sock.sendall(_struct.pack("<I", len(hb)) + hb)
+        if payload is not None and len(payload):
+            self._sock.sendall(payload)
+
+    def _read_exact_into(self, view) -> None:
+        got = 0
+        n = len(view)
+        while got < n:
+            r = self._sock.recv_into(view[got:], n - got)
+            if r == 0:
+                raise ConnectionError("gpu daemon closed the connection")
+            got += r
+
+    def _recv(self, out_buffer=None) -> dict:
+        head = bytearray(4)
+        self._read_exact_into(memoryview(head))
+        (hlen,) = _struct.unpack("<I", bytes(head))
+        hb = bytearray(hlen)
+        self._read_exact_into(memoryview(hb))
+        header = _json.loads(bytes(hb))
+        plen = header.get("plen", 0)
+        if plen:
+            if out_buffer is not None and len(out_buffer) == plen:
+                self._read_exact_into(memoryview(out_buffer).cast("B"))
+            else:
+                tmp = bytearray(plen)
+                self._read_exact_into(memoryview(tmp))
+                header["_payload"] = bytes(tmp)
+        if not header.get("ok", False):
+            raise RuntimeError(header.get("error", "gpu daemon error"))
+        return header
+
+    def _call(self, header: dict, payload=None, out_buffer=None) -> dict:
+        with self._lock:
+            self._send(header, payload)
+            return self._recv(out_buffer)
+
+    # _hipops-compatible surface ----------------------------------------
+    def is_available(self) -> bool:
+        return True
+
+    def upload(self, buffer):
+        mv = memoryview(buffer).cast("B")
+        return self._call({"op": "upload"}, payload=mv)["h"]
+
+    def download(self, h, out) -> None:
+        self._call(
+            {"op": "download", "h": h, "nbytes": memoryview(out).nbytes},
+            out_buffer=memoryview(out).cast("B"),
+        )
+
+    def alloc(self, nbytes):
+        return self._call({"op": "alloc", "nbytes": nbytes})["h"]
+
+    def free(self, h) -> None:
+        self._call({"op": "free", "h": h})
+
+    def rand(self, n, dtype, seed):
+        return self._call({"op": "rand", "n": n, "dtype": dtype, "seed": seed})["h"]
+
+    def unary(self, h, uop, dtype, n):
+        return self._call(
+            {"op": "unary", "h": h, "uop": uop, "dtype": dtype, "n": n}
+        )["h"]
+
+    def binary(self, ha, hb, bop, dtype, n):
+        return self._call(
+            {"op": "binary", "ha": ha, "hb": hb, "bop": bop, "dtype": dtype, "n": n}
+        )["h"]
+
+    def binary_scalar(self, h, scalar, bop, dtype, n):
+        return self._call(
+            {
+                "op": "binary_scalar",
+                "h": h,
+                "scalar": scalar,
+                "bop": bop,
+                "dtype": dtype,
+                "n": n,
+            }
+        )["h"]
+
+    def sum(self, h, dtype, n, square):
+        return self._call(
+            {"op": "sum", "h": h, "dtype": dtype, "n": n, "square": square}
+        )["value"]
+
+    def gemm(self, ha, hb, m, n, k, dtype):
+        return self._call(
+            {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
+        )["h"]
+
+    def synchronize(self) -> None:
+        self._call({"op": "sync"})
+
+    def mem_info(self):
+        r = self._call({"op": "mem_info"})
+        return (r["free"], r["total"])
+
+
 # dtype codes shared with _hipops
 _F32, _F64 = 0, 1
 _UNARY = {"square": 0, "negative": 1, "absolute": 2, "sqrt": 3, "exp": 4}
@@ -39,41 +178,66 @@ _BINARY = {"add": 0, "subtract": 1, "multiply": 2, "divide": 3, "true_divide": 3
 MIN_ELEMS = int(os.environ.get("APP_HIP_NUMPY_MIN_ELEMS", 2_000_000))
 MIN_MATMUL_FLOPS = float(os.environ.get("APP_HIP_NUMPY_MIN_MATMUL_FLOPS", 5e7))
 
-_state = {"ready": False, "failed": None}
+_state = {"backend": None, "failed": None}
+
+
+def daemon_socket():
+    path = os.environ.get("APP_GPU_SERVICE")
+    if path and os.path.exists(path):
+        return path
+    return None
 
 
 def available() -> bool:
-    return _hipops.is_available()
+    return daemon_socket() is not None or _hipops.is_available()
+
+
+def backend():
+    _ensure_ready()
+    return _state["backend"]
 
 
 def _ensure_ready() -> None:
-    if _state["ready"]:
+    if _state["backend"] is not None:
         return
     if _state["failed"]:
         raise RuntimeError(_state["failed"])
     try:
-        _hipops.init(0)
-        _state["ready"] = True
+        env_path = os.environ.get("APP_GPU_SERVICE")
+        if env_path:
+            # a daemon is advertised: give it a moment to come up (its HIP
+            # init runs concurrently with engine startup) before falling
+            # back to an own-context backend
+            import time
+
+            wait = float(os.environ.get("APP_GPU_SERVICE_WAIT", "10"))
+            deadline = time.monotonic() + wait
+            while True:
+                if os.path.exists(env_path):
+                    try:
+                        _state["backend"] = RemoteBackend(env_path)
+                        return
+                    except (OSError, RuntimeError):
+                        pass
+                if time.monotonic() >= deadline or not _hipops.is_available():
+                    break
+                time.sleep(0.05)
+        _state["backend"] = LocalBackend()
     except Exception as e:
         _state["failed"] = str(e)
         raise
 
 
 def warmup() -> None:
-    """Bring up the HIP context + staging pools and touch every hot kernel
-    once (pre-forked warm child calls this before any request)."""
+    """Connect to the GPU daemon (or bring up a local context) and touch
+    the hot path once (pre-forked warm child calls this pre-request)."""
     if not available():
         raise RuntimeError("no AMD GPU visible")
     _ensure_ready()
-    h = _hipops.rand(4096, _F64, 12345)
-    h2 = _hipops.unary(h, _UNARY["square"], _F64, 4096)
-    _hipops.sum(h, _F64, 4096, 1)
-    a = _np.zeros((64, 64), dtype=_np.float32)
-    ha = _hipops.upload(a)
-    hc = _hipops.gemm(ha, ha, 64, 64, 64, _F32)
-    for handle in (h, h2, ha, hc):
-        _hipops.free(handle)
-    _hipops.synchronize()
+    b = _state["backend"]
+    h = b.rand(4096, _F64, 12345)
+    b.sum(h, _F64, 4096, 1)
+    b.free(h)
 
 
 def _dtype_code(dtype):
@@ -121,14 +285,14 @@ class DeviceArray:
     def __del__(self):
         try:
             if self._handle is not None:
-                _hipops.free(self._handle)
+                backend().free(self._handle)
         except Exception:
             pass
 
     def materialize(self) -> _np.ndarray:
         if self._host is None:
             out = _np.empty(self.shape, dtype=self.dtype)
-            _hipops.download(self._handle, out)
+            backend().download(self._handle, out)
             self._host = out
         return self._host
 
@@ -153,7 +317,7 @@ class DeviceArray:
 
     # -- device compute -------------------------------------------------
     def _unary(self, opname):
-        out = _hipops.unary(self._handle, _UNARY[opname], _dtype_code(self.dtype), self.size)
+        out = backend().unary(self._handle, _UNARY[opname], _dtype_code(self.dtype), self.size)
         return DeviceArray(out, self.shape, self.dtype)
 
     def _binary(self, opname, other, reverse=False):
@@ -162,12 +326,12 @@ class DeviceArray:
             if other.shape != self.shape or other.dtype != self.dtype:
                 return NotImplemented
             a, b = (other, self) if reverse else (self, other)
-            out = _hipops.binary(a._handle, b._handle, _BINARY[opname], code, self.size)
+            out = backend().binary(a._handle, b._handle, _BINARY[opname], code, self.size)
             return DeviceArray(out, self.shape, self.dtype)
         if isinstance(other, (int, float)):
             if reverse and opname in ("subtract", "divide", "true_divide"):
                 return NotImplemented  # scalar-first sub/div: fall back
-            out = _hipops.binary_scalar(
+            out = backend().binary_scalar(
                 self._handle, float(other), _BINARY[opname], code, self.size
             )
             return DeviceArray(out, self.shape, self.dtype)
@@ -176,7 +340,7 @@ class DeviceArray:
     def sum(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
-                _hipops.sum(self._handle, _dtype_code(self.dtype), self.size, 0)
+                backend().sum(self._handle, _dtype_code(self.dtype), self.size, 0)
             )
         return self.materialize().sum(axis=axis, **kwargs)
 
@@ -188,7 +352,7 @@ class DeviceArray:
     def square_sum(self):
         """Fused sum(x*x) -- no intermediate array."""
         return self.dtype.type(
-            _hipops.sum(self._handle, _dtype_code(self.dtype), self.size, 1)
+            backend().sum(self._handle, _dtype_code(self.dtype), self.size, 1)
         )
 
     # -- NEP 13: ufuncs --------------------------------------------------
@@ -281,7 +445,7 @@ def _to_device(x) -> "DeviceArray | None":
     if _dtype_code(arr.dtype) is None:
         return None
     arr = _np.ascontiguousarray(arr)
-    return DeviceArray(_hipops.upload(arr), arr.shape, arr.dtype)
+    return DeviceArray(backend().upload(arr), arr.shape, arr.dtype)
 
 
 # ---------------------------------------------------------------------------
@@ -296,7 +460,7 @@ def rand(*shape, seed=None):
         n *= int(s)
     if seed is None:
         seed = int(_np.random.randint(0, 2**63 - 1, dtype=_np.int64))
-    h = _hipops.rand(n, _F64, int(seed))
+    h = backend().rand(n, _F64, int(seed))
     return DeviceArray(h, shape if shape else (), _np.float64)
 
 
@@ -342,7 +506,7 @@ def matmul(a, b, _force=False):
     if da is None or db is None or da.dtype != db.dtype:
         return NotImplemented
     code = _dtype_code(da.dtype)
-    hc = _hipops.gemm(da._handle, db._handle, m, n, k, code)
+    hc = backend().gemm(da._handle, db._handle, m, n, k, code)
     return DeviceArray(hc, (m, n), da.dtype)
 
 
