@@ -299,6 +299,12 @@ static ServerConfig g_cfg;
 // ---------------------------------------------------------------------------
 // zygote management
 // ---------------------------------------------------------------------------
+static double now_ms() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
+}
+
 struct JobState {
   std::mutex mu;
   std::condition_variable cv;
@@ -306,6 +312,8 @@ struct JobState {
   bool done = false;
   long pid = -1;
   int exit_code = -1;
+  double t_started_ms = 0;
+  double t_done_ms = 0;
 };
 
 class Zygote {
@@ -397,10 +405,12 @@ class Zygote {
       const json::Value* pidv = msg.get("pid");
       st->pid = pidv ? (long)pidv->num : -1;
       st->started = true;
+      st->t_started_ms = now_ms();
     } else if (ev->str == "exit") {
       const json::Value* code = msg.get("code");
       st->exit_code = code ? (int)code->num : -1;
       st->done = true;
+      st->t_done_ms = now_ms();
     }
     st->cv.notify_all();
   }
@@ -563,6 +573,10 @@ struct ExecOutcome {
   std::string stdout_text;
   std::string stderr_text;
   int exit_code = -1;
+  // per-stage timings (ms): dispatch = submit -> interpreter running,
+  // run = interpreter start -> exit
+  double t_dispatch_ms = 0;
+  double t_run_ms = 0;
 };
 
 // Cold path: plain fork/exec of python (used when the zygote is down or
@@ -641,6 +655,7 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
   }
   req += "}}";
 
+  double t_submit = now_ms();
   auto st = g_zygote->submit(id, req);
 
   ExecOutcome out;
@@ -666,6 +681,10 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
   }
   g_zygote->drop(id);
   out.exit_code = st->exit_code;
+  double t_started = st->t_started_ms > 0 ? st->t_started_ms : t_submit;
+  double t_done = st->t_done_ms > 0 ? st->t_done_ms : now_ms();
+  out.t_dispatch_ms = t_started - t_submit;
+  out.t_run_ms = t_done - t_started;
   lk.unlock();
   out.stdout_text = read_file(stdout_path);
   out.stderr_text = read_file(stderr_path);
@@ -879,9 +898,14 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   std::vector<std::string> changed;
   scan_changed(workspace, "", start, g_cfg.scan_recursive, changed);
 
+  char timings[160];
+  snprintf(timings, sizeof timings,
+           ",\"timings\":{\"dispatch_ms\":%.2f,\"run_ms\":%.2f,\"scan_ms\":%.2f}",
+           outcome.t_dispatch_ms, outcome.t_run_ms, 0.0);
   std::string resp = "{\"stdout\":" + json::quote(outcome.stdout_text) +
                      ",\"stderr\":" + json::quote(outcome.stderr_text) +
                      ",\"exit_code\":" + std::to_string(outcome.exit_code) +
+                     std::string(timings) +
                      ",\"files\":[";
   for (size_t i = 0; i < changed.size(); i++) {
     if (i) resp += ",";

@@ -40,6 +40,9 @@ class Result:
     stderr: str
     exit_code: int
     files: Mapping[str, str] = field(default_factory=dict)
+    # per-stage timings (ms): sandbox create/upload/exec/download on the
+    # client side plus dispatch/run from the executor server
+    timings: Mapping[str, float] = field(default_factory=dict)
 
 
 class ExecutorError(RuntimeError):
@@ -120,15 +123,21 @@ class SandboxClient:
         timeout: Optional[float] = None,
     ) -> Result:
         """Create a fresh sandbox session, run, tear it down."""
+        import time as _time
+
+        t0 = _time.perf_counter()
         try:
             session = await self.create_sandbox()
         except (httpx.HTTPError, OSError) as e:
             raise ExecutorError(f"sandbox create failed: {e!r}") from e
+        t_create = (_time.perf_counter() - t0) * 1000
         try:
-            return await self.run(
+            result = await self.run(
                 storage, source_code, files=files, env=env, timeout=timeout,
                 session=session,
             )
+            result.timings = {**result.timings, "create_ms": round(t_create, 2)}
+            return result
         finally:
             asyncio.ensure_future(self.delete_sandbox(session))
 
@@ -142,12 +151,18 @@ class SandboxClient:
             if resp.status_code not in (200, 204):
                 raise ExecutorError(f"upload of {path} failed: {resp.status_code}")
 
+        import time as _time
+
+        t0 = _time.perf_counter()
         await asyncio.gather(*(upload(p, h) for p, h in files.items()))
+        t_upload = (_time.perf_counter() - t0) * 1000
 
         body: dict = {"source_code": source_code, "env": dict(env)}
         if timeout is not None:
             body["timeout"] = timeout
+        t0 = _time.perf_counter()
         resp = await self._client.post(f"{prefix}/execute", json=body)
+        t_exec = (_time.perf_counter() - t0) * 1000
         if resp.status_code != 200:
             raise ExecutorError(f"execute failed: {resp.status_code} {resp.text!r}")
         payload = resp.json()
@@ -162,10 +177,19 @@ class SandboxClient:
                         await writer.write(chunk)
                 return path, writer.hash
 
+        t0 = _time.perf_counter()
         stored = dict(await asyncio.gather(*(download(p) for p in payload["files"])))
+        t_download = (_time.perf_counter() - t0) * 1000
+        timings = dict(payload.get("timings") or {})
+        timings.update(
+            upload_ms=round(t_upload, 2),
+            exec_api_ms=round(t_exec, 2),
+            download_ms=round(t_download, 2),
+        )
         return Result(
             stdout=payload["stdout"],
             stderr=payload["stderr"],
             exit_code=payload["exit_code"],
             files=stored,
+            timings=timings,
         )

@@ -33,12 +33,16 @@ async def series(ex, name, code, n, conc):
     sem = asyncio.Semaphore(conc)
     lat = []
 
+    stages = {}
+
     async def one():
         async with sem:
             t = time.perf_counter()
             r = await ex.execute(code)
             assert r.exit_code == 0, r.stderr[:300]
             lat.append(time.perf_counter() - t)
+            for k, v in (r.timings or {}).items():
+                stages.setdefault(k, []).append(v)
 
     t0 = time.perf_counter()
     await asyncio.gather(*(one() for _ in range(n)))
@@ -47,7 +51,10 @@ async def series(ex, name, code, n, conc):
     print(
         f"{name:18s} conc={conc:2d}: {n / dt:6.1f} req/s  "
         f"p50={statistics.median(lat) * 1000:6.0f} ms  "
-        f"min={lat[0] * 1000:5.0f} ms  max={lat[-1] * 1000:6.0f} ms",
+        f"min={lat[0] * 1000:5.0f} ms  max={lat[-1] * 1000:6.0f} ms  "
+        + "  ".join(
+            f"{k}={statistics.median(v):.1f}" for k, v in sorted(stages.items())
+        ),
         flush=True,
     )
 
