@@ -67,6 +67,7 @@ def main() -> None:
     parser.add_argument("--steps", type=int, default=16)
     parser.add_argument("--warmup", type=int, default=4)
     parser.add_argument("--concurrency", type=int, default=4)
+    parser.add_argument("--engines-per-gpu", type=int, default=2)
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
     args = parser.parse_args()
@@ -166,6 +167,7 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
         executor_root=os.path.join(tmp, "executors"),
         executor_backend="local",
         executor_pool_target_length=max(2, args.concurrency),
+        engines_per_gpu=args.engines_per_gpu,
         gpu_count=1 if use_gpu else 0,
         gpu_pinning=False,  # engines inherit this rank's HIP_VISIBLE_DEVICES
         hip_numpy="require" if use_gpu else "off",

@@ -65,6 +65,7 @@ class ApplicationContext:
         return LocalPoolExecutor(
             file_storage=self.file_storage,
             pool_target_length=cfg.executor_pool_target_length,
+            engines_per_gpu=cfg.engines_per_gpu,
             gpu_count=cfg.gpu_count,
             gpu_pinning=cfg.gpu_pinning,
             executor_root=cfg.executor_root,

@@ -70,8 +70,11 @@ class Config(EnvSettings):
     executor_pod_name_prefix: str = "code-executor-"
 
     # ---- local backend ----------------------------------------------------
-    # warm pool of pre-spawned executor-server processes, per GPU
+    # warm pool of pre-forked single-use interpreters, per engine
     executor_pool_target_length: int = 2
+    # engines (executor-server + zygote + GPU daemon) per GPU: raises the
+    # sandbox-management parallelism of one device
+    engines_per_gpu: int = 1
     # root dir for per-executor workspaces + unix sockets (tmpdir if empty)
     executor_root: str = ""
     # path to the executor-server binary ("" = bundled build)

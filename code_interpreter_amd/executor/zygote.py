@@ -158,10 +158,14 @@ def main() -> None:
     buf = b""
 
     while True:
-        # refill: forking is cheap; the expensive HIP init happens inside
-        # the children concurrently
-        while len(warm_pool) < pool_target:
-            warm_pool.append(WarmChild())
+        # refill AT MOST ONE child per tick, and only when no request is
+        # waiting: a fork of a numpy-loaded interpreter costs ~10 ms (page
+        # tables), and a refill burst inside this loop would delay job
+        # dispatch and exit notifications for every in-flight request
+        if len(warm_pool) < pool_target:
+            r, _, _ = select.select([server], [], [], 0)
+            if not r:
+                warm_pool.append(WarmChild())
         ready_count = sum(1 for w in warm_pool if w.poll_ready())
         if ready_count != announced:
             announced = ready_count

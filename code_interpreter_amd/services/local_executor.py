@@ -86,6 +86,7 @@ class LocalPoolExecutor:
         self,
         file_storage: Storage,
         pool_target_length: int = 2,
+        engines_per_gpu: int = 1,
         gpu_count: int = -1,
         gpu_pinning: bool = True,
         executor_root: str = "",
@@ -115,7 +116,7 @@ class LocalPoolExecutor:
         self.hip_numpy = hip_numpy
         self.spawn_ready_timeout = spawn_ready_timeout
 
-        self.n_engines = max(1, gpu_count)
+        self.n_engines = max(1, gpu_count) * max(1, engines_per_gpu)
         self._engines: list[Optional[Engine]] = [None] * self.n_engines
         self._spawn_locks = [asyncio.Lock() for _ in range(self.n_engines)]
         self._rr = 0

@@ -65,6 +65,7 @@ async def main():
     ex = LocalPoolExecutor(
         Storage(os.path.join(tmp, "s")),
         pool_target_length=12,
+        engines_per_gpu=int(os.environ.get("EXP_ENGINES", "1")),
         gpu_count=1 if hip != "off" else 0,
         gpu_pinning=False,
         executor_root=os.path.join(tmp, "e"),
