@@ -851,6 +851,7 @@ static std::string url_decode(const std::string& s) {
 
 static void handle_execute(Conn& conn, const HttpRequest& req,
                            const std::string& workspace) {
+  double t_handler0 = now_ms();
   json::Parser parser(req.body);
   json::Value body = parser.parse();
   if (!parser.ok || !body.is_object() || !body.get("source_code") ||
@@ -898,10 +899,11 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   std::vector<std::string> changed;
   scan_changed(workspace, "", start, g_cfg.scan_recursive, changed);
 
-  char timings[160];
+  char timings[192];
   snprintf(timings, sizeof timings,
-           ",\"timings\":{\"dispatch_ms\":%.2f,\"run_ms\":%.2f,\"scan_ms\":%.2f}",
-           outcome.t_dispatch_ms, outcome.t_run_ms, 0.0);
+           ",\"timings\":{\"dispatch_ms\":%.2f,\"run_ms\":%.2f,"
+           "\"handler_ms\":%.2f}",
+           outcome.t_dispatch_ms, outcome.t_run_ms, now_ms() - t_handler0);
   std::string resp = "{\"stdout\":" + json::quote(outcome.stdout_text) +
                      ",\"stderr\":" + json::quote(outcome.stderr_text) +
                      ",\"exit_code\":" + std::to_string(outcome.exit_code) +
