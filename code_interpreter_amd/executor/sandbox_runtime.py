@@ -226,6 +226,9 @@ def run_user_script(script_path: str) -> int:
     """Run the user's script the way `python script.py` would: fresh
     __main__ globals, argv[0] = script, tracebacks to stderr, exit code 0/1
     (or SystemExit's code). Returns the exit code."""
+    import time as _time
+
+    _t0 = _time.perf_counter()
     install_import_hooks()
 
     with open(script_path, "r", encoding="utf-8", errors="replace") as f:
@@ -240,6 +243,7 @@ def run_user_script(script_path: str) -> int:
         except Exception:
             pass
 
+    _t1 = _time.perf_counter()
     sys.argv = [script_path]
     script_globals = {
         "__name__": "__main__",
@@ -267,9 +271,37 @@ def run_user_script(script_path: str) -> int:
         traceback.print_exception(etype, exc, tb)
         return 1
     finally:
+        _write_child_timings(_t0, _t1, _time.perf_counter())
         try:
             sys.stdout.flush()
             sys.stderr.flush()
         except Exception:
             pass
     return 0
+
+
+def _write_child_timings(t0: float, t1: float, t2: float) -> None:
+    path = os.environ.get("SANDBOX_TIMING_FILE")
+    if not path:
+        return
+    try:
+        rpc_ms, rpc_n = 0.0, 0
+        hipnp = sys.modules.get("hipnp")
+        if hipnp is not None:
+            stats = getattr(hipnp, "RPC_STATS", None)
+            if stats:
+                rpc_ms, rpc_n = stats.get("ms", 0.0), stats.get("n", 0)
+        import json
+
+        with open(path, "w") as f:
+            json.dump(
+                {
+                    "child_setup_ms": round((t1 - t0) * 1000, 2),
+                    "child_exec_ms": round((t2 - t1) * 1000, 2),
+                    "gpu_rpc_ms": round(rpc_ms, 2),
+                    "gpu_rpc_n": rpc_n,
+                },
+                f,
+            )
+    except Exception:
+        pass

@@ -899,15 +899,21 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   std::vector<std::string> changed;
   scan_changed(workspace, "", start, g_cfg.scan_recursive, changed);
 
+  std::string child_t = read_file(stdout_path + ".t");  // child-side phases
   char timings[192];
   snprintf(timings, sizeof timings,
            ",\"timings\":{\"dispatch_ms\":%.2f,\"run_ms\":%.2f,"
-           "\"handler_ms\":%.2f}",
+           "\"handler_ms\":%.2f",
            outcome.t_dispatch_ms, outcome.t_run_ms, now_ms() - t_handler0);
+  std::string timings_str = timings;
+  if (!child_t.empty() && child_t[0] == '{') {
+    timings_str += ",\"child\":" + child_t;
+  }
+  timings_str += "}";
   std::string resp = "{\"stdout\":" + json::quote(outcome.stdout_text) +
                      ",\"stderr\":" + json::quote(outcome.stderr_text) +
                      ",\"exit_code\":" + std::to_string(outcome.exit_code) +
-                     std::string(timings) +
+                     timings_str +
                      ",\"files\":[";
   for (size_t i = 0; i < changed.size(); i++) {
     if (i) resp += ",";
@@ -918,6 +924,7 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   // clean the temp dir
   unlink(script_path.c_str());
   unlink(stdout_path.c_str());
+  unlink((stdout_path + ".t").c_str());
   unlink(stderr_path.c_str());
   rmdir(tmpdir);
 

@@ -61,6 +61,7 @@ def _run_job_in_child(job: dict) -> None:
         sys.stderr = os.fdopen(2, "w", buffering=1, closefd=False)
         os.environ.update(job.get("env") or {})
         os.chdir(job["cwd"])
+        os.environ["SANDBOX_TIMING_FILE"] = job["stdout"] + ".t"
         exit_code = sandbox_runtime.run_user_script(job["script"])
     except SystemExit as e:
         exit_code = e.code if isinstance(e.code, int) else (0 if e.code is None else 1)

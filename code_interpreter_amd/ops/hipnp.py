@@ -103,9 +103,15 @@ class RemoteBackend:
         return header
 
     def _call(self, header: dict, payload=None, out_buffer=None) -> dict:
+        import time as _time
+
+        t0 = _time.perf_counter()
         with self._lock:
             self._send(header, payload)
-            return self._recv(out_buffer)
+            out = self._recv(out_buffer)
+        RPC_STATS["ms"] += (_time.perf_counter() - t0) * 1000
+        RPC_STATS["n"] += 1
+        return out
 
     # _hipops-compatible surface ----------------------------------------
     def is_available(self) -> bool:
@@ -179,6 +185,7 @@ MIN_ELEMS = int(os.environ.get("APP_HIP_NUMPY_MIN_ELEMS", 2_000_000))
 MIN_MATMUL_FLOPS = float(os.environ.get("APP_HIP_NUMPY_MIN_MATMUL_FLOPS", 5e7))
 
 _state = {"backend": None, "failed": None}
+RPC_STATS = {"ms": 0.0, "n": 0}
 
 
 def daemon_socket():

@@ -42,7 +42,11 @@ async def series(ex, name, code, n, conc):
             assert r.exit_code == 0, r.stderr[:300]
             lat.append(time.perf_counter() - t)
             for k, v in (r.timings or {}).items():
-                stages.setdefault(k, []).append(v)
+                if isinstance(v, dict):
+                    for k2, v2 in v.items():
+                        stages.setdefault(k2, []).append(v2)
+                else:
+                    stages.setdefault(k, []).append(v)
 
     t0 = time.perf_counter()
     await asyncio.gather(*(one() for _ in range(n)))
