@@ -876,6 +876,7 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   clock_gettime(CLOCK_REALTIME, &now);
   TimeSpec start{(int64_t)now.tv_sec, (int64_t)now.tv_nsec};
 
+  double t_pre0 = now_ms();
   char tmpl[] = "/tmp/exec.XXXXXX";
   char* tmpdir = mkdtemp(tmpl);
   if (!tmpdir) {
@@ -887,6 +888,7 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   std::string stderr_path = std::string(tmpdir) + "/stderr";
   write_file(script_path, source);
 
+  double t_pre1 = now_ms();
   ExecOutcome outcome;
   if (g_cfg.zygote && g_zygote && g_zygote->alive()) {
     outcome = run_via_zygote(script_path, stdout_path, stderr_path, extra_env,
@@ -896,15 +898,19 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
                        timeout_s, workspace);
   }
 
+  double t_post0 = now_ms();
   std::vector<std::string> changed;
   scan_changed(workspace, "", start, g_cfg.scan_recursive, changed);
+  double t_post1 = now_ms();
 
   std::string child_t = read_file(stdout_path + ".t");  // child-side phases
-  char timings[192];
+  char timings[256];
   snprintf(timings, sizeof timings,
            ",\"timings\":{\"dispatch_ms\":%.2f,\"run_ms\":%.2f,"
-           "\"handler_ms\":%.2f",
-           outcome.t_dispatch_ms, outcome.t_run_ms, now_ms() - t_handler0);
+           "\"handler_ms\":%.2f,\"pre_ms\":%.2f,\"scan_ms\":%.2f,"
+           "\"wait_ms\":%.2f",
+           outcome.t_dispatch_ms, outcome.t_run_ms, now_ms() - t_handler0,
+           t_pre1 - t_pre0, t_post1 - t_post0, t_post0 - t_pre1);
   std::string timings_str = timings;
   if (!child_t.empty() && child_t[0] == '{') {
     timings_str += ",\"child\":" + child_t;
