@@ -140,6 +140,29 @@ def _maybe_install_hip_numpy() -> None:
         # auto mode: CPU numpy is the documented fallback
 
 
+def preload() -> None:
+    """Called in the ZYGOTE before any fork: import everything heavy so
+    children get it copy-on-write. Importing hipnp/_hipops only LOADS the
+    HIP userspace stack (large .so relocations, ~0.3 CPU-s) -- it creates
+    no device context, so it is fork-safe; paying it once in the zygote
+    instead of per child keeps bursts under container CPU quotas (a 16-CPU
+    cgroup quota turns per-child reloads into 100 ms CFS throttle stalls).
+    """
+    install_import_hooks()
+    try:
+        import numpy  # noqa: F401
+    except ImportError:
+        return
+    ops_dir = os.environ.get("APP_OPS_DIR")
+    if ops_dir and ops_dir not in sys.path:
+        sys.path.insert(0, ops_dir)
+    if os.environ.get("APP_HIP_NUMPY", "auto").lower() != "off":
+        try:
+            import hipnp  # noqa: F401  (no HIP context is created here)
+        except Exception:
+            pass
+
+
 def prewarm() -> None:
     """Expensive init in the pre-forked warm child, before any request:
     import numpy, install hooks, and bring up the HIP runtime (device

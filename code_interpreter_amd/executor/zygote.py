@@ -39,6 +39,8 @@ if RUNTIME_DIR not in sys.path:
 
 import sandbox_runtime  # noqa: E402
 
+sandbox_runtime.preload()  # heavy imports once, pre-fork (COW for children)
+
 
 def _run_job_in_child(job: dict) -> None:
     """Executed in the forked child. Never returns."""

@@ -161,6 +161,10 @@ class LocalPoolExecutor:
         sock = os.path.join(root, "exec.sock")
 
         env = dict(os.environ)
+        # sandboxed CPU numpy should not spawn one BLAS thread per host
+        # core: under container CPU quotas that turns into CFS throttling
+        env.setdefault("OPENBLAS_NUM_THREADS", "8")
+        env.setdefault("OMP_NUM_THREADS", "8")
         env.update(
             {
                 "APP_LISTEN_UNIX": sock,
