@@ -64,9 +64,9 @@ def _percentile(sorted_vals, p):
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=16)
-    parser.add_argument("--warmup", type=int, default=4)
-    parser.add_argument("--concurrency", type=int, default=4)
+    parser.add_argument("--steps", type=int, default=256)
+    parser.add_argument("--warmup", type=int, default=32)
+    parser.add_argument("--concurrency", type=int, default=8)
     parser.add_argument("--engines-per-gpu", type=int, default=2)
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
