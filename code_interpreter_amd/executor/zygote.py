@@ -37,9 +37,14 @@ RUNTIME_DIR = os.path.dirname(os.path.abspath(__file__))
 if RUNTIME_DIR not in sys.path:
     sys.path.insert(0, RUNTIME_DIR)
 
+import gc
+
 import sandbox_runtime  # noqa: E402
 
 sandbox_runtime.preload()  # heavy imports once, pre-fork (COW for children)
+# freeze the heap into the permanent generation: forked children's GC does
+# not touch (and COW-copy) the preloaded objects' pages
+gc.freeze()
 
 
 def _run_job_in_child(job: dict) -> None:

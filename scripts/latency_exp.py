@@ -86,6 +86,8 @@ async def main():
         await series(ex, "numpy-gpu-1e8", NUMPY_GPU, 20, 1)
         await series(ex, "numpy-gpu-1e8", NUMPY_GPU, 40, 8)
         await series(ex, "numpy-gpu-small", NUMPY_SMALL_GPU, 40, 8)
+        await series(ex, "sustained-1e8", NUMPY_GPU, 200, 8)
+        await series(ex, "sustained-trivial", TRIVIAL, 200, 8)
     finally:
         await ex.aclose()
 
