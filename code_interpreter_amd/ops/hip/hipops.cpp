@@ -54,6 +54,11 @@ struct State {
   std::unordered_map<int64_t, std::vector<void*>> free_list;
   int64_t pool_bytes = 0;
   uint64_t rand_offset = 0;
+  // allocator stats
+  int64_t outstanding = 0;
+  long mempool_allocs = 0;
+  long fallback_allocs = 0;
+  long oom_trims = 0;
 };
 
 State g;
@@ -74,12 +79,17 @@ void* pool_alloc(int64_t size) {
     void* p = nullptr;
     hipError_t e = hipMallocAsync(&p, size, g.compute);
     if (e == hipErrorOutOfMemory) {
+      g.oom_trims++;
       hipMemPool_t pool = nullptr;
       if (hipDeviceGetDefaultMemPool(&pool, g.device) == hipSuccess)
         (void)hipMemPoolTrimTo(pool, 0);
       e = hipMallocAsync(&p, size, g.compute);
     }
-    if (e == hipSuccess) return p;
+    if (e == hipSuccess) {
+      g.mempool_allocs++;
+      g.outstanding += size;
+      return p;
+    }
     // fall through to plain hipMalloc on persistent failure
   }
   auto it = g.free_list.find(size);
@@ -104,7 +114,10 @@ void* pool_alloc(int64_t size) {
 }
 
 void pool_free(void* p, int64_t size) {
-  if (g.use_mempool && hipFreeAsync(p, g.compute) == hipSuccess) return;
+  if (g.use_mempool && hipFreeAsync(p, g.compute) == hipSuccess) {
+    g.outstanding -= round_size(size);
+    return;
+  }
   g.free_list[round_size(size)].push_back(p);
 }
 
@@ -457,8 +470,10 @@ PyObject* py_mem_info(PyObject*, PyObject*) {
   ensure_init();
   size_t free_b = 0, total_b = 0;
   HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
-  return Py_BuildValue("(KK)", (unsigned long long)free_b,
-                       (unsigned long long)total_b);
+  return Py_BuildValue("(KKLlll)", (unsigned long long)free_b,
+                       (unsigned long long)total_b,
+                       (long long)g.outstanding, g.mempool_allocs,
+                       g.fallback_allocs, g.oom_trims);
   WRAP_END
 }
 

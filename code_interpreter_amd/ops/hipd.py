@@ -139,8 +139,8 @@ class Connection(threading.Thread):
             _hipops.synchronize()
             return {"ok": True}, b""
         if op == "mem_info":
-            free_b, total_b = _hipops.mem_info()
-            return {"ok": True, "free": free_b, "total": total_b}, b""
+            info = _hipops.mem_info()
+            return {"ok": True, "info": list(info)}, b""
         raise ValueError(f"unknown op {op!r}")
 
 

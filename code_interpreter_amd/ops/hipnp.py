@@ -172,8 +172,7 @@ class RemoteBackend:
         self._call({"op": "sync"})
 
     def mem_info(self):
-        r = self._call({"op": "mem_info"})
-        return (r["free"], r["total"])
+        return tuple(self._call({"op": "mem_info"})["info"])
 
 
 # dtype codes shared with _hipops

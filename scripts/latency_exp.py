@@ -15,6 +15,8 @@ sys.path.insert(0, str(REPO))
 from code_interpreter_amd.services.local_executor import LocalPoolExecutor  # noqa
 from code_interpreter_amd.services.storage import Storage  # noqa
 
+sys.path.insert(0, str(REPO / "code_interpreter_amd" / "ops"))
+
 TRIVIAL = "print('x')"
 NUMPY_TOUCH = "import numpy\nprint(numpy.__version__)"
 NUMPY_GPU = (
@@ -63,6 +65,22 @@ async def series(ex, name, code, n, conc):
     )
 
 
+def _print_daemon_stats(ex):
+    import hipnp  # noqa (ops dir on path via package __init__)
+
+    for i, eng in enumerate(ex._engines):
+        if eng is None:
+            continue
+        sock = os.path.join(eng.root, "sandboxes", "gpu.sock")
+        if not os.path.exists(sock):
+            continue
+        try:
+            b = hipnp.RemoteBackend(sock)
+            print(f"engine {i} daemon mem_info: {b.mem_info()}", flush=True)
+        except Exception as e:
+            print(f"engine {i} stats failed: {e}", flush=True)
+
+
 async def main():
     tmp = tempfile.mkdtemp()
     hip = os.environ.get("EXP_HIP", "require")
@@ -87,6 +105,7 @@ async def main():
         await series(ex, "numpy-gpu-1e8", NUMPY_GPU, 40, 8)
         await series(ex, "numpy-gpu-small", NUMPY_SMALL_GPU, 40, 8)
         await series(ex, "sustained-1e8", NUMPY_GPU, 200, 8)
+        _print_daemon_stats(ex)
         await series(ex, "sustained-trivial", TRIVIAL, 200, 8)
     finally:
         await ex.aclose()
