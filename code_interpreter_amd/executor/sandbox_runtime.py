@@ -310,10 +310,12 @@ def _write_child_timings(t0: float, t1: float, t2: float) -> None:
     try:
         rpc_ms, rpc_n = 0.0, 0
         hipnp = sys.modules.get("hipnp")
+        per_op = {}
         if hipnp is not None:
             stats = getattr(hipnp, "RPC_STATS", None)
             if stats:
                 rpc_ms, rpc_n = stats.get("ms", 0.0), stats.get("n", 0)
+                per_op = stats.get("per_op", {})
         import json
 
         with open(path, "w") as f:
@@ -323,6 +325,7 @@ def _write_child_timings(t0: float, t1: float, t2: float) -> None:
                     "child_exec_ms": round((t2 - t1) * 1000, 2),
                     "gpu_rpc_ms": round(rpc_ms, 2),
                     "gpu_rpc_n": rpc_n,
+                    **{f"rpc_{k}": v for k, v in per_op.items()},
                 },
                 f,
             )

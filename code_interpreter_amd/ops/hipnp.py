@@ -109,8 +109,11 @@ class RemoteBackend:
         with self._lock:
             self._send(header, payload)
             out = self._recv(out_buffer)
-        RPC_STATS["ms"] += (_time.perf_counter() - t0) * 1000
+        dt = (_time.perf_counter() - t0) * 1000
+        RPC_STATS["ms"] += dt
         RPC_STATS["n"] += 1
+        per_op = RPC_STATS.setdefault("per_op", {})
+        per_op[header["op"]] = round(per_op.get(header["op"], 0.0) + dt, 2)
         return out
 
     # _hipops-compatible surface ----------------------------------------
