@@ -36,19 +36,24 @@ async def series(ex, name, code, n, conc):
     lat = []
 
     stages = {}
+    rows = []
 
     async def one():
         async with sem:
             t = time.perf_counter()
             r = await ex.execute(code)
             assert r.exit_code == 0, r.stderr[:300]
-            lat.append(time.perf_counter() - t)
+            wall = time.perf_counter() - t
+            lat.append(wall)
+            flat = {}
             for k, v in (r.timings or {}).items():
                 if isinstance(v, dict):
-                    for k2, v2 in v.items():
-                        stages.setdefault(k2, []).append(v2)
+                    flat.update(v)
                 else:
-                    stages.setdefault(k, []).append(v)
+                    flat[k] = v
+            rows.append((wall, flat))
+            for k, v in flat.items():
+                stages.setdefault(k, []).append(v)
 
     t0 = time.perf_counter()
     await asyncio.gather(*(one() for _ in range(n)))
@@ -63,6 +68,12 @@ async def series(ex, name, code, n, conc):
         ),
         flush=True,
     )
+    rows.sort(key=lambda r: -r[0])
+    for wall, flat in rows[:3]:
+        if wall * 1000 > 3 * statistics.median(lat) * 1000 + 50:
+            print(f"   WORST {wall*1000:7.0f} ms: "
+                  + "  ".join(f"{k}={v}" for k, v in sorted(flat.items())),
+                  flush=True)
 
 
 def _print_daemon_stats(ex):
