@@ -147,3 +147,29 @@ def wheelhouse(tmp_path_factory):
 
 def run_async(coro):
     return asyncio.run(coro)
+
+
+@pytest.fixture(scope="session")
+def gpu_executor(tmp_path_factory, executor_bin):
+    """A live 1-GPU local executor pool (skipped when no GPU is visible)."""
+    from code_interpreter_amd.services.local_executor import LocalPoolExecutor
+    from code_interpreter_amd.services.storage import Storage
+
+    sys.path.insert(0, str(REPO_ROOT / "code_interpreter_amd" / "ops"))
+    import _hipops
+
+    if not _hipops.is_available():
+        pytest.skip("no AMD GPU visible")
+
+    tmp = tmp_path_factory.mktemp("gpue")
+    ex = LocalPoolExecutor(
+        Storage(str(tmp / "storage")),
+        pool_target_length=2,
+        gpu_count=1,
+        executor_root=str(tmp / "eng"),
+        hip_numpy="require",
+        dep_install=False,
+        execute_timeout=120.0,
+    )
+    yield ex
+    asyncio.run(ex.aclose())

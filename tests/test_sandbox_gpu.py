@@ -2,40 +2,10 @@
 is actually routed to the gfx950 HIP kernels (no silent CPU fallback)."""
 
 import asyncio
-import tempfile
 
 import pytest
 
-from code_interpreter_amd.services.local_executor import LocalPoolExecutor
-from code_interpreter_amd.services.storage import Storage
-
 pytestmark = pytest.mark.gpu
-
-
-@pytest.fixture(scope="module")
-def gpu_executor(tmp_path_factory, executor_bin):
-    import sys
-    from pathlib import Path
-
-    sys.path.insert(
-        0, str(Path(__file__).resolve().parent.parent / "code_interpreter_amd" / "ops")
-    )
-    import _hipops
-
-    if not _hipops.is_available():
-        pytest.skip("no AMD GPU visible")
-
-    tmp = tmp_path_factory.mktemp("gpue")
-    ex = LocalPoolExecutor(
-        Storage(str(tmp / "storage")),
-        pool_target_length=2,
-        gpu_count=1,
-        executor_root=str(tmp / "eng"),
-        hip_numpy="require",
-        dep_install=False,
-    )
-    yield ex
-    asyncio.run(ex.aclose())
 
 
 def _run(ex, code, **kw):
