@@ -22,6 +22,7 @@ SOURCES = [
     HIP_DIR / "gemm_f32.hip",
     HIP_DIR / "gemm_f64.hip",
     HIP_DIR / "gemm_bf16.hip",
+    HIP_DIR / "gemm_bf16_256.hip",
 ]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")

@@ -50,3 +50,9 @@ void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
 // bf16 in (as uint16 storage), f32 accumulate, bf16 out
 void launch_gemm_bf16(const uint16_t* a, const uint16_t* b, uint16_t* c, int m,
                       int n, int k, hipStream_t stream);
+// fast 256^2-tile 8-phase path (takes B pre-transposed [N][K])
+bool gemm_bf16_256_supported(int m, int n, int k);
+void launch_gemm_bf16_256(const uint16_t* a, const uint16_t* bt, uint16_t* c,
+                          int m, int n, int k, hipStream_t stream);
+void launch_transpose_bf16(const uint16_t* in, uint16_t* out, int k, int n,
+                           hipStream_t stream);

@@ -1,0 +1,304 @@
+// bf16 GEMM, 256^2-tile 8-phase structure for gfx950.
+//
+// The fast MFMA GEMM shape for this chip (CDNA4 guide's verified
+// template class): 256x256 block tile, BK=64, 512 threads as 8 waves,
+// one workgroup per CU (128 KiB LDS), double-buffered K-tiles filled by
+// global_load_lds_dwordx4 (direct-to-LDS DMA, 16 B/lane), raw
+// s_barrier + counted s_waitcnt vmcnt(N) so prefetched half-tiles stay
+// in flight ACROSS barriers, st_16x32 XOR swizzle applied to BOTH the
+// glds source position and the ds_read_b128 offset (same involution) to
+// kill bank conflicts, s_setprio(1) around each MFMA cluster, XCD-aware
+// bijective tile remap.
+//
+// Both operands are consumed K-contiguous: A is row-major [M][K]; B is
+// pre-transposed to Bt[N][K] by transpose_bf16 (driven by the launcher)
+// so the LDS images stay lane-linear for glds.
+//
+// Schedule (4 phases per K-tile X, cooperative 128x128 C-quadrants:
+// all 8 waves work on one quadrant per phase):
+//   ph1: Q(0,0)  reads A-half0+B-half0   stages B-half0 of tile X+1
+//   ph2: Q(0,1)  reads B-half1 (A reuse)  stages A-half1 of tile X+1
+//   ph3: Q(1,1)  reads A-half1 (B reuse)  stages A-half0 of tile X+2
+//   ph4: Q(1,0)  reads B-half0 (A reuse)  stages B-half1 of tile X+2
+//        + s_waitcnt vmcnt(4) (once per K-tile)
+// Slot-reuse: a stage targets a slot only >= 1 barrier after its last
+// read; landing: vmcnt(4) at each ph4 retires every stage older than
+// the last two, which covers every read deadline (B0/A1 of X+1 staged
+// at X.ph1/2 are retired by X.ph4's wait; A0/B1 of X+2 staged at
+// X.ph3/4 are retired by X+1.ph4's wait, read at X+2.ph1/2).
+//
+// Fast path requires M%256==0, N%256==0, K%128==0 (dispatch falls back
+// to the general 128^2 kernel otherwise).
+
+#include "common.h"
+
+#include <hip/hip_bf16.h>
+
+namespace {
+
+constexpr int BM = 256;
+constexpr int BN = 256;
+constexpr int BK = 64;
+constexpr int THREADS = 512;
+
+using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ uint16_t f32_to_bf16_rne(float f) {
+  union {
+    uint32_t u;
+    float f;
+  } cvt;
+  cvt.f = f;
+  uint32_t lsb = (cvt.u >> 16) & 1;
+  cvt.u += 0x7FFF + lsb;
+  return (uint16_t)(cvt.u >> 16);
+}
+
+// st_16x32 swizzle: flip bit5 (32 B) conditioned on bit9 (512 B).
+// Involution; row bits untouched.
+__device__ __forceinline__ unsigned swz(unsigned byte_off) {
+  return byte_off ^ (((byte_off >> 9) & 1u) << 5);
+}
+
+// LDS: [buf(2)][op(2: A=0, Bt=1)][half(2)][128 rows][128 B] = 128 KiB
+constexpr unsigned kHalfBytes = 128 * 128;
+constexpr unsigned kOpBytes = 2 * kHalfBytes;
+constexpr unsigned kBufBytes = 2 * kOpBytes;
+constexpr unsigned kLdsBytes = 2 * kBufBytes;
+
+__device__ __forceinline__ unsigned lds_off(int buf, int op, int half) {
+  return (unsigned)buf * kBufBytes + (unsigned)op * kOpBytes +
+         (unsigned)half * kHalfBytes;
+}
+
+using lds_void = __attribute__((address_space(3))) void;
+using global_void = const __attribute__((address_space(1))) void;
+
+__global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
+    const uint16_t* __restrict__ A,   // [M][K]
+    const uint16_t* __restrict__ Bt,  // [N][K]
+    uint16_t* __restrict__ C,         // [M][N]
+    int M, int N, int K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  int nwg = tiles_m * tiles_n;
+  int wgid = blockIdx.x;
+  {
+    const int nxcd = 8;
+    int q = nwg / nxcd, r = nwg % nxcd;
+    int xcd = wgid % nxcd, idx = wgid / nxcd;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int row0 = (wgid / tiles_n) * BM;
+  const int col0 = (wgid % tiles_n) * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;      // 0..7
+  const int wrow = wave >> 2;     // 0..1: 64-row band inside a quadrant
+  const int wcol = wave & 3;      // 0..3: 32-col band inside a quadrant
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+
+  // acc[quadrant(mq*2+nq)][mt 0..3][nt 0..1]
+  f32x4 acc[4][4][2] = {};
+
+  const uint16_t* op_src[2][2] = {
+      {A + (int64_t)row0 * K, A + (int64_t)(row0 + 128) * K},
+      {Bt + (int64_t)col0 * K, Bt + (int64_t)(col0 + 128) * K},
+  };
+
+  // stage one half-tile (128 rows x 64 k bf16 = 16 KiB): 2 glds per
+  // thread; LDS image is lane-linear, swizzle goes on the SOURCE column
+  auto stage_half = [&](int op, int half, int ktile) {
+    const uint16_t* src = op_src[op][half];
+    unsigned base = lds_off(ktile & 1, op, half);
+    int k0 = ktile * BK;
+#pragma unroll
+    for (int q = 0; q < 2; q++) {
+      int row = q * 64 + (tid >> 3);
+      unsigned image_byte = (unsigned)row * 128 + (unsigned)(tid & 7) * 16;
+      int lcol = (int)((swz(image_byte) >> 4) & 7);
+      const uint16_t* gsrc = src + (int64_t)row * K + k0 + lcol * 8;
+      unsigned dst_off =
+          base + (unsigned)(q * 64 * 128) + (unsigned)(tid >> 6) * 1024u;
+      __builtin_amdgcn_global_load_lds(
+          (global_void*)gsrc, (lds_void*)(smem + dst_off), 16, 0, 0);
+    }
+  };
+
+  bf16x8 a_frag[4][2];  // 4 m-frags x 2 k-chunks (one quadrant, this wave)
+  bf16x8 b_frag[2][2];  // 2 n-frags x 2 k-chunks
+
+  auto load_a = [&](int buf, int mq) {
+    unsigned base = lds_off(buf, 0, mq);
+#pragma unroll
+    for (int t = 0; t < 4; t++)
+#pragma unroll
+      for (int ch = 0; ch < 2; ch++) {
+        unsigned row = (unsigned)(wrow * 64 + t * 16 + l15);
+        a_frag[t][ch] = *reinterpret_cast<const bf16x8*>(
+            &smem[base + swz(row * 128 + (unsigned)(ch * 4 + l4) * 16)]);
+      }
+  };
+  auto load_b = [&](int buf, int nq) {
+    unsigned base = lds_off(buf, 1, nq);
+#pragma unroll
+    for (int t = 0; t < 2; t++)
+#pragma unroll
+      for (int ch = 0; ch < 2; ch++) {
+        unsigned row = (unsigned)(wcol * 32 + t * 16 + l15);
+        b_frag[t][ch] = *reinterpret_cast<const bf16x8*>(
+            &smem[base + swz(row * 128 + (unsigned)(ch * 4 + l4) * 16)]);
+      }
+  };
+
+#define MFMA_QUADRANT(mq, nq)                                              \
+  do {                                                                     \
+    __builtin_amdgcn_s_setprio(1);                                         \
+    _Pragma("unroll") for (int t = 0; t < 4; t++)                          \
+        _Pragma("unroll") for (int n = 0; n < 2; n++)                      \
+            _Pragma("unroll") for (int ch = 0; ch < 2; ch++) {             \
+      acc[(mq)*2 + (nq)][t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(  \
+          a_frag[t][ch], b_frag[n][ch], acc[(mq)*2 + (nq)][t][n], 0, 0, 0); \
+    }                                                                      \
+    __builtin_amdgcn_s_setprio(0);                                         \
+  } while (0)
+
+  const int n_ktiles = K / BK;  // even (K % 128 == 0)
+
+  // prologue: tiles 0 and 1 fully staged, plus tile 2's A0 + B1 (the
+  // halves the steady-state pattern would have staged before tile 0)
+  stage_half(0, 0, 0);
+  stage_half(0, 1, 0);
+  stage_half(1, 0, 0);
+  stage_half(1, 1, 0);
+  stage_half(0, 0, 1);
+  stage_half(0, 1, 1);
+  stage_half(1, 0, 1);
+  stage_half(1, 1, 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (n_ktiles > 2) {
+    stage_half(0, 0, 2);  // A-half0 of tile 2
+    stage_half(1, 1, 2);  // B-half1 of tile 2
+  }
+
+  for (int kt = 0; kt < n_ktiles; kt++) {
+    const int buf = kt & 1;
+    const bool s1 = kt + 1 < n_ktiles;   // stage tile kt+1 halves
+    const bool s2 = kt + 2 < n_ktiles;   // stage tile kt+2 halves
+
+    // ph1: Q(0,0)
+    load_a(buf, 0);
+    load_b(buf, 0);
+    if (s1) stage_half(1, 0, kt + 1);  // B-half0(kt+1)
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    MFMA_QUADRANT(0, 0);
+    __builtin_amdgcn_s_barrier();
+
+    // ph2: Q(0,1)
+    load_b(buf, 1);
+    if (s1) stage_half(0, 1, kt + 1);  // A-half1(kt+1)
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    MFMA_QUADRANT(0, 1);
+    __builtin_amdgcn_s_barrier();
+
+    // ph3: Q(1,1)
+    load_a(buf, 1);
+    if (s2) stage_half(0, 0, kt + 2);  // A-half0(kt+2)
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    MFMA_QUADRANT(1, 1);
+    __builtin_amdgcn_s_barrier();
+
+    // ph4: Q(1,0)
+    load_b(buf, 0);
+    if (s2) stage_half(1, 1, kt + 2);  // B-half1(kt+2)
+    if (s1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    MFMA_QUADRANT(1, 0);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: acc -> bf16 C stores
+#pragma unroll
+  for (int mq = 0; mq < 2; mq++) {
+#pragma unroll
+    for (int nq = 0; nq < 2; nq++) {
+      const int r0 = row0 + mq * 128 + wrow * 64 + 4 * l4;
+      const int c0 = col0 + nq * 128 + wcol * 32 + l15;
+#pragma unroll
+      for (int mt = 0; mt < 4; mt++)
+#pragma unroll
+        for (int nt = 0; nt < 2; nt++)
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            C[(int64_t)(r0 + mt * 16 + r) * N + c0 + nt * 16] =
+                f32_to_bf16_rne(acc[mq * 2 + nq][mt][nt][r]);
+          }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bf16 transpose: out[N][K] = in[K][N]^T. 64x64 LDS tiles (pad 72 u16).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void transpose_bf16_kernel(
+    const uint16_t* __restrict__ in, uint16_t* __restrict__ out, int K,
+    int N) {
+  __shared__ uint16_t tile[64][72];
+  int k0 = blockIdx.x * 64;
+  int n0 = blockIdx.y * 64;
+  int tid = threadIdx.x;
+  int lr = tid >> 3;
+  int lc = (tid & 7) * 8;
+#pragma unroll
+  for (int q = 0; q < 2; q++) {
+    int k = k0 + lr + q * 32;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      int n = n0 + lc + j;
+      tile[lc + j][lr + q * 32] =
+          (k < K && n < N) ? in[(int64_t)k * N + n] : 0;
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int q = 0; q < 2; q++) {
+    int n = n0 + lr + q * 32;
+    if (n >= N) continue;
+    uint16_t* dst = out + (int64_t)n * K + k0 + lc;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      if (k0 + lc + j < K) dst[j] = tile[lr + q * 32][lc + j];
+  }
+}
+
+}  // namespace
+
+void launch_transpose_bf16(const uint16_t* in, uint16_t* out, int k, int n,
+                           hipStream_t stream) {
+  dim3 grid((k + 63) / 64, (n + 63) / 64);
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream, in,
+                     out, k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+bool gemm_bf16_256_supported(int m, int n, int k) {
+  return m % 256 == 0 && n % 256 == 0 && k % 128 == 0;
+}
+
+void launch_gemm_bf16_256(const uint16_t* a, const uint16_t* bt, uint16_t* c,
+                          int m, int n, int k, hipStream_t stream) {
+  int tiles_m = m / BM;
+  int tiles_n = n / BN;
+  hipLaunchKernelGGL(gemm_bf16_256_kernel, dim3(tiles_m * tiles_n),
+                     dim3(THREADS), kLdsBytes, stream, a, bt, c, m, n, k,
+                     tiles_m, tiles_n);
+  HIP_CHECK(hipGetLastError());
+}

@@ -468,7 +468,16 @@ PyObject* py_gemm(PyObject*, PyObject* args) {
   else if (dt == 0)
     launch_gemm_f32((const float*)a.ptr, (const float*)b.ptr, (float*)out, m,
                     n, k, g.compute);
-  else
+  else if (gemm_bf16_256_supported(m, n, k)) {
+    // fast path: pre-transpose B to [N][K] (bandwidth-bound, ~1-2% of
+    // GEMM time) so both operands stream K-contiguous through glds
+    void* bt = pool_alloc((int64_t)n * k * 2);
+    launch_transpose_bf16((const uint16_t*)b.ptr, (uint16_t*)bt, k, n,
+                          g.compute);
+    launch_gemm_bf16_256((const uint16_t*)a.ptr, (const uint16_t*)bt,
+                         (uint16_t*)out, m, n, k, g.compute);
+    pool_free(bt, (int64_t)n * k * 2);
+  } else
     launch_gemm_bf16((const uint16_t*)a.ptr, (const uint16_t*)b.ptr,
                      (uint16_t*)out, m, n, k, g.compute);
   Py_END_ALLOW_THREADS;
