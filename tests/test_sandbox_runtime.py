@@ -1,0 +1,95 @@
+"""Unit tests for the sandbox runtime (import scan, hooks, script runner)
+-- runs in-process on CPU."""
+
+import subprocess
+import sys
+from pathlib import Path
+
+RUNTIME_DIR = Path(__file__).resolve().parent.parent / "code_interpreter_amd" / "executor"
+sys.path.insert(0, str(RUNTIME_DIR))
+
+import sandbox_runtime  # noqa: E402
+
+
+def test_scan_missing_imports_stdlib_ignored():
+    assert sandbox_runtime.scan_missing_imports("import os, sys\nimport json") == []
+
+
+def test_scan_missing_imports_installed_ignored():
+    assert sandbox_runtime.scan_missing_imports("import numpy\nimport scipy") == []
+
+
+def test_scan_missing_imports_finds_missing_with_alias():
+    out = sandbox_runtime.scan_missing_imports(
+        "import definitely_not_installed_xyz\nimport fitz\n"
+    )
+    assert "definitely_not_installed_xyz" in out
+    # alias map: fitz -> pymupdf
+    assert "pymupdf" in out
+
+
+def test_scan_missing_imports_from_import():
+    out = sandbox_runtime.scan_missing_imports("from not_a_module_abc.sub import x")
+    assert out == ["not_a_module_abc"]
+
+
+def test_scan_syntax_error_returns_empty():
+    assert sandbox_runtime.scan_missing_imports("def broken(:") == []
+
+
+def test_run_user_script_exit_codes(tmp_path):
+    # run in a subprocess so the import hook does not leak into pytest
+    code = """
+import sys
+sys.path.insert(0, {runtime!r})
+import sandbox_runtime
+import pathlib
+p = pathlib.Path({tmp!r})
+(p / "ok.py").write_text("print('fine')")
+(p / "boom.py").write_text("raise ValueError('nope')")
+(p / "exit3.py").write_text("import sys; sys.exit(3)")
+assert sandbox_runtime.run_user_script(str(p / "ok.py")) == 0
+assert sandbox_runtime.run_user_script(str(p / "boom.py")) == 1
+assert sandbox_runtime.run_user_script(str(p / "exit3.py")) == 3
+print("ALL_OK")
+""".format(runtime=str(RUNTIME_DIR), tmp=str(tmp_path))
+    r = subprocess.run(
+        [sys.executable, "-c", code], capture_output=True, text=True, timeout=60
+    )
+    assert "ALL_OK" in r.stdout, r.stderr
+
+
+def test_matplotlib_show_captures_plot(tmp_path, executor_bin):
+    """Headless artifact capture parity: plt.show() writes plot.png into
+    the workspace (reference sitecustomize.py:9-12)."""
+    import pytest
+
+    try:
+        import matplotlib  # noqa: F401
+    except ImportError:
+        pytest.skip("matplotlib not installed")
+    import asyncio
+
+    from code_interpreter_amd.services.local_executor import LocalPoolExecutor
+    from code_interpreter_amd.services.storage import Storage
+
+    async def run():
+        ex = LocalPoolExecutor(
+            Storage(str(tmp_path / "s")),
+            pool_target_length=1,
+            gpu_count=0,
+            executor_root=str(tmp_path / "e"),
+            dep_install=False,
+        )
+        try:
+            r = await ex.execute(
+                "import matplotlib\nmatplotlib.use('Agg')\n"
+                "import matplotlib.pyplot as plt\n"
+                "plt.plot([1, 2, 3])\nplt.show()\n"
+            )
+            assert r.exit_code == 0, r.stderr
+            assert "/workspace/plot.png" in r.files
+        finally:
+            await ex.aclose()
+
+    asyncio.run(run())
