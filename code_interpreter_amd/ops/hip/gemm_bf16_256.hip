@@ -180,10 +180,9 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
   stage_half(1, 1, 1);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  if (n_ktiles > 2) {
-    stage_half(0, 0, 2);  // A-half0 of tile 2
-    stage_half(1, 1, 2);  // B-half1 of tile 2
-  }
+  // (no pre-stage of tile 2 here: its slots are tile 0's live buffer;
+  // the steady-state pattern stages them at tile 0's ph3/ph4, and the
+  // vmcnt(4) at tile 1's ph4 retires them before tile 2 reads)
 
   for (int kt = 0; kt < n_ktiles; kt++) {
     const int buf = kt & 1;
