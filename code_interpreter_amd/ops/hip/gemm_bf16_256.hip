@@ -21,6 +21,11 @@
 //   ph3: Q(1,1)  reads A-half1 (B reuse)  stages A-half0 of tile X+2
 //   ph4: Q(1,0)  reads B-half0 (A reuse)  stages B-half1 of tile X+2
 //        + s_waitcnt vmcnt(4) (once per K-tile)
+// One barrier per phase (before the MFMA cluster): it both publishes the
+// staged half-tiles (each wave waits its own glds with the counted vmcnt
+// first) and fences the next phase's reads/stages against this phase's
+// cross-wave reads. A second post-MFMA barrier is redundant: MFMA touches
+// registers only.
 // Slot-reuse: a stage targets a slot only >= 1 barrier after its last
 // read; landing: vmcnt(4) at each ph4 retires every stage older than
 // the last two, which covers every read deadline (B0/A1 of X+1 staged
@@ -196,7 +201,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     MFMA_QUADRANT(0, 0);
-    __builtin_amdgcn_s_barrier();
 
     // ph2: Q(0,1)
     load_b(buf, 1);
@@ -204,7 +208,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     MFMA_QUADRANT(0, 1);
-    __builtin_amdgcn_s_barrier();
 
     // ph3: Q(1,1)
     load_a(buf, 1);
@@ -212,7 +215,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     MFMA_QUADRANT(1, 1);
-    __builtin_amdgcn_s_barrier();
 
     // ph4: Q(1,0)
     load_b(buf, 0);
@@ -221,7 +223,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     MFMA_QUADRANT(1, 0);
-    __builtin_amdgcn_s_barrier();
   }
 
   // epilogue: acc -> bf16 C stores
