@@ -21,11 +21,15 @@
 //   ph3: Q(1,1)  reads A-half1 (B reuse)  stages A-half0 of tile X+2
 //   ph4: Q(1,0)  reads B-half0 (A reuse)  stages B-half1 of tile X+2
 //        + s_waitcnt vmcnt(4) (once per K-tile)
-// One barrier per phase (before the MFMA cluster): it both publishes the
-// staged half-tiles (each wave waits its own glds with the counted vmcnt
-// first) and fences the next phase's reads/stages against this phase's
-// cross-wave reads. A second post-MFMA barrier is redundant: MFMA touches
-// registers only.
+// TWO barriers per K-tile (not per phase): ph4's barrier (after the
+// per-wave vmcnt) publishes the next tile's staged halves before any
+// wave reads them; ph3's barrier fences this tile's ph1/ph2 cross-wave
+// reads before the ph3/ph4 stages overwrite those slots. ph1/ph2 need no
+// barrier: their reads hit the buffer published at the previous ph4, and
+// their stages target slots last read before that same barrier. Each
+// MFMA cluster is guarded by its own per-wave lgkmcnt(0) +
+// sched_barrier(0) (hipcc can hoist register-only MFMAs past an
+// inline-asm wait, guide rule 18).
 // Slot-reuse: a stage targets a slot only >= 1 barrier after its last
 // read; landing: vmcnt(4) at each ph4 retires every stage older than
 // the last two, which covers every read deadline (B0/A1 of X+1 staged
@@ -198,30 +202,33 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     load_a(buf, 0);
     load_b(buf, 0);
     if (s1) stage_half(1, 0, kt + 1);  // B-half0(kt+1)
-    __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
     MFMA_QUADRANT(0, 0);
 
     // ph2: Q(0,1)
     load_b(buf, 1);
     if (s1) stage_half(0, 1, kt + 1);  // A-half1(kt+1)
-    __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
     MFMA_QUADRANT(0, 1);
 
-    // ph3: Q(1,1)
+    // ph3: Q(1,1) -- barrier BEFORE the stage: all waves must be past
+    // their ph1/ph2 reads of the slots ph3/ph4 overwrite
     load_a(buf, 1);
-    if (s2) stage_half(0, 0, kt + 2);  // A-half0(kt+2)
     __builtin_amdgcn_s_barrier();
+    if (s2) stage_half(0, 0, kt + 2);  // A-half0(kt+2)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
     MFMA_QUADRANT(1, 1);
 
-    // ph4: Q(1,0)
+    // ph4: Q(1,0) -- counted vmcnt + publication barrier for next tile
     load_b(buf, 0);
     if (s2) stage_half(1, 1, kt + 2);  // B-half1(kt+2)
     if (s1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
     MFMA_QUADRANT(1, 0);
   }
 

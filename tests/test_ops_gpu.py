@@ -180,7 +180,9 @@ def test_gemm_f64_identity_asymmetric(hip):
     np.testing.assert_allclose(c, b)
 
 
-@pytest.mark.parametrize("shape", [(256, 256, 256), (128, 256, 96)])
+@pytest.mark.parametrize(
+    "shape", [(256, 256, 256), (128, 256, 96), (512, 512, 512), (256, 512, 384)]
+)
 def test_gemm_bf16(hip, shape):
     m, n, k = shape
     rng = np.random.default_rng(7)
