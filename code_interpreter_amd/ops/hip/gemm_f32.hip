@@ -1,20 +1,23 @@
 // f32 GEMM on the gfx950 matrix cores.
 //
 // Uses v_mfma_f32_32x32x2_f32 (exact f32 in/accumulate at the 157 TF f32
-// vector rate -- there is no TF32/xf32 on CDNA4, and this is ~2.4x an f32
-// VALU kernel at identical numerics). Structure: 128x128 block tile,
-// 4 waves as 2x2, each wave 2x2 tiles of 32x32 (16-reg f32 accumulators),
-// BK=32 K-steps staged through LDS, A transposed into LDS at stage time so
-// both MFMA operand reads are bank-conflict-free (A rows padded +1 lane).
+// vector rate -- no TF32/xf32 on CDNA4; ~2.4x an f32 VALU kernel at
+// identical numerics). Structure: 128x128 block tile, 4 waves as 2x2,
+// each wave 2x2 tiles of 32x32 (16-reg f32 accumulators), BK=32 K-steps,
+// double-buffered LDS with the async-stage split (T14): the next tile's
+// global loads are issued BEFORE this tile's MFMA loop (HBM latency
+// hides under compute), and written to the other LDS buffer after a
+// counted drain -- one barrier per K-tile. A is transposed into LDS at
+// write time (As[k][m], +1 pad) so both operand reads are
+// bank-conflict-free.
 //
-// Operand layout for mfma_f32_32x32x2_f32 (one f32 VGPR per lane each):
+// Operand layout for mfma_f32_32x32x2_f32 (one f32 per lane each):
 //   A: lane l supplies A[i = l&31][k = l>>5]
 //   B: lane l supplies B[k = l>>5][j = l&31]
 //   C/D (16 regs): col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
 //
-// Arbitrary M, N, K: out-of-range stage reads are zero-filled, C stores are
-// bounds-guarded. blockIdx is remapped XCD-aware (bijective) so neighbor
-// tiles share an XCD's L2.
+// Arbitrary M, N, K (bounds-checked staging + guarded C stores);
+// XCD-aware bijective tile remap.
 
 #include "common.h"
 
@@ -23,14 +26,13 @@ namespace {
 constexpr int BM = 128;
 constexpr int BN = 128;
 constexpr int BK = 32;
-constexpr int THREADS = 256;  // 4 waves: 2x2
+constexpr int THREADS = 256;
 
 using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
 __global__ __launch_bounds__(THREADS) void gemm_f32_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
-  // bijective XCD-aware tile remap (8 XCDs)
   int nwg = tiles_m * tiles_n;
   int wgid = blockIdx.x;
   {
@@ -44,93 +46,112 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_kernel(
   int row0 = tile_m * BM;
   int col0 = tile_n * BN;
 
-  // LDS: As transposed [BK][BM+1] (pad kills the k-major write conflict),
-  // Bs natural [BK][BN]
-  __shared__ float As[BK][BM + 1];
-  __shared__ float Bs[BK][BN];
+  // double-buffered: As transposed [BK][BM+1], Bs natural [BK][BN]
+  __shared__ float As[2][BK][BM + 1];
+  __shared__ float Bs[2][BK][BN];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;      // 0..3
-  const int wave_m = wave >> 1;   // 0..1
-  const int wave_n = wave & 1;    // 0..1
+  const int wave = tid >> 6;
+  const int wave_m = wave >> 1;
+  const int wave_n = wave & 1;
+  const int l31 = lane & 31;
+  const int lk = lane >> 5;
 
-  // each wave owns a 64x64 output tile: 2x2 MFMA tiles of 32x32
   f32x16 acc[2][2] = {};
 
-  const int l31 = lane & 31;
-  const int lk = lane >> 5;  // 0/1: the k-slot this lane supplies
+  // staging registers: 4 float4 of A (rows tid>>3 + {0,32,64,96}, cols
+  // 4*(tid&7)) and 4 float4 of B (rows tid>>5 + {0,8,16,24}, cols
+  // 4*(tid&31))
+  const int a_m = tid >> 3;
+  const int a_k = (tid & 7) * 4;
+  const int b_k = tid >> 5;
+  const int b_n = (tid & 31) * 4;
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // ---- stage A[row0:row0+128][k0:k0+32] -> As[k][m] (transposed) ----
-    // 256 threads x 4 floats: thread t covers row m = t>>3, cols 4*(t&7)..
-    {
-      int m = tid >> 3;          // 0..31 x4 iterations
-      int kq = (tid & 7) * 4;    // 0,4,..28
-      for (int mm = m; mm < BM; mm += 32) {
-        int gr = row0 + mm;
-        float v0 = 0, v1 = 0, v2 = 0, v3 = 0;
-        if (gr < M) {
-          int gk = k0 + kq;
-          const float* src = A + (int64_t)gr * K + gk;
-          if (gk + 3 < K) {
-            v0 = src[0]; v1 = src[1]; v2 = src[2]; v3 = src[3];
-          } else {
-            if (gk + 0 < K) v0 = src[0];
-            if (gk + 1 < K) v1 = src[1];
-            if (gk + 2 < K) v2 = src[2];
-            if (gk + 3 < K) v3 = src[3];
-          }
-        }
-        As[kq + 0][mm] = v0;
-        As[kq + 1][mm] = v1;
-        As[kq + 2][mm] = v2;
-        As[kq + 3][mm] = v3;
+  float4 a_reg[4], b_reg[4];
+
+  auto issue_loads = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      int gr = row0 + a_m + i * 32;
+      int gk = k0 + a_k;
+      if (gr < M && gk + 3 < K) {
+        a_reg[i] = *reinterpret_cast<const float4*>(&A[(int64_t)gr * K + gk]);
+      } else {
+        float v[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          v[j] = (gr < M && gk + j < K) ? A[(int64_t)gr * K + gk + j] : 0.0f;
+        a_reg[i] = {v[0], v[1], v[2], v[3]};
       }
     }
-    // ---- stage B[k0:k0+32][col0:col0+128] -> Bs[k][n] ----
-    {
-      int kq = tid >> 5;         // 0..7 x4 iterations
-      int n = (tid & 31) * 4;    // 0,4,..124
-      for (int kk = kq; kk < BK; kk += 8) {
-        int gk = k0 + kk;
-        float v0 = 0, v1 = 0, v2 = 0, v3 = 0;
-        if (gk < K) {
-          int gn = col0 + n;
-          const float* src = B + (int64_t)gk * N + gn;
-          if (gn + 3 < N) {
-            v0 = src[0]; v1 = src[1]; v2 = src[2]; v3 = src[3];
-          } else {
-            if (gn + 0 < N) v0 = src[0];
-            if (gn + 1 < N) v1 = src[1];
-            if (gn + 2 < N) v2 = src[2];
-            if (gn + 3 < N) v3 = src[3];
-          }
-        }
-        float4 v = {v0, v1, v2, v3};
-        *reinterpret_cast<float4*>(&Bs[kk][n]) = v;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      int gk = k0 + b_k + i * 8;
+      int gn = col0 + b_n;
+      if (gk < K && gn + 3 < N) {
+        b_reg[i] = *reinterpret_cast<const float4*>(&B[(int64_t)gk * N + gn]);
+      } else {
+        float v[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          v[j] = (gk < K && gn + j < N) ? B[(int64_t)gk * N + gn + j] : 0.0f;
+        b_reg[i] = {v[0], v[1], v[2], v[3]};
       }
     }
-    __syncthreads();
+  };
 
-    // ---- MFMA inner loop: BK/2 = 16 steps of K=2 ----
-    const int am0 = wave_m * 64;
-    const int bn0 = wave_n * 64;
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      As[buf][a_k + 0][a_m + i * 32] = a_reg[i].x;
+      As[buf][a_k + 1][a_m + i * 32] = a_reg[i].y;
+      As[buf][a_k + 2][a_m + i * 32] = a_reg[i].z;
+      As[buf][a_k + 3][a_m + i * 32] = a_reg[i].w;
+    }
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      *reinterpret_cast<float4*>(&Bs[buf][b_k + i * 8][b_n]) = b_reg[i];
+    }
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 64;
+
+  auto compute_tile = [&](int buf) {
 #pragma unroll
     for (int ks = 0; ks < BK; ks += 2) {
-      float a0 = As[ks + lk][am0 + l31];
-      float a1 = As[ks + lk][am0 + 32 + l31];
-      float b0 = Bs[ks + lk][bn0 + l31];
-      float b1 = Bs[ks + lk][bn0 + 32 + l31];
+      float a0 = As[buf][ks + lk][am0 + l31];
+      float a1 = As[buf][ks + lk][am0 + 32 + l31];
+      float b0 = Bs[buf][ks + lk][bn0 + l31];
+      float b1 = Bs[buf][ks + lk][bn0 + 32 + l31];
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
     }
+  };
+
+  // prologue: tile 0 staged synchronously
+  issue_loads(0);
+  write_lds(0);
+  __syncthreads();
+
+  // one barrier per K-tile: the write targets the OTHER buffer, whose
+  // last cross-wave readers were separated by the previous iteration's
+  // barrier; this iteration's readers only touch buf cur
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    bool have_next = k0 + BK < K;
+    if (have_next) issue_loads(k0 + BK);  // HBM latency hides under MFMA
+    compute_tile(cur);
+    if (have_next) {
+      write_lds(cur ^ 1);
+      cur ^= 1;
+    }
     __syncthreads();
   }
 
-  // ---- epilogue: C writes (bounds-guarded) ----
   const int crow0 = row0 + wave_m * 64 + 4 * (lane >> 5);
   const int ccol0 = col0 + wave_n * 64 + l31;
 #pragma unroll
