@@ -265,22 +265,34 @@ class CustomToolExecutor:
             if isinstance(n, (ast.Import, ast.ImportFrom))
         )
         # Imports re-emitted at top level so the executor's dependency
-        # auto-install sees them; user prints are swallowed so stdout is
-        # exactly one JSON document.
+        # auto-install sees them. stdout is suppressed at the FD level
+        # while the tool runs (python-level redirect misses native
+        # libraries -- e.g. RCCL prints a banner straight to fd 1), so
+        # stdout is exactly one JSON document. The __main__ guard keeps
+        # multiprocessing-spawn children (multi-GPU tools) from re-running
+        # the wrapper when they re-import it as __mp_main__.
         script = f"""# tool dependency imports (re-emitted for dependency detection)
 {import_lines}
 
-import contextlib
 import json
+import os
+import sys
 
 import pydantic
 
-with contextlib.redirect_stdout(None):
-    tool_globals = {{}}
-    exec(compile({clean_source!r}, "<custom-tool>", "exec"), tool_globals)
-    tool_result = pydantic.TypeAdapter(tool_globals[{function_def.name!r}]).validate_json({tool_input_json!r})
+if __name__ == "__main__":
+    saved_stdout_fd = os.dup(1)
+    devnull_fd = os.open(os.devnull, os.O_WRONLY)
+    os.dup2(devnull_fd, 1)
+    try:
+        tool_globals = {{}}
+        exec(compile({clean_source!r}, "<custom-tool>", "exec"), tool_globals)
+        tool_result = pydantic.TypeAdapter(tool_globals[{function_def.name!r}]).validate_json({tool_input_json!r})
+    finally:
+        sys.stdout.flush()
+        os.dup2(saved_stdout_fd, 1)
 
-print(json.dumps(tool_result))
+    print(json.dumps(tool_result))
 """
         result = await self.code_executor.execute(source_code=script, env=dict(env))
         if result.exit_code != 0:
