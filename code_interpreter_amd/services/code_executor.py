@@ -24,10 +24,15 @@ kubernetes_code_executor.py:81-149).
 """
 
 import asyncio
+from contextvars import ContextVar
 from dataclasses import dataclass, field
 from typing import Mapping, Optional
 
 import httpx
+
+# request-id propagation into the executor (logged there; SURVEY.md
+# section 5: pass the request id to the pod as a header)
+REQUEST_ID: ContextVar = ContextVar("sandbox_request_id", default="-")
 
 from code_interpreter_amd.services.storage import Storage
 
@@ -160,8 +165,14 @@ class SandboxClient:
         body: dict = {"source_code": source_code, "env": dict(env)}
         if timeout is not None:
             body["timeout"] = timeout
+        headers = {}
+        request_id = REQUEST_ID.get()
+        if request_id and request_id != "-":
+            headers["X-Request-Id"] = request_id
         t0 = _time.perf_counter()
-        resp = await self._client.post(f"{prefix}/execute", json=body)
+        resp = await self._client.post(
+            f"{prefix}/execute", json=body, headers=headers
+        )
         t_exec = (_time.perf_counter() - t0) * 1000
         if resp.status_code != 200:
             raise ExecutorError(f"execute failed: {resp.status_code} {resp.text!r}")

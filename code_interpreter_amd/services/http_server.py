@@ -86,6 +86,9 @@ def create_http_server(
     def set_request_id() -> str:
         request_id = str(uuid.uuid4())
         request_id_context_var.set(request_id)
+        from code_interpreter_amd.services.code_executor import REQUEST_ID
+
+        REQUEST_ID.set(request_id)
         return request_id
 
     @app.post("/v1/execute", response_model=ExecuteResponse)

@@ -51,3 +51,29 @@ def test_reader_validates_hash(tmp_path):
             await storage.read("../../etc/passwd")
 
     asyncio.run(run())
+
+
+def test_storage_gc(tmp_path):
+    import os
+    import time as time_mod
+
+    from code_interpreter_amd.storage_gc import collect
+
+    async def run():
+        storage = Storage(str(tmp_path))
+        old = await storage.write(b"old")
+        new = await storage.write(b"new")
+        return old, new
+
+    old, new = asyncio.run(run())
+    past = time_mod.time() - 100_000
+    os.utime(tmp_path / old, (past, past))
+
+    stats = collect(str(tmp_path), ttl_hours=1.0, dry_run=True)
+    assert stats == {"removed": 1, "kept": 1, "freed_bytes": 3}
+    assert (tmp_path / old).exists()
+
+    stats = collect(str(tmp_path), ttl_hours=1.0)
+    assert stats["removed"] == 1
+    assert not (tmp_path / old).exists()
+    assert (tmp_path / new).exists()
