@@ -137,11 +137,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     }
   };
 
-  bf16x8 a_frag[4][2];   // 4 m-frags x 2 k-chunks (one quadrant, this wave)
-  bf16x8 b_frag0[2][2];  // n-half 0 fragments
-  bf16x8 b_frag1[2][2];  // n-half 1 fragments (separate array: the next
-                         // half's loads issue BETWEEN the current MFMA
-                         // halves, hiding ds_read latency under MFMA)
+  bf16x8 a_frag[4][2];  // 4 m-frags x 2 k-chunks (one quadrant, this wave)
+  bf16x8 b_frag[2][2];  // 2 n-frags x 2 k-chunks
 
   auto load_a = [&](int buf, int mq) {
     unsigned base = lds_off(buf, 0, mq);
@@ -154,27 +151,26 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
             &smem[base + swz(row * 128 + (unsigned)(ch * 4 + l4) * 16)]);
       }
   };
-  auto load_b = [&](int buf, int nq, bf16x8 (&frag)[2][2]) {
+  auto load_b = [&](int buf, int nq) {
     unsigned base = lds_off(buf, 1, nq);
 #pragma unroll
     for (int t = 0; t < 2; t++)
 #pragma unroll
       for (int ch = 0; ch < 2; ch++) {
         unsigned row = (unsigned)(wcol * 32 + t * 16 + l15);
-        frag[t][ch] = *reinterpret_cast<const bf16x8*>(
+        b_frag[t][ch] = *reinterpret_cast<const bf16x8*>(
             &smem[base + swz(row * 128 + (unsigned)(ch * 4 + l4) * 16)]);
       }
   };
 
-// one half of a quadrant's MFMA cluster: m-frags [T0, T0+2) x 2 n x 2 ch
-#define MFMA_HALF(mq, nq, BF, T0)                                          \
+#define MFMA_QUADRANT(mq, nq)                                              \
   do {                                                                     \
     __builtin_amdgcn_s_setprio(1);                                         \
-    _Pragma("unroll") for (int t = (T0); t < (T0) + 2; t++)                \
+    _Pragma("unroll") for (int t = 0; t < 4; t++)                          \
         _Pragma("unroll") for (int n = 0; n < 2; n++)                      \
             _Pragma("unroll") for (int ch = 0; ch < 2; ch++) {             \
       acc[(mq)*2 + (nq)][t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(  \
-          a_frag[t][ch], BF[n][ch], acc[(mq)*2 + (nq)][t][n], 0, 0, 0);    \
+          a_frag[t][ch], b_frag[n][ch], acc[(mq)*2 + (nq)][t][n], 0, 0, 0); \
     }                                                                      \
     __builtin_amdgcn_s_setprio(0);                                         \
   } while (0)
@@ -202,22 +198,20 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     const bool s1 = kt + 1 < n_ktiles;   // stage tile kt+1 halves
     const bool s2 = kt + 2 < n_ktiles;   // stage tile kt+2 halves
 
-    // ph1: Q(0,0); b_frag1 loads issue between the MFMA halves
+    // ph1: Q(0,0)
     load_a(buf, 0);
-    load_b(buf, 0, b_frag0);
+    load_b(buf, 0);
     if (s1) stage_half(1, 0, kt + 1);  // B-half0(kt+1)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
-    MFMA_HALF(0, 0, b_frag0, 0);
-    load_b(buf, 1, b_frag1);
-    MFMA_HALF(0, 0, b_frag0, 2);
+    MFMA_QUADRANT(0, 0);
 
     // ph2: Q(0,1)
+    load_b(buf, 1);
     if (s1) stage_half(0, 1, kt + 1);  // A-half1(kt+1)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
-    MFMA_HALF(0, 1, b_frag1, 0);
-    MFMA_HALF(0, 1, b_frag1, 2);
+    MFMA_QUADRANT(0, 1);
 
     // ph3: Q(1,1) -- barrier BEFORE the stage: all waves must be past
     // their ph1/ph2 reads of the slots ph3/ph4 overwrite
@@ -226,18 +220,16 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
     if (s2) stage_half(0, 0, kt + 2);  // A-half0(kt+2)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
-    MFMA_HALF(1, 1, b_frag1, 0);
-    load_b(buf, 0, b_frag0);
-    MFMA_HALF(1, 1, b_frag1, 2);
+    MFMA_QUADRANT(1, 1);
 
     // ph4: Q(1,0) -- counted vmcnt + publication barrier for next tile
+    load_b(buf, 0);
     if (s2) stage_half(1, 1, kt + 2);  // B-half1(kt+2)
     if (s1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
-    MFMA_HALF(1, 0, b_frag0, 0);
-    MFMA_HALF(1, 0, b_frag0, 2);
+    MFMA_QUADRANT(1, 0);
   }
 
   // epilogue: acc -> bf16 C stores
