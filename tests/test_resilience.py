@@ -1,0 +1,70 @@
+"""Failure-recovery tests: engine crashes are absorbed by respawn +
+whole-execution retry (the reference's equivalent surface is pod-spawn
+retry + k8s GC, kubernetes_code_executor.py:75-79, 191-195)."""
+
+import asyncio
+import os
+import signal
+
+import pytest
+
+from code_interpreter_amd.services.local_executor import LocalPoolExecutor
+from code_interpreter_amd.services.storage import Storage
+
+
+@pytest.fixture
+def executor(tmp_path, executor_bin):
+    ex = LocalPoolExecutor(
+        Storage(str(tmp_path / "s")),
+        pool_target_length=1,
+        gpu_count=0,
+        executor_root=str(tmp_path / "e"),
+        dep_install=False,
+    )
+    yield ex
+    asyncio.run(ex.aclose())
+
+
+def test_engine_crash_recovers(executor):
+    async def run():
+        r = await executor.execute("print('before')")
+        assert r.exit_code == 0
+        # kill the engine out from under the executor
+        engine = executor._engines[0]
+        os.killpg(engine.proc.pid, signal.SIGKILL)
+        engine.proc.wait()
+        # next execution must respawn and succeed (retry layer)
+        r = await executor.execute("print('after')")
+        assert r.exit_code == 0
+        assert r.stdout == "after\n"
+
+    asyncio.run(run())
+
+
+def test_user_timeout_contract(executor):
+    """Hanging user code is bounded by the executor timeout and reported
+    with the reference's exact shape ('', 'Execution timed out', -1)."""
+    executor.execute_timeout = 2.0
+
+    async def run():
+        r = await executor.execute("import time\ntime.sleep(60)")
+        assert r.exit_code == -1
+        assert r.stderr == "Execution timed out"
+        assert r.stdout == ""
+        # the engine survives a timed-out execution
+        r = await executor.execute("print('alive')")
+        assert r.stdout == "alive\n"
+
+    asyncio.run(run())
+
+
+def test_crash_in_user_code_is_contained(executor):
+    async def run():
+        r = await executor.execute(
+            "import ctypes\nctypes.string_at(0)"  # segfault the sandbox
+        )
+        assert r.exit_code != 0
+        r = await executor.execute("print('still up')")
+        assert r.stdout == "still up\n"
+
+    asyncio.run(run())
