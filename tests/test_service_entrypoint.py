@@ -70,3 +70,31 @@ def test_main_entrypoint_and_health_check(tmp_path, executor_bin):
     finally:
         os.killpg(proc.pid, signal.SIGKILL)
         proc.wait()
+
+
+@pytest.mark.slow
+def test_bench_distributed_contract(tmp_path):
+    """The driver launches bench.py under torch.distributed.run for N>1;
+    this is the same invocation on CPU (gloo, 2 ranks, tiny array)."""
+    import json
+
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
+            str(REPO_ROOT / "bench.py"),
+            "--gpus", "2", "--steps", "4", "--warmup", "1",
+            "--array-size", "200000", "--concurrency", "2",
+        ],
+        capture_output=True,
+        text=True,
+        timeout=300,
+        cwd=str(REPO_ROOT),
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    line = [l for l in proc.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["steps"] == 4
+    assert out["value"] > 0
