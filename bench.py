@@ -87,11 +87,17 @@ def main() -> None:
         if use_gpu:
             torch.cuda.set_device(local_rank)
 
-    # pin this rank's service (and its sandbox engines) to one GPU
+    # pin this rank's service (and its sandbox engines) to one GPU; if the
+    # environment already restricts visibility to a list, take this rank's
+    # entry of that list (inheriting the full list would land every rank's
+    # engines on the same device)
     if use_gpu:
-        os.environ["HIP_VISIBLE_DEVICES"] = os.environ.get(
-            "HIP_VISIBLE_DEVICES", str(local_rank)
-        )
+        existing = os.environ.get("HIP_VISIBLE_DEVICES")
+        if existing:
+            ids = [x for x in existing.split(",") if x.strip() != ""]
+            os.environ["HIP_VISIBLE_DEVICES"] = ids[local_rank % len(ids)]
+        else:
+            os.environ["HIP_VISIBLE_DEVICES"] = str(local_rank)
 
     result = asyncio.run(run_rank(args, rank, world_size, use_gpu))
 
