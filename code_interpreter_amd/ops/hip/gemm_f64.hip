@@ -21,7 +21,7 @@ namespace {
 
 constexpr int BM = 64;
 constexpr int BN = 64;
-constexpr int BK = 32;
+constexpr int BK = 16;
 constexpr int THREADS = 256;
 
 using f64x4 = __attribute__((__vector_size__(4 * sizeof(double)))) double;
@@ -54,25 +54,25 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
   f64x4 acc[2][2] = {};
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A[row0:+64][k0:+32]: thread t -> row t>>2, 8 cols at 8*(t&3)
+    // stage A[row0:+64][k0:+16]: thread t -> row t>>2, 4 cols at 4*(t&3)
     {
       int m = tid >> 2;
-      int kq = (tid & 3) * 8;
+      int kq = (tid & 3) * 4;
       int gr = row0 + m;
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
+      for (int j = 0; j < 4; j++) {
         int gk = k0 + kq + j;
         As[m][kq + j] =
             (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
       }
     }
-    // stage B[k0:+32][col0:+64]: thread t -> row t>>3, 8 cols at 8*(t&7)
+    // stage B[k0:+16][col0:+64]: thread t -> row t>>4, 4 cols at 4*(t&15)
     {
-      int kk = tid >> 3;
-      int n = (tid & 7) * 8;
+      int kk = tid >> 4;
+      int n = (tid & 15) * 4;
       int gk = k0 + kk;
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
+      for (int j = 0; j < 4; j++) {
         int gn = col0 + n + j;
         Bs[kk][n + j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
       }

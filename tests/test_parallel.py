@@ -54,9 +54,13 @@ def test_matmul_allreduce_cpu_smoke():
 
 @pytest.mark.gpu
 def test_matmul_allreduce_rccl_single_gpu():
-    import torch
+    # detect the GPU WITHOUT importing torch here: loading torch's bundled
+    # HIP runtime next to the system-ROCm _hipops (initialized by other
+    # tests in this process) makes torch.cuda report no devices; the
+    # spawned workers are fresh processes and see the GPU fine
+    from code_interpreter_amd.utils.gpus import detect_gpu_count
 
-    if not torch.cuda.is_available():
+    if detect_gpu_count() == 0:
         pytest.skip("no GPU")
     stats = allreduce_matmul_bench(size=2048, dtype="bfloat16", world_size=1, iters=3)
     assert stats["device"] == "cuda"
