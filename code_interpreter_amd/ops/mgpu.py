@@ -171,11 +171,13 @@ def allreduce_matmul_bench(
     dtype: str = "bfloat16",
     world_size: Optional[int] = None,
     iters: int = 10,
+    backend: Optional[str] = None,
 ) -> dict:
     """The BASELINE acceptance workload: per-GPU bf16 size^2 matmul with the
     result all-reduced across all GPUs over xGMI. Returns rank 0's stats."""
     return run_distributed(
         _matmul_allreduce_worker,
         world_size=world_size,
+        backend=backend,
         args=(size, dtype, iters),
     )

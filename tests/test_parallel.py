@@ -62,7 +62,9 @@ def test_matmul_allreduce_rccl_single_gpu():
 
     if detect_gpu_count() == 0:
         pytest.skip("no GPU")
-    stats = allreduce_matmul_bench(size=2048, dtype="bfloat16", world_size=1, iters=3)
+    stats = allreduce_matmul_bench(
+        size=2048, dtype="bfloat16", world_size=1, iters=3, backend="nccl"
+    )
     assert stats["device"] == "cuda"
     assert stats["matmul_tflops_per_gpu"] > 10  # hipBLASLt bf16 on MI355X
 
