@@ -255,6 +255,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_bf16_256_kernel(
 // ---------------------------------------------------------------------------
 // bf16 transpose: out[N][K] = in[K][N]^T. 64x64 LDS tiles (pad 72 u16).
 // ---------------------------------------------------------------------------
+using ushort8 = __attribute__((ext_vector_type(8))) unsigned short;
+
 __global__ __launch_bounds__(256) void transpose_bf16_kernel(
     const uint16_t* __restrict__ in, uint16_t* __restrict__ out, int K,
     int N) {
@@ -264,25 +266,40 @@ __global__ __launch_bounds__(256) void transpose_bf16_kernel(
   int tid = threadIdx.x;
   int lr = tid >> 3;
   int lc = (tid & 7) * 8;
+  const bool interior =
+      (k0 + 63 < K) && (n0 + 63 < N);  // vector fast path (16 B I/O)
 #pragma unroll
   for (int q = 0; q < 2; q++) {
     int k = k0 + lr + q * 32;
+    if (interior) {
+      ushort8 v = *reinterpret_cast<const ushort8*>(&in[(int64_t)k * N + n0 + lc]);
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      int n = n0 + lc + j;
-      tile[lc + j][lr + q * 32] =
-          (k < K && n < N) ? in[(int64_t)k * N + n] : 0;
+      for (int j = 0; j < 8; j++) tile[lc + j][lr + q * 32] = v[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        int n = n0 + lc + j;
+        tile[lc + j][lr + q * 32] =
+            (k < K && n < N) ? in[(int64_t)k * N + n] : 0;
+      }
     }
   }
   __syncthreads();
 #pragma unroll
   for (int q = 0; q < 2; q++) {
     int n = n0 + lr + q * 32;
-    if (n >= N) continue;
-    uint16_t* dst = out + (int64_t)n * K + k0 + lc;
+    if (interior) {
+      ushort8 v;
 #pragma unroll
-    for (int j = 0; j < 8; j++)
-      if (k0 + lc + j < K) dst[j] = tile[lr + q * 32][lc + j];
+      for (int j = 0; j < 8; j++) v[j] = tile[lr + q * 32][lc + j];
+      *reinterpret_cast<ushort8*>(&out[(int64_t)n * K + k0 + lc]) = v;
+    } else {
+      if (n >= N) continue;
+      uint16_t* dst = out + (int64_t)n * K + k0 + lc;
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        if (k0 + lc + j < K) dst[j] = tile[lr + q * 32][lc + j];
+    }
   }
 }
 
