@@ -40,8 +40,10 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
   int row0 = (wgid / tiles_n) * BM;
   int col0 = (wgid % tiles_n) * BN;
 
-  __shared__ double As[BM][BK + 1];
-  __shared__ double Bs[BK][BN + 1];
+  // double-buffered (2 x 17 KiB: still 4 workgroups/CU); next tile's
+  // global loads issue before this tile's MFMAs (T14 async-stage split)
+  __shared__ double As[2][BM][BK + 1];
+  __shared__ double Bs[2][BK][BN + 1];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -53,44 +55,61 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
 
   f64x4 acc[2][2] = {};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A[row0:+64][k0:+16]: thread t -> row t>>2, 4 cols at 4*(t&3)
-    {
-      int m = tid >> 2;
-      int kq = (tid & 3) * 4;
-      int gr = row0 + m;
-#pragma unroll
-      for (int j = 0; j < 4; j++) {
-        int gk = k0 + kq + j;
-        As[m][kq + j] =
-            (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
-      }
-    }
-    // stage B[k0:+16][col0:+64]: thread t -> row t>>4, 4 cols at 4*(t&15)
-    {
-      int kk = tid >> 4;
-      int n = (tid & 15) * 4;
-      int gk = k0 + kk;
-#pragma unroll
-      for (int j = 0; j < 4; j++) {
-        int gn = col0 + n + j;
-        Bs[kk][n + j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
-      }
-    }
-    __syncthreads();
+  const int a_m = tid >> 2;
+  const int a_k = (tid & 3) * 4;
+  const int b_k = tid >> 4;
+  const int b_n = (tid & 15) * 4;
 
-    const int am0 = wave_m * 32;
-    const int bn0 = wave_n * 32;
+  double a_reg[4], b_reg[4];
+
+  auto issue_loads = [&](int k0) {
+    int gr = row0 + a_m;
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int gk = k0 + a_k + j;
+      a_reg[j] = (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
+    }
+    int gk = k0 + b_k;
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int gn = col0 + b_n + j;
+      b_reg[j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) As[buf][a_m][a_k + j] = a_reg[j];
+#pragma unroll
+    for (int j = 0; j < 4; j++) Bs[buf][b_k][b_n + j] = b_reg[j];
+  };
+
+  const int am0 = wave_m * 32;
+  const int bn0 = wave_n * 32;
+  auto compute_tile = [&](int buf) {
 #pragma unroll
     for (int ks = 0; ks < BK; ks += 4) {
-      double a0 = As[am0 + l15][ks + lk];
-      double a1 = As[am0 + 16 + l15][ks + lk];
-      double b0 = Bs[ks + lk][bn0 + l15];
-      double b1 = Bs[ks + lk][bn0 + 16 + l15];
+      double a0 = As[buf][am0 + l15][ks + lk];
+      double a1 = As[buf][am0 + 16 + l15][ks + lk];
+      double b0 = Bs[buf][ks + lk][bn0 + l15];
+      double b1 = Bs[buf][ks + lk][bn0 + 16 + l15];
       acc[0][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b1, acc[1][1], 0, 0, 0);
+    }
+  };
+
+  issue_loads(0);
+  write_lds(0);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    bool have_next = k0 + BK < K;
+    if (have_next) issue_loads(k0 + BK);
+    compute_tile(cur);
+    if (have_next) {
+      write_lds(cur ^ 1);
+      cur ^= 1;
     }
     __syncthreads();
   }
