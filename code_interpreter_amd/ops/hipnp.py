@@ -537,6 +537,9 @@ def install(numpy_module, mode: str = "auto") -> None:
 
     np = numpy_module
     orig_rand = np.random.rand
+    orig_random = np.random.random
+    orig_random_sample = np.random.random_sample
+    orig_uniform = np.random.uniform
     orig_matmul = np.matmul
     orig_dot = np.dot
     orig_square = np.square
@@ -553,6 +556,40 @@ def install(numpy_module, mode: str = "auto") -> None:
                 if mode == "require":
                     raise
         return orig_rand(*shape)
+
+    def _size_elems(size):
+        if size is None:
+            return 1
+        if isinstance(size, int):
+            return size
+        n = 1
+        for s in size:
+            n *= int(s)
+        return n
+
+    def patched_random(size=None):
+        if size is not None and _size_elems(size) >= MIN_ELEMS:
+            try:
+                shape = (size,) if isinstance(size, int) else tuple(size)
+                return rand(*shape)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_random(size)
+
+    def patched_uniform(low=0.0, high=1.0, size=None):
+        if (
+            size is not None
+            and _size_elems(size) >= MIN_ELEMS
+            and isinstance(low, (int, float))
+            and isinstance(high, (int, float))
+        ):
+            try:
+                return uniform_device(float(low), float(high), size)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_uniform(low, high, size)
 
     def patched_matmul(a, b, *args, **kwargs):
         if not args and not kwargs:
@@ -613,6 +650,9 @@ def install(numpy_module, mode: str = "auto") -> None:
         return orig_sum(_asarray(x), *args, **kwargs)
 
     np.random.rand = patched_rand
+    np.random.random = patched_random
+    np.random.random_sample = patched_random
+    np.random.uniform = patched_uniform
     np.matmul = patched_matmul
     np.dot = patched_dot
     np.square = patched_square

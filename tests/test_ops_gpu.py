@@ -248,3 +248,13 @@ def test_hipnp_matmul_matches_numpy(hnp):
     np.testing.assert_allclose(
         np.asarray(c), a.astype(np.float64) @ b.astype(np.float64), rtol=1e-4, atol=1e-3
     )
+
+
+def test_hipnp_random_variants(hnp):
+    import hipnp
+
+    u = hipnp.uniform_device(-2.0, 2.0, 3_000_000)
+    host = np.asarray(u)
+    assert -2.0 <= host.min() < -1.9
+    assert 1.9 < host.max() < 2.0
+    np.testing.assert_allclose(host.mean(), 0.0, atol=5e-3)
