@@ -68,6 +68,12 @@ def main() -> None:
     parser.add_argument("--warmup", type=int, default=32)
     parser.add_argument("--concurrency", type=int, default=8)
     parser.add_argument("--engines-per-gpu", type=int, default=2)
+    parser.add_argument(
+        "--http-workers",
+        type=int,
+        default=2,
+        help="control-plane event loops per rank (uvicorn-workers style)",
+    )
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
     args = parser.parse_args()
@@ -165,50 +171,68 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
     from code_interpreter_amd.application_context import ApplicationContext
     from code_interpreter_amd.config import Config
 
-    tmp = tempfile.mkdtemp(prefix=f"bench-r{rank}-")
-    port = _free_port()
-    config = Config(
-        http_listen_addr=f"127.0.0.1:{port}",
-        file_storage_path=os.path.join(tmp, "storage"),
-        executor_root=os.path.join(tmp, "executors"),
-        executor_backend="local",
-        executor_pool_target_length=max(2, args.concurrency),
-        engines_per_gpu=args.engines_per_gpu,
-        gpu_count=1 if use_gpu else 0,
-        gpu_pinning=False,  # engines inherit this rank's HIP_VISIBLE_DEVICES
-        hip_numpy="require" if use_gpu else "off",
-        dep_install=False,
-    )
-    ctx = ApplicationContext(config)
-
-    server = uvicorn.Server(
-        uvicorn.Config(
-            app=ctx.http_server, host="127.0.0.1", port=port, log_config=None
+    # W control-plane workers (uvicorn-workers style): each worker is a
+    # full service instance (own event loop + engines) on its own port;
+    # requests round-robin across them. One asyncio loop saturates around
+    # ~300 req/s of HTTP handling, well below the sandbox pipeline.
+    workers = max(1, args.http_workers)
+    engines_per_worker = max(1, args.engines_per_gpu // workers)
+    stacks = []
+    for w in range(workers):
+        tmp = tempfile.mkdtemp(prefix=f"bench-r{rank}w{w}-")
+        port = _free_port()
+        config = Config(
+            http_listen_addr=f"127.0.0.1:{port}",
+            file_storage_path=os.path.join(tmp, "storage"),
+            executor_root=os.path.join(tmp, "executors"),
+            executor_backend="local",
+            executor_pool_target_length=max(
+                2, args.concurrency // workers
+            ),
+            engines_per_gpu=engines_per_worker,
+            gpu_count=1 if use_gpu else 0,
+            gpu_pinning=False,  # engines inherit the rank's visible device
+            hip_numpy="require" if use_gpu else "off",
+            dep_install=False,
         )
-    )
-    thread = threading.Thread(target=server.run, daemon=True)
-    thread.start()
+        ctx = ApplicationContext(config)
+        server = uvicorn.Server(
+            uvicorn.Config(
+                app=ctx.http_server, host="127.0.0.1", port=port,
+                log_config=None,
+            )
+        )
+        thread = threading.Thread(target=server.run, daemon=True)
+        thread.start()
+        stacks.append({"ctx": ctx, "server": server, "thread": thread, "port": port})
 
     source = WORKLOAD.format(array_size=args.array_size)
-    async with httpx.AsyncClient(
-        base_url=f"http://127.0.0.1:{port}", timeout=300.0
-    ) as client:
-        # wait for the service + at least one engine
-        deadline = time.time() + 120
-        while time.time() < deadline:
-            try:
-                r = await client.post(
-                    "/v1/execute", json={"source_code": "print('ready')"}
-                )
-                if r.status_code == 200 and r.json()["exit_code"] == 0:
-                    break
-            except httpx.HTTPError:
-                pass
-            await asyncio.sleep(0.25)
-        else:
-            raise RuntimeError("service did not become ready")
+    clients = [
+        httpx.AsyncClient(base_url=f"http://127.0.0.1:{s['port']}", timeout=300.0)
+        for s in stacks
+    ]
+    try:
+        # wait for every worker + at least one engine each
+        deadline = time.time() + 180
+        for client in clients:
+            while time.time() < deadline:
+                try:
+                    r = await client.post(
+                        "/v1/execute", json={"source_code": "print('ready')"}
+                    )
+                    if r.status_code == 200 and r.json()["exit_code"] == 0:
+                        break
+                except httpx.HTTPError:
+                    pass
+                await asyncio.sleep(0.25)
+            else:
+                raise RuntimeError("service did not become ready")
+
+        counter = {"n": 0}
 
         async def one_request() -> float:
+            counter["n"] += 1
+            client = clients[counter["n"] % len(clients)]
             t0 = time.perf_counter()
             resp = await client.post("/v1/execute", json={"source_code": source})
             dt = time.perf_counter() - t0
@@ -252,10 +276,14 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
 
             dist.barrier()
         elapsed = time.perf_counter() - t_start
-
-    server.should_exit = True
-    thread.join(timeout=10)
-    await ctx.code_executor.aclose()
+    finally:
+        for client in clients:
+            await client.aclose()
+        for s in stacks:
+            s["server"].should_exit = True
+        for s in stacks:
+            s["thread"].join(timeout=10)
+            await s["ctx"].code_executor.aclose()
 
     latencies.sort()
     return {
