@@ -71,8 +71,9 @@ def main() -> None:
     parser.add_argument(
         "--http-workers",
         type=int,
-        default=2,
-        help="control-plane event loops per rank (uvicorn-workers style)",
+        default=1,
+        help="control-plane event loops per rank (uvicorn-workers style; "
+        "measured: >1 loses to the shared client loop + CPU quota, keep 1)",
     )
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
@@ -177,6 +178,8 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
     # ~300 req/s of HTTP handling, well below the sandbox pipeline.
     workers = max(1, args.http_workers)
     engines_per_worker = max(1, args.engines_per_gpu // workers)
+    if workers == 1:
+        engines_per_worker = args.engines_per_gpu
     stacks = []
     for w in range(workers):
         tmp = tempfile.mkdtemp(prefix=f"bench-r{rank}w{w}-")
