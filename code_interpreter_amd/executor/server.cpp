@@ -849,15 +849,18 @@ static std::string url_decode(const std::string& s) {
   return out;
 }
 
-static void handle_execute(Conn& conn, const HttpRequest& req,
-                           const std::string& workspace) {
+// returns true when changed files were reported (the caller may keep the
+// session alive for downloads); session != nullptr adds it to the response
+static bool handle_execute(Conn& conn, const HttpRequest& req,
+                           const std::string& workspace,
+                           const std::string* session = nullptr) {
   double t_handler0 = now_ms();
   json::Parser parser(req.body);
   json::Value body = parser.parse();
   if (!parser.ok || !body.is_object() || !body.get("source_code") ||
       !body.get("source_code")->is_string()) {
     conn.respond(400, "Bad Request", "{\"error\":\"invalid request body\"}");
-    return;
+    return false;
   }
   const std::string& source = body.get("source_code")->str;
   double timeout_s = g_cfg.default_timeout;
@@ -881,7 +884,7 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   char* tmpdir = mkdtemp(tmpl);
   if (!tmpdir) {
     conn.respond(500, "Internal Server Error", "{\"error\":\"mkdtemp failed\"}");
-    return;
+    return false;
   }
   std::string script_path = std::string(tmpdir) + "/script.py";
   std::string stdout_path = std::string(tmpdir) + "/stdout";
@@ -925,7 +928,11 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
     if (i) resp += ",";
     resp += json::quote(changed[i]);
   }
-  resp += "]}";
+  resp += "]";
+  if (session != nullptr && !changed.empty()) {
+    resp += ",\"session\":" + json::quote(*session);
+  }
+  resp += "}";
 
   // clean the temp dir
   unlink(script_path.c_str());
@@ -935,6 +942,7 @@ static void handle_execute(Conn& conn, const HttpRequest& req,
   rmdir(tmpdir);
 
   conn.respond(200, "OK", resp);
+  return !changed.empty();
 }
 
 // workspace file routes (PUT/GET), shared by the legacy pod-style routes
@@ -992,6 +1000,26 @@ static void handle_conn(int fd) {
                           req.path.substr(strlen("/workspace/")));
     } else if (req.method == "POST" && req.path == "/execute") {
       handle_execute(conn, req, g_cfg.workspace);
+    } else if (req.method == "POST" && req.path == "/execute-ephemeral") {
+      // one-shot: fresh workspace session created, executed and (when no
+      // files changed) deleted within a single request -- saves two
+      // round trips on the hot path; when files DID change the session
+      // id is returned for downloads + explicit DELETE
+      std::string id = random_hex(12);
+      std::string ws = g_sessions_root + "/" + id;
+      mkdirs(ws);
+      {
+        std::lock_guard<std::mutex> lk(g_sandboxes.mu);
+        g_sandboxes.workspaces[id] = ws;
+      }
+      bool had_files = handle_execute(conn, req, ws, &id);
+      if (!had_files) {
+        {
+          std::lock_guard<std::mutex> lk(g_sandboxes.mu);
+          g_sandboxes.workspaces.erase(id);
+        }
+        rmtree(ws);
+      }
     } else if (req.method == "POST" && req.path == "/sandboxes") {
       // fresh single-use workspace session
       std::string id = random_hex(12);

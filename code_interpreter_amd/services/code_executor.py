@@ -127,8 +127,18 @@ class SandboxClient:
         env: Mapping[str, str] = {},
         timeout: Optional[float] = None,
     ) -> Result:
-        """Create a fresh sandbox session, run, tear it down."""
+        """Run in a fresh single-use sandbox session. Without input files
+        the whole round is ONE request (/execute-ephemeral: the engine
+        creates the workspace, runs, and deletes it when no files changed);
+        with input files (or changed outputs) the explicit session routes
+        handle staging, then the session is torn down."""
         import time as _time
+
+        if not files:
+            try:
+                return await self._run_ephemeral(storage, source_code, env, timeout)
+            except (httpx.HTTPError, OSError) as e:
+                raise ExecutorError(f"sandbox request failed: {e!r}") from e
 
         t0 = _time.perf_counter()
         try:
@@ -145,6 +155,55 @@ class SandboxClient:
             return result
         finally:
             asyncio.ensure_future(self.delete_sandbox(session))
+
+    async def _run_ephemeral(self, storage, source_code, env, timeout) -> Result:
+        body: dict = {"source_code": source_code, "env": dict(env)}
+        if timeout is not None:
+            body["timeout"] = timeout
+        headers = {}
+        request_id = REQUEST_ID.get()
+        if request_id and request_id != "-":
+            headers["X-Request-Id"] = request_id
+        import time as _time
+
+        t0 = _time.perf_counter()
+        resp = await self._client.post(
+            "/execute-ephemeral", json=body, headers=headers
+        )
+        t_exec = (_time.perf_counter() - t0) * 1000
+        if resp.status_code != 200:
+            raise ExecutorError(f"execute failed: {resp.status_code} {resp.text!r}")
+        payload = resp.json()
+        session = payload.get("session")
+        stored = {}
+        if payload["files"] and session:
+            try:
+                prefix = f"/sandboxes/{session}"
+
+                async def download(path: str):
+                    async with storage.writer() as writer:
+                        async with self._client.stream(
+                            "GET", f"{prefix}/workspace/{_rel(path)}"
+                        ) as file_resp:
+                            file_resp.raise_for_status()
+                            async for chunk in file_resp.aiter_bytes():
+                                await writer.write(chunk)
+                        return path, writer.hash
+
+                stored = dict(
+                    await asyncio.gather(*(download(p) for p in payload["files"]))
+                )
+            finally:
+                asyncio.ensure_future(self.delete_sandbox(session))
+        timings = dict(payload.get("timings") or {})
+        timings["exec_api_ms"] = round(t_exec, 2)
+        return Result(
+            stdout=payload["stdout"],
+            stderr=payload["stderr"],
+            exit_code=payload["exit_code"],
+            files=stored,
+            timings=timings,
+        )
 
     async def _run(self, storage, source_code, files, env, timeout, prefix) -> Result:
         async def upload(path: str, object_hash: str):

@@ -93,7 +93,7 @@ def create_http_server(
 
     @app.post("/v1/execute", response_model=ExecuteResponse)
     async def execute(request: ExecuteRequest, request_id: str = Depends(set_request_id)):
-        logger.info(
+        logger.debug(
             "Executing code with files %s: %s", request.files, request.source_code
         )
         try:
@@ -105,7 +105,7 @@ def create_http_server(
         except Exception as e:
             logger.exception("Error executing code")
             raise HTTPException(status_code=500, detail=str(e))
-        logger.info("Code execution completed with result %s", result)
+        logger.debug("Code execution completed with result %s", result)
         return ExecuteResponse(
             stdout=result.stdout,
             stderr=result.stderr,
