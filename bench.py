@@ -71,9 +71,9 @@ def main() -> None:
     parser.add_argument(
         "--http-workers",
         type=int,
-        default=1,
-        help="control-plane event loops per rank (uvicorn-workers style; "
-        "measured: >1 loses to the shared client loop + CPU quota, keep 1)",
+        default=2,
+        help="service processes per rank (each is a full `python -m "
+        "code_interpreter_amd` instance -- the real deployment unit)",
     )
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
@@ -165,57 +165,59 @@ def main() -> None:
 
 
 async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
+    import signal
+    import subprocess
+
     import httpx
     import torch
-    import uvicorn
 
-    from code_interpreter_amd.application_context import ApplicationContext
-    from code_interpreter_amd.config import Config
-
-    # W control-plane workers (uvicorn-workers style): each worker is a
-    # full service instance (own event loop + engines) on its own port;
-    # requests round-robin across them. One asyncio loop saturates around
-    # ~300 req/s of HTTP handling, well below the sandbox pipeline.
+    # The service runs as REAL deployment processes (`python -m
+    # code_interpreter_amd`, one or more per rank): serving must not share
+    # this process's GIL with the benchmark client.
     workers = max(1, args.http_workers)
     engines_per_worker = max(1, args.engines_per_gpu // workers)
-    if workers == 1:
-        engines_per_worker = args.engines_per_gpu
-    stacks = []
+    procs = []
+    ports = []
     for w in range(workers):
         tmp = tempfile.mkdtemp(prefix=f"bench-r{rank}w{w}-")
         port = _free_port()
-        config = Config(
-            http_listen_addr=f"127.0.0.1:{port}",
-            file_storage_path=os.path.join(tmp, "storage"),
-            executor_root=os.path.join(tmp, "executors"),
-            executor_backend="local",
-            executor_pool_target_length=max(
-                2, args.concurrency // workers
-            ),
-            engines_per_gpu=engines_per_worker,
-            gpu_count=1 if use_gpu else 0,
-            gpu_pinning=False,  # engines inherit the rank's visible device
-            hip_numpy="require" if use_gpu else "off",
-            dep_install=False,
+        env = dict(os.environ)
+        env.update(
+            {
+                "APP_HTTP_LISTEN_ADDR": f"127.0.0.1:{port}",
+                "APP_GRPC_LISTEN_ADDR": f"127.0.0.1:{_free_port()}",
+                "APP_FILE_STORAGE_PATH": os.path.join(tmp, "storage"),
+                "APP_EXECUTOR_ROOT": os.path.join(tmp, "executors"),
+                "APP_EXECUTOR_BACKEND": "local",
+                "APP_EXECUTOR_POOL_TARGET_LENGTH": str(
+                    max(2, args.concurrency // workers)
+                ),
+                "APP_ENGINES_PER_GPU": str(engines_per_worker),
+                "APP_GPU_COUNT": "1" if use_gpu else "0",
+                "APP_GPU_PINNING": "false",  # inherit the rank's device
+                "APP_HIP_NUMPY": "require" if use_gpu else "off",
+                "APP_DEP_INSTALL": "false",
+                "PYTHONPATH": str(REPO_ROOT),
+            }
         )
-        ctx = ApplicationContext(config)
-        server = uvicorn.Server(
-            uvicorn.Config(
-                app=ctx.http_server, host="127.0.0.1", port=port,
-                log_config=None,
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, "-m", "code_interpreter_amd"],
+                env=env,
+                cwd=str(REPO_ROOT),
+                stdout=subprocess.DEVNULL,
+                stderr=subprocess.DEVNULL,
+                start_new_session=True,
             )
         )
-        thread = threading.Thread(target=server.run, daemon=True)
-        thread.start()
-        stacks.append({"ctx": ctx, "server": server, "thread": thread, "port": port})
+        ports.append(port)
 
     source = WORKLOAD.format(array_size=args.array_size)
     clients = [
-        httpx.AsyncClient(base_url=f"http://127.0.0.1:{s['port']}", timeout=300.0)
-        for s in stacks
+        httpx.AsyncClient(base_url=f"http://127.0.0.1:{p}", timeout=300.0)
+        for p in ports
     ]
     try:
-        # wait for every worker + at least one engine each
         deadline = time.time() + 180
         for client in clients:
             while time.time() < deadline:
@@ -282,11 +284,13 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
     finally:
         for client in clients:
             await client.aclose()
-        for s in stacks:
-            s["server"].should_exit = True
-        for s in stacks:
-            s["thread"].join(timeout=10)
-            await s["ctx"].code_executor.aclose()
+        for proc in procs:
+            try:
+                os.killpg(proc.pid, signal.SIGKILL)
+            except (ProcessLookupError, OSError):
+                proc.kill()
+        for proc in procs:
+            proc.wait()
 
     latencies.sort()
     return {
