@@ -64,14 +64,14 @@ def _percentile(sorted_vals, p):
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=256)
-    parser.add_argument("--warmup", type=int, default=32)
-    parser.add_argument("--concurrency", type=int, default=8)
-    parser.add_argument("--engines-per-gpu", type=int, default=2)
+    parser.add_argument("--steps", type=int, default=512)
+    parser.add_argument("--warmup", type=int, default=64)
+    parser.add_argument("--concurrency", type=int, default=12)
+    parser.add_argument("--engines-per-gpu", type=int, default=3)
     parser.add_argument(
         "--http-workers",
         type=int,
-        default=2,
+        default=3,
         help="service processes per rank (each is a full `python -m "
         "code_interpreter_amd` instance -- the real deployment unit)",
     )
@@ -213,57 +213,90 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
         ports.append(port)
 
     source = WORKLOAD.format(array_size=args.array_size)
-    clients = [
-        httpx.AsyncClient(base_url=f"http://127.0.0.1:{p}", timeout=300.0)
-        for p in ports
+
+    def drive(port: int, n_warm: int, n_timed: int, conc: int, out: dict,
+              start_evt: threading.Event, warm_done: threading.Barrier):
+        """One client thread per service process: its own event loop +
+        connection pool, so the measuring side never serializes the
+        services behind a single GIL/loop."""
+
+        async def main():
+            async with httpx.AsyncClient(
+                base_url=f"http://127.0.0.1:{port}", timeout=300.0
+            ) as client:
+                deadline = time.time() + 180
+                while time.time() < deadline:
+                    try:
+                        r = await client.post(
+                            "/v1/execute", json={"source_code": "print('ready')"}
+                        )
+                        if r.status_code == 200 and r.json()["exit_code"] == 0:
+                            break
+                    except httpx.HTTPError:
+                        pass
+                    await asyncio.sleep(0.25)
+                else:
+                    raise RuntimeError("service did not become ready")
+
+                async def one_request() -> float:
+                    t0 = time.perf_counter()
+                    resp = await client.post(
+                        "/v1/execute", json={"source_code": source}
+                    )
+                    dt = time.perf_counter() - t0
+                    body = resp.json()
+                    if resp.status_code != 200 or body["exit_code"] != 0:
+                        raise RuntimeError(
+                            f"execute failed: {resp.status_code} "
+                            f"{body.get('stderr', '')[:500]}"
+                        )
+                    if "Result:" not in body["stdout"]:
+                        raise RuntimeError(
+                            f"unexpected stdout: {body['stdout'][:200]}"
+                        )
+                    return dt
+
+                async def run_phase(n: int) -> list:
+                    sem = asyncio.Semaphore(conc)
+                    latencies = []
+
+                    async def guarded():
+                        async with sem:
+                            latencies.append(await one_request())
+
+                    await asyncio.gather(*(guarded() for _ in range(n)))
+                    return latencies
+
+                await run_phase(n_warm)
+                warm_done.wait()  # all threads warmed before the clock
+                start_evt.wait()
+                out[port] = await run_phase(n_timed)
+
+        asyncio.run(main())
+
+    # split work across one client thread per service process
+    workers_n = len(ports)
+    per = [args.steps // workers_n] * workers_n
+    for i in range(args.steps % workers_n):
+        per[i] += 1
+    warm_per = max(1, args.warmup // workers_n)
+    conc_per = max(1, args.concurrency // workers_n)
+    results: dict = {}
+    start_evt = threading.Event()
+    warm_done = threading.Barrier(workers_n + 1)
+    threads = [
+        threading.Thread(
+            target=drive,
+            args=(ports[i], warm_per, per[i], conc_per, results, start_evt,
+                  warm_done),
+            daemon=True,
+        )
+        for i in range(workers_n)
     ]
     try:
-        deadline = time.time() + 180
-        for client in clients:
-            while time.time() < deadline:
-                try:
-                    r = await client.post(
-                        "/v1/execute", json={"source_code": "print('ready')"}
-                    )
-                    if r.status_code == 200 and r.json()["exit_code"] == 0:
-                        break
-                except httpx.HTTPError:
-                    pass
-                await asyncio.sleep(0.25)
-            else:
-                raise RuntimeError("service did not become ready")
-
-        counter = {"n": 0}
-
-        async def one_request() -> float:
-            counter["n"] += 1
-            client = clients[counter["n"] % len(clients)]
-            t0 = time.perf_counter()
-            resp = await client.post("/v1/execute", json={"source_code": source})
-            dt = time.perf_counter() - t0
-            body = resp.json()
-            if resp.status_code != 200 or body["exit_code"] != 0:
-                raise RuntimeError(
-                    f"execute failed: {resp.status_code} "
-                    f"{body.get('stderr', '')[:500]}"
-                )
-            if "Result:" not in body["stdout"]:
-                raise RuntimeError(f"unexpected stdout: {body['stdout'][:200]}")
-            return dt
-
-        async def run_phase(n: int) -> list:
-            sem = asyncio.Semaphore(args.concurrency)
-            latencies = []
-
-            async def guarded():
-                async with sem:
-                    latencies.append(await one_request())
-
-            await asyncio.gather(*(guarded() for _ in range(n)))
-            return latencies
-
-        # warmup (untimed)
-        await run_phase(args.warmup)
+        for t in threads:
+            t.start()
+        warm_done.wait()  # every client warmed its service
 
         # timed region, bracketed by barrier + device sync on both sides
         if world_size > 1:
@@ -273,7 +306,9 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
         if use_gpu:
             torch.cuda.synchronize()
         t_start = time.perf_counter()
-        latencies = await run_phase(args.steps)
+        start_evt.set()
+        for t in threads:
+            t.join()
         if use_gpu:
             torch.cuda.synchronize()
         if world_size > 1:
@@ -281,9 +316,12 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
 
             dist.barrier()
         elapsed = time.perf_counter() - t_start
+        latencies = [x for port in results for x in results[port]]
+        if len(latencies) != args.steps:
+            raise RuntimeError(
+                f"client threads returned {len(latencies)} != {args.steps}"
+            )
     finally:
-        for client in clients:
-            await client.aclose()
         for proc in procs:
             try:
                 os.killpg(proc.pid, signal.SIGKILL)
