@@ -61,7 +61,88 @@ def _percentile(sorted_vals, p):
     return sorted_vals[idx]
 
 
+def client_worker(argv) -> None:
+    """Internal: one measuring client process against one service port."""
+    import json as jsonlib
+
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--port", type=int, required=True)
+    parser.add_argument("--warm", type=int, required=True)
+    parser.add_argument("--steps", type=int, required=True)
+    parser.add_argument("--concurrency", type=int, required=True)
+    parser.add_argument("--ready-file", required=True)
+    parser.add_argument("--go-file", required=True)
+    parser.add_argument("--out-file", required=True)
+    parser.add_argument("--source-file", required=True)
+    args = parser.parse_args(argv)
+
+    import httpx
+
+    source = Path(args.source_file).read_text()
+
+    async def main_async():
+        async with httpx.AsyncClient(
+            base_url=f"http://127.0.0.1:{args.port}", timeout=300.0
+        ) as client:
+            deadline = time.time() + 180
+            while time.time() < deadline:
+                try:
+                    r = await client.post(
+                        "/v1/execute", json={"source_code": "print('ready')"}
+                    )
+                    if r.status_code == 200 and r.json()["exit_code"] == 0:
+                        break
+                except httpx.HTTPError:
+                    pass
+                await asyncio.sleep(0.25)
+            else:
+                raise RuntimeError("service did not become ready")
+
+            async def one_request() -> float:
+                t0 = time.perf_counter()
+                resp = await client.post(
+                    "/v1/execute", json={"source_code": source}
+                )
+                dt = time.perf_counter() - t0
+                body = resp.json()
+                if resp.status_code != 200 or body["exit_code"] != 0:
+                    raise RuntimeError(
+                        f"execute failed: {resp.status_code} "
+                        f"{body.get('stderr', '')[:500]}"
+                    )
+                if "Result:" not in body["stdout"]:
+                    raise RuntimeError(f"unexpected stdout: {body['stdout'][:200]}")
+                return dt
+
+            async def run_phase(n: int) -> list:
+                sem = asyncio.Semaphore(args.concurrency)
+                latencies = []
+
+                async def guarded():
+                    async with sem:
+                        latencies.append(await one_request())
+
+                await asyncio.gather(*(guarded() for _ in range(n)))
+                return latencies
+
+            await run_phase(args.warm)
+            Path(args.ready_file).touch()
+            while not os.path.exists(args.go_file):
+                await asyncio.sleep(0.001)
+            t0 = time.monotonic()  # CLOCK_MONOTONIC: comparable across procs
+            latencies = await run_phase(args.steps)
+            t1 = time.monotonic()
+            Path(args.out_file).write_text(
+                jsonlib.dumps({"t0": t0, "t1": t1, "latencies": latencies})
+            )
+
+    asyncio.run(main_async())
+
+
 def main() -> None:
+    if len(sys.argv) > 1 and sys.argv[1] == "--_client":
+        client_worker(sys.argv[2:])
+        return
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=512)
@@ -214,89 +295,51 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
 
     source = WORKLOAD.format(array_size=args.array_size)
 
-    def drive(port: int, n_warm: int, n_timed: int, conc: int, out: dict,
-              start_evt: threading.Event, warm_done: threading.Barrier):
-        """One client thread per service process: its own event loop +
-        connection pool, so the measuring side never serializes the
-        services behind a single GIL/loop."""
-
-        async def main():
-            async with httpx.AsyncClient(
-                base_url=f"http://127.0.0.1:{port}", timeout=300.0
-            ) as client:
-                deadline = time.time() + 180
-                while time.time() < deadline:
-                    try:
-                        r = await client.post(
-                            "/v1/execute", json={"source_code": "print('ready')"}
-                        )
-                        if r.status_code == 200 and r.json()["exit_code"] == 0:
-                            break
-                    except httpx.HTTPError:
-                        pass
-                    await asyncio.sleep(0.25)
-                else:
-                    raise RuntimeError("service did not become ready")
-
-                async def one_request() -> float:
-                    t0 = time.perf_counter()
-                    resp = await client.post(
-                        "/v1/execute", json={"source_code": source}
-                    )
-                    dt = time.perf_counter() - t0
-                    body = resp.json()
-                    if resp.status_code != 200 or body["exit_code"] != 0:
-                        raise RuntimeError(
-                            f"execute failed: {resp.status_code} "
-                            f"{body.get('stderr', '')[:500]}"
-                        )
-                    if "Result:" not in body["stdout"]:
-                        raise RuntimeError(
-                            f"unexpected stdout: {body['stdout'][:200]}"
-                        )
-                    return dt
-
-                async def run_phase(n: int) -> list:
-                    sem = asyncio.Semaphore(conc)
-                    latencies = []
-
-                    async def guarded():
-                        async with sem:
-                            latencies.append(await one_request())
-
-                    await asyncio.gather(*(guarded() for _ in range(n)))
-                    return latencies
-
-                await run_phase(n_warm)
-                warm_done.wait()  # all threads warmed before the clock
-                start_evt.wait()
-                out[port] = await run_phase(n_timed)
-
-        asyncio.run(main())
-
-    # split work across one client thread per service process
+    # one measuring client PROCESS per service process (a python thread
+    # cannot drive a second event loop in parallel -- the GIL): sync via
+    # ready/go files, timestamps on the shared CLOCK_MONOTONIC
     workers_n = len(ports)
     per = [args.steps // workers_n] * workers_n
     for i in range(args.steps % workers_n):
         per[i] += 1
     warm_per = max(1, args.warmup // workers_n)
     conc_per = max(1, args.concurrency // workers_n)
-    results: dict = {}
-    start_evt = threading.Event()
-    warm_done = threading.Barrier(workers_n + 1)
-    threads = [
-        threading.Thread(
-            target=drive,
-            args=(ports[i], warm_per, per[i], conc_per, results, start_evt,
-                  warm_done),
-            daemon=True,
+    ctmp = tempfile.mkdtemp(prefix=f"bench-cli-r{rank}-")
+    src_file = os.path.join(ctmp, "source.py")
+    Path(src_file).write_text(source)
+    go_file = os.path.join(ctmp, "go")
+    client_procs = []
+    out_files = []
+    ready_files = []
+    for i, port in enumerate(ports):
+        out_f = os.path.join(ctmp, f"out{i}.json")
+        ready_f = os.path.join(ctmp, f"ready{i}")
+        out_files.append(out_f)
+        ready_files.append(ready_f)
+        client_procs.append(
+            subprocess.Popen(
+                [
+                    sys.executable, str(REPO_ROOT / "bench.py"), "--_client",
+                    "--port", str(port), "--warm", str(warm_per),
+                    "--steps", str(per[i]), "--concurrency", str(conc_per),
+                    "--ready-file", ready_f, "--go-file", go_file,
+                    "--out-file", out_f, "--source-file", src_file,
+                ],
+                cwd=str(REPO_ROOT),
+                start_new_session=True,
+            )
         )
-        for i in range(workers_n)
-    ]
     try:
-        for t in threads:
-            t.start()
-        warm_done.wait()  # every client warmed its service
+        deadline = time.time() + 240
+        while time.time() < deadline:
+            if all(os.path.exists(f) for f in ready_files):
+                break
+            for proc in client_procs:
+                if proc.poll() not in (None, 0):
+                    raise RuntimeError("client worker died during warmup")
+            await asyncio.sleep(0.05)
+        else:
+            raise RuntimeError("client warmup did not finish")
 
         # timed region, bracketed by barrier + device sync on both sides
         if world_size > 1:
@@ -305,23 +348,33 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
             dist.barrier()
         if use_gpu:
             torch.cuda.synchronize()
-        t_start = time.perf_counter()
-        start_evt.set()
-        for t in threads:
-            t.join()
+        Path(go_file).touch()
+        for proc in client_procs:
+            rc = proc.wait()
+            if rc != 0:
+                raise RuntimeError(f"client worker exited {rc}")
         if use_gpu:
             torch.cuda.synchronize()
         if world_size > 1:
             import torch.distributed as dist
 
             dist.barrier()
-        elapsed = time.perf_counter() - t_start
-        latencies = [x for port in results for x in results[port]]
+
+        t0s, t1s, latencies = [], [], []
+        for out_f in out_files:
+            data = json.loads(Path(out_f).read_text())
+            t0s.append(data["t0"])
+            t1s.append(data["t1"])
+            latencies.extend(data["latencies"])
+        elapsed = max(t1s) - min(t0s)
         if len(latencies) != args.steps:
             raise RuntimeError(
-                f"client threads returned {len(latencies)} != {args.steps}"
+                f"client workers returned {len(latencies)} != {args.steps}"
             )
     finally:
+        for proc in client_procs:
+            if proc.poll() is None:
+                proc.kill()
         for proc in procs:
             try:
                 os.killpg(proc.pid, signal.SIGKILL)
