@@ -473,6 +473,16 @@ def rand(*shape, seed=None):
     return DeviceArray(h, shape if shape else (), _np.float64)
 
 
+def uniform_device(low: float, high: float, size) -> "DeviceArray":
+    """np.random.uniform semantics on-device: rand * (high-low) + low."""
+    shape = (size,) if isinstance(size, int) else tuple(size)
+    x = rand(*shape)
+    if low == 0.0 and high == 1.0:
+        return x
+    scaled = x._binary("multiply", float(high - low))
+    return scaled._binary("add", float(low))
+
+
 def square(x):
     _ensure_ready()
     d = _to_device(x)
