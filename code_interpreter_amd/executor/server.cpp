@@ -321,15 +321,20 @@ class Zygote {
   bool start() {
     int fds[2];
     if (socketpair(AF_UNIX, SOCK_STREAM, 0, fds) != 0) return false;
+    // Everything the child needs is built BEFORE fork(): this may be
+    // called from a monitor thread, and a forked child of a multithreaded
+    // process deadlocks if it allocates while another thread held the
+    // heap lock at fork time -- the child below is exec/exit only.
+    std::string zygote_py = g_cfg.runtime_dir + "/zygote.py";
+    char fd_str[16];
+    snprintf(fd_str, sizeof fd_str, "%d", fds[1]);
+    const char* argv[] = {g_cfg.python.c_str(), "-u", zygote_py.c_str(),
+                          "--fd", fd_str, nullptr};
     pid_t pid = fork();
     if (pid < 0) return false;
     if (pid == 0) {
-      // child: exec python zygote.py --fd N
       close(fds[0]);
-      std::string fd_str = std::to_string(fds[1]);
-      std::string zygote_py = g_cfg.runtime_dir + "/zygote.py";
-      execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), "-u", zygote_py.c_str(),
-             "--fd", fd_str.c_str(), (char*)nullptr);
+      execvp(argv[0], const_cast<char* const*>(argv));
       _exit(127);
     }
     close(fds[1]);
@@ -342,6 +347,7 @@ class Zygote {
 
   bool alive() const { return alive_.load(); }
   bool warm() const { return warm_.load(); }
+  pid_t pid() const { return pid_; }
 
   // Submit a job; returns the JobState used to track it.
   std::shared_ptr<JobState> submit(uint64_t id, const std::string& request_json) {
@@ -656,14 +662,26 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
   req += "}}";
 
   double t_submit = now_ms();
-  auto st = g_zygote->submit(id, req);
+  Zygote* zygote = g_zygote;
+  auto st = zygote->submit(id, req);
 
   ExecOutcome out;
   std::unique_lock<std::mutex> lk(st->mu);
   auto deadline =
       std::chrono::steady_clock::now() +
       std::chrono::milliseconds((int64_t)(timeout_s * 1000));
-  bool finished = st->cv.wait_until(lk, deadline, [&] { return st->done; });
+  bool finished = false;
+  while (!finished && std::chrono::steady_clock::now() < deadline) {
+    finished = st->cv.wait_for(lk, std::chrono::milliseconds(100),
+                               [&] { return st->done; });
+    if (!finished && !st->started && !zygote->alive()) {
+      // the zygote died before dispatching this job: retry cold
+      lk.unlock();
+      zygote->drop(id);
+      return run_cold(script_path, stdout_path, stderr_path, extra_env,
+                      timeout_s, workspace);
+    }
+  }
   if (!finished) {
     // timeout: kill the child's process group (reference parity:
     // ("", "Execution timed out", -1), server.rs:169)
@@ -1142,15 +1160,50 @@ int main(int, char**) {
   // socket never appears, and children fall back (ops/hipnp.py).
   std::string ops_dir = env_or("APP_OPS_DIR", "");
   if (env_or("APP_HIP_DAEMON", "1") != "0" && !ops_dir.empty()) {
-    std::string gpu_sock = g_sessions_root + "/gpu.sock";
-    pid_t pid = fork();
-    if (pid == 0) {
-      std::string script = ops_dir + "/hipd.py";
-      execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), "-u", script.c_str(),
-             "--socket", gpu_sock.c_str(), (char*)nullptr);
-      _exit(127);
+    // spawn + monitor/respawn. The spawner may run from the monitor
+    // thread: argv is fully built pre-fork and the child only exec/exits
+    // (a forked child of a multithreaded process deadlocks if it
+    // allocates while another thread held the heap lock).
+    static std::string gpu_sock = g_sessions_root + "/gpu.sock";
+    static std::string hipd_script = ops_dir + "/hipd.py";
+    auto spawn_daemon = []() -> pid_t {
+      const char* argv[] = {g_cfg.python.c_str(), "-u", hipd_script.c_str(),
+                            "--socket", gpu_sock.c_str(), nullptr};
+      pid_t pid = fork();
+      if (pid == 0) {
+        execvp(argv[0], const_cast<char* const*>(argv));
+        _exit(127);
+      }
+      return pid;
+    };
+    pid_t daemon_pid = spawn_daemon();
+    if (daemon_pid > 0) {
+      setenv("APP_GPU_SERVICE", gpu_sock.c_str(), 1);
+      std::thread([daemon_pid, spawn_daemon]() {
+        int respawns = 0;
+        pid_t pid = daemon_pid;
+        while (true) {
+          int status = 0;
+          if (waitpid(pid, &status, 0) != pid) return;
+          // exit(3) = no GPU visible: do not respawn; remove the socket
+          // and the env hint so sandboxes use their fallback
+          if (WIFEXITED(status) && WEXITSTATUS(status) == 3) {
+            unlink(gpu_sock.c_str());
+            unsetenv("APP_GPU_SERVICE");
+            return;
+          }
+          if (++respawns > 5) {
+            fprintf(stderr, "executor-server: gpu daemon kept dying\n");
+            unlink(gpu_sock.c_str());
+            return;
+          }
+          fprintf(stderr, "executor-server: gpu daemon died, respawning\n");
+          sleep(1);
+          pid = spawn_daemon();
+          if (pid <= 0) return;
+        }
+      }).detach();
     }
-    if (pid > 0) setenv("APP_GPU_SERVICE", gpu_sock.c_str(), 1);
   }
 
   if (g_cfg.zygote) {
@@ -1158,6 +1211,28 @@ int main(int, char**) {
     if (!g_zygote->start()) {
       fprintf(stderr, "executor-server: zygote failed to start; cold mode\n");
       g_cfg.zygote = false;
+    } else {
+      // respawn a dead zygote (cold fork/exec covers the gap meanwhile)
+      std::thread([]() {
+        int respawns = 0;
+        while (true) {
+          while (g_zygote->alive()) sleep(1);
+          int status = 0;
+          waitpid(g_zygote->pid(), &status, 0);  // reap (no zombie)
+          if (++respawns > 5) {
+            fprintf(stderr, "executor-server: zygote kept dying; cold mode\n");
+            return;
+          }
+          fprintf(stderr, "executor-server: zygote died, respawning\n");
+          Zygote* fresh = new Zygote();
+          if (fresh->start()) {
+            g_zygote = fresh;  // old object intentionally leaked (threads
+                               // may still hold job states briefly)
+          } else {
+            return;
+          }
+        }
+      }).detach();
     }
   }
 

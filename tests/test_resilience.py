@@ -68,3 +68,41 @@ def test_crash_in_user_code_is_contained(executor):
         assert r.stdout == "still up\n"
 
     asyncio.run(run())
+
+
+def test_zygote_crash_recovers(executor):
+    """Killing the engine's zygote degrades to cold fork/exec and the
+    monitor respawns it; executions keep succeeding throughout."""
+    import time
+
+    import psutil
+
+    async def run():
+        r = await executor.execute("print('warm path')")
+        assert r.exit_code == 0
+        engine = executor._engines[0]
+        server = psutil.Process(engine.proc.pid)
+        def zygote_children():
+            out = []
+            for c in server.children():
+                try:
+                    if "zygote.py" in " ".join(c.cmdline()):
+                        out.append(c)
+                except (psutil.ZombieProcess, psutil.NoSuchProcess):
+                    continue
+            return out
+
+        zygotes = zygote_children()
+        assert zygotes, "zygote process not found"
+        for z in zygotes:
+            z.kill()
+        # immediately after the kill: cold path must serve
+        r = await executor.execute("print('cold path')")
+        assert r.exit_code == 0 and r.stdout == "cold path\n"
+        # give the monitor time to respawn, then execute again
+        time.sleep(2.5)
+        r = await executor.execute("print('respawned')")
+        assert r.exit_code == 0 and r.stdout == "respawned\n"
+        assert zygote_children(), "zygote was not respawned"
+
+    asyncio.run(run())
