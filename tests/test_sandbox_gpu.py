@@ -107,3 +107,51 @@ def test_fp32_matmul_via_service_is_fast(gpu_executor):
     kind, dt = r.stdout.split()
     assert kind == "DeviceArray"
     assert float(dt) < 0.5, f"matmul path too slow: {dt}s"
+
+
+def test_gpu_daemon_crash_recovers(gpu_executor):
+    """Killing an engine's GPU daemon must not break executions: sandboxes
+    fall back to an own-context backend while the engine's monitor
+    respawns the daemon, then the remote path returns."""
+    import os
+    import signal
+    import time
+
+    import psutil
+
+    code = (
+        "import numpy, hipnp\n"
+        "x = numpy.random.rand(3_000_000)\n"
+        "print(hipnp.backend().name, float(numpy.sum(numpy.square(x))))\n"
+    )
+    r = _run(gpu_executor, code)
+    assert r.exit_code == 0, r.stderr
+    assert r.stdout.split()[0] == "remote"
+
+    # kill every engine's daemon
+    killed = 0
+    for eng in gpu_executor._engines:
+        if eng is None:
+            continue
+        server = psutil.Process(eng.proc.pid)
+        for child in server.children():
+            try:
+                if "hipd.py" in " ".join(child.cmdline()):
+                    os.kill(child.pid, signal.SIGKILL)
+                    killed += 1
+            except (psutil.ZombieProcess, psutil.NoSuchProcess):
+                continue
+    assert killed >= 1, "no gpu daemon found"
+
+    # executions keep succeeding immediately (own-context fallback is
+    # allowed); after the monitor respawns, the remote backend is back
+    deadline = time.time() + 30
+    back = False
+    while time.time() < deadline:
+        r = _run(gpu_executor, code)
+        assert r.exit_code == 0, r.stderr
+        if r.stdout.split()[0] == "remote":
+            back = True
+            break
+        time.sleep(1.0)
+    assert back, "daemon was not respawned"
