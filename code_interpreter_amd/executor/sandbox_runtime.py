@@ -286,7 +286,12 @@ def run_user_script(script_path: str) -> int:
         if e.code is None:
             return 0
         return e.code if isinstance(e.code, int) else 1
-    except BaseException:
+    except BaseException as e:
+        if type(e).__name__ == "GpuBackendLost":
+            # infrastructure failure, not a user bug: exit 113 so the
+            # control plane retries the whole execution in a fresh sandbox
+            print("gpu backend lost; execution will be retried", file=sys.stderr)
+            return 113
         import traceback
 
         etype, exc, tb = sys.exc_info()

@@ -55,16 +55,32 @@ class LocalBackend:
         return getattr(_hipops, item)
 
 
+class GpuBackendLost(RuntimeError):
+    """The GPU daemon connection died while device state was live; the
+    sandbox cannot transparently recover (its handles are gone). The
+    runner maps this to exit code 113, which the control plane treats as
+    an infrastructure failure and retries the whole execution in a fresh
+    sandbox (the same semantics as the reference's pod-death retry)."""
+
+
 class RemoteBackend:
-    """RPC to the engine's GPU daemon (ops/hipd.py) over a unix socket."""
+    """RPC to the engine's GPU daemon (ops/hipd.py) over a unix socket.
+    On connection loss with no live device handles (e.g. a pre-warmed
+    child whose daemon restarted) it transparently reconnects and
+    retries; with live handles it raises GpuBackendLost."""
 
     name = "remote"
 
     def __init__(self, path: str):
-        self._sock = _socket.socket(_socket.AF_UNIX, _socket.SOCK_STREAM)
-        self._sock.connect(path)
+        self._path = path
         self._lock = _threading.Lock()
+        self._live_handles = 0
+        self._connect()
         self._call({"op": "ping"})
+
+    def _connect(self):
+        self._sock = _socket.socket(_socket.AF_UNIX, _socket.SOCK_STREAM)
+        self._sock.connect(self._path)
 
     def _send(self, header: dict, payload=None) -> None:
         if payload is not None:
@@ -102,13 +118,39 @@ class RemoteBackend:
             raise RuntimeError(header.get("error", "gpu daemon error"))
         return header
 
+    def _reconnect_and_retry(self, header, payload, out_buffer):
+        if self._live_handles > 0:
+            raise GpuBackendLost(
+                "gpu daemon connection lost with live device handles"
+            )
+        import time as _time
+
+        wait = float(os.environ.get("APP_GPU_SERVICE_WAIT", "10"))
+        deadline = _time.monotonic() + wait
+        while True:
+            try:
+                self._connect()
+                self._send(header, payload)
+                return self._recv(out_buffer)
+            except (ConnectionError, OSError):
+                if _time.monotonic() >= deadline:
+                    raise GpuBackendLost("gpu daemon unreachable")
+                _time.sleep(0.1)
+
     def _call(self, header: dict, payload=None, out_buffer=None) -> dict:
         import time as _time
 
         t0 = _time.perf_counter()
         with self._lock:
-            self._send(header, payload)
-            out = self._recv(out_buffer)
+            try:
+                self._send(header, payload)
+                out = self._recv(out_buffer)
+            except (ConnectionError, OSError):
+                out = self._reconnect_and_retry(header, payload, out_buffer)
+        if "h" in out:
+            self._live_handles += 1
+        elif header.get("op") == "free":
+            self._live_handles = max(0, self._live_handles - 1)
         dt = (_time.perf_counter() - t0) * 1000
         RPC_STATS["ms"] += dt
         RPC_STATS["n"] += 1

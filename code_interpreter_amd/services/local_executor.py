@@ -245,13 +245,18 @@ class LocalPoolExecutor:
             engine = await self._ensure_engine(idx)
             engine.inflight += 1
             try:
-                return await engine.client.run_single_use(
+                result = await engine.client.run_single_use(
                     self.file_storage,
                     source_code,
                     files=files,
                     env=env,
                     timeout=self.execute_timeout,
                 )
+                if result.exit_code == 113 and "gpu backend lost" in result.stderr:
+                    # sandbox infrastructure failure (GPU daemon restart
+                    # mid-execution): retry in a fresh sandbox
+                    raise ExecutorError("gpu backend lost mid-execution")
+                return result
             except ExecutorError:
                 if not engine.alive():
                     asyncio.ensure_future(self._ensure_engine(idx))
