@@ -106,3 +106,28 @@ def test_zygote_crash_recovers(executor):
         assert zygote_children(), "zygote was not respawned"
 
     asyncio.run(run())
+
+def test_infra_exit_code_triggers_whole_execution_retry(executor, tmp_path):
+    """Exit code 113 + the GpuBackendLost stderr marker (emitted by
+    sandbox_runtime when the GPU daemon dies mid-script with live device
+    handles) must be treated as an infrastructure failure: the executor
+    retries the whole execution in a fresh sandbox instead of returning
+    the failure to the user."""
+    marker = tmp_path / "first_attempt_done"
+    code = (
+        "import os, sys\n"
+        f"m = {str(marker)!r}\n"
+        "if not os.path.exists(m):\n"
+        "    open(m, 'w').close()\n"
+        "    print('gpu backend lost; execution will be retried',"
+        " file=sys.stderr)\n"
+        "    sys.exit(113)\n"
+        "print('second attempt ok')\n"
+    )
+
+    async def run():
+        r = await executor.execute(code)
+        assert r.exit_code == 0, (r.exit_code, r.stderr)
+        assert r.stdout == "second attempt ok\n"
+
+    asyncio.run(run())
