@@ -292,6 +292,11 @@ struct ServerConfig {
   bool zygote = true;
   bool scan_recursive = false;
   double default_timeout = 60.0;
+  // request limits: a sandbox host must stay alive under hostile input,
+  // so header and body sizes are hard-capped (actix enforces similar
+  // defaults in the reference; a hand-written parser has to do it itself)
+  size_t max_header_bytes = 64 * 1024;
+  size_t max_body_bytes = (size_t)1 << 30;  // 1 GiB, APP_MAX_BODY_BYTES
 };
 
 static ServerConfig g_cfg;
@@ -737,6 +742,10 @@ class Conn {
     auto cl = req.headers.find("content-length");
     if (cl != req.headers.end()) {
       size_t len = (size_t)strtoull(cl->second.c_str(), nullptr, 10);
+      if (len > g_cfg.max_body_bytes) {
+        respond(413, "Payload Too Large", "{\"error\":\"body too large\"}");
+        return false;  // cannot resync mid-body: close the connection
+      }
       return read_exact_body(len, req.body);
     }
     return true;  // no body
@@ -757,9 +766,19 @@ class Conn {
     while (true) {
       size_t pos = buf_.find("\r\n\r\n");
       if (pos != std::string::npos) {
+        if (pos + 4 > g_cfg.max_header_bytes) {
+          respond(431, "Request Header Fields Too Large",
+                  "{\"error\":\"headers too large\"}");
+          return false;
+        }
         out = buf_.substr(0, pos + 4);
         buf_.erase(0, pos + 4);
         return true;
+      }
+      if (buf_.size() > g_cfg.max_header_bytes) {
+        respond(431, "Request Header Fields Too Large",
+                "{\"error\":\"headers too large\"}");
+        return false;
       }
       if (!fill()) return false;
     }
@@ -776,9 +795,14 @@ class Conn {
     while (true) {
       size_t pos;
       while ((pos = buf_.find("\r\n")) == std::string::npos) {
+        if (buf_.size() > 4096) return false;  // chunk-size line is tiny
         if (!fill()) return false;
       }
       size_t chunk_len = strtoull(buf_.substr(0, pos).c_str(), nullptr, 16);
+      if (out.size() + chunk_len > g_cfg.max_body_bytes) {
+        respond(413, "Payload Too Large", "{\"error\":\"body too large\"}");
+        return false;
+      }
       buf_.erase(0, pos + 2);
       if (chunk_len == 0) {
         // trailing CRLF (possibly trailers; we accept bare CRLF)
@@ -1135,6 +1159,9 @@ int main(int, char**) {
   g_cfg.python = env_or("APP_PYTHON", "python3");
   g_cfg.zygote = env_or("APP_ZYGOTE", "1") != "0";
   g_cfg.scan_recursive = env_or("APP_SCAN_RECURSIVE", "0") == "1";
+  g_cfg.max_body_bytes =
+      (size_t)strtoull(env_or("APP_MAX_BODY_BYTES", "1073741824").c_str(),
+                       nullptr, 10);
 
   // default runtime dir: the directory containing this binary
   std::string self_dir;
