@@ -76,12 +76,71 @@ class ExecuteCustomToolErrorResponse(BaseModel):
     stderr: str
 
 
+def _make_metrics():
+    """Prometheus counters/histograms when prometheus_client is present
+    (it is in the service image); observability beyond the reference,
+    which has no metrics endpoint. Returns
+    (observe(route, status, seconds), response_factory|None)."""
+    try:
+        from prometheus_client import (
+            CollectorRegistry,
+            Counter,
+            Histogram,
+            generate_latest,
+            CONTENT_TYPE_LATEST,
+        )
+    except ImportError:
+        return (lambda route, status_code, seconds: None), None
+
+    registry = CollectorRegistry()
+    requests_total = Counter(
+        "code_interpreter_requests_total",
+        "API requests by route and status",
+        ["route", "status"],
+        registry=registry,
+    )
+    latency = Histogram(
+        "code_interpreter_request_seconds",
+        "End-to-end request latency by route",
+        ["route"],
+        buckets=(0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1, 2.5, 5, 10, 30, 60),
+        registry=registry,
+    )
+
+    def observe(route: str, status_code: int, seconds: float) -> None:
+        requests_total.labels(route=route, status=str(status_code)).inc()
+        latency.labels(route=route).observe(seconds)
+
+    def render():
+        from fastapi import Response
+
+        return Response(generate_latest(registry), media_type=CONTENT_TYPE_LATEST)
+
+    return observe, render
+
+
 def create_http_server(
     code_executor,
     custom_tool_executor: CustomToolExecutor,
     request_id_context_var: ContextVar,
 ) -> FastAPI:
     app = FastAPI(title="code-interpreter-amd")
+    observe, metrics_response = _make_metrics()
+
+    @app.middleware("http")
+    async def record_metrics(request, call_next):
+        import time
+
+        t0 = time.perf_counter()
+        response = await call_next(request)
+        observe(request.url.path, response.status_code, time.perf_counter() - t0)
+        return response
+
+    if metrics_response is not None:
+
+        @app.get("/metrics")
+        async def metrics():
+            return metrics_response()
 
     def set_request_id() -> str:
         request_id = str(uuid.uuid4())

@@ -311,3 +311,16 @@ def test_ad_hoc_install_from_wheelhouse(tmp_path_factory, executor_bin, wheelhou
             assert "moo says: Hello World" in body["stdout"]
     finally:
         svc.stop()
+
+
+def test_metrics_endpoint(http_client):
+    """/metrics (beyond-reference observability): request counters and
+    latency histograms in Prometheus text format."""
+    r = http_client.post("/v1/execute", json={"source_code": "print('m')"})
+    assert r.status_code == 200
+    m = http_client.get("/metrics")
+    assert m.status_code == 200
+    body = m.text
+    assert "code_interpreter_requests_total" in body
+    assert '/v1/execute",status="200"' in body.replace("route=", "")
+    assert "code_interpreter_request_seconds_bucket" in body
