@@ -398,13 +398,23 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
         for proc in client_procs:
             if proc.poll() is None:
                 proc.kill()
+        # graceful service shutdown (engines flush profiler output and
+        # wind down their process groups); force-kill as a fallback
         for proc in procs:
             try:
-                os.killpg(proc.pid, signal.SIGKILL)
+                os.killpg(proc.pid, signal.SIGTERM)
             except (ProcessLookupError, OSError):
-                proc.kill()
+                proc.terminate()
+        deadline = time.monotonic() + 15.0
         for proc in procs:
-            proc.wait()
+            try:
+                proc.wait(max(0.1, deadline - time.monotonic()))
+            except subprocess.TimeoutExpired:
+                try:
+                    os.killpg(proc.pid, signal.SIGKILL)
+                except (ProcessLookupError, OSError):
+                    proc.kill()
+                proc.wait()
 
     latencies.sort()
     return {

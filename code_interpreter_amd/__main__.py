@@ -3,6 +3,7 @@ asyncio loop (parity: reference __main__.py:22-36)."""
 
 import asyncio
 import logging
+import signal
 
 import uvicorn
 
@@ -24,6 +25,14 @@ async def main() -> None:
         )
     )
 
+    # graceful shutdown: SIGTERM/SIGINT stop the frontends and close the
+    # executor pool (engines get SIGTERM -> their process groups wind
+    # down cleanly); without this a killed service leaks engine trees
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    for sig in (signal.SIGTERM, signal.SIGINT):
+        loop.add_signal_handler(sig, stop.set)
+
     tasks = [asyncio.create_task(uvicorn_server.serve())]
     try:
         grpc_server = ctx.grpc_server
@@ -33,7 +42,16 @@ async def main() -> None:
     except Exception as e:  # gRPC layer is optional at runtime
         logger.warning("gRPC server disabled: %s", e)
 
-    await asyncio.gather(*tasks)
+    stop_task = asyncio.create_task(stop.wait())
+    try:
+        await asyncio.wait(
+            [*tasks, stop_task], return_when=asyncio.FIRST_COMPLETED
+        )
+    finally:
+        for t in (*tasks, stop_task):
+            t.cancel()
+        await asyncio.gather(*tasks, stop_task, return_exceptions=True)
+        await ctx.code_executor.aclose()
 
 
 if __name__ == "__main__":

@@ -36,6 +36,7 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <signal.h>
+#include <sys/prctl.h>
 #include <string.h>
 #include <sys/socket.h>
 #include <sys/stat.h>
@@ -1150,8 +1151,37 @@ static int make_listen_socket() {
   return fd;
 }
 
+// Graceful shutdown: broadcast SIGTERM to the engine's process group
+// (zygote, warm children, GPU daemon all share it -- the engine is
+// spawned with start_new_session), give the daemon a moment to flush
+// (it may be writing profiler output), then exit. Everything here is
+// async-signal-safe.
+static volatile sig_atomic_t g_shutting_down = 0;
+static volatile pid_t g_daemon_pid_for_shutdown = 0;
+
+static void on_terminate(int) {
+  g_shutting_down = 1;
+  signal(SIGTERM, SIG_IGN);  // the pg broadcast below includes ourselves
+  kill(0, SIGTERM);
+  pid_t dpid = g_daemon_pid_for_shutdown;
+  if (dpid > 0) {
+    for (int i = 0; i < 30; i++) {  // <= 3 s for a clean daemon exit
+      if (waitpid(dpid, nullptr, WNOHANG) != 0) break;
+      struct timespec ts = {0, 100 * 1000 * 1000};
+      nanosleep(&ts, nullptr);
+    }
+  }
+  _exit(0);
+}
+
 int main(int, char**) {
   signal(SIGPIPE, SIG_IGN);
+  signal(SIGTERM, on_terminate);
+  signal(SIGINT, on_terminate);
+  // safety net against engine leaks: if the control-plane process that
+  // spawned us dies without calling aclose (SIGKILL, crash), die too
+  prctl(PR_SET_PDEATHSIG, SIGTERM);
+  if (getppid() == 1) raise(SIGTERM);  // parent already gone
 
   g_cfg.listen_addr = env_or("APP_LISTEN_ADDR", "0.0.0.0:8000");
   g_cfg.listen_unix = env_or("APP_LISTEN_UNIX", "");
@@ -1212,6 +1242,7 @@ int main(int, char**) {
         while (true) {
           int status = 0;
           if (waitpid(pid, &status, 0) != pid) return;
+          if (g_shutting_down) return;
           // exit(3) = no GPU visible: do not respawn; remove the socket
           // and the env hint so sandboxes use their fallback
           if (WIFEXITED(status) && WEXITSTATUS(status) == 3) {

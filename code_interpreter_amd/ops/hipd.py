@@ -150,6 +150,13 @@ def main() -> None:
     parser.add_argument("--device", type=int, default=0)
     args = parser.parse_args()
 
+    # clean exit on SIGTERM (engine shutdown broadcast): run atexit and
+    # profiler finalizers -- a SIGKILLed daemon loses any rocprofv3
+    # output it was asked to collect
+    import signal as _signal
+
+    _signal.signal(_signal.SIGTERM, lambda *_: sys.exit(0))
+
     if not _hipops.is_available():
         sys.exit(3)
     _hipops.init(args.device)

@@ -65,17 +65,27 @@ class Engine:
 
     async def aclose(self) -> None:
         await self.client.aclose()
+        # graceful first: the server broadcasts SIGTERM to its process
+        # group (zygote, warm children, GPU daemon) and waits for the
+        # daemon to flush (e.g. rocprofv3 output) before exiting
         try:
-            os.killpg(self.proc.pid, signal.SIGKILL)
+            os.killpg(self.proc.pid, signal.SIGTERM)
         except (ProcessLookupError, PermissionError, OSError):
             try:
-                self.proc.kill()
+                self.proc.terminate()
             except ProcessLookupError:
                 pass
         try:
             await asyncio.to_thread(self.proc.wait, 5)
         except subprocess.TimeoutExpired:
-            pass
+            try:
+                os.killpg(self.proc.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError, OSError):
+                self.proc.kill()
+            try:
+                await asyncio.to_thread(self.proc.wait, 5)
+            except subprocess.TimeoutExpired:
+                pass
         shutil.rmtree(self.root, ignore_errors=True)
 
 
