@@ -22,8 +22,15 @@
 // element dtypes understood by the generic elementwise/reduce entry points
 enum class DType : int { F32 = 0, F64 = 1 };
 
-enum class UnaryOp : int { Square = 0, Neg = 1, Abs = 2, Sqrt = 3, Exp = 4 };
-enum class BinOp : int { Add = 0, Sub = 1, Mul = 2, Div = 3 };
+enum class UnaryOp : int {
+  Square = 0, Neg = 1, Abs = 2, Sqrt = 3, Exp = 4,
+  Log = 5, Sin = 6, Cos = 7, Tanh = 8,
+};
+enum class BinOp : int {
+  Add = 0, Sub = 1, Mul = 2, Div = 3, Max = 4, Min = 5, Pow = 6,
+};
+// full-array reduction flavors (the python-side `mode` int)
+enum class ReduceOp : int { Sum = 0, SumSquares = 1, Max = 2, Min = 3 };
 
 // launchers (defined in kernels_ew.hip / gemm_*.hip); all take raw device
 // pointers and run on `stream`
@@ -36,7 +43,7 @@ void launch_binary_scalar(DType dt, BinOp op, const void* a, double scalar,
 // reduction: out_partials must hold >= reduce_num_partials(n) elements of dt;
 // final scalar (in dt) is written to out_scalar (device ptr) by stage 2
 int reduce_num_partials(int64_t n);
-void launch_sum(DType dt, bool square_inputs, const void* in, void* partials,
+void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
                 void* out_scalar, int64_t n, hipStream_t stream);
 // philox4x32-10 uniform doubles/floats in [0, 1)
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,

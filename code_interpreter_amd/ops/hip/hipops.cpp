@@ -445,20 +445,22 @@ PyObject* py_binary_scalar(PyObject*, PyObject* args) {
   WRAP_END
 }
 
-// sum(handle, dtype, n, square:int) -> float
+// sum(handle, dtype, n, mode:int) -> float
+// mode: 0 = sum, 1 = sum of squares, 2 = max, 3 = min (ReduceOp)
 PyObject* py_sum(PyObject*, PyObject* args) {
   unsigned long long h;
-  int dt, square;
+  int dt, mode;
   long long n;
-  if (!PyArg_ParseTuple(args, "KiLi", &h, &dt, &n, &square)) return nullptr;
+  if (!PyArg_ParseTuple(args, "KiLi", &h, &dt, &n, &mode)) return nullptr;
   WRAP_BEGIN
   ensure_init();
+  if (mode < 0 || mode > 3) throw std::runtime_error("bad reduce mode");
   DType dtype = dtype_from_int(dt);
   DevBuf& in = get_buf(h);
   double result = 0;
   Py_BEGIN_ALLOW_THREADS;
   ensure_reduce_scratch();
-  launch_sum(dtype, square != 0, in.ptr, g.reduce_scratch, g.scalar_dev, n,
+  launch_sum(dtype, (ReduceOp)mode, in.ptr, g.reduce_scratch, g.scalar_dev, n,
              g.compute);
   HIP_CHECK(hipMemcpyAsync(g.scalar_pin, g.scalar_dev, 8,
                            hipMemcpyDeviceToHost, g.compute));
@@ -545,7 +547,7 @@ PyMethodDef methods[] = {
     {"binary", py_binary, METH_VARARGS, "binary(ha, hb, op, dtype, n) -> handle"},
     {"binary_scalar", py_binary_scalar, METH_VARARGS,
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
-    {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, square) -> float"},
+    {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"synchronize", py_synchronize, METH_NOARGS, "sync all streams"},
     {"mem_info", py_mem_info, METH_NOARGS, "(free, total) bytes"},

@@ -40,6 +40,10 @@ __device__ __forceinline__ T apply_unary(T v) {
   if constexpr (OP == (int)UnaryOp::Abs) return v < T(0) ? -v : v;
   if constexpr (OP == (int)UnaryOp::Sqrt) return (T)sqrt((double)v);
   if constexpr (OP == (int)UnaryOp::Exp) return (T)exp((double)v);
+  if constexpr (OP == (int)UnaryOp::Log) return (T)log((double)v);
+  if constexpr (OP == (int)UnaryOp::Sin) return (T)sin((double)v);
+  if constexpr (OP == (int)UnaryOp::Cos) return (T)cos((double)v);
+  if constexpr (OP == (int)UnaryOp::Tanh) return (T)tanh((double)v);
   return v;
 }
 
@@ -49,6 +53,12 @@ __device__ __forceinline__ T apply_bin(T a, T b) {
   if constexpr (OP == (int)BinOp::Sub) return a - b;
   if constexpr (OP == (int)BinOp::Mul) return a * b;
   if constexpr (OP == (int)BinOp::Div) return a / b;
+  // numpy NaN semantics: maximum/minimum propagate NaN from either side
+  if constexpr (OP == (int)BinOp::Max)
+    return (a != a) ? a : ((b != b) ? b : (a > b ? a : b));
+  if constexpr (OP == (int)BinOp::Min)
+    return (a != a) ? a : ((b != b) ? b : (a < b ? a : b));
+  if constexpr (OP == (int)BinOp::Pow) return (T)pow((double)a, (double)b);
   return a;
 }
 
@@ -169,6 +179,69 @@ __global__ void sum_stage2(const T* __restrict__ partials, T* __restrict__ out,
   }
 }
 
+template <bool MAXOP>
+__device__ __forceinline__ double minmax_combine(double acc, double v) {
+  if (v != v) return v;  // NaN propagates (numpy np.max/np.min)
+  if (acc != acc) return acc;
+  if constexpr (MAXOP) return v > acc ? v : acc;
+  else return v < acc ? v : acc;
+}
+
+template <bool MAXOP>
+__device__ __forceinline__ double wave_reduce_minmax(double v) {
+  for (int off = 32; off > 0; off >>= 1)
+    v = minmax_combine<MAXOP>(v, __shfl_down(v, off, 64));
+  return v;
+}
+
+template <typename T, bool MAXOP>
+__global__ void minmax_stage1(const T* __restrict__ in,
+                              T* __restrict__ partials, int64_t n) {
+  using V2 = Vec2<T>;
+  int64_t n2 = n / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  double acc = MAXOP ? -INFINITY : INFINITY;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n2;
+       i += stride) {
+    V2 v = reinterpret_cast<const V2*>(in)[i];
+    acc = minmax_combine<MAXOP>(acc, (double)v.x);
+    acc = minmax_combine<MAXOP>(acc, (double)v.y);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    acc = minmax_combine<MAXOP>(acc, (double)in[n - 1]);
+  }
+  acc = wave_reduce_minmax<MAXOP>(acc);
+  __shared__ double wave_vals[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wave_vals[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double total = MAXOP ? -INFINITY : INFINITY;
+    for (int w = 0; w < (int)blockDim.x / 64; w++)
+      total = minmax_combine<MAXOP>(total, wave_vals[w]);
+    partials[blockIdx.x] = (T)total;
+  }
+}
+
+template <typename T, bool MAXOP>
+__global__ void minmax_stage2(const T* __restrict__ partials,
+                              T* __restrict__ out, int n) {
+  double acc = MAXOP ? -INFINITY : INFINITY;
+  for (int i = threadIdx.x; i < n; i += blockDim.x)
+    acc = minmax_combine<MAXOP>(acc, (double)partials[i]);
+  acc = wave_reduce_minmax<MAXOP>(acc);
+  __shared__ double wave_vals[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wave_vals[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double total = MAXOP ? -INFINITY : INFINITY;
+    for (int w = 0; w < (int)blockDim.x / 64; w++)
+      total = minmax_combine<MAXOP>(total, wave_vals[w]);
+    out[0] = (T)total;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Philox4x32-10 uniform RNG
 // ---------------------------------------------------------------------------
@@ -260,6 +333,7 @@ static void launch_unary_t(UnaryOp op, const T* in, T* out, int64_t n,
                        dim3(kBlock), 0, s, in, out, n);                  \
     break;
     CASE(Square) CASE(Neg) CASE(Abs) CASE(Sqrt) CASE(Exp)
+    CASE(Log) CASE(Sin) CASE(Cos) CASE(Tanh)
 #undef CASE
   }
   HIP_CHECK(hipGetLastError());
@@ -283,7 +357,7 @@ static void launch_binary_t(BinOp op, const T* a, const T* b, T* out,
     hipLaunchKernelGGL((binary_kernel<T, (int)BinOp::OP>), dim3(grid),    \
                        dim3(kBlock), 0, s, a, b, out, n);                 \
     break;
-    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div)
+    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div) CASE(Max) CASE(Min) CASE(Pow)
 #undef CASE
   }
   HIP_CHECK(hipGetLastError());
@@ -309,7 +383,7 @@ static void launch_binary_scalar_t(BinOp op, const T* a, T scalar, T* out,
     hipLaunchKernelGGL((binary_scalar_kernel<T, (int)BinOp::OP>), dim3(grid), \
                        dim3(kBlock), 0, s, a, scalar, out, n);               \
     break;
-    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div)
+    CASE(Add) CASE(Sub) CASE(Mul) CASE(Div) CASE(Max) CASE(Min) CASE(Pow)
 #undef CASE
   }
   HIP_CHECK(hipGetLastError());
@@ -328,28 +402,47 @@ void launch_binary_scalar(DType dt, BinOp op, const void* a, double scalar,
 int reduce_num_partials(int64_t n) { return grid_for(n / 2 + 1); }
 
 template <typename T>
-static void launch_sum_t(bool square, const T* in, T* partials, T* out_scalar,
-                         int64_t n, hipStream_t s) {
+static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
+                         T* out_scalar, int64_t n, hipStream_t s) {
   int grid = grid_for(n / 2 + 1);
-  if (square)
-    hipLaunchKernelGGL((sum_stage1<T, true>), dim3(grid), dim3(kBlock), 0, s,
-                       in, partials, n);
-  else
-    hipLaunchKernelGGL((sum_stage1<T, false>), dim3(grid), dim3(kBlock), 0, s,
-                       in, partials, n);
+  switch (mode) {
+    case ReduceOp::Sum:
+      hipLaunchKernelGGL((sum_stage1<T, false>), dim3(grid), dim3(kBlock), 0,
+                         s, in, partials, n);
+      break;
+    case ReduceOp::SumSquares:
+      hipLaunchKernelGGL((sum_stage1<T, true>), dim3(grid), dim3(kBlock), 0,
+                         s, in, partials, n);
+      break;
+    case ReduceOp::Max:
+      hipLaunchKernelGGL((minmax_stage1<T, true>), dim3(grid), dim3(kBlock),
+                         0, s, in, partials, n);
+      break;
+    case ReduceOp::Min:
+      hipLaunchKernelGGL((minmax_stage1<T, false>), dim3(grid), dim3(kBlock),
+                         0, s, in, partials, n);
+      break;
+  }
   HIP_CHECK(hipGetLastError());
-  hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
-                     out_scalar, grid);
+  if (mode == ReduceOp::Max)
+    hipLaunchKernelGGL((minmax_stage2<T, true>), dim3(1), dim3(kBlock), 0, s,
+                       partials, out_scalar, grid);
+  else if (mode == ReduceOp::Min)
+    hipLaunchKernelGGL((minmax_stage2<T, false>), dim3(1), dim3(kBlock), 0, s,
+                       partials, out_scalar, grid);
+  else
+    hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
+                       out_scalar, grid);
   HIP_CHECK(hipGetLastError());
 }
 
-void launch_sum(DType dt, bool square_inputs, const void* in, void* partials,
+void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
                 void* out_scalar, int64_t n, hipStream_t stream) {
   if (dt == DType::F64)
-    launch_sum_t(square_inputs, (const double*)in, (double*)partials,
+    launch_sum_t(mode, (const double*)in, (double*)partials,
                  (double*)out_scalar, n, stream);
   else
-    launch_sum_t(square_inputs, (const float*)in, (float*)partials,
+    launch_sum_t(mode, (const float*)in, (float*)partials,
                  (float*)out_scalar, n, stream);
 }
 

@@ -222,8 +222,16 @@ class RemoteBackend:
 
 # dtype codes shared with _hipops
 _F32, _F64 = 0, 1
-_UNARY = {"square": 0, "negative": 1, "absolute": 2, "sqrt": 3, "exp": 4}
-_BINARY = {"add": 0, "subtract": 1, "multiply": 2, "divide": 3, "true_divide": 3}
+_UNARY = {
+    "square": 0, "negative": 1, "absolute": 2, "sqrt": 3, "exp": 4,
+    "log": 5, "sin": 6, "cos": 7, "tanh": 8,
+}
+_BINARY = {
+    "add": 0, "subtract": 1, "multiply": 2, "divide": 3, "true_divide": 3,
+    "maximum": 4, "minimum": 5, "power": 6,
+}
+# full-array reduction modes (the extension's ReduceOp)
+_REDUCE_SUM, _REDUCE_SUMSQ, _REDUCE_MAX, _REDUCE_MIN = 0, 1, 2, 3
 
 MIN_ELEMS = int(os.environ.get("APP_HIP_NUMPY_MIN_ELEMS", 2_000_000))
 MIN_MATMUL_FLOPS = float(os.environ.get("APP_HIP_NUMPY_MIN_MATMUL_FLOPS", 5e7))
@@ -380,8 +388,8 @@ class DeviceArray:
             out = backend().binary(a._handle, b._handle, _BINARY[opname], code, self.size)
             return DeviceArray(out, self.shape, self.dtype)
         if isinstance(other, (int, float)):
-            if reverse and opname in ("subtract", "divide", "true_divide"):
-                return NotImplemented  # scalar-first sub/div: fall back
+            if reverse and opname in ("subtract", "divide", "true_divide", "power"):
+                return NotImplemented  # scalar-first sub/div/pow: fall back
             out = backend().binary_scalar(
                 self._handle, float(other), _BINARY[opname], code, self.size
             )
@@ -406,6 +414,39 @@ class DeviceArray:
             backend().sum(self._handle, _dtype_code(self.dtype), self.size, 1)
         )
 
+    def max(self, axis=None, **kwargs):
+        if axis is None and not kwargs.get("keepdims"):
+            return self.dtype.type(
+                backend().sum(
+                    self._handle, _dtype_code(self.dtype), self.size, _REDUCE_MAX
+                )
+            )
+        return self.materialize().max(axis=axis, **kwargs)
+
+    def min(self, axis=None, **kwargs):
+        if axis is None and not kwargs.get("keepdims"):
+            return self.dtype.type(
+                backend().sum(
+                    self._handle, _dtype_code(self.dtype), self.size, _REDUCE_MIN
+                )
+            )
+        return self.materialize().min(axis=axis, **kwargs)
+
+    def var(self, axis=None, ddof=0, **kwargs):
+        """Variance from two device reductions (sum and fused sum(x*x));
+        nothing round-trips through the host."""
+        if axis is None and not kwargs.get("keepdims"):
+            n = self.size
+            s1 = float(self.sum())
+            s2 = float(self.square_sum())
+            return self.dtype.type((s2 - s1 * s1 / n) / (n - ddof))
+        return self.materialize().var(axis=axis, ddof=ddof, **kwargs)
+
+    def std(self, axis=None, ddof=0, **kwargs):
+        if axis is None and not kwargs.get("keepdims"):
+            return self.dtype.type(float(self.var(ddof=ddof)) ** 0.5)
+        return self.materialize().std(axis=axis, ddof=ddof, **kwargs)
+
     # -- NEP 13: ufuncs --------------------------------------------------
     def __array_ufunc__(self, ufunc, method, *inputs, **kwargs):
         if kwargs.get("out") is not None:
@@ -424,9 +465,14 @@ class DeviceArray:
                     r = NotImplemented
                 if r is not NotImplemented:
                     return r
-        elif method == "reduce" and name == "add" and len(inputs) == 1:
+        elif method == "reduce" and len(inputs) == 1:
             if kwargs.get("axis") is None and not kwargs.get("keepdims"):
-                return inputs[0].sum()
+                if name == "add":
+                    return inputs[0].sum()
+                if name == "maximum":
+                    return inputs[0].max()
+                if name == "minimum":
+                    return inputs[0].min()
         return self._fallback_ufunc(ufunc, method, inputs, kwargs)
 
     def _fallback_ufunc(self, ufunc, method, inputs, kwargs):
@@ -449,6 +495,20 @@ class DeviceArray:
                 return r
         if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
             return args[0]._unary("square")
+        _reductions = {
+            _np.max: "max", _np.amax: "max",
+            _np.min: "min", _np.amin: "min",
+            _np.std: "std", _np.var: "var",
+        }
+        meth = _reductions.get(func)
+        if meth and len(args) == 1 and isinstance(args[0], DeviceArray):
+            allowed = {"axis", "ddof", "keepdims"}
+            if set(kwargs) <= allowed:
+                if kwargs.get("axis") is None and not kwargs.get("keepdims"):
+                    call_kwargs = {}
+                    if meth in ("std", "var") and "ddof" in kwargs:
+                        call_kwargs["ddof"] = kwargs["ddof"]
+                    return getattr(args[0], meth)(**call_kwargs)
         # generic fallback: materialize every DeviceArray
         host_args = [
             x.materialize() if isinstance(x, DeviceArray) else x for x in args
@@ -473,6 +533,9 @@ class DeviceArray:
 
     def __truediv__(self, o):
         return self._coerce(self._binary("divide", o))
+
+    def __pow__(self, o):
+        return self._coerce(self._binary("power", o))
 
     def __matmul__(self, o):
         r = matmul(self, o, _force=True)

@@ -258,3 +258,95 @@ def test_hipnp_random_variants(hnp):
     assert -2.0 <= host.min() < -1.9
     assert 1.9 < host.max() < 2.0
     np.testing.assert_allclose(host.mean(), 0.0, atol=5e-3)
+
+
+# ---------------------------------------------------------------------------
+# widened elementwise / reduction coverage
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize(
+    "uop,npop",
+    [
+        (1, np.negative),
+        (2, np.abs),
+        (3, np.sqrt),
+        (4, np.exp),
+        (5, np.log),
+        (6, np.sin),
+        (7, np.cos),
+        (8, np.tanh),
+    ],
+)
+def test_unary_ops_match_numpy(hip, uop, npop):
+    rng = np.random.default_rng(11)
+    a = rng.uniform(0.1, 4.0, 300_001)  # positive domain: valid for log/sqrt
+    h = hip.upload(a)
+    h2 = hip.unary(h, uop, 1, a.size)
+    out = np.empty_like(a)
+    hip.download(h2, out)
+    hip.free(h)
+    hip.free(h2)
+    np.testing.assert_allclose(out, npop(a), rtol=1e-14, atol=1e-15)
+
+
+@pytest.mark.parametrize(
+    "bop,npop", [(3, np.divide), (4, np.maximum), (5, np.minimum), (6, np.power)]
+)
+def test_binary_ops_match_numpy(hip, bop, npop):
+    rng = np.random.default_rng(12)
+    a = rng.uniform(0.5, 2.0, 100_001)
+    b = rng.uniform(0.5, 2.0, 100_001)
+    ha, hb = hip.upload(a), hip.upload(b)
+    hc = hip.binary(ha, hb, bop, 1, a.size)
+    out = np.empty_like(a)
+    hip.download(hc, out)
+    for h in (ha, hb, hc):
+        hip.free(h)
+    np.testing.assert_allclose(out, npop(a, b), rtol=1e-14)
+
+
+def test_reduce_max_min_match_numpy(hip):
+    rng = np.random.default_rng(13)
+    a = rng.standard_normal(5_000_003)
+    h = hip.upload(a)
+    mx = hip.sum(h, 1, a.size, 2)
+    mn = hip.sum(h, 1, a.size, 3)
+    hip.free(h)
+    assert mx == a.max()
+    assert mn == a.min()
+
+
+def test_reduce_max_propagates_nan(hip):
+    a = np.random.default_rng(14).standard_normal(1_000_000)
+    a[777_777] = np.nan
+    h = hip.upload(a)
+    mx = hip.sum(h, 1, a.size, 2)
+    mn = hip.sum(h, 1, a.size, 3)
+    hip.free(h)
+    assert np.isnan(mx) and np.isnan(mn)
+
+
+def test_device_array_reductions_and_ufuncs(hnp):
+    """np.log/np.sin/np.max/np.std on a device-resident array stay on
+    the GPU (no materialize) and match the numpy host reference."""
+    x = hnp.rand(3_000_000, seed=5)
+    host = x.materialize().copy()
+    y = np.log(x + 1.0)
+    assert type(y).__name__ == "DeviceArray"
+    np.testing.assert_allclose(
+        float(np.max(y)), np.log(host + 1.0).max(), rtol=1e-12
+    )
+    np.testing.assert_allclose(
+        float(np.min(y)), np.log(host + 1.0).min(), rtol=1e-12
+    )
+    np.testing.assert_allclose(float(np.std(x)), host.std(), rtol=1e-9)
+    np.testing.assert_allclose(
+        float(np.var(x, ddof=1)), host.var(ddof=1), rtol=1e-9
+    )
+    z = np.maximum(x, 0.5)
+    assert type(z).__name__ == "DeviceArray"
+    np.testing.assert_allclose(
+        float(np.sum(z)), np.maximum(host, 0.5).sum(), rtol=1e-12
+    )
+    w = np.sin(x) ** 2.0 + np.cos(x) ** 2.0
+    assert type(w).__name__ == "DeviceArray"
+    np.testing.assert_allclose(float(np.sum(w)), x.size, rtol=1e-12)

@@ -155,3 +155,33 @@ def test_gpu_daemon_crash_recovers(gpu_executor):
             break
         time.sleep(1.0)
     assert back, "daemon was not respawned"
+
+
+def test_widened_numpy_surface_stays_on_device(gpu_executor):
+    """A realistic analysis chain (rand -> log/sin/maximum -> std/max)
+    keeps every step on the GPU: the intermediate is a DeviceArray and
+    the reductions run as device kernels."""
+    code = (
+        "import numpy, hipnp\n"
+        "x = numpy.random.rand(4_000_000)\n"
+        "y = numpy.log(x + 1.0)\n"
+        "z = numpy.maximum(numpy.sin(y), 0.1)\n"
+        "assert type(z).__name__ == 'DeviceArray', type(z)\n"
+        "stats = (float(numpy.max(z)), float(numpy.min(z)),"
+        " float(numpy.std(z)), float(numpy.sum(z)))\n"
+        "import json; print(json.dumps(stats))\n"
+        "ops = hipnp.RPC_STATS.get('per_op', {})\n"
+        "assert 'unary' in ops and 'binary_scalar' in ops, ops\n"
+    )
+    r = _run(gpu_executor, code)
+    assert r.exit_code == 0, r.stderr
+    import json
+    import numpy as np
+
+    mx, mn, sd, total = json.loads(r.stdout.strip().splitlines()[-1])
+    # host reference of the same chain on the same Philox stream is not
+    # reproducible here (seeded per request); check invariants instead
+    assert 0.1 <= mn <= mx <= 1.0
+    assert 0 < sd < 1
+    assert total == pytest.approx(4_000_000 * (mn + mx) / 2, rel=0.5)
+    assert np is not None
