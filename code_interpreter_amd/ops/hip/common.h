@@ -48,6 +48,10 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 // philox4x32-10 uniform doubles/floats in [0, 1)
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
                          uint64_t offset, hipStream_t stream);
+// normal(mu, sigma) via Philox + Box-Muller (f64 only: numpy's normal
+// family returns float64)
+void launch_rand_normal(void* out, int64_t n, uint64_t seed, uint64_t offset,
+                        double mu, double sigma, hipStream_t stream);
 
 // row-major GEMM: C[M,N] = A[M,K] @ B[K,N]
 void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,

@@ -388,6 +388,24 @@ PyObject* py_rand(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// randn(n, seed, mu, sigma) -> handle (float64 N(mu, sigma^2))
+PyObject* py_randn(PyObject*, PyObject* args) {
+  long long n;
+  unsigned long long seed;
+  double mu, sigma;
+  if (!PyArg_ParseTuple(args, "LKdd", &n, &seed, &mu, &sigma)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  void* dev = nullptr;
+  Py_BEGIN_ALLOW_THREADS;
+  dev = pool_alloc(n * 8);
+  launch_rand_normal(dev, n, seed, g.rand_offset, mu, sigma, g.compute);
+  Py_END_ALLOW_THREADS;
+  g.rand_offset += (uint64_t)n;
+  return PyLong_FromUnsignedLongLong(register_buf(dev, n * 8));
+  WRAP_END
+}
+
 PyObject* py_unary(PyObject*, PyObject* args) {
   unsigned long long h;
   int op, dt;
@@ -543,6 +561,8 @@ PyMethodDef methods[] = {
     {"upload", py_upload, METH_VARARGS, "upload(buffer) -> handle"},
     {"download", py_download, METH_VARARGS, "download(handle, buffer)"},
     {"rand", py_rand, METH_VARARGS, "rand(n, dtype, seed) -> handle"},
+    {"randn", py_randn, METH_VARARGS,
+     "randn(n, seed, mu, sigma) -> handle (f64 normal)"},
     {"unary", py_unary, METH_VARARGS, "unary(h, op, dtype, n) -> handle"},
     {"binary", py_binary, METH_VARARGS, "binary(ha, hb, op, dtype, n) -> handle"},
     {"binary_scalar", py_binary_scalar, METH_VARARGS,

@@ -296,6 +296,37 @@ __global__ void rand_f64_kernel(double* __restrict__ out, int64_t n,
   }
 }
 
+// Box-Muller on Philox pairs: each counter yields two N(mu, sigma^2)
+// doubles. u1 is mapped to (0,1] so log(u1) is always finite.
+__global__ void randn_f64_kernel(double* __restrict__ out, int64_t n,
+                                 uint64_t seed, uint64_t offset, double mu,
+                                 double sigma) {
+  int64_t pairs = (n + 1) / 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const double scale = 1.0 / 9007199254740992.0;  // 2^-53
+  const double two_pi = 6.283185307179586476925286766559;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < pairs;
+       i += stride) {
+    uint32_t r[4];
+    philox10(offset + (uint64_t)i, seed, r);
+    uint64_t u0 = ((uint64_t)r[1] << 32) | r[0];
+    uint64_t u1 = ((uint64_t)r[3] << 32) | r[2];
+    double a = (double)((u0 >> 11) + 1) * scale;  // (0, 1]
+    double b = (double)(u1 >> 11) * scale;        // [0, 1)
+    double radius = sqrt(-2.0 * log(a));
+    double s_v, c_v;
+    sincos(two_pi * b, &s_v, &c_v);
+    double z0 = mu + sigma * (radius * c_v);
+    double z1 = mu + sigma * (radius * s_v);
+    int64_t j = i * 2;
+    if (j + 1 < n) {
+      reinterpret_cast<Vec2<double>*>(out)[i] = {z0, z1};
+    } else if (j < n) {
+      out[j] = z0;
+    }
+  }
+}
+
 __global__ void rand_f32_kernel(float* __restrict__ out, int64_t n,
                                 uint64_t seed, uint64_t offset) {
   int64_t quads = (n + 3) / 4;
@@ -444,6 +475,14 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
   else
     launch_sum_t(mode, (const float*)in, (float*)partials,
                  (float*)out_scalar, n, stream);
+}
+
+void launch_rand_normal(void* out, int64_t n, uint64_t seed, uint64_t offset,
+                        double mu, double sigma, hipStream_t stream) {
+  int grid = grid_for((n + 1) / 2);
+  hipLaunchKernelGGL(randn_f64_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     (double*)out, n, seed, offset, mu, sigma);
+  HIP_CHECK(hipGetLastError());
 }
 
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,

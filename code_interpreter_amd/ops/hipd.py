@@ -114,6 +114,10 @@ class Connection(threading.Thread):
             h = _hipops.rand(m["n"], m["dtype"], m["seed"])
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "randn":
+            h = _hipops.randn(m["n"], m["seed"], m["mu"], m["sigma"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "unary":
             h = _hipops.unary(m["h"], m["uop"], m["dtype"], m["n"])
             self.handles.add(h)

@@ -181,6 +181,11 @@ class RemoteBackend:
     def rand(self, n, dtype, seed):
         return self._call({"op": "rand", "n": n, "dtype": dtype, "seed": seed})["h"]
 
+    def randn(self, n, seed, mu, sigma):
+        return self._call(
+            {"op": "randn", "n": n, "seed": seed, "mu": mu, "sigma": sigma}
+        )["h"]
+
     def unary(self, h, uop, dtype, n):
         return self._call(
             {"op": "unary", "h": h, "uop": uop, "dtype": dtype, "n": n}
@@ -578,6 +583,19 @@ def rand(*shape, seed=None):
     return DeviceArray(h, shape if shape else (), _np.float64)
 
 
+def normal_device(loc: float, scale: float, *shape, seed=None) -> "DeviceArray":
+    """N(loc, scale^2) float64 on-device: Philox + Box-Muller with the
+    affine transform fused into the generating kernel."""
+    _ensure_ready()
+    n = 1
+    for sh in shape:
+        n *= int(sh)
+    if seed is None:
+        seed = int(_np.random.randint(0, 2**63 - 1, dtype=_np.int64))
+    h = backend().randn(n, int(seed), float(loc), float(scale))
+    return DeviceArray(h, shape if shape else (), _np.float64)
+
+
 def uniform_device(low: float, high: float, size) -> "DeviceArray":
     """np.random.uniform semantics on-device: rand * (high-low) + low."""
     shape = (size,) if isinstance(size, int) else tuple(size)
@@ -764,6 +782,50 @@ def install(numpy_module, mode: str = "auto") -> None:
                         raise
         return orig_sum(_asarray(x), *args, **kwargs)
 
+    orig_randn = np.random.randn
+    orig_standard_normal = np.random.standard_normal
+    orig_normal = np.random.normal
+
+    def patched_randn(*shape):
+        n = 1
+        for sh in shape:
+            n *= int(sh)
+        if n >= MIN_ELEMS:
+            try:
+                return normal_device(0.0, 1.0, *shape)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_randn(*shape)
+
+    def patched_standard_normal(size=None):
+        if size is not None and _size_elems(size) >= MIN_ELEMS:
+            try:
+                shape = (size,) if isinstance(size, int) else tuple(size)
+                return normal_device(0.0, 1.0, *shape)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_standard_normal(size)
+
+    def patched_normal(loc=0.0, scale=1.0, size=None):
+        if (
+            size is not None
+            and _size_elems(size) >= MIN_ELEMS
+            and isinstance(loc, (int, float))
+            and isinstance(scale, (int, float))
+        ):
+            try:
+                shape = (size,) if isinstance(size, int) else tuple(size)
+                return normal_device(float(loc), float(scale), *shape)
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_normal(loc, scale, size)
+
+    np.random.randn = patched_randn
+    np.random.standard_normal = patched_standard_normal
+    np.random.normal = patched_normal
     np.random.rand = patched_rand
     np.random.random = patched_random
     np.random.random_sample = patched_random

@@ -350,3 +350,32 @@ def test_device_array_reductions_and_ufuncs(hnp):
     w = np.sin(x) ** 2.0 + np.cos(x) ** 2.0
     assert type(w).__name__ == "DeviceArray"
     np.testing.assert_allclose(float(np.sum(w)), x.size, rtol=1e-12)
+
+
+def test_randn_statistics(hip):
+    import math
+
+    h = hip.randn(10_000_000, 42, 0.0, 1.0)
+    out = np.empty(10_000_000)
+    hip.download(h, out)
+    hip.free(h)
+    assert abs(out.mean()) < 2e-3
+    assert abs(out.std() - 1.0) < 2e-3
+    # standardized 3rd/4th moments: ~N(0,1) skew 0, kurtosis 3
+    assert abs(float((out**3).mean())) < 5e-3
+    assert abs(float((out**4).mean()) - 3.0) < 2e-2
+    assert math.isfinite(out.min()) and math.isfinite(out.max())
+    # affine fusion: N(5, 2^2)
+    h2 = hip.randn(4_000_000, 43, 5.0, 2.0)
+    out2 = np.empty(4_000_000)
+    hip.download(h2, out2)
+    hip.free(h2)
+    assert abs(out2.mean() - 5.0) < 5e-3
+    assert abs(out2.std() - 2.0) < 5e-3
+
+
+def test_hipnp_normal_device(hnp):
+    x = hnp.normal_device(-1.0, 0.5, 3_000_000, seed=9)
+    assert type(x).__name__ == "DeviceArray"
+    assert float(np.mean(x)) == pytest.approx(-1.0, abs=2e-3)
+    assert float(np.std(x)) == pytest.approx(0.5, abs=2e-3)

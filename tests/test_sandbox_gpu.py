@@ -185,3 +185,22 @@ def test_widened_numpy_surface_stays_on_device(gpu_executor):
     assert 0 < sd < 1
     assert total == pytest.approx(4_000_000 * (mn + mx) / 2, rel=0.5)
     assert np is not None
+
+
+def test_normal_family_routed_to_device(gpu_executor):
+    code = (
+        "import numpy, hipnp\n"
+        "a = numpy.random.randn(3_000_000)\n"
+        "b = numpy.random.standard_normal(3_000_000)\n"
+        "c = numpy.random.normal(10.0, 3.0, 3_000_000)\n"
+        "assert type(a).__name__ == 'DeviceArray', type(a)\n"
+        "assert type(b).__name__ == 'DeviceArray', type(b)\n"
+        "assert type(c).__name__ == 'DeviceArray', type(c)\n"
+        "print(float(numpy.mean(a)), float(numpy.std(b)),"
+        " float(numpy.mean(c)), float(numpy.std(c)))\n"
+    )
+    r = _run(gpu_executor, code)
+    assert r.exit_code == 0, r.stderr
+    ma, sb, mc, sc = map(float, r.stdout.split())
+    assert abs(ma) < 5e-3 and abs(sb - 1.0) < 5e-3
+    assert abs(mc - 10.0) < 2e-2 and abs(sc - 3.0) < 2e-2
