@@ -3,7 +3,8 @@
 Two mechanisms, installed by the sandbox runtime (executor/sandbox_runtime
 .py) when a GPU is visible:
 
-1. patched module-level entry points (numpy.random.rand, numpy.matmul,
+1. patched module-level entry points (numpy.random.{rand, random,
+   random_sample, uniform, randn, standard_normal, normal}, numpy.matmul,
    numpy.dot, numpy.square, numpy.sum): above a size threshold the work
    runs on the MI355X through _hipops and the result stays device-resident;
 2. DeviceArray: a duck array (NEP 13/18 __array_ufunc__ +
