@@ -110,7 +110,7 @@ def client_worker(argv) -> None:
                         f"execute failed: {resp.status_code} "
                         f"{body.get('stderr', '')[:500]}"
                     )
-                if "Result:" not in body["stdout"]:
+                if "Execution Time:" not in body["stdout"]:
                     raise RuntimeError(f"unexpected stdout: {body['stdout'][:200]}")
                 return dt
 
@@ -235,7 +235,7 @@ def main() -> None:
         print(
             json.dumps(
                 {
-                    "metric": "exec requests/sec (benchmark-numpy.py)",
+                    "metric": f"exec requests/sec ({args.workload})",
                     "value": round(value, 3),
                     "unit": "req/s",
                     "n_gpus": world_size if use_gpu else 0,
@@ -316,7 +316,12 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
         )
         ports.append(port)
 
-    source = WORKLOAD.format(array_size=args.array_size)
+    if args.workload == "benchmark-numpy.py":
+        source = WORKLOAD.format(array_size=args.array_size)
+    else:
+        # any other named workload is read from examples/ (e.g.
+        # benchmark-fib.py for the CPU-bound reference workload)
+        source = (REPO_ROOT / "examples" / args.workload).read_text()
 
     # one measuring client PROCESS per service process (a python thread
     # cannot drive a second event loop in parallel -- the GIL): sync via
