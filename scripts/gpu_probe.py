@@ -45,7 +45,19 @@ def main():
     # fused square+sum
     t = timed(lambda: _hipops.sum(h, 1, n, 1))
     print(f"sqsum_f64     n=1e8: {t*1e3:7.2f} ms  {8*n/t/1e12:6.2f} TB/s (read)")
+
+    # max reduction (8n read, NaN-propagating)
+    t = timed(lambda: _hipops.sum(h, 1, n, 2))
+    print(f"max_f64       n=1e8: {t*1e3:7.2f} ms  {8*n/t/1e12:6.2f} TB/s (read)")
+
+    # transcendental unary (log): 16n bytes traffic
+    t = timed(lambda: _hipops.free(_hipops.unary(h, 5, 1, n)))
+    print(f"log_f64       n=1e8: {t*1e3:7.2f} ms  {16*n/t/1e12:6.2f} TB/s (r+w)")
     _hipops.free(h)
+
+    # normal RNG (Box-Muller, fused affine): writes 8n bytes
+    t = timed(lambda: _hipops.free(_hipops.randn(n, 3, 0.0, 1.0)))
+    print(f"randn_f64     n=1e8: {t*1e3:7.2f} ms  {8*n/t/1e12:6.2f} TB/s (write)")
 
     # upload 800 MB (pinned staged)
     host = np.random.rand(n)
