@@ -23,7 +23,6 @@ import os
 import socket
 import sys
 import tempfile
-import threading
 import time
 from pathlib import Path
 

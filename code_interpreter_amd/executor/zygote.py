@@ -31,7 +31,6 @@ import select
 import signal
 import socket
 import sys
-import time
 
 RUNTIME_DIR = os.path.dirname(os.path.abspath(__file__))
 if RUNTIME_DIR not in sys.path:

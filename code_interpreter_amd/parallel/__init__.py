@@ -19,3 +19,11 @@ from mgpu import (  # noqa: E402,F401
     run_distributed,
     visible_gpu_count,
 )
+
+__all__ = [
+    "DEFAULT_BUCKET_BYTES",
+    "allreduce_bucketed",
+    "allreduce_matmul_bench",
+    "run_distributed",
+    "visible_gpu_count",
+]
