@@ -196,6 +196,12 @@ def main() -> None:
         dist.init_process_group(backend="nccl" if use_gpu else "gloo")
         if use_gpu:
             torch.cuda.set_device(local_rank)
+            # force CUDA/HIP context creation NOW, while every device is
+            # still visible: the per-rank HIP_VISIBLE_DEVICES restriction
+            # below is for child service processes, and must not change
+            # which device this rank's collectives run on
+            torch.cuda.init()
+            torch.zeros(1, device="cuda")
 
     # pin this rank's service (and its sandbox engines) to one GPU; if the
     # environment already restricts visibility to a list, take this rank's
