@@ -348,9 +348,13 @@ class DeviceArray:
         return str(self.materialize())
 
     def __del__(self):
+        # free on the EXISTING backend only: a destructor running during
+        # interpreter teardown (or after the backend is gone) must never
+        # construct one -- _ensure_ready from GC context can re-init HIP
         try:
-            if self._handle is not None:
-                backend().free(self._handle)
+            b = _state["backend"]
+            if b is not None and self._handle is not None:
+                b.free(self._handle)
         except Exception:
             pass
 
@@ -542,6 +546,17 @@ class DeviceArray:
 
     def __pow__(self, o):
         return self._coerce(self._binary("power", o))
+
+    # scalar-first sub/div/pow have no device kernel ordering: compute on
+    # the host instead of raising (user code does `1.0 / x` freely)
+    def __rsub__(self, o):
+        return o - self.materialize()
+
+    def __rtruediv__(self, o):
+        return o / self.materialize()
+
+    def __rpow__(self, o):
+        return o ** self.materialize()
 
     def __matmul__(self, o):
         r = matmul(self, o, _force=True)
