@@ -23,7 +23,7 @@ KNOWN_NON_CONFIG = {
     "APP_RUNTIME_DIR", "APP_OPS_DIR", "APP_ZYGOTE", "APP_WARM_CHILDREN",
     "APP_SESSIONS_DIR", "APP_HIP_DAEMON", "APP_GPU_SERVICE",
     "APP_GPU_SERVICE_WAIT", "APP_HIP_NUMPY_MIN_ELEMS",
-    "APP_HIP_NUMPY_MIN_MATMUL_FLOPS",
+    "APP_HIP_NUMPY_MIN_MATMUL_FLOPS", "APP_MAX_BODY_BYTES",
 }
 
 
