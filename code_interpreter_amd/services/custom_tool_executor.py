@@ -87,6 +87,9 @@ def _split_source(tool_source_code: str):
         body = ast.parse(clean_source).body
     except SyntaxError as e:
         raise CustomToolParseError([f"Syntax error: {e.msg} on line {e.lineno}"])
+    except ValueError as e:
+        # e.g. null bytes in the source: a client input error, not a 500
+        raise CustomToolParseError([f"Invalid source: {e}"])
     if not body:
         raise CustomToolParseError(
             [
