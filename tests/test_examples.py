@@ -5,6 +5,8 @@ kept executable by CI)."""
 
 from pathlib import Path
 
+import pytest
+
 EXAMPLES = Path(__file__).resolve().parent.parent / "examples"
 
 
@@ -75,3 +77,34 @@ def test_using_imports(http_client):
     out = _execute(http_client, "using_imports.py")
     assert out["exit_code"] == 0, out["stderr"]
     assert "p =" in out["stdout"]
+
+
+@pytest.mark.slow
+def test_mixed_workload_under_load(http_client):
+    """BASELINE config 5: using_imports-style + files-style payloads
+    concurrently -- the import scan, workspace staging and changed-file
+    scan hold up under load with zero errors."""
+    import concurrent.futures
+
+    files_payload = (EXAMPLES / "files.py").read_text()
+    imports_payload = (
+        "import numpy as np\n"
+        "import pandas as pd\n"
+        "from scipy import stats\n"
+        "x = np.arange(100, dtype=float)\n"
+        "print(float(pd.Series(x).mean()), float(stats.sem(x)))\n"
+    )
+
+    def one(i):
+        payload = files_payload if i % 2 == 0 else imports_payload
+        r = http_client.post("/v1/execute", json={"source_code": payload})
+        assert r.status_code == 200, r.text
+        out = r.json()
+        assert out["exit_code"] == 0, out["stderr"]
+        if i % 2 == 0:
+            assert any(p.endswith("note.txt") for p in out["files"])
+        return i
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=8) as pool:
+        done = list(pool.map(one, range(40)))
+    assert len(done) == 40
