@@ -23,6 +23,7 @@
 #include <string>
 #include <thread>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 #include "common.h"
@@ -53,6 +54,9 @@ struct State {
   std::unordered_map<uint64_t, DevBuf> bufs;
   // free-list: size -> free device pointers
   std::unordered_map<int64_t, std::vector<void*>> free_list;
+  // pointers that came from plain hipMalloc (mempool fallback): they must
+  // be released with hipFree, never hipFreeAsync
+  std::unordered_set<void*> plain_allocs;
   int64_t pool_bytes = 0;
   uint64_t rand_offset = 0;
   // allocator stats
@@ -65,6 +69,25 @@ struct State {
 State g;
 
 int64_t round_size(int64_t n) { return (n + 255) & ~255ll; }
+
+// release a cached pointer with the API that matches its provenance
+void release_ptr(void* p) {
+  if (g.plain_allocs.erase(p)) {
+    (void)hipFree(p);
+    return;
+  }
+  if (!g.use_mempool || hipFreeAsync(p, g.compute) != hipSuccess) (void)hipFree(p);
+}
+
+// under memory pressure our exact-size cache is the first thing to give
+// back: hipMemPoolTrimTo cannot reclaim blocks WE are holding
+void flush_free_list() {
+  for (auto& kv : g.free_list)
+    for (void* q : kv.second) release_ptr(q);
+  g.free_list.clear();
+  g.pool_bytes = 0;
+  (void)hipStreamSynchronize(g.compute);
+}
 
 void ensure_init() {
   if (!g.initialized) throw std::runtime_error("_hipops not initialized; call init()");
@@ -97,6 +120,7 @@ void* pool_alloc(int64_t size) {
     hipError_t e = hipMallocAsync(&p, size, g.compute);
     if (e == hipErrorOutOfMemory) {
       g.oom_trims++;
+      flush_free_list();
       hipMemPool_t pool = nullptr;
       if (hipDeviceGetDefaultMemPool(&pool, g.device) == hipSuccess)
         (void)hipMemPoolTrimTo(pool, 0);
@@ -119,15 +143,15 @@ void* pool_alloc(int64_t size) {
   void* p = nullptr;
   hipError_t e = hipMalloc(&p, size);
   if (e == hipErrorOutOfMemory) {
-    // drop the cache and retry once
-    for (auto& kv : g.free_list)
-      for (void* q : kv.second) (void)hipFree(q);
-    g.free_list.clear();
+    // drop the cache (provenance-aware) and retry once
+    flush_free_list();
     e = hipMalloc(&p, size);
   }
   if (e != hipSuccess)
     throw std::runtime_error(std::string("hipMalloc failed: ") +
                              hipGetErrorString(e));
+  g.plain_allocs.insert(p);
+  g.outstanding += size;
   return p;
 }
 
@@ -139,9 +163,7 @@ void pool_free(void* p, int64_t size) {
     g.pool_bytes += size;
     return;
   }
-  if (g.use_mempool && hipFreeAsync(p, g.compute) == hipSuccess) return;
-  g.free_list[size].push_back(p);
-  g.pool_bytes += size;
+  release_ptr(p);
 }
 
 // memcpy into pinned staging is single-thread-bound at ~15 GB/s; split
