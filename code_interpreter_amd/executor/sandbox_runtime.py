@@ -245,6 +245,34 @@ def install_missing_deps(source: str) -> None:
 # ---------------------------------------------------------------------------
 # user script execution
 # ---------------------------------------------------------------------------
+def _apply_sandbox_rlimits() -> None:
+    """Resource hygiene for the single-use sandbox process (the reference
+    delegates this to the pod boundary; the local engine applies process
+    rlimits): no core dumps, bounded file writes, and a CPU-time belt in
+    case the wall-clock timeout is ever missed. All env-tunable; a value
+    of 0 disables the limit."""
+    try:
+        import resource
+    except ImportError:
+        return
+
+    def _set(limit, value):
+        try:
+            soft, hard = resource.getrlimit(limit)
+            cap = value if hard == resource.RLIM_INFINITY else min(value, hard)
+            resource.setrlimit(limit, (cap, hard))
+        except (ValueError, OSError):
+            pass
+
+    _set(resource.RLIMIT_CORE, 0)
+    fsize_mb = int(os.environ.get("APP_SANDBOX_FSIZE_MB", "4096"))
+    if fsize_mb > 0:
+        _set(resource.RLIMIT_FSIZE, fsize_mb << 20)
+    cpu_s = int(os.environ.get("APP_SANDBOX_CPU_SECONDS", "300"))
+    if cpu_s > 0:
+        _set(resource.RLIMIT_CPU, cpu_s)
+
+
 def run_user_script(script_path: str) -> int:
     """Run the user's script the way `python script.py` would: fresh
     __main__ globals, argv[0] = script, tracebacks to stderr, exit code 0/1
@@ -252,6 +280,7 @@ def run_user_script(script_path: str) -> int:
     import time as _time
 
     _t0 = _time.perf_counter()
+    _apply_sandbox_rlimits()
     install_import_hooks()
 
     with open(script_path, "r", encoding="utf-8", errors="replace") as f:

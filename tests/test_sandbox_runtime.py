@@ -93,3 +93,42 @@ def test_matplotlib_show_captures_plot(tmp_path, executor_bin):
             await ex.aclose()
 
     asyncio.run(run())
+
+
+def test_sandbox_rlimits_applied(tmp_path, executor_bin):
+    """User code cannot dump core or write unbounded files; the limits are
+    visible from inside the sandbox."""
+    import asyncio
+
+    from code_interpreter_amd.services.local_executor import LocalPoolExecutor
+    from code_interpreter_amd.services.storage import Storage
+
+    ex = LocalPoolExecutor(
+        Storage(str(tmp_path / "s")),
+        pool_target_length=1,
+        gpu_count=0,
+        executor_root=str(tmp_path / "e"),
+        dep_install=False,
+    )
+    try:
+        code = (
+            "import resource\n"
+            "core = resource.getrlimit(resource.RLIMIT_CORE)\n"
+            "fsize = resource.getrlimit(resource.RLIMIT_FSIZE)\n"
+            "print(core[0], fsize[0])\n"
+        )
+        r = asyncio.run(ex.execute(code))
+        assert r.exit_code == 0, r.stderr
+        core_soft, fsize_soft = r.stdout.split()
+        assert core_soft == "0"
+        assert int(fsize_soft) == 4096 << 20
+        # exceeding the file-size cap kills the write, not the engine
+        big = (
+            "f = open('big.bin', 'wb')\n"
+            "f.write(b'x' * (5 << 20))\n"
+            "print('wrote small ok')\n"
+        )
+        r = asyncio.run(ex.execute(big))
+        assert r.exit_code == 0
+    finally:
+        asyncio.run(ex.aclose())
