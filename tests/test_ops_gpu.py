@@ -412,3 +412,42 @@ def test_allocator_pressure_and_oom_trim(hip):
     s = hip.sum(h, 1, 1_000_000, 0)
     hip.free(h)
     assert 0.45e6 < s < 0.55e6
+
+
+def test_gemm_multirun_race_screen(hip):
+    """3 repetitions x several sizes per dtype: single-run numerics tests
+    miss intermittent synchronization races (one was caught exactly this
+    way in the retired bf16 quadrant variant — profiles/NOTES.md)."""
+    for rep in range(3):
+        for dt, sizes, tol in ((0, (192, 512, 1024), 2e-2), (1, (192, 512, 1024), 1e-9)):
+            dtype = np.float32 if dt == 0 else np.float64
+            for size in sizes:
+                rng = np.random.default_rng(rep * 1000 + size + dt)
+                a = rng.uniform(-1, 1, (size, size)).astype(dtype)
+                b = rng.uniform(-1, 1, (size, size)).astype(dtype)
+                ha, hb = hip.upload(a), hip.upload(b)
+                hc = hip.gemm(ha, hb, size, size, size, dt)
+                out = np.empty_like(a)
+                hip.download(hc, out)
+                for h in (ha, hb, hc):
+                    hip.free(h)
+                ref = a.astype(np.float64) @ b.astype(np.float64)
+                err = np.max(np.abs(out - ref) / (np.abs(ref) + 1.0))
+                assert err < tol, f"rep{rep} dt{dt} {size}: relerr {err}"
+        # bf16: 512 (256-tile path) and 320 (128-tile fallback path)
+        for size in (512, 320):
+            rng = np.random.default_rng(rep * 77 + size)
+            a = f32_to_bf16_np(rng.uniform(-1, 1, (size, size)))
+            b = f32_to_bf16_np(rng.uniform(-1, 1, (size, size)))
+            ha, hb = hip.upload(a), hip.upload(b)
+            hc = hip.gemm(ha, hb, size, size, size, 2)
+            out = np.empty((size, size), dtype=np.uint16)
+            hip.download(hc, out)
+            for h in (ha, hb, hc):
+                hip.free(h)
+            ref = bf16_to_f32_np(a).astype(np.float64) @ bf16_to_f32_np(b).astype(
+                np.float64
+            )
+            got = bf16_to_f32_np(out).astype(np.float64)
+            err = np.max(np.abs(got - ref) / (np.abs(ref) + 1.0))
+            assert err < 2e-2, f"rep{rep} bf16 {size}: relerr {err}"
