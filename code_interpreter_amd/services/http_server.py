@@ -142,6 +142,12 @@ def create_http_server(
         async def metrics():
             return metrics_response()
 
+    @app.get("/healthz")
+    async def healthz():
+        """Cheap liveness for HTTP-only probes (the reference's liveness
+        is the gRPC health_check; both are wired in k8s/)."""
+        return {"status": "ok"}
+
     def set_request_id() -> str:
         request_id = str(uuid.uuid4())
         request_id_context_var.set(request_id)

@@ -324,3 +324,8 @@ def test_metrics_endpoint(http_client):
     assert "code_interpreter_requests_total" in body
     assert '/v1/execute",status="200"' in body.replace("route=", "")
     assert "code_interpreter_request_seconds_bucket" in body
+
+
+def test_healthz(http_client):
+    r = http_client.get("/healthz")
+    assert r.status_code == 200 and r.json() == {"status": "ok"}
