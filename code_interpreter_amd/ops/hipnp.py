@@ -381,8 +381,34 @@ class DeviceArray:
     def __float__(self):
         return float(self.materialize())
 
+    def __bool__(self):
+        # numpy semantics: truthiness only for size-1 arrays, else raise
+        return bool(self.materialize())
+
     def __iter__(self):
         return iter(self.materialize())
+
+    # comparisons materialize and delegate to numpy (the default object
+    # identity would silently return False for `x == 5`-style masks)
+    def __eq__(self, o):
+        return self.materialize() == _asarray_or_scalar(o)
+
+    def __ne__(self, o):
+        return self.materialize() != _asarray_or_scalar(o)
+
+    def __lt__(self, o):
+        return self.materialize() < _asarray_or_scalar(o)
+
+    def __le__(self, o):
+        return self.materialize() <= _asarray_or_scalar(o)
+
+    def __gt__(self, o):
+        return self.materialize() > _asarray_or_scalar(o)
+
+    def __ge__(self, o):
+        return self.materialize() >= _asarray_or_scalar(o)
+
+    __hash__ = None  # mutable-array semantics, same as numpy
 
     # -- device compute -------------------------------------------------
     def _unary(self, opname):
@@ -443,13 +469,16 @@ class DeviceArray:
         return self.materialize().min(axis=axis, **kwargs)
 
     def var(self, axis=None, ddof=0, **kwargs):
-        """Variance from two device reductions (sum and fused sum(x*x));
-        nothing round-trips through the host."""
+        """Two-pass variance entirely on-device: mean, then the fused
+        sum((x-mean)^2). (The one-pass sum-of-squares form cancels
+        catastrophically when |mean| >> std and can go negative.)"""
         if axis is None and not kwargs.get("keepdims"):
             n = self.size
-            s1 = float(self.sum())
-            s2 = float(self.square_sum())
-            return self.dtype.type((s2 - s1 * s1 / n) / (n - ddof))
+            mu = float(self.sum()) / n
+            shifted = self._binary("subtract", mu)
+            if shifted is NotImplemented:
+                return self.materialize().var(ddof=ddof)
+            return self.dtype.type(float(shifted.square_sum()) / (n - ddof))
         return self.materialize().var(axis=axis, ddof=ddof, **kwargs)
 
     def std(self, axis=None, ddof=0, **kwargs):
@@ -570,6 +599,10 @@ class DeviceArray:
 
 def _asarray(x):
     return x.materialize() if isinstance(x, DeviceArray) else _np.asarray(x)
+
+
+def _asarray_or_scalar(x):
+    return x.materialize() if isinstance(x, DeviceArray) else x
 
 
 def _to_device(x) -> "DeviceArray | None":

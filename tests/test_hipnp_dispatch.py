@@ -187,3 +187,24 @@ def test_matmul_chain(fake):
     y = x @ x
     assert isinstance(y, hipnp.DeviceArray)
     np.testing.assert_allclose(y.materialize(), a @ a, rtol=1e-12)
+
+
+def test_var_large_mean_is_stable(fake):
+    host = np.random.default_rng(6).normal(1e9, 1.0, 10_000)
+    x = _device(fake, host)
+    v = float(np.var(x))
+    assert v == pytest.approx(host.var(), rel=1e-6)
+    assert v >= 0
+    assert float(np.std(x)) == pytest.approx(host.std(), rel=1e-6)
+
+
+def test_comparisons_and_truthiness(fake):
+    host = np.array([0.2, 0.7, 0.5])
+    x = _device(fake, host)
+    np.testing.assert_array_equal(x > 0.5, host > 0.5)
+    np.testing.assert_array_equal(x == 0.7, host == 0.7)
+    np.testing.assert_array_equal(x <= 0.5, host <= 0.5)
+    with pytest.raises(ValueError):
+        bool(x)  # size > 1: ambiguous, numpy semantics
+    one = _device(fake, np.array([1.0]))
+    assert bool(one) is True
