@@ -160,7 +160,14 @@ def _namespace_from_imports(imports: list) -> dict:
             if node.module in SAFE_TYPE_MODULES:
                 mod = __import__(node.module, fromlist=[a.name for a in node.names])
                 for alias in node.names:
-                    ns[alias.asname or alias.name] = getattr(mod, alias.name)
+                    if alias.name == "*":  # star import of a safe module
+                        public = getattr(mod, "__all__", None) or [
+                            n for n in vars(mod) if not n.startswith("_")
+                        ]
+                        for n in public:
+                            ns[n] = getattr(mod, n)
+                    else:
+                        ns[alias.asname or alias.name] = getattr(mod, alias.name)
     return ns
 
 

@@ -110,3 +110,12 @@ def test_docstring_colon_in_text_not_directive():
 
 def test_docstring_empty():
     assert parse_docstring("") == ("", "", {})
+
+
+def test_star_import_of_safe_module(parser):
+    tool = parser.parse(
+        "from typing import *\n"
+        "def f(x: List[int]) -> int:\n"
+        "    return len(x)\n"
+    )
+    assert tool.input_schema["properties"]["x"]["type"] == "array"
