@@ -76,6 +76,7 @@ class ApplicationContext:
             dep_install=cfg.dep_install,
             pip_extra_args=cfg.pip_extra_args,
             hip_numpy=cfg.hip_numpy,
+            max_inflight_per_engine=cfg.max_inflight_per_engine,
         )
 
     def _kubernetes_executor(self):

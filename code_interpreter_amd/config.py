@@ -75,6 +75,9 @@ class Config(EnvSettings):
     # engines (executor-server + zygote + GPU daemon) per GPU: raises the
     # sandbox-management parallelism of one device
     engines_per_gpu: int = 1
+    # backpressure: queue (not reject) requests beyond this many in-flight
+    # sandboxes per engine; 0 disables
+    max_inflight_per_engine: int = 8
     # root dir for per-executor workspaces + unix sockets (tmpdir if empty)
     executor_root: str = ""
     # path to the executor-server binary ("" = bundled build)

@@ -108,6 +108,7 @@ class LocalPoolExecutor:
         pip_extra_args: str = "",
         hip_numpy: str = "auto",
         spawn_ready_timeout: float = 60.0,
+        max_inflight_per_engine: int = 8,
     ):
         self.file_storage = file_storage
         if gpu_count < 0:
@@ -131,6 +132,14 @@ class LocalPoolExecutor:
         self._spawn_locks = [asyncio.Lock() for _ in range(self.n_engines)]
         self._rr = 0
         self._closed = False
+        # backpressure: beyond ~8 in-flight sandboxes per engine the fork
+        # path thrashes the CPU quota (measured: throughput AND tail
+        # latency degrade); excess requests queue here instead
+        self._inflight_sem = (
+            asyncio.Semaphore(max_inflight_per_engine * self.n_engines)
+            if max_inflight_per_engine > 0
+            else None
+        )
 
     # -- lifecycle ---------------------------------------------------------
 
@@ -274,4 +283,7 @@ class LocalPoolExecutor:
             finally:
                 engine.inflight -= 1
 
-        return await async_retry(attempt, attempts=3, retry_on=(ExecutorError,))
+        if self._inflight_sem is None:
+            return await async_retry(attempt, attempts=3, retry_on=(ExecutorError,))
+        async with self._inflight_sem:
+            return await async_retry(attempt, attempts=3, retry_on=(ExecutorError,))

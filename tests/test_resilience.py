@@ -131,3 +131,30 @@ def test_infra_exit_code_triggers_whole_execution_retry(executor, tmp_path):
         assert r.stdout == "second attempt ok\n"
 
     asyncio.run(run())
+
+
+def test_backpressure_queues_overload(tmp_path, executor_bin):
+    """With a 1-deep in-flight cap, a burst of concurrent executions is
+    serialized through the queue -- all succeed, none rejected."""
+    ex = LocalPoolExecutor(
+        Storage(str(tmp_path / "s2")),
+        pool_target_length=1,
+        gpu_count=0,
+        executor_root=str(tmp_path / "e2"),
+        dep_install=False,
+        max_inflight_per_engine=1,
+    )
+
+    async def run():
+        results = await asyncio.gather(
+            *(ex.execute(f"print({i})") for i in range(12))
+        )
+        assert [r.exit_code for r in results] == [0] * 12
+        assert sorted(r.stdout for r in results) == sorted(
+            f"{i}\n" for i in range(12)
+        )
+
+    try:
+        asyncio.run(run())
+    finally:
+        asyncio.run(ex.aclose())
