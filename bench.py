@@ -175,6 +175,12 @@ def main() -> None:
         "code_interpreter_amd` instance -- the real deployment unit); "
         "0 = auto from the container CPU quota and world size",
     )
+    parser.add_argument(
+        "--pool-target",
+        type=int,
+        default=0,
+        help="warm sandbox children per engine (0 = concurrency/workers)",
+    )
     parser.add_argument("--array-size", type=int, default=10**8)
     parser.add_argument("--workload", default="benchmark-numpy.py")
     args = parser.parse_args()
@@ -307,7 +313,9 @@ async def run_rank(args, rank: int, world_size: int, use_gpu: bool) -> dict:
                 "APP_EXECUTOR_ROOT": os.path.join(tmp, "executors"),
                 "APP_EXECUTOR_BACKEND": "local",
                 "APP_EXECUTOR_POOL_TARGET_LENGTH": str(
-                    max(2, args.concurrency // workers)
+                    args.pool_target
+                    if args.pool_target > 0
+                    else max(2, args.concurrency // workers)
                 ),
                 "APP_ENGINES_PER_GPU": str(engines_per_worker),
                 "APP_GPU_COUNT": "1" if use_gpu else "0",
