@@ -102,6 +102,39 @@ class Connection(threading.Thread):
             out = bytearray(m["nbytes"])
             _hipops.download(m["h"], out)
             return {"ok": True}, bytes(out)
+        if op == "upload_shm":
+            # large-transfer path: payload arrives as a /dev/shm file the
+            # client wrote (one memcpy each side instead of a socket
+            # stream); path is restricted to /dev/shm
+            path = m["path"]
+            if not path.startswith("/dev/shm/"):
+                raise ValueError("upload_shm path must be under /dev/shm")
+            import mmap as _mmap
+
+            with open(path, "rb") as f:
+                mapped = _mmap.mmap(f.fileno(), m["nbytes"],
+                                    prot=_mmap.PROT_READ)
+                try:
+                    h = _hipops.upload(memoryview(mapped))
+                finally:
+                    mapped.close()
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "download_shm":
+            path = m["path"]
+            if not path.startswith("/dev/shm/"):
+                raise ValueError("download_shm path must be under /dev/shm")
+            import mmap as _mmap
+
+            with open(path, "r+b") as f:
+                f.truncate(m["nbytes"])
+                mapped = _mmap.mmap(f.fileno(), m["nbytes"])
+                try:
+                    _hipops.download(m["h"], memoryview(mapped))
+                    mapped.flush()
+                finally:
+                    mapped.close()
+            return {"ok": True}, b""
         if op == "alloc":
             h = _hipops.alloc(m["nbytes"])
             self.handles.add(h)

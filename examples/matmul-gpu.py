@@ -1,4 +1,7 @@
 # fp32 matmul routed to the MFMA matrix cores (v_mfma_f32_32x32x2_f32).
+# Host-born operands: the timing includes staging 128 MB into the GPU
+# daemon over the /dev/shm fast path (device-born chains, like
+# benchmark-numpy.py, skip staging entirely).
 import numpy as np
 import time
 
