@@ -48,6 +48,9 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 // philox4x32-10 uniform doubles/floats in [0, 1)
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
                          uint64_t offset, hipStream_t stream);
+// dtype conversion (f64 <-> f32), vectorized
+void launch_convert(DType src, DType dst, const void* in, void* out,
+                    int64_t n, hipStream_t stream);
 // normal(mu, sigma) via Philox + Box-Muller (f64 only: numpy's normal
 // family returns float64)
 void launch_rand_normal(void* out, int64_t n, uint64_t seed, uint64_t offset,

@@ -428,6 +428,26 @@ PyObject* py_randn(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// convert(h, src_dt, dst_dt, n) -> handle (f64 <-> f32 cast)
+PyObject* py_convert(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int src, dst;
+  long long n;
+  if (!PyArg_ParseTuple(args, "KiiL", &h, &src, &dst, &n)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DType sdt = dtype_from_int(src), ddt = dtype_from_int(dst);
+  DevBuf& in = get_buf(h);
+  int64_t out_size = n * (ddt == DType::F64 ? 8 : 4);
+  void* out = nullptr;
+  Py_BEGIN_ALLOW_THREADS;
+  out = pool_alloc(out_size);
+  launch_convert(sdt, ddt, in.ptr, out, n, g.compute);
+  Py_END_ALLOW_THREADS;
+  return PyLong_FromUnsignedLongLong(register_buf(out, out_size));
+  WRAP_END
+}
+
 PyObject* py_unary(PyObject*, PyObject* args) {
   unsigned long long h;
   int op, dt;
@@ -586,6 +606,8 @@ PyMethodDef methods[] = {
     {"randn", py_randn, METH_VARARGS,
      "randn(n, seed, mu, sigma) -> handle (f64 normal)"},
     {"unary", py_unary, METH_VARARGS, "unary(h, op, dtype, n) -> handle"},
+    {"convert", py_convert, METH_VARARGS,
+     "convert(h, src_dt, dst_dt, n) -> handle (f64<->f32)"},
     {"binary", py_binary, METH_VARARGS, "binary(ha, hb, op, dtype, n) -> handle"},
     {"binary_scalar", py_binary_scalar, METH_VARARGS,
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},

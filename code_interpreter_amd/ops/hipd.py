@@ -151,6 +151,10 @@ class Connection(threading.Thread):
             h = _hipops.randn(m["n"], m["seed"], m["mu"], m["sigma"])
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "convert":
+            h = _hipops.convert(m["h"], m["src"], m["dst"], m["n"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "unary":
             h = _hipops.unary(m["h"], m["uop"], m["dtype"], m["n"])
             self.handles.add(h)

@@ -247,6 +247,9 @@ class RemoteBackend:
             {"op": "randn", "n": n, "seed": seed, "mu": mu, "sigma": sigma}
         )["h"]
 
+    def convert(self, h, src, dst, n):
+        return self._call({"op": "convert", "h": h, "src": src, "dst": dst, "n": n})["h"]
+
     def unary(self, h, uop, dtype, n):
         return self._call(
             {"op": "unary", "h": h, "uop": uop, "dtype": dtype, "n": n}
@@ -430,6 +433,19 @@ class DeviceArray:
         if dtype is not None and dtype != host.dtype:
             return host.astype(dtype)
         return host
+
+    def astype(self, dtype, **kwargs):
+        """Device-side cast between f32/f64 (a common follow-on to the
+        routed RNG entry points); anything else materializes."""
+        target = _np.dtype(dtype)
+        src = _dtype_code(self.dtype)
+        dst = _dtype_code(target)
+        if src is not None and dst is not None and not kwargs:
+            if src == dst:
+                return self
+            h = backend().convert(self._handle, src, dst, self.size)
+            return DeviceArray(h, self.shape, target)
+        return self.materialize().astype(dtype, **kwargs)
 
     # anything we don't implement: materialize and delegate
     def __getattr__(self, name):

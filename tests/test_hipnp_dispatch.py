@@ -69,6 +69,11 @@ class FakeBackend:
             np.random.default_rng(seed).normal(mu, sigma, n)
         )
 
+    def convert(self, h, src, dst, n):
+        self.calls.append("convert")
+        out = self.bufs[h].view(self._dt(src))[:n].astype(self._dt(dst))
+        return self._new(out)
+
     def unary(self, h, uop, dtype, n):
         self.calls.append("unary")
         return self._new(self._UOPS[uop](self.bufs[h].view(self._dt(dtype))[:n]))
@@ -208,3 +213,16 @@ def test_comparisons_and_truthiness(fake):
         bool(x)  # size > 1: ambiguous, numpy semantics
     one = _device(fake, np.array([1.0]))
     assert bool(one) is True
+
+
+def test_astype_stays_on_device(fake):
+    host = np.random.default_rng(7).random(64)
+    x = _device(fake, host)
+    y = x.astype(np.float32)
+    assert isinstance(y, hipnp.DeviceArray) and y.dtype == np.float32
+    np.testing.assert_allclose(y.materialize(), host.astype(np.float32))
+    z = y.astype(np.float64)
+    assert isinstance(z, hipnp.DeviceArray) and z.dtype == np.float64
+    assert x.astype(np.float64) is x  # same dtype: no copy, no transfer
+    w = x.astype(np.int32)  # unsupported target: host fallback
+    assert isinstance(w, np.ndarray) and w.dtype == np.int32

@@ -451,3 +451,19 @@ def test_gemm_multirun_race_screen(hip):
             got = bf16_to_f32_np(out).astype(np.float64)
             err = np.max(np.abs(got - ref) / (np.abs(ref) + 1.0))
             assert err < 2e-2, f"rep{rep} bf16 {size}: relerr {err}"
+
+
+def test_convert_matches_numpy(hip):
+    rng = np.random.default_rng(21)
+    a = rng.standard_normal(1_000_003)
+    h = hip.upload(a)
+    h32 = hip.convert(h, 1, 0, a.size)
+    out32 = np.empty(a.size, dtype=np.float32)
+    hip.download(h32, out32)
+    np.testing.assert_array_equal(out32, a.astype(np.float32))
+    h64 = hip.convert(h32, 0, 1, a.size)
+    out64 = np.empty(a.size, dtype=np.float64)
+    hip.download(h64, out64)
+    for hh in (h, h32, h64):
+        hip.free(hh)
+    np.testing.assert_array_equal(out64, a.astype(np.float32).astype(np.float64))
