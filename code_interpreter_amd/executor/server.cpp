@@ -994,7 +994,9 @@ static void handle_workspace_io(Conn& conn, const HttpRequest& req,
                                 const std::string& workspace,
                                 const std::string& rel_encoded) {
   std::string rel = url_decode(rel_encoded);
-  if (!safe_rel_path(rel)) {
+  // a path naming a directory (trailing '/') or nothing is a client
+  // error, not a 500 from the failed open
+  if (!safe_rel_path(rel) || rel.empty() || rel.back() == '/') {
     conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
     return;
   }
@@ -1002,10 +1004,15 @@ static void handle_workspace_io(Conn& conn, const HttpRequest& req,
   if (req.method == "PUT") {
     size_t slash = full.rfind('/');
     if (slash != std::string::npos) mkdirs(full.substr(0, slash));
-    if (write_file(full, req.body))
+    if (write_file(full, req.body)) {
       conn.respond(204, "No Content", "");
-    else
+    } else if (errno == EISDIR || errno == ENOTDIR || errno == ENAMETOOLONG ||
+               errno == EINVAL) {
+      // the path itself is unusable: client error, not a server fault
+      conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
+    } else {
       conn.respond(500, "Internal Server Error", "{\"error\":\"write failed\"}");
+    }
   } else if (req.method == "GET") {
     struct stat st;
     if (stat(full.c_str(), &st) != 0 || !S_ISREG(st.st_mode)) {
