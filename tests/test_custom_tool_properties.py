@@ -104,3 +104,33 @@ def test_unsupported_signatures_rejected(parser, name, bad):
     src = f"def {name}({bad}) -> int:\n    return 1\n"
     with pytest.raises(CustomToolParseError):
         parser.parse(src)
+
+
+WORDS = st.from_regex(r"[a-z][a-z ]{0,24}[a-z]|[a-z]", fullmatch=True)
+
+
+@given(
+    st.one_of(st.just(""), WORDS),
+    st.lists(
+        st.tuples(st.from_regex(r"[a-z][a-z_]{0,8}", fullmatch=True), WORDS),
+        max_size=4,
+        unique_by=lambda t: t[0],
+    ),
+    st.one_of(st.none(), WORDS),
+)
+@settings(max_examples=60, deadline=None)
+def test_docstring_roundtrip(description, params, ret):
+    """Constructed ReST docstrings parse back into their parts."""
+    from code_interpreter_amd.services.custom_tool_executor import parse_docstring
+
+    lines = [description]
+    for name, desc in params:
+        lines.append(f":param {name}: {desc}")
+    if ret is not None:
+        lines.append(f":return: {ret}")
+    parsed_desc, parsed_ret, parsed_params = parse_docstring("\n".join(lines))
+    assert parsed_desc == description.strip()
+    for name, desc in params:
+        assert parsed_params.get(name) == desc.strip(), (name, parsed_params)
+    if ret is not None:
+        assert parsed_ret == ret.strip()
