@@ -6,6 +6,8 @@ import glob
 import os
 import subprocess
 
+KFD_NODES = "/sys/class/kfd/kfd/topology/nodes"
+
 
 @functools.lru_cache(maxsize=1)
 def detect_gpu_count() -> int:
@@ -24,7 +26,7 @@ def detect_gpu_count() -> int:
 def _physical_gpu_count() -> int:
     # KFD topology: GPU nodes have non-zero simd_count
     count = 0
-    for props in glob.glob("/sys/class/kfd/kfd/topology/nodes/*/properties"):
+    for props in glob.glob(KFD_NODES + "/*/properties"):
         try:
             with open(props) as f:
                 for line in f:
