@@ -77,3 +77,28 @@ def test_storage_gc(tmp_path):
     assert stats["removed"] == 1
     assert not (tmp_path / old).exists()
     assert (tmp_path / new).exists()
+
+
+def test_storage_gc_ttl(tmp_path):
+    import os
+    import time
+
+    from code_interpreter_amd.storage_gc import collect
+
+    root = tmp_path / "store"
+    root.mkdir()
+    old = root / ("a" * 64)
+    new = root / ("b" * 64)
+    junk = root / "not-an-object.txt"
+    for p in (old, new, junk):
+        p.write_bytes(b"x" * 100)
+    stale = time.time() - 48 * 3600
+    os.utime(old, (stale, stale))
+
+    dry = collect(str(root), ttl_hours=24, dry_run=True)
+    assert dry["removed"] == 1 and old.exists()
+
+    out = collect(str(root), ttl_hours=24)
+    assert out["removed"] == 1 and out["kept"] == 1
+    assert not old.exists()
+    assert new.exists() and junk.exists()  # non-object names untouched
