@@ -274,6 +274,52 @@ inline std::string quote(const std::string& s) {
   return out;
 }
 
+// Replace invalid UTF-8 sequences with U+FFFD so captured stdout/stderr
+// always embeds as valid-UTF-8 JSON (reference parity: server.rs uses
+// String::from_utf8_lossy; without this, user code printing raw bytes
+// turns a 200 into a client-side decode error).
+inline std::string to_valid_utf8(const std::string& s) {
+  std::string out;
+  out.reserve(s.size());
+  size_t i = 0, n = s.size();
+  auto cont = [&](size_t k) {
+    return i + k < n && (static_cast<unsigned char>(s[i + k]) & 0xC0) == 0x80;
+  };
+  while (i < n) {
+    unsigned char c = s[i];
+    if (c < 0x80) {
+      out += (char)c;
+      i += 1;
+    } else if ((c & 0xE0) == 0xC0 && c >= 0xC2 && cont(1)) {
+      out.append(s, i, 2);
+      i += 2;
+    } else if ((c & 0xF0) == 0xE0 && cont(1) && cont(2)) {
+      unsigned char c1 = s[i + 1];
+      // reject overlongs (E0 80-9F) and surrogates (ED A0-BF)
+      if ((c == 0xE0 && c1 < 0xA0) || (c == 0xED && c1 > 0x9F)) {
+        out += "\xEF\xBF\xBD";
+        i += 1;
+      } else {
+        out.append(s, i, 3);
+        i += 3;
+      }
+    } else if ((c & 0xF8) == 0xF0 && c <= 0xF4 && cont(1) && cont(2) && cont(3)) {
+      unsigned char c1 = s[i + 1];
+      if ((c == 0xF0 && c1 < 0x90) || (c == 0xF4 && c1 > 0x8F)) {
+        out += "\xEF\xBF\xBD";
+        i += 1;
+      } else {
+        out.append(s, i, 4);
+        i += 4;
+      }
+    } else {
+      out += "\xEF\xBF\xBD";  // U+FFFD replacement character
+      i += 1;
+    }
+  }
+  return out;
+}
+
 }  // namespace json
 
 // ---------------------------------------------------------------------------
@@ -287,6 +333,7 @@ static std::string env_or(const char* name, const std::string& dflt) {
 struct ServerConfig {
   std::string listen_addr;   // "host:port" or empty
   std::string listen_unix;   // unix socket path or empty
+  int max_connections = 512;  // concurrent handler threads (APP_MAX_CONNECTIONS)
   std::string workspace;
   std::string python;
   std::string runtime_dir;   // where zygote.py lives
@@ -437,7 +484,9 @@ class Zygote {
   std::map<uint64_t, std::shared_ptr<JobState>> jobs_;
 };
 
-static Zygote* g_zygote = nullptr;
+// Held atomically: the respawn monitor swaps it from its own thread while
+// request threads read it (a plain pointer here is a data race).
+static std::atomic<Zygote*> g_zygote{nullptr};
 static std::atomic<uint64_t> g_job_id{1};
 
 // ---------------------------------------------------------------------------
@@ -668,7 +717,7 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
   req += "}}";
 
   double t_submit = now_ms();
-  Zygote* zygote = g_zygote;
+  Zygote* zygote = g_zygote.load();
   auto st = zygote->submit(id, req);
 
   ExecOutcome out;
@@ -697,13 +746,13 @@ static ExecOutcome run_via_zygote(const std::string& script_path,
     }
     // give the zygote a moment to reap and report
     st->cv.wait_for(lk, std::chrono::seconds(5), [&] { return st->done; });
-    g_zygote->drop(id);
+    zygote->drop(id);
     out.stdout_text = "";
     out.stderr_text = "Execution timed out";
     out.exit_code = -1;
     return out;
   }
-  g_zygote->drop(id);
+  zygote->drop(id);
   out.exit_code = st->exit_code;
   double t_started = st->t_started_ms > 0 ? st->t_started_ms : t_submit;
   double t_done = st->t_done_ms > 0 ? st->t_done_ms : now_ms();
@@ -726,30 +775,101 @@ struct HttpRequest {
   bool keep_alive = true;
 };
 
+// How the body will arrive, decided from the headers; the route handler
+// then picks WHERE it goes (memory for JSON routes, straight to the
+// workspace file for PUT -- a 1 GiB upload must not cost 1 GiB of RSS).
+struct BodyPlan {
+  bool chunked = false;
+  bool has_length = false;
+  size_t length = 0;
+  bool has_body() const { return chunked || (has_length && length > 0); }
+};
+
 class Conn {
  public:
   explicit Conn(int fd) : fd_(fd) {}
   ~Conn() { close(fd_); }
 
-  bool read_request(HttpRequest& req) {
+  // Phase 1: headers only; the body stays on the socket until the route
+  // decides its sink. Content-Length above the cap is refused up front.
+  bool read_request_headers(HttpRequest& req, BodyPlan& plan) {
     std::string header_block;
     if (!read_until_headers(header_block)) return false;
     if (!parse_headers(header_block, req)) return false;
 
     auto te = req.headers.find("transfer-encoding");
     if (te != req.headers.end() && te->second.find("chunked") != std::string::npos) {
-      return read_chunked_body(req.body);
+      plan.chunked = true;
+      return true;
     }
     auto cl = req.headers.find("content-length");
     if (cl != req.headers.end()) {
-      size_t len = (size_t)strtoull(cl->second.c_str(), nullptr, 10);
-      if (len > g_cfg.max_body_bytes) {
+      plan.has_length = true;
+      plan.length = (size_t)strtoull(cl->second.c_str(), nullptr, 10);
+      if (plan.length > g_cfg.max_body_bytes) {
         respond(413, "Payload Too Large", "{\"error\":\"body too large\"}");
         return false;  // cannot resync mid-body: close the connection
       }
-      return read_exact_body(len, req.body);
     }
-    return true;  // no body
+    return true;
+  }
+
+  // Phase 2a: body into memory (JSON routes).
+  bool read_body(const BodyPlan& plan, std::string& out) {
+    if (plan.chunked) return read_chunked_body(out);
+    if (plan.has_length) return read_exact_body(plan.length, out);
+    return true;
+  }
+
+  // Phase 2b: body streamed to an fd in bounded chunks (workspace PUT).
+  // On sink failure sets *write_err and keeps draining is NOT attempted
+  // (caller closes the connection).
+  bool read_body_to_fd(const BodyPlan& plan, int fd, bool* write_err) {
+    *write_err = false;
+    auto sink = [&](const char* data, size_t n) -> bool {
+      size_t off = 0;
+      while (off < n) {
+        ssize_t w = ::write(fd, data + off, n - off);
+        if (w <= 0) { *write_err = true; return false; }
+        off += (size_t)w;
+      }
+      return true;
+    };
+    if (plan.chunked) return read_chunked_to_sink(sink);
+    if (plan.has_length) return read_sized_to_sink(plan.length, sink);
+    return true;
+  }
+
+  // Phase 2c: body consumed and discarded (error responses on routes
+  // whose body we never needed, keeping the connection parseable).
+  bool read_body_discard(const BodyPlan& plan) {
+    auto sink = [](const char*, size_t) { return true; };
+    if (plan.chunked) return read_chunked_to_sink(sink);
+    if (plan.has_length) return read_sized_to_sink(plan.length, sink);
+    return true;
+  }
+
+  // Streamed file response: headers + 256 KiB read/send loop, so a
+  // multi-GB workspace download costs O(chunk) memory.
+  void respond_file(const std::string& path, int64_t size) {
+    int fd = open(path.c_str(), O_RDONLY);
+    if (fd < 0) {
+      respond(404, "Not Found", "{\"error\":\"not found\"}");
+      return;
+    }
+    std::string head =
+        "HTTP/1.1 200 OK\r\nContent-Type: application/octet-stream"
+        "\r\nContent-Length: " + std::to_string(size) + "\r\n\r\n";
+    send_all(head);
+    char chunk[262144];
+    int64_t left = size;
+    while (left > 0) {
+      ssize_t n = read(fd, chunk, std::min<int64_t>(left, sizeof chunk));
+      if (n <= 0) break;
+      send_all(std::string(chunk, (size_t)n));
+      left -= n;
+    }
+    close(fd);
   }
 
   void respond(int code, const char* status, const std::string& body,
@@ -791,6 +911,62 @@ class Conn {
     out = buf_.substr(0, len);
     buf_.erase(0, len);
     return true;
+  }
+  template <typename Sink>
+  bool read_sized_to_sink(size_t len, Sink&& sink) {
+    // drain what is already buffered, then stream the rest in chunks
+    size_t from_buf = std::min(buf_.size(), len);
+    if (from_buf) {
+      if (!sink(buf_.data(), from_buf)) return false;
+      buf_.erase(0, from_buf);
+      len -= from_buf;
+    }
+    char chunk[262144];
+    while (len > 0) {
+      ssize_t n = recv(fd_, chunk, std::min(len, sizeof chunk), 0);
+      if (n <= 0) return false;
+      if (!sink(chunk, (size_t)n)) return false;
+      len -= (size_t)n;
+    }
+    return true;
+  }
+  template <typename Sink>
+  bool read_chunked_to_sink(Sink&& sink) {
+    size_t total = 0;
+    while (true) {
+      size_t pos;
+      while ((pos = buf_.find("\r\n")) == std::string::npos) {
+        if (buf_.size() > 4096) return false;  // chunk-size line is tiny
+        if (!fill()) return false;
+      }
+      size_t chunk_len = strtoull(buf_.substr(0, pos).c_str(), nullptr, 16);
+      total += chunk_len;
+      if (total > g_cfg.max_body_bytes) {
+        respond(413, "Payload Too Large", "{\"error\":\"body too large\"}");
+        return false;
+      }
+      buf_.erase(0, pos + 2);
+      if (chunk_len == 0) {
+        while (buf_.size() < 2) {
+          if (!fill()) return false;
+        }
+        buf_.erase(0, 2);
+        return true;
+      }
+      // stream this chunk's payload
+      size_t remaining = chunk_len;
+      while (remaining > 0) {
+        if (buf_.empty() && !fill()) return false;
+        size_t take = std::min(buf_.size(), remaining);
+        if (!sink(buf_.data(), take)) return false;
+        buf_.erase(0, take);
+        remaining -= take;
+      }
+      while (buf_.size() < 2) {
+        if (!fill()) return false;
+      }
+      buf_.erase(0, 2);  // chunk-terminating CRLF
+    }
   }
   bool read_chunked_body(std::string& out) {
     while (true) {
@@ -936,7 +1112,8 @@ static bool handle_execute(Conn& conn, const HttpRequest& req,
 
   double t_pre1 = now_ms();
   ExecOutcome outcome;
-  if (g_cfg.zygote && g_zygote && g_zygote->alive()) {
+  Zygote* zyg = g_zygote.load();
+  if (g_cfg.zygote && zyg && zyg->alive()) {
     outcome = run_via_zygote(script_path, stdout_path, stderr_path, extra_env,
                              timeout_s, workspace);
   } else {
@@ -962,8 +1139,9 @@ static bool handle_execute(Conn& conn, const HttpRequest& req,
     timings_str += ",\"child\":" + child_t;
   }
   timings_str += "}";
-  std::string resp = "{\"stdout\":" + json::quote(outcome.stdout_text) +
-                     ",\"stderr\":" + json::quote(outcome.stderr_text) +
+  std::string resp =
+      "{\"stdout\":" + json::quote(json::to_valid_utf8(outcome.stdout_text)) +
+      ",\"stderr\":" + json::quote(json::to_valid_utf8(outcome.stderr_text)) +
                      ",\"exit_code\":" + std::to_string(outcome.exit_code) +
                      timings_str +
                      ",\"files\":[";
@@ -989,40 +1167,64 @@ static bool handle_execute(Conn& conn, const HttpRequest& req,
 }
 
 // workspace file routes (PUT/GET), shared by the legacy pod-style routes
-// and the per-session routes
-static void handle_workspace_io(Conn& conn, const HttpRequest& req,
+// and the per-session routes. Returns false when the connection can no
+// longer be reused (body left unread / stream aborted).
+// PUT streams the body straight into the destination file and GET
+// streams the file out in chunks (reference parity with server.rs:69-96's
+// streaming; a 1 GiB round trip stays at O(256 KiB) executor memory).
+static bool handle_workspace_io(Conn& conn, const HttpRequest& req,
+                                const BodyPlan& plan,
                                 const std::string& workspace,
                                 const std::string& rel_encoded) {
   std::string rel = url_decode(rel_encoded);
   // a path naming a directory (trailing '/') or nothing is a client
   // error, not a 500 from the failed open
   if (!safe_rel_path(rel) || rel.empty() || rel.back() == '/') {
+    bool drained = conn.read_body_discard(plan);
     conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
-    return;
+    return drained;
   }
   std::string full = workspace + "/" + rel;
   if (req.method == "PUT") {
     size_t slash = full.rfind('/');
     if (slash != std::string::npos) mkdirs(full.substr(0, slash));
-    if (write_file(full, req.body)) {
-      conn.respond(204, "No Content", "");
-    } else if (errno == EISDIR || errno == ENOTDIR || errno == ENAMETOOLONG ||
-               errno == EINVAL) {
-      // the path itself is unusable: client error, not a server fault
-      conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
-    } else {
-      conn.respond(500, "Internal Server Error", "{\"error\":\"write failed\"}");
+    int fd = open(full.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0666);
+    if (fd < 0) {
+      bool drained = conn.read_body_discard(plan);
+      if (errno == EISDIR || errno == ENOTDIR || errno == ENAMETOOLONG ||
+          errno == EINVAL) {
+        conn.respond(400, "Bad Request", "{\"error\":\"bad path\"}");
+      } else {
+        conn.respond(500, "Internal Server Error",
+                     "{\"error\":\"write failed\"}");
+      }
+      return drained;
     }
-  } else if (req.method == "GET") {
+    bool write_err = false;
+    bool ok = conn.read_body_to_fd(plan, fd, &write_err);
+    close(fd);
+    if (!ok) {
+      unlink(full.c_str());  // partial upload must not look complete
+      if (write_err)
+        conn.respond(500, "Internal Server Error",
+                     "{\"error\":\"write failed\"}");
+      return false;
+    }
+    conn.respond(204, "No Content", "");
+    return true;
+  }
+  if (!conn.read_body_discard(plan)) return false;
+  if (req.method == "GET") {
     struct stat st;
     if (stat(full.c_str(), &st) != 0 || !S_ISREG(st.st_mode)) {
       conn.respond(404, "Not Found", "{\"error\":\"not found\"}");
     } else {
-      conn.respond(200, "OK", read_file(full), "application/octet-stream");
+      conn.respond_file(full, (int64_t)st.st_size);
     }
   } else {
     conn.respond(405, "Method Not Allowed", "{\"error\":\"method\"}");
   }
+  return true;
 }
 
 static bool lookup_sandbox(const std::string& id, std::string& workspace) {
@@ -1033,21 +1235,54 @@ static bool lookup_sandbox(const std::string& id, std::string& workspace) {
   return true;
 }
 
+// Bounded concurrency: the acceptor blocks while max_connections handler
+// threads are live (the kernel backlog absorbs the burst), so a
+// connection storm cannot create unbounded detached threads.
+static void handle_conn(int fd);
+
+static std::atomic<int> g_active_conns{0};
+static std::mutex g_conn_gate_mu;
+static std::condition_variable g_conn_gate_cv;
+
+static void handle_conn_counted(int fd) {
+  handle_conn(fd);
+  g_active_conns.fetch_sub(1);
+  g_conn_gate_cv.notify_one();
+}
+
 static void handle_conn(int fd) {
   Conn conn(fd);
   while (true) {
     HttpRequest req;
-    if (!conn.read_request(req)) return;
+    BodyPlan plan;
+    if (!conn.read_request_headers(req, plan)) return;
+
+    // workspace file routes consume their body streaming; every other
+    // route reads it into memory here
+    bool is_ws_route = req.path.rfind("/workspace/", 0) == 0;
+    bool is_session_ws = false;
+    if (!is_ws_route && req.path.rfind("/sandboxes/", 0) == 0) {
+      std::string rest = req.path.substr(strlen("/sandboxes/"));
+      size_t slash = rest.find('/');
+      is_session_ws = slash != std::string::npos &&
+                      rest.compare(slash + 1, strlen("workspace/"),
+                                   "workspace/", strlen("workspace/")) == 0;
+    }
+    if (!is_ws_route && !is_session_ws) {
+      if (!conn.read_body(plan, req.body)) return;
+    }
 
     if (req.method == "GET" && req.path == "/healthz") {
-      bool warm = g_zygote && g_zygote->warm();
+      Zygote* zyg = g_zygote.load();
+      bool warm = zyg && zyg->warm();
       conn.respond(200, "OK",
                    std::string("{\"status\":\"ok\",\"warm\":") +
                        (warm ? "true" : "false") + "}");
     } else if (req.path.rfind("/workspace/", 0) == 0) {
       // legacy pod-style routes: the default workspace
-      handle_workspace_io(conn, req, g_cfg.workspace,
-                          req.path.substr(strlen("/workspace/")));
+      if (!handle_workspace_io(conn, req, plan, g_cfg.workspace,
+                               req.path.substr(strlen("/workspace/"))))
+        return;
     } else if (req.method == "POST" && req.path == "/execute") {
       handle_execute(conn, req, g_cfg.workspace);
     } else if (req.method == "POST" && req.path == "/execute-ephemeral") {
@@ -1087,7 +1322,9 @@ static void handle_conn(int fd) {
                                                                  : slash);
       std::string ws;
       if (!lookup_sandbox(id, ws)) {
+        bool drained = !is_session_ws || conn.read_body_discard(plan);
         conn.respond(404, "Not Found", "{\"error\":\"no such sandbox\"}");
+        if (!drained) return;
       } else if (slash == std::string::npos) {
         if (req.method == "DELETE") {
           {
@@ -1102,7 +1339,9 @@ static void handle_conn(int fd) {
       } else {
         std::string sub = rest.substr(slash + 1);
         if (sub.rfind("workspace/", 0) == 0) {
-          handle_workspace_io(conn, req, ws, sub.substr(strlen("workspace/")));
+          if (!handle_workspace_io(conn, req, plan, ws,
+                                   sub.substr(strlen("workspace/"))))
+            return;
         } else if (req.method == "POST" && sub == "execute") {
           handle_execute(conn, req, ws);
         } else {
@@ -1199,6 +1438,9 @@ int main(int, char**) {
   g_cfg.max_body_bytes =
       (size_t)strtoull(env_or("APP_MAX_BODY_BYTES", "1073741824").c_str(),
                        nullptr, 10);
+  g_cfg.max_connections =
+      atoi(env_or("APP_MAX_CONNECTIONS", "512").c_str());
+  if (g_cfg.max_connections < 1) g_cfg.max_connections = 1;
 
   // default runtime dir: the directory containing this binary
   std::string self_dir;
@@ -1272,8 +1514,9 @@ int main(int, char**) {
   }
 
   if (g_cfg.zygote) {
-    g_zygote = new Zygote();
-    if (!g_zygote->start()) {
+    Zygote* zyg = new Zygote();
+    g_zygote.store(zyg);
+    if (!zyg->start()) {
       fprintf(stderr, "executor-server: zygote failed to start; cold mode\n");
       g_cfg.zygote = false;
     } else {
@@ -1281,9 +1524,10 @@ int main(int, char**) {
       std::thread([]() {
         int respawns = 0;
         while (true) {
-          while (g_zygote->alive()) sleep(1);
+          Zygote* cur = g_zygote.load();
+          while (cur->alive()) sleep(1);
           int status = 0;
-          waitpid(g_zygote->pid(), &status, 0);  // reap (no zombie)
+          waitpid(cur->pid(), &status, 0);  // reap (no zombie)
           if (++respawns > 5) {
             fprintf(stderr, "executor-server: zygote kept dying; cold mode\n");
             return;
@@ -1291,8 +1535,8 @@ int main(int, char**) {
           fprintf(stderr, "executor-server: zygote died, respawning\n");
           Zygote* fresh = new Zygote();
           if (fresh->start()) {
-            g_zygote = fresh;  // old object intentionally leaked (threads
-                               // may still hold job states briefly)
+            g_zygote.store(fresh);  // old object intentionally leaked
+                                    // (request threads may still hold it)
           } else {
             return;
           }
@@ -1309,13 +1553,20 @@ int main(int, char**) {
           g_cfg.workspace.c_str(), (int)g_cfg.zygote);
 
   while (true) {
+    {
+      std::unique_lock<std::mutex> lk(g_conn_gate_mu);
+      g_conn_gate_cv.wait(lk, [] {
+        return g_active_conns.load() < g_cfg.max_connections;
+      });
+    }
     int fd = accept(listen_fd, nullptr, nullptr);
     if (fd < 0) {
       if (errno == EINTR) continue;
       perror("accept");
       break;
     }
-    std::thread(handle_conn, fd).detach();
+    g_active_conns.fetch_add(1);
+    std::thread(handle_conn_counted, fd).detach();
   }
   return 0;
 }

@@ -428,6 +428,27 @@ class DeviceArray:
             self._host = out
         return self._host
 
+    def _mutable_host(self) -> _np.ndarray:
+        """Materialize for in-place mutation: the host copy becomes the
+        source of truth, the (now stale) device buffer is released, and
+        the next device op re-uploads lazily (`_dev_handle`). Repeated
+        `x[i] = v` therefore costs one download total, not one
+        upload per assignment."""
+        host = self.materialize()
+        if self._handle is not None:
+            try:
+                backend().free(self._handle)
+            except Exception:
+                pass
+            self._handle = None
+        return host
+
+    def _dev_handle(self):
+        if self._handle is None:
+            arr = _np.ascontiguousarray(self.materialize())
+            self._handle = backend().upload(arr)
+        return self._handle
+
     def __array__(self, dtype=None, copy=None):
         host = self.materialize()
         if dtype is not None and dtype != host.dtype:
@@ -443,7 +464,7 @@ class DeviceArray:
         if src is not None and dst is not None and not kwargs:
             if src == dst:
                 return self
-            h = backend().convert(self._handle, src, dst, self.size)
+            h = backend().convert(self._dev_handle(), src, dst, self.size)
             return DeviceArray(h, self.shape, target)
         return self.materialize().astype(dtype, **kwargs)
 
@@ -453,6 +474,16 @@ class DeviceArray:
 
     def __getitem__(self, idx):
         return self.materialize()[idx]
+
+    def __setitem__(self, idx, value):
+        # in-place mutation (x[0] = 1, x[x < 0] = 0, ...): CPU numpy
+        # supports it, so the sandbox contract requires it; the host copy
+        # becomes authoritative and the device buffer is dropped
+        if isinstance(idx, DeviceArray):
+            idx = idx.materialize()
+        if isinstance(value, DeviceArray):
+            value = value.materialize()
+        self._mutable_host()[idx] = value
 
     def __float__(self):
         return float(self.materialize())
@@ -488,7 +519,7 @@ class DeviceArray:
 
     # -- device compute -------------------------------------------------
     def _unary(self, opname):
-        out = backend().unary(self._handle, _UNARY[opname], _dtype_code(self.dtype), self.size)
+        out = backend().unary(self._dev_handle(), _UNARY[opname], _dtype_code(self.dtype), self.size)
         return DeviceArray(out, self.shape, self.dtype)
 
     def _binary(self, opname, other, reverse=False):
@@ -497,13 +528,13 @@ class DeviceArray:
             if other.shape != self.shape or other.dtype != self.dtype:
                 return NotImplemented
             a, b = (other, self) if reverse else (self, other)
-            out = backend().binary(a._handle, b._handle, _BINARY[opname], code, self.size)
+            out = backend().binary(a._dev_handle(), b._dev_handle(), _BINARY[opname], code, self.size)
             return DeviceArray(out, self.shape, self.dtype)
         if isinstance(other, (int, float)):
             if reverse and opname in ("subtract", "divide", "true_divide", "power"):
                 return NotImplemented  # scalar-first sub/div/pow: fall back
             out = backend().binary_scalar(
-                self._handle, float(other), _BINARY[opname], code, self.size
+                self._dev_handle(), float(other), _BINARY[opname], code, self.size
             )
             return DeviceArray(out, self.shape, self.dtype)
         return NotImplemented
@@ -511,7 +542,7 @@ class DeviceArray:
     def sum(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
-                backend().sum(self._handle, _dtype_code(self.dtype), self.size, 0)
+                backend().sum(self._dev_handle(), _dtype_code(self.dtype), self.size, 0)
             )
         return self.materialize().sum(axis=axis, **kwargs)
 
@@ -523,14 +554,14 @@ class DeviceArray:
     def square_sum(self):
         """Fused sum(x*x) -- no intermediate array."""
         return self.dtype.type(
-            backend().sum(self._handle, _dtype_code(self.dtype), self.size, 1)
+            backend().sum(self._dev_handle(), _dtype_code(self.dtype), self.size, 1)
         )
 
     def max(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
                 backend().sum(
-                    self._handle, _dtype_code(self.dtype), self.size, _REDUCE_MAX
+                    self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MAX
                 )
             )
         return self.materialize().max(axis=axis, **kwargs)
@@ -539,7 +570,7 @@ class DeviceArray:
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
                 backend().sum(
-                    self._handle, _dtype_code(self.dtype), self.size, _REDUCE_MIN
+                    self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MIN
                 )
             )
         return self.materialize().min(axis=axis, **kwargs)
@@ -594,6 +625,24 @@ class DeviceArray:
         host_inputs = [
             x.materialize() if isinstance(x, DeviceArray) else x for x in inputs
         ]
+        out = kwargs.get("out")
+        if out is not None:
+            # numpy rejects duck-typed out= targets: substitute each
+            # DeviceArray's mutable host copy (updated in place, device
+            # buffer invalidated) and hand the originals back
+            outs = out if isinstance(out, tuple) else (out,)
+            host_outs = tuple(
+                o._mutable_host() if isinstance(o, DeviceArray) else o
+                for o in outs
+            )
+            kwargs = {**kwargs, "out": host_outs}
+            result = getattr(ufunc, method)(*host_inputs, **kwargs)
+            if isinstance(result, tuple):
+                return tuple(
+                    orig if isinstance(orig, DeviceArray) else res
+                    for orig, res in zip(outs, result)
+                )
+            return outs[0] if isinstance(outs[0], DeviceArray) else result
         return getattr(ufunc, method)(*host_inputs, **kwargs)
 
     # -- NEP 18: numpy functions -----------------------------------------
@@ -773,7 +822,7 @@ def matmul(a, b, _force=False):
     if da is None or db is None or da.dtype != db.dtype:
         return NotImplemented
     code = _dtype_code(da.dtype)
-    hc = backend().gemm(da._handle, db._handle, m, n, k, code)
+    hc = backend().gemm(da._dev_handle(), db._dev_handle(), m, n, k, code)
     return DeviceArray(hc, (m, n), da.dtype)
 
 

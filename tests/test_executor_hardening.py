@@ -137,3 +137,63 @@ def test_body_at_cap_still_accepted(hardened):
     r = hardened.client.put("/workspace/cap.bin", content=data)
     assert r.status_code == 204
     assert (Path(hardened.workspace) / "cap.bin").stat().st_size == len(data)
+
+
+def test_invalid_utf8_output_is_lossy_not_500(hardened):
+    # reference parity (String::from_utf8_lossy): raw non-UTF-8 bytes on
+    # stdout/stderr must come back as U+FFFD-substituted text in valid
+    # JSON, not poison the response
+    r = hardened.client.post(
+        "/execute",
+        json={
+            "source_code": (
+                "import sys\n"
+                "sys.stdout.buffer.write(b'\\xff\\xfeok\\x80')\n"
+                "sys.stderr.buffer.write(b'\\xc3')\n"  # truncated 2-byte seq
+            )
+        },
+    )
+    assert r.status_code == 200
+    body = r.json()  # decodes => the JSON is valid UTF-8
+    assert body["exit_code"] == 0
+    assert "ok" in body["stdout"]
+    assert "\ufffd" in body["stdout"]
+    assert body["stderr"].startswith("\ufffd") or "\ufffd" in body["stderr"]
+
+
+def test_utf8_passthrough_unchanged(hardened):
+    r = hardened.client.post(
+        "/execute", json={"source_code": "print('héllo \\u4e16\\u754c')"}
+    )
+    assert r.status_code == 200
+    assert r.json()["stdout"] == "héllo \u4e16\u754c\n"
+
+
+def test_large_put_get_streams_with_bounded_rss(tmp_path, executor_bin):
+    # 96 MiB round trip must not buffer the body: the server's peak RSS
+    # stays far below the payload size (streamed PUT -> file, chunked GET)
+    ex = RawExecutor(tmp_path, executor_bin)
+    try:
+        data = b"z" * (96 << 20)
+        assert ex.client.put("/workspace/big.bin", content=data).status_code == 204
+        got = ex.client.get("/workspace/big.bin")
+        assert got.status_code == 200
+        assert len(got.content) == len(data)
+        hwm_kb = 0
+        with open(f"/proc/{ex.proc.pid}/status") as f:
+            for line in f:
+                if line.startswith("VmHWM:"):
+                    hwm_kb = int(line.split()[1])
+        assert hwm_kb > 0 and hwm_kb < (64 << 10), f"peak RSS {hwm_kb} kB"
+    finally:
+        ex.close()
+
+
+def test_connection_storm_5000_bounded_threads(hardened):
+    # a storm far above APP_MAX_CONNECTIONS must neither kill the server
+    # nor leave it wedged (acceptor gate + kernel backlog absorb it)
+    for _ in range(5000):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.connect(hardened.sock)
+        s.close()
+    assert _alive(hardened)

@@ -226,3 +226,59 @@ def test_astype_stays_on_device(fake):
     assert x.astype(np.float64) is x  # same dtype: no copy, no transfer
     w = x.astype(np.int32)  # unsupported target: host fallback
     assert isinstance(w, np.ndarray) and w.dtype == np.int32
+
+
+def test_setitem_scalar_and_boolean_mask(fake):
+    # x[0] = 1 and x[x < 0] = 0 must work like CPU numpy
+    host = np.random.default_rng(8).normal(0, 1, 256)
+    x = _device(fake, host.copy())
+    x[0] = 42.0
+    x[x < 0] = 0.0
+    expect = host.copy()
+    expect[0] = 42.0
+    expect[expect < 0] = 0.0
+    np.testing.assert_array_equal(np.asarray(x), expect)
+
+
+def test_setitem_then_device_compute_sees_mutation(fake):
+    host = np.ones(64)
+    x = _device(fake, host.copy())
+    x[:32] = 3.0
+    # next device op re-uploads the mutated host copy lazily
+    assert float(x.sum()) == pytest.approx(32 * 3.0 + 32 * 1.0)
+    y = np.square(x)
+    assert isinstance(y, hipnp.DeviceArray)
+    np.testing.assert_array_equal(
+        np.asarray(y), np.concatenate([np.full(32, 9.0), np.ones(32)])
+    )
+
+
+def test_setitem_slice_and_fancy_index(fake):
+    host = np.arange(10.0)
+    x = _device(fake, host.copy())
+    x[2:5] = [7.0, 8.0, 9.0]
+    x[np.array([0, 9])] = -1.0
+    expect = host.copy()
+    expect[2:5] = [7.0, 8.0, 9.0]
+    expect[np.array([0, 9])] = -1.0
+    np.testing.assert_array_equal(np.asarray(x), expect)
+
+
+def test_ufunc_out_devicearray_target(fake):
+    host = np.random.default_rng(9).random(32)
+    x = _device(fake, host.copy())
+    y = _device(fake, np.zeros(32))
+    # out= a DeviceArray: numpy itself rejects duck arrays, the fallback
+    # must materialize the target, mutate it, and hand the wrapper back
+    r = np.add(x, 1.0, out=y)
+    assert r is y
+    np.testing.assert_allclose(np.asarray(y), host + 1.0)
+    # and the mutated target keeps working on the device path
+    assert float(y.sum()) == pytest.approx(float((host + 1.0).sum()))
+
+
+def test_ufunc_out_self_inplace(fake):
+    host = np.random.default_rng(10).random(16)
+    x = _device(fake, host.copy())
+    np.multiply(x, 2.0, out=x)
+    np.testing.assert_allclose(np.asarray(x), host * 2.0)
