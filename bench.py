@@ -124,7 +124,29 @@ def client_worker(argv) -> None:
                 await asyncio.gather(*(guarded() for _ in range(n)))
                 return latencies
 
-            await run_phase(args.warm)
+            # Warm to STEADY STATE, not to a fixed request count: the
+            # first ~100 requests pay pool prefill, allocator first-touch
+            # and zygote warm costs, so a small --warmup (the driver runs
+            # --steps 20 --warmup 5) would otherwise time the cold ramp
+            # (r01: 61 req/s at 20 steps vs 1390 at 512). Floor at the
+            # requested warmup, then run ~0.75 s batches until two
+            # consecutive batches agree within 12%, bounded by
+            # APP_BENCH_WARM_MAX_S.
+            await run_phase(max(1, args.warm))
+            warm_cap = float(os.environ.get("APP_BENCH_WARM_MAX_S", "75"))
+            warm_deadline = time.monotonic() + warm_cap
+            rate = 0.0
+            stable = 0
+            while time.monotonic() < warm_deadline and stable < 2:
+                n = min(512, max(16, int(rate * 0.75))) if rate else 32
+                t0 = time.monotonic()
+                await run_phase(n)
+                new_rate = n / max(1e-9, time.monotonic() - t0)
+                if rate and abs(new_rate - rate) <= 0.12 * rate:
+                    stable += 1
+                else:
+                    stable = 0
+                rate = new_rate
             Path(args.ready_file).touch()
             while not os.path.exists(args.go_file):
                 await asyncio.sleep(0.001)
@@ -136,6 +158,18 @@ def client_worker(argv) -> None:
             )
 
     asyncio.run(main_async())
+
+
+def _rank_visible_device(existing, local_rank: int) -> str:
+    """The HIP_VISIBLE_DEVICES value for one rank's service processes: if
+    the environment already restricts visibility to a list, take this
+    rank's entry of that list (inheriting the full list would land every
+    rank's engines on the same device); else the bare local rank."""
+    if existing:
+        ids = [x for x in existing.split(",") if x.strip() != ""]
+        if ids:
+            return ids[local_rank % len(ids)]
+    return str(local_rank)
 
 
 def _cpu_quota() -> int:
@@ -217,17 +251,11 @@ def main() -> None:
             torch.cuda.init()
             torch.zeros(1, device="cuda")
 
-    # pin this rank's service (and its sandbox engines) to one GPU; if the
-    # environment already restricts visibility to a list, take this rank's
-    # entry of that list (inheriting the full list would land every rank's
-    # engines on the same device)
+    # pin this rank's service (and its sandbox engines) to one GPU
     if use_gpu:
-        existing = os.environ.get("HIP_VISIBLE_DEVICES")
-        if existing:
-            ids = [x for x in existing.split(",") if x.strip() != ""]
-            os.environ["HIP_VISIBLE_DEVICES"] = ids[local_rank % len(ids)]
-        else:
-            os.environ["HIP_VISIBLE_DEVICES"] = str(local_rank)
+        os.environ["HIP_VISIBLE_DEVICES"] = _rank_visible_device(
+            os.environ.get("HIP_VISIBLE_DEVICES"), local_rank
+        )
 
     result = asyncio.run(run_rank(args, rank, world_size, use_gpu))
 

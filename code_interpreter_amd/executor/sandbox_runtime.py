@@ -70,6 +70,12 @@ def _patched_import(name, globals=None, locals=None, fromlist=(), level=0):
 
             clip.write_videofile = quiet_write
             clip._ci_amd_quiet = True
+    elif name == "torch" or name.startswith("torch."):
+        caller = ""
+        if isinstance(globals, dict):
+            caller = globals.get("__name__") or ""
+        if not caller.startswith("torch"):
+            _maybe_install_hip_torch()
     elif name == "numpy" or name.startswith("numpy."):
         # only act on imports from OUTSIDE numpy: by the time a non-numpy
         # caller's `import numpy` returns, numpy is fully initialized
@@ -138,6 +144,41 @@ def _maybe_install_hip_numpy() -> None:
         if mode == "require":
             raise
         # auto mode: CPU numpy is the documented fallback
+
+
+_hiptorch_state = {"attempted": False, "installed": False}
+
+
+def _maybe_install_hip_torch() -> None:
+    """Route the torch matmul family to the hand-written MFMA kernels
+    (ops/hiptorch.py) once the user's `import torch` completes.
+
+    Modes (APP_HIP_TORCH): "auto" (route when a GPU + the extension are
+    available), "require" (raise if the HIP path is unusable), "off".
+    """
+    if _hiptorch_state["attempted"]:
+        return
+    mode = os.environ.get("APP_HIP_TORCH", "auto").lower()
+    if mode == "off":
+        return
+    torch_module = sys.modules.get("torch")
+    # same late-init guard as numpy: torch-internal imports fire this hook
+    # mid-init; "classes" is assigned near the tail of torch/__init__.py
+    if torch_module is None or not (
+        "Tensor" in torch_module.__dict__ and "classes" in torch_module.__dict__
+    ):
+        return
+    _hiptorch_state["attempted"] = True
+    ops_dir = os.environ.get("APP_OPS_DIR")
+    if ops_dir and ops_dir not in sys.path:
+        sys.path.insert(0, ops_dir)
+    try:
+        import hiptorch
+
+        _hiptorch_state["installed"] = hiptorch.install(mode=mode)
+    except Exception:
+        if mode == "require":
+            raise
 
 
 def preload() -> None:

@@ -220,6 +220,20 @@ DType dtype_from_int(int dt) {
   throw std::runtime_error("unsupported dtype code");
 }
 
+// GIL-released device-work region with exception transport: a C++ throw
+// inside Py_BEGIN/END_ALLOW_THREADS would skip the GIL reacquire and hit
+// PyErr_SetString unlocked (undefined behavior); capture and rethrow
+// after the GIL is back instead.
+#define NOGIL_BEGIN           \
+  std::string _nogil_err;     \
+  Py_BEGIN_ALLOW_THREADS;     \
+  try {
+#define NOGIL_END                                               \
+  }                                                             \
+  catch (const std::exception& _e) { _nogil_err = _e.what(); }  \
+  Py_END_ALLOW_THREADS;                                         \
+  if (!_nogil_err.empty()) throw std::runtime_error(_nogil_err);
+
 // C++ exceptions -> Python RuntimeError
 #define WRAP_BEGIN try {
 #define WRAP_END                                  \
@@ -249,7 +263,7 @@ PyObject* py_init(PyObject*, PyObject* args) {
   if (!PyArg_ParseTuple(args, "|i", &device)) return nullptr;
   WRAP_BEGIN
   if (g.initialized) Py_RETURN_NONE;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   HIP_CHECK(hipSetDevice(device));
   HIP_CHECK(hipStreamCreateWithFlags(&g.compute, hipStreamNonBlocking));
   HIP_CHECK(hipStreamCreateWithFlags(&g.copy, hipStreamNonBlocking));
@@ -269,7 +283,7 @@ PyObject* py_init(PyObject*, PyObject* args) {
         g.use_mempool = true;
     }
   }
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   g.device = device;
   g.initialized = true;
   Py_RETURN_NONE;
@@ -282,9 +296,9 @@ PyObject* py_alloc(PyObject*, PyObject* args) {
   WRAP_BEGIN
   ensure_init();
   void* p = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   p = pool_alloc(nbytes);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(p, nbytes));
   WRAP_END
 }
@@ -311,7 +325,7 @@ PyObject* py_upload(PyObject*, PyObject* args) {
   ensure_init();
   int64_t nbytes = view.len;
   void* dev = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   ensure_staging();
   dev = pool_alloc(nbytes);
   // the allocation is stream-ordered on the compute stream; the copy
@@ -335,7 +349,7 @@ PyObject* py_upload(PyObject*, PyObject* args) {
   // compute stream must not run ahead of the upload
   HIP_CHECK(hipEventSynchronize(g.pin_evt[0]));
   HIP_CHECK(hipEventSynchronize(g.pin_evt[1]));
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   PyBuffer_Release(&view);
   return PyLong_FromUnsignedLongLong(register_buf(dev, nbytes));
   WRAP_END
@@ -356,7 +370,7 @@ PyObject* py_download(PyObject*, PyObject* args) {
     PyBuffer_Release(&view);
     throw std::runtime_error("download target too small");
   }
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   ensure_staging();
   HIP_CHECK(hipStreamSynchronize(g.compute));
   char* dst = (char*)view.buf;
@@ -385,7 +399,7 @@ PyObject* py_download(PyObject*, PyObject* args) {
       parallel_memcpy(dst + pending_off[s], g.pin[s], pending_len[s]);
     }
   }
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   PyBuffer_Release(&view);
   Py_RETURN_NONE;
   WRAP_END
@@ -401,10 +415,10 @@ PyObject* py_rand(PyObject*, PyObject* args) {
   DType dtype = dtype_from_int(dt);
   int64_t esize = dtype == DType::F64 ? 8 : 4;
   void* dev = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   dev = pool_alloc(n * esize);
   launch_rand_uniform(dtype, dev, n, seed, g.rand_offset, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   g.rand_offset += (uint64_t)n;  // never reuse counters within a process
   return PyLong_FromUnsignedLongLong(register_buf(dev, n * esize));
   WRAP_END
@@ -419,10 +433,10 @@ PyObject* py_randn(PyObject*, PyObject* args) {
   WRAP_BEGIN
   ensure_init();
   void* dev = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   dev = pool_alloc(n * 8);
   launch_rand_normal(dev, n, seed, g.rand_offset, mu, sigma, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   g.rand_offset += (uint64_t)n;
   return PyLong_FromUnsignedLongLong(register_buf(dev, n * 8));
   WRAP_END
@@ -440,10 +454,10 @@ PyObject* py_convert(PyObject*, PyObject* args) {
   DevBuf& in = get_buf(h);
   int64_t out_size = n * (ddt == DType::F64 ? 8 : 4);
   void* out = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   out = pool_alloc(out_size);
   launch_convert(sdt, ddt, in.ptr, out, n, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(out, out_size));
   WRAP_END
 }
@@ -458,10 +472,10 @@ PyObject* py_unary(PyObject*, PyObject* args) {
   DType dtype = dtype_from_int(dt);
   DevBuf& in = get_buf(h);
   void* out = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   out = pool_alloc(in.size);
   launch_unary(dtype, (UnaryOp)op, in.ptr, out, n, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(out, in.size));
   WRAP_END
 }
@@ -477,10 +491,10 @@ PyObject* py_binary(PyObject*, PyObject* args) {
   DevBuf& a = get_buf(ha);
   DevBuf& b = get_buf(hb);
   void* out = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   out = pool_alloc(a.size);
   launch_binary(dtype, (BinOp)op, a.ptr, b.ptr, out, n, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(out, a.size));
   WRAP_END
 }
@@ -497,10 +511,10 @@ PyObject* py_binary_scalar(PyObject*, PyObject* args) {
   DType dtype = dtype_from_int(dt);
   DevBuf& a = get_buf(ha);
   void* out = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   out = pool_alloc(a.size);
   launch_binary_scalar(dtype, (BinOp)op, a.ptr, scalar, out, n, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(out, a.size));
   WRAP_END
 }
@@ -518,7 +532,7 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   DType dtype = dtype_from_int(dt);
   DevBuf& in = get_buf(h);
   double result = 0;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   ensure_reduce_scratch();
   launch_sum(dtype, (ReduceOp)mode, in.ptr, g.reduce_scratch, g.scalar_dev, n,
              g.compute);
@@ -529,7 +543,7 @@ PyObject* py_sum(PyObject*, PyObject* args) {
     result = *g.scalar_pin;
   else
     result = (double)*(float*)g.scalar_pin;
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyFloat_FromDouble(result);
   WRAP_END
 }
@@ -546,7 +560,7 @@ PyObject* py_gemm(PyObject*, PyObject* args) {
   DevBuf& b = get_buf(hb);
   int64_t esize = dt == 1 ? 8 : (dt == 0 ? 4 : 2);
   void* out = nullptr;
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   out = pool_alloc((int64_t)m * n * esize);
   if (dt == 1)
     launch_gemm_f64((const double*)a.ptr, (const double*)b.ptr, (double*)out,
@@ -566,18 +580,63 @@ PyObject* py_gemm(PyObject*, PyObject* args) {
   } else
     launch_gemm_bf16((const uint16_t*)a.ptr, (const uint16_t*)b.ptr,
                      (uint16_t*)out, m, n, k, g.compute);
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   return PyLong_FromUnsignedLongLong(register_buf(out, (int64_t)m * n * esize));
   WRAP_END
+}
+
+// torch interop: raw-pointer GEMM on the CALLER's stream and context.
+// The sandbox torch hook (ops/hiptorch.py) passes torch-allocated device
+// buffers and torch's current HIP stream, so the hand-written MFMA
+// kernels run inside torch's own context with torch's stream ordering --
+// no _hipops state (init, pools, staging) is touched. dtype codes:
+// 0=f32, 1=f64, 2=bf16. For bf16 the caller passes scratch for the
+// B pre-transpose when the 256^2-tile fast path applies (bt != 0).
+PyObject* py_gemm_raw(PyObject*, PyObject* args) {
+  unsigned long long pa, pb, pc, pbt, stream;
+  int m, n, k, dt;
+  if (!PyArg_ParseTuple(args, "KKKKiiiiK", &pa, &pb, &pc, &pbt, &m, &n, &k,
+                        &dt, &stream))
+    return nullptr;
+  WRAP_BEGIN
+  if (dt < 0 || dt > 2) throw std::runtime_error("gemm_raw: bad dtype code");
+  hipStream_t s = (hipStream_t)stream;
+  NOGIL_BEGIN
+  if (dt == 1)
+    launch_gemm_f64((const double*)pa, (const double*)pb, (double*)pc, m, n,
+                    k, s);
+  else if (dt == 0)
+    launch_gemm_f32((const float*)pa, (const float*)pb, (float*)pc, m, n, k,
+                    s);
+  else if (dt == 2) {
+    if (pbt && gemm_bf16_256_supported(m, n, k)) {
+      launch_transpose_bf16((const uint16_t*)pb, (uint16_t*)pbt, k, n, s);
+      launch_gemm_bf16_256((const uint16_t*)pa, (const uint16_t*)pbt,
+                           (uint16_t*)pc, m, n, k, s);
+    } else {
+      launch_gemm_bf16((const uint16_t*)pa, (const uint16_t*)pb,
+                       (uint16_t*)pc, m, n, k, s);
+    }
+  }
+  NOGIL_END
+  Py_RETURN_NONE;
+  WRAP_END
+}
+
+PyObject* py_gemm_bf16_256_ok(PyObject*, PyObject* args) {
+  int m, n, k;
+  if (!PyArg_ParseTuple(args, "iii", &m, &n, &k)) return nullptr;
+  if (gemm_bf16_256_supported(m, n, k)) Py_RETURN_TRUE;
+  Py_RETURN_FALSE;
 }
 
 PyObject* py_synchronize(PyObject*, PyObject*) {
   WRAP_BEGIN
   ensure_init();
-  Py_BEGIN_ALLOW_THREADS;
+  NOGIL_BEGIN
   HIP_CHECK(hipStreamSynchronize(g.compute));
   HIP_CHECK(hipStreamSynchronize(g.copy));
-  Py_END_ALLOW_THREADS;
+  NOGIL_END
   Py_RETURN_NONE;
   WRAP_END
 }
@@ -613,6 +672,11 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"gemm_raw", py_gemm_raw, METH_VARARGS,
+     "gemm_raw(pa, pb, pc, pbt, m, n, k, dtype, stream): raw-pointer GEMM "
+     "on the caller's stream (torch interop; dtype 0=f32 1=f64 2=bf16)"},
+    {"gemm_bf16_256_ok", py_gemm_bf16_256_ok, METH_VARARGS,
+     "gemm_bf16_256_ok(m, n, k) -> bool (256-tile fast path applies?)"},
     {"synchronize", py_synchronize, METH_NOARGS, "sync all streams"},
     {"mem_info", py_mem_info, METH_NOARGS, "(free, total) bytes"},
     {nullptr, nullptr, 0, nullptr},

@@ -49,6 +49,22 @@ def _entry(rank: int, world_size: int, port: int, backend: str, fn, args, queue)
     try:
         if backend == "nccl":
             torch.cuda.set_device(rank % max(1, torch.cuda.device_count()))
+            # route this rank's matmuls to the hand-written MFMA kernels
+            # (same policy env as the sandbox runtime: APP_HIP_TORCH)
+            mode = os.environ.get("APP_HIP_TORCH", "auto").lower()
+            if mode != "off":
+                try:
+                    import sys
+
+                    ops_dir = os.path.dirname(os.path.abspath(__file__))
+                    if ops_dir not in sys.path:
+                        sys.path.insert(0, ops_dir)
+                    import hiptorch
+
+                    hiptorch.install(mode=mode)
+                except ImportError:
+                    if mode == "require":
+                        raise
         result = fn(rank, world_size, *args)
         if rank == 0 and queue is not None:
             queue.put(result)
