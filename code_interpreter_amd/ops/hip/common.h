@@ -68,5 +68,9 @@ void launch_gemm_bf16(const uint16_t* a, const uint16_t* b, uint16_t* c, int m,
 bool gemm_bf16_256_supported(int m, int n, int k);
 void launch_gemm_bf16_256(const uint16_t* a, const uint16_t* bt, uint16_t* c,
                           int m, int n, int k, hipStream_t stream);
+// variant T: same tile structure, B consumed in native [K][N] via
+// ds_read_b64_tr_b16 (no pre-transpose pass, no Bt scratch)
+void launch_gemm_bf16_256t(const uint16_t* a, const uint16_t* b, uint16_t* c,
+                           int m, int n, int k, hipStream_t stream);
 void launch_transpose_bf16(const uint16_t* in, uint16_t* out, int k, int n,
                            hipStream_t stream);

@@ -214,6 +214,17 @@ DevBuf& get_buf(uint64_t h) {
   return it->second;
 }
 
+// bf16 256-tile variant: "t" (default; tr16 transpose-read, no B
+// pre-transpose) or "b" (the r01 pre-transpose kernel, kept for A/B)
+bool use_bf16_tr16() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("APP_BF16_256_VARIANT");
+    v = (e && e[0] == 'b') ? 0 : 1;
+  }
+  return v == 1;
+}
+
 DType dtype_from_int(int dt) {
   if (dt == 0) return DType::F32;
   if (dt == 1) return DType::F64;
@@ -569,14 +580,20 @@ PyObject* py_gemm(PyObject*, PyObject* args) {
     launch_gemm_f32((const float*)a.ptr, (const float*)b.ptr, (float*)out, m,
                     n, k, g.compute);
   else if (gemm_bf16_256_supported(m, n, k)) {
-    // fast path: pre-transpose B to [N][K] (bandwidth-bound, ~1-2% of
-    // GEMM time) so both operands stream K-contiguous through glds
-    void* bt = pool_alloc((int64_t)n * k * 2);
-    launch_transpose_bf16((const uint16_t*)b.ptr, (uint16_t*)bt, k, n,
-                          g.compute);
-    launch_gemm_bf16_256((const uint16_t*)a.ptr, (const uint16_t*)bt,
-                         (uint16_t*)out, m, n, k, g.compute);
-    pool_free(bt, (int64_t)n * k * 2);
+    if (use_bf16_tr16()) {
+      // fast path: B consumed native [K][N] through the tr16 image
+      launch_gemm_bf16_256t((const uint16_t*)a.ptr, (const uint16_t*)b.ptr,
+                            (uint16_t*)out, m, n, k, g.compute);
+    } else {
+      // r01 path: pre-transpose B to [N][K] so both operands stream
+      // K-contiguous through glds
+      void* bt = pool_alloc((int64_t)n * k * 2);
+      launch_transpose_bf16((const uint16_t*)b.ptr, (uint16_t*)bt, k, n,
+                            g.compute);
+      launch_gemm_bf16_256((const uint16_t*)a.ptr, (const uint16_t*)bt,
+                           (uint16_t*)out, m, n, k, g.compute);
+      pool_free(bt, (int64_t)n * k * 2);
+    }
   } else
     launch_gemm_bf16((const uint16_t*)a.ptr, (const uint16_t*)b.ptr,
                      (uint16_t*)out, m, n, k, g.compute);
@@ -609,7 +626,10 @@ PyObject* py_gemm_raw(PyObject*, PyObject* args) {
     launch_gemm_f32((const float*)pa, (const float*)pb, (float*)pc, m, n, k,
                     s);
   else if (dt == 2) {
-    if (pbt && gemm_bf16_256_supported(m, n, k)) {
+    if (gemm_bf16_256_supported(m, n, k) && use_bf16_tr16()) {
+      launch_gemm_bf16_256t((const uint16_t*)pa, (const uint16_t*)pb,
+                            (uint16_t*)pc, m, n, k, s);
+    } else if (pbt && gemm_bf16_256_supported(m, n, k)) {
       launch_transpose_bf16((const uint16_t*)pb, (uint16_t*)pbt, k, n, s);
       launch_gemm_bf16_256((const uint16_t*)pa, (const uint16_t*)pbt,
                            (uint16_t*)pc, m, n, k, s);

@@ -40,7 +40,7 @@ for rep in range(3):
 sys.exit(1 if bad else 0)
 '''
 
-for variant in ("b",):
+for variant in ("t", "b"):
     env = dict(os.environ, APP_BF16_256_VARIANT=variant)
     r = subprocess.run([sys.executable, "-c", CHECK], env=env, cwd=REPO,
                        capture_output=True, text=True, timeout=600)
