@@ -80,18 +80,32 @@ class PodExecutor:
     # -- pod lifecycle ----------------------------------------------------
 
     async def fill_pool(self) -> None:
-        """Top the warm pod queue up to target (parity:
-        fill_executor_pod_queue, kubernetes_code_executor.py:151-189)."""
-        while len(self.pod_queue) + self.spawning_count < self.pod_queue_target_length:
-            self.spawning_count += 1
+        """Top the warm pod queue up to target with CONCURRENT in-flight
+        spawns (parity: the reference tracks in-flight spawns and runs
+        them simultaneously, kubernetes_code_executor.py:151-189). r01
+        spawned serially and gave up on the first error; with multi-GB
+        ROCm executor images that is minutes of warmup and a fragile
+        prefill. Each spawn failure is absorbed and logged (spawn_pod
+        already retries x3 internally); the others keep going."""
+        need = (
+            self.pod_queue_target_length
+            - len(self.pod_queue)
+            - self.spawning_count
+        )
+        if need <= 0:
+            return
+        self.spawning_count += need
+
+        async def spawn_one() -> None:
             try:
                 pod = await self.spawn_pod()
                 self.pod_queue.append(pod)
             except Exception as e:
                 logger.warning("executor pod spawn failed: %s", e)
-                return
             finally:
                 self.spawning_count -= 1
+
+        await asyncio.gather(*(spawn_one() for _ in range(need)))
 
     # alias kept for reference-shaped call sites
     fill_executor_pod_queue = fill_pool

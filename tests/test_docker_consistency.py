@@ -1,0 +1,98 @@
+"""Static consistency checks between the container images and the code
+they ship (the service image was missing aiohttp and
+could not boot; the executor image's wrappers wrapped binaries that were
+never installed). Docker cannot run in this environment, so these checks
+parse the Dockerfiles against the control plane's ACTUAL imports — they
+rot with the code, not with a hand-maintained list."""
+
+import ast
+import re
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+PKG = REPO / "code_interpreter_amd"
+
+# import name -> pip distribution name where they differ
+PIP_NAMES = {
+    "google": "protobuf",
+    "grpc": "grpcio",
+    "grpc_reflection": "grpcio-reflection",
+}
+# imported lazily/optionally with a guarded fallback; not boot-critical
+OPTIONAL = {"mgpu"}
+
+
+def control_plane_imports():
+    mods = set()
+    for p in PKG.rglob("*.py"):
+        rel = p.relative_to(PKG).parts
+        # the executor/ and ops/ trees run inside the EXECUTOR image
+        if rel[0] in ("executor", "ops"):
+            continue
+        tree = ast.parse(p.read_text())
+        for node in ast.walk(tree):
+            if isinstance(node, ast.Import):
+                for a in node.names:
+                    mods.add(a.name.split(".")[0])
+            elif isinstance(node, ast.ImportFrom) and node.level == 0 and node.module:
+                mods.add(node.module.split(".")[0])
+    stdlib = set(sys.stdlib_module_names)
+    return {
+        PIP_NAMES.get(m, m)
+        for m in mods
+        if m not in stdlib and m != "code_interpreter_amd" and m not in OPTIONAL
+    }
+
+
+def test_service_image_installs_every_control_plane_import():
+    df = (REPO / "docker" / "Dockerfile.service").read_text()
+    pip_block = re.search(
+        r"pip install[^&]*", df.replace("\\\n", " "), re.DOTALL
+    ).group(0)
+    installed = set(re.findall(r"[A-Za-z0-9_\-\[\]]+", pip_block))
+    installed = {p.split("[")[0].lower().replace("_", "-") for p in installed}
+    missing = {
+        d for d in control_plane_imports()
+        if d.lower().replace("_", "-") not in installed
+    }
+    assert not missing, f"Dockerfile.service pip list missing: {sorted(missing)}"
+
+
+def test_service_image_has_kubectl_and_entrypoint():
+    df = (REPO / "docker" / "Dockerfile.service").read_text()
+    assert "kubernetes-client" in df  # provides kubectl for the pod backend
+    assert "code_interpreter_amd" in df
+    assert re.search(r'ENTRYPOINT.*python.*-m.*code_interpreter_amd', df)
+
+
+def test_executor_image_installs_wrapped_binaries():
+    df = (REPO / "docker" / "Dockerfile.executor").read_text()
+    flat = df.replace("\\\n", " ")
+    wrappers = REPO / "code_interpreter_amd" / "executor" / "wrappers"
+    for wrapper in wrappers.iterdir():
+        body = wrapper.read_text()
+        target = re.search(r"exec\s+(\S+)", body).group(1)
+        binary = Path(target).name
+        assert re.search(
+            rf"apt-get install[^&]*\b{binary}\b", flat
+        ), f"wrapper {wrapper.name} wraps {binary}, not installed in image"
+    # pandoc's PDF engine + the pymupdf the alias table promises (fitz)
+    assert "weasyprint" in flat
+    assert "pymupdf" in flat
+    # wrappers actually shipped onto PATH
+    assert "wrappers/" in flat and "/usr/local/bin" in flat
+
+
+def test_executor_image_builds_native_components():
+    df = (REPO / "docker" / "Dockerfile.executor").read_text()
+    assert "make -C" in df  # executor-server built in-image
+    assert "gfx950" in df and "build.py" in df  # HIP kernels built in-image
+    assert "HSA_ENABLE_IPC_MODE_LEGACY=0" in df
+
+
+def test_executor_wrappers_are_executable_sh():
+    wrappers = REPO / "code_interpreter_amd" / "executor" / "wrappers"
+    for wrapper in wrappers.iterdir():
+        first = wrapper.read_text().splitlines()[0]
+        assert first.startswith("#!"), wrapper

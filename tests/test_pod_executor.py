@@ -194,3 +194,63 @@ def test_file_roundtrip_through_pod(fake_cluster, tmp_path):
         assert "/workspace/out.txt" in r.files
 
     asyncio.run(run())
+
+
+def test_fill_pool_spawns_concurrently(tmp_path, executor_bin):
+    """Prefill must overlap spawns (reference parity: in-flight tracking,
+    kubernetes_code_executor.py:151-189): with a create that takes
+    ~0.15 s, filling a 4-pod pool serially would take >= 0.6 s; the
+    concurrent fill finishes in ~one spawn's time."""
+    fake = FakeKubectl(tmp_path, executor_bin)
+    in_flight = {"now": 0, "max": 0}
+    orig_create = fake.create
+
+    async def slow_create(*a, **kw):
+        in_flight["now"] += 1
+        in_flight["max"] = max(in_flight["max"], in_flight["now"])
+        try:
+            await asyncio.sleep(0.15)
+            return await orig_create(*a, **kw)
+        finally:
+            in_flight["now"] -= 1
+
+    fake.create = slow_create
+    try:
+        ex = make_executor(fake, tmp_path, pod_queue_target_length=4)
+
+        async def run():
+            t0 = time.monotonic()
+            await ex.fill_pool()
+            elapsed = time.monotonic() - t0
+            assert len(ex.pod_queue) == 4
+            assert ex.spawning_count == 0
+            assert in_flight["max"] >= 3, "spawns did not overlap"
+            assert elapsed < 0.45, f"serial prefill? {elapsed:.2f}s"
+            await ex.aclose()
+
+        asyncio.run(run())
+    finally:
+        fake.close()
+
+
+def test_fill_pool_absorbs_single_failures(tmp_path, executor_bin):
+    """One pod's spawn failing (even through all its retries) must not
+    abort the rest of the prefill."""
+    # 3 failures = exactly one spawn_pod() exhausting its 3 attempts
+    fake = FakeKubectl(tmp_path, executor_bin, fail_first_creates=3)
+    try:
+        ex = make_executor(fake, tmp_path, pod_queue_target_length=3)
+
+        async def run():
+            await ex.fill_pool()
+            # the failed slot is released; the others filled
+            assert len(ex.pod_queue) >= 2
+            assert ex.spawning_count == 0
+            # a second fill tops up the failed slot
+            await ex.fill_pool()
+            assert len(ex.pod_queue) == 3
+            await ex.aclose()
+
+        asyncio.run(run())
+    finally:
+        fake.close()
