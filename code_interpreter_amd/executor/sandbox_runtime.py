@@ -284,6 +284,47 @@ def install_missing_deps(source: str) -> None:
 
 
 # ---------------------------------------------------------------------------
+# shell escapes: `!cmd` lines (the reference runs scripts under xonsh,
+# server.rs:152-165, whose headline capability is shell escapes; full
+# xonsh is a Python-superset shell we deliberately do not ship -- see
+# MIGRATION.md -- but `!cmd` lines are restored here)
+# ---------------------------------------------------------------------------
+import re as _re
+
+_SHELL_LINE = _re.compile(r"^(\s*)!(?!=)(.+)$")
+
+
+def transform_shell_escapes(source: str):
+    """Rewrite `<indent>!cmd` lines into `<indent>__ci_shell__('cmd')`.
+    Returns the transformed source, or None if no line matched. Only
+    applied when the ORIGINAL source fails to compile as python (a line
+    cannot start with `!` in valid python), so pure-python scripts --
+    including ones with multiline strings containing `!`-leading lines --
+    are never touched."""
+    changed = False
+    out = []
+    for line in source.splitlines():
+        m = _SHELL_LINE.match(line)
+        if m:
+            indent, cmd = m.group(1), m.group(2).strip()
+            out.append(f"{indent}__ci_shell__({cmd!r})")
+            changed = True
+        else:
+            out.append(line)
+    return "\n".join(out) + "\n" if changed else None
+
+
+def _ci_shell(cmd: str) -> int:
+    """Run one shell escape: inherit stdout/stderr, return the exit code
+    (xonsh parity: a failing command does not abort the script)."""
+    import subprocess
+
+    sys.stdout.flush()
+    sys.stderr.flush()
+    return subprocess.run(cmd, shell=True).returncode
+
+
+# ---------------------------------------------------------------------------
 # user script execution
 # ---------------------------------------------------------------------------
 def _apply_sandbox_rlimits() -> None:
@@ -342,14 +383,26 @@ def run_user_script(script_path: str) -> int:
         "__name__": "__main__",
         "__file__": script_path,
         "__builtins__": builtins,
+        "__ci_shell__": _ci_shell,
     }
     try:
         code = compile(source, script_path, "exec")
     except SyntaxError:
-        import traceback
+        # `!cmd` shell escapes are invalid python: retry with the
+        # transform (APP_SHELL_ESCAPES=0 disables)
+        code = None
+        if _env_flag("APP_SHELL_ESCAPES"):
+            transformed = transform_shell_escapes(source)
+            if transformed is not None:
+                try:
+                    code = compile(transformed, script_path, "exec")
+                except SyntaxError:
+                    code = None
+        if code is None:
+            import traceback
 
-        traceback.print_exc(limit=0)
-        return 1
+            traceback.print_exc(limit=0)
+            return 1
     try:
         exec(code, script_globals)
     except SystemExit as e:

@@ -15,8 +15,10 @@
 //    is a job handoff instead of a cold interpreter + HIP start
 //    (the reference pays upm+pip+xonsh cold start per request,
 //    server.rs:126-169 -- including a "TODO ~80ms" it never took).
-//  - scripts run under plain python, not xonsh (deviation; xonsh's `!cmd`
-//    escapes are not supported -- use subprocess).
+//  - scripts run under the python sandbox runtime, not xonsh; xonsh's
+//    headline `!cmd` shell escapes ARE supported via a compile-gated
+//    source transform (sandbox_runtime.transform_shell_escapes);
+//    other xonsh syntax ($VAR, $(...)) is not (MIGRATION.md).
 //  - dependency auto-install (reference: upm guess + pip, server.rs:126-147)
 //    is an AST import scan + pip in the zygote child (depscan in
 //    sandbox_runtime.py), pointed at a wheelhouse via APP_PIP_EXTRA_ARGS.
@@ -648,6 +650,9 @@ static ExecOutcome run_cold(const std::string& script_path,
                             const std::map<std::string, std::string>& extra_env,
                             double timeout_s, const std::string& workspace) {
   ExecOutcome out;
+  // built before fork(): the child of this multithreaded process must
+  // not allocate (heap lock may be held by another thread at fork time)
+  std::string runner = g_cfg.runtime_dir + "/runner.py";
   pid_t pid = fork();
   if (pid < 0) {
     out.stderr_text = "fork failed";
@@ -663,8 +668,11 @@ static ExecOutcome run_cold(const std::string& script_path,
     if (se >= 0) dup2(se, 2);
     for (auto& kv : extra_env) setenv(kv.first.c_str(), kv.second.c_str(), 1);
     if (chdir(workspace.c_str()) != 0) _exit(126);
-    execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), script_path.c_str(),
-           (char*)nullptr);
+    // run through the sandbox runtime (runner.py), not bare python: the
+    // cold path must keep the zygote path's semantics (import hooks,
+    // dep install, rlimits, shell escapes)
+    execlp(g_cfg.python.c_str(), g_cfg.python.c_str(), runner.c_str(),
+           script_path.c_str(), (char*)nullptr);
     _exit(127);
   }
   // wait with timeout

@@ -232,3 +232,65 @@ def test_cold_mode_no_zygote(tmp_path, executor_bin):
         assert resp.json()["exit_code"] == 0
     finally:
         ex.close()
+
+
+# ---------------------------------------------------------------------------
+# `!cmd` shell escapes (reference: xonsh, server.rs:152-165; here a
+# compile-gated source transform in sandbox_runtime)
+# ---------------------------------------------------------------------------
+SHELL_ESCAPE_SCRIPT = """x = 6 * 7
+!echo "shell says $x"
+print("python says", x)
+if x > 10:
+    !echo nested-escape-ran
+!false
+print("still alive after failing command")
+"""
+
+
+def test_shell_escape_lines(raw_executor):
+    r = raw_executor.client.post(
+        "/execute", json={"source_code": SHELL_ESCAPE_SCRIPT}
+    )
+    body = r.json()
+    assert body["exit_code"] == 0, body["stderr"]
+    out = body["stdout"]
+    # $x is SHELL interpolation: undefined in the shell env -> empty
+    assert "shell says" in out
+    assert "python says 42" in out
+    assert "nested-escape-ran" in out
+    assert "still alive after failing command" in out
+
+
+def test_shell_escape_cold_path(tmp_path, executor_bin):
+    # APP_ZYGOTE=0: fork/exec fallback must run the same sandbox runtime
+    ex = RawExecutor(tmp_path, executor_bin, APP_ZYGOTE="0")
+    try:
+        r = ex.client.post(
+            "/execute", json={"source_code": '!echo cold-escape-ok\nprint("py")'}
+        )
+        body = r.json()
+        assert body["exit_code"] == 0, body["stderr"]
+        assert "cold-escape-ok" in body["stdout"]
+        assert "py" in body["stdout"]
+    finally:
+        ex.close()
+
+
+def test_plain_python_with_bang_in_string_untouched(raw_executor):
+    # a multiline string containing a `!`-leading line is valid python:
+    # the transform must never fire
+    src = 's = """\n!not a command\n"""\nprint(repr(s))'
+    r = raw_executor.client.post("/execute", json={"source_code": src})
+    body = r.json()
+    assert body["exit_code"] == 0
+    assert "!not a command" in body["stdout"]
+
+
+def test_real_syntax_error_still_reported(raw_executor):
+    r = raw_executor.client.post(
+        "/execute", json={"source_code": "def broken(:\n    pass"}
+    )
+    body = r.json()
+    assert body["exit_code"] == 1
+    assert "SyntaxError" in body["stderr"]

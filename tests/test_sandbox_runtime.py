@@ -132,3 +132,17 @@ def test_sandbox_rlimits_applied(tmp_path, executor_bin):
         assert r.exit_code == 0
     finally:
         asyncio.run(ex.aclose())
+
+
+def test_transform_shell_escapes_unit():
+    from code_interpreter_amd.executor import sandbox_runtime as rt
+
+    src = 'x = 1\n!ls -la /tmp\nif x:\n    !echo "q uo"\nprint(x)\n'
+    out = rt.transform_shell_escapes(src)
+    assert "__ci_shell__('ls -la /tmp')" in out
+    assert "    __ci_shell__('echo \"q uo\"')" in out
+    assert "print(x)" in out
+    # no escapes -> None (caller keeps the original)
+    assert rt.transform_shell_escapes("print(1)\n") is None
+    # != operator lines must not match
+    assert rt.transform_shell_escapes("a\n!= b\n") is None
