@@ -45,6 +45,10 @@ void launch_binary_scalar(DType dt, BinOp op, const void* a, double scalar,
 int reduce_num_partials(int64_t n);
 void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
                 void* out_scalar, int64_t n, hipStream_t stream);
+// single-axis reduction of a contiguous array viewed [outer][red][inner]
+void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
+                        int64_t outer, int64_t red, int64_t inner,
+                        hipStream_t stream);
 // philox4x32-10 uniform doubles/floats in [0, 1)
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
                          uint64_t offset, hipStream_t stream);

@@ -559,6 +559,66 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// reduce_axis(h, dtype, outer, red, inner, mode) -> handle
+// contiguous [outer][red][inner] reduced over the middle axis
+PyObject* py_reduce_axis(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt, mode;
+  long long outer, red, inner;
+  if (!PyArg_ParseTuple(args, "KiLLLi", &h, &dt, &outer, &red, &inner, &mode))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (mode < 0 || mode > 3) throw std::runtime_error("bad reduce mode");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  int64_t out_size = outer * inner * esize;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(out_size);
+  launch_reduce_axis(dtype, (ReduceOp)mode, in.ptr, out, outer, red, inner,
+                     g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, out_size));
+  WRAP_END
+}
+
+// gemm_batched(hA, hB, batch, m, n, k, dtype) -> handle
+// C[b] = A[b] @ B[b] for contiguous [batch][m][k] x [batch][k][n];
+// per-batch kernel launches enqueue back-to-back on the compute stream
+PyObject* py_gemm_batched(PyObject*, PyObject* args) {
+  unsigned long long ha, hb;
+  int batch, m, n, k, dt;
+  if (!PyArg_ParseTuple(args, "KKiiiii", &ha, &hb, &batch, &m, &n, &k, &dt))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (dt != 0 && dt != 1)
+    throw std::runtime_error("gemm_batched supports f32/f64");
+  DevBuf& a = get_buf(ha);
+  DevBuf& b = get_buf(hb);
+  int64_t esize = dt == 1 ? 8 : 4;
+  int64_t out_size = (int64_t)batch * m * n * esize;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(out_size);
+  for (int i = 0; i < batch; i++) {
+    const char* pa = (const char*)a.ptr + (int64_t)i * m * k * esize;
+    const char* pb = (const char*)b.ptr + (int64_t)i * k * n * esize;
+    char* pc = (char*)out + (int64_t)i * m * n * esize;
+    if (dt == 1)
+      launch_gemm_f64((const double*)pa, (const double*)pb, (double*)pc, m,
+                      n, k, g.compute);
+    else
+      launch_gemm_f32((const float*)pa, (const float*)pb, (float*)pc, m, n,
+                      k, g.compute);
+  }
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, out_size));
+  WRAP_END
+}
+
 // gemm(hA, hB, m, n, k, dtype) -> handle  (row-major C = A @ B)
 PyObject* py_gemm(PyObject*, PyObject* args) {
   unsigned long long ha, hb;
@@ -692,6 +752,10 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"reduce_axis", py_reduce_axis, METH_VARARGS,
+     "reduce_axis(h, dtype, outer, red, inner, mode) -> handle"},
+    {"gemm_batched", py_gemm_batched, METH_VARARGS,
+     "gemm_batched(ha, hb, batch, m, n, k, dtype) -> handle"},
     {"gemm_raw", py_gemm_raw, METH_VARARGS,
      "gemm_raw(pa, pb, pc, pbt, m, n, k, dtype, stream): raw-pointer GEMM "
      "on the caller's stream (torch interop; dtype 0=f32 1=f64 2=bf16)"},

@@ -11,6 +11,8 @@
 //  - RNG is Philox4x32-10 (counter-based): stateless, any index range can
 //    be generated independently at full bandwidth.
 
+#include <algorithm>
+
 #include "common.h"
 
 namespace {
@@ -260,6 +262,71 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// axis-wise reduction: in viewed as [outer][red][inner] -> out[outer][inner]
+// (covers any single-axis reduce of a contiguous N-D array). Two shapes:
+//  - inner > 1: one thread per (outer, inner) output element; lanes read
+//    consecutive `inner` addresses per step -> coalesced;
+//  - inner == 1 (last-axis reduce): one wave64 per slice, lanes stride
+//    the reduced axis, cross-lane butterfly combines.
+// MODE matches ReduceOp: 0 sum, 1 sum-of-squares, 2 max, 3 min.
+// ---------------------------------------------------------------------------
+template <int MODE>
+__device__ __forceinline__ double axis_init() {
+  if constexpr (MODE == 2) return -INFINITY;
+  else if constexpr (MODE == 3) return INFINITY;
+  else return 0.0;
+}
+
+template <int MODE>
+__device__ __forceinline__ double axis_combine(double acc, double v) {
+  if constexpr (MODE == 0) return acc + v;
+  else if constexpr (MODE == 1) return acc + v * v;
+  else return minmax_combine<MODE == 2>(acc, v);
+}
+
+template <int MODE>
+__device__ __forceinline__ double axis_wave_reduce(double v) {
+  if constexpr (MODE <= 1) return wave_reduce_sum<double>(v);
+  else return wave_reduce_minmax<MODE == 2>(v);
+}
+
+template <typename T, int MODE>
+__global__ void reduce_axis_inner_kernel(const T* __restrict__ in,
+                                         T* __restrict__ out, int64_t outer,
+                                         int64_t red, int64_t inner) {
+  int64_t total = outer * inner;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    int64_t o = idx / inner, i = idx % inner;
+    const T* p = in + o * red * inner + i;
+    double acc = axis_init<MODE>();
+    for (int64_t r = 0; r < red; r++)
+      acc = axis_combine<MODE>(acc, (double)p[r * inner]);
+    out[idx] = (T)acc;
+  }
+}
+
+template <typename T, int MODE>
+__global__ void reduce_axis_last_kernel(const T* __restrict__ in,
+                                        T* __restrict__ out, int64_t outer,
+                                        int64_t red) {
+  const int waves_per_block = blockDim.x >> 6;
+  const int lane = threadIdx.x & 63;
+  int64_t stride = (int64_t)gridDim.x * waves_per_block;
+  for (int64_t slice = (int64_t)blockIdx.x * waves_per_block +
+                       (threadIdx.x >> 6);
+       slice < outer; slice += stride) {
+    const T* p = in + slice * red;
+    double acc = axis_init<MODE>();
+    for (int64_t r = lane; r < red; r += 64)
+      acc = axis_combine<MODE>(acc, (double)p[r]);
+    acc = axis_wave_reduce<MODE>(acc);
+    if (lane == 0) out[slice] = (T)acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Philox4x32-10 uniform RNG
 // ---------------------------------------------------------------------------
 __device__ __forceinline__ void philox_round(uint32_t& c0, uint32_t& c1,
@@ -485,6 +552,45 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
   HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_reduce_axis_t(ReduceOp mode, const T* in, T* out,
+                                 int64_t outer, int64_t red, int64_t inner,
+                                 hipStream_t s) {
+  if (inner == 1) {
+    // one wave per slice; >= 2048 workgroups saturates the 8 XCDs
+    int waves_per_block = kBlock / 64;
+    int grid = (int)std::min<int64_t>(
+        (outer + waves_per_block - 1) / waves_per_block, 4096);
+    switch ((int)mode) {
+      case 0: hipLaunchKernelGGL((reduce_axis_last_kernel<T, 0>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red); break;
+      case 1: hipLaunchKernelGGL((reduce_axis_last_kernel<T, 1>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red); break;
+      case 2: hipLaunchKernelGGL((reduce_axis_last_kernel<T, 2>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red); break;
+      default: hipLaunchKernelGGL((reduce_axis_last_kernel<T, 3>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red); break;
+    }
+  } else {
+    int64_t total = outer * inner;
+    int grid = (int)std::min<int64_t>((total + kBlock - 1) / kBlock, 4096);
+    switch ((int)mode) {
+      case 0: hipLaunchKernelGGL((reduce_axis_inner_kernel<T, 0>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red, inner); break;
+      case 1: hipLaunchKernelGGL((reduce_axis_inner_kernel<T, 1>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red, inner); break;
+      case 2: hipLaunchKernelGGL((reduce_axis_inner_kernel<T, 2>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red, inner); break;
+      default: hipLaunchKernelGGL((reduce_axis_inner_kernel<T, 3>), dim3(grid), dim3(kBlock), 0, s, in, out, outer, red, inner); break;
+    }
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
+                        int64_t outer, int64_t red, int64_t inner,
+                        hipStream_t s) {
+  if (dt == DType::F64)
+    launch_reduce_axis_t(mode, (const double*)in, (double*)out, outer, red,
+                         inner, s);
+  else
+    launch_reduce_axis_t(mode, (const float*)in, (float*)out, outer, red,
+                         inner, s);
 }
 
 void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,

@@ -183,6 +183,20 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "reduce_axis":
+            h = _hipops.reduce_axis(
+                self._own(m["h"]), m["dtype"], m["outer"], m["red"],
+                m["inner"], m["mode"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "gemm_batched":
+            h = _hipops.gemm_batched(
+                self._own(m["ha"]), self._own(m["hb"]), m["batch"], m["m"],
+                m["n"], m["k"], m["dtype"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "gemm":
             h = _hipops.gemm(self._own(m["ha"]), self._own(m["hb"]), m["m"], m["n"], m["k"], m["dtype"])
             self.handles.add(h)

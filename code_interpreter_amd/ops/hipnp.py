@@ -282,6 +282,18 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def reduce_axis(self, h, dtype, outer, red, inner, mode):
+        return self._call(
+            {"op": "reduce_axis", "h": h, "dtype": dtype, "outer": outer,
+             "red": red, "inner": inner, "mode": mode}
+        )["h"]
+
+    def gemm_batched(self, ha, hb, batch, m, n, k, dtype):
+        return self._call(
+            {"op": "gemm_batched", "ha": ha, "hb": hb, "batch": batch,
+             "m": m, "n": n, "k": k, "dtype": dtype}
+        )["h"]
+
     def synchronize(self) -> None:
         self._call({"op": "sync"})
 
@@ -539,16 +551,65 @@ class DeviceArray:
             return DeviceArray(out, self.shape, self.dtype)
         return NotImplemented
 
+    def _axis_reduce(self, mode, axis, keepdims=False):
+        """Single-axis reduction on-device: the contiguous array viewed
+        as [outer][red][inner], reduced over the middle. Returns None
+        when not routable (caller falls back to host numpy)."""
+        code = _dtype_code(self.dtype)
+        if code is None:
+            return None
+        if isinstance(axis, (tuple, list)):
+            if len(axis) != 1:
+                return None
+            axis = axis[0]
+        if not isinstance(axis, (int, _np.integer)):
+            return None
+        nd = len(self.shape)
+        axis = int(axis)
+        if axis < 0:
+            axis += nd
+        if not 0 <= axis < nd or self.size == 0:
+            return None
+        outer = 1
+        for sdim in self.shape[:axis]:
+            outer *= sdim
+        red = self.shape[axis]
+        inner = 1
+        for sdim in self.shape[axis + 1:]:
+            inner *= sdim
+        h = backend().reduce_axis(
+            self._dev_handle(), code, outer, red, inner, mode
+        )
+        out_shape = (
+            self.shape[:axis]
+            + ((1,) if keepdims else ())
+            + self.shape[axis + 1:]
+        )
+        return DeviceArray(h, out_shape, self.dtype)
+
     def sum(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
                 backend().sum(self._dev_handle(), _dtype_code(self.dtype), self.size, 0)
             )
+        if axis is not None and set(kwargs) <= {"keepdims"}:
+            r = self._axis_reduce(_REDUCE_SUM, axis, kwargs.get("keepdims", False))
+            if r is not None:
+                return r
         return self.materialize().sum(axis=axis, **kwargs)
 
     def mean(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(float(self.sum()) / self.size)
+        if axis is not None and set(kwargs) <= {"keepdims"}:
+            r = self._axis_reduce(_REDUCE_SUM, axis, kwargs.get("keepdims", False))
+            if r is not None:
+                nd = len(self.shape)
+                ax = axis if not isinstance(axis, (tuple, list)) else axis[0]
+                red = self.shape[int(ax) if int(ax) >= 0 else int(ax) + nd]
+                out = r._binary("multiply", 1.0 / red)
+                if out is not NotImplemented:
+                    return out
         return self.materialize().mean(axis=axis, **kwargs)
 
     def square_sum(self):
@@ -564,6 +625,10 @@ class DeviceArray:
                     self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MAX
                 )
             )
+        if axis is not None and set(kwargs) <= {"keepdims"}:
+            r = self._axis_reduce(_REDUCE_MAX, axis, kwargs.get("keepdims", False))
+            if r is not None:
+                return r
         return self.materialize().max(axis=axis, **kwargs)
 
     def min(self, axis=None, **kwargs):
@@ -573,6 +638,10 @@ class DeviceArray:
                     self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MIN
                 )
             )
+        if axis is not None and set(kwargs) <= {"keepdims"}:
+            r = self._axis_reduce(_REDUCE_MIN, axis, kwargs.get("keepdims", False))
+            if r is not None:
+                return r
         return self.materialize().min(axis=axis, **kwargs)
 
     def var(self, axis=None, ddof=0, **kwargs):
@@ -612,13 +681,14 @@ class DeviceArray:
                 if r is not NotImplemented:
                     return r
         elif method == "reduce" and len(inputs) == 1:
-            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
-                if name == "add":
-                    return inputs[0].sum()
-                if name == "maximum":
-                    return inputs[0].max()
-                if name == "minimum":
-                    return inputs[0].min()
+            axis = kwargs.get("axis")
+            keepdims = bool(kwargs.get("keepdims", False))
+            meth = {"add": "sum", "maximum": "max", "minimum": "min"}.get(name)
+            if meth and set(kwargs) <= {"axis", "keepdims"}:
+                if axis is None and not keepdims:
+                    return getattr(inputs[0], meth)()
+                if axis is not None:
+                    return getattr(inputs[0], meth)(axis=axis, keepdims=keepdims)
         return self._fallback_ufunc(ufunc, method, inputs, kwargs)
 
     def _fallback_ufunc(self, ufunc, method, inputs, kwargs):
@@ -648,11 +718,11 @@ class DeviceArray:
     # -- NEP 18: numpy functions -----------------------------------------
     def __array_function__(self, func, types, args, kwargs):
         if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
-            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
-                return args[0].sum()
+            if set(kwargs) <= {"axis", "keepdims"}:
+                return args[0].sum(**kwargs)
         if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
-            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
-                return args[0].mean()
+            if set(kwargs) <= {"axis", "keepdims"}:
+                return args[0].mean(**kwargs)
         if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
             r = matmul(*args, _force=True)
             if r is not NotImplemented:
@@ -673,6 +743,8 @@ class DeviceArray:
                     if meth in ("std", "var") and "ddof" in kwargs:
                         call_kwargs["ddof"] = kwargs["ddof"]
                     return getattr(args[0], meth)(**call_kwargs)
+                if meth in ("max", "min") and kwargs.get("axis") is not None:
+                    return getattr(args[0], meth)(**kwargs)
         # generic fallback: materialize every DeviceArray
         host_args = [
             x.materialize() if isinstance(x, DeviceArray) else x for x in args
@@ -806,11 +878,19 @@ def square_sum(x):
 
 def matmul(a, b, _force=False):
     """Row-major 2D matmul on the MFMA matrix cores (f32: 32x32x2 f32 MFMA,
-    f64: 16x16x4 f64 MFMA). Returns NotImplemented when the shape/dtype is
+    f64: 16x16x4 f64 MFMA); equal-batch 3D stacks run per-batch GEMMs
+    enqueued back-to-back. Returns NotImplemented when the shape/dtype is
     not routable (caller falls back)."""
     _ensure_ready()
     a_shape = a.shape if hasattr(a, "shape") else _np.asarray(a).shape
     b_shape = b.shape if hasattr(b, "shape") else _np.asarray(b).shape
+    if (
+        len(a_shape) == 3
+        and len(b_shape) == 3
+        and a_shape[0] == b_shape[0]
+        and a_shape[2] == b_shape[1]
+    ):
+        return _matmul_batched(a, b, _force)
     if len(a_shape) != 2 or len(b_shape) != 2 or a_shape[1] != b_shape[0]:
         return NotImplemented
     m, k = a_shape
@@ -824,6 +904,25 @@ def matmul(a, b, _force=False):
     code = _dtype_code(da.dtype)
     hc = backend().gemm(da._dev_handle(), db._dev_handle(), m, n, k, code)
     return DeviceArray(hc, (m, n), da.dtype)
+
+
+def _matmul_batched(a, b, _force=False):
+    """[batch][m][k] @ [batch][k][n] via per-batch device GEMMs."""
+    a_shape = a.shape if hasattr(a, "shape") else _np.asarray(a).shape
+    b_shape = b.shape if hasattr(b, "shape") else _np.asarray(b).shape
+    batch, m, k = a_shape
+    n = b_shape[2]
+    if not _force and 2.0 * batch * m * n * k < MIN_MATMUL_FLOPS:
+        return NotImplemented
+    da = _to_device(a)
+    db = _to_device(b)
+    if da is None or db is None or da.dtype != db.dtype:
+        return NotImplemented
+    code = _dtype_code(da.dtype)
+    hc = backend().gemm_batched(
+        da._dev_handle(), db._dev_handle(), batch, m, n, k, code
+    )
+    return DeviceArray(hc, (batch, m, n), da.dtype)
 
 
 # ---------------------------------------------------------------------------

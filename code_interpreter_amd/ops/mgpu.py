@@ -102,13 +102,28 @@ def run_distributed(
     return queue.get() if not queue.empty() else None
 
 
-def allreduce_bucketed(tensors: List, bucket_bytes: int = DEFAULT_BUCKET_BYTES):
+def allreduce_bucketed(tensors: List, bucket_bytes: Optional[int] = None):
     """All-reduce a list of tensors in flattened buckets sized for xGMI
     ring pipelining (each bucket = one RCCL call; per-link bandwidth bound
     means fewer, larger collectives beat many small ones up to the bucket
-    size that still overlaps)."""
+    size that still overlaps). bucket_bytes=None picks the size from the
+    node topology model (parallel/topology.py): 64 MB default, shrunk so
+    small payloads still pipeline across >= 4 in-flight buckets."""
     import torch
     import torch.distributed as dist
+
+    if bucket_bytes is None:
+        total = sum(t.numel() * t.element_size() for t in tensors)
+        try:
+            from code_interpreter_amd.parallel.topology import (
+                recommended_bucket_bytes,
+            )
+
+            bucket_bytes = recommended_bucket_bytes(
+                total, dist.get_world_size()
+            )
+        except ImportError:  # standalone sandbox import: ops/ only
+            bucket_bytes = DEFAULT_BUCKET_BYTES
 
     bucket: List = []
     used = 0

@@ -20,7 +20,20 @@ from mgpu import (  # noqa: E402,F401
     visible_gpu_count,
 )
 
+from code_interpreter_amd.parallel.topology import (  # noqa: E402,F401
+    XGMI_LINK_GBPS,
+    XGMI_LINKS_PER_GPU,
+    gpu_inventory,
+    recommended_bucket_bytes,
+    ring_allreduce_seconds,
+)
+
 __all__ = [
+    "XGMI_LINK_GBPS",
+    "XGMI_LINKS_PER_GPU",
+    "gpu_inventory",
+    "recommended_bucket_bytes",
+    "ring_allreduce_seconds",
     "DEFAULT_BUCKET_BYTES",
     "allreduce_bucketed",
     "allreduce_matmul_bench",

@@ -159,7 +159,8 @@ def create_http_server(
     @app.post("/v1/execute", response_model=ExecuteResponse)
     async def execute(request: ExecuteRequest, request_id: str = Depends(set_request_id)):
         logger.debug(
-            "Executing code with files %s: %s", request.files, request.source_code
+            "execute: %d input file(s), %d source bytes",
+            len(request.files), len(request.source_code),
         )
         try:
             result = await code_executor.execute(
@@ -168,9 +169,10 @@ def create_http_server(
                 env=request.env,
             )
         except Exception as e:
-            logger.exception("Error executing code")
+            logger.exception("execute failed (request %s)", request_id)
             raise HTTPException(status_code=500, detail=str(e))
-        logger.debug("Code execution completed with result %s", result)
+        logger.debug("execute done: exit=%d, %d changed file(s)",
+                     result.exit_code, len(result.files))
         return ExecuteResponse(
             stdout=result.stdout,
             stderr=result.stderr,
@@ -182,7 +184,7 @@ def create_http_server(
     async def parse_custom_tool(
         request: ParseCustomToolRequest, request_id: str = Depends(set_request_id)
     ):
-        logger.info("Parsing custom tool")
+        logger.info("parse-custom-tool (request %s)", request_id)
         tool = custom_tool_executor.parse(tool_source_code=request.tool_source_code)
         return ParseCustomToolResponse(
             tool_name=tool.name,
@@ -192,7 +194,7 @@ def create_http_server(
 
     @app.exception_handler(CustomToolParseError)
     async def parse_error_handler(request, e: CustomToolParseError):
-        logger.warning("Invalid custom tool: %s", e.errors)
+        logger.warning("custom tool rejected by parser: %s", e.errors)
         return JSONResponse(
             status_code=status.HTTP_400_BAD_REQUEST,
             content=ParseCustomToolErrorResponse(error_messages=e.errors).model_dump(),
@@ -202,7 +204,7 @@ def create_http_server(
     async def execute_custom_tool(
         request: ExecuteCustomToolRequest, request_id: str = Depends(set_request_id)
     ):
-        logger.info("Executing custom tool")
+        logger.info("execute-custom-tool (request %s)", request_id)
         result = await custom_tool_executor.execute(
             tool_source_code=request.tool_source_code,
             tool_input_json=request.tool_input_json,
@@ -212,7 +214,7 @@ def create_http_server(
 
     @app.exception_handler(CustomToolExecuteError)
     async def execute_error_handler(request, e: CustomToolExecuteError):
-        logger.warning("Error executing custom tool: %s", e)
+        logger.warning("custom tool execution returned nonzero: %s", e)
         return JSONResponse(
             status_code=status.HTTP_400_BAD_REQUEST,
             content=ExecuteCustomToolErrorResponse(stderr=e.stderr).model_dump(),

@@ -100,6 +100,27 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def reduce_axis(self, h, dtype, outer, red, inner, mode):
+        self.calls.append("reduce_axis")
+        a = self.bufs[h].view(self._dt(dtype))[: outer * red * inner]
+        a = a.reshape(outer, red, inner)
+        if mode == 0:
+            out = a.sum(axis=1, dtype=np.float64)
+        elif mode == 1:
+            out = np.square(a.astype(np.float64)).sum(axis=1)
+        elif mode == 2:
+            out = a.max(axis=1)
+        else:
+            out = a.min(axis=1)
+        return self._new(out.astype(self._dt(dtype)))
+
+    def gemm_batched(self, ha, hb, batch, m, n, k, dtype):
+        self.calls.append("gemm_batched")
+        dt = self._dt(dtype)
+        a = self.bufs[ha].view(dt).reshape(batch, m, k)
+        b = self.bufs[hb].view(dt).reshape(batch, k, n)
+        return self._new(np.matmul(a, b))
+
     def gemm(self, ha, hb, m, n, k, dtype):
         self.calls.append("gemm")
         dt = self._dt(dtype) if dtype in (0, 1) else None
@@ -282,3 +303,52 @@ def test_ufunc_out_self_inplace(fake):
     x = _device(fake, host.copy())
     np.multiply(x, 2.0, out=x)
     np.testing.assert_allclose(np.asarray(x), host * 2.0)
+
+
+def test_axis_reductions_stay_on_device(fake):
+    host = np.random.default_rng(11).random((6, 8, 10))
+    x = _device(fake, host)
+    for axis in (0, 1, 2, -1):
+        r = x.sum(axis=axis)
+        assert isinstance(r, hipnp.DeviceArray)
+        np.testing.assert_allclose(np.asarray(r), host.sum(axis=axis), rtol=1e-12)
+        np.testing.assert_allclose(
+            np.asarray(np.max(x, axis=axis)), host.max(axis=axis)
+        )
+        np.testing.assert_allclose(
+            np.asarray(np.min(x, axis=axis)), host.min(axis=axis)
+        )
+    np.testing.assert_allclose(
+        np.asarray(x.mean(axis=1)), host.mean(axis=1), rtol=1e-12
+    )
+    assert "reduce_axis" in fake.calls
+
+
+def test_axis_reduction_keepdims_and_np_sum(fake):
+    host = np.random.default_rng(12).random((4, 5))
+    x = _device(fake, host)
+    r = np.sum(x, axis=0, keepdims=True)
+    assert isinstance(r, hipnp.DeviceArray)
+    assert r.shape == (1, 5)
+    np.testing.assert_allclose(np.asarray(r), host.sum(axis=0, keepdims=True))
+    r2 = np.add.reduce(x, axis=1)
+    np.testing.assert_allclose(np.asarray(r2), host.sum(axis=1))
+
+
+def test_axis_reduction_multi_axis_falls_back(fake):
+    host = np.random.default_rng(13).random((3, 4, 5))
+    x = _device(fake, host)
+    r = x.sum(axis=(0, 2))  # two axes: host fallback, same values
+    assert isinstance(r, np.ndarray)
+    np.testing.assert_allclose(r, host.sum(axis=(0, 2)))
+
+
+def test_batched_matmul_on_device(fake):
+    rng = np.random.default_rng(14)
+    a = rng.random((4, 8, 6))
+    b = rng.random((4, 6, 9))
+    r = hipnp.matmul(_device(fake, a), _device(fake, b), _force=True)
+    assert isinstance(r, hipnp.DeviceArray)
+    assert r.shape == (4, 8, 9)
+    np.testing.assert_allclose(np.asarray(r), np.matmul(a, b), rtol=1e-12)
+    assert "gemm_batched" in fake.calls

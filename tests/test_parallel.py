@@ -89,3 +89,33 @@ def test_custom_tool_multi_gpu_job(gpu_executor):
     )
     assert result["device"] == "cuda"
     assert result["matmul_tflops_per_gpu"] > 10
+
+
+def test_topology_bucket_sizing():
+    from code_interpreter_amd.parallel.topology import (
+        DEFAULT_BUCKET_BYTES,
+        MIN_BUCKET_BYTES,
+        recommended_bucket_bytes,
+        ring_allreduce_seconds,
+    )
+
+    # big payloads: the 64 MB default
+    assert recommended_bucket_bytes(1 << 30, 8) == DEFAULT_BUCKET_BYTES
+    # small payloads: shrunk so >= 4 buckets pipeline, floored
+    assert recommended_bucket_bytes(32 << 20, 8) == 8 << 20
+    assert recommended_bucket_bytes(1 << 20, 8) == MIN_BUCKET_BYTES
+    # single rank: no collective, size is moot
+    assert recommended_bucket_bytes(1 << 30, 1) == DEFAULT_BUCKET_BYTES
+
+    # ring model: world=1 free; time scales with 2(N-1)/N
+    assert ring_allreduce_seconds(1 << 30, 1) == 0.0
+    t2 = ring_allreduce_seconds(1 << 30, 2)
+    t8 = ring_allreduce_seconds(1 << 30, 8)
+    assert t8 / t2 == pytest.approx((2 * 7 / 8) / (2 * 1 / 2))
+
+
+def test_gpu_inventory_cpu_safe():
+    from code_interpreter_amd.parallel.topology import gpu_inventory
+
+    inv = gpu_inventory()  # no GPU here: must be an empty list, no raise
+    assert isinstance(inv, list)
