@@ -8,7 +8,7 @@ hipnp.py covers numpy; this module covers torch:
 
 - a global ``TorchFunctionMode`` intercepts the matmul family
   (``torch.matmul``, ``torch.mm``, ``a @ b``) and runs eligible 2-D
-  GEMMs through ``_hipops.gemm_raw`` -- the same MFMA kernels rocprof
+  GEMMs through ``_hipgemm.gemm_raw`` -- the same MFMA kernels rocprof
   shows on the numpy path (gemm_bf16_256b / gemm_f32 / gemm_f64) --
   ON TORCH'S OWN current HIP stream and memory (``data_ptr``), so
   stream ordering and the caching allocator behave exactly as for a
@@ -31,7 +31,13 @@ _OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 if _OPS_DIR not in sys.path:
     sys.path.insert(0, _OPS_DIR)
 
-import _hipops
+# _hipgemm, NOT _hipops: the zygote pre-imports _hipops before forking
+# sandbox children, and a HIP fat-binary registered pre-fork cannot be
+# launched once the child re-initializes the runtime (torch CUDA init) --
+# it segfaults (isolated by scripts/torch_case.py --steps). _hipgemm is
+# never imported by the zygote, so each torch-using child dlopens it
+# fresh and registration happens against the child's live runtime.
+import _hipgemm
 
 # route only when the kernel launch is worth more than its overhead;
 # env-tunable like hipnp's thresholds
@@ -43,7 +49,7 @@ _state = {"mode_obj": None, "torch": None}
 
 
 def available() -> bool:
-    return _hipops.is_available()
+    return _hipgemm.is_available()
 
 
 def _dtype_code(torch, dtype):
@@ -83,7 +89,7 @@ def _try_mm(torch, a, b):
     c = torch.empty((m, n), dtype=a.dtype, device=a.device)
     bt_ptr = 0
     bt = None
-    if dt == 2 and _hipops.gemm_bf16_256_ok(m, n, k):
+    if dt == 2 and _hipgemm.gemm_bf16_256_ok(m, n, k):
         # scratch for the B pre-transpose (the 256^2-tile kernel consumes
         # both operands K-contiguous); torch-allocated on the same stream,
         # so the caching allocator's stream ordering keeps it live until
@@ -91,7 +97,7 @@ def _try_mm(torch, a, b):
         bt = torch.empty((n, k), dtype=a.dtype, device=a.device)
         bt_ptr = bt.data_ptr()
     stream = torch.cuda.current_stream(a.device).cuda_stream
-    _hipops.gemm_raw(
+    _hipgemm.gemm_raw(
         a.data_ptr(), b.data_ptr(), c.data_ptr(), bt_ptr, m, n, k, dt, stream
     )
     del bt
@@ -141,7 +147,7 @@ def install(mode: str = "auto") -> bool:
         return False
     import torch
 
-    if not (torch.cuda.is_available() and _hipops.is_available()):
+    if not (torch.cuda.is_available() and _hipgemm.is_available()):
         if mode == "require":
             raise RuntimeError("APP_HIP_TORCH=require but no AMD GPU is visible")
         return False
