@@ -182,12 +182,23 @@ def _maybe_install_hip_torch() -> None:
 
 
 def preload() -> None:
-    """Called in the ZYGOTE before any fork: import everything heavy so
-    children get it copy-on-write. Importing hipnp/_hipops only LOADS the
-    HIP userspace stack (large .so relocations, ~0.3 CPU-s) -- it creates
-    no device context, so it is fork-safe; paying it once in the zygote
-    instead of per child keeps bursts under container CPU quotas (a 16-CPU
-    cgroup quota turns per-child reloads into 100 ms CFS throttle stalls).
+    """Called in the ZYGOTE before any fork: import everything heavy
+    (numpy, the hooks) so children get it copy-on-write.
+
+    The HIP stack is deliberately NOT touched here -- not as an
+    extension import and not as a bare library load. Measured facts
+    (r01 + scripts/torch_case.py isolation this round):
+    - r01 preloaded _hipops in the zygote to avoid per-child ~0.3-CPU-s
+      HIP userspace reloads bursting the container's CFS quota;
+    - but ANY libamdhip64 instance loaded pre-fork cannot launch kernels
+      post-fork (segfault at first launch). torch survives because it
+      always loads its own bundled runtime post-fork; our extensions
+      bind by soname to the pre-fork instance and die.
+    The resolution is in ops/hipnp.py (_load_hipops): daemon-RPC
+    children never import _hipops at all (no mapping cost, beating the
+    r01 COW trick), and processes that really launch local kernels
+    (LocalBackend, hiptorch's _hipgemm) load the runtime fresh in
+    themselves, post-fork.
     """
     install_import_hooks()
     try:
@@ -197,11 +208,6 @@ def preload() -> None:
     ops_dir = os.environ.get("APP_OPS_DIR")
     if ops_dir and ops_dir not in sys.path:
         sys.path.insert(0, ops_dir)
-    if os.environ.get("APP_HIP_NUMPY", "auto").lower() != "off":
-        try:
-            import hipnp  # noqa: F401  (no HIP context is created here)
-        except Exception:
-            pass
 
 
 def prewarm() -> None:

@@ -34,8 +34,6 @@ _OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 if _OPS_DIR not in sys.path:
     sys.path.insert(0, _OPS_DIR)
 
-import _hipops
-
 import numpy as _np
 
 import json as _json
@@ -44,16 +42,35 @@ import struct as _struct
 import threading as _threading
 
 
+def _load_hipops():
+    """Import _hipops ON DEMAND, never at module import.
+
+    Sandbox children that talk to the GPU daemon (the serving hot path)
+    need no local HIP at all; importing _hipops would map the ~0.3-CPU-s
+    HIP userspace stack per forked child (the r01 CFS-throttle cliff)
+    AND bind our kernels to whichever libamdhip64.so.7 instance is
+    already loaded -- if that instance was loaded before the fork, kernel
+    launches through it segfault (isolated by scripts/torch_case.py:
+    torch, which always uses its own bundled runtime loaded post-fork,
+    is unaffected; our kernels through a pre-fork-loaded runtime die).
+    Loading lazily in exactly the process that launches kernels keeps
+    every binding post-fork and the hot path mapping-free."""
+    import _hipops as m
+
+    return m
+
+
 class LocalBackend:
     """Direct _hipops calls: this process owns the HIP context."""
 
     name = "local"
 
     def __init__(self):
-        _hipops.init(0)
+        self._m = _load_hipops()
+        self._m.init(0)
 
     def __getattr__(self, item):
-        return getattr(_hipops, item)
+        return getattr(self._m, item)
 
 
 class GpuBackendLost(RuntimeError):
@@ -328,8 +345,28 @@ def daemon_socket():
     return None
 
 
+def daemon_configured() -> bool:
+    """A GPU daemon is advertised for this engine (the socket itself may
+    still be coming up -- _ensure_ready waits for it)."""
+    return bool(os.environ.get("APP_GPU_SERVICE"))
+
+
+def _gpu_present() -> bool:
+    try:
+        return _load_hipops().is_available()
+    except ImportError:
+        return False
+
+
 def available() -> bool:
-    return daemon_socket() is not None or _hipops.is_available()
+    # a CONFIGURED daemon answers availability without touching _hipops:
+    # this question is asked in the ZYGOTE (numpy import hook fires during
+    # preload), and probing _hipops there would load the HIP runtime
+    # pre-fork -- poisoning kernel launches in every forked child (see
+    # _load_hipops). Only daemon-less processes probe the local runtime.
+    if daemon_configured():
+        return True
+    return _gpu_present()
 
 
 def backend():
@@ -359,7 +396,7 @@ def _ensure_ready() -> None:
                         return
                     except (OSError, RuntimeError):
                         pass
-                if time.monotonic() >= deadline or not _hipops.is_available():
+                if time.monotonic() >= deadline or not _gpu_present():
                     break
                 time.sleep(0.05)
         _state["backend"] = LocalBackend()

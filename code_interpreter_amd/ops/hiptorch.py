@@ -31,13 +31,27 @@ _OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 if _OPS_DIR not in sys.path:
     sys.path.insert(0, _OPS_DIR)
 
-# _hipgemm, NOT _hipops: the zygote pre-imports _hipops before forking
-# sandbox children, and a HIP fat-binary registered pre-fork cannot be
-# launched once the child re-initializes the runtime (torch CUDA init) --
-# it segfaults (isolated by scripts/torch_case.py --steps). _hipgemm is
-# never imported by the zygote, so each torch-using child dlopens it
-# fresh and registration happens against the child's live runtime.
-import _hipgemm
+# _hipgemm (the GEMM-kernel extension) is imported LAZILY, inside
+# install(), strictly AFTER `import torch`: torch ships its own
+# libamdhip64 (SONAME libamdhip64.so.7) and the dynamic loader binds
+# _hipgemm's soname dependency to the FIRST matching instance -- with
+# torch loaded first that is torch's own runtime, so our kernels
+# register with and launch on exactly the runtime whose streams torch
+# hands us. Binding to any other instance is fragile (and an instance
+# loaded before the zygote fork segfaults at launch -- isolated by
+# scripts/torch_case.py --steps). The zygote never imports this module.
+_hipgemm = None
+
+
+def _ensure_hipgemm():
+    global _hipgemm
+    if _hipgemm is None:
+        import torch  # noqa: F401  (load torch's HIP runtime first)
+
+        import _hipgemm as m
+
+        _hipgemm = m
+    return _hipgemm
 
 # route only when the kernel launch is worth more than its overhead;
 # env-tunable like hipnp's thresholds
@@ -49,7 +63,10 @@ _state = {"mode_obj": None, "torch": None}
 
 
 def available() -> bool:
-    return _hipgemm.is_available()
+    try:
+        return _ensure_hipgemm().is_available()
+    except ImportError:
+        return False
 
 
 def _dtype_code(torch, dtype):
@@ -147,7 +164,7 @@ def install(mode: str = "auto") -> bool:
         return False
     import torch
 
-    if not (torch.cuda.is_available() and _hipgemm.is_available()):
+    if not (torch.cuda.is_available() and _ensure_hipgemm().is_available()):
         if mode == "require":
             raise RuntimeError("APP_HIP_TORCH=require but no AMD GPU is visible")
         return False

@@ -21,20 +21,22 @@ print('ok', float(c.float().abs().sum()))
 
 # step-by-step variant: which call inside the routed path crashes the
 # forked child? (route must be "off" so the import hook stays out)
-SRC_STEPS = """import os, sys
+SRC_STEPS = """import faulthandler, sys
+faulthandler.enable()
+import os
 import torch
 print('t1 torch imported', flush=True)
 sys.path.insert(0, os.environ['APP_OPS_DIR'])
-import _hipops
-print('t2 _hipops imported', flush=True)
-print('t3 is_available', _hipops.is_available(), flush=True)
+import _hipgemm
+print('t2 _hipgemm imported', flush=True)
+print('t3 is_available', _hipgemm.is_available(), flush=True)
 x = torch.randn(1024, 1024, device='cuda', dtype=torch.bfloat16)
 torch.cuda.synchronize()
 print('t4 torch cuda works', flush=True)
-print('t5 256ok', _hipops.gemm_bf16_256_ok(1024, 1024, 1024), flush=True)
+print('t5 256ok', _hipgemm.gemm_bf16_256_ok(1024, 1024, 1024), flush=True)
 c = torch.empty((1024, 1024), dtype=torch.bfloat16, device='cuda')
 stream = torch.cuda.current_stream().cuda_stream
-_hipops.gemm_raw(x.data_ptr(), x.data_ptr(), c.data_ptr(), 0, 1024, 1024, 1024, 2, stream)
+_hipgemm.gemm_raw(x.data_ptr(), x.data_ptr(), c.data_ptr(), 0, 1024, 1024, 1024, 2, stream)
 torch.cuda.synchronize()
 print('t6 gemm_raw done', float(c.float().abs().sum()), flush=True)
 import hiptorch
@@ -60,6 +62,7 @@ async def main():
         dep_install=False,
         execute_timeout=240.0,
         zygote_enabled=zygote,
+        hip_numpy="off" if "--nonp" in sys.argv else "auto",
     )
     print("step3 executor up", flush=True)
     src = SRC_STEPS if "--steps" in sys.argv else SRC
@@ -67,7 +70,7 @@ async def main():
     print(
         "step4 result exit=", r.exit_code,
         "stdout=", repr(r.stdout[:120]),
-        "stderr=", repr(r.stderr[:400]),
+        "stderr=", repr(r.stderr[-1500:]),
         flush=True,
     )
     await ex.aclose()

@@ -488,10 +488,11 @@ def hip_torch(hip):
 @pytest.mark.parametrize(
     "dtype,shape,tol",
     [
+        # shapes sit ABOVE MIN_MM_FLOPS (5e7) so routing must engage
         ("float32", (512, 384, 256), 2e-5),
-        ("float64", (256, 256, 192), 1e-12),
+        ("float64", (512, 384, 256), 1e-12),
         ("bfloat16", (512, 512, 256), 3e-2),   # 256-tile fast path
-        ("bfloat16", (300, 200, 96), 3e-2),    # general bf16 kernel
+        ("bfloat16", (520, 360, 192), 3e-2),   # general bf16 kernel
     ],
 )
 def test_torch_mm_routed_matches_torch(hip_torch, dtype, shape, tol):
