@@ -82,7 +82,14 @@ def _run_job_in_child(job: dict) -> None:
             sys.stderr.flush()
         except Exception:
             pass
-        os._exit(exit_code & 0xFF if exit_code >= 0 else 1)
+        code = exit_code & 0xFF if exit_code >= 0 else 1
+        if os.environ.get("APP_CHILD_CLEAN_EXIT") == "1":
+            # profiling aid: rocprofv3's preloaded tool writes this
+            # child's kernel CSV only from atexit handlers, which
+            # os._exit skips -- a clean interpreter exit keeps the
+            # sandbox's kernel table visible to the profiler
+            sys.exit(code)
+        os._exit(code)
 
 
 class WarmChild:

@@ -505,9 +505,9 @@ def test_torch_mm_routed_matches_torch(hip_torch, dtype, shape, tol):
     before = hiptorch.STATS["mm_routed"]
     c = a @ b
     assert hiptorch.STATS["mm_routed"] == before + 1, "matmul not routed"
-    # plain-torch fp32 CPU reference (same inputs, library-free path)
-    ref = a.float().cpu() @ b.float().cpu()
-    err = (c.float().cpu() - ref).abs().max().item()
+    # plain-torch fp64 CPU reference (same inputs, library-free path)
+    ref = a.double().cpu() @ b.double().cpu()
+    err = (c.double().cpu() - ref).abs().max().item()
     scale = ref.abs().max().item() + 1e-9
     assert err / scale < tol, (err, scale)
 
