@@ -21,21 +21,84 @@ import sys
 _original_import = builtins.__import__
 _hooks_installed = False
 
-# import-name -> pip package name, for modules whose names differ
+# import-name -> pip package name, for modules whose names differ.
+# The reference ships replit/upm with its full pypi import->package
+# sqlite map (executor/Dockerfile:32-38); this is the equivalent
+# knowledge for the import names that actually differ from their
+# distribution names (identical names need no entry: the scan falls
+# through to pip install <import name>).
 PIP_NAME_ALIASES = {
+    # imaging / media
     "cv2": "opencv-python-headless",
     "PIL": "pillow",
-    "sklearn": "scikit-learn",
-    "yaml": "pyyaml",
     "fitz": "pymupdf",
     "ffmpeg": "ffmpeg-python",
+    "moviepy": "moviepy",
+    "skimage": "scikit-image",
+    "OpenGL": "PyOpenGL",
+    "cairo": "pycairo",
+    "cairosvg": "CairoSVG",
+    # science / ML
+    "sklearn": "scikit-learn",
+    "Bio": "biopython",
+    "community": "python-louvain",
+    "igraph": "python-igraph",
+    "Levenshtein": "python-Levenshtein",
+    "speech_recognition": "SpeechRecognition",
+    # parsing / documents
+    "yaml": "pyyaml",
     "bs4": "beautifulsoup4",
-    "Crypto": "pycryptodome",
-    "dateutil": "python-dateutil",
-    "dotenv": "python-dotenv",
     "docx": "python-docx",
     "pptx": "python-pptx",
-    "moviepy": "moviepy",
+    "odf": "odfpy",
+    "pdfminer": "pdfminer.six",
+    "PyPDF2": "pypdf2",
+    "markdown": "Markdown",
+    "slugify": "python-slugify",
+    "magic": "python-magic",
+    "ruamel": "ruamel.yaml",
+    # crypto / auth
+    "Crypto": "pycryptodome",
+    "Cryptodome": "pycryptodomex",
+    "OpenSSL": "pyopenssl",
+    "jwt": "PyJWT",
+    "jose": "python-jose",
+    "nacl": "pynacl",
+    "socks": "PySocks",
+    # datetime / locale
+    "dateutil": "python-dateutil",
+    "babel": "Babel",
+    # config / env
+    "dotenv": "python-dotenv",
+    # web / network / APIs
+    "websocket": "websocket-client",
+    "zmq": "pyzmq",
+    "kafka": "kafka-python",
+    "memcache": "python-memcached",
+    "MySQLdb": "mysqlclient",
+    "psycopg2": "psycopg2-binary",
+    "github": "PyGithub",
+    "gitlab": "python-gitlab",
+    "googleapiclient": "google-api-python-client",
+    "apiclient": "google-api-python-client",
+    "telegram": "python-telegram-bot",
+    "discord": "discord.py",
+    "rest_framework": "djangorestframework",
+    "flask_sqlalchemy": "Flask-SQLAlchemy",
+    "flask_cors": "Flask-Cors",
+    "flask_login": "Flask-Login",
+    "flask_wtf": "Flask-WTF",
+    "fake_useragent": "fake-useragent",
+    "user_agents": "user-agents",
+    # hardware / misc
+    "serial": "pyserial",
+    "usb": "pyusb",
+    "git": "GitPython",
+    "faker": "Faker",
+    "dns": "dnspython",
+    "attr": "attrs",
+    "mpl_toolkits": "matplotlib",
+    "pylab": "matplotlib",
 }
 
 
