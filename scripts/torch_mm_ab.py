@@ -35,7 +35,11 @@ def main():
     size = int(sys.argv[1]) if len(sys.argv) > 1 else 8192
     assert torch.cuda.is_available()
     dev = "cuda"
-    for dtype, name in ((torch.bfloat16, "bf16"), (torch.float32, "f32")):
+    for dtype, name in (
+        (torch.bfloat16, "bf16"),
+        (torch.float32, "f32"),
+        (torch.float64, "f64"),
+    ):
         a = torch.randn(size, size, dtype=dtype, device=dev)
         b = torch.randn(size, size, dtype=dtype, device=dev)
         flops = 2 * size**3
