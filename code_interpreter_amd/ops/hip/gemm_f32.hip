@@ -169,13 +169,166 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// BK=16 variant: half the LDS (33 KB/block) -> 4 blocks/CU -> 4 waves
+// per SIMD. The MFMA-rate probe (scripts/mfma_rate_probe.cpp) measures
+// the f32 pipe ceiling 156 TF reachable from 2 waves/SIMD; the BK=32
+// kernel at exactly 2 waves/SIMD leaves no thread-level slack to cover
+// barrier/staging bubbles -- this variant trades 2x barrier frequency
+// for 2x the waves covering them.
+// ---------------------------------------------------------------------------
+constexpr int BK16 = 16;
+
+__global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  int nwg = tiles_m * tiles_n;
+  int wgid = blockIdx.x;
+  {
+    const int nxcd = 8;
+    int q = nwg / nxcd, r = nwg % nxcd;
+    int xcd = wgid % nxcd, idx = wgid / nxcd;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = wgid / tiles_n;
+  int tile_n = wgid % tiles_n;
+  int row0 = tile_m * BM;
+  int col0 = tile_n * BN;
+
+  __shared__ float As[2][BK16][BM + 1];
+  __shared__ float Bs[2][BK16][BN];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = wave >> 1;
+  const int wave_n = wave & 1;
+  const int l31 = lane & 31;
+  const int lk = lane >> 5;
+
+  f32x16 acc[2][2] = {};
+
+  // staging: BM*BK16/THREADS = 8 f32 of A (2 float4), BK16*BN/THREADS =
+  // 8 f32 of B (2 float4) per thread
+  const int a_m = tid >> 2;         // 64 rows per half
+  const int a_k = (tid & 3) * 4;
+  const int b_k = tid >> 5;         // 8 k-rows per half
+  const int b_n = (tid & 31) * 4;
+
+  float4 a_reg[2], b_reg[2];
+
+  auto issue_loads = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      int gr = row0 + a_m + i * 64;
+      int gk = k0 + a_k;
+      if (gr < M && gk + 3 < K) {
+        a_reg[i] = *reinterpret_cast<const float4*>(&A[(int64_t)gr * K + gk]);
+      } else {
+        float v[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          v[j] = (gr < M && gk + j < K) ? A[(int64_t)gr * K + gk + j] : 0.0f;
+        a_reg[i] = {v[0], v[1], v[2], v[3]};
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      int gk = k0 + b_k + i * 8;
+      int gn = col0 + b_n;
+      if (gk < K && gn + 3 < N) {
+        b_reg[i] = *reinterpret_cast<const float4*>(&B[(int64_t)gk * N + gn]);
+      } else {
+        float v[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          v[j] = (gk < K && gn + j < N) ? B[(int64_t)gk * N + gn + j] : 0.0f;
+        b_reg[i] = {v[0], v[1], v[2], v[3]};
+      }
+    }
+  };
+
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      As[buf][a_k + 0][a_m + i * 64] = a_reg[i].x;
+      As[buf][a_k + 1][a_m + i * 64] = a_reg[i].y;
+      As[buf][a_k + 2][a_m + i * 64] = a_reg[i].z;
+      As[buf][a_k + 3][a_m + i * 64] = a_reg[i].w;
+    }
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      *reinterpret_cast<float4*>(&Bs[buf][b_k + i * 8][b_n]) = b_reg[i];
+    }
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 64;
+
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < BK16; ks += 2) {
+      float a0 = As[buf][ks + lk][am0 + l31];
+      float a1 = As[buf][ks + lk][am0 + 32 + l31];
+      float b0 = Bs[buf][ks + lk][bn0 + l31];
+      float b1 = Bs[buf][ks + lk][bn0 + 32 + l31];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+  };
+
+  issue_loads(0);
+  write_lds(0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK16) {
+    bool have_next = k0 + BK16 < K;
+    if (have_next) issue_loads(k0 + BK16);
+    compute_tile(cur);
+    if (have_next) {
+      write_lds(cur ^ 1);
+      cur ^= 1;
+    }
+    __syncthreads();
+  }
+
+  const int crow0 = row0 + wave_m * 64 + 4 * (lane >> 5);
+  const int ccol0 = col0 + wave_n * 64 + l31;
+#pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 2; nt++) {
+      int col = ccol0 + nt * 32;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        int row = crow0 + mt * 32 + (reg & 3) + 8 * (reg >> 2);
+        if (row < M) C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,
                      int k, hipStream_t stream) {
   int tiles_m = (m + BM - 1) / BM;
   int tiles_n = (n + BN - 1) / BN;
-  hipLaunchKernelGGL(gemm_f32_kernel, dim3(tiles_m * tiles_n), dim3(THREADS),
-                     0, stream, a, b, c, m, n, k, tiles_m, tiles_n);
+  // default: the BK=16 high-occupancy kernel (124.8 TF @8192 vs 108 for
+  // BK=32 -- same-box A/B, profiles/NOTES.md r02); APP_F32_VARIANT=k32
+  // keeps the old kernel for comparison
+  const char* v = getenv("APP_F32_VARIANT");
+  if (v && v[0] == 'k') {
+    hipLaunchKernelGGL(gemm_f32_kernel, dim3(tiles_m * tiles_n), dim3(THREADS),
+                       0, stream, a, b, c, m, n, k, tiles_m, tiles_n);
+  } else {
+    hipLaunchKernelGGL(gemm_f32_bk16_kernel, dim3(tiles_m * tiles_n),
+                       dim3(THREADS), 0, stream, a, b, c, m, n, k, tiles_m,
+                       tiles_n);
+  }
   HIP_CHECK(hipGetLastError());
 }
