@@ -131,10 +131,159 @@ __global__ __launch_bounds__(THREADS) void gemm_f64_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 128x128-tile variant: the 64x64 kernel's arithmetic intensity is
+// 8 flop/B from HBM -> a 64 TF DRAM roofline, and it measures 57.8 TF =
+// 90% of that WRONG roof (rocBLAS reaches 72). Doubling the tile
+// doubles intensity (16 flop/B -> 128 TF cap), moving the bound back to
+// the MFMA pipe. 512 threads = 8 waves as 2(M)x4(N), each wave a 64x32
+// C tile (acc[4][2]); BK=16, double-buffered, 68 KB LDS -> 2 blocks/CU
+// -> 4 waves/SIMD covering staging/barrier bubbles.
+// ---------------------------------------------------------------------------
+constexpr int BM2 = 128;
+constexpr int BN2 = 128;
+constexpr int THREADS2 = 512;
+
+template <int BKT>
+__global__ __launch_bounds__(THREADS2) void gemm_f64_128_kernel(
+    const double* __restrict__ A, const double* __restrict__ B,
+    double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  int nwg = tiles_m * tiles_n;
+  int wgid = blockIdx.x;
+  {
+    const int nxcd = 8;
+    int q = nwg / nxcd, r = nwg % nxcd;
+    int xcd = wgid % nxcd, idx = wgid / nxcd;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int row0 = (wgid / tiles_n) * BM2;
+  int col0 = (wgid % tiles_n) * BN2;
+
+  __shared__ double As[2][BM2][BKT + 1];
+  __shared__ double Bs[2][BKT][BN2 + 1];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;        // 0..7
+  const int wave_m = wave >> 2;     // 0..1 -> 64-row half
+  const int wave_n = wave & 3;      // 0..3 -> 32-col band
+  const int l15 = lane & 15;
+  const int lk = lane >> 4;         // 0..3
+
+  f64x4 acc[4][2] = {};
+
+  // staging: AE = BM2*BKT/THREADS2 f64 of A, BE = BKT*BN2/THREADS2 of B
+  constexpr int AE = BM2 * BKT / THREADS2;
+  constexpr int BE = BKT * BN2 / THREADS2;
+  const int a_m = tid / (BKT / AE);
+  const int a_k = (tid % (BKT / AE)) * AE;
+  const int b_k = tid / (BN2 / BE);
+  const int b_n = (tid % (BN2 / BE)) * BE;
+
+  double a_reg[AE], b_reg[BE];
+
+  auto issue_loads = [&](int k0) {
+    int gr = row0 + a_m;
+#pragma unroll
+    for (int j = 0; j < AE; j++) {
+      int gk = k0 + a_k + j;
+      a_reg[j] = (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
+    }
+    int gk = k0 + b_k;
+#pragma unroll
+    for (int j = 0; j < BE; j++) {
+      int gn = col0 + b_n + j;
+      b_reg[j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < AE; j++) As[buf][a_m][a_k + j] = a_reg[j];
+#pragma unroll
+    for (int j = 0; j < BE; j++) Bs[buf][b_k][b_n + j] = b_reg[j];
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 32;
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < BKT; ks += 4) {
+      double a0 = As[buf][am0 + l15][ks + lk];
+      double a1 = As[buf][am0 + 16 + l15][ks + lk];
+      double a2 = As[buf][am0 + 32 + l15][ks + lk];
+      double a3 = As[buf][am0 + 48 + l15][ks + lk];
+      double b0 = Bs[buf][ks + lk][bn0 + l15];
+      double b1 = Bs[buf][ks + lk][bn0 + 16 + l15];
+      // no two consecutive MFMAs share a source operand: the rate probe
+      // suggests f64 MFMA stalls on back-to-back src reuse
+      acc[0][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b0, acc[0][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b0, acc[2][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b1, acc[3][1], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b0, acc[1][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b0, acc[3][0], 0, 0, 0);
+    }
+  };
+
+  issue_loads(0);
+  write_lds(0);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BKT) {
+    bool have_next = k0 + BKT < K;
+    if (have_next) issue_loads(k0 + BKT);
+    compute_tile(cur);
+    if (have_next) {
+      write_lds(cur ^ 1);
+      cur ^= 1;
+    }
+    __syncthreads();
+  }
+
+  const int crow0 = row0 + wave_m * 64 + (lane >> 4);
+  const int ccol0 = col0 + wave_n * 32 + l15;
+#pragma unroll
+  for (int mt = 0; mt < 4; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 2; nt++) {
+      int col = ccol0 + nt * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int row = crow0 + mt * 16 + 4 * reg;
+        if (row < M) C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
                      int k, hipStream_t stream) {
+  // default: 128^2 tile, BK=16; APP_F64_VARIANT: "s" = the old 64^2
+  // kernel, "8"/"32" = BK sweep on the 128^2 tile (occupancy experiment)
+  const char* v = getenv("APP_F64_VARIANT");
+  if (!(v && v[0] == 's')) {
+    int tiles_m2 = (m + BM2 - 1) / BM2;
+    int tiles_n2 = (n + BN2 - 1) / BN2;
+    if (v && v[0] == '8')
+      hipLaunchKernelGGL((gemm_f64_128_kernel<8>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS2), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    else if (v && v[0] == '3')
+      hipLaunchKernelGGL((gemm_f64_128_kernel<32>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS2), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    else
+      hipLaunchKernelGGL((gemm_f64_128_kernel<16>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS2), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    HIP_CHECK(hipGetLastError());
+    return;
+  }
   int tiles_m = (m + BM - 1) / BM;
   int tiles_n = (n + BN - 1) / BN;
   hipLaunchKernelGGL(gemm_f64_kernel, dim3(tiles_m * tiles_n), dim3(THREADS),
