@@ -87,6 +87,27 @@ PyObject* py_gemm_raw(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// gemm_raw_nt(pa, pw, pc, m, n, k, stream): bf16 C[M,N] = A[M,K] @ W^T
+// where W is [N][K] row-major -- torch's nn.Linear weight layout IS the
+// 256-tile kernel's pre-transposed B operand, so linear routes with
+// ZERO transpose work. 256-supported shapes only (caller checks).
+PyObject* py_gemm_raw_nt(PyObject*, PyObject* args) {
+  unsigned long long pa, pw, pc, stream;
+  int m, n, k;
+  if (!PyArg_ParseTuple(args, "KKKiiiK", &pa, &pw, &pc, &m, &n, &k, &stream))
+    return nullptr;
+  WRAP_BEGIN
+  if (!gemm_bf16_256_supported(m, n, k))
+    throw std::runtime_error("gemm_raw_nt: shape not 256-tile supported");
+  hipStream_t s = (hipStream_t)stream;
+  NOGIL_BEGIN
+  launch_gemm_bf16_256((const uint16_t*)pa, (const uint16_t*)pw,
+                       (uint16_t*)pc, m, n, k, s);
+  NOGIL_END
+  Py_RETURN_NONE;
+  WRAP_END
+}
+
 PyObject* py_gemm_bf16_256_ok(PyObject*, PyObject* args) {
   int m, n, k;
   if (!PyArg_ParseTuple(args, "iii", &m, &n, &k)) return nullptr;
@@ -98,6 +119,8 @@ PyMethodDef methods[] = {
     {"is_available", py_is_available, METH_NOARGS, "GPU present?"},
     {"gemm_raw", py_gemm_raw, METH_VARARGS,
      "gemm_raw(pa, pb, pc, pbt, m, n, k, dtype, stream)"},
+    {"gemm_raw_nt", py_gemm_raw_nt, METH_VARARGS,
+     "gemm_raw_nt(pa, pw, pc, m, n, k, stream): bf16 A @ W^T, W=[N][K]"},
     {"gemm_bf16_256_ok", py_gemm_bf16_256_ok, METH_VARARGS,
      "gemm_bf16_256_ok(m, n, k) -> bool"},
     {nullptr, nullptr, 0, nullptr},

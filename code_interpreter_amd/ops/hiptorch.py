@@ -57,7 +57,8 @@ def _ensure_hipgemm():
 # env-tunable like hipnp's thresholds
 MIN_MM_FLOPS = float(os.environ.get("APP_HIP_TORCH_MIN_FLOPS", 5e7))
 
-STATS = {"mm_routed": 0, "mm_fallback": 0}
+STATS = {"mm_routed": 0, "mm_fallback": 0, "linear_routed": 0,
+         "bmm_routed": 0}
 
 _state = {"mode_obj": None, "torch": None}
 
@@ -122,6 +123,84 @@ def _try_mm(torch, a, b):
     return c
 
 
+def _try_linear(torch, x, w, bias=None):
+    """Route F.linear: y = x @ w.T (+ bias). torch's weight layout
+    [out,in] row-major IS the 256-tile kernel's pre-transposed B
+    operand, so eligible bf16 linears run with zero transpose work."""
+    if not (isinstance(x, torch.Tensor) and isinstance(w, torch.Tensor)):
+        return None
+    if not (x.is_cuda and w.is_cuda) or x.dtype != w.dtype:
+        return None
+    if x.dtype != torch.bfloat16 or w.dim() != 2 or x.dim() < 2:
+        return None
+    if torch.is_grad_enabled() and (
+        x.requires_grad or w.requires_grad
+        or (bias is not None and bias.requires_grad)
+    ):
+        return None
+    n, k = w.shape
+    if x.shape[-1] != k:
+        return None
+    lead = x.shape[:-1]
+    m = 1
+    for d in lead:
+        m *= d
+    if 2.0 * m * n * k < MIN_MM_FLOPS:
+        return None
+    if not _hipgemm.gemm_bf16_256_ok(m, n, k):
+        return None
+    x2 = x.reshape(m, k).contiguous()
+    w = w.contiguous()
+    c = torch.empty((m, n), dtype=x.dtype, device=x.device)
+    stream = torch.cuda.current_stream(x.device).cuda_stream
+    _hipgemm.gemm_raw_nt(x2.data_ptr(), w.data_ptr(), c.data_ptr(), m, n, k,
+                         stream)
+    if bias is not None:
+        c = c + bias
+    STATS["linear_routed"] += 1
+    return c.reshape(*lead, n)
+
+
+def _try_bmm(torch, a, b):
+    """Route equal-batch 3-D matmuls as per-batch GEMM launches enqueued
+    back-to-back on torch's stream."""
+    if not (isinstance(a, torch.Tensor) and isinstance(b, torch.Tensor)):
+        return None
+    if not (a.is_cuda and b.is_cuda) or a.dtype != b.dtype:
+        return None
+    dt = _dtype_code(torch, a.dtype)
+    if dt is None or a.dim() != 3 or b.dim() != 3:
+        return None
+    if a.shape[0] != b.shape[0] or a.shape[2] != b.shape[1]:
+        return None
+    if torch.is_grad_enabled() and (a.requires_grad or b.requires_grad):
+        return None
+    batch, m, k = a.shape
+    n = b.shape[2]
+    if batch == 0 or 2.0 * m * n * k < MIN_MM_FLOPS:
+        return None
+    a = a.contiguous()
+    b = b.contiguous()
+    c = torch.empty((batch, m, n), dtype=a.dtype, device=a.device)
+    es = a.element_size()
+    stream = torch.cuda.current_stream(a.device).cuda_stream
+    for i in range(batch):
+        bt_ptr = 0
+        bt = None
+        if dt == 2 and _hipgemm.gemm_bf16_256_ok(m, n, k):
+            bt = torch.empty((n, k), dtype=a.dtype, device=a.device)
+            bt_ptr = bt.data_ptr()
+        _hipgemm.gemm_raw(
+            a.data_ptr() + i * m * k * es,
+            b.data_ptr() + i * k * n * es,
+            c.data_ptr() + i * m * n * es,
+            bt_ptr, m, n, k, dt, stream,
+        )
+        del bt
+    STATS["bmm_routed"] += 1
+    return c
+
+
 def _make_mode(torch):
     from torch.overrides import TorchFunctionMode
 
@@ -131,6 +210,16 @@ def _make_mode(torch):
         torch.Tensor.matmul,
         torch.Tensor.__matmul__,
     }
+    bmm_funcs = {torch.bmm, torch.Tensor.bmm}
+    linear_funcs = {torch.nn.functional.linear}
+
+    def _guarded(fn, *args):
+        try:
+            return fn(torch, *args)
+        except Exception:
+            if os.environ.get("APP_HIP_TORCH", "auto").lower() == "require":
+                raise
+            return None
 
     class HipMatmulMode(TorchFunctionMode):
         """Global interception of the matmul family; self-excluded while
@@ -140,15 +229,26 @@ def _make_mode(torch):
 
         def __torch_function__(self, func, types, args=(), kwargs=None):
             kwargs = kwargs or {}
+            r = None
             if func in mm_funcs and len(args) == 2 and not kwargs:
-                try:
-                    r = _try_mm(torch, args[0], args[1])
-                except Exception:
-                    if os.environ.get("APP_HIP_TORCH", "auto").lower() == "require":
-                        raise
-                    r = None
-                if r is not None:
-                    return r
+                a, b = args
+                if (
+                    isinstance(a, torch.Tensor) and isinstance(b, torch.Tensor)
+                    and a.dim() == 3 and b.dim() == 3
+                ):
+                    r = _guarded(_try_bmm, a, b)
+                else:
+                    r = _guarded(_try_mm, a, b)
+            elif func in bmm_funcs and len(args) == 2 and not kwargs:
+                r = _guarded(_try_bmm, args[0], args[1])
+            elif func in linear_funcs and 2 <= len(args) <= 3 and set(
+                kwargs
+            ) <= {"bias"}:
+                bias = args[2] if len(args) == 3 else kwargs.get("bias")
+                r = _guarded(_try_linear, args[0], args[1], bias)
+            if r is not None:
+                return r
+            if func in mm_funcs or func in bmm_funcs or func in linear_funcs:
                 STATS["mm_fallback"] += 1
             return func(*args, **kwargs)
 

@@ -100,3 +100,18 @@ def test_require_mode_surfaces_routing_errors(mode, monkeypatch):
     monkeypatch.setenv("APP_HIP_TORCH", "require")
     with pytest.raises(RuntimeError, match="kernel launch failed"):
         torch.matmul(a, a)
+
+
+def test_linear_and_bmm_eligibility_cpu():
+    # CPU tensors: never routed, plain fallbacks intact under the mode
+    m = hiptorch._make_mode(torch)
+    with m:
+        import torch.nn.functional as F
+
+        x = torch.randn(8, 16)
+        w = torch.randn(4, 16)
+        torch.testing.assert_close(F.linear(x, w), x @ w.T)
+        a = torch.randn(3, 8, 8)
+        torch.testing.assert_close(torch.bmm(a, a), torch.matmul(a, a))
+    assert hiptorch._try_linear(torch, x, w) is None
+    assert hiptorch._try_bmm(torch, a, a) is None

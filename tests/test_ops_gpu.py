@@ -608,3 +608,37 @@ def test_device_array_axis_reduction_no_host_roundtrip(hnp):
     total_via_axis = float(r.sum())
     total_direct = float(x.sum())
     assert abs(total_via_axis - total_direct) / abs(total_direct) < 1e-10
+
+
+def test_torch_linear_routed_zero_transpose(hip_torch):
+    torch, hiptorch = hip_torch
+    import torch.nn.functional as F
+
+    x = torch.randn(512, 1024, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(768, 1024, dtype=torch.bfloat16, device="cuda")
+    bias = torch.randn(768, dtype=torch.bfloat16, device="cuda")
+    before = hiptorch.STATS["linear_routed"]
+    y = F.linear(x, w, bias)
+    assert hiptorch.STATS["linear_routed"] == before + 1, "linear not routed"
+    ref = (x.double().cpu() @ w.double().cpu().T) + bias.double().cpu()
+    err = (y.double().cpu() - ref).abs().max().item()
+    assert err / (ref.abs().max().item() + 1e-9) < 3e-2
+    # 3-D input flattens through the same path
+    x3 = torch.randn(4, 128, 1024, dtype=torch.bfloat16, device="cuda")
+    y3 = F.linear(x3, w)
+    assert y3.shape == (4, 128, 768)
+    assert hiptorch.STATS["linear_routed"] == before + 2
+
+
+def test_torch_bmm_routed(hip_torch):
+    torch, hiptorch = hip_torch
+    a = torch.randn(6, 512, 256, dtype=torch.float32, device="cuda")
+    b = torch.randn(6, 256, 384, dtype=torch.float32, device="cuda")
+    before = hiptorch.STATS["bmm_routed"]
+    c1 = torch.bmm(a, b)
+    c2 = torch.matmul(a, b)
+    assert hiptorch.STATS["bmm_routed"] == before + 2, "bmm not routed"
+    ref = torch.matmul(a.double().cpu(), b.double().cpu())
+    for c in (c1, c2):
+        err = (c.double().cpu() - ref).abs().max().item()
+        assert err / (ref.abs().max().item() + 1e-9) < 2e-4
