@@ -71,6 +71,12 @@ class FakeHipops(types.ModuleType):
     def gemm(self, ha, hb, m, n, k, dtype):
         return self._new(b"\0" * (m * n * 8))
 
+    def reduce_axis(self, h, dtype, outer, red, inner, mode):
+        return self._new(b"\0" * (outer * inner * 8))
+
+    def gemm_batched(self, ha, hb, batch, m, n, k, dtype):
+        return self._new(b"\0" * (batch * m * n * 8))
+
     def synchronize(self):
         pass
 
@@ -123,6 +129,10 @@ def test_cross_connection_handle_access_denied(hipd):
         # every handle-consuming op must refuse a foreign handle
         denied = [
             {"op": "download", "h": h, "nbytes": 128},
+            {"op": "reduce_axis", "h": h, "dtype": 1, "outer": 2, "red": 4,
+             "inner": 2, "mode": 0},
+            {"op": "gemm_batched", "ha": h, "hb": h, "batch": 1, "m": 2,
+             "n": 2, "k": 2, "dtype": 1},
             {"op": "free", "h": h},
             {"op": "unary", "h": h, "uop": 0, "dtype": 1, "n": 16},
             {"op": "binary", "ha": h, "hb": h, "bop": 0, "dtype": 1, "n": 16},

@@ -230,3 +230,48 @@ def test_large_host_array_matmul_uses_shm_path(gpu_executor):
     r = _run(gpu_executor, code)
     assert r.exit_code == 0, r.stderr
     assert r.stdout.startswith("ok ")
+
+
+def test_axis_reductions_and_mutation_in_sandbox(gpu_executor):
+    """r02 surface through the FULL sandbox + daemon-RPC path: axis-wise
+    reductions stay device-resident and in-place mutation works."""
+    code = (
+        "import numpy, hipnp\n"
+        "x = numpy.random.rand(2000, 2000)\n"
+        "rows = x.sum(axis=1)\n"
+        "assert type(rows).__name__ == 'DeviceArray', type(rows)\n"
+        "cols = numpy.sum(x, axis=0)\n"
+        "total_r = float(rows.sum()); total_c = float(cols.sum())\n"
+        "assert abs(total_r - total_c) / total_r < 1e-10\n"
+        "m = x.max(axis=1)\n"
+        "assert float(m.min()) <= 1.0\n"
+        "x[0] = 0.5\n"
+        "assert abs(float(x.sum(axis=1)[0]) - 1000.0) < 1e-6\n"
+        "ops = hipnp.RPC_STATS.get('per_op', {})\n"
+        "assert 'reduce_axis' in ops, ops\n"
+        "print('axis-ok', round(total_r))\n"
+    )
+    r = _run(gpu_executor, code)
+    assert r.exit_code == 0, r.stderr
+    assert "axis-ok" in r.stdout
+
+
+def test_torch_routed_in_sandbox(gpu_executor):
+    """User `import torch; a @ b` in a sandbox runs on the hand-written
+    MFMA kernels (hiptorch mode installed by the import hook)."""
+    code = (
+        "import torch\n"
+        "import hiptorch\n"
+        "assert hiptorch._state['mode_obj'] is not None, 'mode not installed'\n"
+        "a = torch.randn(1024, 1024, device='cuda', dtype=torch.bfloat16)\n"
+        "c = a @ a\n"
+        "torch.cuda.synchronize()\n"
+        "assert hiptorch.STATS['mm_routed'] >= 1, hiptorch.STATS\n"
+        "ref = (a.float() @ a.float())\n"
+        "err = (c.float() - ref).abs().max() / ref.abs().max()\n"
+        "assert float(err) < 3e-2, float(err)\n"
+        "print('torch-ok', float(c.float().abs().sum()))\n"
+    )
+    r = _run(gpu_executor, code, env={"APP_HIP_TORCH": "require"})
+    assert r.exit_code == 0, r.stderr
+    assert "torch-ok" in r.stdout
