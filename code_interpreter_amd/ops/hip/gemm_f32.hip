@@ -177,8 +177,7 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_kernel(
 // barrier/staging bubbles -- this variant trades 2x barrier frequency
 // for 2x the waves covering them.
 // ---------------------------------------------------------------------------
-constexpr int BK16 = 16;
-
+template <int BKT>
 __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
@@ -195,8 +194,8 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
   int row0 = tile_m * BM;
   int col0 = tile_n * BN;
 
-  __shared__ float As[2][BK16][BM + 1];
-  __shared__ float Bs[2][BK16][BN];
+  __shared__ float As[2][BKT][BM + 1];
+  __shared__ float Bs[2][BKT][BN];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -208,19 +207,25 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
 
   f32x16 acc[2][2] = {};
 
-  // staging: BM*BK16/THREADS = 8 f32 of A (2 float4), BK16*BN/THREADS =
-  // 8 f32 of B (2 float4) per thread
-  const int a_m = tid >> 2;         // 64 rows per half
-  const int a_k = (tid & 3) * 4;
-  const int b_k = tid >> 5;         // 8 k-rows per half
-  const int b_n = (tid & 31) * 4;
+  // staging: AQ = BM*BKT/THREADS/4 float4 of A, BQ = BKT*BN/THREADS/4
+  // of B per thread (AQ=BQ=2 at BKT=16, 1 at BKT=8)
+  constexpr int AQ = BM * BKT / THREADS / 4;
+  constexpr int BQ = BKT * BN / THREADS / 4;
+  constexpr int KV = BKT / 4;            // float4 columns per A row
+  const int a_m = tid / KV;              // BM/AQ rows per chunk
+  const int a_k = (tid % KV) * 4;
+  constexpr int NV = BN / 4;             // float4 columns per B row
+  const int b_k = tid / NV;
+  const int b_n = (tid % NV) * 4;
+  constexpr int AROWS = THREADS / KV;    // rows covered per i-step
+  constexpr int BROWS = THREADS / NV;
 
-  float4 a_reg[2], b_reg[2];
+  float4 a_reg[AQ], b_reg[BQ];
 
   auto issue_loads = [&](int k0) {
 #pragma unroll
-    for (int i = 0; i < 2; i++) {
-      int gr = row0 + a_m + i * 64;
+    for (int i = 0; i < AQ; i++) {
+      int gr = row0 + a_m + i * AROWS;
       int gk = k0 + a_k;
       if (gr < M && gk + 3 < K) {
         a_reg[i] = *reinterpret_cast<const float4*>(&A[(int64_t)gr * K + gk]);
@@ -233,8 +238,8 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
       }
     }
 #pragma unroll
-    for (int i = 0; i < 2; i++) {
-      int gk = k0 + b_k + i * 8;
+    for (int i = 0; i < BQ; i++) {
+      int gk = k0 + b_k + i * BROWS;
       int gn = col0 + b_n;
       if (gk < K && gn + 3 < N) {
         b_reg[i] = *reinterpret_cast<const float4*>(&B[(int64_t)gk * N + gn]);
@@ -250,15 +255,15 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
 
   auto write_lds = [&](int buf) {
 #pragma unroll
-    for (int i = 0; i < 2; i++) {
-      As[buf][a_k + 0][a_m + i * 64] = a_reg[i].x;
-      As[buf][a_k + 1][a_m + i * 64] = a_reg[i].y;
-      As[buf][a_k + 2][a_m + i * 64] = a_reg[i].z;
-      As[buf][a_k + 3][a_m + i * 64] = a_reg[i].w;
+    for (int i = 0; i < AQ; i++) {
+      As[buf][a_k + 0][a_m + i * AROWS] = a_reg[i].x;
+      As[buf][a_k + 1][a_m + i * AROWS] = a_reg[i].y;
+      As[buf][a_k + 2][a_m + i * AROWS] = a_reg[i].z;
+      As[buf][a_k + 3][a_m + i * AROWS] = a_reg[i].w;
     }
 #pragma unroll
-    for (int i = 0; i < 2; i++) {
-      *reinterpret_cast<float4*>(&Bs[buf][b_k + i * 8][b_n]) = b_reg[i];
+    for (int i = 0; i < BQ; i++) {
+      *reinterpret_cast<float4*>(&Bs[buf][b_k + i * BROWS][b_n]) = b_reg[i];
     }
   };
 
@@ -267,7 +272,7 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
 
   auto compute_tile = [&](int buf) {
 #pragma unroll
-    for (int ks = 0; ks < BK16; ks += 2) {
+    for (int ks = 0; ks < BKT; ks += 2) {
       float a0 = As[buf][ks + lk][am0 + l31];
       float a1 = As[buf][ks + lk][am0 + 32 + l31];
       float b0 = Bs[buf][ks + lk][bn0 + l31];
@@ -284,9 +289,9 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
   __syncthreads();
 
   int cur = 0;
-  for (int k0 = 0; k0 < K; k0 += BK16) {
-    bool have_next = k0 + BK16 < K;
-    if (have_next) issue_loads(k0 + BK16);
+  for (int k0 = 0; k0 < K; k0 += BKT) {
+    bool have_next = k0 + BKT < K;
+    if (have_next) issue_loads(k0 + BKT);
     compute_tile(cur);
     if (have_next) {
       write_lds(cur ^ 1);
@@ -325,8 +330,12 @@ void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,
   if (v && v[0] == 'k') {
     hipLaunchKernelGGL(gemm_f32_kernel, dim3(tiles_m * tiles_n), dim3(THREADS),
                        0, stream, a, b, c, m, n, k, tiles_m, tiles_n);
+  } else if (v && v[0] == '8') {
+    hipLaunchKernelGGL((gemm_f32_bk16_kernel<8>), dim3(tiles_m * tiles_n),
+                       dim3(THREADS), 0, stream, a, b, c, m, n, k, tiles_m,
+                       tiles_n);
   } else {
-    hipLaunchKernelGGL(gemm_f32_bk16_kernel, dim3(tiles_m * tiles_n),
+    hipLaunchKernelGGL((gemm_f32_bk16_kernel<16>), dim3(tiles_m * tiles_n),
                        dim3(THREADS), 0, stream, a, b, c, m, n, k, tiles_m,
                        tiles_n);
   }
