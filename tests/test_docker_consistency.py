@@ -20,7 +20,10 @@ PIP_NAMES = {
     "grpc_reflection": "grpcio-reflection",
 }
 # imported lazily/optionally with a guarded fallback; not boot-critical
-OPTIONAL = {"mgpu"}
+# (torch: parallel/topology.gpu_inventory probes it inside try/except --
+# the CONTROL-PLANE image deliberately ships without the multi-GB torch
+# wheel; torch lives in the EXECUTOR image where user code runs)
+OPTIONAL = {"mgpu", "torch"}
 
 
 def control_plane_imports():
