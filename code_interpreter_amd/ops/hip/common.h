@@ -49,6 +49,10 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// argmax/argmin (numpy tie/NaN semantics): scratch >= 2048*(8+8) bytes,
+// result int64 index written to out_idx (device)
+void launch_argminmax(DType dt, bool maxop, const void* in, void* scratch,
+                      void* out_idx, int64_t n, hipStream_t stream);
 // philox4x32-10 uniform doubles/floats in [0, 1)
 void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
                          uint64_t offset, hipStream_t stream);

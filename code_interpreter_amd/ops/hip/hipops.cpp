@@ -559,6 +559,32 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// argminmax(h, dtype, n, maxop) -> int64 index (numpy tie/NaN semantics)
+PyObject* py_argminmax(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt, maxop;
+  long long n;
+  if (!PyArg_ParseTuple(args, "KiLi", &h, &dt, &n, &maxop)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (n <= 0) throw std::runtime_error("argminmax of empty array");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  long long result = 0;
+  NOGIL_BEGIN
+  ensure_reduce_scratch();
+  // reuse the pinned scalar as the int64 result landing slot
+  launch_argminmax(dtype, maxop != 0, in.ptr, g.reduce_scratch, g.scalar_dev,
+                   n, g.compute);
+  HIP_CHECK(hipMemcpyAsync(g.scalar_pin, g.scalar_dev, 8,
+                           hipMemcpyDeviceToHost, g.compute));
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  result = *(int64_t*)g.scalar_pin;
+  NOGIL_END
+  return PyLong_FromLongLong(result);
+  WRAP_END
+}
+
 // reduce_axis(h, dtype, outer, red, inner, mode) -> handle
 // contiguous [outer][red][inner] reduced over the middle axis
 PyObject* py_reduce_axis(PyObject*, PyObject* args) {
@@ -752,6 +778,8 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"argminmax", py_argminmax, METH_VARARGS,
+     "argminmax(h, dtype, n, maxop) -> int64 index"},
     {"reduce_axis", py_reduce_axis, METH_VARARGS,
      "reduce_axis(h, dtype, outer, red, inner, mode) -> handle"},
     {"gemm_batched", py_gemm_batched, METH_VARARGS,

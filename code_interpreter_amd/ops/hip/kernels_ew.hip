@@ -262,6 +262,89 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// argmax/argmin: two-stage (value, index) reduction. numpy semantics:
+// FIRST occurrence wins ties; NaN wins (propagates) like np.argmax.
+// ---------------------------------------------------------------------------
+template <bool MAXOP>
+__device__ __forceinline__ bool arg_better(double v, int64_t i, double bv,
+                                           int64_t bi) {
+  // NaN beats everything (numpy returns the first NaN's index);
+  // among equals the SMALLER index wins
+  bool v_nan = v != v, b_nan = bv != bv;
+  if (v_nan != b_nan) return v_nan;
+  if (v_nan && b_nan) return i < bi;
+  if (v != bv) return MAXOP ? v > bv : v < bv;
+  return i < bi;
+}
+
+template <typename T, bool MAXOP>
+__global__ void argminmax_stage1(const T* __restrict__ in,
+                                 double* __restrict__ pvals,
+                                 int64_t* __restrict__ pidx, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  double best = MAXOP ? -INFINITY : INFINITY;
+  int64_t besti = 0;
+  bool any = false;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    double v = (double)in[i];
+    if (!any || arg_better<MAXOP>(v, i, best, besti)) {
+      best = v;
+      besti = i;
+      any = true;
+    }
+  }
+  if (!any) besti = n;  // lane saw nothing: neutral (index n loses ties)
+  // wave64 butterfly on (value, index) pairs
+  for (int off = 32; off > 0; off >>= 1) {
+    double ov = __shfl_down(best, off, 64);
+    int64_t oi = __shfl_down(besti, off, 64);
+    if (oi < n && (besti >= n || arg_better<MAXOP>(ov, oi, best, besti))) {
+      best = ov;
+      besti = oi;
+    }
+  }
+  __shared__ double wv[kBlock / 64];
+  __shared__ int64_t wi[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) {
+    wv[wave] = best;
+    wi[wave] = besti;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < (int)blockDim.x / 64; w++) {
+      if (wi[w] < n &&
+          (wi[0] >= n || arg_better<MAXOP>(wv[w], wi[w], wv[0], wi[0]))) {
+        wv[0] = wv[w];
+        wi[0] = wi[w];
+      }
+    }
+    pvals[blockIdx.x] = wv[0];
+    pidx[blockIdx.x] = wi[0];
+  }
+}
+
+template <bool MAXOP>
+__global__ void argminmax_stage2(const double* __restrict__ pvals,
+                                 const int64_t* __restrict__ pidx,
+                                 int64_t* __restrict__ out, int nparts,
+                                 int64_t n) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    double best = pvals[0];
+    int64_t besti = pidx[0];
+    for (int i = 1; i < nparts; i++) {
+      if (pidx[i] < n &&
+          (besti >= n || arg_better<MAXOP>(pvals[i], pidx[i], best, besti))) {
+        best = pvals[i];
+        besti = pidx[i];
+      }
+    }
+    out[0] = besti;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // axis-wise reduction: in viewed as [outer][red][inner] -> out[outer][inner]
 // (covers any single-axis reduce of a contiguous N-D array). Two shapes:
 //  - inner > 1: one thread per (outer, inner) output element; lanes read
@@ -552,6 +635,35 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
   HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_argminmax_t(bool maxop, const T* in, void* scratch,
+                               void* out_idx, int64_t n, hipStream_t s) {
+  // scratch layout: [kBlock doubles][kBlock int64] partials
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  double* pvals = (double*)scratch;
+  int64_t* pidx = (int64_t*)((char*)scratch + 2048 * sizeof(double));
+  if (maxop) {
+    hipLaunchKernelGGL((argminmax_stage1<T, true>), dim3(grid), dim3(kBlock),
+                       0, s, in, pvals, pidx, n);
+    hipLaunchKernelGGL((argminmax_stage2<true>), dim3(1), dim3(64), 0, s,
+                       pvals, pidx, (int64_t*)out_idx, grid, n);
+  } else {
+    hipLaunchKernelGGL((argminmax_stage1<T, false>), dim3(grid), dim3(kBlock),
+                       0, s, in, pvals, pidx, n);
+    hipLaunchKernelGGL((argminmax_stage2<false>), dim3(1), dim3(64), 0, s,
+                       pvals, pidx, (int64_t*)out_idx, grid, n);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_argminmax(DType dt, bool maxop, const void* in, void* scratch,
+                      void* out_idx, int64_t n, hipStream_t s) {
+  if (dt == DType::F64)
+    launch_argminmax_t(maxop, (const double*)in, scratch, out_idx, n, s);
+  else
+    launch_argminmax_t(maxop, (const float*)in, scratch, out_idx, n, s);
 }
 
 template <typename T>

@@ -183,6 +183,11 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "argminmax":
+            v = _hipops.argminmax(
+                self._own(m["h"]), m["dtype"], m["n"], m["maxop"]
+            )
+            return {"ok": True, "value": v}, b""
         if op == "reduce_axis":
             h = _hipops.reduce_axis(
                 self._own(m["h"]), m["dtype"], m["outer"], m["red"],

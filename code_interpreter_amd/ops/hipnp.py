@@ -299,6 +299,12 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def argminmax(self, h, dtype, n, maxop):
+        return self._call(
+            {"op": "argminmax", "h": h, "dtype": dtype, "n": n,
+             "maxop": maxop}
+        )["value"]
+
     def reduce_axis(self, h, dtype, outer, red, inner, mode):
         return self._call(
             {"op": "reduce_axis", "h": h, "dtype": dtype, "outer": outer,
@@ -681,6 +687,24 @@ class DeviceArray:
                 return r
         return self.materialize().min(axis=axis, **kwargs)
 
+    def argmax(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return _np.intp(
+                backend().argminmax(
+                    self._dev_handle(), _dtype_code(self.dtype), self.size, 1
+                )
+            )
+        return self.materialize().argmax(axis=axis, **kwargs)
+
+    def argmin(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return _np.intp(
+                backend().argminmax(
+                    self._dev_handle(), _dtype_code(self.dtype), self.size, 0
+                )
+            )
+        return self.materialize().argmin(axis=axis, **kwargs)
+
     def var(self, axis=None, ddof=0, **kwargs):
         """Two-pass variance entirely on-device: mean, then the fused
         sum((x-mean)^2). (The one-pass sum-of-squares form cancels
@@ -766,6 +790,12 @@ class DeviceArray:
                 return r
         if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
             return args[0]._unary("square")
+        if func in (_np.argmax, _np.argmin) and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ):
+            if set(kwargs) <= {"axis"}:
+                meth = "argmax" if func is _np.argmax else "argmin"
+                return getattr(args[0], meth)(**kwargs)
         _reductions = {
             _np.max: "max", _np.amax: "max",
             _np.min: "min", _np.amin: "min",
@@ -928,6 +958,16 @@ def matmul(a, b, _force=False):
         and a_shape[2] == b_shape[1]
     ):
         return _matmul_batched(a, b, _force)
+    if (
+        len(a_shape) == 1
+        and b_shape == a_shape
+        and a_shape[0] >= MIN_ELEMS
+    ):
+        da, db = _to_device(a), _to_device(b)
+        if da is not None and db is not None and da.dtype == db.dtype:
+            prod = da._binary("multiply", db)
+            if prod is not NotImplemented:
+                return prod.sum()  # 1-D dot: scalar on device
     if len(a_shape) != 2 or len(b_shape) != 2 or a_shape[1] != b_shape[0]:
         return NotImplemented
     m, k = a_shape

@@ -71,6 +71,9 @@ class FakeHipops(types.ModuleType):
     def gemm(self, ha, hb, m, n, k, dtype):
         return self._new(b"\0" * (m * n * 8))
 
+    def argminmax(self, h, dtype, n, maxop):
+        return 0
+
     def reduce_axis(self, h, dtype, outer, red, inner, mode):
         return self._new(b"\0" * (outer * inner * 8))
 
@@ -131,6 +134,7 @@ def test_cross_connection_handle_access_denied(hipd):
             {"op": "download", "h": h, "nbytes": 128},
             {"op": "reduce_axis", "h": h, "dtype": 1, "outer": 2, "red": 4,
              "inner": 2, "mode": 0},
+            {"op": "argminmax", "h": h, "dtype": 1, "n": 16, "maxop": 1},
             {"op": "gemm_batched", "ha": h, "hb": h, "batch": 1, "m": 2,
              "n": 2, "k": 2, "dtype": 1},
             {"op": "free", "h": h},

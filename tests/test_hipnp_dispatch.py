@@ -100,6 +100,11 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def argminmax(self, h, dtype, n, maxop):
+        self.calls.append("argminmax")
+        a = self.bufs[h].view(self._dt(dtype))[:n]
+        return int(a.argmax() if maxop else a.argmin())
+
     def reduce_axis(self, h, dtype, outer, red, inner, mode):
         self.calls.append("reduce_axis")
         a = self.bufs[h].view(self._dt(dtype))[: outer * red * inner]
@@ -352,3 +357,28 @@ def test_batched_matmul_on_device(fake):
     assert r.shape == (4, 8, 9)
     np.testing.assert_allclose(np.asarray(r), np.matmul(a, b), rtol=1e-12)
     assert "gemm_batched" in fake.calls
+
+
+def test_argmax_argmin_on_device(fake):
+    host = np.random.default_rng(15).normal(0, 10, 512)
+    host[100] = host.max() + 5
+    host[200] = host.min() - 5
+    x = _device(fake, host)
+    assert int(x.argmax()) == 100
+    assert int(x.argmin()) == 200
+    assert int(np.argmax(x)) == 100
+    assert int(np.argmin(x)) == 200
+    assert "argminmax" in fake.calls
+    # axis argmax: host fallback, same values
+    h2 = np.random.default_rng(16).random((4, 6))
+    x2 = _device(fake, h2)
+    np.testing.assert_array_equal(np.argmax(x2, axis=1), h2.argmax(axis=1))
+
+
+def test_1d_dot_stays_on_device(fake, monkeypatch):
+    monkeypatch.setattr(hipnp, "MIN_ELEMS", 16)
+    a = np.random.default_rng(17).random(64)
+    b = np.random.default_rng(18).random(64)
+    r = hipnp.matmul(_device(fake, a), _device(fake, b))
+    assert not isinstance(r, hipnp.DeviceArray)  # scalar
+    assert float(r) == pytest.approx(float(a @ b), rel=1e-12)

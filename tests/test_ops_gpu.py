@@ -642,3 +642,29 @@ def test_torch_bmm_routed(hip_torch):
     for c in (c1, c2):
         err = (c.double().cpu() - ref).abs().max().item()
         assert err / (ref.abs().max().item() + 1e-9) < 2e-4
+
+
+def test_argminmax_matches_numpy(hip):
+    rng = np.random.default_rng(21)
+    for dtype, code in ((np.float64, 1), (np.float32, 0)):
+        a = rng.standard_normal(10_000_001).astype(dtype)
+        h = hip.upload(a)
+        amax = hip.argminmax(h, code, a.size, 1)
+        amin = hip.argminmax(h, code, a.size, 0)
+        hip.free(h)
+        assert amax == int(a.argmax())
+        assert amin == int(a.argmin())
+    # ties: first occurrence wins (numpy semantics)
+    t = np.zeros(1_000_000)
+    t[123] = 7.0
+    t[456_789] = 7.0
+    h = hip.upload(t)
+    assert hip.argminmax(h, 1, t.size, 1) == 123
+    assert hip.argminmax(h, 1, t.size, 0) == 0  # first zero
+    hip.free(h)
+    # NaN propagates: numpy returns the first NaN's index
+    t[50_000] = np.nan
+    h2 = hip.upload(t)
+    assert hip.argminmax(h2, 1, t.size, 1) == 50_000
+    assert hip.argminmax(h2, 1, t.size, 0) == 50_000
+    hip.free(h2)
