@@ -20,10 +20,10 @@ sys.path.insert(0, str(OPS_DIR))
 pytestmark = pytest.mark.gpu
 
 
-# must run through the hand-written MFMA kernels, matching torch numerics
-# ---------------------------------------------------------------------------
 @pytest.fixture(scope="module")
-def hip_torch(hip):
+def hip_torch():
+    # torch initializes FIRST in this process (no _hipops dependency):
+    # the order every sandbox child sees
     torch = pytest.importorskip("torch")
     if not torch.cuda.is_available():
         pytest.skip("torch sees no GPU")
