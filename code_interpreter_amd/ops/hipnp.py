@@ -687,6 +687,27 @@ class DeviceArray:
                 return r
         return self.materialize().min(axis=axis, **kwargs)
 
+    def clip(self, a_min=None, a_max=None, **kwargs):
+        """Scalar-bounds clip on device (maximum then minimum chains);
+        array bounds / extra kwargs fall back to host numpy."""
+        if not kwargs and (
+            a_min is None or isinstance(a_min, (int, float))
+        ) and (a_max is None or isinstance(a_max, (int, float))):
+            r = self
+            if a_min is not None:
+                r2 = r._binary("maximum", float(a_min))
+                if r2 is NotImplemented:
+                    return self.materialize().clip(a_min, a_max, **kwargs)
+                r = r2
+            if a_max is not None:
+                r2 = r._binary("minimum", float(a_max))
+                if r2 is NotImplemented:
+                    return self.materialize().clip(a_min, a_max, **kwargs)
+                r = r2
+            if r is not self:
+                return r
+        return self.materialize().clip(a_min, a_max, **kwargs)
+
     def argmax(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return _np.intp(
@@ -790,6 +811,9 @@ class DeviceArray:
                 return r
         if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
             return args[0]._unary("square")
+        if func is _np.clip and len(args) >= 1 and isinstance(args[0], DeviceArray):
+            if len(args) <= 3 and not kwargs:
+                return args[0].clip(*args[1:])
         if func in (_np.argmax, _np.argmin) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ):

@@ -382,3 +382,18 @@ def test_1d_dot_stays_on_device(fake, monkeypatch):
     r = hipnp.matmul(_device(fake, a), _device(fake, b))
     assert not isinstance(r, hipnp.DeviceArray)  # scalar
     assert float(r) == pytest.approx(float(a @ b), rel=1e-12)
+
+
+def test_clip_on_device(fake):
+    host = np.random.default_rng(19).normal(0, 2, 256)
+    x = _device(fake, host)
+    r = np.clip(x, -1.0, 1.0)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_array_equal(np.asarray(r), host.clip(-1.0, 1.0))
+    r2 = x.clip(0.5)  # min only
+    np.testing.assert_array_equal(np.asarray(r2), host.clip(0.5))
+    # array bounds: host fallback, same values
+    bounds = np.full(256, 0.25)
+    r3 = x.clip(bounds, None)
+    assert isinstance(r3, np.ndarray)
+    np.testing.assert_array_equal(r3, host.clip(bounds, None))
