@@ -680,6 +680,44 @@ PyObject* py_gemm(PyObject*, PyObject* args) {
                            (uint16_t*)out, m, n, k, g.compute);
       pool_free(bt, (int64_t)n * k * 2);
     }
+  } else if (2.0 * m * n * k >= 4e9) {
+    // large non-aligned shape: zero-pad to the 256-tile fast path
+    // (measured: the general 128-tile kernel runs ~120-160 TF vs ~1130
+    // for the 256 path at 4k-class shapes; padding costs three pitched
+    // device copies + two memsets at HBM rate). Zero rows/cols
+    // contribute exact zeros, so the extracted C block is bit-identical
+    // to the unpadded computation.
+    int mp = (m + 255) & ~255;
+    int np2 = (n + 255) & ~255;
+    int kp = (k + 127) & ~127;
+    void* ap = pool_alloc((int64_t)mp * kp * 2);
+    void* bp = pool_alloc((int64_t)kp * np2 * 2);
+    void* cp = pool_alloc((int64_t)mp * np2 * 2);
+    HIP_CHECK(hipMemsetAsync(ap, 0, (int64_t)mp * kp * 2, g.compute));
+    HIP_CHECK(hipMemsetAsync(bp, 0, (int64_t)kp * np2 * 2, g.compute));
+    HIP_CHECK(hipMemcpy2DAsync(ap, (size_t)kp * 2, a.ptr, (size_t)k * 2,
+                               (size_t)k * 2, m, hipMemcpyDeviceToDevice,
+                               g.compute));
+    HIP_CHECK(hipMemcpy2DAsync(bp, (size_t)np2 * 2, b.ptr, (size_t)n * 2,
+                               (size_t)n * 2, k, hipMemcpyDeviceToDevice,
+                               g.compute));
+    if (use_bf16_tr16()) {
+      launch_gemm_bf16_256t((const uint16_t*)ap, (const uint16_t*)bp,
+                            (uint16_t*)cp, mp, np2, kp, g.compute);
+    } else {
+      void* bt = pool_alloc((int64_t)np2 * kp * 2);
+      launch_transpose_bf16((const uint16_t*)bp, (uint16_t*)bt, kp, np2,
+                            g.compute);
+      launch_gemm_bf16_256((const uint16_t*)ap, (const uint16_t*)bt,
+                           (uint16_t*)cp, mp, np2, kp, g.compute);
+      pool_free(bt, (int64_t)np2 * kp * 2);
+    }
+    HIP_CHECK(hipMemcpy2DAsync(out, (size_t)n * 2, cp, (size_t)np2 * 2,
+                               (size_t)n * 2, m, hipMemcpyDeviceToDevice,
+                               g.compute));
+    pool_free(ap, (int64_t)mp * kp * 2);
+    pool_free(bp, (int64_t)kp * np2 * 2);
+    pool_free(cp, (int64_t)mp * np2 * 2);
   } else
     launch_gemm_bf16((const uint16_t*)a.ptr, (const uint16_t*)b.ptr,
                      (uint16_t*)out, m, n, k, g.compute);

@@ -102,6 +102,21 @@ def _try_mm(torch, a, b):
     n = b.shape[1]
     if m == 0 or n == 0 or k == 0 or 2.0 * m * n * k < MIN_MM_FLOPS:
         return None
+    if (
+        dt == 2
+        and not _hipgemm.gemm_bf16_256_ok(m, n, k)
+        and 2.0 * m * n * k >= 4e9
+    ):
+        # large non-aligned bf16: zero-pad to the 256-tile fast path
+        # (~9x the general kernel; zero padding is exact)
+        mp, np_, kp = (m + 255) & ~255, (n + 255) & ~255, (k + 127) & ~127
+        a_p = torch.zeros((mp, kp), dtype=a.dtype, device=a.device)
+        a_p[:m, :k] = a
+        b_p = torch.zeros((kp, np_), dtype=b.dtype, device=b.device)
+        b_p[:k, :n] = b
+        c_p = _try_mm(torch, a_p, b_p)
+        if c_p is not None:
+            return c_p[:m, :n]
     a = a.contiguous()
     b = b.contiguous()
     c = torch.empty((m, n), dtype=a.dtype, device=a.device)

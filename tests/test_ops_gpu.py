@@ -561,3 +561,17 @@ def test_argminmax_matches_numpy(hip):
     assert hip.argminmax(h2, 1, t.size, 1) == 50_000
     assert hip.argminmax(h2, 1, t.size, 0) == 50_000
     hip.free(h2)
+
+
+def test_gemm_bf16_padded_path_matches(hip):
+    # above the pad-to-256 threshold with non-aligned dims: the result
+    # must be bit-identical to the (slow) general kernel's math
+    m, n, k = 2000, 1500, 1100
+    rng = np.random.default_rng(22)
+    a = f32_to_bf16_np(rng.uniform(-1, 1, (m, k)).astype(np.float32))
+    b = f32_to_bf16_np(rng.uniform(-1, 1, (k, n)).astype(np.float32))
+    c = bf16_to_f32_np(_gemm_host(hip, a, b, 2, np.uint16))
+    ref = bf16_to_f32_np(a).astype(np.float64) @ bf16_to_f32_np(b).astype(
+        np.float64
+    )
+    np.testing.assert_allclose(c, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k))
