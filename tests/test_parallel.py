@@ -119,3 +119,25 @@ def test_gpu_inventory_cpu_safe():
 
     inv = gpu_inventory()  # no GPU here: must be an empty list, no raise
     assert isinstance(inv, list)
+
+
+def _bucketed_auto(rank, world_size):
+    # spawn-picklable module-level worker
+    import torch
+
+    from code_interpreter_amd.parallel import allreduce_bucketed
+
+    tensors = [torch.full((n,), float(rank + 1)) for n in (64, 4096, 7)]
+    allreduce_bucketed(tensors)  # None -> topology-recommended bucket size
+    expected = world_size * (world_size + 1) / 2
+    for t in tensors:
+        assert torch.allclose(t, torch.full_like(t, expected))
+    return float(tensors[1][0])
+
+
+def test_run_distributed_gloo_world4_bucketed():
+    """World-4 rehearsal of the exact collective pattern the 8-GPU
+    acceptance workload uses (bucketed all-reduce with topology-picked
+    bucket size: the bucket_bytes=None path)."""
+    result = run_distributed(_bucketed_auto, world_size=4, backend="gloo")
+    assert result == 4 * 5 / 2
