@@ -85,7 +85,16 @@ class RemoteBackend:
     """RPC to the engine's GPU daemon (ops/hipd.py) over a unix socket.
     On connection loss with no live device handles (e.g. a pre-warmed
     child whose daemon restarted) it transparently reconnects and
-    retries; with live handles it raises GpuBackendLost."""
+    retries; with live handles it raises GpuBackendLost.
+
+    One connection (and one lock) per sandbox process, BY DESIGN: the
+    daemon's handle-ownership boundary is the connection, so a
+    per-thread connection pool would make device arrays unusable across
+    the user's threads (each thread's handles would be invisible to the
+    others). The cost is that a sandbox's device ops serialize at the
+    RPC layer even if user code is multi-threaded -- an accepted trade
+    for cross-request isolation; compute itself is already serialized
+    per GPU by the daemon's single HIP context."""
 
     name = "remote"
 
