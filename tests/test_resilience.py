@@ -158,3 +158,40 @@ def test_backpressure_queues_overload(tmp_path, executor_bin):
         asyncio.run(run())
     finally:
         asyncio.run(ex.aclose())
+
+
+def test_multi_gpu_engine_pinning_round_robin(tmp_path, executor_bin):
+    """gpu_count=8 on a CPU box: engines still spawn (CPU sandboxes) and
+    each engine process carries its round-robin HIP_VISIBLE_DEVICES pin —
+    the data-parallel fan-out config the driver's 8-GPU run relies on."""
+    import psutil
+
+    from code_interpreter_amd.services.local_executor import LocalPoolExecutor
+    from code_interpreter_amd.services.storage import Storage
+
+    ex = LocalPoolExecutor(
+        Storage(str(tmp_path / "s")),
+        pool_target_length=1,
+        engines_per_gpu=1,
+        gpu_count=8,
+        gpu_pinning=True,
+        executor_root=str(tmp_path / "e"),
+        server_bin=executor_bin,
+        dep_install=False,
+        hip_numpy="off",
+    )
+
+    async def run():
+        assert ex.n_engines == 8
+        # touch a few engines (full 8 would be slow on CI): indexes 0,3,7
+        for idx in (0, 3, 7):
+            engine = await ex._ensure_engine(idx)
+            proc = psutil.Process(engine.proc.pid)
+            env = proc.environ()
+            assert env.get("HIP_VISIBLE_DEVICES") == str(idx % 8), (
+                idx, env.get("HIP_VISIBLE_DEVICES"))
+        r = await ex.execute("print('pinned-ok')")
+        assert r.exit_code == 0 and "pinned-ok" in r.stdout
+        await ex.aclose()
+
+    asyncio.run(run())
