@@ -148,16 +148,32 @@ template <int BKT>
 __global__ __launch_bounds__(THREADS2) void gemm_f64_128_kernel(
     const double* __restrict__ A, const double* __restrict__ B,
     double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
-  int nwg = tiles_m * tiles_n;
-  int wgid = blockIdx.x;
+  int tm, tn;
   {
-    const int nxcd = 8;
-    int q = nwg / nxcd, r = nwg % nxcd;
-    int xcd = wgid % nxcd, idx = wgid / nxcd;
-    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    int nwg = tiles_m * tiles_n;
+    int wgid = blockIdx.x;
+    const int n_st = (tiles_m / 8) * (tiles_n / 4);
+    if (tiles_m % 8 == 0 && tiles_n % 4 == 0 && n_st % 8 == 0) {
+      // 2D XCD supertiling (the bf16 kernel's mapping): each XCD works
+      // whole 8x4-tile supertiles so its L2 re-reads A row-bands 4x and
+      // B columns 8x
+      const int st_cols = tiles_n / 4;
+      int xcd = wgid % 8, idx = wgid / 8;
+      int st = xcd + 8 * (idx >> 5);
+      int p = idx & 31;
+      tm = (st / st_cols) * 8 + (p >> 2);
+      tn = (st % st_cols) * 4 + (p & 3);
+    } else {
+      const int nxcd = 8;
+      int q = nwg / nxcd, r = nwg % nxcd;
+      int xcd = wgid % nxcd, idx = wgid / nxcd;
+      wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+      tm = wgid / tiles_n;
+      tn = wgid % tiles_n;
+    }
   }
-  int row0 = (wgid / tiles_n) * BM2;
-  int col0 = (wgid % tiles_n) * BN2;
+  int row0 = tm * BM2;
+  int col0 = tn * BN2;
 
   __shared__ double As[2][BM2][BKT + 1];
   __shared__ double Bs[2][BKT][BN2 + 1];
