@@ -133,16 +133,20 @@ def client_worker(argv) -> None:
             # consecutive batches agree within 12%, bounded by
             # APP_BENCH_WARM_MAX_S.
             await run_phase(max(1, args.warm))
-            warm_cap = float(os.environ.get("APP_BENCH_WARM_MAX_S", "75"))
-            warm_deadline = time.monotonic() + warm_cap
+            warm_cap = float(os.environ.get("APP_BENCH_WARM_MAX_S", "90"))
+            warm_start = time.monotonic()
+            warm_deadline = warm_start + warm_cap
+            min_warm_s = min(15.0, warm_cap / 3)
             rate = 0.0
             stable = 0
-            while time.monotonic() < warm_deadline and stable < 2:
+            while time.monotonic() < warm_deadline and (
+                stable < 3 or time.monotonic() - warm_start < min_warm_s
+            ):
                 n = min(512, max(16, int(rate * 0.75))) if rate else 32
                 t0 = time.monotonic()
                 await run_phase(n)
                 new_rate = n / max(1e-9, time.monotonic() - t0)
-                if rate and abs(new_rate - rate) <= 0.12 * rate:
+                if rate and abs(new_rate - rate) <= 0.08 * rate:
                     stable += 1
                 else:
                     stable = 0
