@@ -170,3 +170,26 @@ def test_wire_format_roundtrip():
     back = pb.ExecuteRequest.FromString(data)
     assert back.source_code == "x"
     assert dict(back.files) == {"/workspace/a": "h" * 64}
+
+
+def test_cli_client_execute(service, tmp_path, capsys):
+    """The curl-style CLI (python -m code_interpreter_amd.grpc_api.client)
+    executes source against a live service and propagates the exit code."""
+    from code_interpreter_amd.grpc_api.client import main as cli_main
+
+    svc = GrpcServiceUnderTest(service).start()
+    try:
+        addr = f"127.0.0.1:{svc.port}"
+        rc = cli_main(["--addr", addr, "--source", "print(6 * 7)"])
+        out = capsys.readouterr()
+        assert rc == 0
+        assert "42" in out.out
+
+        script = tmp_path / "s.py"
+        script.write_text("import sys\nprint('boom')\nsys.exit(3)\n")
+        rc = cli_main(["--addr", addr, "--file", str(script)])
+        out = capsys.readouterr()
+        assert rc == 3
+        assert "boom" in out.out
+    finally:
+        svc.stop()
