@@ -603,6 +603,19 @@ class DeviceArray:
             return DeviceArray(out, self.shape, self.dtype)
         return NotImplemented
 
+    @staticmethod
+    def _norm_axis(axis, nd):
+        if isinstance(axis, (tuple, list)):
+            if len(axis) != 1:
+                return None
+            axis = axis[0]
+        if not isinstance(axis, (int, _np.integer)):
+            return None
+        axis = int(axis)
+        if axis < 0:
+            axis += nd
+        return axis if 0 <= axis < nd else None
+
     def _axis_reduce(self, mode, axis, keepdims=False):
         """Single-axis reduction on-device: the contiguous array viewed
         as [outer][red][inner], reduced over the middle. Returns None
@@ -610,17 +623,9 @@ class DeviceArray:
         code = _dtype_code(self.dtype)
         if code is None:
             return None
-        if isinstance(axis, (tuple, list)):
-            if len(axis) != 1:
-                return None
-            axis = axis[0]
-        if not isinstance(axis, (int, _np.integer)):
-            return None
         nd = len(self.shape)
-        axis = int(axis)
-        if axis < 0:
-            axis += nd
-        if not 0 <= axis < nd or self.size == 0:
+        axis = self._norm_axis(axis, nd)
+        if axis is None or self.size == 0:
             return None
         outer = 1
         for sdim in self.shape[:axis]:
@@ -629,6 +634,11 @@ class DeviceArray:
         inner = 1
         for sdim in self.shape[axis + 1:]:
             inner *= sdim
+        if inner == 1 and outer < 64:
+            # last-axis reduce with very few slices: the wave-per-slice
+            # kernel would use < 64 of the chip's ~8k wave slots -- let
+            # the caller pick a better path (full-reduce or host)
+            return None
         h = backend().reduce_axis(
             self._dev_handle(), code, outer, red, inner, mode
         )
@@ -639,11 +649,28 @@ class DeviceArray:
         )
         return DeviceArray(h, out_shape, self.dtype)
 
+    def _full_axis_scalar(self, axis, kwargs, compute):
+        """axis reduce of a 1-D array == full reduce: route to the
+        scalar kernel (numpy returns a scalar, or a 1-element array
+        with keepdims)."""
+        if (
+            len(self.shape) == 1
+            and set(kwargs) <= {"keepdims"}
+            and self._norm_axis(axis, 1) == 0
+        ):
+            val = compute()
+            return _np.array([val]) if kwargs.get("keepdims") else val
+        return None
+
     def sum(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(
                 backend().sum(self._dev_handle(), _dtype_code(self.dtype), self.size, 0)
             )
+        if axis is not None:
+            r = self._full_axis_scalar(axis, kwargs, lambda: self.sum())
+            if r is not None:
+                return r
         if axis is not None and set(kwargs) <= {"keepdims"}:
             r = self._axis_reduce(_REDUCE_SUM, axis, kwargs.get("keepdims", False))
             if r is not None:
@@ -653,6 +680,10 @@ class DeviceArray:
     def mean(self, axis=None, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(float(self.sum()) / self.size)
+        if axis is not None:
+            r = self._full_axis_scalar(axis, kwargs, lambda: self.mean())
+            if r is not None:
+                return r
         if axis is not None and set(kwargs) <= {"keepdims"}:
             r = self._axis_reduce(_REDUCE_SUM, axis, kwargs.get("keepdims", False))
             if r is not None:
@@ -677,6 +708,10 @@ class DeviceArray:
                     self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MAX
                 )
             )
+        if axis is not None:
+            r = self._full_axis_scalar(axis, kwargs, lambda: self.max())
+            if r is not None:
+                return r
         if axis is not None and set(kwargs) <= {"keepdims"}:
             r = self._axis_reduce(_REDUCE_MAX, axis, kwargs.get("keepdims", False))
             if r is not None:
@@ -690,6 +725,10 @@ class DeviceArray:
                     self._dev_handle(), _dtype_code(self.dtype), self.size, _REDUCE_MIN
                 )
             )
+        if axis is not None:
+            r = self._full_axis_scalar(axis, kwargs, lambda: self.min())
+            if r is not None:
+                return r
         if axis is not None and set(kwargs) <= {"keepdims"}:
             r = self._axis_reduce(_REDUCE_MIN, axis, kwargs.get("keepdims", False))
             if r is not None:

@@ -311,7 +311,8 @@ def test_ufunc_out_self_inplace(fake):
 
 
 def test_axis_reductions_stay_on_device(fake):
-    host = np.random.default_rng(11).random((6, 8, 10))
+    # outer >= 64 for the last axis so the wave-per-slice path engages
+    host = np.random.default_rng(11).random((16, 8, 10))
     x = _device(fake, host)
     for axis in (0, 1, 2, -1):
         r = x.sum(axis=axis)
@@ -397,3 +398,30 @@ def test_clip_on_device(fake):
     r3 = x.clip(bounds, None)
     assert isinstance(r3, np.ndarray)
     np.testing.assert_array_equal(r3, host.clip(bounds, None))
+
+
+def test_axis_reduce_1d_uses_scalar_path(fake):
+    host = np.random.default_rng(20).random(4096)
+    x = _device(fake, host)
+    before = fake.calls.count("reduce_axis")
+    s0 = x.sum(axis=0)
+    s_neg = np.sum(x, axis=-1)
+    m = x.max(axis=0)
+    k = x.sum(axis=0, keepdims=True)
+    # 1-D axis reduce == full reduce: scalar kernel, never reduce_axis
+    assert fake.calls.count("reduce_axis") == before
+    assert float(s0) == pytest.approx(host.sum(), rel=1e-12)
+    assert float(s_neg) == pytest.approx(host.sum(), rel=1e-12)
+    assert float(m) == host.max()
+    assert isinstance(k, np.ndarray) and k.shape == (1,)
+    np.testing.assert_allclose(k, host.sum(keepdims=True))
+
+
+def test_axis_reduce_few_slices_falls_back(fake):
+    # 4 slices on the last axis: the wave-per-slice kernel would idle
+    # the chip; host fallback keeps the values identical
+    host = np.random.default_rng(21).random((4, 1000))
+    x = _device(fake, host)
+    r = x.sum(axis=1)
+    assert isinstance(r, np.ndarray)
+    np.testing.assert_allclose(r, host.sum(axis=1))
