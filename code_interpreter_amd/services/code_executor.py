@@ -274,10 +274,19 @@ class SandboxClient:
 
     async def _run(self, storage, source_code, files, env, timeout, prefix) -> Result:
         async def upload(path: str, object_hash: str):
-            async with storage.reader(object_hash) as reader:
-                data = await reader.read()
+            # stream in 64 KiB chunks (chunked transfer into the
+            # executor's streaming PUT): a multi-GB input file must not
+            # be buffered whole in the control plane
+            async def chunks():
+                async with storage.reader(object_hash) as reader:
+                    while True:
+                        piece = await reader.read(1 << 16)
+                        if not piece:
+                            break
+                        yield piece
+
             async with self._client().put(
-                f"{prefix}/workspace/{_rel(path)}", data=data
+                f"{prefix}/workspace/{_rel(path)}", data=chunks()
             ) as resp:
                 if resp.status not in (200, 204):
                     raise ExecutorError(f"upload of {path} failed: {resp.status}")
