@@ -49,6 +49,11 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// broadcast binary over [outer][inner]: mode 0 = b[inner] along outer,
+// mode 1 = b[outer] along inner
+void launch_binary_bcast(DType dt, BinOp op, int mode, const void* a,
+                         const void* b, void* out, int64_t outer,
+                         int64_t inner, hipStream_t stream);
 // argmax/argmin (numpy tie/NaN semantics): scratch >= 2048*(8+8) bytes,
 // result int64 index written to out_idx (device)
 void launch_argminmax(DType dt, bool maxop, const void* in, void* scratch,

@@ -575,6 +575,29 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// binary_bcast(ha, hb, bop, dtype, outer, inner, mode) -> handle
+PyObject* py_binary_bcast(PyObject*, PyObject* args) {
+  unsigned long long ha, hb;
+  int op, dt, mode;
+  long long outer, inner;
+  if (!PyArg_ParseTuple(args, "KKiiLLi", &ha, &hb, &op, &dt, &outer, &inner,
+                        &mode))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DType dtype = dtype_from_int(dt);
+  DevBuf& a = get_buf(ha);
+  DevBuf& b = get_buf(hb);
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(a.size);
+  launch_binary_bcast(dtype, (BinOp)op, mode, a.ptr, b.ptr, out, outer,
+                      inner, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, a.size));
+  WRAP_END
+}
+
 // argminmax(h, dtype, n, maxop) -> int64 index (numpy tie/NaN semantics)
 PyObject* py_argminmax(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -833,6 +856,8 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"binary_bcast", py_binary_bcast, METH_VARARGS,
+     "binary_bcast(ha, hb, op, dtype, outer, inner, mode) -> handle"},
     {"argminmax", py_argminmax, METH_VARARGS,
      "argminmax(h, dtype, n, maxop) -> int64 index"},
     {"reduce_axis", py_reduce_axis, METH_VARARGS,

@@ -262,6 +262,36 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// broadcast binary: a viewed [outer][inner] combined with a vector b --
+// mode 0: b[inner] broadcast along outer (row vector, e.g. x - mean0)
+// mode 1: b[outer] broadcast along inner (column vector, keepdims shape)
+// ---------------------------------------------------------------------------
+template <typename T, int MODE>
+__global__ void binary_bcast_kernel(const T* __restrict__ a,
+                                    const T* __restrict__ b,
+                                    T* __restrict__ out, int64_t outer,
+                                    int64_t inner, int op) {
+  int64_t n = outer * inner;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    T av = a[i];
+    T bv = MODE == 0 ? b[i % inner] : b[i / inner];
+    double x = (double)av, y = (double)bv, r;
+    switch (op) {
+      case 0: r = x + y; break;
+      case 1: r = x - y; break;
+      case 2: r = x * y; break;
+      case 3: r = x / y; break;
+      case 4: r = (y != y) ? y : (x != x ? x : (x > y ? x : y)); break;
+      case 5: r = (y != y) ? y : (x != x ? x : (x < y ? x : y)); break;
+      default: r = pow(x, y); break;
+    }
+    out[i] = (T)r;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // argmax/argmin: two-stage (value, index) reduction. numpy semantics:
 // FIRST occurrence wins ties; NaN wins (propagates) like np.argmax.
 // ---------------------------------------------------------------------------
@@ -635,6 +665,32 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
   HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_binary_bcast_t(BinOp op, int mode, const T* a, const T* b,
+                                  T* out, int64_t outer, int64_t inner,
+                                  hipStream_t s) {
+  int64_t n = outer * inner;
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 4096);
+  if (mode == 0)
+    hipLaunchKernelGGL((binary_bcast_kernel<T, 0>), dim3(grid), dim3(kBlock),
+                       0, s, a, b, out, outer, inner, (int)op);
+  else
+    hipLaunchKernelGGL((binary_bcast_kernel<T, 1>), dim3(grid), dim3(kBlock),
+                       0, s, a, b, out, outer, inner, (int)op);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_binary_bcast(DType dt, BinOp op, int mode, const void* a,
+                         const void* b, void* out, int64_t outer,
+                         int64_t inner, hipStream_t s) {
+  if (dt == DType::F64)
+    launch_binary_bcast_t(op, mode, (const double*)a, (const double*)b,
+                          (double*)out, outer, inner, s);
+  else
+    launch_binary_bcast_t(op, mode, (const float*)a, (const float*)b,
+                          (float*)out, outer, inner, s);
 }
 
 template <typename T>

@@ -183,6 +183,13 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "binary_bcast":
+            h = _hipops.binary_bcast(
+                self._own(m["ha"]), self._own(m["hb"]), m["bop"], m["dtype"],
+                m["outer"], m["inner"], m["mode"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "argminmax":
             v = _hipops.argminmax(
                 self._own(m["h"]), m["dtype"], m["n"], m["maxop"]

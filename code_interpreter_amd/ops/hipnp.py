@@ -308,6 +308,12 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def binary_bcast(self, ha, hb, bop, dtype, outer, inner, mode):
+        return self._call(
+            {"op": "binary_bcast", "ha": ha, "hb": hb, "bop": bop,
+             "dtype": dtype, "outer": outer, "inner": inner, "mode": mode}
+        )["h"]
+
     def argminmax(self, h, dtype, n, maxop):
         return self._call(
             {"op": "argminmax", "h": h, "dtype": dtype, "n": n,
@@ -590,6 +596,10 @@ class DeviceArray:
         code = _dtype_code(self.dtype)
         if isinstance(other, DeviceArray):
             if other.shape != self.shape or other.dtype != self.dtype:
+                if not reverse:
+                    r = self._binary_bcast(opname, other)
+                    if r is not NotImplemented:
+                        return r
                 return NotImplemented
             a, b = (other, self) if reverse else (self, other)
             out = backend().binary(a._dev_handle(), b._dev_handle(), _BINARY[opname], code, self.size)
@@ -602,6 +612,30 @@ class DeviceArray:
             )
             return DeviceArray(out, self.shape, self.dtype)
         return NotImplemented
+
+    def _binary_bcast(self, opname, other):
+        """Device broadcasting for the two clean numpy cases:
+        row vector (other.shape == self.shape[-1:], broadcast along the
+        leading axes) and keepdims column (other.shape == self.shape with
+        the last axis == 1). Returns NotImplemented otherwise."""
+        if not isinstance(other, DeviceArray) or other.dtype != self.dtype:
+            return NotImplemented
+        code = _dtype_code(self.dtype)
+        if code is None or len(self.shape) < 2:
+            return NotImplemented
+        inner = self.shape[-1]
+        outer = self.size // max(1, inner)
+        if other.shape == self.shape[-1:]:
+            mode = 0  # b[inner] along outer
+        elif other.shape == self.shape[:-1] + (1,):
+            mode = 1  # b[outer] along inner
+        else:
+            return NotImplemented
+        out = backend().binary_bcast(
+            self._dev_handle(), other._dev_handle(), _BINARY[opname], code,
+            outer, inner, mode,
+        )
+        return DeviceArray(out, self.shape, self.dtype)
 
     @staticmethod
     def _norm_axis(axis, nd):

@@ -100,6 +100,18 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def binary_bcast(self, ha, hb, bop, dtype, outer, inner, mode):
+        self.calls.append("binary_bcast")
+        dt = self._dt(dtype)
+        a = self.bufs[ha].view(dt).reshape(-1)[: outer * inner].reshape(
+            outer, inner)
+        b = self.bufs[hb].view(dt).reshape(-1)
+        if mode == 0:
+            out = self._BOPS[bop](a, b[:inner][None, :])
+        else:
+            out = self._BOPS[bop](a, b[:outer][:, None])
+        return self._new(out.astype(dt))
+
     def argminmax(self, h, dtype, n, maxop):
         self.calls.append("argminmax")
         a = self.bufs[h].view(self._dt(dtype))[:n]
@@ -425,3 +437,28 @@ def test_axis_reduce_few_slices_falls_back(fake):
     r = x.sum(axis=1)
     assert isinstance(r, np.ndarray)
     np.testing.assert_allclose(r, host.sum(axis=1))
+
+
+def test_broadcast_binary_on_device(fake):
+    host = np.random.default_rng(22).random((12, 7))
+    x = _device(fake, host)
+    row = _device(fake, host.mean(axis=0))           # shape (7,)
+    col = _device(fake, host.mean(axis=1)[:, None])  # shape (12, 1)
+    r = x - row
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(np.asarray(r), host - host.mean(axis=0),
+                               rtol=1e-12)
+    r2 = x / col
+    assert isinstance(r2, hipnp.DeviceArray)
+    np.testing.assert_allclose(np.asarray(r2),
+                               host / host.mean(axis=1)[:, None], rtol=1e-12)
+    assert "binary_bcast" in fake.calls
+    # the composed idiom: center by column means, all on device
+    centered = x - x.mean(axis=0, keepdims=False)
+    np.testing.assert_allclose(np.asarray(centered),
+                               host - host.mean(axis=0), rtol=1e-12)
+    # invalid broadcast (12,7) - (12,): numpy raises; the duck array
+    # must not silently compute something else
+    mid = _device(fake, host.mean(axis=1))  # shape (12,)
+    with pytest.raises((ValueError, TypeError)):
+        _ = x - mid

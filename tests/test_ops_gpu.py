@@ -614,3 +614,38 @@ def test_concurrent_daemon_style_ops_are_isolated(hip):
     for t in threads:
         t.join()
     assert not errors, errors[:3]
+
+
+def test_binary_bcast_matches_numpy(hip):
+    rng = np.random.default_rng(40)
+    outer, inner = 4096, 512
+    a = rng.standard_normal((outer, inner))
+    row = rng.standard_normal(inner)
+    col = rng.standard_normal(outer)
+    ha = hip.upload(a)
+    hrow = hip.upload(row)
+    hcol = hip.upload(col)
+    out = np.empty_like(a)
+    # mode 0 (row) subtract; mode 1 (col) divide
+    h = hip.binary_bcast(ha, hrow, 1, 1, outer, inner, 0)
+    hip.download(h, out)
+    hip.free(h)
+    np.testing.assert_allclose(out, a - row[None, :], rtol=1e-12)
+    h = hip.binary_bcast(ha, hcol, 3, 1, outer, inner, 1)
+    hip.download(h, out)
+    hip.free(h)
+    np.testing.assert_allclose(out, a / col[:, None], rtol=1e-12)
+    for hh in (ha, hrow, hcol):
+        hip.free(hh)
+
+
+def test_device_center_normalize_chain(hnp):
+    # the composed idiom entirely on device: (x - mean0) / (std0-ish)
+    x = hnp.rand(2048, 1024, seed=55)
+    mu = x.mean(axis=0)
+    centered = x - mu
+    assert isinstance(centered, hnp.DeviceArray)
+    host = np.asarray(x)
+    np.testing.assert_allclose(
+        np.asarray(centered), host - host.mean(axis=0), rtol=1e-9, atol=1e-12
+    )
