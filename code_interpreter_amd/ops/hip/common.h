@@ -49,6 +49,16 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// boolean masks: compare -> u8, np.where select, masked fill, popcount
+void launch_compare(DType dt, int op, const void* a, const void* b,
+                    double scalar, void* out_u8, int64_t n, hipStream_t s);
+void launch_where(DType dt, const void* mask, const void* pa, double sa,
+                  const void* pb, double sb, void* out, int64_t n,
+                  hipStream_t s);
+void launch_masked_fill(DType dt, void* data, const void* mask, double value,
+                        int64_t n, hipStream_t s);
+void launch_mask_count(const void* mask, void* scratch, void* out_i64,
+                       int64_t n, hipStream_t s);
 // broadcast binary over [outer][inner]: mode 0 = b[inner] along outer,
 // mode 1 = b[outer] along inner
 void launch_binary_bcast(DType dt, BinOp op, int mode, const void* a,

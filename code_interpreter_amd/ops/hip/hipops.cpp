@@ -575,6 +575,95 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// compare(h, dtype, n, cmp_op, hb_or_0, scalar) -> u8 mask handle
+PyObject* py_compare(PyObject*, PyObject* args) {
+  unsigned long long h, hb;
+  int dt, op;
+  long long n;
+  double scalar;
+  if (!PyArg_ParseTuple(args, "KiLiKd", &h, &dt, &n, &op, &hb, &scalar))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (op < 0 || op > 5) throw std::runtime_error("bad compare op");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& a = get_buf(h);
+  void* bptr = hb ? get_buf(hb).ptr : nullptr;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n);
+  launch_compare(dtype, op, a.ptr, bptr, scalar, out, n, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n));
+  WRAP_END
+}
+
+// where(hmask, dtype, n, ha_or_0, sa, hb_or_0, sb) -> handle
+PyObject* py_where(PyObject*, PyObject* args) {
+  unsigned long long hm, ha, hb;
+  int dt;
+  long long n;
+  double sa, sb;
+  if (!PyArg_ParseTuple(args, "KiLKdKd", &hm, &dt, &n, &ha, &sa, &hb, &sb))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DType dtype = dtype_from_int(dt);
+  DevBuf& m = get_buf(hm);
+  void* pa = ha ? get_buf(ha).ptr : nullptr;
+  void* pb = hb ? get_buf(hb).ptr : nullptr;
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n * esize);
+  launch_where(dtype, m.ptr, pa, sa, pb, sb, out, n, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n * esize));
+  WRAP_END
+}
+
+// masked_fill(h, hmask, dtype, n, value): in-place on h's buffer
+PyObject* py_masked_fill(PyObject*, PyObject* args) {
+  unsigned long long h, hm;
+  int dt;
+  long long n;
+  double value;
+  if (!PyArg_ParseTuple(args, "KKiLd", &h, &hm, &dt, &n, &value))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DType dtype = dtype_from_int(dt);
+  DevBuf& a = get_buf(h);
+  DevBuf& m = get_buf(hm);
+  NOGIL_BEGIN
+  launch_masked_fill(dtype, a.ptr, m.ptr, value, n, g.compute);
+  NOGIL_END
+  Py_RETURN_NONE;
+  WRAP_END
+}
+
+// mask_count(hmask, n) -> int64 popcount
+PyObject* py_mask_count(PyObject*, PyObject* args) {
+  unsigned long long hm;
+  long long n;
+  if (!PyArg_ParseTuple(args, "KL", &hm, &n)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DevBuf& m = get_buf(hm);
+  long long result = 0;
+  NOGIL_BEGIN
+  std::lock_guard<std::mutex> reduce_lk(g.reduce_mu);
+  ensure_reduce_scratch();
+  launch_mask_count(m.ptr, g.reduce_scratch, g.scalar_dev, n, g.compute);
+  HIP_CHECK(hipMemcpyAsync(g.scalar_pin, g.scalar_dev, 8,
+                           hipMemcpyDeviceToHost, g.compute));
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  result = *(int64_t*)g.scalar_pin;
+  NOGIL_END
+  return PyLong_FromLongLong(result);
+  WRAP_END
+}
+
 // binary_bcast(ha, hb, bop, dtype, outer, inner, mode) -> handle
 PyObject* py_binary_bcast(PyObject*, PyObject* args) {
   unsigned long long ha, hb;
@@ -856,6 +945,14 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"compare", py_compare, METH_VARARGS,
+     "compare(h, dtype, n, cmp_op, hb_or_0, scalar) -> u8 mask handle"},
+    {"where", py_where, METH_VARARGS,
+     "where(hmask, dtype, n, ha_or_0, sa, hb_or_0, sb) -> handle"},
+    {"masked_fill", py_masked_fill, METH_VARARGS,
+     "masked_fill(h, hmask, dtype, n, value): in place"},
+    {"mask_count", py_mask_count, METH_VARARGS,
+     "mask_count(hmask, n) -> int64"},
     {"binary_bcast", py_binary_bcast, METH_VARARGS,
      "binary_bcast(ha, hb, op, dtype, outer, inner, mode) -> handle"},
     {"argminmax", py_argminmax, METH_VARARGS,

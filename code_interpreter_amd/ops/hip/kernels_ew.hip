@@ -262,6 +262,91 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// boolean-mask ops: comparisons -> u8 masks, select (np.where),
+// masked fill, mask popcount. CmpOp: 0 lt, 1 le, 2 gt, 3 ge, 4 eq, 5 ne
+// (NaN compares false like numpy, except ne where NaN != x is true).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ bool cmp_apply(int op, double x, double y) {
+  switch (op) {
+    case 0: return x < y;
+    case 1: return x <= y;
+    case 2: return x > y;
+    case 3: return x >= y;
+    case 4: return x == y;
+    default: return x != y;
+  }
+}
+
+template <typename T, bool SCALAR>
+__global__ void compare_kernel(const T* __restrict__ a,
+                               const T* __restrict__ b, double scalar,
+                               unsigned char* __restrict__ out, int64_t n,
+                               int op) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    double y = SCALAR ? scalar : (double)b[i];
+    out[i] = cmp_apply(op, (double)a[i], y) ? 1 : 0;
+  }
+}
+
+// out[i] = mask[i] ? a : b, with a/b each independently an array or a
+// scalar (null pointer selects the scalar)
+template <typename T>
+__global__ void where_kernel(const unsigned char* __restrict__ mask,
+                             const T* __restrict__ pa, double sa,
+                             const T* __restrict__ pb, double sb,
+                             T* __restrict__ out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (mask[i])
+      out[i] = pa ? pa[i] : (T)sa;
+    else
+      out[i] = pb ? pb[i] : (T)sb;
+  }
+}
+
+template <typename T>
+__global__ void masked_fill_kernel(T* __restrict__ data,
+                                   const unsigned char* __restrict__ mask,
+                                   double value, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (mask[i]) data[i] = (T)value;
+  }
+}
+
+__global__ void mask_count_stage1(const unsigned char* __restrict__ mask,
+                                  int64_t* __restrict__ partials, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t acc = 0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    acc += mask[i] ? 1 : 0;
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  __shared__ int64_t wsum[kBlock / 64];
+  int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wsum[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int64_t total = 0;
+    for (int w = 0; w < (int)blockDim.x / 64; w++) total += wsum[w];
+    partials[blockIdx.x] = total;
+  }
+}
+
+__global__ void mask_count_stage2(const int64_t* __restrict__ partials,
+                                  int64_t* __restrict__ out, int nparts) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    int64_t total = 0;
+    for (int i = 0; i < nparts; i++) total += partials[i];
+    out[0] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // broadcast binary: a viewed [outer][inner] combined with a vector b --
 // mode 0: b[inner] broadcast along outer (row vector, e.g. x - mean0)
 // mode 1: b[outer] broadcast along inner (column vector, keepdims shape)
@@ -664,6 +749,68 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
   else
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_compare_t(int op, const T* a, const T* b, double scalar,
+                             unsigned char* out, int64_t n, hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 4096);
+  if (b)
+    hipLaunchKernelGGL((compare_kernel<T, false>), dim3(grid), dim3(kBlock),
+                       0, s, a, b, scalar, out, n, op);
+  else
+    hipLaunchKernelGGL((compare_kernel<T, true>), dim3(grid), dim3(kBlock),
+                       0, s, a, b, scalar, out, n, op);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_compare(DType dt, int op, const void* a, const void* b,
+                    double scalar, void* out_u8, int64_t n, hipStream_t s) {
+  if (dt == DType::F64)
+    launch_compare_t(op, (const double*)a, (const double*)b, scalar,
+                     (unsigned char*)out_u8, n, s);
+  else
+    launch_compare_t(op, (const float*)a, (const float*)b, scalar,
+                     (unsigned char*)out_u8, n, s);
+}
+
+void launch_where(DType dt, const void* mask, const void* pa, double sa,
+                  const void* pb, double sb, void* out, int64_t n,
+                  hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 4096);
+  if (dt == DType::F64)
+    hipLaunchKernelGGL((where_kernel<double>), dim3(grid), dim3(kBlock), 0, s,
+                       (const unsigned char*)mask, (const double*)pa, sa,
+                       (const double*)pb, sb, (double*)out, n);
+  else
+    hipLaunchKernelGGL((where_kernel<float>), dim3(grid), dim3(kBlock), 0, s,
+                       (const unsigned char*)mask, (const float*)pa, sa,
+                       (const float*)pb, sb, (float*)out, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_masked_fill(DType dt, void* data, const void* mask, double value,
+                        int64_t n, hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 4096);
+  if (dt == DType::F64)
+    hipLaunchKernelGGL((masked_fill_kernel<double>), dim3(grid), dim3(kBlock),
+                       0, s, (double*)data, (const unsigned char*)mask, value,
+                       n);
+  else
+    hipLaunchKernelGGL((masked_fill_kernel<float>), dim3(grid), dim3(kBlock),
+                       0, s, (float*)data, (const unsigned char*)mask, value,
+                       n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_mask_count(const void* mask, void* scratch, void* out_i64,
+                       int64_t n, hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  hipLaunchKernelGGL(mask_count_stage1, dim3(grid), dim3(kBlock), 0, s,
+                     (const unsigned char*)mask, (int64_t*)scratch, n);
+  hipLaunchKernelGGL(mask_count_stage2, dim3(1), dim3(64), 0, s,
+                     (const int64_t*)scratch, (int64_t*)out_i64, grid);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -183,6 +183,37 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "compare":
+            hb = m.get("hb", 0)
+            if hb:
+                self._own(hb)
+            h = _hipops.compare(
+                self._own(m["h"]), m["dtype"], m["n"], m["cmp"], hb,
+                m.get("scalar", 0.0)
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "where":
+            ha, hb = m.get("ha", 0), m.get("hb", 0)
+            if ha:
+                self._own(ha)
+            if hb:
+                self._own(hb)
+            h = _hipops.where(
+                self._own(m["hm"]), m["dtype"], m["n"], ha,
+                m.get("sa", 0.0), hb, m.get("sb", 0.0)
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
+        if op == "masked_fill":
+            _hipops.masked_fill(
+                self._own(m["h"]), self._own(m["hm"]), m["dtype"], m["n"],
+                m["value"]
+            )
+            return {"ok": True}, b""
+        if op == "mask_count":
+            v = _hipops.mask_count(self._own(m["hm"]), m["n"])
+            return {"ok": True, "value": v}, b""
         if op == "binary_bcast":
             h = _hipops.binary_bcast(
                 self._own(m["ha"]), self._own(m["hb"]), m["bop"], m["dtype"],

@@ -308,6 +308,27 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def compare(self, h, dtype, n, cmp, hb, scalar):
+        return self._call(
+            {"op": "compare", "h": h, "dtype": dtype, "n": n, "cmp": cmp,
+             "hb": hb, "scalar": scalar}
+        )["h"]
+
+    def where(self, hm, dtype, n, ha, sa, hb, sb):
+        return self._call(
+            {"op": "where", "hm": hm, "dtype": dtype, "n": n, "ha": ha,
+             "sa": sa, "hb": hb, "sb": sb}
+        )["h"]
+
+    def masked_fill(self, h, hm, dtype, n, value):
+        self._call(
+            {"op": "masked_fill", "h": h, "hm": hm, "dtype": dtype, "n": n,
+             "value": value}
+        )
+
+    def mask_count(self, hm, n):
+        return self._call({"op": "mask_count", "hm": hm, "n": n})["value"]
+
     def binary_bcast(self, ha, hb, bop, dtype, outer, inner, mode):
         return self._call(
             {"op": "binary_bcast", "ha": ha, "hb": hb, "bop": bop,
@@ -351,6 +372,11 @@ _BINARY = {
 }
 # full-array reduction modes (the extension's ReduceOp)
 _REDUCE_SUM, _REDUCE_SUMSQ, _REDUCE_MAX, _REDUCE_MIN = 0, 1, 2, 3
+# comparison ops (u8 mask kernels)
+_CMP = {
+    "less": 0, "less_equal": 1, "greater": 2, "greater_equal": 3,
+    "equal": 4, "not_equal": 5,
+}
 
 MIN_ELEMS = int(os.environ.get("APP_HIP_NUMPY_MIN_ELEMS", 2_000_000))
 MIN_MATMUL_FLOPS = float(os.environ.get("APP_HIP_NUMPY_MIN_MATMUL_FLOPS", 5e7))
@@ -547,8 +573,34 @@ class DeviceArray:
 
     def __setitem__(self, idx, value):
         # in-place mutation (x[0] = 1, x[x < 0] = 0, ...): CPU numpy
-        # supports it, so the sandbox contract requires it; the host copy
-        # becomes authoritative and the device buffer is dropped
+        # supports it, so the sandbox contract requires it. Boolean-mask
+        # assignment of a scalar runs ON DEVICE (masked_fill on this
+        # array's buffer -- safe, handles are single-owner); everything
+        # else materializes, mutates the host copy, and drops the device
+        # buffer (re-uploaded lazily on the next device op).
+        if (
+            isinstance(value, (int, float))
+            and not isinstance(value, bool)
+            and _dtype_code(self.dtype) is not None
+            and self._handle is not None
+        ):
+            mask = None
+            if isinstance(idx, BoolDeviceArray) and idx.shape == self.shape:
+                mask = idx
+            elif (
+                isinstance(idx, _np.ndarray)
+                and idx.dtype == _np.bool_
+                and idx.shape == self.shape
+            ):
+                mh = backend().upload(_np.ascontiguousarray(idx))
+                mask = BoolDeviceArray(mh, idx.shape, _np.bool_)
+            if mask is not None:
+                backend().masked_fill(
+                    self._handle, mask._dev_handle(),
+                    _dtype_code(self.dtype), self.size, float(value),
+                )
+                self._host = None  # the device buffer is authoritative
+                return
         if isinstance(idx, DeviceArray):
             idx = idx.materialize()
         if isinstance(value, DeviceArray):
@@ -565,25 +617,57 @@ class DeviceArray:
     def __iter__(self):
         return iter(self.materialize())
 
-    # comparisons materialize and delegate to numpy (the default object
-    # identity would silently return False for `x == 5`-style masks)
+    # comparisons produce DEVICE boolean masks (u8 buffers) for scalar
+    # and same-shape operands; anything else materializes and delegates
+    # (the default object identity would silently return False for
+    # `x == 5`-style masks)
+    def _compare(self, cmp_name, other):
+        code = _dtype_code(self.dtype)
+        if code is None:
+            return None
+        if isinstance(other, (bool, _np.bool_)):
+            return None
+        if isinstance(other, (int, float)):
+            h = backend().compare(
+                self._dev_handle(), code, self.size, _CMP[cmp_name], 0,
+                float(other),
+            )
+        elif (
+            isinstance(other, DeviceArray)
+            and other.shape == self.shape
+            and other.dtype == self.dtype
+        ):
+            h = backend().compare(
+                self._dev_handle(), code, self.size, _CMP[cmp_name],
+                other._dev_handle(), 0.0,
+            )
+        else:
+            return None
+        return BoolDeviceArray(h, self.shape, _np.bool_)
+
     def __eq__(self, o):
-        return self.materialize() == _asarray_or_scalar(o)
+        r = self._compare("equal", o)
+        return self.materialize() == _asarray_or_scalar(o) if r is None else r
 
     def __ne__(self, o):
-        return self.materialize() != _asarray_or_scalar(o)
+        r = self._compare("not_equal", o)
+        return self.materialize() != _asarray_or_scalar(o) if r is None else r
 
     def __lt__(self, o):
-        return self.materialize() < _asarray_or_scalar(o)
+        r = self._compare("less", o)
+        return self.materialize() < _asarray_or_scalar(o) if r is None else r
 
     def __le__(self, o):
-        return self.materialize() <= _asarray_or_scalar(o)
+        r = self._compare("less_equal", o)
+        return self.materialize() <= _asarray_or_scalar(o) if r is None else r
 
     def __gt__(self, o):
-        return self.materialize() > _asarray_or_scalar(o)
+        r = self._compare("greater", o)
+        return self.materialize() > _asarray_or_scalar(o) if r is None else r
 
     def __ge__(self, o):
-        return self.materialize() >= _asarray_or_scalar(o)
+        r = self._compare("greater_equal", o)
+        return self.materialize() >= _asarray_or_scalar(o) if r is None else r
 
     __hash__ = None  # mutable-array semantics, same as numpy
 
@@ -832,6 +916,10 @@ class DeviceArray:
             return self._fallback_ufunc(ufunc, method, inputs, kwargs)
         name = ufunc.__name__
         if method == "__call__":
+            if name in _CMP and len(inputs) == 2 and inputs[0] is self:
+                r = self._compare(name, inputs[1])
+                if r is not None:
+                    return r
             if name in _UNARY and len(inputs) == 1 and inputs[0] is self:
                 return self._unary(name)
             if name in _BINARY and len(inputs) == 2:
@@ -893,6 +981,15 @@ class DeviceArray:
                 return r
         if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
             return args[0]._unary("square")
+        if func is _np.where and len(args) == 3 and not kwargs:
+            r = where_device(*args)
+            if r is not NotImplemented:
+                return r
+        if func is _np.count_nonzero and len(args) == 1 and isinstance(
+            args[0], BoolDeviceArray
+        ):
+            if not kwargs:
+                return args[0].sum()
         if func is _np.clip and len(args) >= 1 and isinstance(args[0], DeviceArray):
             if len(args) <= 3 and not kwargs:
                 return args[0].clip(*args[1:])
@@ -965,6 +1062,98 @@ class DeviceArray:
         if r is NotImplemented:
             raise TypeError("unsupported operand for device array")
         return r
+
+
+class BoolDeviceArray(DeviceArray):
+    """Device-resident boolean mask (u8 storage, numpy bool semantics).
+    Arithmetic/reduction kernels are f32/f64-only, so everything except
+    the mask-specific fast paths (popcount sum, np.where selection,
+    masked assignment) materializes to a host bool array."""
+
+    def sum(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return _np.intp(backend().mask_count(self._dev_handle(), self.size))
+        return self.materialize().sum(axis=axis, **kwargs)
+
+    def count_nonzero(self):
+        return self.sum()
+
+    def mean(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return _np.float64(int(self.sum()) / self.size)
+        return self.materialize().mean(axis=axis, **kwargs)
+
+    def any(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return bool(int(self.sum()) > 0)
+        return self.materialize().any(axis=axis, **kwargs)
+
+    def all(self, axis=None, **kwargs):
+        if axis is None and not kwargs:
+            return bool(int(self.sum()) == self.size)
+        return self.materialize().all(axis=axis, **kwargs)
+
+    # f32/f64-only device paths must not see a u8 buffer
+    def _unary(self, opname):
+        raise TypeError("unary op on boolean mask")
+
+    def _binary(self, opname, other, reverse=False):
+        return NotImplemented
+
+    def max(self, axis=None, **kwargs):
+        return self.materialize().max(axis=axis, **kwargs)
+
+    def min(self, axis=None, **kwargs):
+        return self.materialize().min(axis=axis, **kwargs)
+
+    def argmax(self, axis=None, **kwargs):
+        return self.materialize().argmax(axis=axis, **kwargs)
+
+    def argmin(self, axis=None, **kwargs):
+        return self.materialize().argmin(axis=axis, **kwargs)
+
+    def var(self, axis=None, ddof=0, **kwargs):
+        return self.materialize().var(axis=axis, ddof=ddof, **kwargs)
+
+    def std(self, axis=None, ddof=0, **kwargs):
+        return self.materialize().std(axis=axis, ddof=ddof, **kwargs)
+
+    def astype(self, dtype, **kwargs):
+        return self.materialize().astype(dtype, **kwargs)
+
+
+def where_device(mask, a, b):
+    """np.where(mask, a, b) on device; NotImplemented if not routable."""
+    if not isinstance(mask, BoolDeviceArray):
+        return NotImplemented
+
+    def classify(x):
+        if isinstance(x, DeviceArray) and not isinstance(x, BoolDeviceArray):
+            if x.shape == mask.shape and _dtype_code(x.dtype) is not None:
+                return x, None
+            return None, None  # unroutable array
+        if isinstance(x, (bool, _np.bool_)):
+            return None, None
+        if isinstance(x, (int, float)):
+            return None, float(x)
+        return None, None
+
+    da, sa = classify(a)
+    db, sb = classify(b)
+    if (da is None and sa is None) or (db is None and sb is None):
+        return NotImplemented
+    if da is not None and db is not None and da.dtype != db.dtype:
+        return NotImplemented
+    out_dtype = da.dtype if da is not None else (
+        db.dtype if db is not None else _np.dtype(_np.float64)
+    )
+    code = _dtype_code(out_dtype)
+    h = backend().where(
+        mask._dev_handle(), code, mask.size,
+        da._dev_handle() if da is not None else 0, sa or 0.0,
+        db._dev_handle() if db is not None else 0, sb or 0.0,
+    )
+    return DeviceArray(h, mask.shape, out_dtype)
 
 
 def _asarray(x):

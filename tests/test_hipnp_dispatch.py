@@ -100,6 +100,40 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    _CMPS = {
+        0: np.less, 1: np.less_equal, 2: np.greater, 3: np.greater_equal,
+        4: np.equal, 5: np.not_equal,
+    }
+
+    def compare(self, h, dtype, n, cmp, hb, scalar):
+        self.calls.append("compare")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
+        other = (
+            self.bufs[hb].view(self._dt(dtype)).reshape(-1)[:n]
+            if hb else scalar
+        )
+        return self._new(self._CMPS[cmp](a, other).astype(np.uint8))
+
+    def where(self, hm, dtype, n, ha, sa, hb, sb):
+        self.calls.append("where")
+        dt = self._dt(dtype)
+        m = self.bufs[hm].view(np.uint8).reshape(-1)[:n].astype(bool)
+        a = self.bufs[ha].view(dt).reshape(-1)[:n] if ha else np.full(n, sa, dt)
+        b = self.bufs[hb].view(dt).reshape(-1)[:n] if hb else np.full(n, sb, dt)
+        return self._new(np.where(m, a, b).astype(dt))
+
+    def masked_fill(self, h, hm, dtype, n, value):
+        self.calls.append("masked_fill")
+        dt = self._dt(dtype)
+        a = np.array(self.bufs[h].view(dt).reshape(-1)[:n])
+        m = self.bufs[hm].view(np.uint8).reshape(-1)[:n].astype(bool)
+        a[m] = value
+        self.bufs[h] = a
+
+    def mask_count(self, hm, n):
+        self.calls.append("mask_count")
+        return int(self.bufs[hm].view(np.uint8).reshape(-1)[:n].sum())
+
     def binary_bcast(self, ha, hb, bop, dtype, outer, inner, mode):
         self.calls.append("binary_bcast")
         dt = self._dt(dtype)
@@ -462,3 +496,60 @@ def test_broadcast_binary_on_device(fake):
     mid = _device(fake, host.mean(axis=1))  # shape (12,)
     with pytest.raises((ValueError, TypeError)):
         _ = x - mid
+
+
+def test_device_masks_and_where(fake):
+    host = np.random.default_rng(23).normal(0, 1, 512)
+    x = _device(fake, host)
+    m = x < 0
+    assert isinstance(m, hipnp.BoolDeviceArray)
+    np.testing.assert_array_equal(np.asarray(m), host < 0)
+    assert int(m.sum()) == int((host < 0).sum())
+    assert int(np.count_nonzero(m)) == int((host < 0).sum())
+    assert m.any() == (host < 0).any()
+    assert m.all() == (host < 0).all()
+    # np.where: scalar/scalar, array/scalar, array/array
+    r1 = np.where(m, 1.0, 0.0)
+    assert isinstance(r1, hipnp.DeviceArray)
+    np.testing.assert_array_equal(np.asarray(r1), np.where(host < 0, 1.0, 0.0))
+    y = _device(fake, host * 2)
+    r2 = np.where(m, y, 0.5)
+    np.testing.assert_allclose(np.asarray(r2), np.where(host < 0, host * 2, 0.5))
+    r3 = np.where(m, x, y)
+    np.testing.assert_allclose(np.asarray(r3), np.where(host < 0, host, host * 2))
+    assert "compare" in fake.calls and "where" in fake.calls
+
+
+def test_masked_assignment_on_device(fake):
+    host = np.random.default_rng(24).normal(0, 1, 256)
+    x = _device(fake, host.copy())
+    x[x < 0] = 0.0  # the whole idiom stays on device
+    assert "masked_fill" in fake.calls
+    expect = host.copy()
+    expect[expect < 0] = 0.0
+    np.testing.assert_array_equal(np.asarray(x), expect)
+    # follow-on device compute sees the mutation
+    assert float(x.min()) >= 0.0
+    # host numpy bool mask against a device array also routes
+    fake.calls.clear()
+    x2 = _device(fake, host.copy())
+    x2[host > 0.5] = 9.0
+    assert "masked_fill" in fake.calls
+    expect2 = host.copy()
+    expect2[host > 0.5] = 9.0
+    np.testing.assert_array_equal(np.asarray(x2), expect2)
+
+
+def test_mask_comparison_chain_preserves_semantics(fake):
+    host = np.array([1.0, np.nan, -3.0, 0.0])
+    x = _device(fake, host)
+    with np.errstate(invalid="ignore"):
+        for op, ref in [
+            (x > 0, host > 0), (x <= 0, host <= 0), (x == 0, host == 0),
+            (x != 0, host != 0),
+        ]:
+            np.testing.assert_array_equal(np.asarray(op), ref)
+    # equality against an incompatible shape falls back to numpy, which
+    # raises exactly as it would for host arrays (numpy 2.x semantics)
+    with pytest.raises(ValueError):
+        _ = x == np.array([1.0, 2.0])

@@ -649,3 +649,46 @@ def test_device_center_normalize_chain(hnp):
     np.testing.assert_allclose(
         np.asarray(centered), host - host.mean(axis=0), rtol=1e-9, atol=1e-12
     )
+
+
+def test_mask_ops_match_numpy(hip):
+    rng = np.random.default_rng(50)
+    a = rng.standard_normal(2_000_003)
+    b = rng.standard_normal(2_000_003)
+    ha, hb = hip.upload(a), hip.upload(b)
+    # scalar compare + popcount
+    hm = hip.compare(ha, 1, a.size, 0, 0, 0.0)  # a < 0
+    assert hip.mask_count(hm, a.size) == int((a < 0).sum())
+    # where(mask, b, scalar)
+    hw = hip.where(hm, 1, a.size, hb, 0.0, 0, 7.5)
+    out = np.empty_like(a)
+    hip.download(hw, out)
+    np.testing.assert_array_equal(out, np.where(a < 0, b, 7.5))
+    hip.free(hw)
+    # array-vs-array compare
+    hm2 = hip.compare(ha, 1, a.size, 2, hb, 0.0)  # a > b
+    assert hip.mask_count(hm2, a.size) == int((a > b).sum())
+    # masked fill in place
+    hip.masked_fill(ha, hm, 1, a.size, 0.0)
+    hip.download(ha, out)
+    expect = a.copy()
+    expect[a < 0] = 0.0
+    np.testing.assert_array_equal(out, expect)
+    for h in (ha, hb, hm, hm2):
+        hip.free(h)
+
+
+def test_device_threshold_idiom_end_to_end(hnp):
+    # x[x < 0.2] = 0 then count + sum without any host round-trip
+    x = hnp.rand(8_000_000, seed=77)
+    m = x < 0.2
+    assert isinstance(m, hnp.BoolDeviceArray)
+    n_low = int(m.sum())
+    x[m] = 0.0
+    assert x._host is None  # still device-resident
+    host = np.asarray(x)
+    assert int((host == 0.0).sum()) >= n_low  # filled (plus exact zeros)
+    assert abs(n_low / x.size - 0.2) < 0.01
+    y = np.where(x > 0.5, x, 0.0)
+    assert isinstance(y, hnp.DeviceArray)
+    np.testing.assert_allclose(np.asarray(y), np.where(host > 0.5, host, 0.0))
