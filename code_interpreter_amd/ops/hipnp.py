@@ -1391,23 +1391,41 @@ def install(numpy_module, mode: str = "auto") -> None:
                     raise
         return orig_dot(_asarray(a), _asarray(b), *args, **kwargs)
 
-    def patched_square(x, *args, **kwargs):
-        if not args and not kwargs:
-            if isinstance(x, DeviceArray):
-                return x._unary("square")
-            arr = x if isinstance(x, _np.ndarray) else None
-            if (
-                arr is not None
-                and arr.size >= MIN_ELEMS
-                and _dtype_code(arr.dtype) is not None
-                and arr.flags.c_contiguous
-            ):
-                try:
-                    return square(arr)
-                except Exception:
-                    if mode == "require":
-                        raise
-        return orig_square(_asarray(x), *args, **kwargs)
+    def _make_patched_unary(opname, orig):
+        """Module-level unary routing (np.exp/log/sqrt/... on large host
+        arrays): upload at PCIe rate + one kernel beats host
+        transcendental loops ~20-50x at 1e8 elements; DeviceArray inputs
+        stay resident. Same passthrough pattern as patched_square."""
+
+        def patched(x, *args, **kwargs):
+            if not args and not kwargs:
+                if isinstance(x, DeviceArray) and not isinstance(
+                    x, BoolDeviceArray
+                ):
+                    try:
+                        return x._unary(opname)
+                    except Exception:
+                        if mode == "require":
+                            raise
+                arr = x if isinstance(x, _np.ndarray) else None
+                if (
+                    arr is not None
+                    and arr.size >= MIN_ELEMS
+                    and _dtype_code(arr.dtype) is not None
+                    and arr.flags.c_contiguous
+                ):
+                    try:
+                        d = _to_device(arr)
+                        if d is not None:
+                            return d._unary(opname)
+                    except Exception:
+                        if mode == "require":
+                            raise
+            return orig(_asarray(x), *args, **kwargs)
+
+        return patched
+
+    patched_square = _make_patched_unary("square", orig_square)
 
     def patched_sum(x, *args, **kwargs):
         if not args and (not kwargs or set(kwargs) <= {"axis"}) and kwargs.get("axis") is None:
@@ -1479,4 +1497,15 @@ def install(numpy_module, mode: str = "auto") -> None:
     np.dot = patched_dot
     np.square = patched_square
     np.sum = patched_sum
+    # the rest of the hot unary surface, same pattern (np.abs is an
+    # alias of np.absolute; both get the patch)
+    for _uname, _npname in (
+        ("sqrt", "sqrt"), ("exp", "exp"), ("log", "log"), ("sin", "sin"),
+        ("cos", "cos"), ("tanh", "tanh"), ("absolute", "absolute"),
+    ):
+        _orig = getattr(np, _npname)
+        _patched = _make_patched_unary(_uname, _orig)
+        setattr(np, _npname, _patched)
+        if _npname == "absolute":
+            np.abs = _patched
     _installed["done"] = True

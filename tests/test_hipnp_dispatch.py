@@ -553,3 +553,39 @@ def test_mask_comparison_chain_preserves_semantics(fake):
     # raises exactly as it would for host arrays (numpy 2.x semantics)
     with pytest.raises(ValueError):
         _ = x == np.array([1.0, 2.0])
+
+
+def test_module_level_unary_routing(fake, monkeypatch):
+    import types
+
+    np_mod = types.SimpleNamespace(**{
+        name: getattr(np, name) for name in (
+            "sum", "square", "matmul", "dot", "sqrt", "exp", "log", "sin",
+            "cos", "tanh", "absolute", "abs",
+        )
+    })
+    np_mod.random = types.SimpleNamespace(
+        rand=np.random.rand, random=np.random.random,
+        random_sample=np.random.random_sample, uniform=np.random.uniform,
+        randn=np.random.randn, standard_normal=np.random.standard_normal,
+        normal=np.random.normal,
+    )
+    monkeypatch.setitem(hipnp._installed, "done", False)
+    monkeypatch.setattr(hipnp, "MIN_ELEMS", 64)
+    monkeypatch.setattr(hipnp, "available", lambda: True)
+    hipnp.install(np_mod, mode="auto")
+    host = np.random.default_rng(30).random(256) + 0.5
+    for name in ("sqrt", "exp", "log", "sin", "cos", "tanh"):
+        r = getattr(np_mod, name)(host)
+        assert isinstance(r, hipnp.DeviceArray), name
+        np.testing.assert_allclose(
+            np.asarray(r), getattr(np, name)(host), rtol=1e-12)
+    r = np_mod.abs(-host)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(np.asarray(r), host)
+    # small arrays stay on host; kwargs pass through untouched
+    small = np.random.default_rng(31).random(8)
+    assert isinstance(np_mod.exp(small), np.ndarray)
+    out = np.empty(256)
+    got = np_mod.exp(host, out=out)
+    assert got is out
