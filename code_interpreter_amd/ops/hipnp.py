@@ -1021,27 +1021,38 @@ class DeviceArray:
         ]
         return func(*host_args, **kwargs)
 
-    # -- operators -------------------------------------------------------
+    # -- operators: device kernel when routable, otherwise numpy's own
+    # semantics on the host copies (dtype upcasts, broadcasting beyond
+    # the device cases, bool operands, ... must all behave exactly like
+    # CPU numpy -- raising here would break valid user code)
+    def _op_or_host(self, opname, o, host_op, reverse=False):
+        r = self._binary(opname, o, reverse=reverse)
+        if r is not NotImplemented:
+            return r
+        a = self.materialize()
+        b = _asarray_or_scalar(o)
+        return host_op(b, a) if reverse else host_op(a, b)
+
     def __add__(self, o):
-        return self._coerce(self._binary("add", o))
+        return self._op_or_host("add", o, lambda a, b: a + b)
 
     def __radd__(self, o):
-        return self._coerce(self._binary("add", o, reverse=True))
+        return self._op_or_host("add", o, lambda a, b: a + b, reverse=True)
 
     def __sub__(self, o):
-        return self._coerce(self._binary("subtract", o))
+        return self._op_or_host("subtract", o, lambda a, b: a - b)
 
     def __mul__(self, o):
-        return self._coerce(self._binary("multiply", o))
+        return self._op_or_host("multiply", o, lambda a, b: a * b)
 
     def __rmul__(self, o):
-        return self._coerce(self._binary("multiply", o, reverse=True))
+        return self._op_or_host("multiply", o, lambda a, b: a * b, reverse=True)
 
     def __truediv__(self, o):
-        return self._coerce(self._binary("divide", o))
+        return self._op_or_host("divide", o, lambda a, b: a / b)
 
     def __pow__(self, o):
-        return self._coerce(self._binary("power", o))
+        return self._op_or_host("power", o, lambda a, b: a ** b)
 
     # scalar-first sub/div/pow have no device kernel ordering: compute on
     # the host instead of raising (user code does `1.0 / x` freely)
@@ -1058,10 +1069,7 @@ class DeviceArray:
         r = matmul(self, o, _force=True)
         return self.materialize() @ _asarray(o) if r is NotImplemented else r
 
-    def _coerce(self, r):
-        if r is NotImplemented:
-            raise TypeError("unsupported operand for device array")
-        return r
+
 
 
 class BoolDeviceArray(DeviceArray):

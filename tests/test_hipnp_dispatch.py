@@ -589,3 +589,20 @@ def test_module_level_unary_routing(fake, monkeypatch):
     out = np.empty(256)
     got = np_mod.exp(host, out=out)
     assert got is out
+
+
+def test_mixed_dtype_and_mask_arithmetic_falls_back(fake):
+    host = np.random.default_rng(32).random(64)
+    x = _device(fake, host)                         # f64
+    y32 = _device(fake, host.astype(np.float32))    # f32
+    # dtype-mismatched device pair: numpy's upcast semantics, not a raise
+    r = x * y32
+    np.testing.assert_allclose(r, host * host.astype(np.float32))
+    # mask arithmetic: bool participates like numpy (0/1 upcast)
+    m = x > 0.5
+    r2 = x * m
+    np.testing.assert_allclose(np.asarray(r2), host * (host > 0.5))
+    r3 = m * 2.5
+    np.testing.assert_allclose(np.asarray(r3), (host > 0.5) * 2.5)
+    r4 = x + m
+    np.testing.assert_allclose(np.asarray(r4), host + (host > 0.5))
