@@ -1282,6 +1282,8 @@ def matmul(a, b, _force=False):
     if da is None or db is None or da.dtype != db.dtype:
         return NotImplemented
     code = _dtype_code(da.dtype)
+    if code is None:  # e.g. boolean matmul: numpy semantics on host
+        return NotImplemented
     hc = backend().gemm(da._dev_handle(), db._dev_handle(), m, n, k, code)
     return DeviceArray(hc, (m, n), da.dtype)
 
@@ -1299,6 +1301,8 @@ def _matmul_batched(a, b, _force=False):
     if da is None or db is None or da.dtype != db.dtype:
         return NotImplemented
     code = _dtype_code(da.dtype)
+    if code is None:
+        return NotImplemented
     hc = backend().gemm_batched(
         da._dev_handle(), db._dev_handle(), batch, m, n, k, code
     )

@@ -606,3 +606,12 @@ def test_mixed_dtype_and_mask_arithmetic_falls_back(fake):
     np.testing.assert_allclose(np.asarray(r3), (host > 0.5) * 2.5)
     r4 = x + m
     np.testing.assert_allclose(np.asarray(r4), host + (host > 0.5))
+
+
+def test_bool_matmul_falls_back(fake):
+    host = np.random.default_rng(34).random((16, 16))
+    m = _device(fake, host) > 0.5
+    r = hipnp.matmul(m, m, _force=True)
+    assert r is NotImplemented  # caller (patched np.matmul) goes to host
+    got = (m @ m)
+    np.testing.assert_array_equal(np.asarray(got), (host > 0.5) @ (host > 0.5))
