@@ -5,13 +5,18 @@ Two mechanisms, installed by the sandbox runtime (executor/sandbox_runtime
 
 1. patched module-level entry points (numpy.random.{rand, random,
    random_sample, uniform, randn, standard_normal, normal}, numpy.matmul,
-   numpy.dot, numpy.square, numpy.sum): above a size threshold the work
-   runs on the MI355X through _hipops and the result stays device-resident;
+   numpy.dot, numpy.square, numpy.sum, numpy.{sqrt,exp,log,sin,cos,tanh,
+   abs}): above a size threshold the work runs on the MI355X through
+   _hipops and the result stays device-resident;
 2. DeviceArray: a duck array (NEP 13/18 __array_ufunc__ +
    __array_function__) so follow-on numpy calls on a device-resident
-   result keep running on the GPU (numpy.sum(numpy.square(x)) never
-   round-trips the intermediate through host memory); any unsupported
-   operation transparently materializes to a host ndarray and falls back.
+   result keep running on the GPU: elementwise unary/binary (incl.
+   row/column broadcasting), full and axis-wise reductions, argmax/
+   argmin, clip, matmul (2-D / equal-batch 3-D / 1-D dot / pad-to-256
+   bf16), device boolean masks (comparisons, np.where, x[mask] = v,
+   popcount), in-place mutation and ufunc out=. Any unsupported
+   operation transparently materializes to a host ndarray and computes
+   numpy's own result -- identical values either way.
 
 Everything computes the same values user code would get on the CPU (same
 dtype; RNG is Philox instead of MT19937 -- a documented backend change,
