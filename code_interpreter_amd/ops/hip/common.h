@@ -49,6 +49,14 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// histogram selection (median/percentile): range histogram (counts:
+// bins u64 + extra[3] = {nan, below, above}) and range compaction
+void launch_hist_range(DType dt, const void* in, int64_t n, double lo,
+                       double hi, double inv_width, int bins, void* counts,
+                       void* extra, hipStream_t stream);
+void launch_extract_range(DType dt, const void* in, int64_t n, double lo,
+                          double hi, void* out, void* counter, int64_t cap,
+                          hipStream_t stream);
 // boolean masks: compare -> u8, np.where select, masked fill, popcount
 void launch_compare(DType dt, int op, const void* a, const void* b,
                     double scalar, void* out_u8, int64_t n, hipStream_t s);

@@ -575,6 +575,82 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// histogram(h, dtype, n, lo, hi, bins) -> bytes((bins+3) * u64):
+// [bin counts..., nan_count, below_count, above_count]
+PyObject* py_histogram(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt, bins;
+  long long n;
+  double lo, hi;
+  if (!PyArg_ParseTuple(args, "KiLddi", &h, &dt, &n, &lo, &hi, &bins))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (bins < 1 || bins > 4096) throw std::runtime_error("bad bin count");
+  if (!(hi > lo)) throw std::runtime_error("bad histogram range");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  std::vector<unsigned long long> host((size_t)bins + 3, 0);
+  double inv_width = (double)bins / (hi - lo);
+  NOGIL_BEGIN
+  void* counts = pool_alloc((int64_t)(bins + 3) * 8);
+  HIP_CHECK(hipMemsetAsync(counts, 0, (int64_t)(bins + 3) * 8, g.compute));
+  void* extra = (char*)counts + (int64_t)bins * 8;
+  launch_hist_range(dtype, in.ptr, n, lo, hi, inv_width, bins, counts, extra,
+                    g.compute);
+  HIP_CHECK(hipMemcpyAsync(host.data(), counts, (int64_t)(bins + 3) * 8,
+                           hipMemcpyDeviceToHost, g.compute));
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  pool_free(counts, (int64_t)(bins + 3) * 8);
+  NOGIL_END
+  return PyBytes_FromStringAndSize((const char*)host.data(),
+                                   (Py_ssize_t)((bins + 3) * 8));
+  WRAP_END
+}
+
+// extract_range(h, dtype, n, lo, hi, cap) -> (count, bytes(f64 values))
+PyObject* py_extract_range(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt;
+  long long n, cap;
+  double lo, hi;
+  if (!PyArg_ParseTuple(args, "KiLddL", &h, &dt, &n, &lo, &hi, &cap))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (cap < 1 || cap > (1 << 22)) throw std::runtime_error("bad cap");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  unsigned long long found = 0;
+  std::vector<double> host;
+  NOGIL_BEGIN
+  void* out = pool_alloc(cap * 8 + 8);
+  void* counter = (char*)out + cap * 8;
+  HIP_CHECK(hipMemsetAsync(counter, 0, 8, g.compute));
+  launch_extract_range(dtype, in.ptr, n, lo, hi, out, counter, cap,
+                       g.compute);
+  HIP_CHECK(hipMemcpyAsync(&found, counter, 8, hipMemcpyDeviceToHost,
+                           g.compute));
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  unsigned long long take = found < (unsigned long long)cap
+                                ? found
+                                : (unsigned long long)cap;
+  host.resize((size_t)take);
+  if (take) {
+    HIP_CHECK(hipMemcpyAsync(host.data(), out, (int64_t)take * 8,
+                             hipMemcpyDeviceToHost, g.compute));
+    HIP_CHECK(hipStreamSynchronize(g.compute));
+  }
+  pool_free(out, cap * 8 + 8);
+  NOGIL_END
+  PyObject* bytes = PyBytes_FromStringAndSize(
+      (const char*)host.data(), (Py_ssize_t)(host.size() * 8));
+  if (!bytes) return nullptr;
+  PyObject* tuple = Py_BuildValue("(KN)", found, bytes);
+  return tuple;
+  WRAP_END
+}
+
 // compare(h, dtype, n, cmp_op, hb_or_0, scalar) -> u8 mask handle
 PyObject* py_compare(PyObject*, PyObject* args) {
   unsigned long long h, hb;
@@ -945,6 +1021,10 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"histogram", py_histogram, METH_VARARGS,
+     "histogram(h, dtype, n, lo, hi, bins) -> bytes of u64 counts"},
+    {"extract_range", py_extract_range, METH_VARARGS,
+     "extract_range(h, dtype, n, lo, hi, cap) -> (count, f64 bytes)"},
     {"compare", py_compare, METH_VARARGS,
      "compare(h, dtype, n, cmp_op, hb_or_0, scalar) -> u8 mask handle"},
     {"where", py_where, METH_VARARGS,

@@ -262,6 +262,66 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// histogram-selection primitives (np.median/percentile without a sort):
+// range histogram with per-block LDS accumulation, and range-compaction
+// extraction for the exact finish on a narrowed candidate bin.
+// ---------------------------------------------------------------------------
+constexpr int kHistBins = 4096;
+
+template <typename T>
+__global__ void hist_range_kernel(const T* __restrict__ in, int64_t n,
+                                  double lo, double hi, double inv_width,
+                                  int bins,
+                                  unsigned long long* __restrict__ counts,
+                                  unsigned long long* __restrict__ extra) {
+  // extra[0] = NaN count, extra[1] = count below lo, extra[2] = above hi
+  __shared__ unsigned int local[kHistBins];
+  for (int i = threadIdx.x; i < bins; i += blockDim.x) local[i] = 0;
+  __syncthreads();
+  unsigned long long nan_c = 0, below_c = 0, above_c = 0;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    double v = (double)in[i];
+    if (v != v) {
+      nan_c++;
+    } else if (v < lo) {
+      below_c++;
+    } else if (v > hi) {
+      above_c++;
+    } else {
+      int b = (int)((v - lo) * inv_width);
+      if (b >= bins) b = bins - 1;  // v == hi (or fp rounding at the edge)
+      if (b < 0) b = 0;
+      atomicAdd(&local[b], 1u);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < bins; i += blockDim.x)
+    if (local[i]) atomicAdd(&counts[i], (unsigned long long)local[i]);
+  if (nan_c) atomicAdd(&extra[0], nan_c);
+  if (below_c) atomicAdd(&extra[1], below_c);
+  if (above_c) atomicAdd(&extra[2], above_c);
+}
+
+template <typename T>
+__global__ void extract_range_kernel(const T* __restrict__ in, int64_t n,
+                                     double lo, double hi,
+                                     double* __restrict__ out,
+                                     unsigned long long* __restrict__ counter,
+                                     int64_t cap) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    double v = (double)in[i];
+    if (v >= lo && v <= hi) {
+      unsigned long long idx = atomicAdd(counter, 1ull);
+      if ((int64_t)idx < cap) out[idx] = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // boolean-mask ops: comparisons -> u8 masks, select (np.where),
 // masked fill, mask popcount. CmpOp: 0 lt, 1 le, 2 gt, 3 ge, 4 eq, 5 ne
 // (NaN compares false like numpy, except ne where NaN != x is true).
@@ -749,6 +809,38 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
   else
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_hist_range(DType dt, const void* in, int64_t n, double lo,
+                       double hi, double inv_width, int bins, void* counts,
+                       void* extra, hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  if (dt == DType::F64)
+    hipLaunchKernelGGL((hist_range_kernel<double>), dim3(grid), dim3(kBlock),
+                       0, s, (const double*)in, n, lo, hi, inv_width, bins,
+                       (unsigned long long*)counts,
+                       (unsigned long long*)extra);
+  else
+    hipLaunchKernelGGL((hist_range_kernel<float>), dim3(grid), dim3(kBlock),
+                       0, s, (const float*)in, n, lo, hi, inv_width, bins,
+                       (unsigned long long*)counts,
+                       (unsigned long long*)extra);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_extract_range(DType dt, const void* in, int64_t n, double lo,
+                          double hi, void* out, void* counter, int64_t cap,
+                          hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  if (dt == DType::F64)
+    hipLaunchKernelGGL((extract_range_kernel<double>), dim3(grid),
+                       dim3(kBlock), 0, s, (const double*)in, n, lo, hi,
+                       (double*)out, (unsigned long long*)counter, cap);
+  else
+    hipLaunchKernelGGL((extract_range_kernel<float>), dim3(grid),
+                       dim3(kBlock), 0, s, (const float*)in, n, lo, hi,
+                       (double*)out, (unsigned long long*)counter, cap);
   HIP_CHECK(hipGetLastError());
 }
 

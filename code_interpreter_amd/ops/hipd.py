@@ -183,6 +183,18 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "histogram":
+            data = _hipops.histogram(
+                self._own(m["h"]), m["dtype"], m["n"], m["lo"], m["hi"],
+                m["bins"]
+            )
+            return {"ok": True}, data
+        if op == "extract_range":
+            count, data = _hipops.extract_range(
+                self._own(m["h"]), m["dtype"], m["n"], m["lo"], m["hi"],
+                m["cap"]
+            )
+            return {"ok": True, "count": count}, data
         if op == "compare":
             hb = m.get("hb", 0)
             if hb:

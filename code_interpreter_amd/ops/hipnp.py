@@ -313,6 +313,20 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def histogram(self, h, dtype, n, lo, hi, bins):
+        out = self._call(
+            {"op": "histogram", "h": h, "dtype": dtype, "n": n, "lo": lo,
+             "hi": hi, "bins": bins}
+        )
+        return out["_payload"]
+
+    def extract_range(self, h, dtype, n, lo, hi, cap):
+        out = self._call(
+            {"op": "extract_range", "h": h, "dtype": dtype, "n": n,
+             "lo": lo, "hi": hi, "cap": cap}
+        )
+        return out["count"], out.get("_payload", b"")
+
     def compare(self, h, dtype, n, cmp, hb, scalar):
         return self._call(
             {"op": "compare", "h": h, "dtype": dtype, "n": n, "cmp": cmp,
@@ -986,6 +1000,17 @@ class DeviceArray:
                 return r
         if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
             return args[0]._unary("square")
+        if func is _np.median and len(args) == 1 and not kwargs:
+            r = quantile_device(args[0], 0.5)
+            if r is not None:
+                return r
+        if func in (_np.quantile, _np.percentile) and len(args) == 2 and not kwargs:
+            qv = args[1]
+            if isinstance(qv, (int, float)):
+                q = qv / 100.0 if func is _np.percentile else float(qv)
+                r = quantile_device(args[0], q)
+                if r is not None:
+                    return r
         if func is _np.where and len(args) == 3 and not kwargs:
             r = where_device(*args)
             if r is not NotImplemented:
@@ -1133,6 +1158,94 @@ class BoolDeviceArray(DeviceArray):
 
     def astype(self, dtype, **kwargs):
         return self.materialize().astype(dtype, **kwargs)
+
+
+_QUANTILE_BINS = 4096
+_QUANTILE_EXTRACT_CAP = 1 << 20  # 8 MB of f64 candidates, max
+
+
+def _order_stat_device(x, k):
+    """k-th (0-based) order statistic of a DeviceArray via histogram
+    bisection + a final exact extraction. Returns (value, nan_count)."""
+    code = _dtype_code(x.dtype)
+    n = x.size
+    lo = float(x.min())
+    hi = float(x.max())
+    if lo != lo or hi != hi:  # min/max NaN-propagate: NaNs present
+        return float("nan"), 1
+
+    below = 0
+    for _ in range(64):
+        if not hi > lo:
+            return lo, 0
+        counts = _np.frombuffer(
+            backend().histogram(
+                x._dev_handle(), code, n, lo, hi, _QUANTILE_BINS
+            ),
+            dtype=_np.uint64,
+        )
+        nan_count = int(counts[_QUANTILE_BINS])
+        if nan_count:
+            return float("nan"), nan_count
+        width = (hi - lo) / _QUANTILE_BINS
+        cs = counts[:_QUANTILE_BINS].astype(_np.int64).cumsum()
+        rel_k = k - below
+        target_bin = int(_np.searchsorted(cs, rel_k, side="right"))
+        if target_bin >= _QUANTILE_BINS:
+            return hi, 0  # k at the far edge (fp rounding): max wins
+        cum = below + (int(cs[target_bin - 1]) if target_bin else 0)
+        in_bin = int(counts[target_bin])
+        new_lo = lo + target_bin * width
+        new_hi = (
+            lo + (target_bin + 1) * width
+            if target_bin + 1 < _QUANTILE_BINS
+            else hi
+        )
+        if in_bin <= _QUANTILE_EXTRACT_CAP:
+            _, data = backend().extract_range(
+                x._dev_handle(), code, n, new_lo, new_hi,
+                _QUANTILE_EXTRACT_CAP,
+            )
+            vals = _np.sort(_np.frombuffer(data, dtype=_np.float64))
+            if len(vals) == 0:
+                return new_lo, 0
+            idx = min(max(k - cum, 0), len(vals) - 1)
+            return float(vals[idx]), 0
+        below = cum
+        lo, hi = new_lo, new_hi
+    return float("nan"), 0  # did not converge: caller falls back
+
+
+def quantile_device(x, q):
+    """numpy-linear-interpolation quantile of a DeviceArray (axis=None).
+    Returns None when not routable (caller falls back to host)."""
+    if not isinstance(x, DeviceArray) or isinstance(x, BoolDeviceArray):
+        return None
+    if _dtype_code(x.dtype) is None or x.size == 0:
+        return None
+    try:
+        q = float(q)
+    except (TypeError, ValueError):
+        return None
+    if not 0.0 <= q <= 1.0:
+        return None
+    h = (x.size - 1) * q
+    k0 = int(_np.floor(h))
+    k1 = int(_np.ceil(h))
+    v0, nan_c = _order_stat_device(x, k0)
+    if nan_c:
+        return x.dtype.type("nan")
+    if v0 != v0:  # non-convergence sentinel without NaNs: fall back
+        return None
+    if k1 == k0:
+        return x.dtype.type(v0)
+    v1, nan_c = _order_stat_device(x, k1)
+    if nan_c:
+        return _np.float64("nan")
+    if v1 != v1:
+        return None
+    # numpy returns the input's dtype for median/quantile of floats
+    return x.dtype.type(v0 + (h - k0) * (v1 - v0))
 
 
 def where_device(mask, a, b):

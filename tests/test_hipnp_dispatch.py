@@ -100,6 +100,28 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def histogram(self, h, dtype, n, lo, hi, bins):
+        self.calls.append("histogram")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n].astype(np.float64)
+        nan_c = int(np.isnan(a).sum())
+        valid = a[~np.isnan(a)]
+        below = int((valid < lo).sum())
+        above = int((valid > hi).sum())
+        inr = valid[(valid >= lo) & (valid <= hi)]
+        b = np.floor((inr - lo) * (bins / (hi - lo))).astype(np.int64)
+        b = np.clip(b, 0, bins - 1)
+        counts = np.bincount(b, minlength=bins).astype(np.uint64)
+        out = np.concatenate(
+            [counts, np.array([nan_c, below, above], dtype=np.uint64)]
+        )
+        return out.tobytes()
+
+    def extract_range(self, h, dtype, n, lo, hi, cap):
+        self.calls.append("extract_range")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n].astype(np.float64)
+        sel = a[(a >= lo) & (a <= hi)]
+        return len(sel), sel[:cap].tobytes()
+
     _CMPS = {
         0: np.less, 1: np.less_equal, 2: np.greater, 3: np.greater_equal,
         4: np.equal, 5: np.not_equal,
@@ -615,3 +637,37 @@ def test_bool_matmul_falls_back(fake):
     assert r is NotImplemented  # caller (patched np.matmul) goes to host
     got = (m @ m)
     np.testing.assert_array_equal(np.asarray(got), (host > 0.5) @ (host > 0.5))
+
+
+def test_median_percentile_on_device(fake):
+    rng = np.random.default_rng(35)
+    host = rng.normal(0, 3, 100_001)
+    x = _device(fake, host)
+    med = np.median(x)
+    assert med == pytest.approx(np.median(host), rel=0, abs=1e-12)
+    assert "histogram" in fake.calls
+    for q in (0.0, 0.1, 0.25, 0.733, 1.0):
+        got = np.quantile(x, q)
+        assert got == pytest.approx(np.quantile(host, q), rel=0, abs=1e-9), q
+    assert np.percentile(x, 90.0) == pytest.approx(
+        np.percentile(host, 90.0), abs=1e-9)
+    # even length: interpolation between two order stats
+    host2 = rng.random(1000)
+    assert np.median(_device(fake, host2)) == pytest.approx(
+        np.median(host2), abs=1e-12)
+
+
+def test_median_duplicates_and_nan(fake):
+    # duplicate-heavy: the candidate bin never shrinks by narrowing;
+    # extraction must finish it exactly
+    host = np.repeat(np.array([1.0, 2.0, 2.0, 2.0, 9.0]), 5000)
+    x = _device(fake, host)
+    assert float(np.median(x)) == float(np.median(host)) == 2.0
+    # constant array
+    c = _device(fake, np.full(777, 4.25))
+    assert float(np.median(c)) == 4.25
+    # NaN present: numpy returns nan
+    host3 = np.random.default_rng(36).random(512)
+    host3[100] = np.nan
+    with np.errstate(invalid="ignore"):
+        assert np.isnan(np.median(_device(fake, host3)))

@@ -692,3 +692,37 @@ def test_device_threshold_idiom_end_to_end(hnp):
     y = np.where(x > 0.5, x, 0.0)
     assert isinstance(y, hnp.DeviceArray)
     np.testing.assert_allclose(np.asarray(y), np.where(host > 0.5, host, 0.0))
+
+
+def test_median_quantile_device_matches_numpy(hnp):
+    x = hnp.rand(10_000_001, seed=91)
+    host = np.asarray(x)
+    assert float(np.median(x)) == float(np.median(host))
+    for q in (0.01, 0.5, 0.999):
+        assert float(np.quantile(x, q)) == pytest.approx(
+            float(np.quantile(host, q)), rel=0, abs=1e-12), q
+    assert float(np.percentile(x, 75.0)) == pytest.approx(
+        float(np.percentile(host, 75.0)), rel=0, abs=1e-12)
+
+
+def test_histogram_extract_primitives(hip):
+    rng = np.random.default_rng(92)
+    a = rng.normal(0, 1, 3_000_001)
+    h = hip.upload(a)
+    counts = np.frombuffer(
+        hip.histogram(h, 1, a.size, -1.0, 1.0, 1000), dtype=np.uint64
+    )
+    ref, _ = np.histogram(a, bins=1000, range=(-1.0, 1.0))
+    # edge-bin semantics differ one ulp at boundaries; totals must agree
+    assert int(counts[:1000].sum()) == int(((a >= -1) & (a <= 1)).sum())
+    np.testing.assert_allclose(
+        counts[:1000].astype(np.int64), ref, atol=3
+    )
+    assert int(counts[1001]) == int((a < -1).sum())  # below
+    assert int(counts[1002]) == int((a > 1).sum())   # above
+    found, data = hip.extract_range(h, 1, a.size, 0.5, 0.6, 1 << 20)
+    vals = np.frombuffer(data, dtype=np.float64)
+    ref_vals = a[(a >= 0.5) & (a <= 0.6)]
+    assert found == len(ref_vals)
+    np.testing.assert_allclose(np.sort(vals), np.sort(ref_vals))
+    hip.free(h)
