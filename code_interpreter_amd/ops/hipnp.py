@@ -914,7 +914,9 @@ class DeviceArray:
     def var(self, axis=None, ddof=0, **kwargs):
         """Two-pass variance entirely on-device: mean, then the fused
         sum((x-mean)^2). (The one-pass sum-of-squares form cancels
-        catastrophically when |mean| >> std and can go negative.)"""
+        catastrophically when |mean| >> std and can go negative.)
+        2-D axis-wise variance composes the same way from the axis
+        reducers + broadcast subtract."""
         if axis is None and not kwargs.get("keepdims"):
             n = self.size
             mu = float(self.sum()) / n
@@ -922,11 +924,37 @@ class DeviceArray:
             if shifted is NotImplemented:
                 return self.materialize().var(ddof=ddof)
             return self.dtype.type(float(shifted.square_sum()) / (n - ddof))
+        if (
+            axis is not None
+            and len(self.shape) == 2
+            and not kwargs.get("keepdims")
+            and set(kwargs) <= {"keepdims"}
+            and _dtype_code(self.dtype) is not None
+        ):
+            ax = self._norm_axis(axis, 2)
+            if ax is not None:
+                red = self.shape[ax]
+                mu = self.mean(axis=ax, keepdims=(ax == 1))
+                if isinstance(mu, DeviceArray):
+                    centered = self._binary_bcast("subtract", mu)
+                    if centered is not NotImplemented:
+                        sq = centered._unary("square")
+                        total = sq.sum(axis=ax)
+                        if isinstance(total, DeviceArray):
+                            r = total._binary(
+                                "multiply", 1.0 / (red - ddof)
+                            )
+                            if r is not NotImplemented:
+                                return r
         return self.materialize().var(axis=axis, ddof=ddof, **kwargs)
 
     def std(self, axis=None, ddof=0, **kwargs):
         if axis is None and not kwargs.get("keepdims"):
             return self.dtype.type(float(self.var(ddof=ddof)) ** 0.5)
+        if axis is not None and not kwargs.get("keepdims"):
+            v = self.var(axis=axis, ddof=ddof, **kwargs)
+            if isinstance(v, DeviceArray):
+                return v._unary("sqrt")
         return self.materialize().std(axis=axis, ddof=ddof, **kwargs)
 
     # -- NEP 13: ufuncs --------------------------------------------------
@@ -1043,7 +1071,9 @@ class DeviceArray:
                     if meth in ("std", "var") and "ddof" in kwargs:
                         call_kwargs["ddof"] = kwargs["ddof"]
                     return getattr(args[0], meth)(**call_kwargs)
-                if meth in ("max", "min") and kwargs.get("axis") is not None:
+                if meth in ("max", "min", "std", "var") and kwargs.get(
+                    "axis"
+                ) is not None:
                     return getattr(args[0], meth)(**kwargs)
         # generic fallback: materialize every DeviceArray
         host_args = [
