@@ -108,3 +108,12 @@ def test_mixed_workload_under_load(http_client):
     with concurrent.futures.ThreadPoolExecutor(max_workers=8) as pool:
         done = list(pool.map(one, range(40)))
     assert len(done) == 40
+
+
+def test_analysis_example_cpu_fallback(http_client):
+    # the r02 showcase chain must give identical values on plain CPU
+    # numpy (no GPU in this env): the routing layer's fallback contract
+    out = _execute(http_client, "analysis-gpu.py")
+    assert out["exit_code"] == 0, out["stderr"]
+    assert "outliers:" in out["stdout"]
+    assert "kind: ndarray" in out["stdout"]

@@ -127,3 +127,18 @@ def test_torch_bmm_routed(hip_torch):
         assert err / (ref.abs().max().item() + 1e-9) < 2e-4
 
 
+
+
+def test_torch_bf16_nonaligned_pad_path(hip_torch):
+    torch, hiptorch = hip_torch
+    # 4000^3 misses the 256/128 divisibility: the routed path zero-pads
+    # to the fast kernel (bit-identical by construction)
+    a = torch.randn(4000, 4000, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(4000, 4000, dtype=torch.bfloat16, device="cuda")
+    before = hiptorch.STATS["mm_routed"]
+    c = a @ b
+    assert hiptorch.STATS["mm_routed"] == before + 1
+    assert c.shape == (4000, 4000)
+    ref = a.double().cpu() @ b.double().cpu()
+    err = (c.double().cpu() - ref).abs().max().item()
+    assert err / (ref.abs().max().item() + 1e-9) < 3e-2

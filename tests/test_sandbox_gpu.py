@@ -275,3 +275,16 @@ def test_torch_routed_in_sandbox(gpu_executor):
     r = _run(gpu_executor, code, env={"APP_HIP_TORCH": "require"})
     assert r.exit_code == 0, r.stderr
     assert "torch-ok" in r.stdout
+
+
+def test_analysis_example_stays_on_device(gpu_executor):
+    """examples/analysis-gpu.py through the real service: the chain's
+    intermediate must be a DeviceArray and the run completes."""
+    from pathlib import Path
+
+    src = (Path(__file__).resolve().parent.parent / "examples"
+           / "analysis-gpu.py").read_text()
+    r = _run(gpu_executor, src)
+    assert r.exit_code == 0, r.stderr
+    assert "kind: DeviceArray" in r.stdout
+    assert "outliers:" in r.stdout
