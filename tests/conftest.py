@@ -154,11 +154,12 @@ def gpu_executor(tmp_path_factory, executor_bin):
     """A live 1-GPU local executor pool (skipped when no GPU is visible)."""
     from code_interpreter_amd.services.local_executor import LocalPoolExecutor
     from code_interpreter_amd.services.storage import Storage
+    from code_interpreter_amd.utils.gpus import detect_gpu_count
 
-    sys.path.insert(0, str(REPO_ROOT / "code_interpreter_amd" / "ops"))
-    import _hipops
-
-    if not _hipops.is_available():
+    # sysfs probe, NOT _hipops.is_available(): initializing the HIP
+    # runtime in the pytest process before torch makes torch.cuda blind
+    # (NOTES.md binding rules) -- this fixture must stay HIP-free
+    if detect_gpu_count() == 0:
         pytest.skip("no AMD GPU visible")
 
     tmp = tmp_path_factory.mktemp("gpue")
