@@ -575,6 +575,26 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// mask_logic(ha, hb_or_0, n, op) -> u8 mask handle
+PyObject* py_mask_logic(PyObject*, PyObject* args) {
+  unsigned long long ha, hb;
+  long long n;
+  int op;
+  if (!PyArg_ParseTuple(args, "KKLi", &ha, &hb, &n, &op)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (op < 0 || op > 4) throw std::runtime_error("bad mask-logic op");
+  DevBuf& a = get_buf(ha);
+  void* bptr = hb ? get_buf(hb).ptr : nullptr;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n);
+  launch_mask_logic(a.ptr, bptr, out, n, op, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n));
+  WRAP_END
+}
+
 // histogram(h, dtype, n, lo, hi, bins) -> bytes((bins+3) * u64):
 // [bin counts..., nan_count, below_count, above_count]
 PyObject* py_histogram(PyObject*, PyObject* args) {
@@ -1021,6 +1041,8 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"mask_logic", py_mask_logic, METH_VARARGS,
+     "mask_logic(ha, hb_or_0, n, op) -> u8 handle (0 and 1 or 2 xor 3 andnot 4 not)"},
     {"histogram", py_histogram, METH_VARARGS,
      "histogram(h, dtype, n, lo, hi, bins) -> bytes of u64 counts"},
     {"extract_range", py_extract_range, METH_VARARGS,

@@ -262,6 +262,31 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// u8 mask logic: 0=and 1=or 2=xor 3=andnot(a & ~b); unary not via
+// xor with an all-ones operand is avoided -- NOT uses op 4 (b ignored)
+// ---------------------------------------------------------------------------
+__global__ void mask_logic_kernel(const unsigned char* __restrict__ a,
+                                  const unsigned char* __restrict__ b,
+                                  unsigned char* __restrict__ out, int64_t n,
+                                  int op) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    unsigned char av = a[i];
+    unsigned char bv = b ? b[i] : 0;
+    unsigned char r;
+    switch (op) {
+      case 0: r = (av && bv) ? 1 : 0; break;
+      case 1: r = (av || bv) ? 1 : 0; break;
+      case 2: r = (!av != !bv) ? 1 : 0; break;
+      case 3: r = (av && !bv) ? 1 : 0; break;
+      default: r = av ? 0 : 1; break;  // NOT
+    }
+    out[i] = r;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // histogram-selection primitives (np.median/percentile without a sort):
 // range histogram with per-block LDS accumulation, and range-compaction
 // extraction for the exact finish on a narrowed candidate bin.
@@ -809,6 +834,15 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
   else
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_mask_logic(const void* a, const void* b, void* out, int64_t n,
+                       int op, hipStream_t s) {
+  int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 4096);
+  hipLaunchKernelGGL(mask_logic_kernel, dim3(grid), dim3(kBlock), 0, s,
+                     (const unsigned char*)a, (const unsigned char*)b,
+                     (unsigned char*)out, n, op);
   HIP_CHECK(hipGetLastError());
 }
 

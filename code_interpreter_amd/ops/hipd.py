@@ -183,6 +183,13 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "mask_logic":
+            hb = m.get("hb", 0)
+            if hb:
+                self._own(hb)
+            h = _hipops.mask_logic(self._own(m["ha"]), hb, m["n"], m["lop"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "histogram":
             data = _hipops.histogram(
                 self._own(m["h"]), m["dtype"], m["n"], m["lo"], m["hi"],

@@ -313,6 +313,11 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def mask_logic(self, ha, hb, n, lop):
+        return self._call(
+            {"op": "mask_logic", "ha": ha, "hb": hb, "n": n, "lop": lop}
+        )["h"]
+
     def histogram(self, h, dtype, n, lo, hi, bins):
         out = self._call(
             {"op": "histogram", "h": h, "dtype": dtype, "n": n, "lo": lo,
@@ -893,6 +898,29 @@ class DeviceArray:
                 return r
         return self.materialize().clip(a_min, a_max, **kwargs)
 
+    def isnan(self):
+        """Device NaN mask (x != x elementwise)."""
+        r = self._compare("not_equal", self)
+        if r is None:
+            return _np.isnan(self.materialize())
+        return r
+
+    def count_nonzero(self):
+        m = self._compare("not_equal", 0.0)
+        if m is None:
+            return _np.count_nonzero(self.materialize())
+        return m.sum()
+
+    def any(self, axis=None, **kwargs):
+        if axis is None and not kwargs and _dtype_code(self.dtype) is not None:
+            return bool(int(self.count_nonzero()) > 0)
+        return self.materialize().any(axis=axis, **kwargs)
+
+    def all(self, axis=None, **kwargs):
+        if axis is None and not kwargs and _dtype_code(self.dtype) is not None:
+            return bool(int(self.count_nonzero()) == self.size)
+        return self.materialize().all(axis=axis, **kwargs)
+
     def argmax(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return _np.intp(
@@ -963,6 +991,27 @@ class DeviceArray:
             return self._fallback_ufunc(ufunc, method, inputs, kwargs)
         name = ufunc.__name__
         if method == "__call__":
+            if name == "isnan" and len(inputs) == 1 and inputs[0] is self:
+                if isinstance(self, BoolDeviceArray):
+                    return _np.zeros(self.shape, dtype=bool)
+                return self.isnan()
+            if name in (
+                "logical_and", "logical_or", "logical_xor",
+                "bitwise_and", "bitwise_or", "bitwise_xor",
+            ) and len(inputs) == 2 and isinstance(self, BoolDeviceArray):
+                other = inputs[1] if inputs[0] is self else inputs[0]
+                lop = {
+                    "logical_and": 0, "bitwise_and": 0,
+                    "logical_or": 1, "bitwise_or": 1,
+                    "logical_xor": 2, "bitwise_xor": 2,
+                }[name]
+                r = self._logic(other, lop)
+                if r is not NotImplemented:
+                    return r
+            if name == "logical_not" and len(inputs) == 1 and isinstance(
+                self, BoolDeviceArray
+            ):
+                return self._logic(None, 4)
             if name in _CMP and len(inputs) == 2 and inputs[0] is self:
                 r = self._compare(name, inputs[1])
                 if r is not None:
@@ -1044,10 +1093,18 @@ class DeviceArray:
             if r is not NotImplemented:
                 return r
         if func is _np.count_nonzero and len(args) == 1 and isinstance(
-            args[0], BoolDeviceArray
+            args[0], DeviceArray
         ):
             if not kwargs:
-                return args[0].sum()
+                if isinstance(args[0], BoolDeviceArray):
+                    return args[0].sum()
+                return args[0].count_nonzero()
+        if func in (_np.any, _np.all) and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ):
+            if not kwargs:
+                meth = "any" if func is _np.any else "all"
+                return getattr(args[0], meth)()
         if func is _np.clip and len(args) >= 1 and isinstance(args[0], DeviceArray):
             if len(args) <= 3 and not kwargs:
                 return args[0].clip(*args[1:])
@@ -1188,6 +1245,45 @@ class BoolDeviceArray(DeviceArray):
 
     def astype(self, dtype, **kwargs):
         return self.materialize().astype(dtype, **kwargs)
+
+    # mask logic stays on device: (x > 0) & (x < 1), ~mask, mask | other
+    def _logic(self, other, lop):
+        if other is None:
+            h = backend().mask_logic(self._dev_handle(), 0, self.size, lop)
+        elif (
+            isinstance(other, BoolDeviceArray) and other.shape == self.shape
+        ):
+            h = backend().mask_logic(
+                self._dev_handle(), other._dev_handle(), self.size, lop
+            )
+        else:
+            return NotImplemented
+        return BoolDeviceArray(h, self.shape, _np.bool_)
+
+    def __and__(self, o):
+        r = self._logic(o, 0)
+        if r is NotImplemented:
+            return self.materialize() & _asarray_or_scalar(o)
+        return r
+
+    def __or__(self, o):
+        r = self._logic(o, 1)
+        if r is NotImplemented:
+            return self.materialize() | _asarray_or_scalar(o)
+        return r
+
+    def __xor__(self, o):
+        r = self._logic(o, 2)
+        if r is NotImplemented:
+            return self.materialize() ^ _asarray_or_scalar(o)
+        return r
+
+    def __invert__(self):
+        return self._logic(None, 4)
+
+    __rand__ = __and__
+    __ror__ = __or__
+    __rxor__ = __xor__
 
 
 _QUANTILE_BINS = 4096

@@ -100,6 +100,23 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def mask_logic(self, ha, hb, n, lop):
+        self.calls.append("mask_logic")
+        a = self.bufs[ha].view(np.uint8).reshape(-1)[:n].astype(bool)
+        b = (self.bufs[hb].view(np.uint8).reshape(-1)[:n].astype(bool)
+             if hb else None)
+        if lop == 0:
+            out = a & b
+        elif lop == 1:
+            out = a | b
+        elif lop == 2:
+            out = a ^ b
+        elif lop == 3:
+            out = a & ~b
+        else:
+            out = ~a
+        return self._new(out.astype(np.uint8))
+
     def histogram(self, h, dtype, n, lo, hi, bins):
         self.calls.append("histogram")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n].astype(np.float64)
@@ -692,3 +709,33 @@ def test_axis_var_std_on_device(fake):
     h3 = np.random.default_rng(38).random((4, 5, 6))
     x3 = _device(fake, h3)
     np.testing.assert_allclose(x3.var(axis=1), h3.var(axis=1), rtol=1e-10)
+
+
+def test_mask_logic_isnan_any_all(fake):
+    host = np.random.default_rng(40).normal(0, 1, 512)
+    host[7] = np.nan
+    host[100] = 0.0
+    x = _device(fake, host)
+    with np.errstate(invalid="ignore"):
+        band = (x > -0.5) & (x < 0.5)
+        ref_band = (host > -0.5) & (host < 0.5)
+    assert isinstance(band, hipnp.BoolDeviceArray)
+    np.testing.assert_array_equal(np.asarray(band), ref_band)
+    inv = ~band
+    np.testing.assert_array_equal(np.asarray(inv), ~ref_band)
+    with np.errstate(invalid="ignore"):
+        either = (x > 1.0) | (x < -1.0)
+        ref_either = (host > 1.0) | (host < -1.0)
+    np.testing.assert_array_equal(np.asarray(either), ref_either)
+    # isnan mask + data cleaning idiom entirely on device
+    nanmask = np.isnan(x)
+    assert isinstance(nanmask, hipnp.BoolDeviceArray)
+    assert int(nanmask.sum()) == 1
+    x[nanmask] = 0.0
+    assert not np.isnan(np.asarray(x)).any()
+    # any/all/count_nonzero on float arrays route through masks
+    assert x.any() == bool(np.nan_to_num(host).any())
+    assert x.all() is False  # there is a zero
+    assert int(np.count_nonzero(x)) == int(
+        np.count_nonzero(np.nan_to_num(host)))
+    assert "mask_logic" in fake.calls

@@ -726,3 +726,34 @@ def test_histogram_extract_primitives(hip):
     assert found == len(ref_vals)
     np.testing.assert_allclose(np.sort(vals), np.sort(ref_vals))
     hip.free(h)
+
+
+def test_mask_logic_kernels(hip):
+    rng = np.random.default_rng(60)
+    a = rng.standard_normal(1_000_001)
+    h = hip.upload(a)
+    m1 = hip.compare(h, 1, a.size, 2, 0, 0.0)    # a > 0
+    m2 = hip.compare(h, 1, a.size, 0, 0, 0.5)    # a < 0.5
+    band = hip.mask_logic(m1, m2, a.size, 0)     # and
+    ref = (a > 0) & (a < 0.5)
+    assert hip.mask_count(band, a.size) == int(ref.sum())
+    inv = hip.mask_logic(band, 0, a.size, 4)     # not
+    assert hip.mask_count(inv, a.size) == int((~ref).sum())
+    either = hip.mask_logic(m1, m2, a.size, 1)   # or
+    assert hip.mask_count(either, a.size) == int(((a > 0) | (a < 0.5)).sum())
+    for hh in (h, m1, m2, band, inv, either):
+        hip.free(hh)
+
+
+def test_nan_cleaning_idiom_device(hnp):
+    x = hnp.rand(4_000_000, seed=123)
+    # introduce NaNs via device arithmetic: log of negatives
+    y = (x - 0.5)._unary("log")  # log(<0) -> nan
+    m = np.isnan(y)
+    assert isinstance(m, hnp.BoolDeviceArray)
+    n_nan = int(m.sum())
+    assert 0 < n_nan < y.size
+    y[m] = 0.0
+    host = np.asarray(y)
+    assert not np.isnan(host).any()
+    assert int(np.count_nonzero(y)) == int(np.count_nonzero(host))
