@@ -49,6 +49,9 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// cumulative sum (flat): totals scratch >= 2048 doubles
+void launch_cumsum(DType dt, const void* in, void* out, void* totals,
+                   int64_t n, hipStream_t stream);
 // u8 mask logic: 0 and, 1 or, 2 xor, 3 andnot, 4 not (b may be null)
 void launch_mask_logic(const void* a, const void* b, void* out, int64_t n,
                        int op, hipStream_t stream);

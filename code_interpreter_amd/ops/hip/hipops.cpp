@@ -575,6 +575,28 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// cumsum(h, dtype, n) -> handle (same dtype)
+PyObject* py_cumsum(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt;
+  long long n;
+  if (!PyArg_ParseTuple(args, "KiL", &h, &dt, &n)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  void* out = nullptr;
+  NOGIL_BEGIN
+  std::lock_guard<std::mutex> reduce_lk(g.reduce_mu);
+  ensure_reduce_scratch();
+  out = pool_alloc(n * esize);
+  launch_cumsum(dtype, in.ptr, out, g.reduce_scratch, n, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n * esize));
+  WRAP_END
+}
+
 // mask_logic(ha, hb_or_0, n, op) -> u8 mask handle
 PyObject* py_mask_logic(PyObject*, PyObject* args) {
   unsigned long long ha, hb;
@@ -1041,6 +1063,7 @@ PyMethodDef methods[] = {
      "binary_scalar(h, scalar, op, dtype, n) -> handle"},
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
+    {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
     {"mask_logic", py_mask_logic, METH_VARARGS,
      "mask_logic(ha, hb_or_0, n, op) -> u8 handle (0 and 1 or 2 xor 3 andnot 4 not)"},
     {"histogram", py_histogram, METH_VARARGS,

@@ -262,6 +262,72 @@ __global__ void minmax_stage2(const T* __restrict__ partials,
 }
 
 // ---------------------------------------------------------------------------
+// cumsum: three-pass block scan. Pass 1: block b scans its contiguous
+// chunk (per-thread sequential sums -> LDS exclusive scan of thread
+// totals -> rewrite with prefixes) and stores its chunk total. Pass 2:
+// one block exclusive-scans the totals. Pass 3: add each block's offset.
+// Accumulation in double (f32 in -> f32 out, f64 exact); element order
+// within a thread is sequential, so rounding matches numpy to ~1 ulp
+// per partial-regrouping (documented).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void cumsum_block_kernel(const T* __restrict__ in,
+                                    T* __restrict__ out,
+                                    double* __restrict__ totals, int64_t n,
+                                    int64_t chunk) {
+  __shared__ double tsum[kBlock];
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = start + chunk < n ? start + chunk : n;
+  int64_t per = (chunk + blockDim.x - 1) / blockDim.x;
+  int64_t t0 = start + threadIdx.x * per;
+  int64_t t1 = t0 + per < end ? t0 + per : end;
+  double acc = 0.0;
+  for (int64_t i = t0; i < t1; i++) acc += (double)in[i];
+  tsum[threadIdx.x] = acc;
+  __syncthreads();
+  // exclusive scan of thread totals (single thread: kBlock=256 adds,
+  // negligible vs the per-thread sweeps)
+  if (threadIdx.x == 0) {
+    double run = 0.0;
+    for (int i = 0; i < (int)blockDim.x; i++) {
+      double v = tsum[i];
+      tsum[i] = run;
+      run += v;
+    }
+    if (totals) totals[blockIdx.x] = run;
+  }
+  __syncthreads();
+  double run = tsum[threadIdx.x];
+  for (int64_t i = t0; i < t1; i++) {
+    run += (double)in[i];
+    out[i] = (T)run;
+  }
+}
+
+__global__ void cumsum_totals_kernel(double* __restrict__ totals, int nb) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    double run = 0.0;
+    for (int i = 0; i < nb; i++) {
+      double v = totals[i];
+      totals[i] = run;
+      run += v;
+    }
+  }
+}
+
+template <typename T>
+__global__ void cumsum_add_offsets_kernel(T* __restrict__ out,
+                                          const double* __restrict__ totals,
+                                          int64_t n, int64_t chunk) {
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = start + chunk < n ? start + chunk : n;
+  double off = totals[blockIdx.x];
+  if (off == 0.0) return;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    out[i] = (T)((double)out[i] + off);
+}
+
+// ---------------------------------------------------------------------------
 // u8 mask logic: 0=and 1=or 2=xor 3=andnot(a & ~b); unary not via
 // xor with an all-ones operand is avoided -- NOT uses op 4 (b ignored)
 // ---------------------------------------------------------------------------
@@ -835,6 +901,28 @@ static void launch_sum_t(ReduceOp mode, const T* in, T* partials,
     hipLaunchKernelGGL((sum_stage2<T>), dim3(1), dim3(kBlock), 0, s, partials,
                        out_scalar, grid);
   HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_cumsum_t(const T* in, T* out, void* totals, int64_t n,
+                            hipStream_t s) {
+  int nb = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  int64_t chunk = (n + nb - 1) / nb;
+  hipLaunchKernelGGL((cumsum_block_kernel<T>), dim3(nb), dim3(kBlock), 0, s,
+                     in, out, (double*)totals, n, chunk);
+  hipLaunchKernelGGL(cumsum_totals_kernel, dim3(1), dim3(64), 0, s,
+                     (double*)totals, nb);
+  hipLaunchKernelGGL((cumsum_add_offsets_kernel<T>), dim3(nb), dim3(kBlock),
+                     0, s, out, (const double*)totals, n, chunk);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_cumsum(DType dt, const void* in, void* out, void* totals,
+                   int64_t n, hipStream_t s) {
+  if (dt == DType::F64)
+    launch_cumsum_t((const double*)in, (double*)out, totals, n, s);
+  else
+    launch_cumsum_t((const float*)in, (float*)out, totals, n, s);
 }
 
 void launch_mask_logic(const void* a, const void* b, void* out, int64_t n,

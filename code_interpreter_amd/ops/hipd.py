@@ -183,6 +183,10 @@ class Connection(threading.Thread):
         if op == "sum":
             v = _hipops.sum(self._own(m["h"]), m["dtype"], m["n"], m["square"])
             return {"ok": True, "value": v}, b""
+        if op == "cumsum":
+            h = _hipops.cumsum(self._own(m["h"]), m["dtype"], m["n"])
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "mask_logic":
             hb = m.get("hb", 0)
             if hb:

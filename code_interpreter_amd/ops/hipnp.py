@@ -313,6 +313,11 @@ class RemoteBackend:
             {"op": "gemm", "ha": ha, "hb": hb, "m": m, "n": n, "k": k, "dtype": dtype}
         )["h"]
 
+    def cumsum(self, h, dtype, n):
+        return self._call(
+            {"op": "cumsum", "h": h, "dtype": dtype, "n": n}
+        )["h"]
+
     def mask_logic(self, ha, hb, n, lop):
         return self._call(
             {"op": "mask_logic", "ha": ha, "hb": hb, "n": n, "lop": lop}
@@ -898,6 +903,22 @@ class DeviceArray:
                 return r
         return self.materialize().clip(a_min, a_max, **kwargs)
 
+    def cumsum(self, axis=None, **kwargs):
+        """Flat cumulative sum on device (axis=None or the only axis of
+        a 1-D array); accumulation in double. Rounding may differ from
+        numpy's strictly-sequential order by ~1 ulp at block
+        regroupings."""
+        code = _dtype_code(self.dtype)
+        routable = code is not None and not kwargs and (
+            axis is None
+            or (len(self.shape) == 1 and self._norm_axis(axis, 1) == 0)
+        )
+        if routable:
+            h = backend().cumsum(self._dev_handle(), code, self.size)
+            shape = self.shape if axis is not None else (self.size,)
+            return DeviceArray(h, shape, self.dtype)
+        return self.materialize().cumsum(axis=axis, **kwargs)
+
     def isnan(self):
         """Device NaN mask (x != x elementwise)."""
         r = self._compare("not_equal", self)
@@ -1099,6 +1120,11 @@ class DeviceArray:
                 if isinstance(args[0], BoolDeviceArray):
                     return args[0].sum()
                 return args[0].count_nonzero()
+        if func is _np.cumsum and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ):
+            if set(kwargs) <= {"axis"}:
+                return args[0].cumsum(**kwargs)
         if func in (_np.any, _np.all) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ):

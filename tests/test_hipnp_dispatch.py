@@ -100,6 +100,11 @@ class FakeBackend:
             return float(a.max())
         return float(a.min())
 
+    def cumsum(self, h, dtype, n):
+        self.calls.append("cumsum")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
+        return self._new(a.astype(np.float64).cumsum().astype(self._dt(dtype)))
+
     def mask_logic(self, ha, hb, n, lop):
         self.calls.append("mask_logic")
         a = self.bufs[ha].view(np.uint8).reshape(-1)[:n].astype(bool)
@@ -739,3 +744,22 @@ def test_mask_logic_isnan_any_all(fake):
     assert int(np.count_nonzero(x)) == int(
         np.count_nonzero(np.nan_to_num(host)))
     assert "mask_logic" in fake.calls
+
+
+def test_cumsum_on_device(fake):
+    host = np.random.default_rng(41).normal(0, 1, 4096)
+    x = _device(fake, host)
+    r = np.cumsum(x)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(np.asarray(r), host.cumsum(), rtol=1e-12)
+    r2 = x.cumsum(axis=0)
+    np.testing.assert_allclose(np.asarray(r2), host.cumsum(), rtol=1e-12)
+    assert "cumsum" in fake.calls
+    # 2-D with axis: host fallback, same values
+    h2 = np.random.default_rng(42).random((8, 16))
+    x2 = _device(fake, h2)
+    np.testing.assert_allclose(x2.cumsum(axis=1), h2.cumsum(axis=1))
+    # flat cumsum of a 2-D array flattens like numpy
+    flat = np.cumsum(x2)
+    assert isinstance(flat, hipnp.DeviceArray) and flat.shape == (128,)
+    np.testing.assert_allclose(np.asarray(flat), h2.cumsum(), rtol=1e-12)

@@ -757,3 +757,18 @@ def test_nan_cleaning_idiom_device(hnp):
     host = np.asarray(y)
     assert not np.isnan(host).any()
     assert int(np.count_nonzero(y)) == int(np.count_nonzero(host))
+
+
+def test_cumsum_matches_numpy(hip):
+    rng = np.random.default_rng(70)
+    for dtype, code, rtol in ((np.float64, 1, 1e-12), (np.float32, 0, 2e-6)):
+        a = rng.standard_normal(10_000_001).astype(dtype)
+        h = hip.upload(a)
+        hc = hip.cumsum(h, code, a.size)
+        out = np.empty_like(a)
+        hip.download(hc, out)
+        hip.free(h)
+        hip.free(hc)
+        ref = a.astype(np.float64).cumsum()
+        scale = np.abs(ref).max() + 1.0
+        assert np.max(np.abs(out.astype(np.float64) - ref)) / scale < rtol
