@@ -25,6 +25,7 @@ GEMM_SOURCES = [
 SOURCES = [
     HIP_DIR / "hipops.cpp",
     HIP_DIR / "kernels_ew.hip",
+    HIP_DIR / "sort.hip",
     *GEMM_SOURCES,
 ]
 # _hipgemm: the torch-interop module. Built SEPARATELY because zygote

@@ -49,6 +49,12 @@ void launch_sum(DType dt, ReduceOp mode, const void* in, void* partials,
 void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         int64_t outer, int64_t red, int64_t inner,
                         hipStream_t stream);
+// radix sort (np.sort/argsort): see sort.hip header for scratch sizes
+int64_t radix_sort_nchunks(int64_t n);
+void launch_radix_sort(DType dt, const void* in, void* out, void* idx_out,
+                       void* keys_a, void* keys_b, void* idx_a, void* idx_b,
+                       void* counts, void* dig_scratch, int64_t n,
+                       bool want_idx, hipStream_t stream);
 // cumulative sum (flat): totals scratch >= 2048 doubles
 void launch_cumsum(DType dt, const void* in, void* out, void* totals,
                    int64_t n, hipStream_t stream);

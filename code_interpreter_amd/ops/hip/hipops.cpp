@@ -575,6 +575,54 @@ PyObject* py_sum(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// sort(h, dtype, n, want_idx) -> handle | (handle, idx_handle)
+PyObject* py_sort(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt, want_idx;
+  long long n;
+  if (!PyArg_ParseTuple(args, "KiLi", &h, &dt, &n, &want_idx))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (n < 1 || n > (1ll << 31)) throw std::runtime_error("bad sort length");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  int64_t ksize = esize;  // key width == element width
+  int64_t nchunks = radix_sort_nchunks(n);
+  int64_t nseg = (nchunks + 255) / 256;
+  void* out = nullptr;
+  void* idx_res = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n * esize);
+  void* keys_a = pool_alloc(n * ksize);
+  void* keys_b = pool_alloc(n * ksize);
+  void* counts = pool_alloc(nchunks * 256 * 4);
+  void* dig = pool_alloc(256 * 8 + nseg * 256 * 4);
+  void* idx_a = nullptr;
+  void* idx_b = nullptr;
+  if (want_idx) {
+    idx_a = pool_alloc(n * 8);
+    idx_b = pool_alloc(n * 8);
+  }
+  launch_radix_sort(dtype, in.ptr, out, idx_a, keys_a, keys_b, idx_a, idx_b,
+                    counts, dig, n, want_idx != 0, g.compute);
+  pool_free(keys_a, n * ksize);
+  pool_free(keys_b, n * ksize);
+  pool_free(counts, nchunks * 256 * 4);
+  pool_free(dig, 256 * 8 + nseg * 256 * 4);
+  if (want_idx) {
+    idx_res = idx_a;  // result indices live in idx_a after even passes
+    pool_free(idx_b, n * 8);
+  }
+  NOGIL_END
+  unsigned long long hout = register_buf(out, n * esize);
+  if (!want_idx) return PyLong_FromUnsignedLongLong(hout);
+  unsigned long long hidx = register_buf(idx_res, n * 8);
+  return Py_BuildValue("(KK)", hout, hidx);
+  WRAP_END
+}
+
 // cumsum(h, dtype, n) -> handle (same dtype)
 PyObject* py_cumsum(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1064,6 +1112,8 @@ PyMethodDef methods[] = {
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
+    {"sort", py_sort, METH_VARARGS,
+     "sort(h, dtype, n, want_idx) -> handle | (handle, int64 idx handle)"},
     {"mask_logic", py_mask_logic, METH_VARARGS,
      "mask_logic(ha, hb_or_0, n, op) -> u8 handle (0 and 1 or 2 xor 3 andnot 4 not)"},
     {"histogram", py_histogram, METH_VARARGS,

@@ -187,6 +187,17 @@ class Connection(threading.Thread):
             h = _hipops.cumsum(self._own(m["h"]), m["dtype"], m["n"])
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "sort":
+            r = _hipops.sort(
+                self._own(m["h"]), m["dtype"], m["n"], m["want_idx"]
+            )
+            if m["want_idx"]:
+                h, hi = r
+                self.handles.add(h)
+                self.handles.add(hi)
+                return {"ok": True, "h": h, "hi": hi}, b""
+            self.handles.add(r)
+            return {"ok": True, "h": r}, b""
         if op == "mask_logic":
             hb = m.get("hb", 0)
             if hb:

@@ -318,6 +318,15 @@ class RemoteBackend:
             {"op": "cumsum", "h": h, "dtype": dtype, "n": n}
         )["h"]
 
+    def sort(self, h, dtype, n, want_idx):
+        out = self._call(
+            {"op": "sort", "h": h, "dtype": dtype, "n": n,
+             "want_idx": want_idx}
+        )
+        if want_idx:
+            return out["h"], out["hi"]
+        return out["h"]
+
     def mask_logic(self, ha, hb, n, lop):
         return self._call(
             {"op": "mask_logic", "ha": ha, "hb": hb, "n": n, "lop": lop}
@@ -919,6 +928,58 @@ class DeviceArray:
             return DeviceArray(h, shape, self.dtype)
         return self.materialize().cumsum(axis=axis, **kwargs)
 
+    def _sort_routable(self, axis, kind, order, kwargs):
+        """np.sort/argsort route: 1-D f32/f64, default comparator. The
+        device sort is an LSD radix sort, so it is stable -- every numpy
+        `kind` is satisfied."""
+        return (
+            _dtype_code(self.dtype) is not None
+            and not kwargs
+            and order is None
+            and kind in (None, "stable", "quicksort", "mergesort", "heapsort")
+            and len(self.shape) == 1
+            and (axis is None or self._norm_axis(axis, 1) == 0)
+        )
+
+    def _device_sorted(self, want_idx):
+        r = backend().sort(
+            self._dev_handle(), _dtype_code(self.dtype), self.size,
+            1 if want_idx else 0,
+        )
+        if want_idx:
+            h, hi = r
+            return (
+                DeviceArray(h, self.shape, self.dtype),
+                DeviceArray(hi, self.shape, _np.int64),
+            )
+        return DeviceArray(r, self.shape, self.dtype)
+
+    def sort(self, axis=-1, kind=None, order=None, **kwargs):
+        """In-place sort (ndarray.sort contract): the handle is swapped
+        for the sorted buffer."""
+        if self._sort_routable(axis, kind, order, kwargs):
+            res = self._device_sorted(False)
+            old_h, self._handle = self._handle, res._handle
+            res._handle = None  # ownership moved; res.__del__ must not free
+            self._host = None
+            if old_h is not None:
+                try:
+                    backend().free(old_h)
+                except Exception:
+                    pass
+            return None
+        host = self._mutable_host()
+        host.sort(axis=axis, kind=kind, order=order, **kwargs)
+        return None
+
+    def argsort(self, axis=-1, kind=None, order=None, **kwargs):
+        if self._sort_routable(axis, kind, order, kwargs):
+            _, idx = self._device_sorted(True)
+            return idx
+        return self.materialize().argsort(
+            axis=axis, kind=kind, order=order, **kwargs
+        )
+
     def isnan(self):
         """Device NaN mask (x != x elementwise)."""
         r = self._compare("not_equal", self)
@@ -943,7 +1004,7 @@ class DeviceArray:
         return self.materialize().all(axis=axis, **kwargs)
 
     def argmax(self, axis=None, **kwargs):
-        if axis is None and not kwargs:
+        if axis is None and not kwargs and _dtype_code(self.dtype) is not None:
             return _np.intp(
                 backend().argminmax(
                     self._dev_handle(), _dtype_code(self.dtype), self.size, 1
@@ -952,7 +1013,7 @@ class DeviceArray:
         return self.materialize().argmax(axis=axis, **kwargs)
 
     def argmin(self, axis=None, **kwargs):
-        if axis is None and not kwargs:
+        if axis is None and not kwargs and _dtype_code(self.dtype) is not None:
             return _np.intp(
                 backend().argminmax(
                     self._dev_handle(), _dtype_code(self.dtype), self.size, 0
@@ -1140,6 +1201,24 @@ class DeviceArray:
             if set(kwargs) <= {"axis"}:
                 meth = "argmax" if func is _np.argmax else "argmin"
                 return getattr(args[0], meth)(**kwargs)
+        if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ):
+            a = args[0]
+            axis = kwargs.get("axis", -1)
+            kind = kwargs.get("kind")
+            order = kwargs.get("order")
+            extra = {
+                k: v for k, v in kwargs.items()
+                if k not in ("axis", "kind", "order")
+            }
+            if a._sort_routable(axis, kind, order, extra):
+                if func is _np.sort:
+                    return a._device_sorted(False)
+                _, idx = a._device_sorted(True)
+                return idx
+            host = a.materialize()
+            return func(host, **kwargs)
         _reductions = {
             _np.max: "max", _np.amax: "max",
             _np.min: "min", _np.amin: "min",

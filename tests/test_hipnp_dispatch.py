@@ -105,6 +105,14 @@ class FakeBackend:
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
         return self._new(a.astype(np.float64).cumsum().astype(self._dt(dtype)))
 
+    def sort(self, h, dtype, n, want_idx):
+        self.calls.append("sort")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
+        if want_idx:
+            idx = np.argsort(a, kind="stable")
+            return self._new(a[idx]), self._new(idx.astype(np.int64))
+        return self._new(np.sort(a, kind="stable"))
+
     def mask_logic(self, ha, hb, n, lop):
         self.calls.append("mask_logic")
         a = self.bufs[ha].view(np.uint8).reshape(-1)[:n].astype(bool)
@@ -763,3 +771,50 @@ def test_cumsum_on_device(fake):
     flat = np.cumsum(x2)
     assert isinstance(flat, hipnp.DeviceArray) and flat.shape == (128,)
     np.testing.assert_allclose(np.asarray(flat), h2.cumsum(), rtol=1e-12)
+
+
+def test_sort_on_device(fake):
+    host = np.random.default_rng(11).random(1000).astype(np.float64)
+    x = _device(fake, host)
+    s = np.sort(x)
+    assert isinstance(s, hipnp.DeviceArray)
+    assert "sort" in fake.calls
+    np.testing.assert_array_equal(s.materialize(), np.sort(host))
+    # f32 too
+    x32 = _device(fake, host.astype(np.float32))
+    s32 = np.sort(x32)
+    assert isinstance(s32, hipnp.DeviceArray)
+    assert s32.dtype == np.float32
+    np.testing.assert_array_equal(
+        s32.materialize(), np.sort(host.astype(np.float32)))
+
+
+def test_argsort_on_device(fake):
+    host = np.random.default_rng(12).random(800)
+    x = _device(fake, host)
+    idx = np.argsort(x)
+    assert isinstance(idx, hipnp.DeviceArray)
+    assert idx.dtype == np.int64
+    np.testing.assert_array_equal(idx.materialize(), np.argsort(host))
+    # method form
+    idx2 = _device(fake, host).argsort()
+    np.testing.assert_array_equal(np.asarray(idx2), np.argsort(host))
+
+
+def test_sort_method_in_place(fake):
+    host = np.random.default_rng(13).random(500)
+    x = _device(fake, host)
+    r = x.sort()
+    assert r is None  # ndarray.sort contract
+    np.testing.assert_array_equal(x.materialize(), np.sort(host))
+
+
+def test_sort_unroutable_falls_back(fake):
+    host = np.random.default_rng(14).random((20, 30))
+    x = _device(fake, host)
+    s = np.sort(x, axis=0)  # 2-D: host fallback
+    assert isinstance(s, np.ndarray)
+    np.testing.assert_array_equal(s, np.sort(host, axis=0))
+    # any kind routes (stable output satisfies every numpy kind)
+    idx = np.argsort(_device(fake, host[0]), kind="heapsort")
+    np.testing.assert_array_equal(np.asarray(idx), np.argsort(host[0]))

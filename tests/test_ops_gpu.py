@@ -772,3 +772,89 @@ def test_cumsum_matches_numpy(hip):
         ref = a.astype(np.float64).cumsum()
         scale = np.abs(ref).max() + 1.0
         assert np.max(np.abs(out.astype(np.float64) - ref)) / scale < rtol
+
+
+# ---------------------------------------------------------------------------
+# device radix sort (sort.hip)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype,code", [(np.float64, 1), (np.float32, 0)])
+@pytest.mark.parametrize("n", [1, 63, 4097, 1_000_003])
+def test_sort_matches_numpy(hip, dtype, code, n):
+    rng = np.random.default_rng(100 + n)
+    a = (rng.standard_normal(n) * 100).astype(dtype)
+    h = hip.upload(a)
+    hs = hip.sort(h, code, n, 0)
+    out = np.empty(n, dtype)
+    hip.download(hs, out)
+    hip.free(h)
+    hip.free(hs)
+    np.testing.assert_array_equal(out, np.sort(a))
+
+
+@pytest.mark.parametrize("dtype,code", [(np.float64, 1), (np.float32, 0)])
+def test_argsort_stable_matches_numpy(hip, dtype, code):
+    # duplicate-heavy data: a stable LSD radix sort must reproduce
+    # numpy's kind="stable" indices exactly
+    n = 500_000
+    rng = np.random.default_rng(7)
+    a = rng.integers(0, 17, n).astype(dtype)
+    h = hip.upload(a)
+    hs, hi = hip.sort(h, code, n, 1)
+    out = np.empty(n, dtype)
+    idx = np.empty(n, np.int64)
+    hip.download(hs, out)
+    hip.download(hi, idx)
+    for x in (h, hs, hi):
+        hip.free(x)
+    np.testing.assert_array_equal(out, np.sort(a, kind="stable"))
+    np.testing.assert_array_equal(idx, np.argsort(a, kind="stable"))
+
+
+def test_sort_nan_last_both_signs(hip):
+    n = 100_000
+    rng = np.random.default_rng(8)
+    a = rng.standard_normal(n)
+    a[rng.integers(0, n, 500)] = np.nan
+    a[rng.integers(0, n, 500)] = np.copysign(np.nan, -1.0)
+    a[rng.integers(0, n, 100)] = np.inf
+    a[rng.integers(0, n, 100)] = -np.inf
+    h = hip.upload(a)
+    hs = hip.sort(h, 1, n, 0)
+    out = np.empty(n, np.float64)
+    hip.download(hs, out)
+    hip.free(h)
+    hip.free(hs)
+    n_nan = int(np.isnan(a).sum())
+    assert np.isnan(out[n - n_nan:]).all()
+    np.testing.assert_array_equal(out[: n - n_nan],
+                                  np.sort(a)[: n - n_nan])
+
+
+def test_sort_large_1e7(hip):
+    n = 10_000_000
+    rng = np.random.default_rng(9)
+    a = rng.standard_normal(n).astype(np.float32)
+    h = hip.upload(a)
+    hs, hi = hip.sort(h, 0, n, 1)
+    out = np.empty(n, np.float32)
+    idx = np.empty(n, np.int64)
+    hip.download(hs, out)
+    hip.download(hi, idx)
+    for x in (h, hs, hi):
+        hip.free(x)
+    np.testing.assert_array_equal(out, np.sort(a))
+    # indices permute the input into sorted order
+    np.testing.assert_array_equal(a[idx], out)
+
+
+def test_device_array_sort_route(hnp):
+    hnp.install()
+    a = np.random.rand(200_000)
+    x = hnp.DeviceArray(None, a.shape, a.dtype)
+    x._host = a  # host-authoritative; first device op uploads
+    s = np.sort(x)
+    assert isinstance(s, hnp.DeviceArray)
+    np.testing.assert_array_equal(s.materialize(), np.sort(a))
+    idx = np.argsort(x)
+    assert isinstance(idx, hnp.DeviceArray) and idx.dtype == np.int64
+    np.testing.assert_array_equal(idx.materialize(), np.argsort(a))
