@@ -848,10 +848,8 @@ def test_sort_large_1e7(hip):
 
 
 def test_device_array_sort_route(hnp):
-    hnp.install()
-    a = np.random.rand(200_000)
-    x = hnp.DeviceArray(None, a.shape, a.dtype)
-    x._host = a  # host-authoritative; first device op uploads
+    x = hnp.rand(200_000, seed=42)
+    a = np.asarray(x).copy()
     s = np.sort(x)
     assert isinstance(s, hnp.DeviceArray)
     np.testing.assert_array_equal(s.materialize(), np.sort(a))
