@@ -980,6 +980,26 @@ class DeviceArray:
             axis=axis, kind=kind, order=order, **kwargs
         )
 
+    def partition(self, kth, axis=-1, kind="introselect", order=None,
+                  **kwargs):
+        """In-place partition. A fully sorted array satisfies the
+        partition contract for every kth, so the device route reuses
+        the radix sort."""
+        if self._sort_routable(axis, None, order, kwargs):
+            self.sort(axis=axis)
+            return None
+        host = self._mutable_host()
+        host.partition(kth, axis=axis, kind=kind, order=order, **kwargs)
+        return None
+
+    def argpartition(self, kth, axis=-1, kind="introselect", order=None,
+                     **kwargs):
+        if self._sort_routable(axis, None, order, kwargs):
+            return self.argsort(axis=axis)
+        return self.materialize().argpartition(
+            kth, axis=axis, kind=kind, order=order, **kwargs
+        )
+
     def isnan(self):
         """Device NaN mask (x != x elementwise)."""
         r = self._compare("not_equal", self)
@@ -1201,6 +1221,21 @@ class DeviceArray:
             if set(kwargs) <= {"axis"}:
                 meth = "argmax" if func is _np.argmax else "argmin"
                 return getattr(args[0], meth)(**kwargs)
+        if func in (_np.partition, _np.argpartition) and len(args) == 2 \
+                and isinstance(args[0], DeviceArray):
+            a, kth = args
+            axis = kwargs.get("axis", -1)
+            order = kwargs.get("order")
+            extra = {
+                k: v for k, v in kwargs.items()
+                if k not in ("axis", "kind", "order")
+            }
+            if a._sort_routable(axis, None, order, extra):
+                if func is _np.partition:
+                    return a._device_sorted(False)
+                _, idx = a._device_sorted(True)
+                return idx
+            return func(a.materialize(), kth, **kwargs)
         if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ):

@@ -818,3 +818,20 @@ def test_sort_unroutable_falls_back(fake):
     # any kind routes (stable output satisfies every numpy kind)
     idx = np.argsort(_device(fake, host[0]), kind="heapsort")
     np.testing.assert_array_equal(np.asarray(idx), np.argsort(host[0]))
+
+
+def test_partition_routes_to_sort(fake):
+    host = np.random.default_rng(15).random(300)
+    x = _device(fake, host)
+    p = np.partition(x, 50)
+    assert isinstance(p, hipnp.DeviceArray)
+    # a sorted array is a valid partition for any kth
+    np.testing.assert_array_equal(p.materialize(), np.sort(host))
+    idx = np.argpartition(_device(fake, host), 10)
+    assert isinstance(idx, hipnp.DeviceArray)
+    ref = host[np.asarray(idx.materialize())]
+    np.testing.assert_array_equal(ref, np.sort(host))
+    # in-place method form
+    y = _device(fake, host)
+    assert y.partition(25) is None
+    np.testing.assert_array_equal(y.materialize(), np.sort(host))
