@@ -381,6 +381,30 @@ PyObject* py_upload(PyObject*, PyObject* args) {
 }
 
 // download(handle, writable_host_buffer)
+// download_slice(h, byte_offset, nbytes) -> bytes. Small-window reads
+// (x[i], head/tail peeks) must not pull the whole buffer across PCIe.
+PyObject* py_download_slice(PyObject*, PyObject* args) {
+  unsigned long long h;
+  long long off, nbytes;
+  if (!PyArg_ParseTuple(args, "KLL", &h, &off, &nbytes)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DevBuf& buf = get_buf(h);
+  if (off < 0 || nbytes < 0 || off + nbytes > buf.size)
+    throw std::runtime_error("download_slice out of range");
+  PyObject* out = PyBytes_FromStringAndSize(nullptr, (Py_ssize_t)nbytes);
+  if (!out) throw std::bad_alloc();
+  char* dst = PyBytes_AS_STRING(out);
+  NOGIL_BEGIN
+  std::lock_guard<std::mutex> stage_lk(g.stage_mu);
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  HIP_CHECK(hipMemcpy(dst, (char*)buf.ptr + off, (size_t)nbytes,
+                      hipMemcpyDeviceToHost));
+  NOGIL_END
+  return out;
+  WRAP_END
+}
+
 PyObject* py_download(PyObject*, PyObject* args) {
   unsigned long long h;
   PyObject* obj;
@@ -1100,6 +1124,8 @@ PyMethodDef methods[] = {
     {"free", py_free, METH_VARARGS, "free(handle)"},
     {"upload", py_upload, METH_VARARGS, "upload(buffer) -> handle"},
     {"download", py_download, METH_VARARGS, "download(handle, buffer)"},
+    {"download_slice", py_download_slice, METH_VARARGS,
+     "download_slice(handle, byte_offset, nbytes) -> bytes"},
     {"rand", py_rand, METH_VARARGS, "rand(n, dtype, seed) -> handle"},
     {"randn", py_randn, METH_VARARGS,
      "randn(n, seed, mu, sigma) -> handle (f64 normal)"},

@@ -187,6 +187,11 @@ class Connection(threading.Thread):
             h = _hipops.cumsum(self._own(m["h"]), m["dtype"], m["n"])
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "download_slice":
+            data = _hipops.download_slice(
+                self._own(m["h"]), m["off"], m["nbytes"]
+            )
+            return {"ok": True}, data
         if op == "sort":
             r = _hipops.sort(
                 self._own(m["h"]), m["dtype"], m["n"], m["want_idx"]

@@ -318,6 +318,12 @@ class RemoteBackend:
             {"op": "cumsum", "h": h, "dtype": dtype, "n": n}
         )["h"]
 
+    def download_slice(self, h, off, nbytes):
+        out = self._call(
+            {"op": "download_slice", "h": h, "off": off, "nbytes": nbytes}
+        )
+        return out["_payload"]
+
     def sort(self, h, dtype, n, want_idx):
         out = self._call(
             {"op": "sort", "h": h, "dtype": dtype, "n": n,
@@ -606,8 +612,46 @@ class DeviceArray:
     def __getattr__(self, name):
         return getattr(self.materialize(), name)
 
+    # scalar reads of big resident arrays fetch one element, not the
+    # whole buffer (x[i] in numpy returns a copy -- a scalar -- so this
+    # fast path is semantically exact)
+    _SCALAR_FETCH_MIN = 1 << 20  # elements; below this just materialize
+
     def __getitem__(self, idx):
+        if (
+            self._host is None
+            and self._handle is not None
+            and self.size >= self._SCALAR_FETCH_MIN
+            and self.dtype.itemsize in (4, 8)
+        ):
+            flat = self._flat_index(idx)
+            if flat is not None:
+                raw = backend().download_slice(
+                    self._handle, flat * self.dtype.itemsize,
+                    self.dtype.itemsize,
+                )
+                return _np.frombuffer(raw, dtype=self.dtype)[0]
         return self.materialize()[idx]
+
+    def _flat_index(self, idx):
+        """C-contiguous flat offset for an all-integer index, or None."""
+        if isinstance(idx, (int, _np.integer)):
+            idx = (idx,)
+        if not (
+            isinstance(idx, tuple)
+            and len(idx) == len(self.shape)
+            and all(isinstance(i, (int, _np.integer)) for i in idx)
+        ):
+            return None
+        flat = 0
+        for i, dim in zip(idx, self.shape):
+            i = int(i)
+            if i < 0:
+                i += dim
+            if not 0 <= i < dim:
+                return None  # let numpy raise its own IndexError
+            flat = flat * dim + i
+        return flat
 
     def __setitem__(self, idx, value):
         # in-place mutation (x[0] = 1, x[x < 0] = 0, ...): CPU numpy

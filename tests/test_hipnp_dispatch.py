@@ -105,6 +105,11 @@ class FakeBackend:
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
         return self._new(a.astype(np.float64).cumsum().astype(self._dt(dtype)))
 
+    def download_slice(self, h, off, nbytes):
+        self.calls.append("download_slice")
+        raw = self.bufs[h].view(np.uint8).reshape(-1).tobytes()
+        return raw[off:off + nbytes]
+
     def sort(self, h, dtype, n, want_idx):
         self.calls.append("sort")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
@@ -835,3 +840,26 @@ def test_partition_routes_to_sort(fake):
     y = _device(fake, host)
     assert y.partition(25) is None
     np.testing.assert_array_equal(y.materialize(), np.sort(host))
+
+
+def test_scalar_getitem_fetches_one_element(fake):
+    n = hipnp.DeviceArray._SCALAR_FETCH_MIN + 3
+    host = np.random.default_rng(16).random(n)
+    x = _device(fake, host)
+    assert x[5] == host[5]
+    assert x[-1] == host[-1]
+    assert "download_slice" in fake.calls
+    assert "download" not in fake.calls  # whole-buffer path never hit
+    # 2-D all-int index
+    m = hipnp.DeviceArray._SCALAR_FETCH_MIN
+    host2 = np.random.default_rng(17).random((1024, m // 1024))
+    y = _device(fake, host2)
+    assert y[3, 7] == host2[3, 7]
+    assert y[-1, -2] == host2[-1, -2]
+    # non-scalar indexing still materializes and matches numpy
+    np.testing.assert_array_equal(x[2:9], host[2:9])
+    # below-threshold arrays materialize (no RPC per element)
+    small = _device(fake, host[:100])
+    fake.calls.clear()
+    assert small[4] == host[4]
+    assert "download_slice" not in fake.calls

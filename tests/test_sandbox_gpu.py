@@ -288,3 +288,33 @@ def test_analysis_example_stays_on_device(gpu_executor):
     assert r.exit_code == 0, r.stderr
     assert "kind: DeviceArray" in r.stdout
     assert "outliers:" in r.stdout
+
+
+def test_sort_and_scalar_peek_on_device(gpu_executor):
+    """np.sort/np.argsort route to the device radix sort inside a real
+    sandboxed execution, and scalar indexing of a resident array stays
+    a one-element fetch."""
+    r = _run(
+        gpu_executor,
+        "import numpy, time\n"
+        "x = numpy.random.rand(10_000_000)\n"
+        "t0 = time.time()\n"
+        "s = numpy.sort(x)\n"
+        "idx = numpy.argsort(x)\n"
+        "dt = time.time() - t0\n"
+        "print(type(s).__name__, type(idx).__name__)\n"
+        "print(float(s[0]), float(s[-1]))\n"
+        "head = numpy.asarray(s)[:5]\n"
+        "assert (head[:-1] <= head[1:]).all()\n"
+        "ix = numpy.asarray(idx)[:3]\n"
+        "xa = numpy.asarray(x)\n"
+        "assert abs(xa[ix[0]] - float(s[0])) < 1e-12\n"
+        "assert dt < 2.0, f'sort not on device: {dt}'\n"
+        "print('ok')\n",
+    )
+    assert r.exit_code == 0, r.stderr
+    lines = r.stdout.splitlines()
+    assert lines[0] == "DeviceArray DeviceArray"
+    lo, hi = map(float, lines[1].split())
+    assert 0.0 <= lo < 1e-5 and 1 - 1e-5 < hi <= 1.0
+    assert lines[-1] == "ok"
