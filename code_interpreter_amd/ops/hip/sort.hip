@@ -15,6 +15,8 @@
 // Keys ping-pong between two buffers; an optional int64 payload
 // (argsort indices) rides along through the same scatters.
 
+#include <cstring>
+
 #include "common.h"
 
 namespace {
@@ -172,6 +174,108 @@ __global__ void radix_scatter_kernel(const K* __restrict__ in_keys,
   }
 }
 
+// Block-cooperative scatter (default variant): one 4096-element chunk
+// per 256-thread block. The wave-autonomous scatter above issues one
+// 4/8-B global write per element to a near-random line — ~16x write
+// amplification, measured 64% of sort time (profiles/
+// sort_kernel_stats_r02.md). Here the chunk is reordered digit-
+// contiguously through LDS first, so global writes go out in ~16-
+// element runs (64-128 B), and the argsort payload is gathered through
+// a u16 in-chunk origin (the 32 KiB in_idx window stays cache-
+// resident) instead of scattered 8-B stores.
+template <typename K, bool PAYLOAD>
+__global__ void radix_scatter_block_kernel(
+    const K* __restrict__ in_keys, K* __restrict__ out_keys,
+    const long long* __restrict__ in_idx, long long* __restrict__ out_idx,
+    int64_t n, int shift, const unsigned int* __restrict__ bases,
+    int64_t nchunks) {
+  constexpr int kQuarter = kSortChunk / 4;  // elements per wave
+  constexpr int kTiles = kQuarter / 64;     // 64-lane tiles per wave
+  __shared__ K stage[kSortChunk];
+  __shared__ unsigned short origin[PAYLOAD ? kSortChunk : 64];
+  __shared__ unsigned int wcnt[4][256];  // per-wave digit counts -> wave-
+                                         // exclusive prefixes (phase 2)
+  __shared__ unsigned int pref[256];     // chunk-local digit prefix
+  __shared__ unsigned int scan_tmp[256];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t chunk = blockIdx.x;
+  if (chunk >= nchunks) return;
+  const int64_t start = chunk * kSortChunk;
+  const int64_t end = start + kSortChunk < n ? start + kSortChunk : n;
+  const int len = (int)(end - start);
+  const uint64_t lanes_below = ((uint64_t)1 << lane) - 1;
+
+  for (int i = lane; i < 256; i += 64) wcnt[wave][i] = 0;
+  // phase 1: wave-local stable ranks over this wave's quarter
+  K key[kTiles];
+  unsigned short rank[kTiles];
+  const int q0 = wave * kQuarter;
+  for (int t = 0; t < kTiles; t++) {
+    int e = q0 + t * 64 + lane;
+    bool valid = e < len;
+    key[t] = valid ? in_keys[start + e] : (K)0;
+    unsigned d = (unsigned)((key[t] >> shift) & 0xff);
+    uint64_t same = __ballot(valid);
+#pragma unroll
+    for (int b = 0; b < 8; b++) {
+      uint64_t bal = __ballot(valid && ((d >> b) & 1u));
+      same &= ((d >> b) & 1u) ? bal : ~bal;
+    }
+    if (valid) {
+      unsigned r = (unsigned)__popcll(same & lanes_below);
+      rank[t] = (unsigned short)(wcnt[wave][d] + r);
+      if (r == 0) wcnt[wave][d] += (unsigned)__popcll(same);
+    }
+  }
+  __syncthreads();
+  // phase 2 (one thread per digit): wave-exclusive prefixes in wcnt,
+  // chunk totals -> exclusive digit prefix in pref (Hillis-Steele)
+  {
+    const int d = threadIdx.x;
+    unsigned c0 = wcnt[0][d], c1 = wcnt[1][d], c2 = wcnt[2][d];
+    unsigned tot = c0 + c1 + c2 + wcnt[3][d];
+    wcnt[0][d] = 0;
+    wcnt[1][d] = c0;
+    wcnt[2][d] = c0 + c1;
+    wcnt[3][d] = c0 + c1 + c2;
+    unsigned v = tot;
+    scan_tmp[d] = v;
+    for (int off = 1; off < 256; off <<= 1) {
+      __syncthreads();
+      unsigned add = d >= off ? scan_tmp[d - off] : 0;
+      __syncthreads();
+      scan_tmp[d] = v = v + add;
+    }
+    pref[d] = v - tot;  // inclusive -> exclusive
+  }
+  __syncthreads();
+  // phase 3: scatter into LDS, digit-contiguous
+  for (int t = 0; t < kTiles; t++) {
+    int e = q0 + t * 64 + lane;
+    if (e < len) {
+      unsigned d = (unsigned)((key[t] >> shift) & 0xff);
+      unsigned pos = pref[d] + wcnt[wave][d] + rank[t];
+      stage[pos] = key[t];
+      if (PAYLOAD) origin[pos] = (unsigned short)e;
+    }
+  }
+  __syncthreads();
+  // phase 4: linear readout -> coalesced digit-run global writes
+  const unsigned int* base_row = bases + chunk * 256;
+  for (int t = 0; t < kTiles; t++) {
+    int li = (int)threadIdx.x + t * 256;
+    if (li < len) {
+      K k = stage[li];
+      unsigned d = (unsigned)((k >> shift) & 0xff);
+      unsigned gpos = base_row[d] + (unsigned)li - pref[d];
+      out_keys[gpos] = k;
+      if (PAYLOAD) out_idx[gpos] = in_idx[start + origin[li]];
+    }
+  }
+}
+
 template <typename T, typename K>
 __global__ void sort_encode_kernel(const T* __restrict__ in,
                                    K* __restrict__ keys,
@@ -199,6 +303,17 @@ __global__ void sort_decode_kernel(const K* __restrict__ keys,
     else
       out[i] = (T)f32_of_key32(keys[i]);
   }
+}
+
+// APP_SORT_VARIANT=wave selects the wave-autonomous scatter for A/B
+// measurement; default is the LDS block scatter
+static bool use_block_scatter() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("APP_SORT_VARIANT");
+    v = (e && strcmp(e, "wave") == 0) ? 0 : 1;
+  }
+  return v == 1;
 }
 
 template <typename T, typename K>
@@ -232,7 +347,18 @@ static void radix_sort_impl(const T* in, T* out, long long* idx_out,
                        segsum, nseg, (unsigned long long*)dig_scratch);
     hipLaunchKernelGGL(radix_fold_kernel, dim3(grid_s), dim3(256), 0, s,
                        (unsigned int*)counts, nchunks, nseg, segsum);
-    if (want_idx)
+    if (use_block_scatter()) {
+      if (want_idx)
+        hipLaunchKernelGGL((radix_scatter_block_kernel<K, true>),
+                           dim3((unsigned)nchunks), dim3(256), 0, s, ka, kb,
+                           ia, ib, n, shift, (const unsigned int*)counts,
+                           nchunks);
+      else
+        hipLaunchKernelGGL((radix_scatter_block_kernel<K, false>),
+                           dim3((unsigned)nchunks), dim3(256), 0, s, ka, kb,
+                           nullptr, nullptr, n, shift,
+                           (const unsigned int*)counts, nchunks);
+    } else if (want_idx)
       hipLaunchKernelGGL((radix_scatter_kernel<K, true>), dim3(grid_c),
                          dim3(256), 0, s, ka, kb, ia, ib, n, shift,
                          (const unsigned int*)counts, nchunks);
