@@ -275,17 +275,308 @@ __global__ __launch_bounds__(THREADS2) void gemm_f64_128_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 3-stage pipelined variant: the double-buffered loop above ends every
+// iteration with a ds_write convoy right before the barrier (the whole
+// block parks until the slowest wave finishes staging). With THREE LDS
+// buffers the staging writes for tile k+1 issue BEFORE tile k's MFMAs,
+// so the compiler interleaves them into the MFMA/ds_read stream, and
+// the barrier only fences buffer reuse two tiles later. BK=8 keeps the
+// triple buffer at ~52 KB -> 3 blocks/CU (6 waves/SIMD vs 4).
+// ---------------------------------------------------------------------------
+template <int BKT>
+__global__ __launch_bounds__(THREADS2) void gemm_f64_128p_kernel(
+    const double* __restrict__ A, const double* __restrict__ B,
+    double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  int tm, tn;
+  {
+    int nwg = tiles_m * tiles_n;
+    int wgid = blockIdx.x;
+    const int n_st = (tiles_m / 8) * (tiles_n / 4);
+    if (tiles_m % 8 == 0 && tiles_n % 4 == 0 && n_st % 8 == 0) {
+      const int st_cols = tiles_n / 4;
+      int xcd = wgid % 8, idx = wgid / 8;
+      int st = xcd + 8 * (idx >> 5);
+      int p = idx & 31;
+      tm = (st / st_cols) * 8 + (p >> 2);
+      tn = (st % st_cols) * 4 + (p & 3);
+    } else {
+      const int nxcd = 8;
+      int q = nwg / nxcd, r = nwg % nxcd;
+      int xcd = wgid % nxcd, idx = wgid / nxcd;
+      wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+      tm = wgid / tiles_n;
+      tn = wgid % tiles_n;
+    }
+  }
+  int row0 = tm * BM2;
+  int col0 = tn * BN2;
+
+  __shared__ double As[3][BM2][BKT + 1];
+  __shared__ double Bs[3][BKT][BN2 + 1];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int l15 = lane & 15;
+  const int lk = lane >> 4;
+
+  f64x4 acc[4][2] = {};
+
+  constexpr int AE = BM2 * BKT / THREADS2;
+  constexpr int BE = BKT * BN2 / THREADS2;
+  const int a_m = tid / (BKT / AE);
+  const int a_k = (tid % (BKT / AE)) * AE;
+  const int b_k = tid / (BN2 / BE);
+  const int b_n = (tid % (BN2 / BE)) * BE;
+
+  double a_reg[AE], b_reg[BE];
+
+  auto issue_loads = [&](int k0) {
+    int gr = row0 + a_m;
+#pragma unroll
+    for (int j = 0; j < AE; j++) {
+      int gk = k0 + a_k + j;
+      a_reg[j] = (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
+    }
+    int gk = k0 + b_k;
+#pragma unroll
+    for (int j = 0; j < BE; j++) {
+      int gn = col0 + b_n + j;
+      b_reg[j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < AE; j++) As[buf][a_m][a_k + j] = a_reg[j];
+#pragma unroll
+    for (int j = 0; j < BE; j++) Bs[buf][b_k][b_n + j] = b_reg[j];
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 32;
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < BKT; ks += 4) {
+      double a0 = As[buf][am0 + l15][ks + lk];
+      double a1 = As[buf][am0 + 16 + l15][ks + lk];
+      double a2 = As[buf][am0 + 32 + l15][ks + lk];
+      double a3 = As[buf][am0 + 48 + l15][ks + lk];
+      double b0 = Bs[buf][ks + lk][bn0 + l15];
+      double b1 = Bs[buf][ks + lk][bn0 + 16 + l15];
+      acc[0][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b0, acc[0][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b0, acc[2][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b1, acc[3][1], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b0, acc[1][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b0, acc[3][0], 0, 0, 0);
+    }
+  };
+
+  // prologue: buffer 0 staged and visible, buffer 1's loads in regs
+  issue_loads(0);
+  write_lds(0);
+  if (BKT < K) issue_loads(BKT);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BKT) {
+    bool have_next = k0 + BKT < K;
+    if (have_next) {
+      // stage k0+BKT NOW (its loads are already in registers); its
+      // buffer was last read two barriers ago, so this is safe, and
+      // these ds_writes overlap the MFMA stream below
+      write_lds((cur + 1) % 3);
+      if (k0 + 2 * BKT < K) issue_loads(k0 + 2 * BKT);
+    }
+    compute_tile(cur);
+    cur = (cur + 1) % 3;
+    __syncthreads();
+  }
+
+  const int crow0 = row0 + wave_m * 64 + (lane >> 4);
+  const int ccol0 = col0 + wave_n * 32 + l15;
+#pragma unroll
+  for (int mt = 0; mt < 4; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 2; nt++) {
+      int col = ccol0 + nt * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int row = crow0 + mt * 16 + 4 * reg;
+        if (row < M) C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 4-wave variant: 128^2 block as 2x2 waves of 64x64 C each -> 16 MFMAs
+// per 8 ds_reads (2:1, vs 1.33:1 for the 8-wave shape). Probes whether
+// the residual bubble is operand-read latency. BK=8, 3 LDS buffers
+// (~52 KB -> 3 blocks/CU = 3 waves/SIMD; acc = 128 VGPRs).
+// ---------------------------------------------------------------------------
+constexpr int THREADS3 = 256;
+
+template <int BKT>
+__global__ __launch_bounds__(THREADS3) void gemm_f64_128w_kernel(
+    const double* __restrict__ A, const double* __restrict__ B,
+    double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  int tm, tn;
+  {
+    int nwg = tiles_m * tiles_n;
+    int wgid = blockIdx.x;
+    const int n_st = (tiles_m / 8) * (tiles_n / 4);
+    if (tiles_m % 8 == 0 && tiles_n % 4 == 0 && n_st % 8 == 0) {
+      const int st_cols = tiles_n / 4;
+      int xcd = wgid % 8, idx = wgid / 8;
+      int st = xcd + 8 * (idx >> 5);
+      int p = idx & 31;
+      tm = (st / st_cols) * 8 + (p >> 2);
+      tn = (st % st_cols) * 4 + (p & 3);
+    } else {
+      const int nxcd = 8;
+      int q = nwg / nxcd, r = nwg % nxcd;
+      int xcd = wgid % nxcd, idx = wgid / nxcd;
+      wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+      tm = wgid / tiles_n;
+      tn = wgid % tiles_n;
+    }
+  }
+  int row0 = tm * BM2;
+  int col0 = tn * BN2;
+
+  __shared__ double As[3][BM2][BKT + 1];
+  __shared__ double Bs[3][BKT][BN2 + 1];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;        // 0..3
+  const int wave_m = wave >> 1;     // 2x2 arrangement
+  const int wave_n = wave & 1;
+  const int l15 = lane & 15;
+  const int lk = lane >> 4;
+
+  f64x4 acc[4][4] = {};
+
+  constexpr int AE = BM2 * BKT / THREADS3;
+  constexpr int BE = BKT * BN2 / THREADS3;
+  const int a_m = tid / (BKT / AE);
+  const int a_k = (tid % (BKT / AE)) * AE;
+  const int b_k = tid / (BN2 / BE);
+  const int b_n = (tid % (BN2 / BE)) * BE;
+
+  double a_reg[AE], b_reg[BE];
+
+  auto issue_loads = [&](int k0) {
+    int gr = row0 + a_m;
+#pragma unroll
+    for (int j = 0; j < AE; j++) {
+      int gk = k0 + a_k + j;
+      a_reg[j] = (gr < M && gk < K) ? A[(int64_t)gr * K + gk] : 0.0;
+    }
+    int gk = k0 + b_k;
+#pragma unroll
+    for (int j = 0; j < BE; j++) {
+      int gn = col0 + b_n + j;
+      b_reg[j] = (gk < K && gn < N) ? B[(int64_t)gk * N + gn] : 0.0;
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < AE; j++) As[buf][a_m][a_k + j] = a_reg[j];
+#pragma unroll
+    for (int j = 0; j < BE; j++) Bs[buf][b_k][b_n + j] = b_reg[j];
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 64;
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < BKT; ks += 4) {
+      double a0 = As[buf][am0 + l15][ks + lk];
+      double a1 = As[buf][am0 + 16 + l15][ks + lk];
+      double a2 = As[buf][am0 + 32 + l15][ks + lk];
+      double a3 = As[buf][am0 + 48 + l15][ks + lk];
+      double b0 = Bs[buf][ks + lk][bn0 + l15];
+      double b1 = Bs[buf][ks + lk][bn0 + 16 + l15];
+      double b2 = Bs[buf][ks + lk][bn0 + 32 + l15];
+      double b3 = Bs[buf][ks + lk][bn0 + 48 + l15];
+#pragma unroll
+      for (int d = 0; d < 4; d++) {
+        // diagonal order: consecutive MFMAs never share a source
+        acc[0][d] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+            a0, d == 0 ? b0 : d == 1 ? b1 : d == 2 ? b2 : b3, acc[0][d], 0, 0, 0);
+        acc[1][(d + 1) & 3] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+            a1, d == 3 ? b0 : d == 0 ? b1 : d == 1 ? b2 : b3, acc[1][(d + 1) & 3], 0, 0, 0);
+        acc[2][(d + 2) & 3] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+            a2, d == 2 ? b0 : d == 3 ? b1 : d == 0 ? b2 : b3, acc[2][(d + 2) & 3], 0, 0, 0);
+        acc[3][(d + 3) & 3] = __builtin_amdgcn_mfma_f64_16x16x4f64(
+            a3, d == 1 ? b0 : d == 2 ? b1 : d == 3 ? b2 : b3, acc[3][(d + 3) & 3], 0, 0, 0);
+      }
+    }
+  };
+
+  issue_loads(0);
+  write_lds(0);
+  if (BKT < K) issue_loads(BKT);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BKT) {
+    bool have_next = k0 + BKT < K;
+    if (have_next) {
+      write_lds((cur + 1) % 3);
+      if (k0 + 2 * BKT < K) issue_loads(k0 + 2 * BKT);
+    }
+    compute_tile(cur);
+    cur = (cur + 1) % 3;
+    __syncthreads();
+  }
+
+  const int crow0 = row0 + wave_m * 64 + (lane >> 4);
+  const int ccol0 = col0 + wave_n * 64 + l15;
+#pragma unroll
+  for (int mt = 0; mt < 4; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 4; nt++) {
+      int col = ccol0 + nt * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int row = crow0 + mt * 16 + 4 * reg;
+        if (row < M) C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
                      int k, hipStream_t stream) {
-  // default: 128^2 tile, BK=16; APP_F64_VARIANT: "s" = the old 64^2
-  // kernel, "8"/"32" = BK sweep on the 128^2 tile (occupancy experiment)
+  // default: 128^2 tile, BK=8, 3-buffer write-early pipeline (63.7 TF
+  // @8192^3 — best of the r02 ladder; profiles/NOTES.md). APP_F64_VARIANT:
+  // "d" = BK=16 double-buffer (prior default), "s" = the old 64^2 kernel,
+  // "8"/"32" = BK sweep on the double-buffered 128^2 tile, "q" = BK=16
+  // 3-buffer (1 block/CU: occupancy loss), "w" = 4-wave 64x64 wave tiles
+  // (2:1 MFMA:ds_read but 3 waves/SIMD: latency-coverage loss)
   const char* v = getenv("APP_F64_VARIANT");
   if (!(v && v[0] == 's')) {
     int tiles_m2 = (m + BM2 - 1) / BM2;
     int tiles_n2 = (n + BN2 - 1) / BN2;
-    if (v && v[0] == '8')
+    if (v && v[0] == 'w')
+      hipLaunchKernelGGL((gemm_f64_128w_kernel<8>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS3), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    else if (v && v[0] == 'q')
+      hipLaunchKernelGGL((gemm_f64_128p_kernel<16>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS2), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    else if (v && v[0] == '8')
       hipLaunchKernelGGL((gemm_f64_128_kernel<8>), dim3(tiles_m2 * tiles_n2),
                          dim3(THREADS2), 0, stream, a, b, c, m, n, k,
                          tiles_m2, tiles_n2);
@@ -293,8 +584,12 @@ void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
       hipLaunchKernelGGL((gemm_f64_128_kernel<32>), dim3(tiles_m2 * tiles_n2),
                          dim3(THREADS2), 0, stream, a, b, c, m, n, k,
                          tiles_m2, tiles_n2);
-    else
+    else if (v && v[0] == 'd')
       hipLaunchKernelGGL((gemm_f64_128_kernel<16>), dim3(tiles_m2 * tiles_n2),
+                         dim3(THREADS2), 0, stream, a, b, c, m, n, k,
+                         tiles_m2, tiles_n2);
+    else
+      hipLaunchKernelGGL((gemm_f64_128p_kernel<8>), dim3(tiles_m2 * tiles_n2),
                          dim3(THREADS2), 0, stream, a, b, c, m, n, k,
                          tiles_m2, tiles_n2);
     HIP_CHECK(hipGetLastError());
