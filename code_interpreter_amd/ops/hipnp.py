@@ -1280,6 +1280,27 @@ class DeviceArray:
                 _, idx = a._device_sorted(True)
                 return idx
             return func(a.materialize(), kth, **kwargs)
+        if func is _np.unique and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and not kwargs:
+            a = args[0]
+            if a._sort_routable(None, None, None, {}):
+                # numpy's unique IS sort + adjacent-compare
+                # (lib/arraysetops); NaNs sort last and collapse to one
+                # (equal_nan=True is the numpy>=1.21 default)
+                host = a._device_sorted(False).materialize()
+                if host.size == 0:
+                    return host
+                keep = _np.empty(host.size, dtype=bool)
+                keep[0] = True
+                _np.not_equal(host[1:], host[:-1], out=keep[1:])
+                n_nan = int(_np.isnan(host[-1:])[0] and
+                            _np.isnan(host).sum())
+                if n_nan > 1:
+                    keep[host.size - n_nan + 1:] = False
+                    keep[host.size - n_nan] = True
+                return host[keep]
+            return _np.unique(a.materialize())
         if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ):

@@ -863,3 +863,18 @@ def test_scalar_getitem_fetches_one_element(fake):
     fake.calls.clear()
     assert small[4] == host[4]
     assert "download_slice" not in fake.calls
+
+
+def test_unique_via_device_sort(fake):
+    host = np.random.default_rng(18).integers(0, 50, 2000).astype(np.float64)
+    host[7] = np.nan
+    host[99] = np.nan
+    x = _device(fake, host)
+    u = np.unique(x)
+    assert "sort" in fake.calls
+    np.testing.assert_array_equal(u, np.unique(host))
+    # kwargs fall back to host numpy untouched
+    vals, counts = np.unique(_device(fake, host), return_counts=True)
+    rv, rc = np.unique(host, return_counts=True)
+    np.testing.assert_array_equal(vals, rv)
+    np.testing.assert_array_equal(counts, rc)

@@ -856,3 +856,11 @@ def test_device_array_sort_route(hnp):
     idx = np.argsort(x)
     assert isinstance(idx, hnp.DeviceArray) and idx.dtype == np.int64
     np.testing.assert_array_equal(idx.materialize(), np.argsort(a))
+
+
+def test_unique_via_device_sort_gpu(hnp):
+    x = hnp.rand(1_000_000, seed=5)
+    q = (x * 100.0).astype(np.float64)  # duplicate-heavy after floor-ish
+    host = np.asarray(q).copy()
+    u = np.unique(q)
+    np.testing.assert_array_equal(u, np.unique(host))
