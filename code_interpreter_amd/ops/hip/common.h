@@ -51,6 +51,11 @@ void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         hipStream_t stream);
 // radix sort (np.sort/argsort): see sort.hip header for scratch sizes
 int64_t radix_sort_nchunks(int64_t n);
+void launch_radix_sort_rows(DType dt, const void* in, void* out,
+                            void* idx_out, void* keys_a, void* keys_b,
+                            void* idx_a, void* idx_b, void* counts,
+                            void* dig_scratch, int64_t rows, int64_t cols,
+                            bool want_idx, hipStream_t s);
 void launch_radix_sort(DType dt, const void* in, void* out, void* idx_out,
                        void* keys_a, void* keys_b, void* idx_a, void* idx_b,
                        void* counts, void* dig_scratch, int64_t n,

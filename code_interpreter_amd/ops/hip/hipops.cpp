@@ -647,6 +647,53 @@ PyObject* py_sort(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// sort2d(h, dtype, rows, cols, want_idx): independent stable sort of
+// every row (np.sort/np.argsort axis=-1 on 2-D). Returns handle |
+// (handle, int64 per-row-position handle).
+PyObject* py_sort2d(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt, want_idx;
+  long long rows, cols;
+  if (!PyArg_ParseTuple(args, "KiLLi", &h, &dt, &rows, &cols, &want_idx))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (rows < 1 || cols < 1 || rows * cols > (1ll << 31))
+    throw std::runtime_error("bad sort2d shape");
+  const int64_t n = rows * cols;
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  int64_t nchunks = radix_sort_nchunks(n);
+  int64_t nseg = (nchunks + 255) / 256;
+  void* out = nullptr;
+  void* idx_res = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n * esize);
+  void* keys_a = pool_alloc(n * esize);
+  void* keys_b = pool_alloc(n * esize);
+  void* idx_a = pool_alloc(n * 8);
+  void* idx_b = pool_alloc(n * 8);
+  void* counts = pool_alloc(nchunks * 256 * 4);
+  void* dig = pool_alloc(256 * 8 + nseg * 256 * 4);
+  if (want_idx) idx_res = pool_alloc(n * 8);
+  launch_radix_sort_rows(dtype, in.ptr, out, idx_res, keys_a, keys_b, idx_a,
+                         idx_b, counts, dig, rows, cols, want_idx != 0,
+                         g.compute);
+  pool_free(keys_a, n * esize);
+  pool_free(keys_b, n * esize);
+  pool_free(idx_a, n * 8);
+  pool_free(idx_b, n * 8);
+  pool_free(counts, nchunks * 256 * 4);
+  pool_free(dig, 256 * 8 + nseg * 256 * 4);
+  NOGIL_END
+  unsigned long long hout = register_buf(out, n * esize);
+  if (!want_idx) return PyLong_FromUnsignedLongLong(hout);
+  unsigned long long hidx = register_buf(idx_res, n * 8);
+  return Py_BuildValue("(KK)", hout, hidx);
+  WRAP_END
+}
+
 // cumsum(h, dtype, n) -> handle (same dtype)
 PyObject* py_cumsum(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1138,6 +1185,8 @@ PyMethodDef methods[] = {
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
+    {"sort2d", py_sort2d, METH_VARARGS,
+     "sort2d(h, dtype, rows, cols, want_idx) -> handle | (handle, idx)"},
     {"sort", py_sort, METH_VARARGS,
      "sort(h, dtype, n, want_idx) -> handle | (handle, int64 idx handle)"},
     {"mask_logic", py_mask_logic, METH_VARARGS,

@@ -44,9 +44,13 @@ __device__ __forceinline__ float f32_of_key32(uint32_t k) {
   return __uint_as_float(u);
 }
 
-template <typename K>
-__global__ void radix_hist_kernel(const K* __restrict__ keys, int64_t n,
-                                  int shift,
+// ROWDIG: digit taken from the element's row id (idx[i] / L) instead of
+// the key -- the row passes of the 2-D (axis=-1) sort. LSD stability
+// makes "key passes, then row passes" == an independent sort per row.
+template <typename K, bool ROWDIG>
+__global__ void radix_hist_kernel(const K* __restrict__ keys,
+                                  const long long* __restrict__ idx,
+                                  unsigned int L, int64_t n, int shift,
                                   unsigned int* __restrict__ counts,
                                   int64_t nchunks) {
   __shared__ unsigned int hist[kWavesPerBlock][256];
@@ -60,7 +64,8 @@ __global__ void radix_hist_kernel(const K* __restrict__ keys, int64_t n,
     int64_t start = chunk * kSortChunk;
     int64_t end = start + kSortChunk < n ? start + kSortChunk : n;
     for (int64_t i = start + lane; i < end; i += 64) {
-      unsigned d = (unsigned)((keys[i] >> shift) & 0xff);
+      unsigned d = ROWDIG ? (((unsigned)idx[i] / L) >> shift) & 0xff
+                          : (unsigned)((keys[i] >> shift) & 0xff);
       atomicAdd(&hist[wave][d], 1u);
     }
     unsigned int* out = counts + chunk * 256;
@@ -183,12 +188,13 @@ __global__ void radix_scatter_kernel(const K* __restrict__ in_keys,
 // element runs (64-128 B), and the argsort payload is gathered through
 // a u16 in-chunk origin (the 32 KiB in_idx window stays cache-
 // resident) instead of scattered 8-B stores.
-template <typename K, bool PAYLOAD>
+template <typename K, bool PAYLOAD, bool ROWDIG = false>
 __global__ void radix_scatter_block_kernel(
     const K* __restrict__ in_keys, K* __restrict__ out_keys,
     const long long* __restrict__ in_idx, long long* __restrict__ out_idx,
     int64_t n, int shift, const unsigned int* __restrict__ bases,
-    int64_t nchunks) {
+    int64_t nchunks, unsigned int L = 0) {
+  static_assert(!ROWDIG || PAYLOAD, "row digits come from the payload");
   constexpr int kQuarter = kSortChunk / 4;  // elements per wave
   constexpr int kTiles = kQuarter / 64;     // 64-lane tiles per wave
   __shared__ K stage[kSortChunk];
@@ -216,7 +222,11 @@ __global__ void radix_scatter_block_kernel(
     int e = q0 + t * 64 + lane;
     bool valid = e < len;
     key[t] = valid ? in_keys[start + e] : (K)0;
-    unsigned d = (unsigned)((key[t] >> shift) & 0xff);
+    unsigned d;
+    if (ROWDIG)  // valid-guarded: in_idx[start+e] is OOB past len
+      d = valid ? (((unsigned)in_idx[start + e] / L) >> shift) & 0xff : 0u;
+    else
+      d = (unsigned)((key[t] >> shift) & 0xff);
     uint64_t same = __ballot(valid);
 #pragma unroll
     for (int b = 0; b < 8; b++) {
@@ -255,7 +265,9 @@ __global__ void radix_scatter_block_kernel(
   for (int t = 0; t < kTiles; t++) {
     int e = q0 + t * 64 + lane;
     if (e < len) {
-      unsigned d = (unsigned)((key[t] >> shift) & 0xff);
+      unsigned d = ROWDIG
+          ? (((unsigned)in_idx[start + e] / L) >> shift) & 0xff
+          : (unsigned)((key[t] >> shift) & 0xff);
       unsigned pos = pref[d] + wcnt[wave][d] + rank[t];
       stage[pos] = key[t];
       if (PAYLOAD) origin[pos] = (unsigned short)e;
@@ -268,10 +280,12 @@ __global__ void radix_scatter_block_kernel(
     int li = (int)threadIdx.x + t * 256;
     if (li < len) {
       K k = stage[li];
-      unsigned d = (unsigned)((k >> shift) & 0xff);
+      long long iv = PAYLOAD ? in_idx[start + origin[li]] : 0;
+      unsigned d = ROWDIG ? (((unsigned)iv / L) >> shift) & 0xff
+                          : (unsigned)((k >> shift) & 0xff);
       unsigned gpos = base_row[d] + (unsigned)li - pref[d];
       out_keys[gpos] = k;
-      if (PAYLOAD) out_idx[gpos] = in_idx[start + origin[li]];
+      if (PAYLOAD) out_idx[gpos] = iv;
     }
   }
 }
@@ -339,8 +353,9 @@ static void radix_sort_impl(const T* in, T* out, long long* idx_out,
   int grid_s = (int)((nseg * 256 + 255) / 256);
   for (int p = 0; p < passes; p++) {
     int shift = p * 8;
-    hipLaunchKernelGGL((radix_hist_kernel<K>), dim3(grid_c), dim3(256), 0, s,
-                       ka, n, shift, (unsigned int*)counts, nchunks);
+    hipLaunchKernelGGL((radix_hist_kernel<K, false>), dim3(grid_c), dim3(256),
+                       0, s, ka, nullptr, 0u, n, shift,
+                       (unsigned int*)counts, nchunks);
     hipLaunchKernelGGL(radix_segsum_kernel, dim3(grid_s), dim3(256), 0, s,
                        (const unsigned int*)counts, nchunks, nseg, segsum);
     hipLaunchKernelGGL(radix_segscan_kernel, dim3(1), dim3(256), 0, s,
@@ -379,6 +394,88 @@ static void radix_sort_impl(const T* in, T* out, long long* idx_out,
   HIP_CHECK(hipGetLastError());
 }
 
+// per-row argsort positions: idx holds flat positions in [0, R*L);
+// the user-facing index is position-within-row
+__global__ void rowpos_kernel(const long long* __restrict__ idx,
+                              long long* __restrict__ out, unsigned int L,
+                              int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (long long)((unsigned)idx[i] % L);
+}
+
+// 2-D axis=-1 sort: LSD key passes with the flat index as payload, then
+// LSD passes over the row id (idx / L). Stability makes the composite
+// equal an independent stable sort of every row. Row passes use the
+// block scatter unconditionally (the wave variant has no ROWDIG mode).
+template <typename T, typename K>
+static void radix_sort_rows_impl(const T* in, T* out, long long* idx_out,
+                                 void* keys_a, void* keys_b, void* idx_a,
+                                 void* idx_b, void* counts, void* dig_scratch,
+                                 int64_t rows, int64_t cols, bool want_idx,
+                                 hipStream_t s) {
+  const int64_t n = rows * cols;
+  int64_t nchunks = (n + kSortChunk - 1) / kSortChunk;
+  int grid_c = (int)((nchunks + kWavesPerBlock - 1) / kWavesPerBlock);
+  int grid_e = (int)std::min<int64_t>((n + 255) / 256, 4096);
+
+  K* ka = (K*)keys_a;
+  K* kb = (K*)keys_b;
+  long long* ia = (long long*)idx_a;
+  long long* ib = (long long*)idx_b;
+
+  hipLaunchKernelGGL((sort_encode_kernel<T, K>), dim3(grid_e), dim3(256), 0,
+                     s, in, ka, ia, n, 1);
+  int row_passes = 1;
+  while ((rows - 1) >> (8 * row_passes)) row_passes++;
+  const int passes = (int)sizeof(K) + row_passes;
+  int64_t nseg = (nchunks + kSeg - 1) / kSeg;
+  unsigned int* segsum = (unsigned int*)((char*)dig_scratch + 256 * 8);
+  int grid_s = (int)((nseg * 256 + 255) / 256);
+  const unsigned int L = (unsigned int)cols;
+  for (int p = 0; p < passes; p++) {
+    bool rowdig = p >= (int)sizeof(K);
+    int shift = (rowdig ? p - (int)sizeof(K) : p) * 8;
+    if (rowdig)
+      hipLaunchKernelGGL((radix_hist_kernel<K, true>), dim3(grid_c),
+                         dim3(256), 0, s, ka, ia, L, n, shift,
+                         (unsigned int*)counts, nchunks);
+    else
+      hipLaunchKernelGGL((radix_hist_kernel<K, false>), dim3(grid_c),
+                         dim3(256), 0, s, ka, nullptr, 0u, n, shift,
+                         (unsigned int*)counts, nchunks);
+    hipLaunchKernelGGL(radix_segsum_kernel, dim3(grid_s), dim3(256), 0, s,
+                       (const unsigned int*)counts, nchunks, nseg, segsum);
+    hipLaunchKernelGGL(radix_segscan_kernel, dim3(1), dim3(256), 0, s,
+                       segsum, nseg, (unsigned long long*)dig_scratch);
+    hipLaunchKernelGGL(radix_fold_kernel, dim3(grid_s), dim3(256), 0, s,
+                       (unsigned int*)counts, nchunks, nseg, segsum);
+    if (rowdig)
+      hipLaunchKernelGGL((radix_scatter_block_kernel<K, true, true>),
+                         dim3((unsigned)nchunks), dim3(256), 0, s, ka, kb,
+                         ia, ib, n, shift, (const unsigned int*)counts,
+                         nchunks, L);
+    else if (use_block_scatter())
+      hipLaunchKernelGGL((radix_scatter_block_kernel<K, true>),
+                         dim3((unsigned)nchunks), dim3(256), 0, s, ka, kb,
+                         ia, ib, n, shift, (const unsigned int*)counts,
+                         nchunks);
+    else
+      hipLaunchKernelGGL((radix_scatter_kernel<K, true>), dim3(grid_c),
+                         dim3(256), 0, s, ka, kb, ia, ib, n, shift,
+                         (const unsigned int*)counts, nchunks);
+    K* tk = ka; ka = kb; kb = tk;
+    long long* ti = ia; ia = ib; ib = ti;
+  }
+  hipLaunchKernelGGL((sort_decode_kernel<T, K>), dim3(grid_e), dim3(256), 0,
+                     s, ka, out, n);
+  if (want_idx)
+    hipLaunchKernelGGL(rowpos_kernel, dim3(grid_e), dim3(256), 0, s, ia,
+                       idx_out, L, n);
+  HIP_CHECK(hipGetLastError());
+}
+
 }  // namespace
 
 // scratch requirements (bytes), all caller-allocated:
@@ -387,6 +484,23 @@ static void radix_sort_impl(const T* in, T* out, long long* idx_out,
 //   counts:        nchunks(n) * 256 * 4
 //   dig_scratch:   256*8 + nseg(n) * 256 * 4   (dig[256] then segsum)
 int64_t radix_sort_nchunks(int64_t n) { return (n + kSortChunk - 1) / kSortChunk; }
+
+// 2-D axis=-1 variant: idx_a/idx_b are REQUIRED (the row id rides in
+// the payload); idx_out only read when want_idx. rows*cols <= 2^31.
+void launch_radix_sort_rows(DType dt, const void* in, void* out,
+                            void* idx_out, void* keys_a, void* keys_b,
+                            void* idx_a, void* idx_b, void* counts,
+                            void* dig_scratch, int64_t rows, int64_t cols,
+                            bool want_idx, hipStream_t s) {
+  if (dt == DType::F64)
+    radix_sort_rows_impl<double, uint64_t>(
+        (const double*)in, (double*)out, (long long*)idx_out, keys_a, keys_b,
+        idx_a, idx_b, counts, dig_scratch, rows, cols, want_idx, s);
+  else
+    radix_sort_rows_impl<float, uint32_t>(
+        (const float*)in, (float*)out, (long long*)idx_out, keys_a, keys_b,
+        idx_a, idx_b, counts, dig_scratch, rows, cols, want_idx, s);
+}
 
 void launch_radix_sort(DType dt, const void* in, void* out, void* idx_out,
                        void* keys_a, void* keys_b, void* idx_a, void* idx_b,

@@ -192,6 +192,18 @@ class Connection(threading.Thread):
                 self._own(m["h"]), m["off"], m["nbytes"]
             )
             return {"ok": True}, data
+        if op == "sort2d":
+            r = _hipops.sort2d(
+                self._own(m["h"]), m["dtype"], m["rows"], m["cols"],
+                m["want_idx"]
+            )
+            if m["want_idx"]:
+                h, hi = r
+                self.handles.add(h)
+                self.handles.add(hi)
+                return {"ok": True, "h": h, "hi": hi}, b""
+            self.handles.add(r)
+            return {"ok": True, "h": r}, b""
         if op == "sort":
             r = _hipops.sort(
                 self._own(m["h"]), m["dtype"], m["n"], m["want_idx"]

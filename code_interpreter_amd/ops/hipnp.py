@@ -333,6 +333,15 @@ class RemoteBackend:
             return out["h"], out["hi"]
         return out["h"]
 
+    def sort2d(self, h, dtype, rows, cols, want_idx):
+        out = self._call(
+            {"op": "sort2d", "h": h, "dtype": dtype, "rows": rows,
+             "cols": cols, "want_idx": want_idx}
+        )
+        if want_idx:
+            return out["h"], out["hi"]
+        return out["h"]
+
     def mask_logic(self, ha, hb, n, lop):
         return self._call(
             {"op": "mask_logic", "ha": ha, "hb": hb, "n": n, "lop": lop}
@@ -973,23 +982,34 @@ class DeviceArray:
         return self.materialize().cumsum(axis=axis, **kwargs)
 
     def _sort_routable(self, axis, kind, order, kwargs):
-        """np.sort/argsort route: 1-D f32/f64, default comparator. The
-        device sort is an LSD radix sort, so it is stable -- every numpy
-        `kind` is satisfied."""
-        return (
-            _dtype_code(self.dtype) is not None
-            and not kwargs
-            and order is None
-            and kind in (None, "stable", "quicksort", "mergesort", "heapsort")
-            and len(self.shape) == 1
-            and (axis is None or self._norm_axis(axis, 1) == 0)
-        )
+        """np.sort/argsort route: f32/f64, default comparator, flat 1-D
+        or 2-D along the last axis. The device sort is an LSD radix
+        sort, so it is stable -- every numpy `kind` is satisfied."""
+        if (
+            _dtype_code(self.dtype) is None
+            or kwargs
+            or order is not None
+            or kind not in (None, "stable", "quicksort", "mergesort",
+                            "heapsort")
+        ):
+            return False
+        if len(self.shape) == 1:
+            return axis is None or self._norm_axis(axis, 1) == 0
+        if len(self.shape) == 2 and axis is not None:
+            return self._norm_axis(axis, 2) == 1 and self.size <= (1 << 31)
+        return False
 
     def _device_sorted(self, want_idx):
-        r = backend().sort(
-            self._dev_handle(), _dtype_code(self.dtype), self.size,
-            1 if want_idx else 0,
-        )
+        code = _dtype_code(self.dtype)
+        if len(self.shape) == 2:
+            rows, cols = self.shape
+            r = backend().sort2d(
+                self._dev_handle(), code, rows, cols, 1 if want_idx else 0
+            )
+        else:
+            r = backend().sort(
+                self._dev_handle(), code, self.size, 1 if want_idx else 0
+            )
         if want_idx:
             h, hi = r
             return (

@@ -118,6 +118,16 @@ class FakeBackend:
             return self._new(a[idx]), self._new(idx.astype(np.int64))
         return self._new(np.sort(a, kind="stable"))
 
+    def sort2d(self, h, dtype, rows, cols, want_idx):
+        self.calls.append("sort2d")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: rows * cols]
+        a = a.reshape(rows, cols)
+        if want_idx:
+            idx = np.argsort(a, axis=-1, kind="stable")
+            return (self._new(np.take_along_axis(a, idx, -1)),
+                    self._new(idx.astype(np.int64)))
+        return self._new(np.sort(a, axis=-1, kind="stable"))
+
     def mask_logic(self, ha, hb, n, lop):
         self.calls.append("mask_logic")
         a = self.bufs[ha].view(np.uint8).reshape(-1)[:n].astype(bool)
@@ -878,3 +888,24 @@ def test_unique_via_device_sort(fake):
     rv, rc = np.unique(host, return_counts=True)
     np.testing.assert_array_equal(vals, rv)
     np.testing.assert_array_equal(counts, rc)
+
+
+def test_sort2d_rows_on_device(fake):
+    host = np.random.default_rng(19).random((40, 64))
+    x = _device(fake, host)
+    srt = np.sort(x, axis=-1)
+    assert isinstance(srt, hipnp.DeviceArray) and srt.shape == (40, 64)
+    assert "sort2d" in fake.calls
+    np.testing.assert_array_equal(srt.materialize(), np.sort(host, axis=-1))
+    idx = np.argsort(_device(fake, host), axis=1)
+    assert isinstance(idx, hipnp.DeviceArray) and idx.dtype == np.int64
+    np.testing.assert_array_equal(
+        idx.materialize(), np.argsort(host, axis=1, kind="stable"))
+    # axis=0 falls back to host
+    s0 = np.sort(_device(fake, host), axis=0)
+    assert isinstance(s0, np.ndarray)
+    np.testing.assert_array_equal(s0, np.sort(host, axis=0))
+    # in-place method, 2-D default axis
+    y = _device(fake, host)
+    assert y.sort() is None
+    np.testing.assert_array_equal(y.materialize(), np.sort(host, axis=-1))

@@ -864,3 +864,51 @@ def test_unique_via_device_sort_gpu(hnp):
     host = np.asarray(q).copy()
     u = np.unique(q)
     np.testing.assert_array_equal(u, np.unique(host))
+
+
+@pytest.mark.parametrize("dtype,code", [(np.float64, 1), (np.float32, 0)])
+@pytest.mark.parametrize("shape", [(1, 5), (3, 4097), (517, 1931), (70000, 37)])
+def test_sort2d_matches_numpy(hip, dtype, code, shape):
+    rows, cols = shape
+    rng = np.random.default_rng(rows * 31 + cols)
+    a = (rng.standard_normal((rows, cols)) * 50).astype(dtype)
+    h = hip.upload(a)
+    hs, hi = hip.sort2d(h, code, rows, cols, 1)
+    out = np.empty((rows, cols), dtype)
+    idx = np.empty((rows, cols), np.int64)
+    hip.download(hs, out)
+    hip.download(hi, idx)
+    for x in (h, hs, hi):
+        hip.free(x)
+    np.testing.assert_array_equal(out, np.sort(a, axis=-1))
+    np.testing.assert_array_equal(
+        idx, np.argsort(a, axis=-1, kind="stable"))
+
+
+def test_sort2d_stability_and_nan(hip):
+    rows, cols = 200, 5000
+    rng = np.random.default_rng(23)
+    a = rng.integers(0, 9, (rows, cols)).astype(np.float64)
+    a[rng.random((rows, cols)) < 0.01] = np.nan
+    h = hip.upload(a)
+    hs, hi = hip.sort2d(h, 1, rows, cols, 1)
+    out = np.empty((rows, cols))
+    idx = np.empty((rows, cols), np.int64)
+    hip.download(hs, out)
+    hip.download(hi, idx)
+    for x in (h, hs, hi):
+        hip.free(x)
+    ref = np.sort(a, axis=-1)
+    nan_ref = np.isnan(ref)
+    assert np.array_equal(np.isnan(out), nan_ref)
+    np.testing.assert_array_equal(out[~nan_ref], ref[~nan_ref])
+    np.testing.assert_array_equal(
+        idx, np.argsort(a, axis=-1, kind="stable"))
+
+
+def test_device_array_sort2d_route(hnp):
+    x = hnp.rand(300, 4000, seed=77)
+    a = np.asarray(x).copy()
+    s = np.sort(x, axis=-1)
+    assert isinstance(s, hnp.DeviceArray)
+    np.testing.assert_array_equal(s.materialize(), np.sort(a, axis=-1))
