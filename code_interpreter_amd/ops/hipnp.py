@@ -1339,6 +1339,36 @@ class DeviceArray:
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func is _np.linalg.norm and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ):
+            a = args[0]
+            ord_ = kwargs.get("ord")
+            axis = kwargs.get("axis")
+            code = _dtype_code(a.dtype)
+            extra = set(kwargs) - {"ord", "axis"}
+            if code is not None and not extra:
+                if axis is None and (
+                    ord_ is None
+                    or (ord_ == 2 and a.ndim == 1)
+                    or (ord_ == "fro" and a.ndim == 2)
+                ):
+                    # flat 2-norm / Frobenius: fused square+sum kernel
+                    return a.dtype.type(float(a.square_sum()) ** 0.5)
+                if (
+                    isinstance(axis, int)
+                    and ord_ in (None, 2)
+                    and a.ndim == 2
+                ):
+                    sq = a._unary("square")
+                    if sq is not NotImplemented:
+                        ssum = sq.sum(axis=axis)
+                        if isinstance(ssum, DeviceArray):
+                            r = ssum._unary("sqrt")
+                            if r is not NotImplemented:
+                                return r
+            host = a.materialize()
+            return _np.linalg.norm(host, **kwargs)
         _reductions = {
             _np.max: "max", _np.amax: "max",
             _np.min: "min", _np.amin: "min",

@@ -909,3 +909,30 @@ def test_sort2d_rows_on_device(fake):
     y = _device(fake, host)
     assert y.sort() is None
     np.testing.assert_array_equal(y.materialize(), np.sort(host, axis=-1))
+
+
+def test_linalg_norm_on_device(fake):
+    host = np.random.default_rng(20).random(5000) - 0.3
+    x = _device(fake, host)
+    assert float(np.linalg.norm(x)) == pytest.approx(
+        np.linalg.norm(host), rel=1e-12)
+    assert float(np.linalg.norm(x, ord=2)) == pytest.approx(
+        np.linalg.norm(host, 2), rel=1e-12)
+    m = np.random.default_rng(21).random((200, 300))
+    y = _device(fake, m)
+    assert float(np.linalg.norm(y)) == pytest.approx(
+        np.linalg.norm(m), rel=1e-12)
+    assert float(np.linalg.norm(y, ord="fro")) == pytest.approx(
+        np.linalg.norm(m, "fro"), rel=1e-12)
+    r = np.linalg.norm(y, axis=1)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(
+        r.materialize(), np.linalg.norm(m, axis=1), rtol=1e-12)
+    r0 = np.linalg.norm(y, axis=0)
+    np.testing.assert_allclose(
+        np.asarray(r0), np.linalg.norm(m, axis=0), rtol=1e-12)
+    # unsupported ord falls back to host numpy
+    assert float(np.linalg.norm(x, ord=1)) == pytest.approx(
+        np.linalg.norm(host, 1), rel=1e-12)
+    assert float(np.linalg.norm(y, ord=2)) == pytest.approx(
+        np.linalg.norm(m, 2), rel=1e-9)  # spectral: host SVD

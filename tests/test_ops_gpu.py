@@ -912,3 +912,16 @@ def test_device_array_sort2d_route(hnp):
     s = np.sort(x, axis=-1)
     assert isinstance(s, hnp.DeviceArray)
     np.testing.assert_array_equal(s.materialize(), np.sort(a, axis=-1))
+
+
+def test_linalg_norm_gpu(hnp):
+    x = hnp.rand(2_000_000, seed=31)
+    a = np.asarray(x).copy()
+    np.testing.assert_allclose(
+        float(np.linalg.norm(x)), np.linalg.norm(a), rtol=1e-10)
+    m = hnp.rand(1000, 2000, seed=32)
+    ma = np.asarray(m).copy()
+    r = np.linalg.norm(m, axis=1)
+    assert isinstance(r, hnp.DeviceArray)
+    np.testing.assert_allclose(
+        r.materialize(), np.linalg.norm(ma, axis=1), rtol=1e-10)
