@@ -405,6 +405,35 @@ PyObject* py_download_slice(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// download_strided(h, byte_offset, stride_bytes, elem_bytes, count)
+// -> bytes. One pitched D2H copy: extracts a column of a row-major
+// matrix (row-quantile picks after the 2-D sort) without pulling the
+// whole buffer.
+PyObject* py_download_strided(PyObject*, PyObject* args) {
+  unsigned long long h;
+  long long off, stride, esz, count;
+  if (!PyArg_ParseTuple(args, "KLLLL", &h, &off, &stride, &esz, &count))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DevBuf& buf = get_buf(h);
+  if (off < 0 || stride < esz || esz <= 0 || count < 0 ||
+      (count > 0 && off + (count - 1) * stride + esz > buf.size))
+    throw std::runtime_error("download_strided out of range");
+  PyObject* out = PyBytes_FromStringAndSize(nullptr, (Py_ssize_t)(esz * count));
+  if (!out) throw std::bad_alloc();
+  char* dst = PyBytes_AS_STRING(out);
+  NOGIL_BEGIN
+  std::lock_guard<std::mutex> stage_lk(g.stage_mu);
+  HIP_CHECK(hipStreamSynchronize(g.compute));
+  HIP_CHECK(hipMemcpy2D(dst, (size_t)esz, (char*)buf.ptr + off,
+                        (size_t)stride, (size_t)esz, (size_t)count,
+                        hipMemcpyDeviceToHost));
+  NOGIL_END
+  return out;
+  WRAP_END
+}
+
 PyObject* py_download(PyObject*, PyObject* args) {
   unsigned long long h;
   PyObject* obj;
@@ -1171,6 +1200,8 @@ PyMethodDef methods[] = {
     {"free", py_free, METH_VARARGS, "free(handle)"},
     {"upload", py_upload, METH_VARARGS, "upload(buffer) -> handle"},
     {"download", py_download, METH_VARARGS, "download(handle, buffer)"},
+    {"download_strided", py_download_strided, METH_VARARGS,
+     "download_strided(handle, off, stride, elem_bytes, count) -> bytes"},
     {"download_slice", py_download_slice, METH_VARARGS,
      "download_slice(handle, byte_offset, nbytes) -> bytes"},
     {"rand", py_rand, METH_VARARGS, "rand(n, dtype, seed) -> handle"},

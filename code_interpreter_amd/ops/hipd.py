@@ -187,6 +187,12 @@ class Connection(threading.Thread):
             h = _hipops.cumsum(self._own(m["h"]), m["dtype"], m["n"])
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "download_strided":
+            data = _hipops.download_strided(
+                self._own(m["h"]), m["off"], m["stride"], m["esz"],
+                m["count"]
+            )
+            return {"ok": True}, data
         if op == "download_slice":
             data = _hipops.download_slice(
                 self._own(m["h"]), m["off"], m["nbytes"]

@@ -324,6 +324,13 @@ class RemoteBackend:
         )
         return out["_payload"]
 
+    def download_strided(self, h, off, stride, esz, count):
+        out = self._call(
+            {"op": "download_strided", "h": h, "off": off, "stride": stride,
+             "esz": esz, "count": count}
+        )
+        return out["_payload"]
+
     def sort(self, h, dtype, n, want_idx):
         out = self._call(
             {"op": "sort", "h": h, "dtype": dtype, "n": n,
@@ -1247,6 +1254,14 @@ class DeviceArray:
             r = quantile_device(args[0], 0.5)
             if r is not None:
                 return r
+        if func is _np.median and len(args) == 1 and set(kwargs) == {"axis"}:
+            if kwargs["axis"] in (1, -1):
+                r = quantile_rows_device(args[0], 0.5)
+                if r is not None:
+                    return r
+            a0 = args[0]
+            if isinstance(a0, DeviceArray):
+                return _np.median(a0.materialize(), **kwargs)
         if func in (_np.quantile, _np.percentile) and len(args) == 2 and not kwargs:
             qv = args[1]
             if isinstance(qv, (int, float)):
@@ -1254,6 +1269,18 @@ class DeviceArray:
                 r = quantile_device(args[0], q)
                 if r is not None:
                     return r
+        if func in (_np.quantile, _np.percentile) and len(args) == 2 and set(
+            kwargs
+        ) == {"axis"}:
+            qv = args[1]
+            if isinstance(qv, (int, float)) and kwargs["axis"] in (1, -1):
+                q = qv / 100.0 if func is _np.percentile else float(qv)
+                r = quantile_rows_device(args[0], q)
+                if r is not None:
+                    return r
+            a0 = args[0]
+            if isinstance(a0, DeviceArray):
+                return func(a0.materialize(), qv, **kwargs)
         if func is _np.where and len(args) == 3 and not kwargs:
             r = where_device(*args)
             if r is not NotImplemented:
@@ -1595,6 +1622,39 @@ def _order_stat_device(x, k):
         below = cum
         lo, hi = new_lo, new_hi
     return float("nan"), 0  # did not converge: caller falls back
+
+
+def quantile_rows_device(x, q):
+    """Per-row quantile (axis=-1) of a 2-D DeviceArray: device row sort,
+    then ONE pitched column download per interpolation endpoint (R
+    elements, not R*L). numpy-linear interpolation; rows containing NaN
+    yield NaN (numpy parity, minus numpy's RuntimeWarning). Returns a
+    host ndarray of length R, or None when not routable."""
+    if not isinstance(x, DeviceArray) or len(x.shape) != 2:
+        return None
+    if _dtype_code(x.dtype) is None or not 0.0 <= q <= 1.0:
+        return None
+    rows, cols = x.shape
+    if cols < 1 or rows < 1 or x.size > (1 << 31):
+        return None
+    srt = x._device_sorted(False)
+    esz = x.dtype.itemsize
+
+    def col(k):
+        raw = backend().download_strided(
+            srt._dev_handle(), k * esz, cols * esz, esz, rows
+        )
+        return _np.frombuffer(raw, dtype=x.dtype).copy()
+
+    pos = q * (cols - 1)
+    k0 = int(_np.floor(pos))
+    k1 = min(k0 + 1, cols - 1)
+    frac = x.dtype.type(pos - k0)
+    c0 = col(k0)
+    out = c0 if k1 == k0 or frac == 0 else c0 + (col(k1) - c0) * frac
+    top = col(cols - 1)  # NaNs sort last: top column flags NaN rows
+    out[_np.isnan(top)] = _np.nan
+    return out
 
 
 def quantile_device(x, q):

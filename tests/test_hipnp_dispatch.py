@@ -118,6 +118,14 @@ class FakeBackend:
             return self._new(a[idx]), self._new(idx.astype(np.int64))
         return self._new(np.sort(a, kind="stable"))
 
+    def download_strided(self, h, off, stride, esz, count):
+        self.calls.append("download_strided")
+        raw = self.bufs[h].view(np.uint8).reshape(-1).tobytes()
+        return b"".join(
+            raw[off + i * stride: off + i * stride + esz]
+            for i in range(count)
+        )
+
     def sort2d(self, h, dtype, rows, cols, want_idx):
         self.calls.append("sort2d")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: rows * cols]
@@ -936,3 +944,32 @@ def test_linalg_norm_on_device(fake):
         np.linalg.norm(host, 1), rel=1e-12)
     assert float(np.linalg.norm(y, ord=2)) == pytest.approx(
         np.linalg.norm(m, 2), rel=1e-9)  # spectral: host SVD
+
+
+def test_row_median_quantile_on_device(fake):
+    host = np.random.default_rng(24).random((37, 101))
+    x = _device(fake, host)
+    med = np.median(x, axis=1)
+    assert isinstance(med, np.ndarray)
+    assert "sort2d" in fake.calls and "download_strided" in fake.calls
+    assert "download" not in fake.calls  # no full materialize
+    np.testing.assert_allclose(med, np.median(host, axis=1), rtol=1e-12)
+    for q in (0.0, 0.25, 0.9, 1.0):
+        np.testing.assert_allclose(
+            np.quantile(_device(fake, host), q, axis=-1),
+            np.quantile(host, q, axis=-1), rtol=1e-12)
+    np.testing.assert_allclose(
+        np.percentile(_device(fake, host), 75, axis=1),
+        np.percentile(host, 75, axis=1), rtol=1e-12)
+    # NaN rows propagate NaN
+    h2 = host.copy()
+    h2[3, 7] = np.nan
+    h2[10, 0] = np.nan
+    got = np.median(_device(fake, h2), axis=1)
+    assert np.isnan(got[3]) and np.isnan(got[10])
+    ok = ~np.isnan(np.median(h2, axis=1))
+    np.testing.assert_allclose(got[ok], np.median(h2, axis=1)[ok], rtol=1e-12)
+    # axis=0 falls back to host numpy
+    np.testing.assert_allclose(
+        np.median(_device(fake, host), axis=0),
+        np.median(host, axis=0), rtol=1e-12)

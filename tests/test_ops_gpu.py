@@ -925,3 +925,13 @@ def test_linalg_norm_gpu(hnp):
     assert isinstance(r, hnp.DeviceArray)
     np.testing.assert_allclose(
         r.materialize(), np.linalg.norm(ma, axis=1), rtol=1e-10)
+
+
+def test_row_median_gpu(hnp):
+    m = hnp.rand(2000, 3000, seed=41)
+    a = np.asarray(m).copy()
+    np.testing.assert_allclose(
+        np.median(m, axis=1), np.median(a, axis=1), rtol=1e-12)
+    np.testing.assert_allclose(
+        np.quantile(m, 0.9, axis=-1), np.quantile(a, 0.9, axis=-1),
+        rtol=1e-12)
