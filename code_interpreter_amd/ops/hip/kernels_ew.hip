@@ -784,6 +784,34 @@ __global__ void rand_f32_kernel(float* __restrict__ out, int64_t n,
   }
 }
 
+// LDS-tiled 2-D transpose: 32x32 element tiles, 256 threads (each
+// thread moves 4 elements), +1-padded LDS so both the row-coalesced
+// read and the row-coalesced (transposed) write are conflict-free.
+template <typename T>
+__global__ void transpose_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                 int64_t rows, int64_t cols,
+                                 int64_t tiles_c) {
+  __shared__ T tile[32][33];
+  int64_t tr = blockIdx.x / tiles_c;
+  int64_t tc = blockIdx.x % tiles_c;
+  int64_t r0 = tr * 32, c0 = tc * 32;
+  int tx = threadIdx.x & 31;
+  int ty = threadIdx.x >> 5;  // 0..7
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    int64_t r = r0 + ty + i * 8;
+    int64_t c = c0 + tx;
+    if (r < rows && c < cols) tile[ty + i * 8][tx] = in[r * cols + c];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    int64_t r = c0 + ty + i * 8;   // output row = input col
+    int64_t c = r0 + tx;           // output col = input row
+    if (r < cols && c < rows) out[r * rows + c] = tile[tx][ty + i * 8];
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1167,5 +1195,20 @@ void launch_rand_uniform(DType dt, void* out, int64_t n, uint64_t seed,
     hipLaunchKernelGGL(rand_f32_kernel, dim3(grid), dim3(kBlock), 0, stream,
                        (float*)out, n, seed, offset);
   }
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_transpose(DType dt, const void* in, void* out, int64_t rows,
+                      int64_t cols, hipStream_t stream) {
+  int64_t tiles_r = (rows + 31) / 32;
+  int64_t tiles_c = (cols + 31) / 32;
+  if (dt == DType::F64)
+    hipLaunchKernelGGL(transpose_kernel<double>, dim3(tiles_r * tiles_c),
+                       dim3(256), 0, stream, (const double*)in, (double*)out,
+                       rows, cols, tiles_c);
+  else
+    hipLaunchKernelGGL(transpose_kernel<float>, dim3(tiles_r * tiles_c),
+                       dim3(256), 0, stream, (const float*)in, (float*)out,
+                       rows, cols, tiles_c);
   HIP_CHECK(hipGetLastError());
 }

@@ -340,6 +340,12 @@ class RemoteBackend:
             return out["h"], out["hi"]
         return out["h"]
 
+    def transpose(self, h, dtype, rows, cols):
+        return self._call(
+            {"op": "transpose", "h": h, "dtype": dtype, "rows": rows,
+             "cols": cols}
+        )["h"]
+
     def sort2d(self, h, dtype, rows, cols, want_idx):
         out = self._call(
             {"op": "sort2d", "h": h, "dtype": dtype, "rows": rows,
@@ -988,10 +994,33 @@ class DeviceArray:
             return DeviceArray(h, shape, self.dtype)
         return self.materialize().cumsum(axis=axis, **kwargs)
 
+    def _device_transposed(self):
+        """Device 2-D transpose (LDS-tiled). Itemsize-based: also moves
+        int64 index matrices through the f64-width kernel (pure data
+        movement)."""
+        rows, cols = self.shape
+        code = _F64 if self.dtype.itemsize == 8 else _F32
+        h = backend().transpose(self._dev_handle(), code, rows, cols)
+        return DeviceArray(h, (cols, rows), self.dtype)
+
+    def transpose(self, *axes):
+        if len(self.shape) == 2 and self.dtype.itemsize in (4, 8) and (
+            not axes or axes == (1, 0) or axes == ((1, 0),)
+        ):
+            return self._device_transposed()
+        return self.materialize().transpose(*axes)
+
+    @property
+    def T(self):
+        if len(self.shape) < 2:
+            return self
+        return self.transpose()
+
     def _sort_routable(self, axis, kind, order, kwargs):
         """np.sort/argsort route: f32/f64, default comparator, flat 1-D
-        or 2-D along the last axis. The device sort is an LSD radix
-        sort, so it is stable -- every numpy `kind` is satisfied."""
+        or 2-D along either axis (axis=0 goes through the device
+        transpose). The device sort is an LSD radix sort, so it is
+        stable -- every numpy `kind` is satisfied."""
         if (
             _dtype_code(self.dtype) is None
             or kwargs
@@ -1003,11 +1032,22 @@ class DeviceArray:
         if len(self.shape) == 1:
             return axis is None or self._norm_axis(axis, 1) == 0
         if len(self.shape) == 2 and axis is not None:
-            return self._norm_axis(axis, 2) == 1 and self.size <= (1 << 31)
+            return (
+                self._norm_axis(axis, 2) in (0, 1)
+                and self.size <= (1 << 31)
+            )
         return False
 
-    def _device_sorted(self, want_idx):
+    def _device_sorted(self, want_idx, axis=None):
+        """axis=None: flat (1-D) or per-row (2-D). axis=0 on 2-D:
+        transpose -> row sort -> transpose back."""
         code = _dtype_code(self.dtype)
+        if len(self.shape) == 2 and axis == 0:
+            t = self._device_transposed()
+            if want_idx:
+                sv, si = t._device_sorted(True)
+                return sv._device_transposed(), si._device_transposed()
+            return t._device_sorted(False)._device_transposed()
         if len(self.shape) == 2:
             rows, cols = self.shape
             r = backend().sort2d(
@@ -1025,11 +1065,16 @@ class DeviceArray:
             )
         return DeviceArray(r, self.shape, self.dtype)
 
+    def _sort_axis01(self, axis):
+        if len(self.shape) == 2 and axis is not None:
+            return self._norm_axis(axis, 2)
+        return None
+
     def sort(self, axis=-1, kind=None, order=None, **kwargs):
         """In-place sort (ndarray.sort contract): the handle is swapped
         for the sorted buffer."""
         if self._sort_routable(axis, kind, order, kwargs):
-            res = self._device_sorted(False)
+            res = self._device_sorted(False, axis=self._sort_axis01(axis))
             old_h, self._handle = self._handle, res._handle
             res._handle = None  # ownership moved; res.__del__ must not free
             self._host = None
@@ -1045,7 +1090,7 @@ class DeviceArray:
 
     def argsort(self, axis=-1, kind=None, order=None, **kwargs):
         if self._sort_routable(axis, kind, order, kwargs):
-            _, idx = self._device_sorted(True)
+            _, idx = self._device_sorted(True, axis=self._sort_axis01(axis))
             return idx
         return self.materialize().argsort(
             axis=axis, kind=kind, order=order, **kwargs
@@ -1259,6 +1304,10 @@ class DeviceArray:
                 r = quantile_rows_device(args[0], 0.5)
                 if r is not None:
                     return r
+            elif kwargs["axis"] == 0:
+                r = quantile_cols_device(args[0], 0.5)
+                if r is not None:
+                    return r
             a0 = args[0]
             if isinstance(a0, DeviceArray):
                 return _np.median(a0.materialize(), **kwargs)
@@ -1273,9 +1322,12 @@ class DeviceArray:
             kwargs
         ) == {"axis"}:
             qv = args[1]
-            if isinstance(qv, (int, float)) and kwargs["axis"] in (1, -1):
+            if isinstance(qv, (int, float)) and kwargs["axis"] in (0, 1, -1):
                 q = qv / 100.0 if func is _np.percentile else float(qv)
-                r = quantile_rows_device(args[0], q)
+                if kwargs["axis"] == 0:
+                    r = quantile_cols_device(args[0], q)
+                else:
+                    r = quantile_rows_device(args[0], q)
                 if r is not None:
                     return r
             a0 = args[0]
@@ -1322,9 +1374,10 @@ class DeviceArray:
                 if k not in ("axis", "kind", "order")
             }
             if a._sort_routable(axis, None, order, extra):
+                ax01 = a._sort_axis01(axis)
                 if func is _np.partition:
-                    return a._device_sorted(False)
-                _, idx = a._device_sorted(True)
+                    return a._device_sorted(False, axis=ax01)
+                _, idx = a._device_sorted(True, axis=ax01)
                 return idx
             return func(a.materialize(), kth, **kwargs)
         if func is _np.unique and len(args) == 1 and isinstance(
@@ -1360,12 +1413,19 @@ class DeviceArray:
                 if k not in ("axis", "kind", "order")
             }
             if a._sort_routable(axis, kind, order, extra):
+                ax01 = a._sort_axis01(axis)
                 if func is _np.sort:
-                    return a._device_sorted(False)
-                _, idx = a._device_sorted(True)
+                    return a._device_sorted(False, axis=ax01)
+                _, idx = a._device_sorted(True, axis=ax01)
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func is _np.transpose and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"axes"}:
+            return args[0].transpose(*(
+                (kwargs["axes"],) if kwargs.get("axes") is not None else ()
+            ))
         if func is _np.linalg.norm and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ):
@@ -1655,6 +1715,16 @@ def quantile_rows_device(x, q):
     top = col(cols - 1)  # NaNs sort last: top column flags NaN rows
     out[_np.isnan(top)] = _np.nan
     return out
+
+
+def quantile_cols_device(x, q):
+    """Per-column quantile (axis=0): device transpose, then the row
+    path."""
+    if not isinstance(x, DeviceArray) or len(x.shape) != 2:
+        return None
+    if _dtype_code(x.dtype) is None or not 0.0 <= q <= 1.0:
+        return None
+    return quantile_rows_device(x._device_transposed(), q)
 
 
 def quantile_device(x, q):

@@ -126,6 +126,11 @@ class FakeBackend:
             for i in range(count)
         )
 
+    def transpose(self, h, dtype, rows, cols):
+        self.calls.append("transpose")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: rows * cols]
+        return self._new(np.ascontiguousarray(a.reshape(rows, cols).T))
+
     def sort2d(self, h, dtype, rows, cols, want_idx):
         self.calls.append("sort2d")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: rows * cols]
@@ -833,9 +838,9 @@ def test_sort_method_in_place(fake):
 
 
 def test_sort_unroutable_falls_back(fake):
-    host = np.random.default_rng(14).random((20, 30))
+    host = np.random.default_rng(14).random((2, 3, 4))
     x = _device(fake, host)
-    s = np.sort(x, axis=0)  # 2-D: host fallback
+    s = np.sort(x, axis=0)  # 3-D: host fallback
     assert isinstance(s, np.ndarray)
     np.testing.assert_array_equal(s, np.sort(host, axis=0))
     # any kind routes (stable output satisfies every numpy kind)
@@ -909,10 +914,10 @@ def test_sort2d_rows_on_device(fake):
     assert isinstance(idx, hipnp.DeviceArray) and idx.dtype == np.int64
     np.testing.assert_array_equal(
         idx.materialize(), np.argsort(host, axis=1, kind="stable"))
-    # axis=0 falls back to host
+    # axis=0 routes through the device transpose
     s0 = np.sort(_device(fake, host), axis=0)
-    assert isinstance(s0, np.ndarray)
-    np.testing.assert_array_equal(s0, np.sort(host, axis=0))
+    assert isinstance(s0, hipnp.DeviceArray)
+    np.testing.assert_array_equal(s0.materialize(), np.sort(host, axis=0))
     # in-place method, 2-D default axis
     y = _device(fake, host)
     assert y.sort() is None
@@ -973,3 +978,34 @@ def test_row_median_quantile_on_device(fake):
     np.testing.assert_allclose(
         np.median(_device(fake, host), axis=0),
         np.median(host, axis=0), rtol=1e-12)
+
+
+def test_transpose_and_axis0_sort(fake):
+    host = np.random.default_rng(25).random((30, 47))
+    x = _device(fake, host)
+    t = np.transpose(x)
+    assert isinstance(t, hipnp.DeviceArray) and t.shape == (47, 30)
+    np.testing.assert_array_equal(t.materialize(), host.T)
+    t2 = _device(fake, host).T
+    assert isinstance(t2, hipnp.DeviceArray)
+    np.testing.assert_array_equal(np.asarray(t2), host.T)
+    s0 = np.sort(_device(fake, host), axis=0)
+    assert isinstance(s0, hipnp.DeviceArray)
+    np.testing.assert_array_equal(s0.materialize(), np.sort(host, axis=0))
+    i0 = np.argsort(_device(fake, host), axis=0)
+    assert isinstance(i0, hipnp.DeviceArray) and i0.dtype == np.int64
+    np.testing.assert_array_equal(
+        i0.materialize(), np.argsort(host, axis=0, kind="stable"))
+
+
+def test_axis0_median_quantile(fake):
+    host = np.random.default_rng(26).random((61, 33))
+    np.testing.assert_allclose(
+        np.median(_device(fake, host), axis=0),
+        np.median(host, axis=0), rtol=1e-12)
+    np.testing.assert_allclose(
+        np.quantile(_device(fake, host), 0.3, axis=0),
+        np.quantile(host, 0.3, axis=0), rtol=1e-12)
+    np.testing.assert_allclose(
+        np.percentile(_device(fake, host), 99, axis=0),
+        np.percentile(host, 99, axis=0), rtol=1e-12)

@@ -935,3 +935,19 @@ def test_row_median_gpu(hnp):
     np.testing.assert_allclose(
         np.quantile(m, 0.9, axis=-1), np.quantile(a, 0.9, axis=-1),
         rtol=1e-12)
+
+
+def test_transpose_axis0_gpu(hnp):
+    m = hnp.rand(1500, 700, seed=55)
+    a = np.asarray(m).copy()
+    t = np.transpose(m)
+    assert isinstance(t, hnp.DeviceArray) and t.shape == (700, 1500)
+    np.testing.assert_array_equal(t.materialize(), a.T)
+    s0 = np.sort(m, axis=0)
+    assert isinstance(s0, hnp.DeviceArray)
+    np.testing.assert_array_equal(s0.materialize(), np.sort(a, axis=0))
+    i0 = np.argsort(m, axis=0)
+    np.testing.assert_array_equal(
+        i0.materialize(), np.argsort(a, axis=0, kind="stable"))
+    np.testing.assert_allclose(
+        np.median(m, axis=0), np.median(a, axis=0), rtol=1e-12)
