@@ -723,6 +723,29 @@ PyObject* py_sort2d(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// cumsum2d(h, dtype, rows, cols) -> handle (row-wise inclusive scan)
+PyObject* py_cumsum2d(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt;
+  long long rows, cols;
+  if (!PyArg_ParseTuple(args, "KiLL", &h, &dt, &rows, &cols)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (rows < 1 || cols < 1) throw std::runtime_error("bad cumsum2d shape");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  int64_t n = rows * cols;
+  if (n * esize > in.size) throw std::runtime_error("cumsum2d oob");
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n * esize);
+  launch_cumsum_rows(dtype, in.ptr, out, rows, cols, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n * esize));
+  WRAP_END
+}
+
 // transpose(h, dtype, rows, cols) -> handle ([cols][rows] result)
 PyObject* py_transpose(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1239,6 +1262,8 @@ PyMethodDef methods[] = {
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
+    {"cumsum2d", py_cumsum2d, METH_VARARGS,
+     "cumsum2d(h, dtype, rows, cols) -> handle (row-wise scan)"},
     {"transpose", py_transpose, METH_VARARGS,
      "transpose(h, dtype, rows, cols) -> handle"},
     {"sort2d", py_sort2d, METH_VARARGS,

@@ -812,6 +812,53 @@ __global__ void transpose_kernel(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+// row-wise inclusive prefix sum (np.cumsum axis=-1 on 2-D): one block
+// per row, 1024-element chunks (4 thread-serial elements + a 256-wide
+// Hillis-Steele over thread sums), running double carry across chunks
+// -- accumulation in double like the flat cumsum.
+template <typename T>
+__global__ void cumsum_rows_kernel(const T* __restrict__ in,
+                                   T* __restrict__ out, int64_t rows,
+                                   int64_t cols) {
+  __shared__ double sums[256];
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * cols;
+  double carry = 0.0;
+  for (int64_t chunk0 = 0; chunk0 < cols; chunk0 += 1024) {
+    int64_t e0 = chunk0 + tid * 4;
+    double v[4];
+    double s = 0.0;
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int64_t e = e0 + j;
+      v[j] = e < cols ? (double)in[base + e] : 0.0;
+      s += v[j];
+    }
+    sums[tid] = s;
+    double x = s;
+    for (int off = 1; off < 256; off <<= 1) {
+      __syncthreads();
+      double add = tid >= off ? sums[tid - off] : 0.0;
+      __syncthreads();
+      x += add;
+      sums[tid] = x;
+    }
+    __syncthreads();
+    double total = sums[255];
+    double run = x - s + carry;  // exclusive prefix for this thread
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int64_t e = e0 + j;
+      if (e < cols) {
+        run += v[j];
+        out[base + e] = (T)run;
+      }
+    }
+    carry += total;
+    __syncthreads();  // sums reused next chunk
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1210,5 +1257,18 @@ void launch_transpose(DType dt, const void* in, void* out, int64_t rows,
     hipLaunchKernelGGL(transpose_kernel<float>, dim3(tiles_r * tiles_c),
                        dim3(256), 0, stream, (const float*)in, (float*)out,
                        rows, cols, tiles_c);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_cumsum_rows(DType dt, const void* in, void* out, int64_t rows,
+                        int64_t cols, hipStream_t stream) {
+  if (dt == DType::F64)
+    hipLaunchKernelGGL(cumsum_rows_kernel<double>, dim3((unsigned)rows),
+                       dim3(256), 0, stream, (const double*)in, (double*)out,
+                       rows, cols);
+  else
+    hipLaunchKernelGGL(cumsum_rows_kernel<float>, dim3((unsigned)rows),
+                       dim3(256), 0, stream, (const float*)in, (float*)out,
+                       rows, cols);
   HIP_CHECK(hipGetLastError());
 }

@@ -340,6 +340,12 @@ class RemoteBackend:
             return out["h"], out["hi"]
         return out["h"]
 
+    def cumsum2d(self, h, dtype, rows, cols):
+        return self._call(
+            {"op": "cumsum2d", "h": h, "dtype": dtype, "rows": rows,
+             "cols": cols}
+        )["h"]
+
     def transpose(self, h, dtype, rows, cols):
         return self._call(
             {"op": "transpose", "h": h, "dtype": dtype, "rows": rows,
@@ -979,19 +985,29 @@ class DeviceArray:
         return self.materialize().clip(a_min, a_max, **kwargs)
 
     def cumsum(self, axis=None, **kwargs):
-        """Flat cumulative sum on device (axis=None or the only axis of
-        a 1-D array); accumulation in double. Rounding may differ from
-        numpy's strictly-sequential order by ~1 ulp at block
-        regroupings."""
+        """Cumulative sum on device: flat (axis=None / 1-D), or 2-D
+        along either axis (axis=0 through the device transpose);
+        accumulation in double. Rounding may differ from numpy's
+        strictly-sequential order by ~1 ulp at block regroupings."""
         code = _dtype_code(self.dtype)
-        routable = code is not None and not kwargs and (
-            axis is None
-            or (len(self.shape) == 1 and self._norm_axis(axis, 1) == 0)
-        )
-        if routable:
-            h = backend().cumsum(self._dev_handle(), code, self.size)
-            shape = self.shape if axis is not None else (self.size,)
-            return DeviceArray(h, shape, self.dtype)
+        if code is not None and not kwargs:
+            if axis is None or (
+                len(self.shape) == 1 and self._norm_axis(axis, 1) == 0
+            ):
+                h = backend().cumsum(self._dev_handle(), code, self.size)
+                shape = self.shape if axis is not None else (self.size,)
+                return DeviceArray(h, shape, self.dtype)
+            if len(self.shape) == 2 and axis is not None:
+                ax = self._norm_axis(axis, 2)
+                if ax == 1:
+                    rows, cols = self.shape
+                    h = backend().cumsum2d(
+                        self._dev_handle(), code, rows, cols
+                    )
+                    return DeviceArray(h, self.shape, self.dtype)
+                if ax == 0:
+                    t = self._device_transposed()
+                    return t.cumsum(axis=1)._device_transposed()
         return self.materialize().cumsum(axis=axis, **kwargs)
 
     def _device_transposed(self):

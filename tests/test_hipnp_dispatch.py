@@ -126,6 +126,13 @@ class FakeBackend:
             for i in range(count)
         )
 
+    def cumsum2d(self, h, dtype, rows, cols):
+        self.calls.append("cumsum2d")
+        dt = self._dt(dtype)
+        a = self.bufs[h].view(dt).reshape(-1)[: rows * cols]
+        a = a.reshape(rows, cols)
+        return self._new(a.astype(np.float64).cumsum(axis=1).astype(dt))
+
     def transpose(self, h, dtype, rows, cols):
         self.calls.append("transpose")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: rows * cols]
@@ -1009,3 +1016,20 @@ def test_axis0_median_quantile(fake):
     np.testing.assert_allclose(
         np.percentile(_device(fake, host), 99, axis=0),
         np.percentile(host, 99, axis=0), rtol=1e-12)
+
+
+def test_cumsum_axis_on_device(fake):
+    host = np.random.default_rng(27).random((25, 80))
+    x = _device(fake, host)
+    r1 = np.cumsum(x, axis=1)
+    assert isinstance(r1, hipnp.DeviceArray) and r1.shape == host.shape
+    assert "cumsum2d" in fake.calls
+    np.testing.assert_allclose(
+        r1.materialize(), np.cumsum(host, axis=1), rtol=1e-12)
+    r0 = np.cumsum(_device(fake, host), axis=0)
+    assert isinstance(r0, hipnp.DeviceArray)
+    np.testing.assert_allclose(
+        r0.materialize(), np.cumsum(host, axis=0), rtol=1e-12)
+    rm1 = _device(fake, host).cumsum(axis=-1)
+    np.testing.assert_allclose(
+        np.asarray(rm1), np.cumsum(host, axis=-1), rtol=1e-12)

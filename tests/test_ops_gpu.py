@@ -951,3 +951,30 @@ def test_transpose_axis0_gpu(hnp):
         i0.materialize(), np.argsort(a, axis=0, kind="stable"))
     np.testing.assert_allclose(
         np.median(m, axis=0), np.median(a, axis=0), rtol=1e-12)
+
+
+@pytest.mark.parametrize("shape", [(1, 7), (300, 4099), (10000, 500)])
+def test_cumsum2d_matches_numpy(hip, shape):
+    rows, cols = shape
+    rng = np.random.default_rng(rows + cols)
+    a = rng.standard_normal((rows, cols))
+    h = hip.upload(a)
+    hc = hip.cumsum2d(h, 1, rows, cols)
+    out = np.empty((rows, cols))
+    hip.download(hc, out)
+    hip.free(h)
+    hip.free(hc)
+    np.testing.assert_allclose(out, np.cumsum(a, axis=1), rtol=1e-12,
+                               atol=1e-12 * cols)
+
+
+def test_cumsum_axis_routes_gpu(hnp):
+    m = hnp.rand(800, 1200, seed=66)
+    a = np.asarray(m).copy()
+    r = np.cumsum(m, axis=1)
+    assert isinstance(r, hnp.DeviceArray)
+    np.testing.assert_allclose(r.materialize(), np.cumsum(a, axis=1),
+                               rtol=1e-10)
+    r0 = np.cumsum(m, axis=0)
+    np.testing.assert_allclose(r0.materialize(), np.cumsum(a, axis=0),
+                               rtol=1e-10)
