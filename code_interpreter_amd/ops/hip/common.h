@@ -73,8 +73,8 @@ void launch_mask_logic(const void* a, const void* b, void* out, int64_t n,
 // histogram selection (median/percentile): range histogram (counts:
 // bins u64 + extra[3] = {nan, below, above}) and range compaction
 void launch_hist_range(DType dt, const void* in, int64_t n, double lo,
-                       double hi, double inv_width, int bins, void* counts,
-                       void* extra, hipStream_t stream);
+                       double hi, double inv_width, int bins, int exact,
+                       void* counts, void* extra, hipStream_t s);
 void launch_extract_range(DType dt, const void* in, int64_t n, double lo,
                           double hi, void* out, void* counter, int64_t cap,
                           hipStream_t stream);

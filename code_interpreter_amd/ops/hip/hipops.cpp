@@ -815,10 +815,11 @@ PyObject* py_mask_logic(PyObject*, PyObject* args) {
 // [bin counts..., nan_count, below_count, above_count]
 PyObject* py_histogram(PyObject*, PyObject* args) {
   unsigned long long h;
-  int dt, bins;
+  int dt, bins, exact = 0;
   long long n;
   double lo, hi;
-  if (!PyArg_ParseTuple(args, "KiLddi", &h, &dt, &n, &lo, &hi, &bins))
+  if (!PyArg_ParseTuple(args, "KiLddi|i", &h, &dt, &n, &lo, &hi, &bins,
+                        &exact))
     return nullptr;
   WRAP_BEGIN
   ensure_init();
@@ -832,8 +833,8 @@ PyObject* py_histogram(PyObject*, PyObject* args) {
   void* counts = pool_alloc((int64_t)(bins + 3) * 8);
   HIP_CHECK(hipMemsetAsync(counts, 0, (int64_t)(bins + 3) * 8, g.compute));
   void* extra = (char*)counts + (int64_t)bins * 8;
-  launch_hist_range(dtype, in.ptr, n, lo, hi, inv_width, bins, counts, extra,
-                    g.compute);
+  launch_hist_range(dtype, in.ptr, n, lo, hi, inv_width, bins, exact,
+                    counts, extra, g.compute);
   HIP_CHECK(hipMemcpyAsync(host.data(), counts, (int64_t)(bins + 3) * 8,
                            hipMemcpyDeviceToHost, g.compute));
   HIP_CHECK(hipStreamSynchronize(g.compute));

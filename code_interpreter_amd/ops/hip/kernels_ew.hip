@@ -359,10 +359,14 @@ __global__ void mask_logic_kernel(const unsigned char* __restrict__ a,
 // ---------------------------------------------------------------------------
 constexpr int kHistBins = 4096;
 
+// exact != 0: after the float bin estimate, correct against the actual
+// linspace edges (edge_i = lo + i*step, step = (hi-lo)/bins) the way
+// numpy's histogram does -- boundary-landing values (integer data on
+// integer edges) then bin EXACTLY like numpy.
 template <typename T>
 __global__ void hist_range_kernel(const T* __restrict__ in, int64_t n,
                                   double lo, double hi, double inv_width,
-                                  int bins,
+                                  int bins, int exact, double step,
                                   unsigned long long* __restrict__ counts,
                                   unsigned long long* __restrict__ extra) {
   // extra[0] = NaN count, extra[1] = count below lo, extra[2] = above hi
@@ -384,6 +388,11 @@ __global__ void hist_range_kernel(const T* __restrict__ in, int64_t n,
       int b = (int)((v - lo) * inv_width);
       if (b >= bins) b = bins - 1;  // v == hi (or fp rounding at the edge)
       if (b < 0) b = 0;
+      if (exact) {
+        if (v < lo + b * step) b--;
+        else if (b != bins - 1 && v >= lo + (b + 1) * step) b++;
+        if (b < 0) b = 0;
+      }
       atomicAdd(&local[b], 1u);
     }
   }
@@ -1010,18 +1019,19 @@ void launch_mask_logic(const void* a, const void* b, void* out, int64_t n,
 }
 
 void launch_hist_range(DType dt, const void* in, int64_t n, double lo,
-                       double hi, double inv_width, int bins, void* counts,
-                       void* extra, hipStream_t s) {
+                       double hi, double inv_width, int bins, int exact,
+                       void* counts, void* extra, hipStream_t s) {
   int grid = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 2048);
+  double step = (hi - lo) / bins;  // numpy linspace's step
   if (dt == DType::F64)
     hipLaunchKernelGGL((hist_range_kernel<double>), dim3(grid), dim3(kBlock),
                        0, s, (const double*)in, n, lo, hi, inv_width, bins,
-                       (unsigned long long*)counts,
+                       exact, step, (unsigned long long*)counts,
                        (unsigned long long*)extra);
   else
     hipLaunchKernelGGL((hist_range_kernel<float>), dim3(grid), dim3(kBlock),
                        0, s, (const float*)in, n, lo, hi, inv_width, bins,
-                       (unsigned long long*)counts,
+                       exact, step, (unsigned long long*)counts,
                        (unsigned long long*)extra);
   HIP_CHECK(hipGetLastError());
 }

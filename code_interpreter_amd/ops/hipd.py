@@ -243,7 +243,7 @@ class Connection(threading.Thread):
         if op == "histogram":
             data = _hipops.histogram(
                 self._own(m["h"]), m["dtype"], m["n"], m["lo"], m["hi"],
-                m["bins"]
+                m["bins"], m.get("exact", 0)
             )
             return {"ok": True}, data
         if op == "extract_range":

@@ -366,10 +366,10 @@ class RemoteBackend:
             {"op": "mask_logic", "ha": ha, "hb": hb, "n": n, "lop": lop}
         )["h"]
 
-    def histogram(self, h, dtype, n, lo, hi, bins):
+    def histogram(self, h, dtype, n, lo, hi, bins, exact=0):
         out = self._call(
             {"op": "histogram", "h": h, "dtype": dtype, "n": n, "lo": lo,
-             "hi": hi, "bins": bins}
+             "hi": hi, "bins": bins, "exact": exact}
         )
         return out["_payload"]
 
@@ -1436,6 +1436,35 @@ class DeviceArray:
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func is _np.histogram and 1 <= len(args) <= 2 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"bins", "range"}:
+            a = args[0]
+            bins = args[1] if len(args) == 2 else kwargs.get("bins", 10)
+            rng = kwargs.get("range")
+            code = _dtype_code(a.dtype)
+            if (
+                code is not None
+                and isinstance(bins, (int, _np.integer))
+                and 1 <= int(bins) <= 4096
+            ):
+                bins = int(bins)
+                if rng is not None:
+                    lo, hi = float(rng[0]), float(rng[1])
+                else:
+                    lo, hi = float(a.min()), float(a.max())
+                if _np.isfinite(lo) and _np.isfinite(hi) and hi >= lo:
+                    if hi == lo:  # numpy expands degenerate ranges
+                        lo, hi = lo - 0.5, hi + 0.5
+                    raw = backend().histogram(
+                        a._dev_handle(), code, a.size, lo, hi, bins, 1
+                    )
+                    counts = _np.frombuffer(raw, dtype=_np.uint64)
+                    hist = counts[:bins].astype(_np.int64)
+                    edges = _np.linspace(lo, hi, bins + 1)
+                    return hist, edges
+            host = a.materialize()
+            return _np.histogram(host, *args[1:], **kwargs)
         if func is _np.transpose and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and set(kwargs) <= {"axes"}:

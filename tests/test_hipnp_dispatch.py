@@ -165,8 +165,22 @@ class FakeBackend:
             out = ~a
         return self._new(out.astype(np.uint8))
 
-    def histogram(self, h, dtype, n, lo, hi, bins):
+    def histogram(self, h, dtype, n, lo, hi, bins, exact=0):
         self.calls.append("histogram")
+        if exact:
+            a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
+            a = a.astype(np.float64)
+            nan_c = int(np.isnan(a).sum())
+            valid = a[~np.isnan(a)]
+            below = int((valid < lo).sum())
+            above = int((valid > hi).sum())
+            counts, _ = np.histogram(
+                valid[(valid >= lo) & (valid <= hi)], bins, (lo, hi))
+            out = np.concatenate([
+                counts.astype(np.uint64),
+                np.array([nan_c, below, above], dtype=np.uint64),
+            ])
+            return out.tobytes()
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n].astype(np.float64)
         nan_c = int(np.isnan(a).sum())
         valid = a[~np.isnan(a)]
@@ -1033,3 +1047,27 @@ def test_cumsum_axis_on_device(fake):
     rm1 = _device(fake, host).cumsum(axis=-1)
     np.testing.assert_allclose(
         np.asarray(rm1), np.cumsum(host, axis=-1), rtol=1e-12)
+
+
+def test_histogram_on_device(fake):
+    host = np.random.default_rng(28).random(5000) * 10
+    x = _device(fake, host)
+    hist, edges = np.histogram(x)
+    rh, re = np.histogram(host)
+    np.testing.assert_array_equal(hist, rh)
+    np.testing.assert_allclose(edges, re, rtol=0)
+    hist, edges = np.histogram(_device(fake, host), bins=50, range=(2, 8))
+    rh, re = np.histogram(host, bins=50, range=(2, 8))
+    np.testing.assert_array_equal(hist, rh)
+    np.testing.assert_allclose(edges, re, rtol=0)
+    # integer-valued data on integer edges: the numpy edge-correction case
+    iv = np.random.default_rng(29).integers(0, 10, 3000).astype(np.float64)
+    hist, edges = np.histogram(_device(fake, iv), bins=10, range=(0, 9))
+    rh, re = np.histogram(iv, bins=10, range=(0, 9))
+    np.testing.assert_array_equal(hist, rh)
+    # bins array falls back to host numpy
+    be = np.array([0.0, 1.0, 5.0, 10.0])
+    hist, edges = np.histogram(_device(fake, host), bins=be)
+    rh, re = np.histogram(host, bins=be)
+    np.testing.assert_array_equal(hist, rh)
+    np.testing.assert_array_equal(edges, re)

@@ -978,3 +978,19 @@ def test_cumsum_axis_routes_gpu(hnp):
     r0 = np.cumsum(m, axis=0)
     np.testing.assert_allclose(r0.materialize(), np.cumsum(a, axis=0),
                                rtol=1e-10)
+
+
+def test_histogram_gpu_exact(hnp):
+    x = hnp.rand(5_000_000, seed=71)
+    a = np.asarray(x).copy()
+    hist, edges = np.histogram(x, bins=64)
+    rh, re = np.histogram(a, bins=64)
+    np.testing.assert_array_equal(hist, rh)
+    np.testing.assert_allclose(edges, re, rtol=0)
+    # integer-valued data on integer edges: exact numpy binning
+    iv = (x * 10.0).astype(np.float64)
+    q = np.floor(iv.materialize())
+    hist, _ = np.histogram(hnp.DeviceArray(
+        hnp.backend().upload(q), q.shape, q.dtype), bins=10, range=(0, 9))
+    rh, _ = np.histogram(q, bins=10, range=(0, 9))
+    np.testing.assert_array_equal(hist, rh)
