@@ -1436,6 +1436,13 @@ class DeviceArray:
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func in (_np.cov, _np.corrcoef) and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and not kwargs:
+            r = (cov_device if func is _np.cov else corrcoef_device)(args[0])
+            if r is not None:
+                return r
+            return func(args[0].materialize())
         if func is _np.histogram and 1 <= len(args) <= 2 and isinstance(
             args[0], DeviceArray
         ) and set(kwargs) <= {"bins", "range"}:
@@ -1760,6 +1767,55 @@ def quantile_rows_device(x, q):
     top = col(cols - 1)  # NaNs sort last: top column flags NaN rows
     out[_np.isnan(top)] = _np.nan
     return out
+
+
+def cov_device(m):
+    """np.cov for a 2-D DeviceArray (rowvar=True, ddof=1 defaults),
+    composed from device ops: row means -> broadcast center -> X @ X.T
+    via the device transpose + GEMM. Returns None when not routable."""
+    if not isinstance(m, DeviceArray) or len(m.shape) != 2:
+        return None
+    if _dtype_code(m.dtype) is None:
+        return None
+    nvar, nobs = m.shape
+    if nobs < 2:
+        return None
+    mu = m.mean(axis=1, keepdims=True)
+    if not isinstance(mu, DeviceArray):
+        return None  # small-outer host fallback in the axis reducer
+    xc = m._binary_bcast("subtract", mu)
+    if xc is NotImplemented:
+        return None
+    prod = matmul(xc, xc._device_transposed(), _force=True)
+    if prod is NotImplemented or not isinstance(prod, DeviceArray):
+        return None
+    r = prod._binary("divide", float(nobs - 1))
+    return None if r is NotImplemented else r
+
+
+def corrcoef_device(m):
+    """np.corrcoef via cov_device: diagonal extracted with ONE strided
+    download, normalization and the final [-1, 1] clip on device."""
+    c = cov_device(m)
+    if c is None:
+        return None
+    n = c.shape[0]
+    esz = c.dtype.itemsize
+    raw = backend().download_strided(
+        c._dev_handle(), 0, (n + 1) * esz, esz, n
+    )
+    d = _np.sqrt(_np.frombuffer(raw, dtype=c.dtype))
+    drow = DeviceArray(backend().upload(_np.ascontiguousarray(d)),
+                       (n,), c.dtype)
+    dcol = DeviceArray(backend().upload(_np.ascontiguousarray(d)),
+                       (n, 1), c.dtype)
+    r = c._binary_bcast("divide", dcol)
+    if r is NotImplemented:
+        return None
+    r = r._binary_bcast("divide", drow)
+    if r is NotImplemented:
+        return None
+    return r.clip(-1.0, 1.0)
 
 
 def quantile_cols_device(x, q):

@@ -1071,3 +1071,19 @@ def test_histogram_on_device(fake):
     rh, re = np.histogram(host, bins=be)
     np.testing.assert_array_equal(hist, rh)
     np.testing.assert_array_equal(edges, re)
+
+
+def test_cov_corrcoef_on_device(fake):
+    host = np.random.default_rng(30).random((80, 500))
+    x = _device(fake, host)
+    c = np.cov(x)
+    assert isinstance(c, hipnp.DeviceArray) and c.shape == (80, 80)
+    np.testing.assert_allclose(c.materialize(), np.cov(host), rtol=1e-10)
+    r = np.corrcoef(_device(fake, host))
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(
+        r.materialize(), np.corrcoef(host), rtol=1e-9, atol=1e-12)
+    # small matrices fall back to host numpy transparently
+    small = np.random.default_rng(31).random((5, 30))
+    np.testing.assert_allclose(
+        np.cov(_device(fake, small)), np.cov(small), rtol=1e-12)

@@ -994,3 +994,14 @@ def test_histogram_gpu_exact(hnp):
         hnp.backend().upload(q), q.shape, q.dtype), bins=10, range=(0, 9))
     rh, _ = np.histogram(q, bins=10, range=(0, 9))
     np.testing.assert_array_equal(hist, rh)
+
+
+def test_cov_corrcoef_gpu(hnp):
+    m = hnp.rand(128, 5000, seed=81)
+    a = np.asarray(m).copy()
+    c = np.cov(m)
+    assert isinstance(c, hnp.DeviceArray)
+    np.testing.assert_allclose(c.materialize(), np.cov(a), rtol=1e-9)
+    r = np.corrcoef(m)
+    np.testing.assert_allclose(
+        r.materialize(), np.corrcoef(a), rtol=1e-8, atol=1e-12)
