@@ -723,6 +723,29 @@ PyObject* py_sort2d(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// diff(h, dtype, outer, inner) -> handle (outer x (inner-1))
+PyObject* py_diff(PyObject*, PyObject* args) {
+  unsigned long long h;
+  int dt;
+  long long outer, inner;
+  if (!PyArg_ParseTuple(args, "KiLL", &h, &dt, &outer, &inner)) return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (outer < 1 || inner < 2) throw std::runtime_error("bad diff shape");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& in = get_buf(h);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  if (outer * inner * esize > in.size) throw std::runtime_error("diff oob");
+  int64_t n_out = outer * (inner - 1);
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(n_out * esize);
+  launch_diff(dtype, in.ptr, out, outer, inner, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, n_out * esize));
+  WRAP_END
+}
+
 // cumsum2d(h, dtype, rows, cols) -> handle (row-wise inclusive scan)
 PyObject* py_cumsum2d(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1263,6 +1286,8 @@ PyMethodDef methods[] = {
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
+    {"diff", py_diff, METH_VARARGS,
+     "diff(h, dtype, outer, inner) -> handle"},
     {"cumsum2d", py_cumsum2d, METH_VARARGS,
      "cumsum2d(h, dtype, rows, cols) -> handle (row-wise scan)"},
     {"transpose", py_transpose, METH_VARARGS,

@@ -868,6 +868,20 @@ __global__ void cumsum_rows_kernel(const T* __restrict__ in,
   }
 }
 
+// np.diff along the last axis: out[o][i] = in[o][i+1] - in[o][i]
+template <typename T>
+__global__ void diff_kernel(const T* __restrict__ in, T* __restrict__ out,
+                            int64_t outer, int64_t inner) {
+  int64_t n_out = outer * (inner - 1);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_out;
+       t += stride) {
+    int64_t o = t / (inner - 1);
+    int64_t i = t % (inner - 1);
+    out[t] = in[o * inner + i + 1] - in[o * inner + i];
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1280,5 +1294,18 @@ void launch_cumsum_rows(DType dt, const void* in, void* out, int64_t rows,
     hipLaunchKernelGGL(cumsum_rows_kernel<float>, dim3((unsigned)rows),
                        dim3(256), 0, stream, (const float*)in, (float*)out,
                        rows, cols);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_diff(DType dt, const void* in, void* out, int64_t outer,
+                 int64_t inner, hipStream_t stream) {
+  int64_t n_out = outer * (inner - 1);
+  int grid = (int)std::min<int64_t>((n_out + 255) / 256, 4096);
+  if (dt == DType::F64)
+    hipLaunchKernelGGL(diff_kernel<double>, dim3(grid), dim3(256), 0, stream,
+                       (const double*)in, (double*)out, outer, inner);
+  else
+    hipLaunchKernelGGL(diff_kernel<float>, dim3(grid), dim3(256), 0, stream,
+                       (const float*)in, (float*)out, outer, inner);
   HIP_CHECK(hipGetLastError());
 }

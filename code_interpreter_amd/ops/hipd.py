@@ -198,6 +198,12 @@ class Connection(threading.Thread):
                 self._own(m["h"]), m["off"], m["nbytes"]
             )
             return {"ok": True}, data
+        if op == "diff":
+            h = _hipops.diff(
+                self._own(m["h"]), m["dtype"], m["outer"], m["inner"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "cumsum2d":
             h = _hipops.cumsum2d(
                 self._own(m["h"]), m["dtype"], m["rows"], m["cols"]

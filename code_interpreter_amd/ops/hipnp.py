@@ -340,6 +340,12 @@ class RemoteBackend:
             return out["h"], out["hi"]
         return out["h"]
 
+    def diff(self, h, dtype, outer, inner):
+        return self._call(
+            {"op": "diff", "h": h, "dtype": dtype, "outer": outer,
+             "inner": inner}
+        )["h"]
+
     def cumsum2d(self, h, dtype, rows, cols):
         return self._call(
             {"op": "cumsum2d", "h": h, "dtype": dtype, "rows": rows,
@@ -1436,6 +1442,30 @@ class DeviceArray:
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func is _np.diff and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"axis", "n"} and kwargs.get("n", 1) == 1:
+            a = args[0]
+            axis = kwargs.get("axis", -1)
+            code = _dtype_code(a.dtype)
+            if code is not None and len(a.shape) in (1, 2):
+                nd = len(a.shape)
+                ax = a._norm_axis(axis, nd)
+                if nd == 1 and ax == 0 and a.shape[0] >= 2:
+                    h = backend().diff(a._dev_handle(), code, 1, a.shape[0])
+                    return DeviceArray(h, (a.shape[0] - 1,), a.dtype)
+                if nd == 2 and ax == 1 and a.shape[1] >= 2:
+                    rows, cols = a.shape
+                    h = backend().diff(a._dev_handle(), code, rows, cols)
+                    return DeviceArray(h, (rows, cols - 1), a.dtype)
+                if nd == 2 and ax == 0 and a.shape[0] >= 2:
+                    t = a._device_transposed()
+                    rows, cols = t.shape
+                    h = backend().diff(t._dev_handle(), code, rows, cols)
+                    return DeviceArray(
+                        h, (rows, cols - 1), a.dtype
+                    )._device_transposed()
+            return _np.diff(a.materialize(), **kwargs)
         if func in (_np.cov, _np.corrcoef) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and not kwargs:

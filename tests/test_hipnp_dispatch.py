@@ -126,6 +126,11 @@ class FakeBackend:
             for i in range(count)
         )
 
+    def diff(self, h, dtype, outer, inner):
+        self.calls.append("diff")
+        a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: outer * inner]
+        return self._new(np.diff(a.reshape(outer, inner), axis=-1))
+
     def cumsum2d(self, h, dtype, rows, cols):
         self.calls.append("cumsum2d")
         dt = self._dt(dtype)
@@ -1087,3 +1092,21 @@ def test_cov_corrcoef_on_device(fake):
     small = np.random.default_rng(31).random((5, 30))
     np.testing.assert_allclose(
         np.cov(_device(fake, small)), np.cov(small), rtol=1e-12)
+
+
+def test_diff_on_device(fake):
+    host = np.random.default_rng(32).random(4000)
+    x = _device(fake, host)
+    d = np.diff(x)
+    assert isinstance(d, hipnp.DeviceArray) and d.shape == (3999,)
+    np.testing.assert_allclose(d.materialize(), np.diff(host), rtol=1e-12)
+    m = np.random.default_rng(33).random((20, 60))
+    d1 = np.diff(_device(fake, m), axis=1)
+    np.testing.assert_allclose(np.asarray(d1), np.diff(m, axis=1), rtol=1e-12)
+    d0 = np.diff(_device(fake, m), axis=0)
+    assert isinstance(d0, hipnp.DeviceArray) and d0.shape == (19, 60)
+    np.testing.assert_allclose(
+        d0.materialize(), np.diff(m, axis=0), rtol=1e-12)
+    # n=2 falls back to host numpy
+    np.testing.assert_allclose(
+        np.diff(_device(fake, host), n=2), np.diff(host, n=2), rtol=1e-12)

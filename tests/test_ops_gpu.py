@@ -1005,3 +1005,15 @@ def test_cov_corrcoef_gpu(hnp):
     r = np.corrcoef(m)
     np.testing.assert_allclose(
         r.materialize(), np.corrcoef(a), rtol=1e-8, atol=1e-12)
+
+
+def test_diff_gpu(hnp):
+    x = hnp.rand(3_000_000, seed=91)
+    a = np.asarray(x).copy()
+    d = np.diff(x)
+    assert isinstance(d, hnp.DeviceArray)
+    np.testing.assert_allclose(d.materialize(), np.diff(a), rtol=1e-12)
+    m = hnp.rand(500, 2000, seed=92)
+    ma = np.asarray(m).copy()
+    np.testing.assert_allclose(
+        np.diff(m, axis=0).materialize(), np.diff(ma, axis=0), rtol=1e-12)
