@@ -318,3 +318,32 @@ def test_sort_and_scalar_peek_on_device(gpu_executor):
     lo, hi = map(float, lines[1].split())
     assert 0.0 <= lo < 1e-5 and 1 - 1e-5 < hi <= 1.0
     assert lines[-1] == "ok"
+
+
+def test_analytics_surface_on_device(gpu_executor):
+    """The round-2 analytics widening (transpose, axis sort, row
+    median, histogram, cov, diff, cumsum axis) works inside a real
+    sandboxed execution and stays fast (i.e. on device)."""
+    r = _run(
+        gpu_executor,
+        "import numpy, time\n"
+        "x = numpy.random.rand(2000, 3000)\n"
+        "t0 = time.time()\n"
+        "med = numpy.median(x, axis=1)\n"
+        "hist, edges = numpy.histogram(x, bins=32)\n"
+        "cs = numpy.cumsum(x, axis=1)\n"
+        "d = numpy.diff(x, axis=1)\n"
+        "t = x.T\n"
+        "s0 = numpy.sort(x, axis=0)\n"
+        "dt = time.time() - t0\n"
+        "print(type(cs).__name__, type(t).__name__, type(s0).__name__)\n"
+        "assert med.shape == (2000,) and abs(float(med.mean()) - 0.5) < 0.01\n"
+        "assert int(hist.sum()) == x.size\n"
+        "assert t.shape == (3000, 2000)\n"
+        "assert dt < 3.0, f'not on device: {dt}'\n"
+        "print('ok', round(float(edges[0]), 3), round(float(edges[-1]), 3))\n",
+    )
+    assert r.exit_code == 0, r.stderr
+    lines = r.stdout.splitlines()
+    assert lines[0] == "DeviceArray DeviceArray DeviceArray"
+    assert lines[1].startswith("ok 0.0")
