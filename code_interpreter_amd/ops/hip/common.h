@@ -51,6 +51,8 @@ void launch_reduce_axis(DType dt, ReduceOp mode, const void* in, void* out,
                         hipStream_t stream);
 // radix sort (np.sort/argsort): see sort.hip header for scratch sizes
 int64_t radix_sort_nchunks(int64_t n);
+void launch_searchsorted(DType dt, const void* a, int64_t n, const void* v,
+                         int64_t m, int right, void* out, hipStream_t stream);
 void launch_diff(DType dt, const void* in, void* out, int64_t outer,
                  int64_t inner, hipStream_t stream);
 void launch_cumsum_rows(DType dt, const void* in, void* out, int64_t rows,

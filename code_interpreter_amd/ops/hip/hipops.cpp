@@ -723,6 +723,31 @@ PyObject* py_sort2d(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// searchsorted(ha, n, hv, m, dtype, right) -> int64 handle (m indices)
+PyObject* py_searchsorted(PyObject*, PyObject* args) {
+  unsigned long long ha, hv;
+  int dt, right;
+  long long n, m;
+  if (!PyArg_ParseTuple(args, "KLKLii", &ha, &n, &hv, &m, &dt, &right))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  if (n < 0 || m < 1) throw std::runtime_error("bad searchsorted sizes");
+  DType dtype = dtype_from_int(dt);
+  DevBuf& a = get_buf(ha);
+  DevBuf& v = get_buf(hv);
+  int64_t esize = dtype == DType::F64 ? 8 : 4;
+  if (n * esize > a.size || m * esize > v.size)
+    throw std::runtime_error("searchsorted oob");
+  void* out = nullptr;
+  NOGIL_BEGIN
+  out = pool_alloc(m * 8);
+  launch_searchsorted(dtype, a.ptr, n, v.ptr, m, right, out, g.compute);
+  NOGIL_END
+  return PyLong_FromUnsignedLongLong(register_buf(out, m * 8));
+  WRAP_END
+}
+
 // diff(h, dtype, outer, inner) -> handle (outer x (inner-1))
 PyObject* py_diff(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1286,6 +1311,8 @@ PyMethodDef methods[] = {
     {"sum", py_sum, METH_VARARGS, "sum(h, dtype, n, mode) -> float (mode 0=sum 1=sumsq 2=max 3=min)"},
     {"gemm", py_gemm, METH_VARARGS, "gemm(ha, hb, m, n, k, dtype) -> handle"},
     {"cumsum", py_cumsum, METH_VARARGS, "cumsum(h, dtype, n) -> handle"},
+    {"searchsorted", py_searchsorted, METH_VARARGS,
+     "searchsorted(ha, n, hv, m, dtype, right) -> int64 handle"},
     {"diff", py_diff, METH_VARARGS,
      "diff(h, dtype, outer, inner) -> handle"},
     {"cumsum2d", py_cumsum2d, METH_VARARGS,

@@ -882,6 +882,35 @@ __global__ void diff_kernel(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+// np.searchsorted: one thread per query, binary search with numpy's
+// sort ordering (NaN compares greater than everything, so NaN queries
+// land among the trailing NaNs exactly like numpy)
+template <typename T>
+__device__ __forceinline__ bool np_lt(T x, T y) {
+  return x < y || (y != y && x == x);
+}
+
+template <typename T, bool RIGHT>
+__global__ void searchsorted_kernel(const T* __restrict__ a, int64_t n,
+                                    const T* __restrict__ v, int64_t m,
+                                    long long* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < m;
+       t += stride) {
+    T val = v[t];
+    int64_t lo = 0, hi = n;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      bool go_right = RIGHT ? !np_lt(val, a[mid]) : np_lt(a[mid], val);
+      if (go_right)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    out[t] = lo;
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1307,5 +1336,31 @@ void launch_diff(DType dt, const void* in, void* out, int64_t outer,
   else
     hipLaunchKernelGGL(diff_kernel<float>, dim3(grid), dim3(256), 0, stream,
                        (const float*)in, (float*)out, outer, inner);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_searchsorted(DType dt, const void* a, int64_t n, const void* v,
+                         int64_t m, int right, void* out,
+                         hipStream_t stream) {
+  int grid = (int)std::min<int64_t>((m + 255) / 256, 4096);
+  if (dt == DType::F64) {
+    if (right)
+      hipLaunchKernelGGL((searchsorted_kernel<double, true>), dim3(grid),
+                         dim3(256), 0, stream, (const double*)a, n,
+                         (const double*)v, m, (long long*)out);
+    else
+      hipLaunchKernelGGL((searchsorted_kernel<double, false>), dim3(grid),
+                         dim3(256), 0, stream, (const double*)a, n,
+                         (const double*)v, m, (long long*)out);
+  } else {
+    if (right)
+      hipLaunchKernelGGL((searchsorted_kernel<float, true>), dim3(grid),
+                         dim3(256), 0, stream, (const float*)a, n,
+                         (const float*)v, m, (long long*)out);
+    else
+      hipLaunchKernelGGL((searchsorted_kernel<float, false>), dim3(grid),
+                         dim3(256), 0, stream, (const float*)a, n,
+                         (const float*)v, m, (long long*)out);
+  }
   HIP_CHECK(hipGetLastError());
 }

@@ -198,6 +198,13 @@ class Connection(threading.Thread):
                 self._own(m["h"]), m["off"], m["nbytes"]
             )
             return {"ok": True}, data
+        if op == "searchsorted":
+            h = _hipops.searchsorted(
+                self._own(m["ha"]), m["n"], self._own(m["hv"]), m["m"],
+                m["dtype"], m["right"]
+            )
+            self.handles.add(h)
+            return {"ok": True, "h": h}, b""
         if op == "diff":
             h = _hipops.diff(
                 self._own(m["h"]), m["dtype"], m["outer"], m["inner"]

@@ -340,6 +340,12 @@ class RemoteBackend:
             return out["h"], out["hi"]
         return out["h"]
 
+    def searchsorted(self, ha, n, hv, m, dtype, right):
+        return self._call(
+            {"op": "searchsorted", "ha": ha, "n": n, "hv": hv, "m": m,
+             "dtype": dtype, "right": right}
+        )["h"]
+
     def diff(self, h, dtype, outer, inner):
         return self._call(
             {"op": "diff", "h": h, "dtype": dtype, "outer": outer,
@@ -1442,6 +1448,55 @@ class DeviceArray:
                 return idx
             host = a.materialize()
             return func(host, **kwargs)
+        if func is _np.searchsorted and len(args) in (2, 3) and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"side"}:
+            a = args[0]
+            v = args[1]
+            side = args[2] if len(args) == 3 else kwargs.get("side", "left")
+            code = _dtype_code(a.dtype)
+            if (
+                code is not None
+                and len(a.shape) == 1
+                and side in ("left", "right")
+            ):
+                scalar_q = _np.isscalar(v) or (
+                    isinstance(v, _np.ndarray) and v.ndim == 0
+                )
+                if isinstance(v, DeviceArray):
+                    if v.dtype == a.dtype and len(v.shape) >= 1:
+                        hv, m, vshape = v._dev_handle(), v.size, v.shape
+                    else:
+                        hv = None
+                elif scalar_q or isinstance(v, (list, tuple, _np.ndarray)):
+                    host_v = _np.ascontiguousarray(
+                        _np.asarray(v, dtype=a.dtype).reshape(-1)
+                    )
+                    if host_v.size >= 1:
+                        # wrapped so the staging buffer is freed on GC
+                        _tmp_q = DeviceArray(
+                            backend().upload(host_v), host_v.shape, a.dtype
+                        )
+                        hv, m, vshape = _tmp_q._handle, \
+                            host_v.size, _np.asarray(v).shape
+                    else:
+                        hv = None
+                else:
+                    hv = None
+                if hv is not None:
+                    h = backend().searchsorted(
+                        a._dev_handle(), a.shape[0], hv, m, code,
+                        1 if side == "right" else 0,
+                    )
+                    res = DeviceArray(h, (m,), _np.int64)
+                    if scalar_q:
+                        out = _np.intp(res.materialize()[0])
+                        return out
+                    if vshape != (m,):
+                        return res.materialize().reshape(vshape)
+                    return res
+            host = a.materialize()
+            return _np.searchsorted(host, *args[1:], **kwargs)
         if func is _np.diff and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and set(kwargs) <= {"axis", "n"} and kwargs.get("n", 1) == 1:

@@ -126,6 +126,14 @@ class FakeBackend:
             for i in range(count)
         )
 
+    def searchsorted(self, ha, n, hv, m, dtype, right):
+        self.calls.append("searchsorted")
+        dt = self._dt(dtype)
+        a = self.bufs[ha].view(dt).reshape(-1)[:n]
+        v = self.bufs[hv].view(dt).reshape(-1)[:m]
+        side = "right" if right else "left"
+        return self._new(np.searchsorted(a, v, side=side).astype(np.int64))
+
     def diff(self, h, dtype, outer, inner):
         self.calls.append("diff")
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[: outer * inner]
@@ -1110,3 +1118,24 @@ def test_diff_on_device(fake):
     # n=2 falls back to host numpy
     np.testing.assert_allclose(
         np.diff(_device(fake, host), n=2), np.diff(host, n=2), rtol=1e-12)
+
+
+def test_searchsorted_on_device(fake):
+    a = np.sort(np.random.default_rng(34).random(10000))
+    q = np.random.default_rng(35).random(500)
+    x = _device(fake, a)
+    r = np.searchsorted(x, _device(fake, q))
+    assert isinstance(r, hipnp.DeviceArray) and r.dtype == np.int64
+    np.testing.assert_array_equal(r.materialize(), np.searchsorted(a, q))
+    # host queries, right side
+    r2 = np.searchsorted(_device(fake, a), q, side="right")
+    np.testing.assert_array_equal(
+        np.asarray(r2), np.searchsorted(a, q, side="right"))
+    # scalar query -> scalar result
+    rs = np.searchsorted(_device(fake, a), 0.5)
+    assert np.isscalar(rs) or np.asarray(rs).ndim == 0
+    assert int(rs) == int(np.searchsorted(a, 0.5))
+    # 2-D queries keep their shape
+    q2 = q[:100].reshape(10, 10)
+    r3 = np.searchsorted(_device(fake, a), q2)
+    np.testing.assert_array_equal(np.asarray(r3), np.searchsorted(a, q2))

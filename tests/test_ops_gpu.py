@@ -1017,3 +1017,17 @@ def test_diff_gpu(hnp):
     ma = np.asarray(m).copy()
     np.testing.assert_allclose(
         np.diff(m, axis=0).materialize(), np.diff(ma, axis=0), rtol=1e-12)
+
+
+def test_searchsorted_gpu(hnp):
+    x = hnp.rand(5_000_000, seed=61)
+    s = np.sort(x)  # device sort
+    a = np.asarray(s).copy()
+    q = hnp.rand(100_000, seed=62)
+    qa = np.asarray(q).copy()
+    r = np.searchsorted(s, q)
+    assert isinstance(r, hnp.DeviceArray)
+    np.testing.assert_array_equal(r.materialize(), np.searchsorted(a, qa))
+    r2 = np.searchsorted(s, qa[:50], side="right")
+    np.testing.assert_array_equal(
+        np.asarray(r2), np.searchsorted(a, qa[:50], side="right"))
