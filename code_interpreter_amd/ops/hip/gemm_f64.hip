@@ -284,7 +284,7 @@ __global__ __launch_bounds__(THREADS2) void gemm_f64_128_kernel(
 // the barrier only fences buffer reuse two tiles later. BK=8 keeps the
 // triple buffer at ~52 KB -> 3 blocks/CU (6 waves/SIMD vs 4).
 // ---------------------------------------------------------------------------
-template <int BKT>
+template <int BKT, bool PRIO = false>
 __global__ __launch_bounds__(THREADS2) void gemm_f64_128p_kernel(
     const double* __restrict__ A, const double* __restrict__ B,
     double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
@@ -358,6 +358,7 @@ __global__ __launch_bounds__(THREADS2) void gemm_f64_128p_kernel(
   const int am0 = wave_m * 64;
   const int bn0 = wave_n * 32;
   auto compute_tile = [&](int buf) {
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < BKT; ks += 4) {
       double a0 = As[buf][am0 + l15][ks + lk];
@@ -375,6 +376,7 @@ __global__ __launch_bounds__(THREADS2) void gemm_f64_128p_kernel(
       acc[2][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b1, acc[2][1], 0, 0, 0);
       acc[3][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b0, acc[3][0], 0, 0, 0);
     }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
   };
 
   // prologue: buffer 0 staged and visible, buffer 1's loads in regs
@@ -568,7 +570,11 @@ void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
   if (!(v && v[0] == 's')) {
     int tiles_m2 = (m + BM2 - 1) / BM2;
     int tiles_n2 = (n + BN2 - 1) / BN2;
-    if (v && v[0] == 'w')
+    if (v && v[0] == 'P')
+      hipLaunchKernelGGL((gemm_f64_128p_kernel<8, true>),
+                         dim3(tiles_m2 * tiles_n2), dim3(THREADS2), 0, stream,
+                         a, b, c, m, n, k, tiles_m2, tiles_n2);
+    else if (v && v[0] == 'w')
       hipLaunchKernelGGL((gemm_f64_128w_kernel<8>), dim3(tiles_m2 * tiles_n2),
                          dim3(THREADS3), 0, stream, a, b, c, m, n, k,
                          tiles_m2, tiles_n2);
