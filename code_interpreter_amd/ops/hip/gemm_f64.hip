@@ -556,6 +556,167 @@ __global__ __launch_bounds__(THREADS3) void gemm_f64_128w_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// glds variant: stages A and B via global_load_lds (16 B/lane LDS-DMA,
+// lane-linear), eliminating the staging registers AND the ds_write
+// convoy -- the PMC breakdown shows the shipped pipeline is VGPR-capped
+// at 4 waves/SIMD with 23% of wave cycles parked on vmcnt/barriers.
+// Without a_reg/b_reg the kernel fits 5 waves/SIMD. glds's lane-linear
+// LDS image cannot take the +1 padding, so the operand layout is
+// XOR-swizzled instead:
+//   A slot(r, kpair) = r*4 + (kpair ^ ((r>>2)&3))   (16-B slots)
+//   B slot(k, cpair) = k*64 + (cpair ^ 8k)
+// Both make every 64-lane ds_read_b64 operand fetch hit each qword
+// bank exactly twice (the b64 minimum). Requires M,N %128==0, K%8==0;
+// the launcher falls back to the 'p' kernel otherwise.
+// ---------------------------------------------------------------------------
+using lds_void_f64 = __attribute__((address_space(3))) void;
+using global_void_f64 = const __attribute__((address_space(1))) void;
+
+template <int MINB, int NBUF = 3>
+__global__ __launch_bounds__(THREADS2, MINB) void gemm_f64_glds_kernel(
+    const double* __restrict__ A, const double* __restrict__ B,
+    double* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x 16 KB
+  int tm, tn;
+  {
+    int nwg = tiles_m * tiles_n;
+    int wgid = blockIdx.x;
+    const int n_st = (tiles_m / 8) * (tiles_n / 4);
+    if (tiles_m % 8 == 0 && tiles_n % 4 == 0 && n_st % 8 == 0) {
+      const int st_cols = tiles_n / 4;
+      int xcd = wgid % 8, idx = wgid / 8;
+      int st = xcd + 8 * (idx >> 5);
+      int p = idx & 31;
+      tm = (st / st_cols) * 8 + (p >> 2);
+      tn = (st % st_cols) * 4 + (p & 3);
+    } else {
+      const int nxcd = 8;
+      int q = nwg / nxcd, r = nwg % nxcd;
+      int xcd = wgid % nxcd, idx = wgid / nxcd;
+      wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+      tm = wgid / tiles_n;
+      tn = wgid % tiles_n;
+    }
+  }
+  const int row0 = tm * BM2;
+  const int col0 = tn * BN2;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;    // 0..7
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int l15 = lane & 15;
+  const int lk = lane >> 4;
+
+  f64x4 acc[4][2] = {};
+
+  // per-thread staging sources (fixed across tiles except k0)
+  const int a_r = tid >> 2;
+  const int a_q = (tid & 3) ^ ((tid >> 4) & 3);
+  const double* a_src0 = A + (int64_t)(row0 + a_r) * K + 2 * a_q;
+  const int b_cq = lane ^ (wave * 8);
+  const double* b_src0 = B + (int64_t)wave * N + col0 + 2 * b_cq;
+
+  auto stage = [&](int ktile) {
+    int buf = ktile % NBUF;
+    int k0 = ktile * 8;
+    unsigned dst = (unsigned)buf * 16384u + (unsigned)wave * 1024u;
+    __builtin_amdgcn_global_load_lds(
+        (global_void_f64*)(a_src0 + k0), (lds_void_f64*)(smem + dst), 16, 0,
+        0);
+    __builtin_amdgcn_global_load_lds(
+        (global_void_f64*)(b_src0 + (int64_t)k0 * N),
+        (lds_void_f64*)(smem + 8192u + dst), 16, 0, 0);
+  };
+
+  auto a_val = [&](int buf, int row, int k) -> double {
+    unsigned slot = (unsigned)row * 4u +
+                    (unsigned)((k >> 1) ^ ((row >> 2) & 3));
+    return *reinterpret_cast<const double*>(
+        smem + (unsigned)buf * 16384u + slot * 16u + (unsigned)(k & 1) * 8u);
+  };
+  auto b_val = [&](int buf, int k, int col) -> double {
+    unsigned slot = (unsigned)k * 64u + (unsigned)((col >> 1) ^ (k * 8));
+    return *reinterpret_cast<const double*>(
+        smem + (unsigned)buf * 16384u + 8192u + slot * 16u +
+        (unsigned)(col & 1) * 8u);
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 32;
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < 8; ks += 4) {
+      int k = ks + lk;
+      double a0 = a_val(buf, am0 + l15, k);
+      double a1 = a_val(buf, am0 + 16 + l15, k);
+      double a2 = a_val(buf, am0 + 32 + l15, k);
+      double a3 = a_val(buf, am0 + 48 + l15, k);
+      double b0 = b_val(buf, k, bn0 + l15);
+      double b1 = b_val(buf, k, bn0 + 16 + l15);
+      acc[0][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b0, acc[0][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b1, acc[1][1], 0, 0, 0);
+      acc[2][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b0, acc[2][0], 0, 0, 0);
+      acc[3][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b1, acc[3][1], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a1, b0, acc[1][0], 0, 0, 0);
+      acc[2][1] = __builtin_amdgcn_mfma_f64_16x16x4f64(a2, b1, acc[2][1], 0, 0, 0);
+      acc[3][0] = __builtin_amdgcn_mfma_f64_16x16x4f64(a3, b0, acc[3][0], 0, 0, 0);
+    }
+  };
+
+  const int n_ktiles = K / 8;
+  const int depth = NBUF - 1;  // tiles in flight beyond the current one
+  for (int t = 0; t < depth && t < n_ktiles; t++) stage(t);
+  {
+    int inflight = n_ktiles > 1 ? 2 * (std::min(depth, n_ktiles) - 1) : 0;
+    if (inflight >= 6)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (inflight >= 4)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (inflight >= 2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  for (int kt = 0; kt < n_ktiles; kt++) {
+    if (kt + depth < n_ktiles) stage(kt + depth);
+    compute_tile(kt % NBUF);
+    if (kt + 1 < n_ktiles) {
+      // retire everything except the loads issued after tile kt+1's
+      int pending = 0;
+      for (int t = kt + 2; t <= kt + depth && t < n_ktiles; t++) pending += 2;
+      if (pending >= 6)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      else if (pending >= 4)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else if (pending >= 2)
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  const int crow0 = row0 + wave_m * 64 + (lane >> 4);
+  const int ccol0 = col0 + wave_n * 32 + l15;
+#pragma unroll
+  for (int mt = 0; mt < 4; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 2; nt++) {
+      int col = ccol0 + nt * 16;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int row = crow0 + mt * 16 + 4 * reg;
+        C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
@@ -570,6 +731,37 @@ void launch_gemm_f64(const double* a, const double* b, double* c, int m, int n,
   if (!(v && v[0] == 's')) {
     int tiles_m2 = (m + BM2 - 1) / BM2;
     int tiles_n2 = (n + BN2 - 1) / BN2;
+    bool full_tiles = (m % BM2 == 0) && (n % BN2 == 0) && (k % 8 == 0);
+    // default for full 128-multiple tiles: the 4-buffer glds kernel
+    // (67.7 TF @8192^3 same-box vs 63.5 for the register-staged
+    // pipeline; depth ladder in profiles/NOTES.md)
+    if (full_tiles && !v) {
+      hipLaunchKernelGGL((gemm_f64_glds_kernel<1, 4>),
+                         dim3(tiles_m2 * tiles_n2), dim3(THREADS2),
+                         4 * 16384, stream, a, b, c, m, n, k, tiles_m2,
+                         tiles_n2);
+      HIP_CHECK(hipGetLastError());
+      return;
+    }
+    if (v && v[0] == 'g' && full_tiles) {
+      if (v[1] == '5')
+        hipLaunchKernelGGL((gemm_f64_glds_kernel<1, 5>),
+                           dim3(tiles_m2 * tiles_n2), dim3(THREADS2),
+                           5 * 16384, stream, a, b, c, m, n, k, tiles_m2,
+                           tiles_n2);
+      else if (v[1] == '4')
+        hipLaunchKernelGGL((gemm_f64_glds_kernel<1, 4>),
+                           dim3(tiles_m2 * tiles_n2), dim3(THREADS2),
+                           4 * 16384, stream, a, b, c, m, n, k, tiles_m2,
+                           tiles_n2);
+      else
+        hipLaunchKernelGGL((gemm_f64_glds_kernel<1, 3>),
+                           dim3(tiles_m2 * tiles_n2), dim3(THREADS2),
+                           3 * 16384, stream, a, b, c, m, n, k, tiles_m2,
+                           tiles_n2);
+      HIP_CHECK(hipGetLastError());
+      return;
+    }
     if (v && v[0] == 'P')
       hipLaunchKernelGGL((gemm_f64_128p_kernel<8, true>),
                          dim3(tiles_m2 * tiles_n2), dim3(THREADS2), 0, stream,
