@@ -331,6 +331,159 @@ __global__ __launch_bounds__(THREADS) void gemm_f32_bk16_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// B-glds variant: B staged via global_load_lds into the PLAIN [BK][128]
+// image (glds of row-major B rows needs no swizzle: the 64-lane b32
+// operand read is exactly 2-way, the minimum), removing B's staging
+// registers and ds_write convoy; A keeps the register path (the
+// 32x32x2f32 A-read's k-parity split fights any 16-B-slot glds image).
+// Full 128-multiple tiles only; the launcher falls back otherwise.
+// ---------------------------------------------------------------------------
+using lds_void_f32 = __attribute__((address_space(3))) void;
+using global_void_f32 = const __attribute__((address_space(1))) void;
+
+template <int BKT>
+__global__ __launch_bounds__(THREADS) void gemm_f32_gldsb_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char bsmem[];  // 2 x BKT*512
+  int tm, tn;
+  {
+    int nwg = tiles_m * tiles_n;
+    int wgid = blockIdx.x;
+    const int n_st = (tiles_m / 8) * (tiles_n / 4);
+    if (tiles_m % 8 == 0 && tiles_n % 4 == 0 && n_st % 8 == 0) {
+      const int st_cols = tiles_n / 4;
+      int xcd = wgid % 8, idx = wgid / 8;
+      int st = xcd + 8 * (idx >> 5);
+      int p = idx & 31;
+      tm = (st / st_cols) * 8 + (p >> 2);
+      tn = (st % st_cols) * 4 + (p & 3);
+    } else {
+      const int nxcd = 8;
+      int q = nwg / nxcd, r = nwg % nxcd;
+      int xcd = wgid % nxcd, idx = wgid / nxcd;
+      wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+      tm = wgid / tiles_n;
+      tn = wgid % tiles_n;
+    }
+  }
+  int row0 = tm * BM;
+  int col0 = tn * BN;
+
+  __shared__ float As[2][BKT][BM + 1];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wave_m = wave >> 1;
+  const int wave_n = wave & 1;
+  const int l31 = lane & 31;
+  const int lk = lane >> 5;
+
+  f32x16 acc[2][2] = {};
+
+  constexpr int AQ = BM * BKT / THREADS / 4;
+  constexpr int KV = BKT / 4;
+  const int a_m = tid / KV;
+  const int a_k = (tid % KV) * 4;
+  constexpr int AROWS = THREADS / KV;
+
+  float4 a_reg[AQ];
+
+  auto issue_a = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < AQ; i++) {
+      int gr = row0 + a_m + i * AROWS;
+      a_reg[i] =
+          *reinterpret_cast<const float4*>(&A[(int64_t)gr * K + k0 + a_k]);
+    }
+  };
+  auto write_a = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < AQ; i++) {
+      As[buf][a_k + 0][a_m + i * AROWS] = a_reg[i].x;
+      As[buf][a_k + 1][a_m + i * AROWS] = a_reg[i].y;
+      As[buf][a_k + 2][a_m + i * AROWS] = a_reg[i].z;
+      As[buf][a_k + 3][a_m + i * AROWS] = a_reg[i].w;
+    }
+  };
+  // B: BKT*128 floats = BKT*32 16-B slots; slot s covers
+  // (k = s>>5, n4 = s&31); wave w's glds instr j covers
+  // s = j*256 + w*64 + lane (plain layout: lds addr = s*16)
+  constexpr unsigned kBBuf = (unsigned)BKT * 512u;
+  auto stage_b = [&](int ktile) {
+    unsigned base = (unsigned)(ktile & 1) * kBBuf;
+    int k0 = ktile * BKT;
+#pragma unroll
+    for (int j = 0; j < BKT * 32 / THREADS; j++) {
+      int slot = j * THREADS + wave * 64;  // + lane, implicit in glds
+      int k = (slot + lane) >> 5;
+      int n4 = (slot + lane) & 31;
+      const float* gsrc = B + (int64_t)(k0 + k) * N + col0 + n4 * 4;
+      __builtin_amdgcn_global_load_lds(
+          (global_void_f32*)gsrc,
+          (lds_void_f32*)(bsmem + base + (unsigned)slot * 16u), 16, 0, 0);
+    }
+  };
+  auto b_val = [&](int buf, int k, int n) -> float {
+    return *reinterpret_cast<const float*>(
+        bsmem + (unsigned)buf * kBBuf + (unsigned)k * 512u + (unsigned)n * 4u);
+  };
+
+  const int am0 = wave_m * 64;
+  const int bn0 = wave_n * 64;
+  auto compute_tile = [&](int buf) {
+#pragma unroll
+    for (int ks = 0; ks < BKT; ks += 2) {
+      float a0 = As[buf][ks + lk][am0 + l31];
+      float a1 = As[buf][ks + lk][am0 + 32 + l31];
+      float b0 = b_val(buf, ks + lk, bn0 + l31);
+      float b1 = b_val(buf, ks + lk, bn0 + 32 + l31);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+  };
+
+  stage_b(0);
+  issue_a(0);
+  write_a(0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BKT) {
+    bool have_next = k0 + BKT < K;
+    if (have_next) {
+      stage_b((k0 / BKT) + 1);  // overlaps the MFMA stream below
+      issue_a(k0 + BKT);
+    }
+    compute_tile(cur);
+    if (have_next) {
+      write_a(cur ^ 1);
+      cur ^= 1;
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const int crow0 = row0 + wave_m * 64 + 4 * (lane >> 5);
+  const int ccol0 = col0 + wave_n * 64 + l31;
+#pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+#pragma unroll
+    for (int nt = 0; nt < 2; nt++) {
+      int col = ccol0 + nt * 32;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        int row = crow0 + mt * 32 + (reg & 3) + 8 * (reg >> 2);
+        C[(int64_t)row * N + col] = acc[mt][nt][reg];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,
@@ -341,6 +494,14 @@ void launch_gemm_f32(const float* a, const float* b, float* c, int m, int n,
   // BK=32 -- same-box A/B, profiles/NOTES.md r02); APP_F32_VARIANT=k32
   // keeps the old kernel for comparison
   const char* v = getenv("APP_F32_VARIANT");
+  bool full_tiles = (m % BM == 0) && (n % BN == 0) && (k % 16 == 0);
+  if (v && v[0] == 'g' && full_tiles) {
+    hipLaunchKernelGGL((gemm_f32_gldsb_kernel<16>), dim3(tiles_m * tiles_n),
+                       dim3(THREADS), 2 * 16 * 512, stream, a, b, c, m, n, k,
+                       tiles_m, tiles_n);
+    HIP_CHECK(hipGetLastError());
+    return;
+  }
   if (v && v[0] == 'k') {
     hipLaunchKernelGGL(gemm_f32_kernel, dim3(tiles_m * tiles_n), dim3(THREADS),
                        0, stream, a, b, c, m, n, k, tiles_m, tiles_n);
