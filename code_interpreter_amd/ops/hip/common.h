@@ -25,6 +25,10 @@ enum class DType : int { F32 = 0, F64 = 1 };
 enum class UnaryOp : int {
   Square = 0, Neg = 1, Abs = 2, Sqrt = 3, Exp = 4,
   Log = 5, Sin = 6, Cos = 7, Tanh = 8,
+  Floor = 9, Ceil = 10, Rint = 11, Trunc = 12, Sign = 13,
+  Log2 = 14, Log10 = 15, Exp2 = 16, Expm1 = 17, Log1p = 18,
+  Cbrt = 19, Tan = 20, Arcsin = 21, Arccos = 22, Arctan = 23,
+  Sinh = 24, Cosh = 25,
 };
 enum class BinOp : int {
   Add = 0, Sub = 1, Mul = 2, Div = 3, Max = 4, Min = 5, Pow = 6,

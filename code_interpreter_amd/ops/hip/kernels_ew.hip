@@ -46,6 +46,25 @@ __device__ __forceinline__ T apply_unary(T v) {
   if constexpr (OP == (int)UnaryOp::Sin) return (T)sin((double)v);
   if constexpr (OP == (int)UnaryOp::Cos) return (T)cos((double)v);
   if constexpr (OP == (int)UnaryOp::Tanh) return (T)tanh((double)v);
+  if constexpr (OP == (int)UnaryOp::Floor) return (T)floor((double)v);
+  if constexpr (OP == (int)UnaryOp::Ceil) return (T)ceil((double)v);
+  // numpy round/rint: round-half-to-even
+  if constexpr (OP == (int)UnaryOp::Rint) return (T)rint((double)v);
+  if constexpr (OP == (int)UnaryOp::Trunc) return (T)trunc((double)v);
+  if constexpr (OP == (int)UnaryOp::Sign)
+    return v != v ? v : (v > T(0) ? T(1) : (v < T(0) ? T(-1) : v));
+  if constexpr (OP == (int)UnaryOp::Log2) return (T)log2((double)v);
+  if constexpr (OP == (int)UnaryOp::Log10) return (T)log10((double)v);
+  if constexpr (OP == (int)UnaryOp::Exp2) return (T)exp2((double)v);
+  if constexpr (OP == (int)UnaryOp::Expm1) return (T)expm1((double)v);
+  if constexpr (OP == (int)UnaryOp::Log1p) return (T)log1p((double)v);
+  if constexpr (OP == (int)UnaryOp::Cbrt) return (T)cbrt((double)v);
+  if constexpr (OP == (int)UnaryOp::Tan) return (T)tan((double)v);
+  if constexpr (OP == (int)UnaryOp::Arcsin) return (T)asin((double)v);
+  if constexpr (OP == (int)UnaryOp::Arccos) return (T)acos((double)v);
+  if constexpr (OP == (int)UnaryOp::Arctan) return (T)atan((double)v);
+  if constexpr (OP == (int)UnaryOp::Sinh) return (T)sinh((double)v);
+  if constexpr (OP == (int)UnaryOp::Cosh) return (T)cosh((double)v);
   return v;
 }
 
@@ -928,6 +947,10 @@ static void launch_unary_t(UnaryOp op, const T* in, T* out, int64_t n,
     break;
     CASE(Square) CASE(Neg) CASE(Abs) CASE(Sqrt) CASE(Exp)
     CASE(Log) CASE(Sin) CASE(Cos) CASE(Tanh)
+    CASE(Floor) CASE(Ceil) CASE(Rint) CASE(Trunc) CASE(Sign)
+    CASE(Log2) CASE(Log10) CASE(Exp2) CASE(Expm1) CASE(Log1p)
+    CASE(Cbrt) CASE(Tan) CASE(Arcsin) CASE(Arccos) CASE(Arctan)
+    CASE(Sinh) CASE(Cosh)
 #undef CASE
   }
   HIP_CHECK(hipGetLastError());

@@ -449,6 +449,10 @@ _F32, _F64 = 0, 1
 _UNARY = {
     "square": 0, "negative": 1, "absolute": 2, "sqrt": 3, "exp": 4,
     "log": 5, "sin": 6, "cos": 7, "tanh": 8,
+    "floor": 9, "ceil": 10, "rint": 11, "trunc": 12, "sign": 13,
+    "log2": 14, "log10": 15, "exp2": 16, "expm1": 17, "log1p": 18,
+    "cbrt": 19, "tan": 20, "arcsin": 21, "arccos": 22, "arctan": 23,
+    "sinh": 24, "cosh": 25,
 }
 _BINARY = {
     "add": 0, "subtract": 1, "multiply": 2, "divide": 3, "true_divide": 3,
@@ -1144,6 +1148,11 @@ class DeviceArray:
             kth, axis=axis, kind=kind, order=order, **kwargs
         )
 
+    def round(self, decimals=0, **kwargs):
+        if decimals == 0 and not kwargs and _dtype_code(self.dtype) is not None:
+            return self._unary("rint")
+        return self.materialize().round(decimals, **kwargs)
+
     def isnan(self):
         """Device NaN mask (x != x elementwise)."""
         r = self._compare("not_equal", self)
@@ -1262,7 +1271,8 @@ class DeviceArray:
                 r = self._compare(name, inputs[1])
                 if r is not None:
                     return r
-            if name in _UNARY and len(inputs) == 1 and inputs[0] is self:
+            if name in _UNARY and len(inputs) == 1 and inputs[0] is self \
+                    and _dtype_code(self.dtype) is not None:
                 return self._unary(name)
             if name in _BINARY and len(inputs) == 2:
                 a, b = inputs
@@ -1557,6 +1567,11 @@ class DeviceArray:
                     return hist, edges
             host = a.materialize()
             return _np.histogram(host, *args[1:], **kwargs)
+        if func in (_np.round, _np.around) and len(args) >= 1 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"decimals"}:
+            dec = args[1] if len(args) > 1 else kwargs.get("decimals", 0)
+            return args[0].round(dec)
         if func is _np.transpose and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and set(kwargs) <= {"axes"}:

@@ -25,6 +25,10 @@ class FakeBackend:
     _UOPS = {
         0: np.square, 1: np.negative, 2: np.abs, 3: np.sqrt, 4: np.exp,
         5: np.log, 6: np.sin, 7: np.cos, 8: np.tanh,
+        9: np.floor, 10: np.ceil, 11: np.rint, 12: np.trunc, 13: np.sign,
+        14: np.log2, 15: np.log10, 16: np.exp2, 17: np.expm1, 18: np.log1p,
+        19: np.cbrt, 20: np.tan, 21: np.arcsin, 22: np.arccos,
+        23: np.arctan, 24: np.sinh, 25: np.cosh,
     }
     _BOPS = {
         0: np.add, 1: np.subtract, 2: np.multiply, 3: np.divide,
@@ -367,9 +371,9 @@ def test_ufunc_reduce_methods(fake):
 def test_unsupported_ufunc_materializes(fake):
     host = np.random.default_rng(4).random(64)
     x = _device(fake, host)
-    out = np.arctan(x)  # not in the device op set
+    out = np.arctan2(x, x)  # not in the device op set
     assert isinstance(out, np.ndarray)
-    np.testing.assert_allclose(out, np.arctan(host), rtol=1e-12)
+    np.testing.assert_allclose(out, np.arctan2(host, host), rtol=1e-12)
 
 
 def test_matmul_chain(fake):
@@ -1139,3 +1143,25 @@ def test_searchsorted_on_device(fake):
     q2 = q[:100].reshape(10, 10)
     r3 = np.searchsorted(_device(fake, a), q2)
     np.testing.assert_array_equal(np.asarray(r3), np.searchsorted(a, q2))
+
+
+def test_widened_unary_set(fake):
+    host = np.random.default_rng(36).random(3000) * 1.9 - 0.95
+    x = _device(fake, host)
+    for f in (np.floor, np.ceil, np.rint, np.trunc, np.sign, np.expm1,
+              np.log1p, np.cbrt, np.tan, np.arcsin, np.arccos, np.arctan,
+              np.sinh, np.cosh):
+        r = f(x)
+        assert isinstance(r, hipnp.DeviceArray), f.__name__
+        np.testing.assert_allclose(
+            r.materialize(), f(host), rtol=1e-12, err_msg=f.__name__)
+    pos = _device(fake, host + 1.0)
+    for f in (np.log2, np.log10, np.exp2):
+        np.testing.assert_allclose(
+            f(pos).materialize(), f(host + 1.0), rtol=1e-12)
+    r = np.round(_device(fake, host * 10))
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_array_equal(r.materialize(), np.round(host * 10))
+    # round with decimals falls back to host values
+    np.testing.assert_allclose(
+        np.round(_device(fake, host), 2), np.round(host, 2), rtol=0)

@@ -1031,3 +1031,21 @@ def test_searchsorted_gpu(hnp):
     r2 = np.searchsorted(s, qa[:50], side="right")
     np.testing.assert_array_equal(
         np.asarray(r2), np.searchsorted(a, qa[:50], side="right"))
+
+
+def test_widened_unary_set_gpu(hnp):
+    x = hnp.rand(2_000_000, seed=93)
+    a = np.asarray(x).copy()
+    y = (x - 0.5) * 1.9  # (-0.95, 0.95): safe for arcsin/arccos
+    ya = (a - 0.5) * 1.9
+    for f in (np.floor, np.ceil, np.rint, np.trunc, np.sign, np.expm1,
+              np.log1p, np.cbrt, np.tan, np.arcsin, np.arccos, np.arctan,
+              np.sinh, np.cosh, np.log2, np.log10, np.exp2):
+        src, ref = (y, ya) if f not in (np.log2, np.log10) else (x, a)
+        r = f(src)
+        assert isinstance(r, hnp.DeviceArray), f.__name__
+        np.testing.assert_allclose(
+            r.materialize(), f(ref), rtol=1e-12, atol=1e-15,
+            err_msg=f.__name__)
+    r = np.round(y * 10)
+    np.testing.assert_array_equal(r.materialize(), np.round(ya * 10))
