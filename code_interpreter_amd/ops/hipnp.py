@@ -1531,6 +1531,20 @@ class DeviceArray:
                         h, (rows, cols - 1), a.dtype
                     )._device_transposed()
             return _np.diff(a.materialize(), **kwargs)
+        _nan_kinds = {
+            _np.nansum: "sum", _np.nanmean: "mean", _np.nanmax: "max",
+            _np.nanmin: "min", _np.nanstd: "std", _np.nanvar: "var",
+        }
+        if func in _nan_kinds and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"ddof"}:
+            kind = _nan_kinds[func]
+            ddof = kwargs.get("ddof", 0) if kind in ("std", "var") else 0
+            if not kwargs or kind in ("std", "var"):
+                r = nan_reduce_device(args[0], kind, ddof)
+                if r is not None:
+                    return r
+            return func(args[0].materialize(), **kwargs)
         if func in (_np.cov, _np.corrcoef) and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and not kwargs:
@@ -1867,6 +1881,54 @@ def quantile_rows_device(x, q):
     top = col(cols - 1)  # NaNs sort last: top column flags NaN rows
     out[_np.isnan(top)] = _np.nan
     return out
+
+
+def nan_reduce_device(x, kind, ddof=0):
+    """Flat NaN-ignoring reductions (np.nansum/nanmean/nanmax/nanmin/
+    nanstd/nanvar) composed from the device mask ops: isnan mask ->
+    popcount -> where-replace -> fused reduce. Returns None when not
+    routable (caller falls back to host numpy, including the all-NaN
+    warning cases)."""
+    if not isinstance(x, DeviceArray) or _dtype_code(x.dtype) is None:
+        return None
+    mask = x.isnan()
+    if not isinstance(mask, BoolDeviceArray):
+        return None
+    n_nan = int(mask.sum())
+    n_valid = x.size - n_nan
+    if kind in ("max", "min"):
+        if n_valid == 0:
+            return None  # numpy warns and returns nan
+        if n_nan == 0:
+            return x.max() if kind == "max" else x.min()
+        repl = -_np.inf if kind == "max" else _np.inf
+        cleaned = where_device(mask, repl, x)
+        if cleaned is NotImplemented:
+            return None
+        return cleaned.max() if kind == "max" else cleaned.min()
+    cleaned = x if n_nan == 0 else where_device(mask, 0.0, x)
+    if cleaned is NotImplemented:
+        return None
+    total = float(cleaned.sum())
+    if kind == "sum":
+        return x.dtype.type(total)
+    if n_valid == 0:
+        return None
+    mu = total / n_valid
+    if kind == "mean":
+        return x.dtype.type(mu)
+    centered = x._binary("subtract", mu)
+    if centered is NotImplemented:
+        return None
+    cc = centered if n_nan == 0 else where_device(mask, 0.0, centered)
+    if cc is NotImplemented:
+        return None
+    if n_valid - ddof <= 0:
+        return None
+    var = float(cc.square_sum()) / (n_valid - ddof)
+    if kind == "var":
+        return x.dtype.type(var)
+    return x.dtype.type(var ** 0.5)  # std
 
 
 def cov_device(m):

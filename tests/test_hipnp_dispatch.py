@@ -1165,3 +1165,32 @@ def test_widened_unary_set(fake):
     # round with decimals falls back to host values
     np.testing.assert_allclose(
         np.round(_device(fake, host), 2), np.round(host, 2), rtol=0)
+
+
+def test_nan_reductions_on_device(fake):
+    host = np.random.default_rng(37).random(5000)
+    host[np.random.default_rng(38).integers(0, 5000, 200)] = np.nan
+    x = _device(fake, host)
+    assert float(np.nansum(x)) == pytest.approx(np.nansum(host), rel=1e-12)
+    assert float(np.nanmean(_device(fake, host))) == pytest.approx(
+        np.nanmean(host), rel=1e-12)
+    assert float(np.nanmax(_device(fake, host))) == pytest.approx(
+        np.nanmax(host), rel=1e-12)
+    assert float(np.nanmin(_device(fake, host))) == pytest.approx(
+        np.nanmin(host), rel=1e-12)
+    assert float(np.nanstd(_device(fake, host))) == pytest.approx(
+        np.nanstd(host), rel=1e-10)
+    assert float(np.nanvar(_device(fake, host), ddof=1)) == pytest.approx(
+        np.nanvar(host, ddof=1), rel=1e-10)
+    # no-NaN input takes the plain fused path
+    clean = np.random.default_rng(39).random(3000)
+    assert float(np.nansum(_device(fake, clean))) == pytest.approx(
+        clean.sum(), rel=1e-12)
+    # all-NaN falls back to host numpy (warning + nan)
+    alln = np.full(100, np.nan)
+    with np.errstate(all="ignore"):
+        import warnings
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            r = np.nanmax(_device(fake, alln))
+    assert np.isnan(r)

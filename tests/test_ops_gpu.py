@@ -1049,3 +1049,18 @@ def test_widened_unary_set_gpu(hnp):
             err_msg=f.__name__)
     r = np.round(y * 10)
     np.testing.assert_array_equal(r.materialize(), np.round(ya * 10))
+
+
+def test_nan_reductions_gpu(hnp):
+    x = hnp.rand(3_000_000, seed=94)
+    a = np.asarray(x).copy()
+    a[::1000] = np.nan
+    xh = hnp.DeviceArray(hnp.backend().upload(a), a.shape, a.dtype)
+    np.testing.assert_allclose(
+        float(np.nansum(xh)), np.nansum(a), rtol=1e-10)
+    np.testing.assert_allclose(
+        float(np.nanmean(xh)), np.nanmean(a), rtol=1e-10)
+    np.testing.assert_allclose(
+        float(np.nanmax(xh)), np.nanmax(a), rtol=1e-12)
+    np.testing.assert_allclose(
+        float(np.nanstd(xh)), np.nanstd(a), rtol=1e-8)
