@@ -80,6 +80,38 @@ class FakeHipops(types.ModuleType):
     def gemm_batched(self, ha, hb, batch, m, n, k, dtype):
         return self._new(b"\0" * (batch * m * n * 8))
 
+    def sort(self, h, dtype, n, want_idx):
+        if want_idx:
+            return self._new(self.bufs[h]), self._new(b"\0" * (n * 8))
+        return self._new(self.bufs[h])
+
+    def sort2d(self, h, dtype, rows, cols, want_idx):
+        if want_idx:
+            return (self._new(self.bufs[h]),
+                    self._new(b"\0" * (rows * cols * 8)))
+        return self._new(self.bufs[h])
+
+    def transpose(self, h, dtype, rows, cols):
+        return self._new(self.bufs[h])
+
+    def cumsum2d(self, h, dtype, rows, cols):
+        return self._new(self.bufs[h])
+
+    def diff(self, h, dtype, outer, inner):
+        return self._new(self.bufs[h])
+
+    def searchsorted(self, ha, n, hv, m, dtype, right):
+        return self._new(b"\0" * (m * 8))
+
+    def download_slice(self, h, off, nbytes):
+        return self.bufs[h][off:off + nbytes]
+
+    def download_strided(self, h, off, stride, esz, count):
+        return b"\0" * (esz * count)
+
+    def histogram(self, h, dtype, n, lo, hi, bins, exact=0):
+        return b"\0" * ((bins + 3) * 8)
+
     def synchronize(self):
         pass
 
@@ -146,6 +178,19 @@ def test_cross_connection_handle_access_denied(hipd):
             {"op": "convert", "h": h, "src": 1, "dst": 0, "n": 16},
             {"op": "gemm", "ha": h, "hb": h, "m": 4, "n": 4, "k": 4,
              "dtype": 1},
+            {"op": "sort", "h": h, "dtype": 1, "n": 16, "want_idx": 1},
+            {"op": "sort2d", "h": h, "dtype": 1, "rows": 4, "cols": 4,
+             "want_idx": 0},
+            {"op": "transpose", "h": h, "dtype": 1, "rows": 4, "cols": 4},
+            {"op": "cumsum2d", "h": h, "dtype": 1, "rows": 4, "cols": 4},
+            {"op": "diff", "h": h, "dtype": 1, "outer": 1, "inner": 16},
+            {"op": "searchsorted", "ha": h, "n": 16, "hv": h, "m": 4,
+             "dtype": 1, "right": 0},
+            {"op": "download_slice", "h": h, "off": 0, "nbytes": 8},
+            {"op": "download_strided", "h": h, "off": 0, "stride": 16,
+             "esz": 8, "count": 2},
+            {"op": "histogram", "h": h, "dtype": 1, "n": 16, "lo": 0.0,
+             "hi": 1.0, "bins": 8, "exact": 1},
         ]
         for msg in denied:
             resp, _ = b.call(msg)
