@@ -1194,3 +1194,28 @@ def test_nan_reductions_on_device(fake):
             warnings.simplefilter("ignore")
             r = np.nanmax(_device(fake, alln))
     assert np.isnan(r)
+
+
+def test_reshape_ravel_on_device(fake):
+    host = np.random.default_rng(40).random(1200)
+    x = _device(fake, host)
+    r = x.reshape(30, 40)
+    assert isinstance(r, hipnp.DeviceArray) and r.shape == (30, 40)
+    assert "download" not in fake.calls
+    np.testing.assert_array_equal(r.materialize(), host.reshape(30, 40))
+    r2 = np.reshape(_device(fake, host), (40, -1))
+    assert isinstance(r2, hipnp.DeviceArray) and r2.shape == (40, 30)
+    np.testing.assert_array_equal(r2.materialize(), host.reshape(40, 30))
+    m = _device(fake, host.reshape(30, 40))
+    f = m.ravel()
+    assert isinstance(f, hipnp.DeviceArray) and f.shape == (1200,)
+    np.testing.assert_array_equal(f.materialize(), host)
+    fl = np.ravel(_device(fake, host.reshape(30, 40)))
+    assert isinstance(fl, hipnp.DeviceArray)
+    # bad shape falls back to numpy's own error
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        _device(fake, host).reshape(7, 7)
+    # order='F' falls back to host semantics
+    rf = _device(fake, host.reshape(30, 40)).reshape(40, 30, order="F")
+    np.testing.assert_array_equal(rf, host.reshape(30, 40).reshape(40, 30, order="F"))

@@ -1064,3 +1064,19 @@ def test_nan_reductions_gpu(hnp):
         float(np.nanmax(xh)), np.nanmax(a), rtol=1e-12)
     np.testing.assert_allclose(
         float(np.nanstd(xh)), np.nanstd(a), rtol=1e-8)
+
+
+def test_reshape_ravel_gpu(hnp):
+    x = hnp.rand(1200, 1000, seed=95)
+    a = np.asarray(x).copy()
+    f = hnp.rand(1200, 1000, seed=95).ravel()
+    assert isinstance(f, hnp.DeviceArray) and f.shape == (1_200_000,)
+    np.testing.assert_array_equal(f.materialize(), a.ravel())
+    r = hnp.rand(1_000_000, seed=96)
+    ra = np.asarray(r).copy()
+    m = r.reshape(1000, -1)
+    assert isinstance(m, hnp.DeviceArray) and m.shape == (1000, 1000)
+    # stays routable: mean over the new axes on device
+    np.testing.assert_allclose(
+        np.asarray(m.mean(axis=0)), ra.reshape(1000, 1000).mean(axis=0),
+        rtol=1e-12)
