@@ -1376,6 +1376,87 @@ class DeviceArray:
         if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
             if set(kwargs) <= {"axis", "keepdims"}:
                 return args[0].mean(**kwargs)
+        if func is _np.trace and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and not kwargs:
+            a = args[0]
+            if (
+                len(a.shape) == 2
+                and _dtype_code(a.dtype) is not None
+                and a._host is None
+            ):
+                n = min(a.shape)
+                esz = a.dtype.itemsize
+                raw = backend().download_strided(
+                    a._dev_handle(), 0, (a.shape[1] + 1) * esz, esz, n
+                )
+                return a.dtype.type(
+                    _np.frombuffer(raw, dtype=a.dtype).sum()
+                )
+            return _np.trace(a.materialize())
+        if func is _np.outer and len(args) == 2 and not kwargs:
+            x, y = args
+            if (
+                isinstance(x, DeviceArray)
+                and isinstance(y, DeviceArray)
+                and len(x.shape) == 1
+                and len(y.shape) == 1
+                and x.dtype == y.dtype
+                and _dtype_code(x.dtype) is not None
+            ):
+                h = backend().gemm(
+                    x._dev_handle(), y._dev_handle(), x.size, y.size, 1,
+                    _dtype_code(x.dtype),
+                )
+                return DeviceArray(h, (x.size, y.size), x.dtype)
+            host = [
+                v.materialize() if isinstance(v, DeviceArray) else v
+                for v in args
+            ]
+            return _np.outer(*host)
+        if func is _np.einsum and len(args) >= 2 and isinstance(
+            args[0], str
+        ) and not kwargs:
+            # the common contractions, mapped onto existing device ops;
+            # anything else materializes below
+            sub = args[0].replace(" ", "")
+            ops_ = args[1:]
+            try:
+                if sub in ("ij,jk->ik", "ij,jk") and len(ops_) == 2:
+                    r = matmul(ops_[0], ops_[1], _force=True)
+                    if r is not NotImplemented:
+                        return r
+                elif sub in ("bij,bjk->bik", "bij,bjk") and len(ops_) == 2:
+                    r = matmul(ops_[0], ops_[1], _force=True)
+                    if r is not NotImplemented:
+                        return r
+                elif sub in ("i,i->", "i,i") and len(ops_) == 2:
+                    r = matmul(ops_[0], ops_[1], _force=True)
+                    if r is not NotImplemented:
+                        return r
+                elif sub in ("ij->ji",) and len(ops_) == 1 and isinstance(
+                    ops_[0], DeviceArray
+                ):
+                    return ops_[0].transpose()
+                elif sub in ("ij->", "i->") and len(ops_) == 1 and isinstance(
+                    ops_[0], DeviceArray
+                ):
+                    return ops_[0].sum()
+                elif sub == "ii" and len(ops_) == 1:
+                    return DeviceArray.__array_function__(
+                        ops_[0], _np.trace, (DeviceArray,), (ops_[0],), {}
+                    )
+                elif sub in ("i,j->ij", "i,j") and len(ops_) == 2:
+                    return DeviceArray.__array_function__(
+                        self, _np.outer, (DeviceArray,), tuple(ops_), {}
+                    )
+            except Exception:
+                pass
+            host = [
+                v.materialize() if isinstance(v, DeviceArray) else v
+                for v in ops_
+            ]
+            return _np.einsum(sub, *host)
         if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
             r = matmul(*args, _force=True)
             if r is not NotImplemented:

@@ -1219,3 +1219,35 @@ def test_reshape_ravel_on_device(fake):
     # order='F' falls back to host semantics
     rf = _device(fake, host.reshape(30, 40)).reshape(40, 30, order="F")
     np.testing.assert_array_equal(rf, host.reshape(30, 40).reshape(40, 30, order="F"))
+
+
+def test_einsum_outer_trace_on_device(fake):
+    a2 = np.random.default_rng(41).random((64, 80))
+    b2 = np.random.default_rng(42).random((80, 48))
+    x, y = _device(fake, a2), _device(fake, b2)
+    r = np.einsum("ij,jk->ik", x, y)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(r.materialize(), a2 @ b2, rtol=1e-12)
+    # transpose / full-sum / trace
+    t = np.einsum("ij->ji", _device(fake, a2))
+    assert isinstance(t, hipnp.DeviceArray)
+    np.testing.assert_array_equal(t.materialize(), a2.T)
+    assert float(np.einsum("ij->", _device(fake, a2))) == pytest.approx(
+        a2.sum(), rel=1e-12)
+    sq = np.random.default_rng(43).random((50, 50))
+    assert float(np.trace(_device(fake, sq))) == pytest.approx(
+        np.trace(sq), rel=1e-12)
+    assert float(np.einsum("ii", _device(fake, sq))) == pytest.approx(
+        np.trace(sq), rel=1e-12)
+    # outer
+    v = np.random.default_rng(44).random(70)
+    w = np.random.default_rng(45).random(90)
+    o = np.outer(_device(fake, v), _device(fake, w))
+    assert isinstance(o, hipnp.DeviceArray) and o.shape == (70, 90)
+    np.testing.assert_allclose(o.materialize(), np.outer(v, w), rtol=1e-12)
+    o2 = np.einsum("i,j->ij", _device(fake, v), _device(fake, w))
+    np.testing.assert_allclose(
+        np.asarray(o2), np.outer(v, w), rtol=1e-12)
+    # unsupported pattern falls back with identical values
+    r3 = np.einsum("ij,ij->i", _device(fake, a2), _device(fake, a2))
+    np.testing.assert_allclose(r3, np.einsum("ij,ij->i", a2, a2), rtol=1e-12)

@@ -1080,3 +1080,21 @@ def test_reshape_ravel_gpu(hnp):
     np.testing.assert_allclose(
         np.asarray(m.mean(axis=0)), ra.reshape(1000, 1000).mean(axis=0),
         rtol=1e-12)
+
+
+def test_einsum_outer_trace_gpu(hnp):
+    x = hnp.rand(300, 400, seed=97)
+    y = hnp.rand(400, 200, seed=98)
+    xa, ya = np.asarray(x).copy(), np.asarray(y).copy()
+    r = np.einsum("ij,jk->ik", x, y)
+    assert isinstance(r, hnp.DeviceArray)
+    np.testing.assert_allclose(r.materialize(), xa @ ya, rtol=1e-12)
+    v = hnp.rand(3000, seed=99)
+    w = hnp.rand(2000, seed=100)
+    o = np.outer(v, w)
+    assert isinstance(o, hnp.DeviceArray)
+    np.testing.assert_allclose(
+        o.materialize(), np.outer(np.asarray(v), np.asarray(w)), rtol=1e-12)
+    sq = hnp.rand(800, 800, seed=101)
+    np.testing.assert_allclose(
+        float(np.trace(sq)), np.trace(np.asarray(sq)), rtol=1e-10)
