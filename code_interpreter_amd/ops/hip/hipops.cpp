@@ -794,6 +794,29 @@ PyObject* py_cumsum2d(PyObject*, PyObject* args) {
   WRAP_END
 }
 
+// copy_d2d(dst, dst_off, src, src_off, nbytes): device-to-device copy
+// window (np.concatenate/stack building blocks)
+PyObject* py_copy_d2d(PyObject*, PyObject* args) {
+  unsigned long long hd, hs;
+  long long doff, soff, nbytes;
+  if (!PyArg_ParseTuple(args, "KLKLL", &hd, &doff, &hs, &soff, &nbytes))
+    return nullptr;
+  WRAP_BEGIN
+  ensure_init();
+  DevBuf& dst = get_buf(hd);
+  DevBuf& src = get_buf(hs);
+  if (doff < 0 || soff < 0 || nbytes < 0 || doff + nbytes > dst.size ||
+      soff + nbytes > src.size)
+    throw std::runtime_error("copy_d2d out of range");
+  NOGIL_BEGIN
+  HIP_CHECK(hipMemcpyAsync((char*)dst.ptr + doff, (char*)src.ptr + soff,
+                           (size_t)nbytes, hipMemcpyDeviceToDevice,
+                           g.compute));
+  NOGIL_END
+  Py_RETURN_NONE;
+  WRAP_END
+}
+
 // transpose(h, dtype, rows, cols) -> handle ([cols][rows] result)
 PyObject* py_transpose(PyObject*, PyObject* args) {
   unsigned long long h;
@@ -1317,6 +1340,8 @@ PyMethodDef methods[] = {
      "diff(h, dtype, outer, inner) -> handle"},
     {"cumsum2d", py_cumsum2d, METH_VARARGS,
      "cumsum2d(h, dtype, rows, cols) -> handle (row-wise scan)"},
+    {"copy_d2d", py_copy_d2d, METH_VARARGS,
+     "copy_d2d(dst, dst_off, src, src_off, nbytes)"},
     {"transpose", py_transpose, METH_VARARGS,
      "transpose(h, dtype, rows, cols) -> handle"},
     {"sort2d", py_sort2d, METH_VARARGS,

@@ -217,6 +217,12 @@ class Connection(threading.Thread):
             )
             self.handles.add(h)
             return {"ok": True, "h": h}, b""
+        if op == "copy_d2d":
+            _hipops.copy_d2d(
+                self._own(m["hd"]), m["doff"], self._own(m["hs"]),
+                m["soff"], m["nbytes"]
+            )
+            return {"ok": True}, b""
         if op == "transpose":
             h = _hipops.transpose(
                 self._own(m["h"]), m["dtype"], m["rows"], m["cols"]

@@ -358,6 +358,12 @@ class RemoteBackend:
              "cols": cols}
         )["h"]
 
+    def copy_d2d(self, hd, doff, hs, soff, nbytes):
+        self._call(
+            {"op": "copy_d2d", "hd": hd, "doff": doff, "hs": hs,
+             "soff": soff, "nbytes": nbytes}
+        )
+
     def transpose(self, h, dtype, rows, cols):
         return self._call(
             {"op": "transpose", "h": h, "dtype": dtype, "rows": rows,
@@ -1716,6 +1722,62 @@ class DeviceArray:
         ) and set(kwargs) <= {"decimals"}:
             dec = args[1] if len(args) > 1 else kwargs.get("decimals", 0)
             return args[0].round(dec)
+        if func in (_np.concatenate, _np.vstack, _np.hstack, _np.stack) \
+                and len(args) == 1 and isinstance(args[0], (list, tuple)):
+            axis = kwargs.get("axis", 0)
+            parts = args[0]
+            routable = (
+                set(kwargs) <= {"axis"}
+                and axis in (0, None)
+                and len(parts) >= 1
+                and all(isinstance(p, DeviceArray) for p in parts)
+                and len({p.dtype for p in parts}) == 1
+                and _dtype_code(parts[0].dtype) is not None
+                and all(p._host is None for p in parts)
+            )
+            if routable:
+                shapes = [p.shape for p in parts]
+                nd = len(shapes[0])
+                if func is _np.concatenate and axis is None:
+                    out_shape = (sum(p.size for p in parts),)
+                elif func in (_np.concatenate, _np.vstack):
+                    base = shapes[0][1:] if nd > 0 else ()
+                    if func is _np.vstack and nd == 1:
+                        # vstack of 1-D rows -> (k, n)
+                        if len({sh for sh in shapes}) == 1:
+                            out_shape = (len(parts), shapes[0][0])
+                        else:
+                            out_shape = None
+                    elif all(sh[1:] == base for sh in shapes):
+                        out_shape = (sum(sh[0] for sh in shapes),) + base
+                    else:
+                        out_shape = None
+                elif func is _np.hstack and nd == 1:
+                    out_shape = (sum(sh[0] for sh in shapes),)
+                elif func is _np.stack and len(set(shapes)) == 1:
+                    out_shape = (len(parts),) + shapes[0]
+                else:
+                    out_shape = None
+                if out_shape is not None:
+                    esz = parts[0].dtype.itemsize
+                    total = 1
+                    for d in out_shape:
+                        total *= d
+                    hd = backend().alloc(total * esz)
+                    out = DeviceArray(hd, out_shape, parts[0].dtype)
+                    off = 0
+                    for part in parts:
+                        backend().copy_d2d(
+                            hd, off, part._dev_handle(), 0,
+                            part.size * esz,
+                        )
+                        off += part.size * esz
+                    return out
+            host = [
+                p.materialize() if isinstance(p, DeviceArray) else p
+                for p in parts
+            ]
+            return func(host, **kwargs)
         if func in (_np.reshape, _np.ravel) and len(args) >= 1 and isinstance(
             args[0], DeviceArray
         ) and not kwargs:

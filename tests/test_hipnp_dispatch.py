@@ -109,6 +109,16 @@ class FakeBackend:
         a = self.bufs[h].view(self._dt(dtype)).reshape(-1)[:n]
         return self._new(a.astype(np.float64).cumsum().astype(self._dt(dtype)))
 
+    def alloc(self, nbytes):
+        self.calls.append("alloc")
+        return self._new(np.zeros(nbytes, dtype=np.uint8))
+
+    def copy_d2d(self, hd, doff, hs, soff, nbytes):
+        self.calls.append("copy_d2d")
+        dst = self.bufs[hd].view(np.uint8).reshape(-1)
+        src = self.bufs[hs].view(np.uint8).reshape(-1)
+        dst[doff:doff + nbytes] = src[soff:soff + nbytes]
+
     def download_slice(self, h, off, nbytes):
         self.calls.append("download_slice")
         raw = self.bufs[h].view(np.uint8).reshape(-1).tobytes()
@@ -1251,3 +1261,31 @@ def test_einsum_outer_trace_on_device(fake):
     # unsupported pattern falls back with identical values
     r3 = np.einsum("ij,ij->i", _device(fake, a2), _device(fake, a2))
     np.testing.assert_allclose(r3, np.einsum("ij,ij->i", a2, a2), rtol=1e-12)
+
+
+def test_concatenate_family_on_device(fake):
+    a = np.random.default_rng(46).random(500)
+    b = np.random.default_rng(47).random(300)
+    x, y = _device(fake, a), _device(fake, b)
+    c = np.concatenate([x, y])
+    assert isinstance(c, hipnp.DeviceArray) and c.shape == (800,)
+    assert "copy_d2d" in fake.calls and "download" not in fake.calls
+    np.testing.assert_array_equal(c.materialize(), np.concatenate([a, b]))
+    h = np.hstack([_device(fake, a), _device(fake, b)])
+    assert isinstance(h, hipnp.DeviceArray)
+    np.testing.assert_array_equal(h.materialize(), np.hstack([a, b]))
+    m1 = np.random.default_rng(48).random((20, 40))
+    m2 = np.random.default_rng(49).random((30, 40))
+    v = np.vstack([_device(fake, m1), _device(fake, m2)])
+    assert isinstance(v, hipnp.DeviceArray) and v.shape == (50, 40)
+    np.testing.assert_array_equal(v.materialize(), np.vstack([m1, m2]))
+    st = np.stack([_device(fake, m1), _device(fake, m1)])
+    assert isinstance(st, hipnp.DeviceArray) and st.shape == (2, 20, 40)
+    np.testing.assert_array_equal(st.materialize(), np.stack([m1, m1]))
+    # axis=1 falls back to host numpy
+    c1 = np.concatenate([_device(fake, m1), _device(fake, m1)], axis=1)
+    assert isinstance(c1, np.ndarray)
+    np.testing.assert_array_equal(c1, np.concatenate([m1, m1], axis=1))
+    # mixed host/device falls back with identical values
+    c2 = np.concatenate([_device(fake, a), b])
+    np.testing.assert_array_equal(np.asarray(c2), np.concatenate([a, b]))
