@@ -1068,9 +1068,9 @@ def test_nan_reductions_gpu(hnp):
 
 def test_reshape_ravel_gpu(hnp):
     x = hnp.rand(1200, 1000, seed=95)
-    a = np.asarray(x).copy()
-    f = hnp.rand(1200, 1000, seed=95).ravel()
+    f = x.ravel()  # before any materialize: device path
     assert isinstance(f, hnp.DeviceArray) and f.shape == (1_200_000,)
+    a = np.asarray(x).copy()
     np.testing.assert_array_equal(f.materialize(), a.ravel())
     r = hnp.rand(1_000_000, seed=96)
     ra = np.asarray(r).copy()
