@@ -1073,9 +1073,9 @@ def test_reshape_ravel_gpu(hnp):
     a = np.asarray(x).copy()
     np.testing.assert_array_equal(f.materialize(), a.ravel())
     r = hnp.rand(1_000_000, seed=96)
-    ra = np.asarray(r).copy()
-    m = r.reshape(1000, -1)
+    m = r.reshape(1000, -1)  # before materialize: device path
     assert isinstance(m, hnp.DeviceArray) and m.shape == (1000, 1000)
+    ra = np.asarray(r).copy()
     # stays routable: mean over the new axes on device
     np.testing.assert_allclose(
         np.asarray(m.mean(axis=0)), ra.reshape(1000, 1000).mean(axis=0),
