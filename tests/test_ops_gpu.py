@@ -1103,16 +1103,17 @@ def test_einsum_outer_trace_gpu(hnp):
 def test_concatenate_family_gpu(hnp):
     a = hnp.rand(2_000_000, seed=102)
     b = hnp.rand(1_000_000, seed=103)
+    c = np.concatenate([a, b])  # before materialize: device path
     aa, ba = np.asarray(a).copy(), np.asarray(b).copy()
-    c = np.concatenate([a, b])
     assert isinstance(c, hnp.DeviceArray) and c.shape == (3_000_000,)
     np.testing.assert_array_equal(c.materialize(), np.concatenate([aa, ba]))
     m1 = hnp.rand(500, 1000, seed=104)
     m2 = hnp.rand(700, 1000, seed=105)
     v = np.vstack([m1, m2])
     assert isinstance(v, hnp.DeviceArray) and v.shape == (1200, 1000)
+    vm = v.materialize()
     np.testing.assert_array_equal(
-        v.materialize(), np.vstack([np.asarray(m1), np.asarray(m2)]))
+        vm, np.vstack([np.asarray(m1), np.asarray(m2)]))
     # concatenated result keeps computing on device
     np.testing.assert_allclose(
         float(c.sum()), aa.sum() + ba.sum(), rtol=1e-10)
