@@ -1492,6 +1492,14 @@ class DeviceArray:
                 r = quantile_device(args[0], q)
                 if r is not None:
                     return r
+            elif isinstance(qv, (list, tuple, _np.ndarray)):
+                qarr = _np.asarray(qv, dtype=_np.float64).reshape(-1)
+                if qarr.size and _np.all((qarr >= 0) & (qarr <= 100)):
+                    qs = qarr / 100.0 if func is _np.percentile else qarr
+                    if _np.all(qs <= 1.0):
+                        r = quantile_list_device(args[0], qs)
+                        if r is not None:
+                            return r.reshape(_np.asarray(qv).shape)
         if func in (_np.quantile, _np.percentile) and len(args) == 2 and set(
             kwargs
         ) == {"axis"}:
@@ -2177,6 +2185,47 @@ def corrcoef_device(m):
     if r is NotImplemented:
         return None
     return r.clip(-1.0, 1.0)
+
+
+def quantile_list_device(x, qs):
+    """np.quantile/percentile with an ARRAY of quantiles (axis=None):
+    ONE device sort, then a 16-byte download_slice per interpolation
+    endpoint — exact numpy-linear values at any number of q's."""
+    if not isinstance(x, DeviceArray) or _dtype_code(x.dtype) is None:
+        return None
+    if x.size < 1:
+        return None
+    if len(x.shape) == 1:
+        flat = x
+    else:
+        flat = x.ravel()  # device copy with its own handle
+        if not isinstance(flat, DeviceArray):
+            return None
+    srt = flat._device_sorted(False)
+    esz = x.dtype.itemsize
+    n = x.size
+    # NaNs sort last: any NaN makes every quantile NaN (numpy parity)
+    top = _np.frombuffer(
+        backend().download_slice(srt._dev_handle(), (n - 1) * esz, esz),
+        dtype=x.dtype,
+    )[0]
+    out = _np.empty(len(qs), dtype=x.dtype)
+    if _np.isnan(top):
+        out[:] = _np.nan
+        return out
+    for i, q in enumerate(qs):
+        pos = float(q) * (n - 1)
+        k0 = int(_np.floor(pos))
+        k1 = min(k0 + 1, n - 1)
+        frac = pos - k0
+        raw = backend().download_slice(
+            srt._dev_handle(), k0 * esz, (k1 - k0 + 1) * esz
+        )
+        vals = _np.frombuffer(raw, dtype=x.dtype)
+        v0 = vals[0]
+        v1 = vals[-1]
+        out[i] = v0 if frac == 0 else v0 + (v1 - v0) * x.dtype.type(frac)
+    return out
 
 
 def quantile_cols_device(x, q):

@@ -1289,3 +1289,24 @@ def test_concatenate_family_on_device(fake):
     # mixed host/device falls back with identical values
     c2 = np.concatenate([_device(fake, a), b])
     np.testing.assert_array_equal(np.asarray(c2), np.concatenate([a, b]))
+
+
+def test_quantile_array_q_on_device(fake):
+    host = np.random.default_rng(50).random(8000)
+    x = _device(fake, host)
+    qs = [0.0, 0.1, 0.5, 0.9, 1.0]
+    r = np.quantile(x, qs)
+    np.testing.assert_allclose(r, np.quantile(host, qs), rtol=1e-12)
+    assert "sort" in fake.calls
+    r2 = np.percentile(_device(fake, host), [5, 25, 75, 95])
+    np.testing.assert_allclose(
+        r2, np.percentile(host, [5, 25, 75, 95]), rtol=1e-12)
+    # 2-D input flattens like numpy
+    m = host.reshape(80, 100)
+    r3 = np.quantile(_device(fake, m), [0.25, 0.75])
+    np.testing.assert_allclose(r3, np.quantile(m, [0.25, 0.75]), rtol=1e-12)
+    # NaN poisons every quantile (numpy parity)
+    h2 = host.copy()
+    h2[17] = np.nan
+    r4 = np.quantile(_device(fake, h2), [0.5, 0.9])
+    assert np.isnan(r4).all()

@@ -1117,3 +1117,11 @@ def test_concatenate_family_gpu(hnp):
     # concatenated result keeps computing on device
     np.testing.assert_allclose(
         float(c.sum()), aa.sum() + ba.sum(), rtol=1e-10)
+
+
+def test_quantile_array_q_gpu(hnp):
+    x = hnp.rand(4_000_000, seed=106)
+    qs = [0.0, 0.25, 0.5, 0.75, 0.99, 1.0]
+    r = np.quantile(x, qs)
+    a = np.asarray(x)
+    np.testing.assert_allclose(r, np.quantile(a, qs), rtol=1e-12)
