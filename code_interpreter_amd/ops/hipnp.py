@@ -1675,6 +1675,69 @@ class DeviceArray:
                         h, (rows, cols - 1), a.dtype
                     )._device_transposed()
             return _np.diff(a.materialize(), **kwargs)
+        if func is _np.ptp and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and not kwargs:
+            a = args[0]
+            if _dtype_code(a.dtype) is not None:
+                return a.dtype.type(float(a.max()) - float(a.min()))
+            return _np.ptp(a.materialize())
+        if func is _np.average and len(args) == 1 and set(kwargs) <= {
+            "weights"
+        }:
+            a = args[0]
+            w = kwargs.get("weights")
+            if (
+                isinstance(a, DeviceArray)
+                and _dtype_code(a.dtype) is not None
+            ):
+                if w is None:
+                    return a.dtype.type(float(a.sum()) / a.size)
+                if (
+                    isinstance(w, DeviceArray)
+                    and w.shape == a.shape
+                    and w.dtype == a.dtype
+                ):
+                    num = a._binary("multiply", w)
+                    if num is not NotImplemented:
+                        den = float(w.sum())
+                        if den != 0:
+                            return a.dtype.type(float(num.sum()) / den)
+            host_w = w.materialize() if isinstance(w, DeviceArray) else w
+            host_a = a.materialize() if isinstance(a, DeviceArray) else a
+            return _np.average(host_a, weights=host_w)
+        if func in (_np.isclose, _np.allclose) and len(args) == 2 and set(
+            kwargs
+        ) <= {"rtol", "atol", "equal_nan"} and not kwargs.get("equal_nan"):
+            a, b = args
+            rtol = float(kwargs.get("rtol", 1e-05))
+            atol = float(kwargs.get("atol", 1e-08))
+            if (
+                isinstance(a, DeviceArray)
+                and isinstance(b, DeviceArray)
+                and a.shape == b.shape
+                and a.dtype == b.dtype
+                and _dtype_code(a.dtype) is not None
+            ):
+                # |a-b| <= atol + rtol*|b|, NaN comparisons are False
+                diff = a._binary("subtract", b)
+                if diff is not NotImplemented:
+                    tol = b._unary("absolute")._binary("multiply", rtol)
+                    if tol is not NotImplemented:
+                        tol = tol._binary("add", atol)
+                    if tol is not NotImplemented:
+                        mask = diff._unary("absolute")._compare(
+                            "less_equal", tol
+                        )
+                        if mask is not None:
+                            if func is _np.allclose:
+                                return bool(mask.all())
+                            return mask
+            host = [
+                v.materialize() if isinstance(v, DeviceArray) else v
+                for v in args
+            ]
+            return func(*host, **kwargs)
         _nan_kinds = {
             _np.nansum: "sum", _np.nanmean: "mean", _np.nanmax: "max",
             _np.nanmin: "min", _np.nanstd: "std", _np.nanvar: "var",

@@ -1310,3 +1310,28 @@ def test_quantile_array_q_on_device(fake):
     h2[17] = np.nan
     r4 = np.quantile(_device(fake, h2), [0.5, 0.9])
     assert np.isnan(r4).all()
+
+
+def test_ptp_average_isclose_on_device(fake):
+    a = np.random.default_rng(51).random(4000)
+    w = np.random.default_rng(52).random(4000)
+    x, xw = _device(fake, a), _device(fake, w)
+    assert float(np.ptp(x)) == pytest.approx(np.ptp(a), rel=1e-12)
+    assert float(np.average(_device(fake, a))) == pytest.approx(
+        np.average(a), rel=1e-12)
+    assert float(np.average(_device(fake, a), weights=xw)) == pytest.approx(
+        np.average(a, weights=w), rel=1e-12)
+    b = a + np.random.default_rng(53).normal(0, 1e-9, 4000)
+    y = _device(fake, b)
+    m = np.isclose(_device(fake, a), y)
+    assert isinstance(m, hipnp.BoolDeviceArray)
+    np.testing.assert_array_equal(np.asarray(m), np.isclose(a, b))
+    assert bool(np.allclose(_device(fake, a), _device(fake, b))) == bool(
+        np.allclose(a, b))
+    far = _device(fake, a + 1.0)
+    assert not np.allclose(_device(fake, a), far)
+    # NaN never close (equal_nan=False default)
+    an = a.copy(); an[5] = np.nan
+    bn = a.copy(); bn[5] = np.nan
+    m2 = np.isclose(_device(fake, an), _device(fake, bn))
+    assert not np.asarray(m2)[5]

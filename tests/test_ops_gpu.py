@@ -1125,3 +1125,15 @@ def test_quantile_array_q_gpu(hnp):
     r = np.quantile(x, qs)
     a = np.asarray(x)
     np.testing.assert_allclose(r, np.quantile(a, qs), rtol=1e-12)
+
+
+def test_ptp_average_isclose_gpu(hnp):
+    x = hnp.rand(2_000_000, seed=107)
+    w = hnp.rand(2_000_000, seed=108)
+    r_ptp = float(np.ptp(x))
+    r_avg = float(np.average(x, weights=w))
+    close_same = bool(np.allclose(x, x))
+    a, wa = np.asarray(x), np.asarray(w)
+    assert r_ptp == pytest.approx(np.ptp(a), rel=1e-12)
+    assert r_avg == pytest.approx(np.average(a, weights=wa), rel=1e-10)
+    assert close_same
