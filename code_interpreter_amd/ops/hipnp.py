@@ -1012,7 +1012,7 @@ class DeviceArray:
         accumulation in double. Rounding may differ from numpy's
         strictly-sequential order by ~1 ulp at block regroupings."""
         code = _dtype_code(self.dtype)
-        if code is not None and not kwargs:
+        if code is not None and not kwargs and self.size > 0:
             if axis is None or (
                 len(self.shape) == 1 and self._norm_axis(axis, 1) == 0
             ):
@@ -1110,6 +1110,7 @@ class DeviceArray:
         stable -- every numpy `kind` is satisfied."""
         if (
             _dtype_code(self.dtype) is None
+            or self.size == 0
             or kwargs
             or order is not None
             or kind not in (None, "stable", "quicksort", "mergesort",
@@ -1768,6 +1769,7 @@ class DeviceArray:
             code = _dtype_code(a.dtype)
             if (
                 code is not None
+                and a.size >= 1
                 and isinstance(bins, (int, _np.integer))
                 and 1 <= int(bins) <= 4096
             ):
