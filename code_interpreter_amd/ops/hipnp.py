@@ -1323,6 +1323,10 @@ class DeviceArray:
                 self, BoolDeviceArray
             ):
                 return self._logic(None, 4)
+            if name == "matmul" and len(inputs) == 2:
+                r = matmul(inputs[0], inputs[1], _force=True)
+                if r is not NotImplemented:
+                    return r
             if name in _CMP and len(inputs) == 2 and inputs[0] is self:
                 r = self._compare(name, inputs[1])
                 if r is not None:
@@ -2476,6 +2480,40 @@ def matmul(a, b, _force=False):
             prod = da._binary("multiply", db)
             if prod is not NotImplemented:
                 return prod.sum()  # 1-D dot: scalar on device
+    if (
+        len(a_shape) == 2
+        and len(b_shape) == 1
+        and a_shape[1] == b_shape[0]
+    ):
+        # matvec: x is bit-identical to a (k, 1) column matrix
+        m, k = a_shape
+        if _force or 2.0 * m * k >= MIN_MATMUL_FLOPS:
+            da, db = _to_device(a), _to_device(b)
+            if da is not None and db is not None and da.dtype == db.dtype:
+                code = _dtype_code(da.dtype)
+                if code is not None:
+                    hc = backend().gemm(
+                        da._dev_handle(), db._dev_handle(), m, 1, k, code
+                    )
+                    return DeviceArray(hc, (m,), da.dtype)
+        return NotImplemented
+    if (
+        len(a_shape) == 1
+        and len(b_shape) == 2
+        and a_shape[0] == b_shape[0]
+    ):
+        # vecmat: x is a (1, k) row matrix
+        k, n = b_shape
+        if _force or 2.0 * n * k >= MIN_MATMUL_FLOPS:
+            da, db = _to_device(a), _to_device(b)
+            if da is not None and db is not None and da.dtype == db.dtype:
+                code = _dtype_code(da.dtype)
+                if code is not None:
+                    hc = backend().gemm(
+                        da._dev_handle(), db._dev_handle(), 1, n, k, code
+                    )
+                    return DeviceArray(hc, (n,), da.dtype)
+        return NotImplemented
     if len(a_shape) != 2 or len(b_shape) != 2 or a_shape[1] != b_shape[0]:
         return NotImplemented
     m, k = a_shape

@@ -1335,3 +1335,18 @@ def test_ptp_average_isclose_on_device(fake):
     bn = a.copy(); bn[5] = np.nan
     m2 = np.isclose(_device(fake, an), _device(fake, bn))
     assert not np.asarray(m2)[5]
+
+
+def test_matvec_vecmat_on_device(fake):
+    A = np.random.default_rng(54).random((60, 80))
+    v = np.random.default_rng(55).random(80)
+    u = np.random.default_rng(56).random(60)
+    da, dv, du = _device(fake, A), _device(fake, v), _device(fake, u)
+    r = np.matmul(da, dv)
+    assert isinstance(r, hipnp.DeviceArray) and r.shape == (60,)
+    np.testing.assert_allclose(r.materialize(), A @ v, rtol=1e-12)
+    r2 = du @ da
+    assert isinstance(r2, hipnp.DeviceArray) and r2.shape == (80,)
+    np.testing.assert_allclose(r2.materialize(), u @ A, rtol=1e-12)
+    r3 = np.dot(_device(fake, A), _device(fake, v))
+    np.testing.assert_allclose(np.asarray(r3), A @ v, rtol=1e-12)

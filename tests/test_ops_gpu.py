@@ -1137,3 +1137,17 @@ def test_ptp_average_isclose_gpu(hnp):
     assert r_ptp == pytest.approx(np.ptp(a), rel=1e-12)
     assert r_avg == pytest.approx(np.average(a, weights=wa), rel=1e-10)
     assert close_same
+
+
+def test_matvec_vecmat_gpu(hnp):
+    A = hnp.rand(2000, 3000, seed=109)
+    v = hnp.rand(3000, seed=110)
+    r = A @ v
+    assert isinstance(r, hnp.DeviceArray) and r.shape == (2000,)
+    Aa, va = np.asarray(A), np.asarray(v)
+    np.testing.assert_allclose(r.materialize(), Aa @ va, rtol=1e-10)
+    u = hnp.rand(2000, seed=111)
+    r2 = u @ A
+    assert isinstance(r2, hnp.DeviceArray) and r2.shape == (3000,)
+    np.testing.assert_allclose(
+        r2.materialize(), np.asarray(u) @ Aa, rtol=1e-10)
