@@ -117,3 +117,15 @@ def test_analysis_example_cpu_fallback(http_client):
     assert out["exit_code"] == 0, out["stderr"]
     assert "outliers:" in out["stdout"]
     assert "kind: ndarray" in out["stdout"]
+
+
+def test_analytics_surface(http_client):
+    out = _execute(http_client, "analytics.py")
+    assert out["exit_code"] == 0, out["stderr"]
+    lines = out["stdout"].splitlines()
+    assert lines[2].startswith("hist total: 2000000")
+    assert lines[3].endswith("True")
+    assert lines[4].endswith("True")
+    assert lines[5].endswith("True")
+    assert lines[6].endswith("True")
+    assert lines[7].endswith("True")
