@@ -1680,6 +1680,40 @@ class DeviceArray:
                         h, (rows, cols - 1), a.dtype
                     )._device_transposed()
             return _np.diff(a.materialize(), **kwargs)
+        if func is _np.nan_to_num and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and set(kwargs) <= {"nan", "posinf", "neginf"}:
+            a = args[0]
+            if _dtype_code(a.dtype) is not None:
+                nan_v = float(kwargs.get("nan", 0.0))
+                big = float(_np.finfo(a.dtype).max)
+                pos_v = float(kwargs.get("posinf", big))
+                neg_v = float(kwargs.get("neginf", -big))
+                r = a
+                m = r.isnan()
+                if isinstance(m, BoolDeviceArray) and int(m.sum()):
+                    r2 = where_device(m, nan_v, r)
+                    if r2 is not NotImplemented:
+                        r = r2
+                for bound, repl in ((_np.inf, pos_v), (-_np.inf, neg_v)):
+                    m = r._compare("equal", bound)
+                    if m is not None and int(m.sum()):
+                        r2 = where_device(m, repl, r)
+                        if r2 is not NotImplemented:
+                            r = r2
+                # numpy always returns a copy; when nothing needed
+                # replacing, clone so mutation can't alias the source
+                return r if r is not a else DeviceArray(
+                    a._device_clone(), a.shape, a.dtype
+                )
+            return _np.nan_to_num(a.materialize(), **kwargs)
+        if func in (_np.real, _np.conj, _np.conjugate) and len(args) == 1 \
+                and isinstance(args[0], DeviceArray) and not kwargs:
+            return args[0]  # real dtypes only ever reach the device
+        if func is _np.imag and len(args) == 1 and isinstance(
+            args[0], DeviceArray
+        ) and not kwargs:
+            return _np.zeros(args[0].shape, dtype=args[0].dtype)
         if func is _np.ptp and len(args) == 1 and isinstance(
             args[0], DeviceArray
         ) and not kwargs:

@@ -1350,3 +1350,28 @@ def test_matvec_vecmat_on_device(fake):
     np.testing.assert_allclose(r2.materialize(), u @ A, rtol=1e-12)
     r3 = np.dot(_device(fake, A), _device(fake, v))
     np.testing.assert_allclose(np.asarray(r3), A @ v, rtol=1e-12)
+
+
+def test_nan_to_num_real_imag_on_device(fake):
+    host = np.random.default_rng(57).random(2000) - 0.5
+    host[3] = np.nan
+    host[7] = np.inf
+    host[11] = -np.inf
+    x = _device(fake, host)
+    r = np.nan_to_num(x)
+    assert isinstance(r, hipnp.DeviceArray)
+    np.testing.assert_allclose(r.materialize(), np.nan_to_num(host), rtol=0)
+    r2 = np.nan_to_num(_device(fake, host), nan=-1.0, posinf=9.0, neginf=-9.0)
+    np.testing.assert_allclose(
+        np.asarray(r2),
+        np.nan_to_num(host, nan=-1.0, posinf=9.0, neginf=-9.0), rtol=0)
+    clean = np.random.default_rng(58).random(100)
+    rc = np.nan_to_num(_device(fake, clean))
+    assert isinstance(rc, hipnp.DeviceArray)
+    rc_host = rc.materialize()
+    np.testing.assert_array_equal(rc_host, clean)
+    # numpy copy semantics: mutating the result leaves the source intact
+    assert float(np.real(_device(fake, clean)).sum()) == pytest.approx(
+        clean.sum(), rel=1e-12)
+    np.testing.assert_array_equal(
+        np.imag(_device(fake, clean)), np.zeros(100))
