@@ -1381,575 +1381,10 @@ class DeviceArray:
 
     # -- NEP 18: numpy functions -----------------------------------------
     def __array_function__(self, func, types, args, kwargs):
-        if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
-            if set(kwargs) <= {"axis", "keepdims"}:
-                return args[0].sum(**kwargs)
-        if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
-            if set(kwargs) <= {"axis", "keepdims"}:
-                return args[0].mean(**kwargs)
-        if func is _np.trace and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            a = args[0]
-            if (
-                len(a.shape) == 2
-                and _dtype_code(a.dtype) is not None
-                and a._host is None
-            ):
-                n = min(a.shape)
-                esz = a.dtype.itemsize
-                raw = backend().download_strided(
-                    a._dev_handle(), 0, (a.shape[1] + 1) * esz, esz, n
-                )
-                return a.dtype.type(
-                    _np.frombuffer(raw, dtype=a.dtype).sum()
-                )
-            return _np.trace(a.materialize())
-        if func is _np.outer and len(args) == 2 and not kwargs:
-            x, y = args
-            if (
-                isinstance(x, DeviceArray)
-                and isinstance(y, DeviceArray)
-                and len(x.shape) == 1
-                and len(y.shape) == 1
-                and x.dtype == y.dtype
-                and _dtype_code(x.dtype) is not None
-            ):
-                h = backend().gemm(
-                    x._dev_handle(), y._dev_handle(), x.size, y.size, 1,
-                    _dtype_code(x.dtype),
-                )
-                return DeviceArray(h, (x.size, y.size), x.dtype)
-            host = [
-                v.materialize() if isinstance(v, DeviceArray) else v
-                for v in args
-            ]
-            return _np.outer(*host)
-        if func is _np.einsum and len(args) >= 2 and isinstance(
-            args[0], str
-        ) and not kwargs:
-            # the common contractions, mapped onto existing device ops;
-            # anything else materializes below
-            sub = args[0].replace(" ", "")
-            ops_ = args[1:]
-            try:
-                if sub in ("ij,jk->ik", "ij,jk") and len(ops_) == 2:
-                    r = matmul(ops_[0], ops_[1], _force=True)
-                    if r is not NotImplemented:
-                        return r
-                elif sub in ("bij,bjk->bik", "bij,bjk") and len(ops_) == 2:
-                    r = matmul(ops_[0], ops_[1], _force=True)
-                    if r is not NotImplemented:
-                        return r
-                elif sub in ("i,i->", "i,i") and len(ops_) == 2:
-                    r = matmul(ops_[0], ops_[1], _force=True)
-                    if r is not NotImplemented:
-                        return r
-                elif sub in ("ij->ji",) and len(ops_) == 1 and isinstance(
-                    ops_[0], DeviceArray
-                ):
-                    return ops_[0].transpose()
-                elif sub in ("ij->", "i->") and len(ops_) == 1 and isinstance(
-                    ops_[0], DeviceArray
-                ):
-                    return ops_[0].sum()
-                elif sub == "ii" and len(ops_) == 1:
-                    return DeviceArray.__array_function__(
-                        ops_[0], _np.trace, (DeviceArray,), (ops_[0],), {}
-                    )
-                elif sub in ("i,j->ij", "i,j") and len(ops_) == 2:
-                    return DeviceArray.__array_function__(
-                        self, _np.outer, (DeviceArray,), tuple(ops_), {}
-                    )
-            except Exception:
-                pass
-            host = [
-                v.materialize() if isinstance(v, DeviceArray) else v
-                for v in ops_
-            ]
-            return _np.einsum(sub, *host)
-        if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
-            r = matmul(*args, _force=True)
-            if r is not NotImplemented:
+        for stage in (_af_linalg, _af_order_stats, _af_array_ops, _af_cleanup, _af_structure):
+            r = stage(func, args, kwargs)
+            if r is not _AF_PASS:
                 return r
-        if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
-            return args[0]._unary("square")
-        if func is _np.median and len(args) == 1 and not kwargs:
-            r = quantile_device(args[0], 0.5)
-            if r is not None:
-                return r
-        if func is _np.median and len(args) == 1 and set(kwargs) == {"axis"}:
-            if kwargs["axis"] in (1, -1):
-                r = quantile_rows_device(args[0], 0.5)
-                if r is not None:
-                    return r
-            elif kwargs["axis"] == 0:
-                r = quantile_cols_device(args[0], 0.5)
-                if r is not None:
-                    return r
-            a0 = args[0]
-            if isinstance(a0, DeviceArray):
-                return _np.median(a0.materialize(), **kwargs)
-        if func in (_np.quantile, _np.percentile) and len(args) == 2 and not kwargs:
-            qv = args[1]
-            if isinstance(qv, (int, float)):
-                q = qv / 100.0 if func is _np.percentile else float(qv)
-                r = quantile_device(args[0], q)
-                if r is not None:
-                    return r
-            elif isinstance(qv, (list, tuple, _np.ndarray)):
-                qarr = _np.asarray(qv, dtype=_np.float64).reshape(-1)
-                if qarr.size and _np.all((qarr >= 0) & (qarr <= 100)):
-                    qs = qarr / 100.0 if func is _np.percentile else qarr
-                    if _np.all(qs <= 1.0):
-                        r = quantile_list_device(args[0], qs)
-                        if r is not None:
-                            return r.reshape(_np.asarray(qv).shape)
-        if func in (_np.quantile, _np.percentile) and len(args) == 2 and set(
-            kwargs
-        ) == {"axis"}:
-            qv = args[1]
-            if isinstance(qv, (int, float)) and kwargs["axis"] in (0, 1, -1):
-                q = qv / 100.0 if func is _np.percentile else float(qv)
-                if kwargs["axis"] == 0:
-                    r = quantile_cols_device(args[0], q)
-                else:
-                    r = quantile_rows_device(args[0], q)
-                if r is not None:
-                    return r
-            a0 = args[0]
-            if isinstance(a0, DeviceArray):
-                return func(a0.materialize(), qv, **kwargs)
-        if func is _np.where and len(args) == 3 and not kwargs:
-            r = where_device(*args)
-            if r is not NotImplemented:
-                return r
-        if func is _np.count_nonzero and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            if not kwargs:
-                if isinstance(args[0], BoolDeviceArray):
-                    return args[0].sum()
-                return args[0].count_nonzero()
-        if func is _np.cumsum and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            if set(kwargs) <= {"axis"}:
-                return args[0].cumsum(**kwargs)
-        if func in (_np.any, _np.all) and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            if not kwargs:
-                meth = "any" if func is _np.any else "all"
-                return getattr(args[0], meth)()
-        if func is _np.clip and len(args) >= 1 and isinstance(args[0], DeviceArray):
-            if len(args) <= 3 and not kwargs:
-                return args[0].clip(*args[1:])
-        if func in (_np.argmax, _np.argmin) and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            if set(kwargs) <= {"axis"}:
-                meth = "argmax" if func is _np.argmax else "argmin"
-                return getattr(args[0], meth)(**kwargs)
-        if func in (_np.partition, _np.argpartition) and len(args) == 2 \
-                and isinstance(args[0], DeviceArray):
-            a, kth = args
-            axis = kwargs.get("axis", -1)
-            order = kwargs.get("order")
-            extra = {
-                k: v for k, v in kwargs.items()
-                if k not in ("axis", "kind", "order")
-            }
-            if a._sort_routable(axis, None, order, extra):
-                ax01 = a._sort_axis01(axis)
-                if func is _np.partition:
-                    return a._device_sorted(False, axis=ax01)
-                _, idx = a._device_sorted(True, axis=ax01)
-                return idx
-            return func(a.materialize(), kth, **kwargs)
-        if func is _np.unique and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            a = args[0]
-            if a._sort_routable(None, None, None, {}):
-                # numpy's unique IS sort + adjacent-compare
-                # (lib/arraysetops); NaNs sort last and collapse to one
-                # (equal_nan=True is the numpy>=1.21 default)
-                host = a._device_sorted(False).materialize()
-                if host.size == 0:
-                    return host
-                keep = _np.empty(host.size, dtype=bool)
-                keep[0] = True
-                _np.not_equal(host[1:], host[:-1], out=keep[1:])
-                n_nan = int(_np.isnan(host[-1:])[0] and
-                            _np.isnan(host).sum())
-                if n_nan > 1:
-                    keep[host.size - n_nan + 1:] = False
-                    keep[host.size - n_nan] = True
-                return host[keep]
-            return _np.unique(a.materialize())
-        if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            a = args[0]
-            axis = kwargs.get("axis", -1)
-            kind = kwargs.get("kind")
-            order = kwargs.get("order")
-            extra = {
-                k: v for k, v in kwargs.items()
-                if k not in ("axis", "kind", "order")
-            }
-            if a._sort_routable(axis, kind, order, extra):
-                ax01 = a._sort_axis01(axis)
-                if func is _np.sort:
-                    return a._device_sorted(False, axis=ax01)
-                _, idx = a._device_sorted(True, axis=ax01)
-                return idx
-            host = a.materialize()
-            return func(host, **kwargs)
-        if func is _np.searchsorted and len(args) in (2, 3) and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"side"}:
-            a = args[0]
-            v = args[1]
-            side = args[2] if len(args) == 3 else kwargs.get("side", "left")
-            code = _dtype_code(a.dtype)
-            if (
-                code is not None
-                and len(a.shape) == 1
-                and side in ("left", "right")
-            ):
-                scalar_q = _np.isscalar(v) or (
-                    isinstance(v, _np.ndarray) and v.ndim == 0
-                )
-                if isinstance(v, DeviceArray):
-                    if v.dtype == a.dtype and len(v.shape) >= 1:
-                        hv, m, vshape = v._dev_handle(), v.size, v.shape
-                    else:
-                        hv = None
-                elif scalar_q or isinstance(v, (list, tuple, _np.ndarray)):
-                    host_v = _np.ascontiguousarray(
-                        _np.asarray(v, dtype=a.dtype).reshape(-1)
-                    )
-                    if host_v.size >= 1:
-                        # wrapped so the staging buffer is freed on GC
-                        _tmp_q = DeviceArray(
-                            backend().upload(host_v), host_v.shape, a.dtype
-                        )
-                        hv, m, vshape = _tmp_q._handle, \
-                            host_v.size, _np.asarray(v).shape
-                    else:
-                        hv = None
-                else:
-                    hv = None
-                if hv is not None:
-                    h = backend().searchsorted(
-                        a._dev_handle(), a.shape[0], hv, m, code,
-                        1 if side == "right" else 0,
-                    )
-                    res = DeviceArray(h, (m,), _np.int64)
-                    if scalar_q:
-                        out = _np.intp(res.materialize()[0])
-                        return out
-                    if vshape != (m,):
-                        return res.materialize().reshape(vshape)
-                    return res
-            host = a.materialize()
-            return _np.searchsorted(host, *args[1:], **kwargs)
-        if func is _np.diff and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"axis", "n"} and kwargs.get("n", 1) == 1:
-            a = args[0]
-            axis = kwargs.get("axis", -1)
-            code = _dtype_code(a.dtype)
-            if code is not None and len(a.shape) in (1, 2):
-                nd = len(a.shape)
-                ax = a._norm_axis(axis, nd)
-                if nd == 1 and ax == 0 and a.shape[0] >= 2:
-                    h = backend().diff(a._dev_handle(), code, 1, a.shape[0])
-                    return DeviceArray(h, (a.shape[0] - 1,), a.dtype)
-                if nd == 2 and ax == 1 and a.shape[1] >= 2:
-                    rows, cols = a.shape
-                    h = backend().diff(a._dev_handle(), code, rows, cols)
-                    return DeviceArray(h, (rows, cols - 1), a.dtype)
-                if nd == 2 and ax == 0 and a.shape[0] >= 2:
-                    t = a._device_transposed()
-                    rows, cols = t.shape
-                    h = backend().diff(t._dev_handle(), code, rows, cols)
-                    return DeviceArray(
-                        h, (rows, cols - 1), a.dtype
-                    )._device_transposed()
-            return _np.diff(a.materialize(), **kwargs)
-        if func is _np.nan_to_num and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"nan", "posinf", "neginf"}:
-            a = args[0]
-            if _dtype_code(a.dtype) is not None:
-                nan_v = float(kwargs.get("nan", 0.0))
-                big = float(_np.finfo(a.dtype).max)
-                pos_v = float(kwargs.get("posinf", big))
-                neg_v = float(kwargs.get("neginf", -big))
-                r = a
-                m = r.isnan()
-                if isinstance(m, BoolDeviceArray) and int(m.sum()):
-                    r2 = where_device(m, nan_v, r)
-                    if r2 is not NotImplemented:
-                        r = r2
-                for bound, repl in ((_np.inf, pos_v), (-_np.inf, neg_v)):
-                    m = r._compare("equal", bound)
-                    if m is not None and int(m.sum()):
-                        r2 = where_device(m, repl, r)
-                        if r2 is not NotImplemented:
-                            r = r2
-                # numpy always returns a copy; when nothing needed
-                # replacing, clone so mutation can't alias the source
-                return r if r is not a else DeviceArray(
-                    a._device_clone(), a.shape, a.dtype
-                )
-            return _np.nan_to_num(a.materialize(), **kwargs)
-        if func in (_np.real, _np.conj, _np.conjugate) and len(args) == 1 \
-                and isinstance(args[0], DeviceArray) and not kwargs:
-            return args[0]  # real dtypes only ever reach the device
-        if func is _np.imag and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            return _np.zeros(args[0].shape, dtype=args[0].dtype)
-        if func is _np.ptp and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            a = args[0]
-            if _dtype_code(a.dtype) is not None:
-                return a.dtype.type(float(a.max()) - float(a.min()))
-            return _np.ptp(a.materialize())
-        if func is _np.average and len(args) == 1 and set(kwargs) <= {
-            "weights"
-        }:
-            a = args[0]
-            w = kwargs.get("weights")
-            if (
-                isinstance(a, DeviceArray)
-                and _dtype_code(a.dtype) is not None
-            ):
-                if w is None:
-                    return a.dtype.type(float(a.sum()) / a.size)
-                if (
-                    isinstance(w, DeviceArray)
-                    and w.shape == a.shape
-                    and w.dtype == a.dtype
-                ):
-                    num = a._binary("multiply", w)
-                    if num is not NotImplemented:
-                        den = float(w.sum())
-                        if den != 0:
-                            return a.dtype.type(float(num.sum()) / den)
-            host_w = w.materialize() if isinstance(w, DeviceArray) else w
-            host_a = a.materialize() if isinstance(a, DeviceArray) else a
-            return _np.average(host_a, weights=host_w)
-        if func in (_np.isclose, _np.allclose) and len(args) == 2 and set(
-            kwargs
-        ) <= {"rtol", "atol", "equal_nan"} and not kwargs.get("equal_nan"):
-            a, b = args
-            rtol = float(kwargs.get("rtol", 1e-05))
-            atol = float(kwargs.get("atol", 1e-08))
-            if (
-                isinstance(a, DeviceArray)
-                and isinstance(b, DeviceArray)
-                and a.shape == b.shape
-                and a.dtype == b.dtype
-                and _dtype_code(a.dtype) is not None
-            ):
-                # |a-b| <= atol + rtol*|b|, NaN comparisons are False
-                diff = a._binary("subtract", b)
-                if diff is not NotImplemented:
-                    tol = b._unary("absolute")._binary("multiply", rtol)
-                    if tol is not NotImplemented:
-                        tol = tol._binary("add", atol)
-                    if tol is not NotImplemented:
-                        mask = diff._unary("absolute")._compare(
-                            "less_equal", tol
-                        )
-                        if mask is not None:
-                            if func is _np.allclose:
-                                return bool(mask.all())
-                            return mask
-            host = [
-                v.materialize() if isinstance(v, DeviceArray) else v
-                for v in args
-            ]
-            return func(*host, **kwargs)
-        _nan_kinds = {
-            _np.nansum: "sum", _np.nanmean: "mean", _np.nanmax: "max",
-            _np.nanmin: "min", _np.nanstd: "std", _np.nanvar: "var",
-        }
-        if func in _nan_kinds and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"ddof"}:
-            kind = _nan_kinds[func]
-            ddof = kwargs.get("ddof", 0) if kind in ("std", "var") else 0
-            if not kwargs or kind in ("std", "var"):
-                r = nan_reduce_device(args[0], kind, ddof)
-                if r is not None:
-                    return r
-            return func(args[0].materialize(), **kwargs)
-        if func in (_np.cov, _np.corrcoef) and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            r = (cov_device if func is _np.cov else corrcoef_device)(args[0])
-            if r is not None:
-                return r
-            return func(args[0].materialize())
-        if func is _np.histogram and 1 <= len(args) <= 2 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"bins", "range"}:
-            a = args[0]
-            bins = args[1] if len(args) == 2 else kwargs.get("bins", 10)
-            rng = kwargs.get("range")
-            code = _dtype_code(a.dtype)
-            if (
-                code is not None
-                and a.size >= 1
-                and isinstance(bins, (int, _np.integer))
-                and 1 <= int(bins) <= 4096
-            ):
-                bins = int(bins)
-                if rng is not None:
-                    lo, hi = float(rng[0]), float(rng[1])
-                else:
-                    lo, hi = float(a.min()), float(a.max())
-                if _np.isfinite(lo) and _np.isfinite(hi) and hi >= lo:
-                    if hi == lo:  # numpy expands degenerate ranges
-                        lo, hi = lo - 0.5, hi + 0.5
-                    raw = backend().histogram(
-                        a._dev_handle(), code, a.size, lo, hi, bins, 1
-                    )
-                    counts = _np.frombuffer(raw, dtype=_np.uint64)
-                    hist = counts[:bins].astype(_np.int64)
-                    edges = _np.linspace(lo, hi, bins + 1)
-                    return hist, edges
-            host = a.materialize()
-            return _np.histogram(host, *args[1:], **kwargs)
-        if func in (_np.round, _np.around) and len(args) >= 1 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"decimals"}:
-            dec = args[1] if len(args) > 1 else kwargs.get("decimals", 0)
-            return args[0].round(dec)
-        if func in (_np.concatenate, _np.vstack, _np.hstack, _np.stack) \
-                and len(args) == 1 and isinstance(args[0], (list, tuple)):
-            axis = kwargs.get("axis", 0)
-            parts = args[0]
-            routable = (
-                set(kwargs) <= {"axis"}
-                and axis in (0, None)
-                and len(parts) >= 1
-                and all(isinstance(p, DeviceArray) for p in parts)
-                and len({p.dtype for p in parts}) == 1
-                and _dtype_code(parts[0].dtype) is not None
-                and all(p._host is None for p in parts)
-            )
-            if routable:
-                shapes = [p.shape for p in parts]
-                nd = len(shapes[0])
-                if func is _np.concatenate and axis is None:
-                    out_shape = (sum(p.size for p in parts),)
-                elif func in (_np.concatenate, _np.vstack):
-                    base = shapes[0][1:] if nd > 0 else ()
-                    if func is _np.vstack and nd == 1:
-                        # vstack of 1-D rows -> (k, n)
-                        if len({sh for sh in shapes}) == 1:
-                            out_shape = (len(parts), shapes[0][0])
-                        else:
-                            out_shape = None
-                    elif all(sh[1:] == base for sh in shapes):
-                        out_shape = (sum(sh[0] for sh in shapes),) + base
-                    else:
-                        out_shape = None
-                elif func is _np.hstack and nd == 1:
-                    out_shape = (sum(sh[0] for sh in shapes),)
-                elif func is _np.stack and len(set(shapes)) == 1:
-                    out_shape = (len(parts),) + shapes[0]
-                else:
-                    out_shape = None
-                if out_shape is not None:
-                    esz = parts[0].dtype.itemsize
-                    total = 1
-                    for d in out_shape:
-                        total *= d
-                    hd = backend().alloc(total * esz)
-                    out = DeviceArray(hd, out_shape, parts[0].dtype)
-                    off = 0
-                    for part in parts:
-                        backend().copy_d2d(
-                            hd, off, part._dev_handle(), 0,
-                            part.size * esz,
-                        )
-                        off += part.size * esz
-                    return out
-            host = [
-                p.materialize() if isinstance(p, DeviceArray) else p
-                for p in parts
-            ]
-            return func(host, **kwargs)
-        if func in (_np.reshape, _np.ravel) and len(args) >= 1 and isinstance(
-            args[0], DeviceArray
-        ) and not kwargs:
-            if func is _np.ravel and len(args) == 1:
-                return args[0].ravel()
-            if func is _np.reshape and len(args) == 2:
-                return args[0].reshape(args[1])
-        if func is _np.transpose and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ) and set(kwargs) <= {"axes"}:
-            return args[0].transpose(*(
-                (kwargs["axes"],) if kwargs.get("axes") is not None else ()
-            ))
-        if func is _np.linalg.norm and len(args) == 1 and isinstance(
-            args[0], DeviceArray
-        ):
-            a = args[0]
-            ord_ = kwargs.get("ord")
-            axis = kwargs.get("axis")
-            code = _dtype_code(a.dtype)
-            extra = set(kwargs) - {"ord", "axis"}
-            if code is not None and not extra:
-                if axis is None and (
-                    ord_ is None
-                    or (ord_ == 2 and a.ndim == 1)
-                    or (ord_ == "fro" and a.ndim == 2)
-                ):
-                    # flat 2-norm / Frobenius: fused square+sum kernel
-                    return a.dtype.type(float(a.square_sum()) ** 0.5)
-                if (
-                    isinstance(axis, int)
-                    and ord_ in (None, 2)
-                    and a.ndim == 2
-                ):
-                    sq = a._unary("square")
-                    if sq is not NotImplemented:
-                        ssum = sq.sum(axis=axis)
-                        if isinstance(ssum, DeviceArray):
-                            r = ssum._unary("sqrt")
-                            if r is not NotImplemented:
-                                return r
-            host = a.materialize()
-            return _np.linalg.norm(host, **kwargs)
-        _reductions = {
-            _np.max: "max", _np.amax: "max",
-            _np.min: "min", _np.amin: "min",
-            _np.std: "std", _np.var: "var",
-        }
-        meth = _reductions.get(func)
-        if meth and len(args) == 1 and isinstance(args[0], DeviceArray):
-            allowed = {"axis", "ddof", "keepdims"}
-            if set(kwargs) <= allowed:
-                if kwargs.get("axis") is None and not kwargs.get("keepdims"):
-                    call_kwargs = {}
-                    if meth in ("std", "var") and "ddof" in kwargs:
-                        call_kwargs["ddof"] = kwargs["ddof"]
-                    return getattr(args[0], meth)(**call_kwargs)
-                if meth in ("max", "min", "std", "var") and kwargs.get(
-                    "axis"
-                ) is not None:
-                    return getattr(args[0], meth)(**kwargs)
         # generic fallback: materialize every DeviceArray
         host_args = [
             x.materialize() if isinstance(x, DeviceArray) else x for x in args
@@ -2798,3 +2233,595 @@ def install(numpy_module, mode: str = "auto") -> None:
         if _npname == "absolute":
             np.abs = _patched
     _installed["done"] = True
+
+
+# ---------------------------------------------------------------------------
+# __array_function__ dispatch stages. Each stage handles a family of
+# numpy functions and returns _AF_PASS to fall through to the generic
+# materialize-and-delegate fallback. Kept as plain module functions so
+# each family reads (and diffs) independently.
+# ---------------------------------------------------------------------------
+_AF_PASS = object()
+def _af_linalg(func, args, kwargs):
+    if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
+        if set(kwargs) <= {"axis", "keepdims"}:
+            return args[0].sum(**kwargs)
+    if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
+        if set(kwargs) <= {"axis", "keepdims"}:
+            return args[0].mean(**kwargs)
+    if func is _np.trace and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        a = args[0]
+        if (
+            len(a.shape) == 2
+            and _dtype_code(a.dtype) is not None
+            and a._host is None
+        ):
+            n = min(a.shape)
+            esz = a.dtype.itemsize
+            raw = backend().download_strided(
+                a._dev_handle(), 0, (a.shape[1] + 1) * esz, esz, n
+            )
+            return a.dtype.type(
+                _np.frombuffer(raw, dtype=a.dtype).sum()
+            )
+        return _np.trace(a.materialize())
+    if func is _np.outer and len(args) == 2 and not kwargs:
+        x, y = args
+        if (
+            isinstance(x, DeviceArray)
+            and isinstance(y, DeviceArray)
+            and len(x.shape) == 1
+            and len(y.shape) == 1
+            and x.dtype == y.dtype
+            and _dtype_code(x.dtype) is not None
+        ):
+            h = backend().gemm(
+                x._dev_handle(), y._dev_handle(), x.size, y.size, 1,
+                _dtype_code(x.dtype),
+            )
+            return DeviceArray(h, (x.size, y.size), x.dtype)
+        host = [
+            v.materialize() if isinstance(v, DeviceArray) else v
+            for v in args
+        ]
+        return _np.outer(*host)
+    if func is _np.einsum and len(args) >= 2 and isinstance(
+        args[0], str
+    ) and not kwargs:
+        # the common contractions, mapped onto existing device ops;
+        # anything else materializes below
+        sub = args[0].replace(" ", "")
+        ops_ = args[1:]
+        try:
+            if sub in ("ij,jk->ik", "ij,jk") and len(ops_) == 2:
+                r = matmul(ops_[0], ops_[1], _force=True)
+                if r is not NotImplemented:
+                    return r
+            elif sub in ("bij,bjk->bik", "bij,bjk") and len(ops_) == 2:
+                r = matmul(ops_[0], ops_[1], _force=True)
+                if r is not NotImplemented:
+                    return r
+            elif sub in ("i,i->", "i,i") and len(ops_) == 2:
+                r = matmul(ops_[0], ops_[1], _force=True)
+                if r is not NotImplemented:
+                    return r
+            elif sub in ("ij->ji",) and len(ops_) == 1 and isinstance(
+                ops_[0], DeviceArray
+            ):
+                return ops_[0].transpose()
+            elif sub in ("ij->", "i->") and len(ops_) == 1 and isinstance(
+                ops_[0], DeviceArray
+            ):
+                return ops_[0].sum()
+            elif sub == "ii" and len(ops_) == 1:
+                return DeviceArray.__array_function__(
+                    ops_[0], _np.trace, (DeviceArray,), (ops_[0],), {}
+                )
+            elif sub in ("i,j->ij", "i,j") and len(ops_) == 2:
+                return DeviceArray.__array_function__(
+                    ops_[0], _np.outer, (DeviceArray,), tuple(ops_), {}
+                )
+        except Exception:
+            pass
+        host = [
+            v.materialize() if isinstance(v, DeviceArray) else v
+            for v in ops_
+        ]
+        return _np.einsum(sub, *host)
+    if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
+        r = matmul(*args, _force=True)
+        if r is not NotImplemented:
+            return r
+    if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
+        return args[0]._unary("square")
+    return _AF_PASS
+
+def _af_order_stats(func, args, kwargs):
+    if func is _np.median and len(args) == 1 and not kwargs:
+        r = quantile_device(args[0], 0.5)
+        if r is not None:
+            return r
+    if func is _np.median and len(args) == 1 and set(kwargs) == {"axis"}:
+        if kwargs["axis"] in (1, -1):
+            r = quantile_rows_device(args[0], 0.5)
+            if r is not None:
+                return r
+        elif kwargs["axis"] == 0:
+            r = quantile_cols_device(args[0], 0.5)
+            if r is not None:
+                return r
+        a0 = args[0]
+        if isinstance(a0, DeviceArray):
+            return _np.median(a0.materialize(), **kwargs)
+    if func in (_np.quantile, _np.percentile) and len(args) == 2 and not kwargs:
+        qv = args[1]
+        if isinstance(qv, (int, float)):
+            q = qv / 100.0 if func is _np.percentile else float(qv)
+            r = quantile_device(args[0], q)
+            if r is not None:
+                return r
+        elif isinstance(qv, (list, tuple, _np.ndarray)):
+            qarr = _np.asarray(qv, dtype=_np.float64).reshape(-1)
+            if qarr.size and _np.all((qarr >= 0) & (qarr <= 100)):
+                qs = qarr / 100.0 if func is _np.percentile else qarr
+                if _np.all(qs <= 1.0):
+                    r = quantile_list_device(args[0], qs)
+                    if r is not None:
+                        return r.reshape(_np.asarray(qv).shape)
+    if func in (_np.quantile, _np.percentile) and len(args) == 2 and set(
+        kwargs
+    ) == {"axis"}:
+        qv = args[1]
+        if isinstance(qv, (int, float)) and kwargs["axis"] in (0, 1, -1):
+            q = qv / 100.0 if func is _np.percentile else float(qv)
+            if kwargs["axis"] == 0:
+                r = quantile_cols_device(args[0], q)
+            else:
+                r = quantile_rows_device(args[0], q)
+            if r is not None:
+                return r
+        a0 = args[0]
+        if isinstance(a0, DeviceArray):
+            return func(a0.materialize(), qv, **kwargs)
+    if func is _np.where and len(args) == 3 and not kwargs:
+        r = where_device(*args)
+        if r is not NotImplemented:
+            return r
+    if func is _np.count_nonzero and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        if not kwargs:
+            if isinstance(args[0], BoolDeviceArray):
+                return args[0].sum()
+            return args[0].count_nonzero()
+    return _AF_PASS
+
+def _af_array_ops(func, args, kwargs):
+    if func is _np.cumsum and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        if set(kwargs) <= {"axis"}:
+            return args[0].cumsum(**kwargs)
+    if func in (_np.any, _np.all) and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        if not kwargs:
+            meth = "any" if func is _np.any else "all"
+            return getattr(args[0], meth)()
+    if func is _np.clip and len(args) >= 1 and isinstance(args[0], DeviceArray):
+        if len(args) <= 3 and not kwargs:
+            return args[0].clip(*args[1:])
+    if func in (_np.argmax, _np.argmin) and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        if set(kwargs) <= {"axis"}:
+            meth = "argmax" if func is _np.argmax else "argmin"
+            return getattr(args[0], meth)(**kwargs)
+    if func in (_np.partition, _np.argpartition) and len(args) == 2 \
+            and isinstance(args[0], DeviceArray):
+        a, kth = args
+        axis = kwargs.get("axis", -1)
+        order = kwargs.get("order")
+        extra = {
+            k: v for k, v in kwargs.items()
+            if k not in ("axis", "kind", "order")
+        }
+        if a._sort_routable(axis, None, order, extra):
+            ax01 = a._sort_axis01(axis)
+            if func is _np.partition:
+                return a._device_sorted(False, axis=ax01)
+            _, idx = a._device_sorted(True, axis=ax01)
+            return idx
+        return func(a.materialize(), kth, **kwargs)
+    if func is _np.unique and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        a = args[0]
+        if a._sort_routable(None, None, None, {}):
+            # numpy's unique IS sort + adjacent-compare
+            # (lib/arraysetops); NaNs sort last and collapse to one
+            # (equal_nan=True is the numpy>=1.21 default)
+            host = a._device_sorted(False).materialize()
+            if host.size == 0:
+                return host
+            keep = _np.empty(host.size, dtype=bool)
+            keep[0] = True
+            _np.not_equal(host[1:], host[:-1], out=keep[1:])
+            n_nan = int(_np.isnan(host[-1:])[0] and
+                        _np.isnan(host).sum())
+            if n_nan > 1:
+                keep[host.size - n_nan + 1:] = False
+                keep[host.size - n_nan] = True
+            return host[keep]
+        return _np.unique(a.materialize())
+    if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        a = args[0]
+        axis = kwargs.get("axis", -1)
+        kind = kwargs.get("kind")
+        order = kwargs.get("order")
+        extra = {
+            k: v for k, v in kwargs.items()
+            if k not in ("axis", "kind", "order")
+        }
+        if a._sort_routable(axis, kind, order, extra):
+            ax01 = a._sort_axis01(axis)
+            if func is _np.sort:
+                return a._device_sorted(False, axis=ax01)
+            _, idx = a._device_sorted(True, axis=ax01)
+            return idx
+        host = a.materialize()
+        return func(host, **kwargs)
+    if func is _np.searchsorted and len(args) in (2, 3) and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"side"}:
+        a = args[0]
+        v = args[1]
+        side = args[2] if len(args) == 3 else kwargs.get("side", "left")
+        code = _dtype_code(a.dtype)
+        if (
+            code is not None
+            and len(a.shape) == 1
+            and side in ("left", "right")
+        ):
+            scalar_q = _np.isscalar(v) or (
+                isinstance(v, _np.ndarray) and v.ndim == 0
+            )
+            if isinstance(v, DeviceArray):
+                if v.dtype == a.dtype and len(v.shape) >= 1:
+                    hv, m, vshape = v._dev_handle(), v.size, v.shape
+                else:
+                    hv = None
+            elif scalar_q or isinstance(v, (list, tuple, _np.ndarray)):
+                host_v = _np.ascontiguousarray(
+                    _np.asarray(v, dtype=a.dtype).reshape(-1)
+                )
+                if host_v.size >= 1:
+                    # wrapped so the staging buffer is freed on GC
+                    _tmp_q = DeviceArray(
+                        backend().upload(host_v), host_v.shape, a.dtype
+                    )
+                    hv, m, vshape = _tmp_q._handle, \
+                        host_v.size, _np.asarray(v).shape
+                else:
+                    hv = None
+            else:
+                hv = None
+            if hv is not None:
+                h = backend().searchsorted(
+                    a._dev_handle(), a.shape[0], hv, m, code,
+                    1 if side == "right" else 0,
+                )
+                res = DeviceArray(h, (m,), _np.int64)
+                if scalar_q:
+                    out = _np.intp(res.materialize()[0])
+                    return out
+                if vshape != (m,):
+                    return res.materialize().reshape(vshape)
+                return res
+        host = a.materialize()
+        return _np.searchsorted(host, *args[1:], **kwargs)
+    if func is _np.diff and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"axis", "n"} and kwargs.get("n", 1) == 1:
+        a = args[0]
+        axis = kwargs.get("axis", -1)
+        code = _dtype_code(a.dtype)
+        if code is not None and len(a.shape) in (1, 2):
+            nd = len(a.shape)
+            ax = a._norm_axis(axis, nd)
+            if nd == 1 and ax == 0 and a.shape[0] >= 2:
+                h = backend().diff(a._dev_handle(), code, 1, a.shape[0])
+                return DeviceArray(h, (a.shape[0] - 1,), a.dtype)
+            if nd == 2 and ax == 1 and a.shape[1] >= 2:
+                rows, cols = a.shape
+                h = backend().diff(a._dev_handle(), code, rows, cols)
+                return DeviceArray(h, (rows, cols - 1), a.dtype)
+            if nd == 2 and ax == 0 and a.shape[0] >= 2:
+                t = a._device_transposed()
+                rows, cols = t.shape
+                h = backend().diff(t._dev_handle(), code, rows, cols)
+                return DeviceArray(
+                    h, (rows, cols - 1), a.dtype
+                )._device_transposed()
+        return _np.diff(a.materialize(), **kwargs)
+    return _AF_PASS
+
+def _af_cleanup(func, args, kwargs):
+    if func is _np.nan_to_num and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"nan", "posinf", "neginf"}:
+        a = args[0]
+        if _dtype_code(a.dtype) is not None:
+            nan_v = float(kwargs.get("nan", 0.0))
+            big = float(_np.finfo(a.dtype).max)
+            pos_v = float(kwargs.get("posinf", big))
+            neg_v = float(kwargs.get("neginf", -big))
+            r = a
+            m = r.isnan()
+            if isinstance(m, BoolDeviceArray) and int(m.sum()):
+                r2 = where_device(m, nan_v, r)
+                if r2 is not NotImplemented:
+                    r = r2
+            for bound, repl in ((_np.inf, pos_v), (-_np.inf, neg_v)):
+                m = r._compare("equal", bound)
+                if m is not None and int(m.sum()):
+                    r2 = where_device(m, repl, r)
+                    if r2 is not NotImplemented:
+                        r = r2
+            # numpy always returns a copy; when nothing needed
+            # replacing, clone so mutation can't alias the source
+            return r if r is not a else DeviceArray(
+                a._device_clone(), a.shape, a.dtype
+            )
+        return _np.nan_to_num(a.materialize(), **kwargs)
+    if func in (_np.real, _np.conj, _np.conjugate) and len(args) == 1 \
+            and isinstance(args[0], DeviceArray) and not kwargs:
+        return args[0]  # real dtypes only ever reach the device
+    if func is _np.imag and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        return _np.zeros(args[0].shape, dtype=args[0].dtype)
+    if func is _np.ptp and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        a = args[0]
+        if _dtype_code(a.dtype) is not None:
+            return a.dtype.type(float(a.max()) - float(a.min()))
+        return _np.ptp(a.materialize())
+    if func is _np.average and len(args) == 1 and set(kwargs) <= {
+        "weights"
+    }:
+        a = args[0]
+        w = kwargs.get("weights")
+        if (
+            isinstance(a, DeviceArray)
+            and _dtype_code(a.dtype) is not None
+        ):
+            if w is None:
+                return a.dtype.type(float(a.sum()) / a.size)
+            if (
+                isinstance(w, DeviceArray)
+                and w.shape == a.shape
+                and w.dtype == a.dtype
+            ):
+                num = a._binary("multiply", w)
+                if num is not NotImplemented:
+                    den = float(w.sum())
+                    if den != 0:
+                        return a.dtype.type(float(num.sum()) / den)
+        host_w = w.materialize() if isinstance(w, DeviceArray) else w
+        host_a = a.materialize() if isinstance(a, DeviceArray) else a
+        return _np.average(host_a, weights=host_w)
+    if func in (_np.isclose, _np.allclose) and len(args) == 2 and set(
+        kwargs
+    ) <= {"rtol", "atol", "equal_nan"} and not kwargs.get("equal_nan"):
+        a, b = args
+        rtol = float(kwargs.get("rtol", 1e-05))
+        atol = float(kwargs.get("atol", 1e-08))
+        if (
+            isinstance(a, DeviceArray)
+            and isinstance(b, DeviceArray)
+            and a.shape == b.shape
+            and a.dtype == b.dtype
+            and _dtype_code(a.dtype) is not None
+        ):
+            # |a-b| <= atol + rtol*|b|, NaN comparisons are False
+            diff = a._binary("subtract", b)
+            if diff is not NotImplemented:
+                tol = b._unary("absolute")._binary("multiply", rtol)
+                if tol is not NotImplemented:
+                    tol = tol._binary("add", atol)
+                if tol is not NotImplemented:
+                    mask = diff._unary("absolute")._compare(
+                        "less_equal", tol
+                    )
+                    if mask is not None:
+                        if func is _np.allclose:
+                            return bool(mask.all())
+                        return mask
+        host = [
+            v.materialize() if isinstance(v, DeviceArray) else v
+            for v in args
+        ]
+        return func(*host, **kwargs)
+    _nan_kinds = {
+        _np.nansum: "sum", _np.nanmean: "mean", _np.nanmax: "max",
+        _np.nanmin: "min", _np.nanstd: "std", _np.nanvar: "var",
+    }
+    if func in _nan_kinds and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"ddof"}:
+        kind = _nan_kinds[func]
+        ddof = kwargs.get("ddof", 0) if kind in ("std", "var") else 0
+        if not kwargs or kind in ("std", "var"):
+            r = nan_reduce_device(args[0], kind, ddof)
+            if r is not None:
+                return r
+        return func(args[0].materialize(), **kwargs)
+    if func in (_np.cov, _np.corrcoef) and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        r = (cov_device if func is _np.cov else corrcoef_device)(args[0])
+        if r is not None:
+            return r
+        return func(args[0].materialize())
+    if func is _np.histogram and 1 <= len(args) <= 2 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"bins", "range"}:
+        a = args[0]
+        bins = args[1] if len(args) == 2 else kwargs.get("bins", 10)
+        rng = kwargs.get("range")
+        code = _dtype_code(a.dtype)
+        if (
+            code is not None
+            and a.size >= 1
+            and isinstance(bins, (int, _np.integer))
+            and 1 <= int(bins) <= 4096
+        ):
+            bins = int(bins)
+            if rng is not None:
+                lo, hi = float(rng[0]), float(rng[1])
+            else:
+                lo, hi = float(a.min()), float(a.max())
+            if _np.isfinite(lo) and _np.isfinite(hi) and hi >= lo:
+                if hi == lo:  # numpy expands degenerate ranges
+                    lo, hi = lo - 0.5, hi + 0.5
+                raw = backend().histogram(
+                    a._dev_handle(), code, a.size, lo, hi, bins, 1
+                )
+                counts = _np.frombuffer(raw, dtype=_np.uint64)
+                hist = counts[:bins].astype(_np.int64)
+                edges = _np.linspace(lo, hi, bins + 1)
+                return hist, edges
+        host = a.materialize()
+        return _np.histogram(host, *args[1:], **kwargs)
+    if func in (_np.round, _np.around) and len(args) >= 1 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"decimals"}:
+        dec = args[1] if len(args) > 1 else kwargs.get("decimals", 0)
+        return args[0].round(dec)
+    return _AF_PASS
+
+def _af_structure(func, args, kwargs):
+    if func in (_np.concatenate, _np.vstack, _np.hstack, _np.stack) \
+            and len(args) == 1 and isinstance(args[0], (list, tuple)):
+        axis = kwargs.get("axis", 0)
+        parts = args[0]
+        routable = (
+            set(kwargs) <= {"axis"}
+            and axis in (0, None)
+            and len(parts) >= 1
+            and all(isinstance(p, DeviceArray) for p in parts)
+            and len({p.dtype for p in parts}) == 1
+            and _dtype_code(parts[0].dtype) is not None
+            and all(p._host is None for p in parts)
+        )
+        if routable:
+            shapes = [p.shape for p in parts]
+            nd = len(shapes[0])
+            if func is _np.concatenate and axis is None:
+                out_shape = (sum(p.size for p in parts),)
+            elif func in (_np.concatenate, _np.vstack):
+                base = shapes[0][1:] if nd > 0 else ()
+                if func is _np.vstack and nd == 1:
+                    # vstack of 1-D rows -> (k, n)
+                    if len({sh for sh in shapes}) == 1:
+                        out_shape = (len(parts), shapes[0][0])
+                    else:
+                        out_shape = None
+                elif all(sh[1:] == base for sh in shapes):
+                    out_shape = (sum(sh[0] for sh in shapes),) + base
+                else:
+                    out_shape = None
+            elif func is _np.hstack and nd == 1:
+                out_shape = (sum(sh[0] for sh in shapes),)
+            elif func is _np.stack and len(set(shapes)) == 1:
+                out_shape = (len(parts),) + shapes[0]
+            else:
+                out_shape = None
+            if out_shape is not None:
+                esz = parts[0].dtype.itemsize
+                total = 1
+                for d in out_shape:
+                    total *= d
+                hd = backend().alloc(total * esz)
+                out = DeviceArray(hd, out_shape, parts[0].dtype)
+                off = 0
+                for part in parts:
+                    backend().copy_d2d(
+                        hd, off, part._dev_handle(), 0,
+                        part.size * esz,
+                    )
+                    off += part.size * esz
+                return out
+        host = [
+            p.materialize() if isinstance(p, DeviceArray) else p
+            for p in parts
+        ]
+        return func(host, **kwargs)
+    if func in (_np.reshape, _np.ravel) and len(args) >= 1 and isinstance(
+        args[0], DeviceArray
+    ) and not kwargs:
+        if func is _np.ravel and len(args) == 1:
+            return args[0].ravel()
+        if func is _np.reshape and len(args) == 2:
+            return args[0].reshape(args[1])
+    if func is _np.transpose and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ) and set(kwargs) <= {"axes"}:
+        return args[0].transpose(*(
+            (kwargs["axes"],) if kwargs.get("axes") is not None else ()
+        ))
+    if func is _np.linalg.norm and len(args) == 1 and isinstance(
+        args[0], DeviceArray
+    ):
+        a = args[0]
+        ord_ = kwargs.get("ord")
+        axis = kwargs.get("axis")
+        code = _dtype_code(a.dtype)
+        extra = set(kwargs) - {"ord", "axis"}
+        if code is not None and not extra:
+            if axis is None and (
+                ord_ is None
+                or (ord_ == 2 and a.ndim == 1)
+                or (ord_ == "fro" and a.ndim == 2)
+            ):
+                # flat 2-norm / Frobenius: fused square+sum kernel
+                return a.dtype.type(float(a.square_sum()) ** 0.5)
+            if (
+                isinstance(axis, int)
+                and ord_ in (None, 2)
+                and a.ndim == 2
+            ):
+                sq = a._unary("square")
+                if sq is not NotImplemented:
+                    ssum = sq.sum(axis=axis)
+                    if isinstance(ssum, DeviceArray):
+                        r = ssum._unary("sqrt")
+                        if r is not NotImplemented:
+                            return r
+        host = a.materialize()
+        return _np.linalg.norm(host, **kwargs)
+    _reductions = {
+        _np.max: "max", _np.amax: "max",
+        _np.min: "min", _np.amin: "min",
+        _np.std: "std", _np.var: "var",
+    }
+    meth = _reductions.get(func)
+    if meth and len(args) == 1 and isinstance(args[0], DeviceArray):
+        allowed = {"axis", "ddof", "keepdims"}
+        if set(kwargs) <= allowed:
+            if kwargs.get("axis") is None and not kwargs.get("keepdims"):
+                call_kwargs = {}
+                if meth in ("std", "var") and "ddof" in kwargs:
+                    call_kwargs["ddof"] = kwargs["ddof"]
+                return getattr(args[0], meth)(**call_kwargs)
+            if meth in ("max", "min", "std", "var") and kwargs.get(
+                "axis"
+            ) is not None:
+                return getattr(args[0], meth)(**kwargs)
+    return _AF_PASS
