@@ -2221,6 +2221,75 @@ def install(numpy_module, mode: str = "auto") -> None:
     np.dot = patched_dot
     np.square = patched_square
     np.sum = patched_sum
+    orig_sort = np.sort
+    orig_argsort = np.argsort
+    orig_median = np.median
+
+    def _promote_big(x):
+        """Upload a large host ndarray (PCIe upload + device op beats a
+        host O(n log n) pass by 10-100x at >=MIN_ELEMS)."""
+        if (
+            isinstance(x, _np.ndarray)
+            and x.size >= MIN_ELEMS
+            and _dtype_code(x.dtype) is not None
+            and x.flags.c_contiguous
+        ):
+            return _to_device(x)
+        return None
+
+    def patched_sort(a, axis=-1, kind=None, order=None, **kw):
+        # numpy 2.x adds a keyword-only `stable`; stable=True is what the
+        # radix sort IS, so it routes too
+        routable_kw = not kw or (set(kw) == {"stable"})
+        if not isinstance(a, DeviceArray) and order is None and routable_kw:
+            try:
+                d = _promote_big(a)
+                if d is not None and d._sort_routable(axis, kind, order, {}):
+                    return d._device_sorted(False, axis=d._sort_axis01(axis))
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_sort(a, axis=axis, kind=kind, order=order, **kw)
+
+    def patched_argsort(a, axis=-1, kind=None, order=None, **kw):
+        routable_kw = not kw or (set(kw) == {"stable"})
+        if not isinstance(a, DeviceArray) and order is None and routable_kw:
+            try:
+                d = _promote_big(a)
+                if d is not None and d._sort_routable(axis, kind, order, {}):
+                    _, idx = d._device_sorted(
+                        True, axis=d._sort_axis01(axis)
+                    )
+                    return idx
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_argsort(a, axis=axis, kind=kind, order=order, **kw)
+
+    def patched_median(a, axis=None, **kwargs):
+        if not isinstance(a, DeviceArray) and not kwargs:
+            try:
+                d = _promote_big(a)
+                if d is not None:
+                    if axis is None:
+                        r = quantile_device(d, 0.5)
+                    elif axis in (1, -1) and len(d.shape) == 2:
+                        r = quantile_rows_device(d, 0.5)
+                    elif axis == 0 and len(d.shape) == 2:
+                        r = quantile_cols_device(d, 0.5)
+                    else:
+                        r = None
+                    if r is not None:
+                        return r
+            except Exception:
+                if mode == "require":
+                    raise
+        return orig_median(a, axis=axis, **kwargs)
+
+    np.sort = patched_sort
+    np.argsort = patched_argsort
+    np.median = patched_median
+
     # the rest of the hot unary surface, same pattern (np.abs is an
     # alias of np.absolute; both get the patch)
     for _uname, _npname in (

@@ -689,7 +689,7 @@ def test_module_level_unary_routing(fake, monkeypatch):
     np_mod = types.SimpleNamespace(**{
         name: getattr(np, name) for name in (
             "sum", "square", "matmul", "dot", "sqrt", "exp", "log", "sin",
-            "cos", "tanh", "absolute", "abs",
+            "cos", "tanh", "absolute", "abs", "sort", "argsort", "median",
         )
     })
     np_mod.random = types.SimpleNamespace(
@@ -717,6 +717,18 @@ def test_module_level_unary_routing(fake, monkeypatch):
     out = np.empty(256)
     got = np_mod.exp(host, out=out)
     assert got is out
+    # big HOST arrays promote through the patched order-stat entry points
+    big = np.random.default_rng(33).random(256)
+    srt = np_mod.sort(big)
+    assert isinstance(srt, hipnp.DeviceArray)
+    np.testing.assert_array_equal(np.asarray(srt), np.sort(big))
+    idx = np_mod.argsort(big)
+    assert isinstance(idx, hipnp.DeviceArray)
+    np.testing.assert_array_equal(np.asarray(idx), np.argsort(big))
+    med = np_mod.median(big)
+    assert med == pytest.approx(np.median(big), abs=1e-12)
+    # small host arrays stay host; structured order passes through
+    assert isinstance(np_mod.sort(small), np.ndarray)
 
 
 def test_mixed_dtype_and_mask_arithmetic_falls_back(fake):
