@@ -2311,6 +2311,16 @@ def install(numpy_module, mode: str = "auto") -> None:
 # each family reads (and diffs) independently.
 # ---------------------------------------------------------------------------
 _AF_PASS = object()
+# identity snapshots taken at import time: install() REPLACES
+# numpy.sort/argsort/median module attributes, but the NEP-18 protocol
+# dispatches with the ORIGINAL function object (the wrappers call the
+# saved originals), so handlers must recognize both
+_NP_SORT0 = _np.sort
+_NP_ARGSORT0 = _np.argsort
+_NP_MEDIAN0 = _np.median
+_SORT_FUNCS = (_NP_SORT0, _np.sort)
+_ARGSORT_FUNCS = (_NP_ARGSORT0, _np.argsort)
+_MEDIAN_FUNCS = (_NP_MEDIAN0, _np.median)
 def _af_linalg(func, args, kwargs):
     if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
         if set(kwargs) <= {"axis", "keepdims"}:
@@ -2408,11 +2418,11 @@ def _af_linalg(func, args, kwargs):
     return _AF_PASS
 
 def _af_order_stats(func, args, kwargs):
-    if func is _np.median and len(args) == 1 and not kwargs:
+    if func in _MEDIAN_FUNCS and len(args) == 1 and not kwargs:
         r = quantile_device(args[0], 0.5)
         if r is not None:
             return r
-    if func is _np.median and len(args) == 1 and set(kwargs) == {"axis"}:
+    if func in _MEDIAN_FUNCS and len(args) == 1 and set(kwargs) == {"axis"}:
         if kwargs["axis"] in (1, -1):
             r = quantile_rows_device(args[0], 0.5)
             if r is not None:
@@ -2525,9 +2535,8 @@ def _af_array_ops(func, args, kwargs):
                 keep[host.size - n_nan] = True
             return host[keep]
         return _np.unique(a.materialize())
-    if func in (_np.sort, _np.argsort) and len(args) == 1 and isinstance(
-        args[0], DeviceArray
-    ):
+    if func in _SORT_FUNCS + _ARGSORT_FUNCS and len(args) == 1 \
+            and isinstance(args[0], DeviceArray):
         a = args[0]
         axis = kwargs.get("axis", -1)
         kind = kwargs.get("kind")
@@ -2538,7 +2547,7 @@ def _af_array_ops(func, args, kwargs):
         }
         if a._sort_routable(axis, kind, order, extra):
             ax01 = a._sort_axis01(axis)
-            if func is _np.sort:
+            if func in _SORT_FUNCS:
                 return a._device_sorted(False, axis=ax01)
             _, idx = a._device_sorted(True, axis=ax01)
             return idx

@@ -1387,3 +1387,22 @@ def test_nan_to_num_real_imag_on_device(fake):
         clean.sum(), rel=1e-12)
     np.testing.assert_array_equal(
         np.imag(_device(fake, clean)), np.zeros(100))
+
+
+def test_protocol_survives_entry_point_patching(fake, monkeypatch):
+    """install() replaces numpy.sort/argsort/median; the NEP-18 protocol
+    still dispatches with the ORIGINAL function objects, which must keep
+    routing (regression: sandbox numpy.sort silently went to host)."""
+    orig_sort, orig_argsort, orig_median = np.sort, np.argsort, np.median
+    monkeypatch.setattr(np, "sort", lambda *a, **k: orig_sort(*a, **k))
+    monkeypatch.setattr(np, "argsort", lambda *a, **k: orig_argsort(*a, **k))
+    monkeypatch.setattr(np, "median", lambda *a, **k: orig_median(*a, **k))
+    host = np.random.default_rng(59).random(1000)
+    x = _device(fake, host)
+    s = np.sort(x)  # wrapper -> orig -> protocol with func=orig
+    assert isinstance(s, hipnp.DeviceArray)
+    np.testing.assert_array_equal(s.materialize(), orig_sort(host))
+    idx = np.argsort(_device(fake, host))
+    assert isinstance(idx, hipnp.DeviceArray)
+    med = np.median(_device(fake, host))
+    assert med == pytest.approx(orig_median(host), abs=1e-12)
