@@ -2152,9 +2152,13 @@ def install(numpy_module, mode: str = "auto") -> None:
     patched_square = _make_patched_unary("square", orig_square)
 
     def patched_sum(x, *args, **kwargs):
+        if isinstance(x, DeviceArray) and not args and set(kwargs) <= {
+            "axis", "keepdims"
+        }:
+            # the method handles axis/keepdims on device and falls back
+            # itself; going through orig_sum would materialize
+            return x.sum(**kwargs)
         if not args and (not kwargs or set(kwargs) <= {"axis"}) and kwargs.get("axis") is None:
-            if isinstance(x, DeviceArray):
-                return x.sum()
             arr = x if isinstance(x, _np.ndarray) else None
             if (
                 arr is not None
@@ -2321,8 +2325,11 @@ _NP_MEDIAN0 = _np.median
 _SORT_FUNCS = (_NP_SORT0, _np.sort)
 _ARGSORT_FUNCS = (_NP_ARGSORT0, _np.argsort)
 _MEDIAN_FUNCS = (_NP_MEDIAN0, _np.median)
+_SUM_FUNCS = (_np.sum,)
+_MATMUL_FUNCS = (_np.matmul, _np.dot)
+_SQUARE_FUNCS = (_np.square,)
 def _af_linalg(func, args, kwargs):
-    if func is _np.sum and len(args) == 1 and isinstance(args[0], DeviceArray):
+    if func in _SUM_FUNCS and len(args) == 1 and isinstance(args[0], DeviceArray):
         if set(kwargs) <= {"axis", "keepdims"}:
             return args[0].sum(**kwargs)
     if func is _np.mean and len(args) == 1 and isinstance(args[0], DeviceArray):
@@ -2409,11 +2416,11 @@ def _af_linalg(func, args, kwargs):
             for v in ops_
         ]
         return _np.einsum(sub, *host)
-    if func in (_np.matmul, _np.dot) and len(args) == 2 and not kwargs:
+    if func in _MATMUL_FUNCS and len(args) == 2 and not kwargs:
         r = matmul(*args, _force=True)
         if r is not NotImplemented:
             return r
-    if func is _np.square and len(args) == 1 and isinstance(args[0], DeviceArray):
+    if func in _SQUARE_FUNCS and len(args) == 1 and isinstance(args[0], DeviceArray):
         return args[0]._unary("square")
     return _AF_PASS
 

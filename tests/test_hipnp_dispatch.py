@@ -1406,3 +1406,32 @@ def test_protocol_survives_entry_point_patching(fake, monkeypatch):
     assert isinstance(idx, hipnp.DeviceArray)
     med = np.median(_device(fake, host))
     assert med == pytest.approx(orig_median(host), abs=1e-12)
+
+
+def test_patched_sum_routes_axis_reductions(fake, monkeypatch):
+    """np.sum(device_2d, axis=...) through the PATCHED module entry point
+    must use the device axis reducer, not materialize (regression)."""
+    import types
+    np_mod = types.SimpleNamespace(sum=np.sum)
+    np_mod.random = types.SimpleNamespace()
+    monkeypatch.setitem(hipnp._installed, "done", False)
+    monkeypatch.setattr(hipnp, "MIN_ELEMS", 64)
+    monkeypatch.setattr(hipnp, "available", lambda: True)
+    # minimal attrs install() touches
+    for name in ("square", "matmul", "dot", "sqrt", "exp", "log", "sin",
+                 "cos", "tanh", "absolute", "abs", "sort", "argsort",
+                 "median"):
+        setattr(np_mod, name, getattr(np, name))
+    for name in ("rand", "random", "random_sample", "uniform", "randn",
+                 "standard_normal", "normal"):
+        setattr(np_mod.random, name, getattr(np.random, name))
+    hipnp.install(np_mod, mode="auto")
+    host = np.random.default_rng(60).random((128, 96))
+    x = _device(fake, host)
+    fake.calls.clear()
+    r = np_mod.sum(x, axis=0)
+    assert isinstance(r, hipnp.DeviceArray)
+    assert "reduce_axis" in fake.calls and "download" not in fake.calls
+    np.testing.assert_allclose(
+        r.materialize(), host.sum(axis=0), rtol=1e-12)
+    assert float(np_mod.sum(x)) == pytest.approx(host.sum(), rel=1e-12)
