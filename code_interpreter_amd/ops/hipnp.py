@@ -2294,6 +2294,39 @@ def install(numpy_module, mode: str = "auto") -> None:
     np.argsort = patched_argsort
     np.median = patched_median
 
+    def _make_patched_scalar_reduce(meth, orig):
+        """np.mean/std/var/max/min on large HOST arrays: upload + one
+        fused device reduce beats the host pass 4-40x; returns the same
+        scalar. DeviceArray inputs go through the method (device axis
+        handling + fallback)."""
+
+        def patched(x, *args, **kwargs):
+            if isinstance(x, DeviceArray) and not args:
+                allowed = {"axis", "keepdims"} | (
+                    {"ddof"} if meth in ("std", "var") else set()
+                )
+                if set(kwargs) <= allowed:
+                    return getattr(x, meth)(**kwargs)
+            if not args and not kwargs:
+                try:
+                    d = _promote_big(x)
+                    if d is not None:
+                        return getattr(d, meth)()
+                except Exception:
+                    if mode == "require":
+                        raise
+            return orig(_asarray(x), *args, **kwargs)
+
+        return patched
+
+    for _meth, _names in (
+        ("mean", ("mean",)), ("std", ("std",)), ("var", ("var",)),
+        ("max", ("max", "amax")), ("min", ("min", "amin")),
+    ):
+        for _n in _names:
+            setattr(np, _n, _make_patched_scalar_reduce(
+                _meth, getattr(np, _n)))
+
     # the rest of the hot unary surface, same pattern (np.abs is an
     # alias of np.absolute; both get the patch)
     for _uname, _npname in (

@@ -690,6 +690,7 @@ def test_module_level_unary_routing(fake, monkeypatch):
         name: getattr(np, name) for name in (
             "sum", "square", "matmul", "dot", "sqrt", "exp", "log", "sin",
             "cos", "tanh", "absolute", "abs", "sort", "argsort", "median",
+            "mean", "std", "var", "max", "amax", "min", "amin",
         )
     })
     np_mod.random = types.SimpleNamespace(
@@ -729,6 +730,17 @@ def test_module_level_unary_routing(fake, monkeypatch):
     assert med == pytest.approx(np.median(big), abs=1e-12)
     # small host arrays stay host; structured order passes through
     assert isinstance(np_mod.sort(small), np.ndarray)
+    # scalar reductions promote big host arrays and match numpy
+    assert float(np_mod.mean(big)) == pytest.approx(big.mean(), rel=1e-12)
+    assert float(np_mod.std(big)) == pytest.approx(big.std(), rel=1e-9)
+    assert float(np_mod.max(big)) == pytest.approx(big.max(), rel=0)
+    assert float(np_mod.amin(big)) == pytest.approx(big.min(), rel=0)
+    assert isinstance(np_mod.mean(small), float) or np.isscalar(
+        np_mod.mean(small))
+    # DeviceArray + axis kwarg goes through the device method
+    da = _device(fake, np.random.default_rng(61).random((70, 80)))
+    r = np_mod.mean(da, axis=0)
+    assert isinstance(r, hipnp.DeviceArray)
 
 
 def test_mixed_dtype_and_mask_arithmetic_falls_back(fake):
@@ -1420,7 +1432,8 @@ def test_patched_sum_routes_axis_reductions(fake, monkeypatch):
     # minimal attrs install() touches
     for name in ("square", "matmul", "dot", "sqrt", "exp", "log", "sin",
                  "cos", "tanh", "absolute", "abs", "sort", "argsort",
-                 "median"):
+                 "median", "mean", "std", "var", "max", "amax", "min",
+                 "amin"):
         setattr(np_mod, name, getattr(np, name))
     for name in ("rand", "random", "random_sample", "uniform", "randn",
                  "standard_normal", "normal"):
