@@ -10,13 +10,19 @@ Two mechanisms, installed by the sandbox runtime (executor/sandbox_runtime
    _hipops and the result stays device-resident;
 2. DeviceArray: a duck array (NEP 13/18 __array_ufunc__ +
    __array_function__) so follow-on numpy calls on a device-resident
-   result keep running on the GPU: elementwise unary/binary (incl.
-   row/column broadcasting), full and axis-wise reductions, argmax/
-   argmin, clip, matmul (2-D / equal-batch 3-D / 1-D dot / pad-to-256
-   bf16), device boolean masks (comparisons, np.where, x[mask] = v,
-   popcount), in-place mutation and ufunc out=. Any unsupported
-   operation transparently materializes to a host ndarray and computes
-   numpy's own result -- identical values either way.
+   result keep running on the GPU: 26 elementwise unaries, binary ops
+   (incl. row/column broadcasting), full and axis-wise reductions (plus
+   the nan* family), argmax/argmin, clip, matmul (2-D / matvec /
+   equal-batch 3-D / 1-D dot / pad-to-256 bf16), sort/argsort/
+   partition/unique/searchsorted (stable radix sort, 2-D both axes),
+   median/quantile/percentile (flat, per-row/column, array q),
+   histogram, cumsum (flat/2-D), transpose/.T, reshape/ravel, diff,
+   cov/corrcoef, einsum common contractions, outer/trace, linalg.norm,
+   concatenate/stack family, nan_to_num, isclose/allclose, ptp,
+   weighted average, device boolean masks (comparisons, np.where,
+   x[mask] = v, popcount), in-place mutation and ufunc out=. Any
+   unsupported operation transparently materializes to a host ndarray
+   and computes numpy's own result -- identical values either way.
 
 Everything computes the same values user code would get on the CPU (same
 dtype; RNG is Philox instead of MT19937 -- a documented backend change,
