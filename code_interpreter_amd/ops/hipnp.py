@@ -2648,6 +2648,33 @@ def _af_array_ops(func, args, kwargs):
                 return res
         host = a.materialize()
         return _np.searchsorted(host, *args[1:], **kwargs)
+    if func is _np.digitize and len(args) == 2 and set(kwargs) <= {
+        "right"
+    }:
+        # numpy: digitize(x, bins) == searchsorted(bins, x, side=...)
+        # for monotonically increasing bins
+        x, bins = args
+        right = bool(kwargs.get("right", False))
+        if isinstance(x, DeviceArray) and len(x.shape) == 1:
+            b = _np.asarray(
+                bins.materialize() if isinstance(bins, DeviceArray) else bins
+            )
+            if (
+                b.ndim == 1
+                and b.size >= 1
+                and _np.all(b[1:] >= b[:-1])  # increasing bins only
+                and _dtype_code(x.dtype) is not None
+            ):
+                db = _to_device(_np.ascontiguousarray(b.astype(x.dtype)))
+                if db is not None:
+                    h = backend().searchsorted(
+                        db._dev_handle(), b.size, x._dev_handle(), x.size,
+                        _dtype_code(x.dtype), 0 if right else 1,
+                    )
+                    return DeviceArray(h, x.shape, _np.int64)
+        host_x = x.materialize() if isinstance(x, DeviceArray) else x
+        host_b = bins.materialize() if isinstance(bins, DeviceArray) else bins
+        return _np.digitize(host_x, host_b, **kwargs)
     if func is _np.diff and len(args) == 1 and isinstance(
         args[0], DeviceArray
     ) and set(kwargs) <= {"axis", "n"} and kwargs.get("n", 1) == 1:

@@ -1448,3 +1448,18 @@ def test_patched_sum_routes_axis_reductions(fake, monkeypatch):
     np.testing.assert_allclose(
         r.materialize(), host.sum(axis=0), rtol=1e-12)
     assert float(np_mod.sum(x)) == pytest.approx(host.sum(), rel=1e-12)
+
+
+def test_digitize_on_device(fake):
+    host = np.random.default_rng(62).random(3000) * 10
+    bins = np.array([1.0, 2.5, 5.0, 7.5, 9.0])
+    x = _device(fake, host)
+    r = np.digitize(x, bins)
+    assert isinstance(r, hipnp.DeviceArray) and r.dtype == np.int64
+    np.testing.assert_array_equal(r.materialize(), np.digitize(host, bins))
+    r2 = np.digitize(_device(fake, host), bins, right=True)
+    np.testing.assert_array_equal(
+        np.asarray(r2), np.digitize(host, bins, right=True))
+    # decreasing bins fall back to host numpy
+    r3 = np.digitize(_device(fake, host), bins[::-1].copy())
+    np.testing.assert_array_equal(r3, np.digitize(host, bins[::-1]))
