@@ -1601,6 +1601,51 @@ def _order_stat_device(x, k):
     return float("nan"), 0  # did not converge: caller falls back
 
 
+def quantile_rows_device_multi(x, qs):
+    """Per-row quantiles for an ARRAY of q's: ONE row sort, then two
+    pitched column downloads per q. Returns [len(qs)][rows] like numpy
+    (quantiles axis leads)."""
+    if not isinstance(x, DeviceArray) or len(x.shape) != 2:
+        return None
+    if _dtype_code(x.dtype) is None:
+        return None
+    rows, cols = x.shape
+    if cols < 1 or rows < 1 or x.size > (1 << 31):
+        return None
+    srt = x._device_sorted(False)
+    esz = x.dtype.itemsize
+
+    def col(k):
+        raw = backend().download_strided(
+            srt._dev_handle(), k * esz, cols * esz, esz, rows
+        )
+        return _np.frombuffer(raw, dtype=x.dtype).copy()
+
+    top = col(cols - 1)
+    nan_rows = _np.isnan(top)
+    out = _np.empty((len(qs), rows), dtype=x.dtype)
+    cache = {}
+    for i, q in enumerate(qs):
+        if not 0.0 <= float(q) <= 1.0:
+            return None
+        pos = float(q) * (cols - 1)
+        k0 = int(_np.floor(pos))
+        k1 = min(k0 + 1, cols - 1)
+        frac = x.dtype.type(pos - k0)
+        if k0 not in cache:
+            cache[k0] = col(k0)
+        c0 = cache[k0]
+        if k1 == k0 or frac == 0:
+            row = c0.copy()
+        else:
+            if k1 not in cache:
+                cache[k1] = col(k1)
+            row = c0 + (cache[k1] - c0) * frac
+        row[nan_rows] = _np.nan
+        out[i] = row
+    return out
+
+
 def quantile_rows_device(x, q):
     """Per-row quantile (axis=-1) of a 2-D DeviceArray: device row sort,
     then ONE pitched column download per interpolation endpoint (R
@@ -2507,6 +2552,21 @@ def _af_order_stats(func, args, kwargs):
                 r = quantile_rows_device(args[0], q)
             if r is not None:
                 return r
+        elif isinstance(qv, (list, tuple, _np.ndarray)) and kwargs[
+            "axis"
+        ] in (0, 1, -1):
+            qarr = _np.asarray(qv, dtype=_np.float64).reshape(-1)
+            if qarr.size and _np.all((qarr >= 0) & (qarr <= 100)):
+                qs = qarr / 100.0 if func is _np.percentile else qarr
+                if _np.all(qs <= 1.0):
+                    a0 = args[0]
+                    if kwargs["axis"] == 0 and isinstance(
+                        a0, DeviceArray
+                    ) and len(a0.shape) == 2:
+                        a0 = a0._device_transposed()
+                    r = quantile_rows_device_multi(a0, qs)
+                    if r is not None:
+                        return r
         a0 = args[0]
         if isinstance(a0, DeviceArray):
             return func(a0.materialize(), qv, **kwargs)

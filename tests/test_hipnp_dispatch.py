@@ -1463,3 +1463,22 @@ def test_digitize_on_device(fake):
     # decreasing bins fall back to host numpy
     r3 = np.digitize(_device(fake, host), bins[::-1].copy())
     np.testing.assert_array_equal(r3, np.digitize(host, bins[::-1]))
+
+
+def test_quantile_array_q_axis_on_device(fake):
+    host = np.random.default_rng(63).random((40, 90))
+    x = _device(fake, host)
+    qs = [0.0, 0.25, 0.5, 1.0]
+    r = np.quantile(x, qs, axis=1)
+    np.testing.assert_allclose(r, np.quantile(host, qs, axis=1), rtol=1e-12)
+    r0 = np.percentile(_device(fake, host), [10, 90], axis=0)
+    np.testing.assert_allclose(
+        r0, np.percentile(host, [10, 90], axis=0), rtol=1e-12)
+    # NaN rows poison their outputs at every q
+    h2 = host.copy()
+    h2[7, 3] = np.nan
+    r2 = np.quantile(_device(fake, h2), [0.5, 0.9], axis=1)
+    ref = np.quantile(h2, [0.5, 0.9], axis=1)
+    assert np.isnan(r2[:, 7]).all()
+    np.testing.assert_allclose(
+        r2[~np.isnan(ref)], ref[~np.isnan(ref)], rtol=1e-12)
