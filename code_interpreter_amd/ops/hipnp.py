@@ -1817,6 +1817,46 @@ def quantile_list_device(x, qs):
     return out
 
 
+def nanquantile_device(x, qs):
+    """Flat NaN-ignoring quantiles: one device sort (NaNs sort last),
+    quantile positions over the leading non-NaN run, two-element
+    download_slice per endpoint. Returns an array aligned with qs, or
+    None when not routable (incl. the all-NaN warning case)."""
+    if not isinstance(x, DeviceArray) or _dtype_code(x.dtype) is None:
+        return None
+    if x.size < 1:
+        return None
+    mask = x.isnan()
+    if not isinstance(mask, BoolDeviceArray):
+        return None
+    n_valid = x.size - int(mask.sum())
+    if n_valid == 0:
+        return None  # numpy warns and returns nan
+    if len(x.shape) == 1:
+        flat = x
+    else:
+        flat = x.ravel()
+        if not isinstance(flat, DeviceArray):
+            return None
+    srt = flat._device_sorted(False)
+    esz = x.dtype.itemsize
+    out = _np.empty(len(qs), dtype=x.dtype)
+    for i, q in enumerate(qs):
+        if not 0.0 <= float(q) <= 1.0:
+            return None
+        pos = float(q) * (n_valid - 1)
+        k0 = int(_np.floor(pos))
+        k1 = min(k0 + 1, n_valid - 1)
+        frac = pos - k0
+        raw = backend().download_slice(
+            srt._dev_handle(), k0 * esz, (k1 - k0 + 1) * esz
+        )
+        vals = _np.frombuffer(raw, dtype=x.dtype)
+        v0, v1 = vals[0], vals[-1]
+        out[i] = v0 if frac == 0 else v0 + (v1 - v0) * x.dtype.type(frac)
+    return out
+
+
 def quantile_cols_device(x, q):
     """Per-column quantile (axis=0): device transpose, then the row
     path."""
@@ -2859,6 +2899,49 @@ def _af_cleanup(func, args, kwargs):
             for v in args
         ]
         return func(*host, **kwargs)
+    if func in (_np.nanmedian, _np.nanquantile, _np.nanpercentile) \
+            and not kwargs and len(args) in (1, 2) and isinstance(
+        args[0], DeviceArray
+    ):
+        a = args[0]
+        qs = None
+        scalar_q = True
+        if func is _np.nanmedian and len(args) == 1:
+            qs = [0.5]
+        elif len(args) == 2:
+            qv = args[1]
+            scale = 100.0 if func is _np.nanpercentile else 1.0
+            if isinstance(qv, (int, float)):
+                qs = [float(qv) / scale]
+            elif isinstance(qv, (list, tuple, _np.ndarray)):
+                qs = (_np.asarray(qv, dtype=_np.float64).reshape(-1)
+                      / scale).tolist()
+                scalar_q = False
+        if qs:
+            r = nanquantile_device(a, qs)
+            if r is not None:
+                if scalar_q:
+                    return a.dtype.type(r[0])
+                return r.reshape(_np.asarray(args[1]).shape)
+        return func(a.materialize(), *args[1:], **kwargs)
+    if func in (_np.nanargmax, _np.nanargmin) and len(args) == 1 \
+            and isinstance(args[0], DeviceArray) and not kwargs:
+        a = args[0]
+        if _dtype_code(a.dtype) is not None:
+            mask = a.isnan()
+            if isinstance(mask, BoolDeviceArray):
+                n_nan = int(mask.sum())
+                if n_nan == 0:
+                    return (a.argmax() if func is _np.nanargmax
+                            else a.argmin())
+                if n_nan < a.size:
+                    repl = -_np.inf if func is _np.nanargmax else _np.inf
+                    cleaned = where_device(mask, repl, a)
+                    if cleaned is not NotImplemented:
+                        return (cleaned.argmax()
+                                if func is _np.nanargmax
+                                else cleaned.argmin())
+        return func(args[0].materialize())
     _nan_kinds = {
         _np.nansum: "sum", _np.nanmean: "mean", _np.nanmax: "max",
         _np.nanmin: "min", _np.nanstd: "std", _np.nanvar: "var",

@@ -1482,3 +1482,26 @@ def test_quantile_array_q_axis_on_device(fake):
     assert np.isnan(r2[:, 7]).all()
     np.testing.assert_allclose(
         r2[~np.isnan(ref)], ref[~np.isnan(ref)], rtol=1e-12)
+
+
+def test_nanquantile_family_on_device(fake):
+    host = np.random.default_rng(64).random(6000)
+    host[np.random.default_rng(65).integers(0, 6000, 300)] = np.nan
+    x = _device(fake, host)
+    fake.calls.clear()
+    assert float(np.nanmedian(x)) == pytest.approx(
+        np.nanmedian(host), abs=1e-12)
+    assert "sort" in fake.calls  # routed, not the materialize fallback
+    assert float(np.nanquantile(_device(fake, host), 0.9)) == pytest.approx(
+        np.nanquantile(host, 0.9), abs=1e-12)
+    r = np.nanpercentile(_device(fake, host), [10, 50, 95])
+    np.testing.assert_allclose(
+        r, np.nanpercentile(host, [10, 50, 95]), rtol=1e-12)
+    # 2-D flattens; nanarg* match numpy
+    m = host[:5980].reshape(46, 130)
+    assert float(np.nanmedian(_device(fake, m))) == pytest.approx(
+        np.nanmedian(m), abs=1e-12)
+    assert int(np.nanargmax(_device(fake, host))) == int(np.nanargmax(host))
+    assert int(np.nanargmin(_device(fake, host))) == int(np.nanargmin(host))
+    clean = np.random.default_rng(66).random(500)
+    assert int(np.nanargmax(_device(fake, clean))) == int(clean.argmax())
