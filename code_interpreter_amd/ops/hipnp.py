@@ -1457,7 +1457,24 @@ class BoolDeviceArray(DeviceArray):
     def sum(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return _np.intp(backend().mask_count(self._dev_handle(), self.size))
+        r = self._axis_count(axis, kwargs)
+        if r is not None:
+            return r
         return self.materialize().sum(axis=axis, **kwargs)
+
+    def _axis_count(self, axis, kwargs):
+        """Per-axis True counts ((x > 0).sum(axis=1)): mask -> 0/1 f64
+        via the select kernel, axis reduce, downloaded as numpy's int64
+        (the result is outer-size small)."""
+        if kwargs or self.size == 0:
+            return None
+        ones = where_device(self, 1.0, 0.0)
+        if ones is NotImplemented:
+            return None
+        r = ones._axis_reduce(_REDUCE_SUM, axis)
+        if r is None:
+            return None
+        return r.materialize().astype(_np.int64)
 
     def count_nonzero(self):
         return self.sum()
@@ -1465,16 +1482,27 @@ class BoolDeviceArray(DeviceArray):
     def mean(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return _np.float64(int(self.sum()) / self.size)
+        r = self._axis_count(axis, kwargs)
+        if r is not None:
+            red = self.shape[self._norm_axis(axis, len(self.shape))]
+            return r / float(red)
         return self.materialize().mean(axis=axis, **kwargs)
 
     def any(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return bool(int(self.sum()) > 0)
+        r = self._axis_count(axis, kwargs)
+        if r is not None:
+            return r > 0
         return self.materialize().any(axis=axis, **kwargs)
 
     def all(self, axis=None, **kwargs):
         if axis is None and not kwargs:
             return bool(int(self.sum()) == self.size)
+        r = self._axis_count(axis, kwargs)
+        if r is not None:
+            red = self.shape[self._norm_axis(axis, len(self.shape))]
+            return r == red
         return self.materialize().all(axis=axis, **kwargs)
 
     # f32/f64-only device paths must not see a u8 buffer

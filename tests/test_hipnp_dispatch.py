@@ -1505,3 +1505,18 @@ def test_nanquantile_family_on_device(fake):
     assert int(np.nanargmin(_device(fake, host))) == int(np.nanargmin(host))
     clean = np.random.default_rng(66).random(500)
     assert int(np.nanargmax(_device(fake, clean))) == int(clean.argmax())
+
+
+def test_mask_axis_reductions(fake):
+    host = np.random.default_rng(67).random((80, 120))
+    m = _device(fake, host) > 0.5
+    ref = host > 0.5
+    r = m.sum(axis=1)
+    assert r.dtype == np.int64
+    np.testing.assert_array_equal(r, ref.sum(axis=1))
+    np.testing.assert_array_equal(m.sum(axis=0), ref.sum(axis=0))
+    np.testing.assert_allclose(m.mean(axis=1), ref.mean(axis=1), rtol=1e-12)
+    np.testing.assert_array_equal(m.any(axis=1), ref.any(axis=1))
+    np.testing.assert_array_equal(m.all(axis=0), ref.all(axis=0))
+    # whole-mask paths unchanged
+    assert int(m.sum()) == int(ref.sum())
