@@ -361,15 +361,21 @@ def install_missing_deps(source: str) -> None:
 import re as _re
 
 _SHELL_LINE = _re.compile(r"^(\s*)!(?!=)(.+)$")
+# $(cmd) capture expressions (xonsh returns the command's stdout as a
+# string); conservative: no nested parens/newlines inside
+_SHELL_CAPTURE = _re.compile(r"\$\(([^()\n]+)\)")
 
 
 def transform_shell_escapes(source: str):
-    """Rewrite `<indent>!cmd` lines into `<indent>__ci_shell__('cmd')`.
-    Returns the transformed source, or None if no line matched. Only
-    applied when the ORIGINAL source fails to compile as python (a line
-    cannot start with `!` in valid python), so pure-python scripts --
-    including ones with multiline strings containing `!`-leading lines --
-    are never touched."""
+    """Rewrite xonsh-style shell escapes: `<indent>!cmd` lines into
+    `<indent>__ci_shell__('cmd')`, and `$(cmd)` expressions into
+    `__ci_shell_capture('cmd')` (returning the command's stdout like
+    xonsh). Returns the transformed source, or None if nothing matched.
+    Only applied when the ORIGINAL source fails to compile as python
+    (neither form is valid python syntax), so pure-python scripts --
+    including ones with strings containing `!` lines or `$(...)` -- are
+    never touched; and the transformed source must itself compile or
+    the user sees the original SyntaxError."""
     changed = False
     out = []
     for line in source.splitlines():
@@ -379,8 +385,29 @@ def transform_shell_escapes(source: str):
             out.append(f"{indent}__ci_shell__({cmd!r})")
             changed = True
         else:
-            out.append(line)
+            new_line, n = _SHELL_CAPTURE.subn(
+                lambda mm: f"__ci_shell_capture__({mm.group(1).strip()!r})",
+                line,
+            )
+            if n:
+                changed = True
+                out.append(new_line)
+            else:
+                out.append(line)
     return "\n".join(out) + "\n" if changed else None
+
+
+def _ci_shell_capture(cmd: str) -> str:
+    """$(cmd): run and return stdout as a string (xonsh capture
+    semantics; stderr passes through)."""
+    import subprocess
+
+    sys.stdout.flush()
+    sys.stderr.flush()
+    r = subprocess.run(cmd, shell=True, capture_output=True, text=True)
+    if r.stderr:
+        sys.stderr.write(r.stderr)
+    return r.stdout
 
 
 def _ci_shell(cmd: str) -> int:
@@ -453,6 +480,7 @@ def run_user_script(script_path: str) -> int:
         "__file__": script_path,
         "__builtins__": builtins,
         "__ci_shell__": _ci_shell,
+        "__ci_shell_capture__": _ci_shell_capture,
     }
     try:
         code = compile(source, script_path, "exec")

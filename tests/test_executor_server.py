@@ -294,3 +294,17 @@ def test_real_syntax_error_still_reported(raw_executor):
     body = r.json()
     assert body["exit_code"] == 1
     assert "SyntaxError" in body["stderr"]
+
+
+def test_shell_capture_expression(raw_executor):
+    src = (
+        'listing = $(echo alpha beta)\n'
+        'print("captured:", listing.strip())\n'
+        'n = len($(echo one two three).split())\n'
+        'print("words:", n)\n'
+    )
+    r = raw_executor.client.post("/execute", json={"source_code": src})
+    body = r.json()
+    assert body["exit_code"] == 0, body["stderr"]
+    assert "captured: alpha beta" in body["stdout"]
+    assert "words: 3" in body["stdout"]

@@ -146,3 +146,18 @@ def test_transform_shell_escapes_unit():
     assert rt.transform_shell_escapes("print(1)\n") is None
     # != operator lines must not match
     assert rt.transform_shell_escapes("a\n!= b\n") is None
+
+
+def test_transform_shell_capture_unit():
+    import importlib
+    rt = importlib.import_module("code_interpreter_amd.executor.sandbox_runtime")
+    out = rt.transform_shell_escapes('files = $(ls /tmp)\nprint(files)\n')
+    assert "__ci_shell_capture__('ls /tmp')" in out
+    # mixed with ! lines
+    out2 = rt.transform_shell_escapes('!mkdir -p /tmp/x\nv = $(echo hi)\n')
+    assert "__ci_shell__('mkdir -p /tmp/x')" in out2
+    assert "__ci_shell_capture__('echo hi')" in out2
+    # pure python with no escapes stays untouched
+    assert rt.transform_shell_escapes("x = f(1)\n") is None
+    # capture runner returns stdout
+    assert rt._ci_shell_capture("echo hello").strip() == "hello"
