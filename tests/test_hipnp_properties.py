@@ -1,0 +1,127 @@
+"""Property-based parity: random programs of routed numpy operations on
+a DeviceArray (fake numpy backend) must produce numpy's own values.
+Complements the example-based dispatch tests by searching the edge
+space (shapes, axes, NaNs, duplicates) automatically."""
+
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+OPS_DIR = Path(__file__).resolve().parent.parent / "code_interpreter_amd" / "ops"
+sys.path.insert(0, str(OPS_DIR))
+
+import hipnp  # noqa: E402
+
+from test_hipnp_dispatch import FakeBackend  # noqa: E402
+
+
+@pytest.fixture
+def fake(monkeypatch):
+    backend = FakeBackend()
+    monkeypatch.setitem(hipnp._state, "backend", backend)
+    monkeypatch.setitem(hipnp._state, "failed", None)
+    return backend
+
+
+def _dev(fake, arr):
+    arr = np.ascontiguousarray(arr)
+    h = fake._new(arr.copy())
+    return hipnp.DeviceArray(h, arr.shape, arr.dtype)
+
+
+arrays_1d = st.builds(
+    lambda seed, n, with_nan: _mk(seed, (n,), with_nan),
+    st.integers(0, 2**31 - 1), st.integers(1, 400), st.booleans(),
+)
+arrays_2d = st.builds(
+    lambda seed, r, c, with_nan: _mk(seed, (r, c), with_nan),
+    st.integers(0, 2**31 - 1), st.integers(1, 40), st.integers(1, 40),
+    st.booleans(),
+)
+
+
+def _mk(seed, shape, with_nan):
+    rng = np.random.default_rng(seed)
+    # duplicate-heavy integers exercise sort stability and ties
+    a = rng.integers(-5, 6, shape).astype(np.float64)
+    if with_nan:
+        m = rng.random(shape) < 0.15
+        a[m] = np.nan
+    return a
+
+
+@settings(max_examples=60, deadline=None)
+@given(a=arrays_1d)
+def test_sort_argsort_parity(a):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        np.testing.assert_array_equal(
+            np.asarray(np.sort(_dev(fake, a))), np.sort(a))
+        np.testing.assert_array_equal(
+            np.asarray(np.argsort(_dev(fake, a))),
+            np.argsort(a, kind="stable"))
+        u = np.unique(_dev(fake, a))
+        np.testing.assert_array_equal(np.asarray(u), np.unique(a))
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=60, deadline=None)
+@given(a=arrays_2d, axis=st.sampled_from([0, 1, -1]))
+def test_axis_sort_cumsum_parity(a, axis):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        np.testing.assert_array_equal(
+            np.asarray(np.sort(_dev(fake, a), axis=axis)),
+            np.sort(a, axis=axis))
+        np.testing.assert_allclose(
+            np.asarray(np.cumsum(_dev(fake, a), axis=axis)),
+            np.cumsum(a, axis=axis), rtol=1e-12)
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=60, deadline=None)
+@given(a=arrays_1d, q=st.floats(0.0, 1.0))
+def test_quantile_parity(a, q):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        got = np.quantile(_dev(fake, a), q)
+        ref = np.quantile(a, q)
+        if np.isnan(ref):
+            assert np.isnan(got)
+        else:
+            assert got == pytest.approx(ref, abs=1e-12)
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=40, deadline=None)
+@given(a=arrays_2d)
+def test_histogram_and_mask_parity(a):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        if np.isnan(a).any():
+            # numpy raises on NaN autorange; the device route must too
+            with pytest.raises(ValueError):
+                np.histogram(_dev(fake, a), bins=7)
+        else:
+            hist, edges = np.histogram(_dev(fake, a), bins=7)
+            rh, re = np.histogram(a, bins=7)
+            np.testing.assert_array_equal(hist, rh)
+            np.testing.assert_allclose(edges, re, rtol=0)
+        with np.errstate(invalid="ignore"):
+            m = _dev(fake, a) > 0
+            ref = a > 0
+            np.testing.assert_array_equal(np.asarray(m.sum(axis=1)),
+                                          ref.sum(axis=1))
+            assert int(m.sum()) == int(ref.sum())
+    finally:
+        hipnp._state["backend"] = None
