@@ -125,3 +125,64 @@ def test_histogram_and_mask_parity(a):
             assert int(m.sum()) == int(ref.sum())
     finally:
         hipnp._state["backend"] = None
+
+
+@settings(max_examples=50, deadline=None)
+@given(a=arrays_1d, b=arrays_1d, side=st.sampled_from(["left", "right"]))
+def test_searchsorted_digitize_parity(a, b, side):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        srt = np.sort(a)
+        got = np.searchsorted(_dev(fake, srt), _dev(fake, b), side=side)
+        np.testing.assert_array_equal(
+            np.asarray(got), np.searchsorted(srt, b, side=side))
+        bins = np.unique(a[~np.isnan(a)])
+        if bins.size:
+            got2 = np.digitize(_dev(fake, b), bins)
+            np.testing.assert_array_equal(
+                np.asarray(got2), np.digitize(b, bins))
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=50, deadline=None)
+@given(a=arrays_2d, b=arrays_2d)
+def test_concat_reshape_parity(a, b):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        if a.shape[1] == b.shape[1]:
+            got = np.vstack([_dev(fake, a), _dev(fake, b)])
+            np.testing.assert_array_equal(
+                np.asarray(got), np.vstack([a, b]))
+        flat = _dev(fake, a).ravel()
+        np.testing.assert_array_equal(np.asarray(flat), a.ravel())
+        r = _dev(fake, a).reshape(-1, a.shape[0])
+        np.testing.assert_array_equal(
+            np.asarray(r), a.reshape(-1, a.shape[0]))
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=50, deadline=None)
+@given(a=arrays_1d)
+def test_nan_reduction_parity(a):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        import warnings
+        with warnings.catch_warnings(), np.errstate(all="ignore"):
+            warnings.simplefilter("ignore")
+            for dev_f, ref_f in (
+                (np.nansum, np.nansum), (np.nanmean, np.nanmean),
+                (np.nanmedian, np.nanmedian),
+            ):
+                got = dev_f(_dev(fake, a))
+                ref = ref_f(a)
+                if np.isnan(ref):
+                    assert np.isnan(got)
+                else:
+                    assert float(got) == pytest.approx(ref, abs=1e-10)
+    finally:
+        hipnp._state["backend"] = None
