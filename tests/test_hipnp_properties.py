@@ -186,3 +186,41 @@ def test_nan_reduction_parity(a):
                     assert float(got) == pytest.approx(ref, abs=1e-10)
     finally:
         hipnp._state["backend"] = None
+
+
+@settings(max_examples=50, deadline=None)
+@given(a=arrays_1d, rtol=st.sampled_from([1e-5, 1e-9, 0.5]))
+def test_isclose_allclose_parity(a, rtol):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        b = a + np.where(np.isnan(a), 0.0, 1e-7)
+        with np.errstate(invalid="ignore"):
+            got = np.isclose(_dev(fake, a), _dev(fake, b), rtol=rtol)
+            ref = np.isclose(a, b, rtol=rtol)
+            np.testing.assert_array_equal(np.asarray(got), ref)
+            assert bool(np.allclose(_dev(fake, a), _dev(fake, b),
+                                    rtol=rtol)) == bool(
+                np.allclose(a, b, rtol=rtol))
+    finally:
+        hipnp._state["backend"] = None
+
+
+@settings(max_examples=40, deadline=None)
+@given(a=arrays_2d)
+def test_median_quantile_axis_parity(a):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        import warnings
+        with warnings.catch_warnings(), np.errstate(all="ignore"):
+            warnings.simplefilter("ignore")
+            for axis in (0, 1):
+                got = np.median(_dev(fake, a), axis=axis)
+                ref = np.median(a, axis=axis)
+                nans = np.isnan(ref)
+                assert np.array_equal(np.isnan(np.asarray(got)), nans)
+                np.testing.assert_allclose(
+                    np.asarray(got)[~nans], ref[~nans], rtol=1e-12)
+    finally:
+        hipnp._state["backend"] = None
