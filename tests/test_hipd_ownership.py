@@ -80,6 +80,12 @@ class FakeHipops(types.ModuleType):
     def gemm_batched(self, ha, hb, batch, m, n, k, dtype):
         return self._new(b"\0" * (batch * m * n * 8))
 
+    def cumsum(self, h, dtype, n):
+        return self._new(self.bufs[h])
+
+    def copy_d2d(self, hd, doff, hs, soff, nbytes):
+        pass
+
     def sort(self, h, dtype, n, want_idx):
         if want_idx:
             return self._new(self.bufs[h]), self._new(b"\0" * (n * 8))
