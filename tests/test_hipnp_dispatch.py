@@ -1520,3 +1520,34 @@ def test_mask_axis_reductions(fake):
     np.testing.assert_array_equal(m.all(axis=0), ref.all(axis=0))
     # whole-mask paths unchanged
     assert int(m.sum()) == int(ref.sum())
+
+
+def test_install_idempotent(fake, monkeypatch):
+    import types
+    names = ("sum", "square", "matmul", "dot", "sqrt", "exp", "log", "sin",
+             "cos", "tanh", "absolute", "abs", "sort", "argsort", "median",
+             "mean", "std", "var", "max", "amax", "min", "amin")
+    np_mod = types.SimpleNamespace(**{n: getattr(np, n) for n in names})
+    np_mod.random = types.SimpleNamespace(**{
+        n: getattr(np.random, n) for n in (
+            "rand", "random", "random_sample", "uniform", "randn",
+            "standard_normal", "normal")
+    })
+    monkeypatch.setitem(hipnp._installed, "done", False)
+    monkeypatch.setattr(hipnp, "MIN_ELEMS", 64)
+    monkeypatch.setattr(hipnp, "available", lambda: True)
+    hipnp.install(np_mod, mode="auto")
+    first = np_mod.sort
+    hipnp.install(np_mod, mode="auto")  # second call must be a no-op
+    assert np_mod.sort is first  # not double-wrapped
+
+
+def test_device_array_gc_frees_backend_handle(fake):
+    host = np.random.default_rng(68).random(256)
+    x = _device(fake, host)
+    h = x._handle
+    assert h in fake.bufs
+    del x
+    import gc
+    gc.collect()
+    assert h not in fake.bufs  # freed through the backend
