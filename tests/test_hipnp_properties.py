@@ -224,3 +224,29 @@ def test_median_quantile_axis_parity(a):
                     np.asarray(got)[~nans], ref[~nans], rtol=1e-12)
     finally:
         hipnp._state["backend"] = None
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    seed=st.integers(0, 2**31 - 1),
+    m=st.integers(1, 24), k=st.integers(1, 24), n=st.integers(1, 24),
+)
+def test_matmul_shapes_parity(seed, m, k, n):
+    fake = FakeBackend()
+    hipnp._state["backend"] = fake
+    try:
+        rng = np.random.default_rng(seed)
+        A = rng.standard_normal((m, k))
+        B = rng.standard_normal((k, n))
+        x = rng.standard_normal(k)
+        got = np.asarray(_dev(fake, A) @ _dev(fake, B))
+        np.testing.assert_allclose(got, A @ B, rtol=1e-10, atol=1e-12)
+        got_mv = np.asarray(_dev(fake, A) @ _dev(fake, x))
+        np.testing.assert_allclose(got_mv, A @ x, rtol=1e-10, atol=1e-12)
+        got_vm = np.asarray(_dev(fake, x) @ _dev(fake, B))
+        np.testing.assert_allclose(got_vm, x @ B, rtol=1e-10, atol=1e-12)
+        got_e = np.asarray(np.einsum("ij,jk->ik", _dev(fake, A),
+                                     _dev(fake, B)))
+        np.testing.assert_allclose(got_e, A @ B, rtol=1e-10, atol=1e-12)
+    finally:
+        hipnp._state["backend"] = None
